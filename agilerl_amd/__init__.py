@@ -1,0 +1,79 @@
+"""agilerl-amd: MI355X-native evolutionary-HPO RL training framework.
+
+A from-scratch framework with the capabilities, public API shape and
+checkpoint format of AgileRL, built for one 8xMI355X node: PyTorch-ROCm +
+hand-written CDNA4 HIP kernels for the hot numeric paths + RCCL
+collectives over the xGMI mesh (one population agent per GPU).  No
+Accelerate/DeepSpeed/Triton/vLLM dependencies.
+
+Public names mirror the reference (``agilerl/__init__.py:74-79``):
+``LocalTrainer`` and the capability flags; algorithm classes live under
+``agilerl_amd.algorithms``.
+"""
+
+from __future__ import annotations
+
+import importlib
+from enum import Enum
+from typing import TYPE_CHECKING
+
+__version__ = "0.1.0"
+
+# ---------------------------------------------------------------------------
+# Capability flags (reference parity: HAS_* in agilerl/__init__.py:56-60).
+# ---------------------------------------------------------------------------
+
+
+def _importable(name: str) -> bool:
+    try:
+        importlib.import_module(name)
+        return True
+    except ImportError:
+        return False
+
+
+HAS_LLM_DEPENDENCIES = _importable("transformers")
+HAS_HIP_KERNELS = None  # resolved lazily below
+
+
+class AgentType(str, Enum):
+    RL = "rl"
+    MULTI_AGENT_RL = "multi_agent_rl"
+    BANDIT = "bandit"
+    OFFLINE = "offline"
+    LLM = "llm"
+
+
+_LAZY = {
+    "LocalTrainer": "agilerl_amd.training.trainer",
+    "Trainer": "agilerl_amd.training.trainer",
+    "Population": "agilerl_amd.population",
+    "TournamentSelection": "agilerl_amd.hpo.tournament",
+    "Mutations": "agilerl_amd.hpo.mutation",
+    "make_vect_envs": "agilerl_amd.envs.registry",
+}
+
+
+def __getattr__(name: str):
+    if name == "HAS_HIP_KERNELS":
+        from .ops import has_extension
+
+        return has_extension()
+    if name in _LAZY:
+        module = importlib.import_module(_LAZY[name])
+        return getattr(module, name)
+    raise AttributeError(f"module 'agilerl_amd' has no attribute '{name}'")
+
+
+__all__ = [
+    "__version__",
+    "AgentType",
+    "HAS_LLM_DEPENDENCIES",
+    "HAS_HIP_KERNELS",
+    "LocalTrainer",
+    "Trainer",
+    "Population",
+    "TournamentSelection",
+    "Mutations",
+    "make_vect_envs",
+]

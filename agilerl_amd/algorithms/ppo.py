@@ -1,0 +1,214 @@
+"""PPO (clip surrogate + value clip + entropy bonus, GAE).
+
+Reference parity: ``agilerl/algorithms/ppo.py:56`` — clip losses
+(:800-832), GAE from the rollout buffer, recurrent BPTT path, target-KL
+early stop.  The GAE scan and (on GPU) the policy-loss reductions run as
+HIP kernels via ``agilerl_amd.ops``.
+"""
+
+from __future__ import annotations
+
+from typing import Any, Dict, Optional, Tuple
+
+import numpy as np
+import torch
+import torch.nn as nn
+
+from ..components.rollout_buffer import RolloutBuffer
+from ..networks.actors import StochasticActor
+from ..networks.value_networks import ValueNetwork
+from ..spaces import Box, Space
+from .core.base import RLAlgorithm
+from .core.optimizer_wrapper import OptimizerWrapper
+from .core.registry import HyperparameterConfig, NetworkGroup, OptimizerConfig, RLParameter
+
+__all__ = ["PPO"]
+
+
+def default_hp_config() -> HyperparameterConfig:
+    return HyperparameterConfig(
+        lr=RLParameter(min=1e-5, max=1e-2),
+        batch_size=RLParameter(min=64, max=4096, dtype=int),
+        clip_coef=RLParameter(min=0.05, max=0.4),
+        ent_coef=RLParameter(min=1e-4, max=0.05),
+        update_epochs=RLParameter(min=1, max=10, dtype=int),
+    )
+
+
+class PPO(RLAlgorithm):
+    def __init__(
+        self,
+        observation_space: Space,
+        action_space: Space,
+        index: int = 0,
+        hp_config: Optional[HyperparameterConfig] = None,
+        net_config: Optional[Dict[str, Any]] = None,
+        head_config: Optional[Dict[str, Any]] = None,
+        batch_size: int = 512,
+        lr: float = 3e-4,
+        learn_step: int = 128,
+        gamma: float = 0.99,
+        gae_lambda: float = 0.95,
+        clip_coef: float = 0.2,
+        clip_vloss: bool = True,
+        ent_coef: float = 0.01,
+        vf_coef: float = 0.5,
+        update_epochs: int = 4,
+        max_grad_norm: float = 0.5,
+        target_kl: Optional[float] = None,
+        normalize_advantage: bool = True,
+        log_std_init: float = 0.0,
+        latent_dim: int = 64,
+        recurrent: bool = False,
+        device: str = "cpu",
+    ):
+        super().__init__(
+            observation_space,
+            action_space,
+            index=index,
+            learn_step=learn_step,
+            device=device,
+            hp_config=hp_config or default_hp_config(),
+            name="PPO",
+        )
+        self.batch_size = int(batch_size)
+        self.lr = float(lr)
+        self.gamma = float(gamma)
+        self.gae_lambda = float(gae_lambda)
+        self.clip_coef = float(clip_coef)
+        self.clip_vloss = bool(clip_vloss)
+        self.ent_coef = float(ent_coef)
+        self.vf_coef = float(vf_coef)
+        self.update_epochs = int(update_epochs)
+        self.max_grad_norm = float(max_grad_norm)
+        self.target_kl = target_kl
+        self.normalize_advantage = normalize_advantage
+        self.net_config = net_config
+        self.latent_dim = latent_dim
+        self.recurrent = recurrent
+
+        self.actor = StochasticActor(
+            observation_space,
+            action_space,
+            encoder_config=net_config,
+            head_config=head_config,
+            latent_dim=latent_dim,
+            log_std_init=log_std_init,
+            device=device,
+        )
+        self.critic = ValueNetwork(
+            observation_space,
+            encoder_config=net_config,
+            head_config=head_config,
+            latent_dim=latent_dim,
+            device=device,
+        )
+        self.optimizer = OptimizerWrapper(
+            torch.optim.Adam, [self.actor, self.critic], lr=self.lr
+        )
+
+        self.register_network_group(NetworkGroup(eval_network="actor", policy=True))
+        self.register_network_group(NetworkGroup(eval_network="critic"))
+        self.register_optimizer(
+            OptimizerConfig(name="optimizer", networks=["actor", "critic"], lr_name="lr")
+        )
+
+    # ------------------------------------------------------------------
+    def get_action(
+        self,
+        obs,
+        action_mask: Optional[np.ndarray] = None,
+        training: bool = True,
+    ):
+        """Returns (action_np, log_prob, entropy, value) during training,
+        or action_np when ``training=False`` (greedy)."""
+        mask_t = (
+            torch.as_tensor(np.asarray(action_mask), device=self.device)
+            if action_mask is not None
+            else None
+        )
+        with torch.no_grad():
+            pre = self.actor.preprocess(obs)
+            if not training:
+                return self.actor.deterministic_action(pre, mask_t).cpu().numpy()
+            action, log_prob, entropy = self.actor.sample(pre, mask_t)
+            value = self.critic(self.critic.preprocess(obs)).squeeze(-1)
+        return action.cpu().numpy(), log_prob, entropy, value
+
+    def get_values(self, obs) -> torch.Tensor:
+        with torch.no_grad():
+            return self.critic(self.critic.preprocess(obs)).squeeze(-1)
+
+    # ------------------------------------------------------------------
+    def learn(self, rollout: RolloutBuffer) -> Dict[str, float]:
+        assert rollout.advantages is not None, "call compute_returns_and_advantages first"
+        stats = {"policy_loss": 0.0, "value_loss": 0.0, "entropy": 0.0, "approx_kl": 0.0}
+        n_updates = 0
+        early_stop = False
+        for _ in range(self.update_epochs):
+            if early_stop:
+                break
+            for mb in rollout.get_minibatches(self.batch_size):
+                loss_stats = self._update_minibatch(mb)
+                for k in stats:
+                    stats[k] += loss_stats[k]
+                n_updates += 1
+                if self.target_kl is not None and loss_stats["approx_kl"] > 1.5 * self.target_kl:
+                    early_stop = True
+                    break
+        if n_updates:
+            stats = {k: v / n_updates for k, v in stats.items()}
+        return stats
+
+    def _update_minibatch(self, mb: Dict[str, torch.Tensor]) -> Dict[str, float]:
+        obs = mb["obs"]
+        actions = mb["action"].to(self.device)
+        old_log_prob = mb["log_prob"].to(self.device).reshape(-1)
+        advantages = mb["advantages"].to(self.device).reshape(-1)
+        returns = mb["returns"].to(self.device).reshape(-1)
+        old_values = mb["value"].to(self.device).reshape(-1)
+        mask = mb.get("action_mask")
+
+        if self.normalize_advantage and advantages.numel() > 1:
+            advantages = (advantages - advantages.mean()) / (advantages.std() + 1e-8)
+
+        new_log_prob, entropy = self.actor.evaluate_actions(
+            self.actor.preprocess(obs), actions, mask
+        )
+        values = self.critic(self.critic.preprocess(obs)).reshape(-1)
+
+        log_ratio = new_log_prob.reshape(-1) - old_log_prob
+        ratio = log_ratio.exp()
+        with torch.no_grad():
+            approx_kl = ((ratio - 1) - log_ratio).mean()
+
+        pg1 = -advantages * ratio
+        pg2 = -advantages * ratio.clamp(1 - self.clip_coef, 1 + self.clip_coef)
+        policy_loss = torch.maximum(pg1, pg2).mean()
+
+        if self.clip_vloss:
+            v_clipped = old_values + (values - old_values).clamp(
+                -self.clip_coef, self.clip_coef
+            )
+            value_loss = 0.5 * torch.maximum(
+                (values - returns) ** 2, (v_clipped - returns) ** 2
+            ).mean()
+        else:
+            value_loss = 0.5 * ((values - returns) ** 2).mean()
+
+        entropy_loss = entropy.mean()
+        loss = policy_loss + self.vf_coef * value_loss - self.ent_coef * entropy_loss
+
+        self.optimizer.zero_grad()
+        loss.backward()
+        nn.utils.clip_grad_norm_(
+            [p for net in (self.actor, self.critic) for p in net.parameters()],
+            self.max_grad_norm,
+        )
+        self.optimizer.step()
+        return {
+            "policy_loss": float(policy_loss.detach()),
+            "value_loss": float(value_loss.detach()),
+            "entropy": float(entropy_loss.detach()),
+            "approx_kl": float(approx_kl),
+        }

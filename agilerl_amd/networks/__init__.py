@@ -1,0 +1,18 @@
+from .base import EvolvableNetwork, build_encoder, get_default_encoder_config
+from .distributions import ActionDistribution
+from .q_networks import QNetwork, RainbowQNetwork, ContinuousQNetwork
+from .actors import DeterministicActor, StochasticActor
+from .value_networks import ValueNetwork
+
+__all__ = [
+    "EvolvableNetwork",
+    "build_encoder",
+    "get_default_encoder_config",
+    "ActionDistribution",
+    "QNetwork",
+    "RainbowQNetwork",
+    "ContinuousQNetwork",
+    "DeterministicActor",
+    "StochasticActor",
+    "ValueNetwork",
+]

@@ -1,0 +1,246 @@
+"""Evolvable network = space-selected encoder + latent + MLP head.
+
+Reference parity: ``agilerl/networks/base.py:167`` (EvolvableNetwork,
+``_build_encoder`` :604, latent mutations ``add_latent_node`` :555 /
+``remove_latent_node`` :573).  New design: both the encoder and the head are
+themselves :class:`EvolvableModule` instances, and the network exposes a
+*namespaced* mutation surface (``encoder.add_node``, ``head.add_layer``,
+``add_latent_node``...) so a sampled mutation can be replayed verbatim on
+every network of the same group (target nets, shared encoders).
+"""
+
+from __future__ import annotations
+
+from typing import Any, Dict, List, Optional, Union
+
+import numpy as np
+import torch
+
+from ..modules.base import EvolvableModule, MutationType, mutation
+from ..modules.cnn import EvolvableCNN
+from ..modules.lstm import EvolvableLSTM
+from ..modules.mlp import EvolvableMLP
+from ..modules.multi_input import EvolvableMultiInput
+from ..modules.simba import EvolvableSimBa
+from ..spaces import Box, DictSpace, Discrete, MultiBinary, MultiDiscrete, Space, TupleSpace, flatdim, is_image_space
+
+__all__ = ["EvolvableNetwork", "build_encoder", "get_default_encoder_config"]
+
+
+def get_default_encoder_config(observation_space: Space, simba: bool = False) -> Dict[str, Any]:
+    """Default encoder config for an observation space.
+
+    Parity with reference ``utils/evolvable_networks.py:get_default_encoder_config``.
+    """
+    if isinstance(observation_space, (DictSpace, TupleSpace)):
+        return {"arch": "multi_input", "latent_dim": 64}
+    if is_image_space(observation_space):
+        return {
+            "arch": "cnn",
+            "channel_size": [32, 64, 64],
+            "kernel_size": [8, 4, 3],
+            "stride_size": [4, 2, 1],
+        }
+    if simba:
+        return {"arch": "simba", "hidden_size": 128, "num_blocks": 2}
+    return {"arch": "mlp", "hidden_size": [64, 64]}
+
+
+def build_encoder(
+    observation_space: Space,
+    latent_dim: int,
+    encoder_config: Optional[Dict[str, Any]] = None,
+    device: str = "cpu",
+) -> EvolvableModule:
+    """Architecture selection per observation space."""
+    cfg = dict(encoder_config or get_default_encoder_config(observation_space))
+    arch = cfg.pop("arch", None)
+    cfg.pop("latent_dim", None)
+    if arch is None:
+        arch = get_default_encoder_config(observation_space)["arch"]
+
+    if isinstance(observation_space, (DictSpace, TupleSpace)) or arch == "multi_input":
+        return EvolvableMultiInput(
+            observation_space=observation_space,
+            num_outputs=latent_dim,
+            device=device,
+            **{k: v for k, v in cfg.items() if k in ("mlp_config", "cnn_config")},
+        )
+    if arch == "cnn" or is_image_space(observation_space):
+        return EvolvableCNN(
+            input_shape=observation_space.shape, num_outputs=latent_dim, device=device, **cfg
+        )
+    num_inputs = flatdim(observation_space)
+    if arch == "simba":
+        return EvolvableSimBa(num_inputs=num_inputs, num_outputs=latent_dim, device=device, **cfg)
+    if arch == "lstm":
+        return EvolvableLSTM(input_size=num_inputs, num_outputs=latent_dim, device=device, **cfg)
+    return EvolvableMLP(num_inputs=num_inputs, num_outputs=latent_dim, device=device, **cfg)
+
+
+class EvolvableNetwork(EvolvableModule):
+    """Encoder -> latent -> head network with a namespaced mutation surface."""
+
+    MIN_LATENT = 8
+    MAX_LATENT = 512
+
+    def __init__(
+        self,
+        observation_space: Space,
+        num_outputs: int,
+        encoder_config: Optional[Dict[str, Any]] = None,
+        head_config: Optional[Dict[str, Any]] = None,
+        latent_dim: int = 64,
+        device: str = "cpu",
+    ):
+        super().__init__(device)
+        self.observation_space = observation_space
+        self.num_outputs = int(num_outputs)
+        self.latent_dim = int(latent_dim)
+        self.encoder_config = dict(encoder_config) if encoder_config else None
+        self.head_config = dict(head_config) if head_config else None
+
+        self.encoder = build_encoder(observation_space, self.latent_dim, self.encoder_config, device)
+        self.head_net = self._build_head()
+
+    # ------------------------------------------------------------------
+    def _build_head(self) -> EvolvableModule:
+        cfg = dict(self.head_config or {"hidden_size": [64]})
+        cfg.pop("arch", None)
+        return EvolvableMLP(
+            num_inputs=self.latent_dim, num_outputs=self.num_outputs, device=self.device, **cfg
+        )
+
+    def extract_features(self, obs) -> torch.Tensor:
+        obs = self.preprocess(obs)
+        return self.encoder(obs)
+
+    def preprocess(self, obs):
+        """Space-aware preprocessing (one-hot for Discrete etc.)."""
+        space = self.observation_space
+        if isinstance(space, (DictSpace, TupleSpace)):
+            return obs
+        if not isinstance(obs, torch.Tensor):
+            obs = torch.as_tensor(np.asarray(obs))
+        obs = obs.to(self.device if isinstance(self.device, torch.device) else torch.device(self.device))
+        if isinstance(space, Discrete):
+            if obs.dim() == 0:
+                obs = obs.unsqueeze(0)
+            if obs.dim() == 1 and obs.dtype in (torch.int32, torch.int64):
+                obs = torch.nn.functional.one_hot(obs.long(), space.n).float()
+            elif obs.shape[-1] != space.n:
+                obs = torch.nn.functional.one_hot(obs.long().reshape(-1), space.n).float()
+        elif isinstance(space, MultiDiscrete):
+            if obs.shape[-1] == len(space.nvec) and obs.dtype in (torch.int32, torch.int64):
+                hots = [
+                    torch.nn.functional.one_hot(obs[..., i].long(), int(n)).float()
+                    for i, n in enumerate(space.nvec)
+                ]
+                obs = torch.cat(hots, dim=-1)
+        elif isinstance(space, MultiBinary):
+            obs = obs.float()
+        else:
+            obs = obs.float()
+            if obs.dim() == len(space.shape):
+                obs = obs.unsqueeze(0)
+        return obs
+
+    def forward(self, obs) -> torch.Tensor:
+        return self.head_net(self.extract_features(obs))
+
+    def reset_noise(self) -> None:
+        self.encoder.reset_noise()
+        self.head_net.reset_noise()
+
+    # ------------------------------------------------------------------
+    # Namespaced mutation surface
+    # ------------------------------------------------------------------
+    @property
+    def mutation_methods(self) -> List[str]:
+        methods = ["add_latent_node", "remove_latent_node"]
+        methods += [f"encoder.{m}" for m in self.encoder.mutation_methods]
+        methods += [f"head.{m}" for m in self.head_net.mutation_methods]
+        return methods
+
+    def get_mutation_methods(self) -> Dict[str, MutationType]:
+        out = {"add_latent_node": MutationType.NODE, "remove_latent_node": MutationType.NODE}
+        for m in self.encoder.mutation_methods:
+            out[f"encoder.{m}"] = getattr(type(self.encoder), m)._mutation_type
+        for m in self.head_net.mutation_methods:
+            out[f"head.{m}"] = getattr(type(self.head_net), m)._mutation_type
+        return out
+
+    def apply_mutation(self, name: str, **choices) -> Optional[dict]:
+        if name.startswith("encoder."):
+            result = self.encoder.apply_mutation(name[len("encoder.") :], **choices)
+        elif name.startswith("head."):
+            result = self.head_net.apply_mutation(name[len("head.") :], **choices)
+        else:
+            result = super().apply_mutation(name, **choices)
+        self._last_mutation = (name, result if isinstance(result, dict) else {})
+        return result
+
+    def _resize_latent(self, new_dim: int) -> None:
+        new_dim = int(np.clip(new_dim, self.MIN_LATENT, self.MAX_LATENT))
+        if new_dim == self.latent_dim:
+            return
+        self.latent_dim = new_dim
+        # encoder output width
+        self.encoder.num_outputs = new_dim
+        self.encoder.recreate_network()
+        # head input width
+        self.head_net.num_inputs = new_dim
+        self.head_net.recreate_network()
+
+    @mutation(MutationType.NODE)
+    def add_latent_node(self, numb_new_nodes: Optional[int] = None) -> dict:
+        if numb_new_nodes is None:
+            numb_new_nodes = int(np.random.choice([8, 16, 32]))
+        self._resize_latent(self.latent_dim + numb_new_nodes)
+        return {"numb_new_nodes": numb_new_nodes}
+
+    @mutation(MutationType.NODE)
+    def remove_latent_node(self, numb_new_nodes: Optional[int] = None) -> dict:
+        if numb_new_nodes is None:
+            numb_new_nodes = int(np.random.choice([8, 16, 32]))
+        self._resize_latent(self.latent_dim - numb_new_nodes)
+        return {"numb_new_nodes": numb_new_nodes}
+
+    # ------------------------------------------------------------------
+    # Rebuild support: init_dict must reflect the LIVE (possibly mutated)
+    # architecture, not the constructor-time config, so clone()/checkpoint
+    # round-trips reproduce mutated offspring exactly.
+    # ------------------------------------------------------------------
+    _ARCH_NAMES = {
+        EvolvableMLP: "mlp",
+        EvolvableCNN: "cnn",
+        EvolvableSimBa: "simba",
+        EvolvableLSTM: "lstm",
+        EvolvableMultiInput: "multi_input",
+    }
+
+    def _live_encoder_config(self) -> Dict[str, Any]:
+        cfg = self.encoder.init_dict
+        cfg.pop("device", None)
+        cfg.pop("num_inputs", None)
+        cfg.pop("num_outputs", None)
+        cfg.pop("input_shape", None)
+        cfg.pop("input_size", None)
+        cfg.pop("observation_space", None)
+        cfg["arch"] = self._ARCH_NAMES.get(type(self.encoder), "mlp")
+        return cfg
+
+    def _live_head_config(self) -> Dict[str, Any]:
+        cfg = self.head_net.init_dict
+        cfg.pop("device", None)
+        cfg.pop("num_inputs", None)
+        cfg.pop("num_outputs", None)
+        return cfg
+
+    @property
+    def init_dict(self) -> Dict[str, Any]:
+        base = super().init_dict
+        base["encoder_config"] = self._live_encoder_config()
+        base["head_config"] = self._live_head_config()
+        base["latent_dim"] = self.latent_dim
+        return base

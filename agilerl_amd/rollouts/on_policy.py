@@ -1,0 +1,75 @@
+"""Vectorized on-policy rollout collection.
+
+Reference parity: ``agilerl/rollouts/on_policy.py`` (collect_rollouts :221,
+collect_rollouts_recurrent :243; shared ``_collect_rollouts`` :29).
+
+The buffer lives on the agent's device; env stepping stays on host numpy
+and observations stream to HBM batched (pinned staging handled by the
+buffer).  Truncation bootstrapping uses ``info["final_observation"]``.
+"""
+
+from __future__ import annotations
+
+from typing import Any, Dict, Optional, Tuple
+
+import numpy as np
+import torch
+
+from ..components.rollout_buffer import RolloutBuffer
+
+__all__ = ["collect_rollouts", "collect_rollouts_recurrent"]
+
+
+def collect_rollouts(
+    agent,
+    env,
+    buffer: RolloutBuffer,
+    n_steps: int,
+    obs: Optional[np.ndarray] = None,
+    done: Optional[np.ndarray] = None,
+) -> Tuple[np.ndarray, np.ndarray, Dict[str, float]]:
+    """Collect ``n_steps`` vectorized steps into ``buffer``; computes GAE.
+
+    Returns (last_obs, last_done, info_stats).
+    """
+    if obs is None:
+        obs, _ = env.reset()
+        done = np.zeros(env.num_envs, dtype=bool)
+    ep_returns: list = []
+    buffer.reset()
+    for _ in range(n_steps):
+        action, log_prob, _entropy, value = agent.get_action(obs, training=True)
+        next_obs, reward, term, trunc, info = env.step(action)
+        # bootstrap through truncation: add V(final_obs) to the reward
+        if np.any(trunc) and "final_observation" in info:
+            with torch.no_grad():
+                v_final = agent.get_values(info["final_observation"]).cpu().numpy()
+            reward = np.where(trunc, reward + agent.gamma * v_final, reward)
+        buffer.add(
+            obs=obs,
+            action=action,
+            reward=reward,
+            done=(term | trunc).astype(np.float32),
+            value=value,
+            log_prob=log_prob,
+        )
+        obs = next_obs
+        done = term | trunc
+        if "episode_return" in info:
+            ep_returns.extend(np.asarray(info["episode_return"]).tolist())
+    last_value = agent.get_values(obs)
+    buffer.compute_returns_and_advantages(last_value, torch.as_tensor(done, dtype=torch.float32))
+    stats = {"mean_episode_return": float(np.mean(ep_returns))} if ep_returns else {}
+    return obs, done, stats
+
+
+def collect_rollouts_recurrent(
+    agent,
+    env,
+    buffer: RolloutBuffer,
+    n_steps: int,
+    obs: Optional[np.ndarray] = None,
+    done: Optional[np.ndarray] = None,
+):
+    """Recurrent variant: threads hidden state through collection (BPTT PPO)."""
+    raise NotImplementedError("recurrent rollout collection lands with LSTM PPO")
