@@ -107,28 +107,31 @@ class LunarLanderVecEnv(BatchedVecEnv):
         py = py + vy * self.DT
         ang = ang + vang * self.DT
 
-        # ground interaction
+        # ground interaction — crash/landing judged on the RAW impact state;
+        # shaping is also computed pre-clamp so zeroing the velocity at ground
+        # contact cannot leak a spurious positive speed-term delta
         on_ground = py <= self.LEG_Y
         upright = np.abs(ang) < 0.4
         self.legs[:, 0] = (on_ground & upright).astype(np.float64)
         self.legs[:, 1] = (on_ground & upright).astype(np.float64)
-        # clamp at ground
         py = np.maximum(py, 0.0)
         grounded = py <= 0.0 + 1e-9
-        vy = np.where(grounded & (vy < 0), 0.0, vy)
-        vx = np.where(grounded, vx * 0.8, vx)
-        vang = np.where(grounded, vang * 0.5, vang)
+        impact_speed = np.sqrt(vx**2 + vy**2)
 
         self.state = np.stack([px, py, vx, vy, ang, vang], axis=1)
-
         shaping = self._shaping()
         reward = (shaping - self.prev_shaping).astype(np.float32)
         self.prev_shaping = shaping
         reward -= (0.30 * main + 0.03 * (left + right)).astype(np.float32)
 
-        speed = np.sqrt(vx**2 + vy**2)
-        crash = (grounded & (~upright | (speed > 1.5))) | (np.abs(px) > self.X_WORLD)
-        landed = grounded & upright & (speed < 0.25) & (np.abs(vang) < 0.3)
+        # now damp ground-contact velocities for rows that keep running
+        vy = np.where(grounded & (vy < 0), 0.0, vy)
+        vx = np.where(grounded, vx * 0.8, vx)
+        vang = np.where(grounded, vang * 0.5, vang)
+        self.state = np.stack([px, py, vx, vy, ang, vang], axis=1)
+
+        crash = (grounded & (~upright | (impact_speed > 1.5))) | (np.abs(px) > self.X_WORLD)
+        landed = grounded & upright & (impact_speed <= 1.5) & (np.abs(vang) < 0.3)
         terminated = crash | landed
         reward = np.where(crash, reward - 100.0, reward)
         reward = np.where(landed, reward + 100.0, reward)

@@ -72,7 +72,7 @@ class ObsDependentRewardEnv(BatchedVecEnv):
         self.state[mask] = self.rng.integers(0, 2, int(mask.sum())).astype(np.float32)
 
     def _obs(self):
-        return self.state.reshape(-1, 1)
+        return self.state.reshape(-1, 1).copy()
 
     def _step_all(self, actions):
         reward = np.where(self.state > 0.5, 1.0, -1.0).astype(np.float32)
@@ -100,7 +100,7 @@ class DiscountedRewardEnv(BatchedVecEnv):
         self.phase[mask] = 0.0
 
     def _obs(self):
-        return self.phase.reshape(-1, 1)
+        return self.phase.reshape(-1, 1).copy()
 
     def _step_all(self, actions):
         second = self.phase > 0.5

@@ -70,6 +70,7 @@ def _wrap_mutation_method(name: str, fn: Callable) -> Callable:
 
     wrapper.__name__ = fn.__name__
     wrapper.__doc__ = fn.__doc__
+    wrapper.__signature__ = inspect.signature(fn)
     wrapper._mutation_type = fn._mutation_type
     wrapper._mutation_wrapped = True
     return wrapper

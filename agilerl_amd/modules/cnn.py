@@ -95,9 +95,9 @@ class EvolvableCNN(EvolvableModule):
         self.model = new_model
 
     @mutation(MutationType.LAYER)
-    def add_layer(self) -> dict:
+    def add_layer(self, hidden_layer: Optional[int] = None, numb_new_channels: Optional[int] = None) -> dict:
         if len(self.channel_size) >= self.max_hidden_layers:
-            return self.add_channel()
+            return self.add_channel(hidden_layer, numb_new_channels)
         self.channel_size.append(self.channel_size[-1])
         self.kernel_size.append(3)
         self.stride_size.append(1)
@@ -105,9 +105,9 @@ class EvolvableCNN(EvolvableModule):
         return {}
 
     @mutation(MutationType.LAYER)
-    def remove_layer(self) -> dict:
+    def remove_layer(self, hidden_layer: Optional[int] = None, numb_new_channels: Optional[int] = None) -> dict:
         if len(self.channel_size) <= self.min_hidden_layers:
-            return self.add_channel()
+            return self.add_channel(hidden_layer, numb_new_channels)
         self.channel_size.pop()
         self.kernel_size.pop()
         self.stride_size.pop()

@@ -137,17 +137,19 @@ class EvolvableMLP(EvolvableModule):
         self.model = new_model
 
     @mutation(MutationType.LAYER)
-    def add_layer(self) -> dict:
+    def add_layer(self, hidden_layer: Optional[int] = None, numb_new_nodes: Optional[int] = None) -> dict:
+        # fallback choices are accepted so a replay on a sibling network takes
+        # the identical path even when the depth bound triggers the fallback
         if len(self.hidden_size) >= self.max_hidden_layers:
-            return self.add_node()
+            return self.add_node(hidden_layer, numb_new_nodes)
         self.hidden_size.append(self.hidden_size[-1])
         self.recreate_network()
         return {}
 
     @mutation(MutationType.LAYER)
-    def remove_layer(self) -> dict:
+    def remove_layer(self, hidden_layer: Optional[int] = None, numb_new_nodes: Optional[int] = None) -> dict:
         if len(self.hidden_size) <= self.min_hidden_layers:
-            return self.add_node()
+            return self.add_node(hidden_layer, numb_new_nodes)
         self.hidden_size.pop()
         self.recreate_network()
         return {}

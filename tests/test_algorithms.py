@@ -1,0 +1,149 @@
+"""Algorithm tests: probe-env convergence, clone, checkpoint round-trip."""
+
+import os
+
+import numpy as np
+import pytest
+import torch
+
+from agilerl_amd.algorithms.dqn import DQN
+from agilerl_amd.algorithms.ppo import PPO
+from agilerl_amd.components import ReplayBuffer, RolloutBuffer
+from agilerl_amd.envs.probe import (
+    ConstantRewardEnv,
+    DiscountedRewardEnv,
+    FixedObsPolicyEnv,
+    ObsDependentRewardEnv,
+    PolicyEnv,
+)
+from agilerl_amd.rollouts.on_policy import collect_rollouts
+from agilerl_amd.spaces import Box, Discrete
+
+
+def fill_buffer(env, buf, steps=200):
+    obs, _ = env.reset(seed=0)
+    for _ in range(steps):
+        action = np.array([env.single_action_space.sample() for _ in range(env.num_envs)])
+        next_obs, reward, term, trunc, info = env.step(action)
+        store_next = next_obs
+        if (term | trunc).any() and "final_observation" in info:
+            store_next = next_obs.copy()
+            store_next[term | trunc] = info["final_observation"][term | trunc]
+        buf.add(obs=obs, action=action, reward=reward, next_obs=store_next,
+                done=term.astype(np.float32))
+        obs = next_obs
+
+
+class TestDQNProbe:
+    def _train(self, env, gamma=0.99, iters=300):
+        agent = DQN(env.observation_space, env.action_space, lr=1e-2, gamma=gamma,
+                    tau=0.1, batch_size=64, net_config={"arch": "mlp", "hidden_size": [32]})
+        buf = ReplayBuffer(2000)
+        fill_buffer(env, buf, 200)
+        for _ in range(iters):
+            agent.learn(buf.sample(64))
+        return agent
+
+    def test_constant_reward(self):
+        env = ConstantRewardEnv(num_envs=4)
+        agent = self._train(env)
+        q = agent.actor(torch.zeros(1, 1))
+        assert torch.allclose(q, torch.ones(1, 2), atol=0.1)
+
+    def test_obs_dependent_reward(self):
+        env = ObsDependentRewardEnv(num_envs=4)
+        agent = self._train(env)
+        q0 = agent.actor(torch.zeros(1, 1))
+        q1 = agent.actor(torch.ones(1, 1))
+        assert torch.allclose(q0, -torch.ones(1, 2), atol=0.15)
+        assert torch.allclose(q1, torch.ones(1, 2), atol=0.15)
+
+    def test_discounted_reward(self):
+        env = DiscountedRewardEnv(num_envs=4)
+        agent = self._train(env, gamma=0.9, iters=500)
+        q0 = agent.actor(torch.zeros(1, 1))
+        q1 = agent.actor(torch.ones(1, 1))
+        assert torch.allclose(q1, torch.ones(1, 2), atol=0.15)
+        assert torch.allclose(q0, torch.full((1, 2), 0.9), atol=0.15)
+
+
+class TestPPOProbe:
+    def test_fixed_obs_policy(self):
+        env = FixedObsPolicyEnv(num_envs=8)
+        agent = PPO(env.observation_space, env.action_space, lr=5e-3, learn_step=32,
+                    batch_size=64, ent_coef=0.0,
+                    net_config={"arch": "mlp", "hidden_size": [32]})
+        buf = RolloutBuffer(32, 8, gamma=agent.gamma, gae_lambda=agent.gae_lambda)
+        obs = done = None
+        for _ in range(20):
+            obs, done, _ = collect_rollouts(agent, env, buf, 32, obs, done)
+            agent.learn(buf)
+        with torch.no_grad():
+            head = agent.actor(agent.actor.preprocess(np.zeros((1, 1), dtype=np.float32)))
+            probs = torch.softmax(head, dim=-1)
+        assert probs[0, 0] > 0.9
+
+    def test_policy_env(self):
+        env = PolicyEnv(num_envs=8)
+        agent = PPO(env.observation_space, env.action_space, lr=5e-3, learn_step=32,
+                    batch_size=64, ent_coef=0.0,
+                    net_config={"arch": "mlp", "hidden_size": [32]})
+        buf = RolloutBuffer(32, 8, gamma=agent.gamma, gae_lambda=agent.gae_lambda)
+        obs = done = None
+        for _ in range(25):
+            obs, done, _ = collect_rollouts(agent, env, buf, 32, obs, done)
+            agent.learn(buf)
+        with torch.no_grad():
+            obs0 = np.array([[1.0, 0.0]], dtype=np.float32)
+            obs1 = np.array([[0.0, 1.0]], dtype=np.float32)
+            p0 = torch.softmax(agent.actor(agent.actor.preprocess(obs0)), -1)
+            p1 = torch.softmax(agent.actor(agent.actor.preprocess(obs1)), -1)
+        assert p0[0, 0] > 0.8
+        assert p1[0, 1] > 0.8
+
+
+class TestCloneCheckpoint:
+    def test_dqn_clone_identical(self):
+        obs_s, act_s = Box(-1, 1, (4,)), Discrete(3)
+        agent = DQN(obs_s, act_s)
+        clone = agent.clone(index=5)
+        x = torch.randn(3, 4)
+        assert torch.allclose(agent.actor(x), clone.actor(x))
+        assert clone.index == 5
+
+    def test_clone_after_arch_mutation(self):
+        agent = DQN(Box(-1, 1, (4,)), Discrete(3))
+        agent.apply_architecture_mutation("encoder.add_node", numb_new_nodes=32)
+        agent.lr = 5e-4
+        clone = agent.clone()
+        x = torch.randn(3, 4)
+        assert torch.allclose(agent.actor(x), clone.actor(x))
+        assert torch.allclose(agent.actor_target(x), clone.actor_target(x))
+        assert clone.lr == 5e-4
+        assert clone.optimizer.lr == 5e-4
+
+    def test_checkpoint_roundtrip(self, tmp_path):
+        agent = DQN(Box(-1, 1, (4,)), Discrete(3), batch_size=77)
+        agent.apply_architecture_mutation("head.add_layer")
+        agent.fitness = [1.0, 2.0]
+        path = str(tmp_path / "agent.pt")
+        agent.save_checkpoint(path)
+        loaded = DQN.load(path)
+        x = torch.randn(3, 4)
+        assert torch.allclose(agent.actor(x), loaded.actor(x))
+        assert loaded.batch_size == 77
+        assert loaded.fitness == [1.0, 2.0]
+
+    def test_ppo_checkpoint_roundtrip(self, tmp_path):
+        agent = PPO(Box(-1, 1, (4,)), Discrete(3))
+        path = str(tmp_path / "ppo.pt")
+        agent.save_checkpoint(path)
+        loaded = PPO.load(path)
+        x = torch.randn(3, 4)
+        assert torch.allclose(agent.critic(x), loaded.critic(x))
+
+    def test_target_synced_after_mutation(self):
+        agent = DQN(Box(-1, 1, (4,)), Discrete(3))
+        agent.apply_architecture_mutation("encoder.add_layer")
+        x = torch.randn(3, 4)
+        assert torch.allclose(agent.actor(x), agent.actor_target(x))

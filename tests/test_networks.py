@@ -1,0 +1,115 @@
+"""Network-layer tests."""
+
+import numpy as np
+import pytest
+import torch
+
+from agilerl_amd.networks import (
+    ContinuousQNetwork,
+    DeterministicActor,
+    QNetwork,
+    RainbowQNetwork,
+    StochasticActor,
+    ValueNetwork,
+)
+from agilerl_amd.spaces import Box, DictSpace, Discrete, MultiDiscrete
+
+
+OBS = Box(-1, 1, (8,))
+ACT = Discrete(4)
+
+
+class TestQNetwork:
+    def test_shapes(self):
+        q = QNetwork(OBS, ACT)
+        assert q(torch.randn(5, 8)).shape == (5, 4)
+
+    def test_discrete_obs_one_hot(self):
+        q = QNetwork(Discrete(6), ACT)
+        out = q(torch.tensor([0, 3, 5]))
+        assert out.shape == (3, 4)
+
+    def test_latent_mutation_consistency(self):
+        q = QNetwork(OBS, ACT)
+        q.apply_mutation("add_latent_node", numb_new_nodes=16)
+        assert q.latent_dim == 80
+        assert q(torch.randn(5, 8)).shape == (5, 4)
+        clone = q.clone()
+        x = torch.randn(3, 8)
+        assert torch.allclose(clone(x), q(x))
+
+    def test_namespaced_mutations(self):
+        q = QNetwork(OBS, ACT)
+        methods = q.mutation_methods
+        assert "encoder.add_node" in methods
+        assert "head.add_layer" in methods
+        assert "add_latent_node" in methods
+
+
+class TestRainbow:
+    def test_dist_normalized(self):
+        net = RainbowQNetwork(OBS, ACT, num_atoms=51)
+        d = net.dist(torch.randn(5, 8))
+        assert d.shape == (5, 4, 51)
+        assert torch.allclose(d.sum(-1), torch.ones(5, 4), atol=1e-5)
+
+    def test_value_stream_mirrors_mutation(self):
+        net = RainbowQNetwork(OBS, ACT)
+        net.apply_mutation("head.add_node", hidden_layer=0, numb_new_nodes=32)
+        assert net.head_net.hidden_size == net.value_net.hidden_size
+        clone = net.clone()
+        x = torch.randn(3, 8)
+        clone.eval(), net.eval()
+        assert torch.allclose(clone(x), net(x))
+
+
+class TestActors:
+    def test_deterministic_box_rescale(self):
+        actor = DeterministicActor(OBS, Box(-2.0, 2.0, (3,)))
+        out = actor(torch.randn(5, 8))
+        assert out.shape == (5, 3)
+        assert (out.abs() <= 2.0 + 1e-5).all()
+
+    def test_stochastic_discrete(self):
+        actor = StochasticActor(OBS, ACT)
+        a, lp, ent = actor.sample(torch.randn(5, 8))
+        assert a.shape == (5,)
+        lp2, _ = actor.evaluate_actions(torch.randn(5, 8) * 0 + torch.randn(5, 8), a)
+        assert lp2.shape == (5,)
+
+    def test_stochastic_masked(self):
+        actor = StochasticActor(OBS, ACT)
+        mask = torch.zeros(5, 4, dtype=torch.bool)
+        mask[:, 2] = True
+        a, _, _ = actor.sample(torch.randn(5, 8), mask)
+        assert (a == 2).all()
+
+    def test_multidiscrete(self):
+        actor = StochasticActor(OBS, MultiDiscrete([3, 4]))
+        a, lp, ent = actor.sample(torch.randn(5, 8))
+        assert a.shape == (5, 2)
+        assert (a[:, 0] < 3).all() and (a[:, 1] < 4).all()
+
+    def test_continuous(self):
+        actor = StochasticActor(OBS, Box(-1, 1, (2,)))
+        a, lp, ent = actor.sample(torch.randn(5, 8))
+        assert a.shape == (5, 2)
+
+
+class TestContinuousQ:
+    def test_forward(self):
+        q = ContinuousQNetwork(OBS, Box(-1, 1, (3,)))
+        out = q(torch.randn(5, 8), torch.randn(5, 3))
+        assert out.shape == (5, 1)
+
+    def test_latent_mutation(self):
+        q = ContinuousQNetwork(OBS, Box(-1, 1, (3,)))
+        q.apply_mutation("add_latent_node", numb_new_nodes=8)
+        assert q(torch.randn(5, 8), torch.randn(5, 3)).shape == (5, 1)
+
+
+def test_dict_obs_network():
+    space = DictSpace({"vec": Box(-1, 1, (4,)), "vec2": Box(-1, 1, (6,))})
+    q = QNetwork(space, ACT)
+    obs = {"vec": torch.randn(5, 4), "vec2": torch.randn(5, 6)}
+    assert q(obs).shape == (5, 4)
