@@ -295,6 +295,8 @@ class EvolvableAlgorithm(metaclass=AlgorithmMeta):
             if hasattr(self, opt_name):
                 getattr(self, opt_name).load_state_dict(opt_state)
         for k, v in ckpt["attributes"].items():
+            if k == "device":  # receiving agent keeps its own device
+                continue
             setattr(self, k, v)
 
     @classmethod

@@ -69,6 +69,7 @@ class BatchedVecEnv(VecEnv):
         self.num_envs = int(num_envs)
         self.rng = np.random.default_rng(seed)
         self._elapsed = np.zeros(self.num_envs, dtype=np.int64)
+        self._ep_return = np.zeros(self.num_envs, dtype=np.float64)
 
     # -- subclass hooks -------------------------------------------------
     def _reset_rows(self, mask: np.ndarray) -> None:
@@ -87,6 +88,7 @@ class BatchedVecEnv(VecEnv):
             self.rng = np.random.default_rng(seed)
         self._reset_rows(np.ones(self.num_envs, dtype=bool))
         self._elapsed[:] = 0
+        self._ep_return[:] = 0.0
         return self._obs(), {}
 
     def step(self, actions):
@@ -102,10 +104,13 @@ class BatchedVecEnv(VecEnv):
         truncated = truncated & ~terminated
         done = terminated | truncated
         obs = self._obs()
+        self._ep_return += reward
         if done.any():
             info = dict(info)
             info["final_observation"] = obs.copy()
+            info["episode_return"] = self._ep_return[done].copy()
             self._reset_rows(done)
             self._elapsed[done] = 0
+            self._ep_return[done] = 0.0
             obs = self._obs()
         return obs, reward, terminated, truncated, info

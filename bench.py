@@ -1,0 +1,210 @@
+#!/usr/bin/env python3
+"""Headline benchmark: PPO LunarLander pop=8 evolutionary HPO.
+
+BASELINE.json metric: env steps/sec (whole node) for PPO LunarLander
+pop=8 at 1/2/4/8 GPUs (one agent per GPU at N=8; agents round-robin on
+fewer GPUs).  One bench "step" = every population agent runs one
+collect(learn_step x num_envs env-steps) + PPO-update cycle; an
+evolution round (fitness all-gather + rank-0 tournament plan broadcast +
+winner weight transfer + mutation, all over RCCL/xGMI) fires every
+EVO_EVERY steps INSIDE the timed region.
+
+Launch (the driver does this for N>1):
+  python -m torch.distributed.run --nnodes=1 --nproc-per-node N \
+      --master-addr 127.0.0.1 --master-port P bench.py --gpus N --steps K --warmup W
+
+Single line of JSON on rank 0 at the end (driver contract).
+"""
+
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import sys
+import time
+
+import numpy as np
+import torch
+
+sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
+
+from agilerl_amd.algorithms.ppo import PPO  # noqa: E402
+from agilerl_amd.components.rollout_buffer import RolloutBuffer  # noqa: E402
+from agilerl_amd.envs import LunarLanderVecEnv  # noqa: E402
+from agilerl_amd.hpo import Mutations, TournamentSelection  # noqa: E402
+from agilerl_amd.parallel import DistributedPopulation, DistributedState, barrier  # noqa: E402
+from agilerl_amd.rollouts.on_policy import collect_rollouts  # noqa: E402
+
+POP_SIZE = 8
+NUM_ENVS = 64
+LEARN_STEP = 128  # rollout length per cycle
+EVO_EVERY = 4
+
+
+def parse_args():
+    p = argparse.ArgumentParser()
+    p.add_argument("--gpus", type=int, default=1)
+    p.add_argument("--steps", type=int, default=8)
+    p.add_argument("--warmup", type=int, default=2)
+    p.add_argument("--num-envs", type=int, default=NUM_ENVS)
+    p.add_argument("--learn-step", type=int, default=LEARN_STEP)
+    p.add_argument("--pop-size", type=int, default=POP_SIZE)
+    return p.parse_args()
+
+
+class BenchRunner:
+    def __init__(self, args):
+        self.args = args
+        self.state = DistributedState.get()
+        self.device = self.state.device
+        seed = 1234 + self.state.rank
+        np.random.seed(seed)
+        torch.manual_seed(seed)
+
+        def factory(index: int) -> PPO:
+            return PPO(
+                observation_space=LunarLanderVecEnv(1).single_observation_space,
+                action_space=LunarLanderVecEnv(1).single_action_space,
+                index=index,
+                learn_step=args.learn_step,
+                batch_size=2048,
+                lr=3e-4,
+                update_epochs=4,
+                net_config={"arch": "mlp", "hidden_size": [64, 64]},
+                device=self.device,
+            )
+
+        self.pop = DistributedPopulation(factory, args.pop_size)
+        self.tournament = TournamentSelection(tournament_size=2, elitism=True,
+                                              rng=np.random.default_rng(7))
+        self.mutations = Mutations(
+            no_mutation=0.4, architecture=0.2, parameters=0.1, activation=0.1,
+            rl_hp=0.2, rand_seed=seed, device=self.device,
+        )
+        # per-slot envs + rollout buffers + carried state
+        self.envs = {}
+        self.buffers = {}
+        self.carried = {}
+        self.fit_window = {}
+        for slot in self.pop.local_indices:
+            self._init_slot(slot)
+        self.step_count = 0
+
+    def _init_slot(self, slot: int) -> None:
+        self.envs[slot] = LunarLanderVecEnv(self.args.num_envs, seed=1000 + slot)
+        agent = self.pop.agents[slot]
+        self.buffers[slot] = RolloutBuffer(
+            capacity=agent.learn_step, num_envs=self.args.num_envs,
+            device=self.device, gamma=agent.gamma, gae_lambda=agent.gae_lambda,
+        )
+        self.carried[slot] = (None, None)
+        self.fit_window[slot] = []
+
+    def bench_step(self) -> int:
+        """One population cycle; returns env steps consumed (this rank)."""
+        steps = 0
+        for slot in self.pop.local_indices:
+            agent = self.pop.agents[slot]
+            buffer = self.buffers[slot]
+            if buffer.capacity != agent.learn_step:
+                buffer = RolloutBuffer(
+                    capacity=agent.learn_step, num_envs=self.args.num_envs,
+                    device=self.device, gamma=agent.gamma, gae_lambda=agent.gae_lambda,
+                )
+                self.buffers[slot] = buffer
+            obs, done = self.carried[slot]
+            env = self.envs[slot]
+            obs, done, stats = collect_rollouts(agent, env, buffer, agent.learn_step, obs, done)
+            self.carried[slot] = (obs, done)
+            agent.learn(buffer)
+            n = agent.learn_step * self.args.num_envs
+            agent.steps[-1] += n
+            steps += n
+            if "mean_episode_return" in stats:
+                self.fit_window[slot].append(stats["mean_episode_return"])
+                self.fit_window[slot] = self.fit_window[slot][-5:]
+                agent.fitness.append(float(np.mean(self.fit_window[slot])))
+        self.step_count += 1
+        if self.step_count % EVO_EVERY == 0:
+            self.pop.evolve(self.tournament, self.mutations)
+            # re-bind env/buffer state for (possibly new) local slots
+            for slot in self.pop.local_indices:
+                if slot not in self.envs:
+                    self._init_slot(slot)
+                else:
+                    self.carried[slot] = (None, None)
+        return steps
+
+
+def main():
+    args = parse_args()
+    runner = BenchRunner(args)
+    state = runner.state
+    use_cuda = torch.cuda.is_available()
+
+    for _ in range(args.warmup):
+        runner.bench_step()
+
+    barrier()
+    if use_cuda:
+        torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    local_steps = 0
+    for _ in range(args.steps):
+        local_steps += runner.bench_step()
+    if use_cuda:
+        torch.cuda.synchronize()
+    barrier()
+    elapsed = time.perf_counter() - t0
+
+    # max elapsed over ranks; total env steps over ranks
+    if state.is_distributed:
+        import torch.distributed as dist
+
+        t = torch.tensor([elapsed, float(local_steps)])
+        if state.backend == "nccl":
+            t = t.to(state.device)
+        dist.all_reduce(t[0:1], op=dist.ReduceOp.MAX)
+        dist.all_reduce(t[1:2], op=dist.ReduceOp.SUM)
+        elapsed = float(t[0])
+        total_steps = float(t[1])
+    else:
+        total_steps = float(local_steps)
+
+    if state.is_main:
+        value = total_steps / elapsed
+        result = {
+            "metric": "env_steps_per_sec",
+            "value": value,
+            "unit": "steps/s",
+            "n_gpus": state.world_size,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": elapsed / args.steps * 1000.0,
+            "higher_is_better": True,
+            "scaling": "strong",
+            "vs_baseline": None,
+            "dtype": "fp32",
+            "data": "synthetic (first-party Box2D-free LunarLander dynamics, same obs/action spaces)",
+            "config": {
+                "model": "PPO mlp[64,64] evolutionary-HPO",
+                "env": "LunarLander-v2 (first-party vectorized reimpl)",
+                "global_batch": args.pop_size * args.num_envs * args.learn_step,
+                "seq_len": args.learn_step,
+                "num_envs_per_agent": args.num_envs,
+                "pop_size": args.pop_size,
+                "evo_every": EVO_EVERY,
+                "parallelism": f"population-parallel dp{state.world_size} (one agent per GPU at 8)",
+            },
+        }
+        print(json.dumps(result), flush=True)
+
+    if state.is_distributed:
+        import torch.distributed as dist
+
+        dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()

@@ -1,0 +1,205 @@
+"""HIP kernel numerics vs eager fp32 PyTorch references (gpu-marked)."""
+
+import numpy as np
+import pytest
+import torch
+
+from agilerl_amd import ops
+from agilerl_amd.ops.backend import extension
+
+pytestmark = pytest.mark.gpu
+
+DEV = "cuda:0"
+
+
+@pytest.fixture(scope="module", autouse=True)
+def _require_ext():
+    assert extension() is not None, "HIP extension must be built on GPU boxes"
+
+
+class TestGaeScan:
+    def test_matches_cpu(self):
+        T, N = 64, 129
+        rewards = torch.randn(T, N)
+        values = torch.randn(T, N)
+        dones = (torch.rand(T, N) < 0.1).float()
+        last_value = torch.randn(N)
+        adv_cpu, ret_cpu = ops.gae_scan(rewards, values, dones, last_value, 0.99, 0.95)
+        adv_gpu, ret_gpu = ops.gae_scan(
+            rewards.to(DEV), values.to(DEV), dones.to(DEV), last_value.to(DEV), 0.99, 0.95
+        )
+        torch.testing.assert_close(adv_gpu.cpu(), adv_cpu, rtol=1e-4, atol=1e-5)
+        torch.testing.assert_close(ret_gpu.cpu(), ret_cpu, rtol=1e-4, atol=1e-5)
+
+
+class TestNStep:
+    def test_matches_cpu(self):
+        B, n = 257, 5
+        rewards = torch.randn(B, n)
+        dones = (torch.rand(B, n) < 0.2).float()
+        r_cpu, s_cpu = ops.nstep_scan(rewards, dones, 0.97)
+        r_gpu, s_gpu = ops.nstep_scan(rewards.to(DEV), dones.to(DEV), 0.97)
+        torch.testing.assert_close(r_gpu.cpu(), r_cpu, rtol=1e-5, atol=1e-6)
+        torch.testing.assert_close(s_gpu.cpu(), s_cpu)
+
+
+class TestC51:
+    def test_matches_cpu(self):
+        B, A = 128, 51
+        dist = torch.softmax(torch.randn(B, A), -1)
+        rewards = torch.randn(B) * 5
+        dones = (torch.rand(B) < 0.3).float()
+        support = torch.linspace(-10, 10, A)
+        p_cpu = ops.c51_project(dist, rewards, dones, support, 0.99, -10, 10)
+        p_gpu = ops.c51_project(
+            dist.to(DEV), rewards.to(DEV), dones.to(DEV), support.to(DEV), 0.99, -10, 10
+        )
+        torch.testing.assert_close(p_gpu.cpu(), p_cpu, rtol=1e-4, atol=1e-5)
+        torch.testing.assert_close(p_gpu.sum(-1).cpu(), torch.ones(B), rtol=1e-4, atol=1e-5)
+
+
+class TestPolyak:
+    def test_matches_foreach(self):
+        tgt = [torch.randn(100, 37, device=DEV), torch.randn(5, device=DEV), torch.randn(64, device=DEV)]
+        src = [torch.randn_like(t) for t in tgt]
+        expected = [t + 0.1 * (s - t) for t, s in zip(tgt, src)]
+        ops.polyak_update_(tgt, src, 0.1)
+        for t, e in zip(tgt, expected):
+            torch.testing.assert_close(t, e, rtol=1e-5, atol=1e-6)
+
+
+class TestNoisyLinear:
+    def test_forward_backward_match(self):
+        torch.manual_seed(0)
+        B, I, O = 64, 37, 29
+        x = torch.randn(B, I, device=DEV, requires_grad=True)
+        w_mu = torch.randn(O, I, device=DEV, requires_grad=True)
+        w_sig = torch.rand(O, I, device=DEV, requires_grad=True)
+        w_eps = torch.randn(O, I, device=DEV)
+        b_mu = torch.randn(O, device=DEV, requires_grad=True)
+        b_sig = torch.rand(O, device=DEV, requires_grad=True)
+        b_eps = torch.randn(O, device=DEV)
+
+        out = ops.noisy_linear(x, w_mu, w_sig, w_eps, b_mu, b_sig, b_eps)
+        ref = torch.nn.functional.linear(x, w_mu + w_sig * w_eps, b_mu + b_sig * b_eps)
+        torch.testing.assert_close(out, ref, rtol=1e-4, atol=1e-5)
+
+        g = torch.randn_like(out)
+        grads = torch.autograd.grad(out, [x, w_mu, w_sig, b_mu, b_sig], g, retain_graph=True)
+        ref_grads = torch.autograd.grad(ref, [x, w_mu, w_sig, b_mu, b_sig], g)
+        for a, b in zip(grads, ref_grads):
+            torch.testing.assert_close(a, b, rtol=1e-3, atol=1e-4)
+
+
+class TestGroupAdvantage:
+    def test_matches_cpu(self):
+        r = torch.randn(16 * 8)
+        a_cpu = ops.group_advantage(r, 8)
+        a_gpu = ops.group_advantage(r.to(DEV), 8)
+        torch.testing.assert_close(a_gpu.cpu(), a_cpu, rtol=1e-4, atol=1e-5)
+
+    def test_large_group(self):
+        r = torch.randn(4 * 300)
+        a_cpu = ops.group_advantage(r, 300)
+        a_gpu = ops.group_advantage(r.to(DEV), 300)
+        torch.testing.assert_close(a_gpu.cpu(), a_cpu, rtol=1e-4, atol=1e-4)
+
+
+class TestRowLseGather:
+    def test_matches_eager(self):
+        ext = extension()
+        rows, V = 128, 32000
+        logits = torch.randn(rows, V, device=DEV)
+        targets = torch.randint(0, V, (rows,), device=DEV)
+        lp, lse = ext.row_lse_gather(logits, targets, 1.0)
+        ref_lse = torch.logsumexp(logits, dim=-1)
+        ref_lp = logits.gather(1, targets.unsqueeze(1)).squeeze(1) - ref_lse
+        torch.testing.assert_close(lse, ref_lse, rtol=1e-4, atol=1e-4)
+        torch.testing.assert_close(lp, ref_lp, rtol=1e-4, atol=1e-4)
+
+    def test_temperature(self):
+        ext = extension()
+        rows, V = 16, 1000
+        logits = torch.randn(rows, V, device=DEV)
+        targets = torch.randint(0, V, (rows,), device=DEV)
+        lp, _ = ext.row_lse_gather(logits, targets, 2.0)
+        scaled = logits / 2.0
+        ref = scaled.gather(1, targets.unsqueeze(1)).squeeze(1) - torch.logsumexp(scaled, -1)
+        torch.testing.assert_close(lp, ref, rtol=1e-4, atol=1e-4)
+
+    def test_softmax_bwd(self):
+        ext = extension()
+        rows, V = 32, 5000
+        logits = torch.randn(rows, V, device=DEV, requires_grad=True)
+        targets = torch.randint(0, V, (rows,), device=DEV)
+        g = torch.randn(rows, device=DEV)
+        lse_ref = torch.logsumexp(logits, -1)
+        lp_ref = logits.gather(1, targets.unsqueeze(1)).squeeze(1) - lse_ref
+        (ref_grad,) = torch.autograd.grad(lp_ref, logits, g)
+
+        work = logits.detach().clone()
+        ext.row_softmax_bwd_(work, targets, lse_ref.detach(), g, 1.0)
+        torch.testing.assert_close(work, ref_grad, rtol=1e-3, atol=1e-5)
+
+
+class TestGrpoTokenLoss:
+    def _eager(self, logp, old_logp, ref_logp, adv, mask, lo, hi, kl, cispo):
+        logp = logp.detach().requires_grad_(True)
+        ratio = (logp - old_logp).exp()
+        if cispo:
+            w = ratio.clamp(lo, hi).detach()
+            loss = -w * adv * logp
+        else:
+            loss = -torch.minimum(ratio * adv, ratio.clamp(lo, hi) * adv)
+        if ref_logp is not None and kl:
+            d = ref_logp - logp
+            loss = loss + kl * (d.exp() - d - 1)
+        loss = loss * mask
+        (dl,) = torch.autograd.grad(loss.sum(), logp)
+        return loss.detach(), dl
+
+    @pytest.mark.parametrize("cispo", [False, True])
+    @pytest.mark.parametrize("use_ref", [False, True])
+    def test_matches_eager(self, cispo, use_ref):
+        ext = extension()
+        N = 4096
+        logp = torch.randn(N, device=DEV) * 0.5
+        old = logp + torch.randn(N, device=DEV) * 0.2
+        ref = logp + torch.randn(N, device=DEV) * 0.2 if use_ref else None
+        adv = torch.randn(N, device=DEV)
+        mask = (torch.rand(N, device=DEV) < 0.8).float()
+        loss, dl = ext.grpo_token_loss(logp, old, ref, adv, mask, 0.8, 1.2, 0.04 if use_ref else 0.0, cispo)
+        ref_loss, ref_dl = self._eager(logp, old, ref, adv, mask, 0.8, 1.2, 0.04 if use_ref else 0.0, cispo)
+        torch.testing.assert_close(loss, ref_loss, rtol=1e-4, atol=1e-5)
+        torch.testing.assert_close(dl, ref_dl, rtol=1e-4, atol=1e-5)
+
+
+class TestAgentOnGpu:
+    def test_dqn_learn_gpu(self):
+        from agilerl_amd.algorithms.dqn import DQN
+        from agilerl_amd.spaces import Box, Discrete
+
+        agent = DQN(Box(-1, 1, (8,)), Discrete(4), device=DEV)
+        batch = {
+            "obs": torch.randn(64, 8, device=DEV),
+            "action": torch.randint(0, 4, (64,), device=DEV),
+            "reward": torch.randn(64, device=DEV),
+            "next_obs": torch.randn(64, 8, device=DEV),
+            "done": torch.zeros(64, device=DEV),
+        }
+        loss = agent.learn(batch)
+        assert np.isfinite(loss)
+
+    def test_ppo_cycle_gpu(self):
+        from agilerl_amd.algorithms.ppo import PPO
+        from agilerl_amd.components.rollout_buffer import RolloutBuffer
+        from agilerl_amd.envs import LunarLanderVecEnv
+        from agilerl_amd.rollouts.on_policy import collect_rollouts
+
+        env = LunarLanderVecEnv(8, seed=0)
+        agent = PPO(env.single_observation_space, env.single_action_space,
+                    learn_step=16, batch_size=64, device=DEV)
+        buf = RolloutBuffer(16, 8, device=DEV, gamma=agent.gamma, gae_lambda=agent.gae_lambda)
+        obs, done, _ = collect_rollouts(agent, env, buf, 16)
+        stats = agent.learn(buf)
+        assert np.isfinite(stats["policy_loss"])
