@@ -1,0 +1,21 @@
+from .dqn import DQN
+from .dqn_rainbow import RainbowDQN
+from .cqn import CQN
+from .ddpg import DDPG
+from .td3 import TD3
+from .ppo import PPO
+from .maddpg import MADDPG
+from .matd3 import MATD3
+from .ippo import IPPO
+
+__all__ = [
+    "DQN",
+    "RainbowDQN",
+    "CQN",
+    "DDPG",
+    "TD3",
+    "PPO",
+    "MADDPG",
+    "MATD3",
+    "IPPO",
+]

@@ -1,0 +1,260 @@
+"""MADDPG — multi-agent DDPG with centralized critics.
+
+Reference parity: ``agilerl/algorithms/maddpg.py:61`` — per-agent actors,
+centralized critics over concat(all obs, all actions) (``learn`` :694,
+stacked actions :722-728), Gumbel-Softmax for discrete action spaces.
+
+MI355X notes: per-agent networks live in :class:`ModuleDict`-s so
+architecture mutations broadcast consistently; the per-agent critic/actor
+losses are summed into ONE backward pass each (one HIP launch sequence
+instead of n_agents sequential graphs); polyak runs as a single fused
+kernel over every target parameter.
+"""
+
+from __future__ import annotations
+
+from typing import Any, Dict, List, Optional, Tuple
+
+import numpy as np
+import torch
+import torch.nn.functional as F
+
+from .. import ops
+from ..modules.base import ModuleDict
+from ..networks.actors import DeterministicActor
+from ..networks.q_networks import ContinuousQNetwork
+from ..spaces import Box, Discrete, Space, flatdim
+from .core.base import MultiAgentRLAlgorithm
+from .core.optimizer_wrapper import OptimizerWrapper
+from .core.registry import HyperparameterConfig, NetworkGroup, OptimizerConfig, RLParameter
+
+__all__ = ["MADDPG"]
+
+
+def default_hp_config() -> HyperparameterConfig:
+    return HyperparameterConfig(
+        lr_actor=RLParameter(min=1e-5, max=1e-2),
+        lr_critic=RLParameter(min=1e-5, max=1e-2),
+        batch_size=RLParameter(min=16, max=1024, dtype=int),
+        learn_step=RLParameter(min=1, max=16, dtype=int),
+    )
+
+
+class MADDPG(MultiAgentRLAlgorithm):
+    def __init__(
+        self,
+        observation_spaces: Dict[str, Space],
+        action_spaces: Dict[str, Space],
+        agent_ids: Optional[List[str]] = None,
+        index: int = 0,
+        hp_config: Optional[HyperparameterConfig] = None,
+        net_config: Optional[Dict[str, Any]] = None,
+        head_config: Optional[Dict[str, Any]] = None,
+        batch_size: int = 64,
+        lr_actor: float = 1e-4,
+        lr_critic: float = 1e-3,
+        learn_step: int = 5,
+        gamma: float = 0.95,
+        tau: float = 1e-2,
+        expl_noise: float = 0.1,
+        latent_dim: int = 64,
+        device: str = "cpu",
+    ):
+        super().__init__(
+            observation_spaces, action_spaces, agent_ids=agent_ids, index=index,
+            learn_step=learn_step, device=device,
+            hp_config=hp_config or default_hp_config(), name="MADDPG",
+        )
+        self.batch_size = int(batch_size)
+        self.lr_actor = float(lr_actor)
+        self.lr_critic = float(lr_critic)
+        self.gamma = float(gamma)
+        self.tau = float(tau)
+        self.expl_noise = float(expl_noise)
+        self.net_config = net_config
+        self.latent_dim = latent_dim
+
+        # joint (centralized) spaces for the critics
+        self.joint_obs_dim = sum(flatdim(sp) for sp in self.observation_spaces.values())
+        self.joint_action_dim = sum(self._raw_action_dim(sp) for sp in self.action_spaces.values())
+        joint_space = Box(-np.inf, np.inf, (self.joint_obs_dim,))
+
+        self.actors = ModuleDict(
+            {
+                aid: DeterministicActor(
+                    self.observation_spaces[aid], self.action_spaces[aid],
+                    encoder_config=net_config, head_config=head_config,
+                    latent_dim=latent_dim, device=device,
+                )
+                for aid in self.agent_ids
+            },
+            device=device,
+        )
+        self.actor_targets = self.actors.clone()
+        self.critics = ModuleDict(
+            {
+                aid: ContinuousQNetwork(
+                    joint_space, Box(-1.0, 1.0, (self.joint_action_dim,)),
+                    encoder_config=net_config, head_config=head_config,
+                    latent_dim=latent_dim, action_dim=self.joint_action_dim, device=device,
+                )
+                for aid in self.agent_ids
+            },
+            device=device,
+        )
+        self.critic_targets = self.critics.clone()
+        for net in (self.actor_targets, self.critic_targets):
+            for p in net.parameters():
+                p.requires_grad = False
+
+        self.actor_optimizer = OptimizerWrapper(
+            torch.optim.Adam, [self.actors], lr=self.lr_actor, multiagent=True
+        )
+        self.critic_optimizer = OptimizerWrapper(
+            torch.optim.Adam, [self.critics], lr=self.lr_critic, multiagent=True
+        )
+
+        self.register_network_group(
+            NetworkGroup(eval_network="actors", shared_networks=["actor_targets"], policy=True, multiagent=True)
+        )
+        self.register_network_group(
+            NetworkGroup(eval_network="critics", shared_networks=["critic_targets"], multiagent=True)
+        )
+        self.register_optimizer(
+            OptimizerConfig(name="actor_optimizer", networks=["actors"], lr_name="lr_actor")
+        )
+        self.register_optimizer(
+            OptimizerConfig(name="critic_optimizer", networks=["critics"], lr_name="lr_critic")
+        )
+        self.register_mutation_hook("_sync_targets_after_mutation")
+
+    # ------------------------------------------------------------------
+    @staticmethod
+    def _raw_action_dim(space: Space) -> int:
+        return space.n if isinstance(space, Discrete) else flatdim(space)
+
+    def _sync_targets_after_mutation(self) -> None:
+        self.actor_targets.load_state_dict(self.actors.state_dict())
+        self.critic_targets.load_state_dict(self.critics.state_dict())
+        for net in (self.actor_targets, self.critic_targets):
+            for p in net.parameters():
+                p.requires_grad = False
+
+    # ------------------------------------------------------------------
+    def get_action(
+        self, obs: Dict[str, np.ndarray], training: bool = True, **kwargs
+    ) -> Tuple[Dict[str, np.ndarray], Dict[str, np.ndarray]]:
+        """Returns (env_actions, raw_actions).  Raw actions are what the
+        centralized critics consume (one-hot / continuous vectors)."""
+        env_actions, raw_actions = {}, {}
+        with torch.no_grad():
+            for aid in self.agent_ids:
+                actor = self.actors[aid]
+                actor.train(training)
+                out = actor(actor.preprocess(obs[aid]))
+                space = self.action_spaces[aid]
+                if isinstance(space, Discrete):
+                    raw = out  # gumbel-softmax one-hot (train) / hard one-hot (eval)
+                    env_actions[aid] = raw.argmax(-1).cpu().numpy()
+                else:
+                    raw = out
+                    a = out.cpu().numpy()
+                    if training and self.expl_noise > 0:
+                        a = a + np.random.normal(0, self.expl_noise, a.shape)
+                    env_actions[aid] = np.clip(a, space.low, space.high)
+                    raw = torch.as_tensor(env_actions[aid], device=out.device, dtype=out.dtype)
+                raw_actions[aid] = raw.cpu().numpy()
+        return env_actions, raw_actions
+
+    # ------------------------------------------------------------------
+    def _joint(self, d: Dict[str, torch.Tensor]) -> torch.Tensor:
+        return torch.cat([d[aid].reshape(d[aid].shape[0], -1) for aid in self.agent_ids], dim=1)
+
+    def _to_dev(self, d) -> Dict[str, torch.Tensor]:
+        return {
+            aid: torch.as_tensor(np.asarray(v)).float().to(self.device) for aid, v in d.items()
+        }
+
+    def learn(self, experiences: Dict[str, Dict[str, torch.Tensor]]) -> float:
+        obs = self._to_dev(experiences["obs"])
+        actions = self._to_dev(experiences["action"])
+        rewards = self._to_dev(experiences["reward"])
+        next_obs = self._to_dev(experiences["next_obs"])
+        dones = self._to_dev(experiences["done"])
+
+        joint_obs = self._joint(obs)
+        joint_actions = self._joint(actions)
+        with torch.no_grad():
+            next_raw = {}
+            for aid in self.agent_ids:
+                tgt = self.actor_targets[aid]
+                tgt.train()
+                next_raw[aid] = tgt(tgt.preprocess(next_obs[aid]))
+            joint_next_obs = self._joint(next_obs)
+            joint_next_actions = self._joint(next_raw)
+
+        # --- critics: one summed backward over all agents
+        critic_loss = 0.0
+        for aid in self.agent_ids:
+            with torch.no_grad():
+                q_next = self.critic_targets[aid](
+                    self.critic_targets[aid].preprocess(joint_next_obs), joint_next_actions
+                )
+                y = rewards[aid].reshape(-1, 1) + self.gamma * (
+                    1.0 - dones[aid].reshape(-1, 1)
+                ) * q_next
+            q = self.critics[aid](self.critics[aid].preprocess(joint_obs), joint_actions)
+            critic_loss = critic_loss + F.mse_loss(q, y)
+        self.critic_optimizer.zero_grad()
+        critic_loss.backward()
+        self.critic_optimizer.step()
+
+        # --- actors: each agent's action column replaced by its live actor
+        actor_loss = 0.0
+        current_raw = {}
+        for aid in self.agent_ids:
+            actor = self.actors[aid]
+            actor.train()
+            current_raw[aid] = actor(actor.preprocess(obs[aid]))
+        for aid in self.agent_ids:
+            cols = [
+                current_raw[a] if a == aid else actions[a].reshape(actions[a].shape[0], -1)
+                for a in self.agent_ids
+            ]
+            joint_a = torch.cat([c.reshape(c.shape[0], -1) for c in cols], dim=1)
+            q = self.critics[aid](self.critics[aid].preprocess(joint_obs), joint_a)
+            actor_loss = actor_loss + (-q.mean())
+        self.actor_optimizer.zero_grad()
+        actor_loss.backward()
+        self.actor_optimizer.step()
+
+        self.soft_update()
+        return float(critic_loss.detach()) / self.n_agents
+
+    def soft_update(self) -> None:
+        ops.polyak_update_(
+            list(self.actor_targets.parameters()) + list(self.critic_targets.parameters()),
+            list(self.actors.parameters()) + list(self.critics.parameters()),
+            self.tau,
+        )
+
+    # ------------------------------------------------------------------
+    def test(self, env, max_steps: Optional[int] = None, loop: int = 3, **kwargs) -> float:
+        with torch.no_grad():
+            totals = []
+            for _ in range(loop):
+                obs, _ = env.reset()
+                ep_rew = np.zeros(env.num_envs)
+                steps = 0
+                while True:
+                    env_actions, _ = self.get_action(obs, training=False)
+                    obs, rewards, term, trunc, _ = env.step(env_actions)
+                    ep_rew += np.mean([rewards[a] for a in self.agent_ids], axis=0)
+                    steps += 1
+                    done = np.any([term[a] | trunc[a] for a in self.agent_ids], axis=0)
+                    if done.all() or (max_steps is not None and steps >= max_steps):
+                        break
+                totals.append(ep_rew.mean())
+        fitness = float(np.mean(totals))
+        self.fitness.append(fitness)
+        return fitness

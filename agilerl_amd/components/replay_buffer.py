@@ -75,11 +75,9 @@ class ReplayBuffer:
     def _coerce(self, data: Dict[str, Any]) -> Tuple[Dict[str, Any], int]:
         """Convert to tensors, normalize to a leading batch dim."""
         data = {k: to_tensor(v) for k, v in data.items() if v is not None}
-        ref = data.get("reward")
-        if ref is None:
-            ref = next(iter(data.values()))
-            while isinstance(ref, dict):
-                ref = next(iter(ref.values()))
+        ref = data.get("reward", next(iter(data.values())))
+        while isinstance(ref, dict):
+            ref = next(iter(ref.values()))
         if ref.dim() == 0:
             data = tree_map(lambda t: t.unsqueeze(0), data)
             batch = 1

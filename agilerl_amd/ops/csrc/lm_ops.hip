@@ -32,35 +32,39 @@ __global__ void row_lse_gather_kernel(
   const float* row = logits + (long)r * V;
   int tid = threadIdx.x;
   int nthreads = blockDim.x;
+  int nwaves = nthreads / WAVE;
+  int wid = tid / WAVE;
+  int lane = tid & (WAVE - 1);
+  __shared__ float red[16];
+  __shared__ float bcast;
 
   // pass 1: max
   float m = -INFINITY;
   for (long j = tid; j < V; j += nthreads) m = fmaxf(m, row[j] * inv_temp);
-  __shared__ float red[16];
-  // wave reduce
   #pragma unroll
   for (int off = WAVE / 2; off > 0; off >>= 1) m = fmaxf(m, __shfl_down(m, off));
-  if ((tid & (WAVE - 1)) == 0) red[tid / WAVE] = m;
+  if (lane == 0) red[wid] = m;
   __syncthreads();
-  if (tid < nthreads / WAVE) m = red[tid]; else m = -INFINITY;
-  #pragma unroll
-  for (int off = 8; off > 0; off >>= 1) m = fmaxf(m, __shfl_down(m, off));
-  m = __shfl(m, 0);
+  if (tid == 0) {
+    float mm = red[0];
+    for (int w = 1; w < nwaves; ++w) mm = fmaxf(mm, red[w]);
+    bcast = mm;
+  }
+  __syncthreads();
+  m = bcast;
+  __syncthreads();  // red[] reused below
 
   // pass 2: sum of exp
   float s = 0.f;
   for (long j = tid; j < V; j += nthreads) s += __expf(row[j] * inv_temp - m);
   #pragma unroll
   for (int off = WAVE / 2; off > 0; off >>= 1) s += __shfl_down(s, off);
-  if ((tid & (WAVE - 1)) == 0) red[tid / WAVE] = s;
+  if (lane == 0) red[wid] = s;
   __syncthreads();
-  if (tid < nthreads / WAVE) s = red[tid]; else s = 0.f;
-  #pragma unroll
-  for (int off = 8; off > 0; off >>= 1) s += __shfl_down(s, off);
-  s = __shfl(s, 0);
-
   if (tid == 0) {
-    float l = m + __logf(s);
+    float ss = 0.f;
+    for (int w = 0; w < nwaves; ++w) ss += red[w];
+    float l = m + __logf(ss);
     lse[r] = l;
     logprob[r] = row[targets[r]] * inv_temp - l;
   }
