@@ -7,6 +7,8 @@ from .ppo import PPO
 from .maddpg import MADDPG
 from .matd3 import MATD3
 from .ippo import IPPO
+from .neural_ucb import NeuralUCB
+from .neural_ts import NeuralTS
 
 __all__ = [
     "DQN",
@@ -18,4 +20,6 @@ __all__ = [
     "MADDPG",
     "MATD3",
     "IPPO",
+    "NeuralUCB",
+    "NeuralTS",
 ]

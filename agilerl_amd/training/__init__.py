@@ -1,0 +1,17 @@
+from .trainer import Trainer, LocalTrainer
+from .train_off_policy import train_off_policy, save_population_checkpoint
+from .train_on_policy import train_on_policy
+from .train_multi_agent_off_policy import train_multi_agent_off_policy
+from .train_multi_agent_on_policy import train_multi_agent_on_policy
+from .train_bandits import train_bandits
+
+__all__ = [
+    "Trainer",
+    "LocalTrainer",
+    "train_off_policy",
+    "train_on_policy",
+    "train_multi_agent_off_policy",
+    "train_multi_agent_on_policy",
+    "train_bandits",
+    "save_population_checkpoint",
+]

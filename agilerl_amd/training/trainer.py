@@ -1,0 +1,270 @@
+"""LocalTrainer: manifest -> env + population + HPO -> workload loop.
+
+Reference parity: ``agilerl/training/trainer.py:102`` (Trainer,
+``from_manifest`` :250, ``train`` :301; LocalTrainer :318 resolves env,
+population, buffers, Mutations and selection strategy then dispatches to
+the per-workload loop :768-870).
+"""
+
+from __future__ import annotations
+
+import importlib
+from typing import Any, Dict, List, Optional, Union
+
+import numpy as np
+
+from ..components.replay_buffer import MultiStepReplayBuffer, PrioritizedReplayBuffer, ReplayBuffer
+from ..envs.registry import make_vect_envs
+from ..hpo.mutation import Mutations
+from ..hpo.tournament import TournamentSelection
+from ..models.manifest import (
+    ALGO_REGISTRY,
+    TrainingManifest,
+    algo_workload,
+    resolve_algo_class,
+)
+
+__all__ = ["Trainer", "LocalTrainer"]
+
+PZ_REGISTRY = {
+    "simple_speaker_listener_v4": "agilerl_amd.envs.mpe.SpeakerListenerVecEnv",
+    "simple_speaker_listener": "agilerl_amd.envs.mpe.SpeakerListenerVecEnv",
+    "simple_spread_v3": "agilerl_amd.envs.mpe.SimpleSpreadVecEnv",
+    "simple_spread": "agilerl_amd.envs.mpe.SimpleSpreadVecEnv",
+}
+
+
+def _import_path(path: str):
+    module, name = path.rsplit(".", 1)
+    return getattr(importlib.import_module(module), name)
+
+
+class Trainer:
+    """Base trainer: holds a validated manifest."""
+
+    def __init__(self, manifest: TrainingManifest, device: str = "cpu", loggers: Optional[List] = None):
+        self.manifest = manifest
+        self.device = device
+        self.loggers = loggers
+
+    @classmethod
+    def from_manifest(
+        cls,
+        manifest: Union[str, Dict[str, Any], TrainingManifest],
+        device: str = "cpu",
+        **kwargs,
+    ) -> "Trainer":
+        if isinstance(manifest, str):
+            manifest = TrainingManifest.from_yaml(manifest)
+        elif isinstance(manifest, dict):
+            manifest = TrainingManifest.model_validate(manifest)
+        return cls(manifest, device=device, **kwargs)
+
+    def to_manifest(self) -> TrainingManifest:
+        return self.manifest
+
+    def train(self):  # pragma: no cover - interface
+        raise NotImplementedError
+
+
+class LocalTrainer(Trainer):
+    """Runs the training loop on this machine (single- or multi-GPU)."""
+
+    # ------------------------------------------------------------------
+    def _make_env(self):
+        spec = self.manifest.env_spec()
+        if spec.type == "gym" or spec.type == "offline":
+            return make_vect_envs(spec.env_id, num_envs=getattr(spec, "num_envs", 8),
+                                  seed=self.manifest.training.seed,
+                                  **getattr(spec, "env_kwargs", {}))
+        if spec.type == "pettingzoo":
+            env_path = PZ_REGISTRY.get(spec.env_id)
+            if env_path is None:
+                raise KeyError(
+                    f"Unknown multi-agent env '{spec.env_id}'. Known: {sorted(PZ_REGISTRY)}"
+                )
+            env_cls = _import_path(env_path)
+            return env_cls(
+                num_envs=spec.num_envs,
+                seed=self.manifest.training.seed,
+                continuous_actions=spec.continuous_actions,
+                **spec.env_kwargs,
+            )
+        raise NotImplementedError(f"env type {spec.type} is handled by the LLM trainer path")
+
+    def _make_population(self, env) -> List:
+        m = self.manifest
+        algo_cls = resolve_algo_class(m.algorithm.name)
+        workload = algo_workload(m.algorithm.name)
+        hps = dict(m.algorithm.hyperparameters)
+        net = m.network
+        net_config = dict(net.encoder_config) or None
+        if net.arch and net_config is not None:
+            net_config.setdefault("arch", net.arch)
+        elif net.arch:
+            net_config = {"arch": net.arch}
+        kwargs: Dict[str, Any] = dict(hps)
+        if net_config:
+            kwargs["net_config"] = net_config
+        if net.head_config:
+            kwargs["head_config"] = dict(net.head_config)
+        kwargs["latent_dim"] = net.latent_dim
+
+        if workload.startswith("multi_agent"):
+            return algo_cls.population(
+                m.training.pop_size,
+                env.observation_spaces,
+                env.action_spaces,
+                agent_ids=env.agents,
+                device=self.device,
+                **kwargs,
+            )
+        return algo_cls.population(
+            m.training.pop_size,
+            env.single_observation_space,
+            env.single_action_space,
+            device=self.device,
+            **kwargs,
+        )
+
+    def _make_buffer(self) -> ReplayBuffer:
+        spec = self.manifest.replay_buffer
+        if spec.per:
+            return PrioritizedReplayBuffer(
+                spec.max_size, alpha=spec.alpha, device=self.device,
+                storage_device=spec.storage_device, n_step=spec.n_step,
+            )
+        if spec.n_step > 1:
+            return MultiStepReplayBuffer(
+                spec.max_size, n_step=spec.n_step, device=self.device,
+                storage_device=spec.storage_device,
+            )
+        return ReplayBuffer(spec.max_size, device=self.device, storage_device=spec.storage_device)
+
+    def _make_hpo(self):
+        m = self.manifest
+        tournament = TournamentSelection(
+            tournament_size=m.selection_strategy.tournament_size,
+            elitism=m.selection_strategy.elitism,
+        )
+        p = m.mutation.probabilities
+        mutations = Mutations(
+            no_mutation=p.no_mutation,
+            architecture=p.architecture,
+            new_layer_prob=m.mutation.new_layer_prob,
+            parameters=p.parameters,
+            activation=p.activation,
+            rl_hp=p.rl_hp,
+            mutation_sd=m.mutation.mutation_sd,
+            activation_selection=m.mutation.activation_selection,
+            mutate_elite=m.mutation.mutate_elite,
+            rand_seed=m.mutation.rand_seed,
+            device=self.device,
+        )
+        return tournament, mutations
+
+    # ------------------------------------------------------------------
+    def train(self):
+        m = self.manifest
+        workload = algo_workload(m.algorithm.name)
+        if m.training.seed is not None:
+            np.random.seed(m.training.seed)
+            import torch
+
+            torch.manual_seed(m.training.seed)
+
+        if workload.startswith("llm"):
+            return self._train_llm(workload)
+
+        if workload == "bandit":
+            from .train_bandits import train_bandits
+
+            env = self._make_bandit_env()
+            pop = self._make_bandit_population(env)
+            tournament, mutations = self._make_hpo()
+            t = m.training
+            return train_bandits(
+                env, m.environment.get("env_id", "bandit"), m.algorithm.name, pop,
+                max_steps=t.max_steps, evo_steps=t.evo_steps, eval_steps=t.eval_steps,
+                eval_loop=t.eval_loop, target=t.target, tournament=tournament,
+                mutation=mutations, loggers=self.loggers,
+                max_wall_seconds=t.max_wall_seconds,
+            )
+
+        env = self._make_env()
+        pop = self._make_population(env)
+        tournament, mutations = self._make_hpo()
+        t = m.training
+        common = dict(
+            max_steps=t.max_steps,
+            evo_steps=t.evo_steps,
+            eval_steps=t.eval_steps,
+            eval_loop=t.eval_loop,
+            target=t.target,
+            tournament=tournament,
+            mutation=mutations,
+            checkpoint=t.checkpoint,
+            checkpoint_path=t.checkpoint_path,
+            loggers=self.loggers,
+            max_wall_seconds=t.max_wall_seconds,
+        )
+        if workload == "off_policy" or workload == "offline":
+            from .train_off_policy import train_off_policy
+
+            return train_off_policy(
+                env, m.environment.get("env_id", "env"), m.algorithm.name, pop,
+                self._make_buffer(),
+                learning_delay=t.learning_delay,
+                eps_start=t.eps_start, eps_end=t.eps_end, eps_decay=t.eps_decay,
+                save_elite=t.save_elite, elite_path=t.elite_path,
+                overwrite_checkpoints=t.overwrite_checkpoints,
+                **common,
+            )
+        if workload == "on_policy":
+            from .train_on_policy import train_on_policy
+
+            return train_on_policy(
+                env, m.environment.get("env_id", "env"), m.algorithm.name, pop,
+                save_elite=t.save_elite, elite_path=t.elite_path,
+                overwrite_checkpoints=t.overwrite_checkpoints,
+                **common,
+            )
+        if workload == "multi_agent_off_policy":
+            from .train_multi_agent_off_policy import train_multi_agent_off_policy
+
+            return train_multi_agent_off_policy(
+                env, m.environment.get("env_id", "env"), m.algorithm.name, pop,
+                self._make_buffer(), learning_delay=t.learning_delay, **common,
+            )
+        if workload == "multi_agent_on_policy":
+            from .train_multi_agent_on_policy import train_multi_agent_on_policy
+
+            return train_multi_agent_on_policy(
+                env, m.environment.get("env_id", "env"), m.algorithm.name, pop, **common,
+            )
+        raise NotImplementedError(workload)
+
+    # ------------------------------------------------------------------
+    def _make_bandit_env(self):
+        spec = self.manifest.env_spec()
+        from ..envs.bandit import SyntheticBanditEnv
+
+        if spec.context_dim is not None:
+            return SyntheticBanditEnv(
+                context_dim=spec.context_dim, num_arms=spec.num_arms or 4,
+                seed=self.manifest.training.seed, **spec.env_kwargs,
+            )
+        raise KeyError("bandit env requires context_dim/num_arms or a dataset")
+
+    def _make_bandit_population(self, env):
+        m = self.manifest
+        algo_cls = resolve_algo_class(m.algorithm.name)
+        return algo_cls.population(
+            m.training.pop_size, env.observation_space, env.action_space,
+            device=self.device, **dict(m.algorithm.hyperparameters),
+        )
+
+    def _train_llm(self, workload: str):
+        from .llm import run_llm_workload
+
+        return run_llm_workload(self, workload)
