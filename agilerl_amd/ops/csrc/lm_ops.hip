@@ -86,8 +86,10 @@ std::vector<torch::Tensor> row_lse_gather(
 }
 
 // ---------------------------------------------------------------------------
-// Backward of the row softmax-gather: dlogits = (softmax - onehot) * dlp
-// computed in-place over the recomputed logits chunk (saves one (N,V)
+// Backward of the row logprob-gather: the forward emits
+// logp = logits[t]*it - logsumexp(logits*it), so
+// d(logp)/d(logits[j]) = it * (onehot[j==t] - softmax[j]).
+// Computed in-place over the recomputed logits chunk (saves one (N,V)
 // allocation + full re-read).  logits is overwritten with the gradient.
 // ---------------------------------------------------------------------------
 __global__ void row_softmax_bwd_kernel(
@@ -102,7 +104,7 @@ __global__ void row_softmax_bwd_kernel(
     int r = i / V;
     long j = i % V;
     float p = __expf(logits[i] * inv_temp - lse[r]);
-    float g = (p - (j == targets[r] ? 1.f : 0.f)) * grad_logprob[r] * inv_temp;
+    float g = ((j == targets[r] ? 1.f : 0.f) - p) * grad_logprob[r] * inv_temp;
     logits[i] = g;
   }
 }

@@ -1,17 +1,99 @@
-"""LLM fine-tuning workload loops (GRPO-family reasoning, SFT, DPO, multiturn)."""
+"""LLM fine-tuning workload loops + manifest adapter.
+
+Reference parity: ``agilerl/training/llm/`` (reasoning :42, sft :30,
+preference :33, multiturn :43) and the trainer dispatch
+(``trainer.py:768-870``).
+"""
+
+from __future__ import annotations
+
+from typing import Any, Dict
+
+from .reasoning import finetune_llm_reasoning
+from .sft import finetune_llm_sft
+from .preference import finetune_llm_preference
+
+__all__ = [
+    "finetune_llm_reasoning",
+    "finetune_llm_sft",
+    "finetune_llm_preference",
+    "run_llm_workload",
+]
 
 
 def run_llm_workload(trainer, workload: str):
+    """Build env + population from the trainer's manifest and dispatch."""
+    import importlib
+
+    from ...models.manifest import resolve_algo_class
+
+    m = trainer.manifest
+    spec = m.env_spec()
+    algo_cls = resolve_algo_class(m.algorithm.name)
+    hps = dict(m.algorithm.hyperparameters)
+    model_kwargs = {
+        k: hps.pop(k)
+        for k in ("model_config", "model_name_or_path", "lora_config")
+        if k in hps
+    }
+    pop = algo_cls.population(
+        m.training.pop_size, device=trainer.device, **model_kwargs, **hps
+    )
+    vocab_size = pop[0].model.config.vocab_size
+
+    env_kwargs = dict(spec.env_kwargs)
     if workload == "llm_reasoning":
-        from .reasoning import finetune_llm_reasoning
+        from ...llm_envs.reasoning import TokenReasoningGym
 
-        return finetune_llm_reasoning(trainer)
-    if workload == "llm_sft":
-        from .sft import finetune_llm_sft
+        reward_fn = None
+        if spec.reward_fn:
+            mod, fn = spec.reward_fn.rsplit(".", 1)
+            reward_fn = getattr(importlib.import_module(mod), fn)
+        env = TokenReasoningGym(
+            vocab_size=vocab_size,
+            prompt_len=spec.max_prompt_tokens,
+            data_batch_size=spec.data_batch_size,
+            group_size=spec.group_size,
+            reward_fn=reward_fn,
+            **env_kwargs,
+        )
+        loop = finetune_llm_reasoning
+    elif workload == "llm_sft":
+        from ...llm_envs.sft import SyntheticSFTGym
 
-        return finetune_llm_sft(trainer)
-    if workload == "llm_preference":
-        from .preference import finetune_llm_preference
+        env = SyntheticSFTGym(
+            vocab_size=vocab_size,
+            prompt_len=spec.max_prompt_tokens,
+            completion_len=spec.max_completion_tokens,
+            data_batch_size=spec.data_batch_size,
+            **env_kwargs,
+        )
+        loop = finetune_llm_sft
+    elif workload == "llm_preference":
+        from ...llm_envs.preference import SyntheticPreferenceGym
 
-        return finetune_llm_preference(trainer)
-    raise NotImplementedError(workload)
+        env = SyntheticPreferenceGym(
+            vocab_size=vocab_size,
+            prompt_len=spec.max_prompt_tokens,
+            completion_len=spec.max_completion_tokens,
+            data_batch_size=spec.data_batch_size,
+            **env_kwargs,
+        )
+        loop = finetune_llm_preference
+    else:
+        raise NotImplementedError(workload)
+
+    tournament, mutations = trainer._make_hpo()
+    t = m.training
+    return loop(
+        env,
+        pop,
+        max_steps=t.max_steps,
+        evo_steps=t.evo_steps,
+        eval_loop=t.eval_loop,
+        target=t.target,
+        tournament=tournament,
+        mutation=mutations,
+        loggers=trainer.loggers,
+        max_wall_seconds=t.max_wall_seconds,
+    )

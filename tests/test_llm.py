@@ -1,0 +1,297 @@
+"""LLM stack tests: LoRA, fused logprobs, GRPO/SFT/DPO on a tiny random Llama."""
+
+import numpy as np
+import pytest
+import torch
+
+from agilerl_amd import ops
+from agilerl_amd.llm import (
+    LoraConfig,
+    adapter_state_dict,
+    add_adapter,
+    apply_lora,
+    load_adapter_state_dict,
+    set_active_adapter,
+)
+from agilerl_amd.ops.fused_logprobs import fused_linear_logprobs
+from agilerl_amd.ops.grpo_loss import grpo_policy_loss
+
+TINY = dict(
+    model_type="llama", vocab_size=128, hidden_size=64, intermediate_size=128,
+    num_hidden_layers=2, num_attention_heads=4, num_key_value_heads=2,
+    max_position_embeddings=256, pad_token_id=0,
+)
+
+
+def tiny_agent(cls, **kw):
+    kw.setdefault("dtype", torch.float32)
+    kw.setdefault("lora_config", {"r": 4, "lora_alpha": 8})
+    kw.setdefault("micro_batch_size", 4)
+    return cls(model_config=dict(TINY), **kw)
+
+
+class TestLora:
+    def _model(self):
+        import torch.nn as nn
+
+        class M(nn.Module):
+            def __init__(self):
+                super().__init__()
+                self.q_proj = nn.Linear(16, 16)
+                self.other = nn.Linear(16, 16)
+
+            def forward(self, x):
+                return self.q_proj(x) + self.other(x)
+
+        return M()
+
+    def test_wrap_and_zero_delta(self):
+        m = self._model()
+        x = torch.randn(3, 16)
+        y0 = m(x)
+        apply_lora(m, LoraConfig(r=4, target_modules=["q_proj"]), adapters=["actor"])
+        y1 = m(x)
+        torch.testing.assert_close(y0, y1)  # B=0 init -> identity delta
+
+    def test_adapter_isolation(self):
+        m = self._model()
+        apply_lora(m, LoraConfig(r=4, target_modules=["q_proj"]), adapters=["a", "b"])
+        x = torch.randn(3, 16)
+        with torch.no_grad():
+            m.q_proj.lora_B["a"].fill_(1.0)
+        set_active_adapter(m, "a")
+        ya = m(x)
+        set_active_adapter(m, "b")
+        yb = m(x)
+        set_active_adapter(m, None)
+        yn = m(x)
+        assert not torch.allclose(ya, yb)
+        torch.testing.assert_close(yb, yn)  # b still zero-delta
+
+    def test_state_roundtrip(self):
+        m = self._model()
+        apply_lora(m, LoraConfig(r=4, target_modules=["q_proj"]), adapters=["a"])
+        with torch.no_grad():
+            m.q_proj.lora_A["a"].normal_()
+            m.q_proj.lora_B["a"].normal_()
+        state = adapter_state_dict(m, "a")
+        add_adapter(m, "c")
+        load_adapter_state_dict(m, "c", state)
+        torch.testing.assert_close(m.q_proj.lora_A["a"], m.q_proj.lora_A["c"])
+
+    def test_save_load_dir(self, tmp_path):
+        from agilerl_amd.llm import load_adapter, save_adapter
+
+        m = self._model()
+        apply_lora(m, LoraConfig(r=4, target_modules=["q_proj"]), adapters=["a"])
+        with torch.no_grad():
+            m.q_proj.lora_B["a"].normal_()
+        save_adapter(m, "a", str(tmp_path / "ad"))
+        assert (tmp_path / "ad" / "adapter_model.safetensors").exists()
+        assert (tmp_path / "ad" / "adapter_config.json").exists()
+        m2 = self._model()
+        apply_lora(m2, LoraConfig(r=4, target_modules=["q_proj"]), adapters=["a"])
+        load_adapter(m2, "a", str(tmp_path / "ad"))
+        torch.testing.assert_close(m.q_proj.lora_B["a"], m2.q_proj.lora_B["a"])
+
+
+class TestFusedLogprobs:
+    def test_matches_naive(self):
+        N, H, V = 37, 16, 50
+        hidden = torch.randn(N, H, requires_grad=True)
+        weight = torch.randn(V, H, requires_grad=True)
+        targets = torch.randint(0, V, (N,))
+        lp = fused_linear_logprobs(hidden, weight, targets, chunk_rows=8)
+        logits = hidden @ weight.t()
+        ref = torch.log_softmax(logits, -1).gather(1, targets.unsqueeze(1)).squeeze(1)
+        torch.testing.assert_close(lp, ref, rtol=1e-4, atol=1e-5)
+
+    def test_backward_matches(self):
+        N, H, V = 21, 8, 30
+        hidden = torch.randn(N, H, requires_grad=True)
+        weight = torch.randn(V, H, requires_grad=True)
+        targets = torch.randint(0, V, (N,))
+        g = torch.randn(N)
+        lp = fused_linear_logprobs(hidden, weight, targets, chunk_rows=7)
+        gh, gw = torch.autograd.grad(lp, [hidden, weight], g)
+        h2 = hidden.detach().requires_grad_(True)
+        w2 = weight.detach().requires_grad_(True)
+        ref = torch.log_softmax(h2 @ w2.t(), -1).gather(1, targets.unsqueeze(1)).squeeze(1)
+        rgh, rgw = torch.autograd.grad(ref, [h2, w2], g)
+        torch.testing.assert_close(gh, rgh, rtol=1e-4, atol=1e-5)
+        torch.testing.assert_close(gw, rgw, rtol=1e-4, atol=1e-5)
+
+    def test_temperature(self):
+        N, H, V = 8, 8, 20
+        hidden = torch.randn(N, H)
+        weight = torch.randn(V, H)
+        targets = torch.randint(0, V, (N,))
+        lp = fused_linear_logprobs(hidden, weight, targets, temperature=2.0)
+        ref = torch.log_softmax(hidden @ weight.t() / 2.0, -1).gather(1, targets.unsqueeze(1)).squeeze(1)
+        torch.testing.assert_close(lp, ref, rtol=1e-4, atol=1e-5)
+
+
+class TestGrpoLoss:
+    def test_grad_direction(self):
+        torch.manual_seed(0)
+        logp = torch.randn(4, 6, requires_grad=True)
+        old = logp.detach() + 0.01 * torch.randn(4, 6)
+        adv = torch.ones(4, 6)
+        mask = torch.ones(4, 6)
+        loss = grpo_policy_loss(logp, old, adv, mask)
+        (g,) = torch.autograd.grad(loss, logp)
+        # positive advantage -> gradient pushes logp UP (negative grad of loss)
+        assert (g < 0).all()
+
+    def test_sequence_norm(self):
+        logp = torch.zeros(2, 4, requires_grad=True)
+        old = torch.zeros(2, 4)
+        adv = torch.ones(2, 4)
+        mask = torch.tensor([[1.0, 1, 1, 1], [1, 0, 0, 0]])
+        l_tok = grpo_policy_loss(logp, old, adv, mask, loss_norm="token")
+        l_seq = grpo_policy_loss(logp, old, adv, mask, loss_norm="sequence")
+        assert float(l_tok) == pytest.approx(-1.0)
+        assert float(l_seq) == pytest.approx(-1.0)
+
+
+class TestGRPOAgent:
+    def test_generate_and_learn(self):
+        from agilerl_amd.algorithms.llm.grpo import GRPO
+        from agilerl_amd.llm_envs import TokenReasoningGym, make_grpo_experiences
+
+        agent = tiny_agent(GRPO, group_size=4, lr=1e-3, beta=0.04, max_completion_tokens=8)
+        env = TokenReasoningGym(vocab_size=128, prompt_len=8, data_batch_size=2, group_size=4, seed=0)
+        prompts = env.reset()
+        seqs = agent.get_action(prompts)
+        assert seqs.shape[0] == 8 and seqs.shape[1] > 8
+        rewards = env.score(seqs)
+        exp = make_grpo_experiences(env, seqs, rewards)
+        stats = agent.learn(exp)
+        assert np.isfinite(stats["loss"])
+        assert stats["kl"] >= -1e-5
+
+    def test_clone_shares_base_and_copies_adapter(self):
+        from agilerl_amd.algorithms.llm.grpo import GRPO
+
+        agent = tiny_agent(GRPO)
+        clone = agent.clone(index=1)
+        assert clone.model is agent.model
+        assert clone.adapter_name != agent.adapter_name
+        s1 = adapter_state_dict(agent.model, agent.adapter_name)
+        s2 = adapter_state_dict(agent.model, clone.adapter_name)
+        for k in s1:
+            torch.testing.assert_close(s1[k], s2[k])
+
+    def test_population_and_cleanup(self):
+        from agilerl_amd.algorithms.llm.grpo import GRPO
+
+        pop = GRPO.population(3, model_config=dict(TINY), dtype=torch.float32,
+                              lora_config={"r": 4}, group_size=2)
+        assert all(p.model is pop[0].model for p in pop)
+        names = {p.adapter_name for p in pop}
+        assert len(names) == 3
+        pop[2].clean_up()
+        from agilerl_amd.llm.lora import iter_lora_modules
+
+        for _, mod in iter_lora_modules(pop[0].model):
+            assert pop[2].adapter_name not in mod.lora_A
+
+    def test_checkpoint_dir(self, tmp_path):
+        from agilerl_amd.algorithms.llm.grpo import GRPO
+
+        agent = tiny_agent(GRPO)
+        with torch.no_grad():
+            for p in agent.policy_network.parameters():
+                p.normal_()
+        path = str(tmp_path / "llm_ckpt")
+        agent.save_checkpoint(path)
+        import os
+
+        assert os.path.exists(os.path.join(path, "actor", "adapter_model.safetensors"))
+        agent2 = tiny_agent(GRPO)
+        agent2.load_checkpoint(path)
+        s1 = adapter_state_dict(agent.model, agent.adapter_name)
+        s2 = adapter_state_dict(agent2.model, agent2.adapter_name)
+        for k in s1:
+            torch.testing.assert_close(s1[k], s2[k])
+
+    def test_cispo_and_gspo(self):
+        from agilerl_amd.algorithms.llm.cispo import CISPO
+        from agilerl_amd.algorithms.llm.gspo import GSPO
+        from agilerl_amd.llm_envs import TokenReasoningGym, make_grpo_experiences
+
+        for cls in (CISPO, GSPO):
+            agent = tiny_agent(cls, group_size=2, max_completion_tokens=4)
+            env = TokenReasoningGym(vocab_size=128, prompt_len=6, data_batch_size=2, group_size=2)
+            prompts = env.reset()
+            seqs = agent.get_action(prompts)
+            exp = make_grpo_experiences(env, seqs, env.score(seqs))
+            stats = agent.learn(exp)
+            assert np.isfinite(stats["loss"])
+
+
+class TestSFTAndDPO:
+    def test_sft_reduces_loss(self):
+        from agilerl_amd.algorithms.llm.sft import SFT
+        from agilerl_amd.llm_envs import SyntheticSFTGym
+
+        agent = tiny_agent(SFT, lr=5e-3)
+        env = SyntheticSFTGym(vocab_size=128, prompt_len=6, completion_len=6,
+                              data_batch_size=8, seed=0)
+        first = agent.learn(env.sample())["loss"]
+        for _ in range(15):
+            last = agent.learn(env.sample())["loss"]
+        assert last < first
+
+    def test_dpo_margin_grows(self):
+        from agilerl_amd.algorithms.llm.dpo import DPO
+        from agilerl_amd.llm_envs import SyntheticPreferenceGym
+
+        agent = tiny_agent(DPO, lr=5e-3, beta=0.5)
+        env = SyntheticPreferenceGym(vocab_size=128, prompt_len=6, completion_len=6,
+                                     data_batch_size=8, seed=0)
+        for _ in range(10):
+            stats = agent.learn(env.sample())
+        assert stats["margin"] > 0
+        acc = agent.test(env)
+        assert acc > 0.5
+
+
+class TestLLMTrainingLoop:
+    def test_reasoning_loop_with_evolution(self):
+        from agilerl_amd.algorithms.llm.grpo import GRPO
+        from agilerl_amd.hpo import Mutations, TournamentSelection
+        from agilerl_amd.llm_envs import TokenReasoningGym
+        from agilerl_amd.training.llm import finetune_llm_reasoning
+
+        pop = GRPO.population(2, model_config=dict(TINY), dtype=torch.float32,
+                              lora_config={"r": 4}, group_size=2,
+                              max_completion_tokens=4, lr=1e-3)
+        env = TokenReasoningGym(vocab_size=128, prompt_len=6, data_batch_size=2, group_size=2)
+        agents, hist = finetune_llm_reasoning(
+            env, pop, max_steps=4, evo_steps=2, eval_loop=1,
+            tournament=TournamentSelection(2, True),
+            mutation=Mutations(no_mutation=0.5, architecture=0.0, parameters=0.0,
+                               activation=0.0, rl_hp=0.5, rand_seed=0),
+            verbose=False,
+        )
+        assert len(agents) == 2
+        assert len(hist) >= 1
+
+    def test_llm_manifest(self):
+        from agilerl_amd.training.trainer import LocalTrainer
+
+        m = {
+            "algorithm": {"name": "GRPO", "hyperparameters": {
+                "model_config": dict(TINY), "dtype": "float32",
+                "lora_config": {"r": 4}, "group_size": 2,
+                "max_completion_tokens": 4, "lr": 1e-3,
+            }},
+            "environment": {"type": "llm", "env_type": "reasoning",
+                            "data_batch_size": 2, "group_size": 2, "max_prompt_tokens": 6},
+            "training": {"max_steps": 2, "pop_size": 1, "evo_steps": 1},
+        }
+        # dtype strings resolve in the trainer path
+        m["algorithm"]["hyperparameters"]["dtype"] = torch.float32
+        agents, hist = LocalTrainer.from_manifest(m).train()
+        assert agents[0].algo == "GRPO"

@@ -1,0 +1,73 @@
+"""GPU LLM-path tests (tiny random Llama, bf16, fused kernels)."""
+
+import numpy as np
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+DEV = "cuda:0"
+
+TINY = dict(
+    model_type="llama", vocab_size=512, hidden_size=128, intermediate_size=256,
+    num_hidden_layers=2, num_attention_heads=4, num_key_value_heads=2,
+    max_position_embeddings=512, pad_token_id=0,
+)
+
+
+class TestFusedLogprobsGpu:
+    def test_bf16_matches_fp32_reference(self):
+        from agilerl_amd.ops.fused_logprobs import fused_linear_logprobs
+
+        N, H, V = 64, 128, 32000
+        hidden = torch.randn(N, H, device=DEV, dtype=torch.bfloat16, requires_grad=True)
+        weight = torch.randn(V, H, device=DEV, dtype=torch.bfloat16, requires_grad=True)
+        targets = torch.randint(0, V, (N,), device=DEV)
+        lp = fused_linear_logprobs(hidden, weight, targets, chunk_rows=32)
+        ref = torch.log_softmax((hidden.float() @ weight.float().t()), -1).gather(
+            1, targets.unsqueeze(1)
+        ).squeeze(1)
+        torch.testing.assert_close(lp, ref, rtol=5e-2, atol=5e-2)
+
+        g = torch.randn(N, device=DEV)
+        gh, gw = torch.autograd.grad(lp, [hidden, weight], g, retain_graph=True)
+        h2 = hidden.detach().float().requires_grad_(True)
+        w2 = weight.detach().float().requires_grad_(True)
+        ref2 = torch.log_softmax(h2 @ w2.t(), -1).gather(1, targets.unsqueeze(1)).squeeze(1)
+        rgh, rgw = torch.autograd.grad(ref2, [h2, w2], g)
+        torch.testing.assert_close(gh.float(), rgh, rtol=0.1, atol=0.1)
+
+
+class TestGRPOGpu:
+    def test_grpo_learn_bf16(self):
+        from agilerl_amd.algorithms.llm.grpo import GRPO
+        from agilerl_amd.llm_envs import TokenReasoningGym, make_grpo_experiences
+
+        agent = GRPO(
+            model_config=dict(TINY), dtype=torch.bfloat16,
+            lora_config={"r": 4}, group_size=4, micro_batch_size=8,
+            beta=0.04, lr=1e-4, max_completion_tokens=16, device=DEV,
+        )
+        env = TokenReasoningGym(vocab_size=512, prompt_len=16, data_batch_size=2, group_size=4)
+        prompts = env.reset()
+        seqs = agent.get_action(prompts)
+        rewards = env.score(seqs)
+        exp = make_grpo_experiences(env, seqs, rewards)
+        stats = agent.learn(exp)
+        assert np.isfinite(stats["loss"])
+        # adapter params actually moved
+        norms = [float(p.float().norm()) for p in agent.policy_network.parameters()]
+        assert any(n > 0 for n in norms)
+
+    def test_sft_gpu(self):
+        from agilerl_amd.algorithms.llm.sft import SFT
+        from agilerl_amd.llm_envs import SyntheticSFTGym
+
+        agent = SFT(model_config=dict(TINY), dtype=torch.bfloat16,
+                    lora_config={"r": 4}, micro_batch_size=8, lr=1e-3, device=DEV)
+        env = SyntheticSFTGym(vocab_size=512, prompt_len=8, completion_len=8,
+                              data_batch_size=16, seed=0)
+        first = agent.learn(env.sample())["loss"]
+        for _ in range(10):
+            last = agent.learn(env.sample())["loss"]
+        assert last < first
