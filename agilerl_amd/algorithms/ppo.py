@@ -139,6 +139,17 @@ class PPO(RLAlgorithm):
         with torch.no_grad():
             return self.critic(self.critic.preprocess(obs)).squeeze(-1)
 
+    def get_action_device(self, obs_t: torch.Tensor, training: bool = True):
+        """Device-native action path (TorchVecEnv): tensors in, tensors out —
+        no host round-trips in the collect loop."""
+        with torch.no_grad():
+            pre = self.actor.preprocess(obs_t)
+            if not training:
+                return self.actor.deterministic_action(pre)
+            action, log_prob, entropy = self.actor.sample(pre)
+            value = self.critic(self.critic.preprocess(obs_t)).squeeze(-1)
+        return action, log_prob, entropy, value
+
     # ------------------------------------------------------------------
     def learn(self, rollout: RolloutBuffer) -> Dict[str, float]:
         assert rollout.advantages is not None, "call compute_returns_and_advantages first"

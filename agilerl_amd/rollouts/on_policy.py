@@ -17,7 +17,7 @@ import torch
 
 from ..components.rollout_buffer import RolloutBuffer
 
-__all__ = ["collect_rollouts", "collect_rollouts_recurrent"]
+__all__ = ["collect_rollouts", "collect_rollouts_device", "collect_rollouts_recurrent"]
 
 
 def collect_rollouts(
@@ -60,6 +60,50 @@ def collect_rollouts(
     last_value = agent.get_values(obs)
     buffer.compute_returns_and_advantages(last_value, torch.as_tensor(done, dtype=torch.float32))
     stats = {"mean_episode_return": float(np.mean(ep_returns))} if ep_returns else {}
+    return obs, done, stats
+
+
+def collect_rollouts_device(
+    agent,
+    env,
+    buffer: RolloutBuffer,
+    n_steps: int,
+    obs: Optional[torch.Tensor] = None,
+    done: Optional[torch.Tensor] = None,
+) -> Tuple[torch.Tensor, torch.Tensor, Dict[str, float]]:
+    """Device-native collection for ``TorchVecEnv`` (obs/actions/rewards
+    never leave HBM; one host sync per rollout for the episode stats)."""
+    if obs is None:
+        obs, _ = env.reset()
+        done = torch.zeros(env.num_envs, dtype=torch.bool, device=obs.device)
+    buffer.reset()
+    ep_sum = torch.zeros((), device=obs.device)
+    ep_cnt = torch.zeros((), device=obs.device)
+    for _ in range(n_steps):
+        action, log_prob, _entropy, value = agent.get_action_device(obs)
+        next_obs, reward, term, trunc, info = env.step(action)
+        # branchless truncation bootstrap: V(final_obs) only credited on trunc rows
+        v_final = agent.get_values(info["final_observation"])
+        reward = torch.where(trunc, reward + agent.gamma * v_final, reward)
+        done_t = info["done_mask"]
+        buffer.add(
+            obs=obs,
+            action=action,
+            reward=reward,
+            done=done_t.float(),
+            value=value,
+            log_prob=log_prob,
+        )
+        ep_sum = ep_sum + (info["episode_return"] * done_t.float()).sum()
+        ep_cnt = ep_cnt + done_t.float().sum()
+        obs = next_obs
+        done = done_t
+    last_value = agent.get_values(obs)
+    buffer.compute_returns_and_advantages(last_value, done.float())
+    stats = {}
+    cnt = float(ep_cnt)
+    if cnt > 0:
+        stats["mean_episode_return"] = float(ep_sum) / cnt
     return obs, done, stats
 
 

@@ -34,7 +34,7 @@ from agilerl_amd.components.rollout_buffer import RolloutBuffer  # noqa: E402
 from agilerl_amd.envs import LunarLanderVecEnv  # noqa: E402
 from agilerl_amd.hpo import Mutations, TournamentSelection  # noqa: E402
 from agilerl_amd.parallel import DistributedPopulation, DistributedState, barrier  # noqa: E402
-from agilerl_amd.rollouts.on_policy import collect_rollouts  # noqa: E402
+from agilerl_amd.rollouts.on_policy import collect_rollouts, collect_rollouts_device  # noqa: E402
 
 POP_SIZE = 8
 NUM_ENVS = 64
@@ -100,7 +100,15 @@ class BenchRunner:
         self.step_count = 0
 
     def _init_slot(self, slot: int) -> None:
-        self.envs[slot] = LunarLanderVecEnv(self.args.num_envs, seed=1000 + slot)
+        if torch.cuda.is_available():
+            # device-resident env: the collect loop never leaves HBM
+            from agilerl_amd.envs.torch_envs import LunarLanderTorchVecEnv
+
+            self.envs[slot] = LunarLanderTorchVecEnv(
+                self.args.num_envs, device=self.device, seed=1000 + slot
+            )
+        else:
+            self.envs[slot] = LunarLanderVecEnv(self.args.num_envs, seed=1000 + slot)
         agent = self.pop.agents[slot]
         self.buffers[slot] = RolloutBuffer(
             capacity=agent.learn_step, num_envs=self.args.num_envs,
@@ -123,7 +131,10 @@ class BenchRunner:
                 self.buffers[slot] = buffer
             obs, done = self.carried[slot]
             env = self.envs[slot]
-            obs, done, stats = collect_rollouts(agent, env, buffer, agent.learn_step, obs, done)
+            collect = (
+                collect_rollouts_device if getattr(env, "is_torch", False) else collect_rollouts
+            )
+            obs, done, stats = collect(agent, env, buffer, agent.learn_step, obs, done)
             self.carried[slot] = (obs, done)
             agent.learn(buffer)
             n = agent.learn_step * self.args.num_envs
