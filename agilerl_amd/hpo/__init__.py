@@ -1,4 +1,5 @@
 from .tournament import TournamentSelection
 from .mutation import Mutations
+from .multi_frequency import MultiFrequencySelection
 
-__all__ = ["TournamentSelection", "Mutations"]
+__all__ = ["TournamentSelection", "Mutations", "MultiFrequencySelection"]

@@ -4,6 +4,7 @@ from .train_on_policy import train_on_policy
 from .train_multi_agent_off_policy import train_multi_agent_off_policy
 from .train_multi_agent_on_policy import train_multi_agent_on_policy
 from .train_bandits import train_bandits
+from .train_offline import train_offline, load_transitions_into_buffer
 
 __all__ = [
     "Trainer",
@@ -13,5 +14,7 @@ __all__ = [
     "train_multi_agent_off_policy",
     "train_multi_agent_on_policy",
     "train_bandits",
+    "train_offline",
+    "load_transitions_into_buffer",
     "save_population_checkpoint",
 ]

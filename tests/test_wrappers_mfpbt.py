@@ -1,0 +1,156 @@
+"""Wrapper, MF-PBT and offline-training tests."""
+
+import numpy as np
+import pytest
+import torch
+
+from agilerl_amd.algorithms import CQN, DQN, MADDPG
+from agilerl_amd.envs import CartPoleVecEnv
+from agilerl_amd.envs.mpe import SpeakerListenerVecEnv
+from agilerl_amd.hpo import MultiFrequencySelection, Mutations, TournamentSelection
+from agilerl_amd.spaces import Box, Discrete
+from agilerl_amd.wrappers import AgentWrapper, MakeEvolvable, RSNorm, Skill
+
+
+class TestRSNorm:
+    def test_normalizes_and_proxies(self):
+        agent = DQN(Box(-1, 1, (4,)), Discrete(2))
+        wrapped = RSNorm(agent)
+        assert wrapped.batch_size == agent.batch_size  # attr passthrough
+        obs = np.random.rand(8, 4).astype(np.float32) * 100  # big scale
+        for _ in range(5):
+            wrapped.get_action(obs + np.random.rand(8, 4).astype(np.float32))
+        assert wrapped.rms.count > 1
+        normed = wrapped._norm(obs)
+        assert np.abs(normed).max() <= 10.0
+
+    def test_learn_normalizes_obs(self):
+        agent = DQN(Box(-1, 1, (4,)), Discrete(2))
+        wrapped = RSNorm(agent)
+        wrapped.rms.update(np.random.rand(100, 4) * 50)
+        batch = {
+            "obs": torch.rand(16, 4) * 50,
+            "action": torch.randint(0, 2, (16,)),
+            "reward": torch.rand(16),
+            "next_obs": torch.rand(16, 4) * 50,
+            "done": torch.zeros(16),
+        }
+        loss = wrapped.learn(batch)
+        assert np.isfinite(loss)
+
+    def test_clone_keeps_stats(self):
+        agent = DQN(Box(-1, 1, (4,)), Discrete(2))
+        wrapped = RSNorm(agent)
+        wrapped.rms.update(np.random.rand(50, 4))
+        clone = wrapped.clone(index=3)
+        assert isinstance(clone, RSNorm)
+        np.testing.assert_allclose(clone.rms.mean, wrapped.rms.mean)
+        assert clone.index == 3
+
+    def test_end_to_end_training(self):
+        from agilerl_amd.components import ReplayBuffer
+        from agilerl_amd.training import train_off_policy
+
+        env = CartPoleVecEnv(4, seed=0)
+        pop = [RSNorm(a) for a in DQN.population(2, env.observation_space, env.action_space,
+                                                 batch_size=32)]
+        agents, hist = train_off_policy(
+            env, "cp", "DQN", pop, ReplayBuffer(2000),
+            max_steps=600, evo_steps=200, eval_loop=1,
+            tournament=TournamentSelection(2, True),
+            mutation=Mutations(no_mutation=0.5, architecture=0.2, parameters=0.1,
+                               activation=0.0, rl_hp=0.2, rand_seed=0),
+            verbose=False,
+        )
+        assert all(isinstance(a, RSNorm) for a in agents)
+
+
+class TestSkill:
+    def test_reward_shaping(self):
+        class UpSkill(Skill):
+            def skill_reward(self, obs, reward, terminated, truncated, info):
+                return reward * 2.0, terminated, truncated
+
+        env = UpSkill(CartPoleVecEnv(2, seed=0))
+        env.reset()
+        _, r, _, _, _ = env.step(np.zeros(2, dtype=int))
+        np.testing.assert_allclose(r, 2.0)
+
+
+class TestMakeEvolvable:
+    def test_mlp_conversion(self):
+        net = torch.nn.Sequential(
+            torch.nn.Linear(4, 32), torch.nn.ReLU(), torch.nn.Linear(32, 2)
+        )
+        evo = MakeEvolvable(net, torch.randn(1, 4))
+        x = torch.randn(5, 4)
+        torch.testing.assert_close(evo(x), net(x))
+        evo.add_node(hidden_layer=0, numb_new_nodes=16)
+        assert evo(x).shape == (5, 2)
+
+    def test_conv_fallback(self):
+        net = torch.nn.Sequential(torch.nn.Conv2d(3, 8, 3), torch.nn.Flatten(),
+                                  torch.nn.Linear(8 * 6 * 6, 2))
+        evo = MakeEvolvable(net, torch.randn(1, 3, 8, 8))
+        from agilerl_amd.modules import EvolvableWrapper
+
+        assert isinstance(evo, EvolvableWrapper)
+
+
+class TestMFPBT:
+    def test_frequencies_gate_evolution(self):
+        sel = MultiFrequencySelection(frequencies=(1, 2), rng=np.random.default_rng(0))
+        fits = np.array([0.0, 1.0, 2.0, 3.0])
+        # generation 1: only subpop 0 (slots 0,1) evolves
+        plan1 = sel.compute_plan(fits.copy(), 4)
+        assert plan1[2] == 2 or plan1[2] == 3  # slot may receive migration only
+        # generation 2: both evolve
+        plan2 = sel.compute_plan(fits.copy(), 4)
+        assert len(plan2) == 4
+
+    def test_select_returns_population(self):
+        pop = DQN.population(4, Box(-1, 1, (4,)), Discrete(2))
+        for i, a in enumerate(pop):
+            a.fitness.append(float(i))
+        sel = MultiFrequencySelection(frequencies=(1, 2), rng=np.random.default_rng(0))
+        elite, new_pop = sel.select(pop)
+        assert elite is pop[3]
+        assert len(new_pop) == 4
+        assert [a.index for a in new_pop] == [0, 1, 2, 3]
+
+
+class TestOffline:
+    def test_cqn_offline_training(self):
+        from agilerl_amd.training import train_offline
+
+        env = CartPoleVecEnv(4, seed=0)
+        # gather a random-policy dataset
+        obs_l, act_l, rew_l, next_l, term_l = [], [], [], [], []
+        obs, _ = env.reset()
+        for _ in range(100):
+            a = np.random.randint(0, 2, 4)
+            next_obs, r, te, tr, info = env.step(a)
+            obs_l.append(obs)
+            act_l.append(a)
+            rew_l.append(r)
+            next_l.append(next_obs)
+            term_l.append(te)
+            obs = next_obs
+        dataset = {
+            "observations": np.concatenate(obs_l),
+            "actions": np.concatenate(act_l),
+            "rewards": np.concatenate(rew_l),
+            "next_observations": np.concatenate(next_l),
+            "terminals": np.concatenate(term_l),
+        }
+        pop = CQN.population(2, env.observation_space, env.action_space, batch_size=32,
+                             net_config={"arch": "mlp", "hidden_size": [32]})
+        agents, hist = train_offline(
+            env, "cp", dataset, "CQN", pop,
+            max_steps=100, evo_steps=50, eval_loop=1,
+            tournament=TournamentSelection(2, True),
+            mutation=Mutations(no_mutation=0.6, architecture=0.2, parameters=0.0,
+                               activation=0.0, rl_hp=0.2, rand_seed=0),
+            verbose=False,
+        )
+        assert len(hist) >= 1
