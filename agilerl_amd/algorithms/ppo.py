@@ -151,15 +151,21 @@ class PPO(RLAlgorithm):
         return action, log_prob, entropy, value
 
     # ------------------------------------------------------------------
-    def learn(self, rollout: RolloutBuffer) -> Dict[str, float]:
-        assert rollout.advantages is not None, "call compute_returns_and_advantages first"
+    def learn(self, rollout) -> Dict[str, float]:
+        """``rollout``: a RolloutBuffer, or a flat dict of (T*N, ...) tensors
+        (the hipGraph collector's output) with advantages/returns included."""
+        if isinstance(rollout, dict):
+            minibatches = lambda: self._flat_minibatches(rollout)
+        else:
+            assert rollout.advantages is not None, "call compute_returns_and_advantages first"
+            minibatches = lambda: rollout.get_minibatches(self.batch_size)
         stats = {"policy_loss": 0.0, "value_loss": 0.0, "entropy": 0.0, "approx_kl": 0.0}
         n_updates = 0
         early_stop = False
         for _ in range(self.update_epochs):
             if early_stop:
                 break
-            for mb in rollout.get_minibatches(self.batch_size):
+            for mb in minibatches():
                 loss_stats = self._update_minibatch(mb)
                 for k in stats:
                     stats[k] += loss_stats[k]
@@ -170,6 +176,13 @@ class PPO(RLAlgorithm):
         if n_updates:
             stats = {k: v / n_updates for k, v in stats.items()}
         return stats
+
+    def _flat_minibatches(self, flat: Dict[str, torch.Tensor]):
+        n = flat["advantages"].shape[0]
+        idx = torch.randperm(n, device=flat["advantages"].device)
+        for start in range(0, n, self.batch_size):
+            sel = idx[start : start + self.batch_size]
+            yield {k: v[sel] for k, v in flat.items()}
 
     def _update_minibatch(self, mb: Dict[str, torch.Tensor]) -> Dict[str, float]:
         obs = mb["obs"]

@@ -45,7 +45,10 @@ class TorchVecEnv:
         raise NotImplementedError
 
     def _rand(self, *shape, low=0.0, high=1.0) -> torch.Tensor:
-        u = torch.rand(*shape, generator=self.gen, device=self.device)
+        # graph_safe: hipGraph capture requires the default (graph-registered)
+        # CUDA generator rather than a user Generator object
+        gen = None if getattr(self, "graph_safe", False) else self.gen
+        u = torch.rand(*shape, generator=gen, device=self.device)
         return low + u * (high - low)
 
     # API ----------------------------------------------------------------
@@ -68,14 +71,16 @@ class TorchVecEnv:
         obs = self._obs()
         self._ep_return += reward
         info: Dict = {}
-        # branchless auto-reset: masked reset every step (no host sync)
+        # branchless auto-reset: masked reset every step (no host sync);
+        # all state updates are IN-PLACE so the whole step is
+        # hipGraph-capturable (stable tensor addresses across replays)
         info["final_observation"] = obs
         info["episode_return"] = self._ep_return.clone()
         info["done_mask"] = done
         self._reset_rows(done)
-        self._elapsed = torch.where(done, torch.zeros_like(self._elapsed), self._elapsed)
-        self._ep_return = torch.where(done, torch.zeros_like(self._ep_return), self._ep_return)
-        obs = torch.where(done.unsqueeze(-1), self._obs(), obs) if done.dtype == torch.bool else obs
+        self._elapsed.copy_(torch.where(done, torch.zeros_like(self._elapsed), self._elapsed))
+        self._ep_return.copy_(torch.where(done, torch.zeros_like(self._ep_return), self._ep_return))
+        obs = torch.where(done.unsqueeze(-1), self._obs(), obs)
         return obs, reward, terminated, truncated, info
 
 
@@ -114,9 +119,9 @@ class LunarLanderTorchVecEnv(TorchVecEnv):
         s[:, 4] = self._rand(N, low=-0.15, high=0.15)
         s[:, 5] = self._rand(N, low=-0.3, high=0.3)
         m = mask.unsqueeze(1)
-        self.state = torch.where(m, s, self.state)
-        self.legs = torch.where(mask, torch.zeros_like(self.legs), self.legs)
-        self.prev_shaping = torch.where(mask, self._shaping(), self.prev_shaping)
+        self.state.copy_(torch.where(m, s, self.state))
+        self.legs.copy_(torch.where(mask, torch.zeros_like(self.legs), self.legs))
+        self.prev_shaping.copy_(torch.where(mask, self._shaping(), self.prev_shaping))
 
     def _obs(self) -> torch.Tensor:
         px, py, vx, vy, ang, vang = self.state.unbind(1)
@@ -156,21 +161,21 @@ class LunarLanderTorchVecEnv(TorchVecEnv):
 
         on_ground = py <= self.LEG_Y
         upright = ang.abs() < 0.4
-        self.legs = (on_ground & upright).float()
+        self.legs.copy_((on_ground & upright).float())
         py = py.clamp(min=0.0)
         grounded = py <= 1e-9
         impact_speed = torch.sqrt(vx**2 + vy**2)
 
-        self.state = torch.stack([px, py, vx, vy, ang, vang], dim=1)
+        self.state.copy_(torch.stack([px, py, vx, vy, ang, vang], dim=1))
         shaping = self._shaping()
         reward = shaping - self.prev_shaping
-        self.prev_shaping = shaping
+        self.prev_shaping.copy_(shaping)
         reward = reward - (0.30 * main + 0.03 * (left + right))
 
         vy = torch.where(grounded & (vy < 0), torch.zeros_like(vy), vy)
         vx = torch.where(grounded, vx * 0.8, vx)
         vang = torch.where(grounded, vang * 0.5, vang)
-        self.state = torch.stack([px, py, vx, vy, ang, vang], dim=1)
+        self.state.copy_(torch.stack([px, py, vx, vy, ang, vang], dim=1))
 
         crash = (grounded & (~upright | (impact_speed > 1.5))) | (px.abs() > self.X_WORLD)
         landed = grounded & upright & (impact_speed <= 1.5) & (vang.abs() < 0.3)
@@ -194,7 +199,7 @@ class CartPoleTorchVecEnv(TorchVecEnv):
 
     def _reset_rows(self, mask: torch.Tensor) -> None:
         s = self._rand(self.num_envs, 4, low=-0.05, high=0.05)
-        self.state = torch.where(mask.unsqueeze(1), s, self.state)
+        self.state.copy_(torch.where(mask.unsqueeze(1), s, self.state))
 
     def _obs(self) -> torch.Tensor:
         return self.state.clone()
@@ -210,6 +215,6 @@ class CartPoleTorchVecEnv(TorchVecEnv):
         x_dot = x_dot + 0.02 * xacc
         theta = theta + 0.02 * theta_dot
         theta_dot = theta_dot + 0.02 * thetaacc
-        self.state = torch.stack([x, x_dot, theta, theta_dot], dim=1)
+        self.state.copy_(torch.stack([x, x_dot, theta, theta_dot], dim=1))
         terminated = (x.abs() > 2.4) | (theta.abs() > 12 * 3.14159 / 180)
         return torch.ones(self.num_envs, device=self.device), terminated

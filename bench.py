@@ -58,6 +58,7 @@ def parse_args():
     p.add_argument("--seq-len", type=int, default=1024, help="grpo: prompt+completion length")
     p.add_argument("--grpo-batch", type=int, default=32, help="grpo: sequences per step per rank")
     p.add_argument("--model-size", choices=["8b", "tiny"], default="8b")
+    p.add_argument("--no-graph", action="store_true", help="disable hipGraph collector")
     return p.parse_args()
 
 
@@ -95,6 +96,8 @@ class BenchRunner:
         self.buffers = {}
         self.carried = {}
         self.fit_window = {}
+        self.collectors = {}
+        self.use_graph = torch.cuda.is_available() and not args.no_graph
         for slot in self.pop.local_indices:
             self._init_slot(slot)
         self.step_count = 0
@@ -116,6 +119,11 @@ class BenchRunner:
         )
         self.carried[slot] = (None, None)
         self.fit_window[slot] = []
+        self.collectors.pop(slot, None)
+        if self.use_graph:
+            from agilerl_amd.rollouts.graph_collector import GraphedPPOCollector
+
+            self.collectors[slot] = GraphedPPOCollector(agent, self.envs[slot], agent.learn_step)
 
     def bench_step(self) -> int:
         """One population cycle; returns env steps consumed (this rank)."""
@@ -129,14 +137,24 @@ class BenchRunner:
                     device=self.device, gamma=agent.gamma, gae_lambda=agent.gae_lambda,
                 )
                 self.buffers[slot] = buffer
-            obs, done = self.carried[slot]
             env = self.envs[slot]
-            collect = (
-                collect_rollouts_device if getattr(env, "is_torch", False) else collect_rollouts
-            )
-            obs, done, stats = collect(agent, env, buffer, agent.learn_step, obs, done)
-            self.carried[slot] = (obs, done)
-            agent.learn(buffer)
+            if self.use_graph:
+                collector = self.collectors.get(slot)
+                if collector is None or collector.agent is not agent or collector.n_steps != agent.learn_step:
+                    from agilerl_amd.rollouts.graph_collector import GraphedPPOCollector
+
+                    collector = GraphedPPOCollector(agent, env, agent.learn_step)
+                    self.collectors[slot] = collector
+                flat, stats = collector.collect()
+                agent.learn(flat)
+            else:
+                obs, done = self.carried[slot]
+                collect = (
+                    collect_rollouts_device if getattr(env, "is_torch", False) else collect_rollouts
+                )
+                obs, done, stats = collect(agent, env, buffer, agent.learn_step, obs, done)
+                self.carried[slot] = (obs, done)
+                agent.learn(buffer)
             n = agent.learn_step * self.args.num_envs
             agent.steps[-1] += n
             steps += n
