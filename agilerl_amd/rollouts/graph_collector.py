@@ -139,7 +139,9 @@ class GraphedPPOCollector:
             last_value, agent.gamma, agent.gae_lambda, self.done,
         )
         T, N = self.storage["reward"].shape
-        flat = {k: v.reshape(T * N, *v.shape[2:]) for k, v in self.storage.items()}
+        # clone: the storage is reused by the next collect; returning views
+        # would silently mutate a held rollout (caught by the aliasing test)
+        flat = {k: v.reshape(T * N, *v.shape[2:]).clone() for k, v in self.storage.items()}
         flat["advantages"] = adv.reshape(-1)
         flat["returns"] = ret.reshape(-1)
         stats = {}

@@ -71,13 +71,17 @@ class BenchRunner:
         np.random.seed(seed)
         torch.manual_seed(seed)
 
+        # minibatch sized to the rollout so big env counts keep the GPU in
+        # few large GEMMs instead of hundreds of small update steps
+        mb_size = max(2048, args.num_envs * args.learn_step // 16)
+
         def factory(index: int) -> PPO:
             return PPO(
                 observation_space=LunarLanderVecEnv(1).single_observation_space,
                 action_space=LunarLanderVecEnv(1).single_action_space,
                 index=index,
                 learn_step=args.learn_step,
-                batch_size=2048,
+                batch_size=mb_size,
                 lr=3e-4,
                 update_epochs=4,
                 net_config={"arch": "mlp", "hidden_size": [64, 64]},
