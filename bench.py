@@ -47,7 +47,10 @@ def parse_args():
     p.add_argument("--gpus", type=int, default=1)
     p.add_argument("--steps", type=int, default=8)
     p.add_argument("--warmup", type=int, default=2)
-    p.add_argument("--num-envs", type=int, default=NUM_ENVS)
+    p.add_argument(
+        "--num-envs", type=int, default=None,
+        help="envs per agent (default: 16384 on GPU via the hipGraph collector, 64 on CPU)",
+    )
     p.add_argument("--learn-step", type=int, default=LEARN_STEP)
     p.add_argument("--pop-size", type=int, default=POP_SIZE)
     p.add_argument(
@@ -294,6 +297,8 @@ def run_grpo(args):
 
 def main():
     args = parse_args()
+    if args.num_envs is None:
+        args.num_envs = 16384 if torch.cuda.is_available() else NUM_ENVS
     if args.workload == "grpo":
         return run_grpo(args)
     runner = BenchRunner(args)
