@@ -86,6 +86,12 @@ class PPO(RLAlgorithm):
         self.net_config = net_config
         self.latent_dim = latent_dim
         self.recurrent = recurrent
+        self.sequence_length = 16
+        if recurrent and (net_config is None or net_config.get("arch") != "lstm"):
+            net_config = dict(net_config or {})
+            net_config["arch"] = "lstm"
+            net_config.setdefault("hidden_state_size", 64)
+            self.net_config = net_config
 
         self.actor = StochasticActor(
             observation_space,
@@ -139,6 +145,77 @@ class PPO(RLAlgorithm):
         with torch.no_grad():
             return self.critic(self.critic.preprocess(obs)).squeeze(-1)
 
+    # ------------------------------------------------------------------
+    # Recurrent (BPTT) path
+    # ------------------------------------------------------------------
+    def init_hidden(self, batch_size: int) -> Dict[str, torch.Tensor]:
+        ha, ca = self.actor.initial_hidden(batch_size)
+        hc, cc = self.critic.initial_hidden(batch_size)
+        return {"ha": ha, "ca": ca, "hc": hc, "cc": cc}
+
+    def get_action_recurrent(
+        self, obs, hidden: Dict[str, torch.Tensor], training: bool = True
+    ):
+        with torch.no_grad():
+            head_out, (ha, ca) = self.actor.forward_step(obs, (hidden["ha"], hidden["ca"]))
+            if not training:
+                action = self.actor.dist_layer.mode(head_out)
+                return action.cpu().numpy(), {"ha": ha, "ca": ca, "hc": hidden["hc"], "cc": hidden["cc"]}
+            action, log_prob, _ent = self.actor.dist_layer.sample(head_out)
+            v_out, (hc, cc) = self.critic.forward_step(obs, (hidden["hc"], hidden["cc"]))
+        new_hidden = {"ha": ha, "ca": ca, "hc": hc, "cc": cc}
+        return action.cpu().numpy(), log_prob, v_out.squeeze(-1), new_hidden
+
+    def _learn_recurrent(self, rollout: RolloutBuffer) -> Dict[str, float]:
+        stats = {"policy_loss": 0.0, "value_loss": 0.0, "entropy": 0.0, "approx_kl": 0.0}
+        n = 0
+        L = self.sequence_length
+        for _ in range(self.update_epochs):
+            seq_bs = max(self.batch_size // L, 1)
+            for mb in rollout.get_sequence_minibatches(L, seq_bs):
+                obs_seq = mb["obs"].float()  # (B, L, F)
+                B = obs_seq.shape[0]
+                hs = mb["hidden_state"]
+                h0a = (hs["ha"][:, 0].transpose(0, 1).contiguous(), hs["ca"][:, 0].transpose(0, 1).contiguous())
+                h0c = (hs["hc"][:, 0].transpose(0, 1).contiguous(), hs["cc"][:, 0].transpose(0, 1).contiguous())
+                head_seq = self.actor.forward_sequence(obs_seq, h0a)
+                v_seq = self.critic.forward_sequence(obs_seq, h0c).reshape(B * L)
+                flat_out = head_seq.reshape(B * L, -1)
+                actions = mb["action"].reshape(B * L, *mb["action"].shape[2:])
+                log_prob, entropy = self.actor.dist_layer.log_prob_entropy(flat_out, actions)
+
+                adv = mb["advantages"].reshape(-1)
+                if self.normalize_advantage and adv.numel() > 1:
+                    adv = (adv - adv.mean()) / (adv.std() + 1e-8)
+                old_lp = mb["log_prob"].reshape(-1)
+                returns = mb["returns"].reshape(-1)
+                old_values = mb["value"].reshape(-1)
+                log_ratio = log_prob.reshape(-1) - old_lp
+                ratio = log_ratio.exp()
+                pg = torch.maximum(
+                    -adv * ratio, -adv * ratio.clamp(1 - self.clip_coef, 1 + self.clip_coef)
+                ).mean()
+                if self.clip_vloss:
+                    v_clip = old_values + (v_seq - old_values).clamp(-self.clip_coef, self.clip_coef)
+                    vloss = 0.5 * torch.maximum((v_seq - returns) ** 2, (v_clip - returns) ** 2).mean()
+                else:
+                    vloss = 0.5 * ((v_seq - returns) ** 2).mean()
+                loss = pg + self.vf_coef * vloss - self.ent_coef * entropy.mean()
+                self.optimizer.zero_grad()
+                loss.backward()
+                nn.utils.clip_grad_norm_(
+                    [p for net in (self.actor, self.critic) for p in net.parameters()],
+                    self.max_grad_norm,
+                )
+                self.optimizer.step()
+                with torch.no_grad():
+                    stats["approx_kl"] += float(((ratio - 1) - log_ratio).mean())
+                stats["policy_loss"] += float(pg.detach())
+                stats["value_loss"] += float(vloss.detach())
+                stats["entropy"] += float(entropy.mean().detach())
+                n += 1
+        return {k: v / max(n, 1) for k, v in stats.items()}
+
     def get_action_device(self, obs_t: torch.Tensor, training: bool = True):
         """Device-native action path (TorchVecEnv): tensors in, tensors out —
         no host round-trips in the collect loop."""
@@ -154,6 +231,8 @@ class PPO(RLAlgorithm):
     def learn(self, rollout) -> Dict[str, float]:
         """``rollout``: a RolloutBuffer, or a flat dict of (T*N, ...) tensors
         (the hipGraph collector's output) with advantages/returns included."""
+        if self.recurrent and not isinstance(rollout, dict):
+            return self._learn_recurrent(rollout)
         if isinstance(rollout, dict):
             minibatches = lambda: self._flat_minibatches(rollout)
         else:
@@ -176,6 +255,30 @@ class PPO(RLAlgorithm):
         if n_updates:
             stats = {k: v / n_updates for k, v in stats.items()}
         return stats
+
+    def test(self, env, max_steps: Optional[int] = None, loop: int = 3, **kwargs) -> float:
+        if not self.recurrent:
+            return super().test(env, max_steps=max_steps, loop=loop, **kwargs)
+        with torch.no_grad():
+            rewards = []
+            for _ in range(loop):
+                obs, _ = env.reset()
+                hidden = self.init_hidden(env.num_envs)
+                done_mask = np.zeros(env.num_envs, dtype=bool)
+                ep_rew = np.zeros(env.num_envs)
+                steps = 0
+                while not done_mask.all():
+                    action, hidden = self.get_action_recurrent(obs, hidden, training=False)
+                    obs, rew, term, trunc, _ = env.step(action)
+                    ep_rew += np.asarray(rew) * (~done_mask)
+                    done_mask |= np.asarray(term) | np.asarray(trunc)
+                    steps += 1
+                    if max_steps is not None and steps >= max_steps:
+                        break
+                rewards.append(ep_rew.mean())
+        fitness = float(np.mean(rewards))
+        self.fitness.append(fitness)
+        return fitness
 
     def _flat_minibatches(self, flat: Dict[str, torch.Tensor]):
         n = flat["advantages"].shape[0]

@@ -70,6 +70,15 @@ class EvolvableLSTM(EvolvableModule):
         h = torch.zeros(self.num_layers, batch_size, self.hidden_state_size, device=self.device)
         return h, h.clone()
 
+    def step(
+        self, x: torch.Tensor, hidden: Optional[Tuple[torch.Tensor, torch.Tensor]] = None
+    ) -> Tuple[torch.Tensor, Tuple[torch.Tensor, torch.Tensor]]:
+        """Single-timestep recurrence: x (B, F) -> (features (B, out), hidden)."""
+        if hidden is None:
+            hidden = self.initial_hidden(x.shape[0])
+        out, new_hidden = self.lstm(x.float().unsqueeze(1), hidden)
+        return self.proj(out[:, -1]), new_hidden
+
     @property
     def output_size(self) -> int:
         return self.num_outputs

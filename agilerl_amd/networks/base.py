@@ -148,6 +148,29 @@ class EvolvableNetwork(EvolvableModule):
     def forward(self, obs) -> torch.Tensor:
         return self.head_net(self.extract_features(obs))
 
+    # ------------------------------------------------------------------
+    # Recurrent support (LSTM encoders)
+    # ------------------------------------------------------------------
+    @property
+    def is_recurrent(self) -> bool:
+        return isinstance(self.encoder, EvolvableLSTM)
+
+    def initial_hidden(self, batch_size: int):
+        return self.encoder.initial_hidden(batch_size)
+
+    def forward_step(self, obs, hidden):
+        """One recurrent step: (head_out (B, out), new_hidden)."""
+        pre = self.preprocess(obs)
+        feats, new_hidden = self.encoder.step(pre, hidden)
+        return self.head_net(feats), new_hidden
+
+    def forward_sequence(self, obs_seq: torch.Tensor, hidden0):
+        """BPTT: obs_seq (B, T, F) with initial hidden -> head_out (B, T, out)."""
+        pre = obs_seq.float()
+        feats_seq, _ = self.encoder.forward_sequence(pre, hidden0)
+        B, T = feats_seq.shape[:2]
+        return self.head_net(feats_seq.reshape(B * T, -1)).reshape(B, T, -1)
+
     def reset_noise(self) -> None:
         self.encoder.reset_noise()
         self.head_net.reset_noise()

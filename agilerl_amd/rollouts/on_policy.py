@@ -114,6 +114,44 @@ def collect_rollouts_recurrent(
     n_steps: int,
     obs: Optional[np.ndarray] = None,
     done: Optional[np.ndarray] = None,
+    hidden: Optional[Dict[str, torch.Tensor]] = None,
 ):
-    """Recurrent variant: threads hidden state through collection (BPTT PPO)."""
-    raise NotImplementedError("recurrent rollout collection lands with LSTM PPO")
+    """Recurrent variant: threads LSTM hidden state through collection and
+    stores the per-step hidden for BPTT sequence minibatches (reference
+    rollouts/on_policy.py:243).  Hidden is zeroed for finished env rows."""
+    if obs is None:
+        obs, _ = env.reset()
+        done = np.zeros(env.num_envs, dtype=bool)
+        hidden = agent.init_hidden(env.num_envs)
+    buffer.reset()
+    ep_returns: list = []
+    for _ in range(n_steps):
+        # store pre-step hidden (B, L, H) so sequences can rebuild h0
+        stored_hidden = {k: v.transpose(0, 1).contiguous() for k, v in hidden.items()}
+        action, log_prob, value, hidden = agent.get_action_recurrent(obs, hidden)
+        next_obs, reward, term, trunc, info = env.step(action)
+        done_now = term | trunc
+        buffer.add(
+            obs=obs,
+            action=action,
+            reward=reward,
+            done=done_now.astype(np.float32),
+            value=value,
+            log_prob=log_prob,
+            hidden_state=stored_hidden,
+        )
+        # zero hidden rows for finished episodes
+        if done_now.any():
+            mask = torch.as_tensor(~done_now, dtype=torch.float32, device=hidden["ha"].device)
+            hidden = {k: v * mask.view(1, -1, 1) for k, v in hidden.items()}
+        obs = next_obs
+        done = done_now
+        if "episode_return" in info:
+            ep_returns.extend(np.asarray(info["episode_return"]).tolist())
+    with torch.no_grad():
+        v_out, _ = agent.critic.forward_step(obs, (hidden["hc"], hidden["cc"]))
+    buffer.compute_returns_and_advantages(
+        v_out.squeeze(-1), torch.as_tensor(done, dtype=torch.float32)
+    )
+    stats = {"mean_episode_return": float(np.mean(ep_returns))} if ep_returns else {}
+    return obs, done, hidden, stats

@@ -78,12 +78,20 @@ def train_on_policy(
                 )
                 buffers[id(agent)] = buffer
             steps_this_cycle = 0
-            obs, done = None, None
+            obs, done, hidden = None, None, None
             learn_stats = []
+            recurrent = getattr(agent, "recurrent", False)
             while steps_this_cycle < evo_steps:
-                obs, done, _stats = collect_rollouts(
-                    agent, env, buffer, agent.learn_step, obs, done
-                )
+                if recurrent:
+                    from ..rollouts.on_policy import collect_rollouts_recurrent
+
+                    obs, done, hidden, _stats = collect_rollouts_recurrent(
+                        agent, env, buffer, agent.learn_step, obs, done, hidden
+                    )
+                else:
+                    obs, done, _stats = collect_rollouts(
+                        agent, env, buffer, agent.learn_step, obs, done
+                    )
                 stats = agent.learn(buffer)
                 learn_stats.append(stats)
                 steps_this_cycle += agent.learn_step * num_envs
