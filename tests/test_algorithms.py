@@ -84,13 +84,13 @@ class TestPPOProbe:
         assert probs[0, 0] > 0.9
 
     def test_policy_env(self):
-        env = PolicyEnv(num_envs=8)
+        env = PolicyEnv(num_envs=8, seed=3)
         agent = PPO(env.observation_space, env.action_space, lr=5e-3, learn_step=32,
                     batch_size=64, ent_coef=0.0,
                     net_config={"arch": "mlp", "hidden_size": [32]})
         buf = RolloutBuffer(32, 8, gamma=agent.gamma, gae_lambda=agent.gae_lambda)
         obs = done = None
-        for _ in range(25):
+        for _ in range(40):
             obs, done, _ = collect_rollouts(agent, env, buf, 32, obs, done)
             agent.learn(buf)
         with torch.no_grad():
