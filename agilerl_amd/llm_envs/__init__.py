@@ -2,6 +2,7 @@ from .base import LLMEnvBase, make_grpo_experiences
 from .reasoning import ReasoningGym, TokenReasoningGym
 from .sft import SFTGym, SyntheticSFTGym
 from .preference import PreferenceGym, SyntheticPreferenceGym
+from .multiturn import MultiTurnTokenEnv, TokenGuessEnv, SyncMultiTurnVecEnv
 
 __all__ = [
     "LLMEnvBase",
@@ -12,4 +13,7 @@ __all__ = [
     "SyntheticSFTGym",
     "PreferenceGym",
     "SyntheticPreferenceGym",
+    "MultiTurnTokenEnv",
+    "TokenGuessEnv",
+    "SyncMultiTurnVecEnv",
 ]

@@ -58,6 +58,10 @@ ALGO_REGISTRY: Dict[str, Dict[str, str]] = {
     "CISPO": {"cls": "agilerl_amd.algorithms.llm.cispo.CISPO", "workload": "llm_reasoning"},
     "SFT": {"cls": "agilerl_amd.algorithms.llm.sft.SFT", "workload": "llm_sft"},
     "DPO": {"cls": "agilerl_amd.algorithms.llm.dpo.DPO", "workload": "llm_preference"},
+    "PPOLLM": {"cls": "agilerl_amd.algorithms.llm.ppo_llm.PPOLLM", "workload": "llm_reasoning"},
+    "LLM_PPO": {"cls": "agilerl_amd.algorithms.llm.ppo_llm.PPOLLM", "workload": "llm_reasoning"},
+    "ReinforceLLM": {"cls": "agilerl_amd.algorithms.llm.reinforce_llm.ReinforceLLM", "workload": "llm_reasoning"},
+    "LLM_REINFORCE": {"cls": "agilerl_amd.algorithms.llm.reinforce_llm.ReinforceLLM", "workload": "llm_reasoning"},
 }
 
 

@@ -295,3 +295,84 @@ class TestLLMTrainingLoop:
         m["algorithm"]["hyperparameters"]["dtype"] = torch.float32
         agents, hist = LocalTrainer.from_manifest(m).train()
         assert agents[0].algo == "GRPO"
+
+
+class TestMultiTurn:
+    def test_sync_vec_env_trajectories(self):
+        from agilerl_amd.llm_envs.multiturn import SyncMultiTurnVecEnv, TokenGuessEnv
+
+        env = SyncMultiTurnVecEnv(lambda: TokenGuessEnv(128, prompt_len=6, max_turns=2),
+                                  data_batch_size=2, group_size=2, max_turns=2, seed=0)
+        prompts = env.reset()
+        assert prompts["input_ids"].shape[0] == 4
+        # fake completions: 3 tokens each turn
+        fake = torch.cat([prompts["input_ids"],
+                          torch.randint(1, 128, (4, 3))], dim=1)
+        prompts2, done = env.step(fake)
+        assert not done and prompts2["input_ids"].shape[1] > prompts["input_ids"].shape[1]
+        fake2 = torch.cat([prompts2["input_ids"], torch.randint(1, 128, (4, 3))], dim=1)
+        _, done = env.step(fake2)
+        assert done
+        traj = env.get_trajectories()
+        assert traj["action_mask"].sum() > 0
+        assert traj["turn_ids"].max() == 1
+        assert traj["rewards"].shape == (4,)
+
+    def test_multiturn_rollout_and_learn(self):
+        from agilerl_amd.algorithms.llm.grpo import GRPO
+        from agilerl_amd.llm_envs.multiturn import SyncMultiTurnVecEnv, TokenGuessEnv
+        from agilerl_amd.training.llm.multiturn import rollout_multiturn
+
+        agent = tiny_agent(GRPO, group_size=2, max_completion_tokens=4, lr=1e-3)
+        env = SyncMultiTurnVecEnv(lambda: TokenGuessEnv(128, prompt_len=6, max_turns=2),
+                                  data_batch_size=2, group_size=2, max_turns=2, seed=0)
+        traj = rollout_multiturn(agent, env)
+        stats = agent.learn(traj)
+        assert np.isfinite(stats["loss"])
+
+    def test_multiturn_loop(self):
+        from agilerl_amd.algorithms.llm.grpo import GRPO
+        from agilerl_amd.hpo import Mutations, TournamentSelection
+        from agilerl_amd.llm_envs.multiturn import SyncMultiTurnVecEnv, TokenGuessEnv
+        from agilerl_amd.training.llm.multiturn import finetune_llm_multiturn
+
+        pop = GRPO.population(2, model_config=dict(TINY), dtype=torch.float32,
+                              lora_config={"r": 4}, group_size=2,
+                              max_completion_tokens=4, lr=1e-3)
+        env = SyncMultiTurnVecEnv(lambda: TokenGuessEnv(128, prompt_len=6, max_turns=2),
+                                  data_batch_size=2, group_size=2, max_turns=2, seed=0)
+        agents, hist = finetune_llm_multiturn(
+            env, pop, max_steps=2, evo_steps=1,
+            tournament=TournamentSelection(2, True),
+            mutation=Mutations(no_mutation=0.5, architecture=0, parameters=0,
+                               activation=0, rl_hp=0.5, rand_seed=0),
+            verbose=False)
+        assert len(hist) >= 1
+
+
+class TestPPOLLMAndReinforce:
+    def test_ppollm_learn_and_clone(self):
+        from agilerl_amd.algorithms.llm.ppo_llm import PPOLLM
+        from agilerl_amd.llm_envs import TokenReasoningGym, make_grpo_experiences
+
+        agent = tiny_agent(PPOLLM, lr=1e-3, max_completion_tokens=4)
+        env = TokenReasoningGym(vocab_size=128, prompt_len=6, data_batch_size=4, group_size=1, seed=0)
+        p = env.reset()
+        seqs = agent.get_action(p)
+        stats = agent.learn(make_grpo_experiences(env, seqs, env.score(seqs)))
+        assert np.isfinite(stats["loss"]) and np.isfinite(stats["value_loss"])
+        clone = agent.clone(1)
+        assert clone.value_head is not agent.value_head
+        torch.testing.assert_close(
+            clone.value_head.weight, agent.value_head.weight
+        )
+
+    def test_reinforce_rloo_advantage(self):
+        from agilerl_amd.algorithms.llm.reinforce_llm import ReinforceLLM
+
+        agent = tiny_agent(ReinforceLLM, group_size=4, max_completion_tokens=4)
+        r = torch.tensor([1.0, 0.0, 0.0, 0.0])
+        adv = agent._calculate_advantages(r)
+        # RLOO: a_0 = r_0 - mean(others) = 1 - 0 = 1
+        assert adv[0].item() == pytest.approx(1.0, abs=1e-5)
+        assert adv[1].item() == pytest.approx(-1.0 / 3.0, abs=1e-5)
