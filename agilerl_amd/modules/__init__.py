@@ -22,6 +22,8 @@ from .cnn import EvolvableCNN
 from .lstm import EvolvableLSTM
 from .simba import EvolvableSimBa
 from .multi_input import EvolvableMultiInput
+from .gpt import EvolvableGPT, CausalSelfAttention, GPTBlock
+from .bert import EvolvableBERT
 
 __all__ = [
     "EvolvableModule",
@@ -45,4 +47,8 @@ __all__ = [
     "EvolvableLSTM",
     "EvolvableSimBa",
     "EvolvableMultiInput",
+    "EvolvableGPT",
+    "CausalSelfAttention",
+    "GPTBlock",
+    "EvolvableBERT",
 ]
