@@ -1,0 +1,325 @@
+"""Async (subprocess) vectorization for arbitrary Python environments.
+
+Reference parity: ``agilerl/vector/pz_async_vec_env.py:101``
+(AsyncPettingZooVecEnv: one process per env :173, command pipes,
+shared-memory observation buffers :759, worker loop :938, error
+propagation :573) plus a single-agent ``AsyncVectorEnv`` (the reference
+delegates that case to Gymnasium, which is not a dependency here).
+
+First-party batched envs (``agilerl_amd.envs``) are the fast path; these
+wrappers exist for user-supplied envs that cannot be vectorized natively.
+Observations travel through one shared-memory block per agent key
+(zero-copy reads into a numpy view); commands/results use pipes.
+"""
+
+from __future__ import annotations
+
+import multiprocessing as mp
+import traceback
+from typing import Any, Callable, Dict, List, Optional, Tuple
+
+import numpy as np
+
+__all__ = ["AsyncVectorEnv", "AsyncPettingZooVecEnv", "DummyVecEnv"]
+
+
+def _flat_size(shape) -> int:
+    return int(np.prod(shape)) if shape else 1
+
+
+# ---------------------------------------------------------------------------
+# Workers
+# ---------------------------------------------------------------------------
+
+def _sa_worker(idx, env_fn, pipe, shm, obs_shape, error_queue):
+    """Single-agent worker: writes observations into its shared-memory slot."""
+    try:
+        env = env_fn()
+        n = _flat_size(obs_shape)
+        view = np.frombuffer(shm.get_obj(), dtype=np.float32)[idx * n : (idx + 1) * n]
+        while True:
+            cmd, data = pipe.recv()
+            if cmd == "reset":
+                obs, info = env.reset(seed=data)
+                view[:] = np.asarray(obs, dtype=np.float32).reshape(-1)
+                pipe.send(("ok", info))
+            elif cmd == "step":
+                obs, reward, term, trunc, info = env.step(data)
+                if term or trunc:
+                    info = dict(info or {})
+                    info["final_observation"] = np.asarray(obs, dtype=np.float32)
+                    obs, _ = env.reset()
+                view[:] = np.asarray(obs, dtype=np.float32).reshape(-1)
+                pipe.send(("ok", (float(reward), bool(term), bool(trunc), info)))
+            elif cmd == "close":
+                pipe.send(("ok", None))
+                break
+    except Exception:
+        error_queue.put((idx, traceback.format_exc()))
+        pipe.send(("error", None))
+
+
+def _pz_worker(idx, env_fn, pipe, shms, obs_shapes, error_queue):
+    """PettingZoo-parallel worker: per-agent shared-memory observation slots."""
+    try:
+        env = env_fn()
+        views = {}
+        for agent, shape in obs_shapes.items():
+            n = _flat_size(shape)
+            views[agent] = np.frombuffer(shms[agent].get_obj(), dtype=np.float32)[
+                idx * n : (idx + 1) * n
+            ]
+
+        def write_obs(obs):
+            for agent, shape in obs_shapes.items():
+                if agent in obs and obs[agent] is not None:
+                    views[agent][:] = np.asarray(obs[agent], dtype=np.float32).reshape(-1)
+                else:  # dead/inactive agent placeholder (reference :791-830)
+                    views[agent][:] = 0.0
+
+        while True:
+            cmd, data = pipe.recv()
+            if cmd == "reset":
+                obs, info = env.reset(seed=data)
+                write_obs(obs)
+                pipe.send(("ok", info))
+            elif cmd == "step":
+                obs, rewards, terms, truncs, info = env.step(data)
+                done = all(terms.get(a, False) or truncs.get(a, False) for a in obs_shapes)
+                if done:
+                    obs, _ = env.reset()
+                write_obs(obs)
+                pipe.send(("ok", (rewards, terms, truncs, info)))
+            elif cmd == "close":
+                pipe.send(("ok", None))
+                break
+    except Exception:
+        error_queue.put((idx, traceback.format_exc()))
+        pipe.send(("error", None))
+
+
+# ---------------------------------------------------------------------------
+# Async single-agent
+# ---------------------------------------------------------------------------
+
+class AsyncVectorEnv:
+    def __init__(self, env_fns: List[Callable], context: str = "spawn"):
+        self.num_envs = len(env_fns)
+        ctx = mp.get_context(context)
+        probe = env_fns[0]()
+        self.single_observation_space = probe.observation_space
+        self.single_action_space = probe.action_space
+        obs_shape = tuple(np.asarray(probe.reset()[0]).shape)
+        probe.close() if hasattr(probe, "close") else None
+        self._obs_shape = obs_shape
+        n = _flat_size(obs_shape)
+        self._shm = ctx.Array("f", self.num_envs * n)
+        self._error_queue = ctx.Queue()
+        self._pipes, self._procs = [], []
+        for i, fn in enumerate(env_fns):
+            parent, child = ctx.Pipe()
+            p = ctx.Process(
+                target=_sa_worker, args=(i, fn, child, self._shm, obs_shape, self._error_queue),
+                daemon=True,
+            )
+            p.start()
+            self._pipes.append(parent)
+            self._procs.append(p)
+
+    def _obs_view(self) -> np.ndarray:
+        flat = np.frombuffer(self._shm.get_obj(), dtype=np.float32)
+        return flat.reshape(self.num_envs, *self._obs_shape).copy()
+
+    def _recv_all(self):
+        out = []
+        for pipe in self._pipes:
+            status, payload = pipe.recv()
+            if status == "error":
+                self._raise_errors()
+            out.append(payload)
+        return out
+
+    def _raise_errors(self):
+        errs = []
+        while not self._error_queue.empty():
+            errs.append(self._error_queue.get())
+        self.close()
+        raise RuntimeError(f"async env worker(s) crashed: {errs}")
+
+    def reset(self, seed: Optional[int] = None):
+        for i, pipe in enumerate(self._pipes):
+            pipe.send(("reset", None if seed is None else seed + i))
+        infos = self._recv_all()
+        return self._obs_view(), {}
+
+    def step(self, actions):
+        for pipe, a in zip(self._pipes, np.asarray(actions)):
+            pipe.send(("step", a.item() if np.ndim(a) == 0 else a))
+        results = self._recv_all()
+        rewards = np.array([r[0] for r in results], dtype=np.float32)
+        terms = np.array([r[1] for r in results])
+        truncs = np.array([r[2] for r in results])
+        info: Dict[str, Any] = {}
+        finals = [r[3].get("final_observation") for r in results]
+        if any(f is not None for f in finals):
+            obs_now = self._obs_view()
+            fo = obs_now.copy()
+            for i, f in enumerate(finals):
+                if f is not None:
+                    fo[i] = f
+            info["final_observation"] = fo
+        return self._obs_view(), rewards, terms, truncs, info
+
+    def close(self):
+        for pipe in self._pipes:
+            try:
+                pipe.send(("close", None))
+            except (BrokenPipeError, OSError):
+                pass
+        for p in self._procs:
+            p.join(timeout=5)
+            if p.is_alive():
+                p.terminate()
+
+
+# ---------------------------------------------------------------------------
+# Async PettingZoo-parallel
+# ---------------------------------------------------------------------------
+
+class AsyncPettingZooVecEnv:
+    def __init__(self, env_fns: List[Callable], context: str = "spawn"):
+        self.num_envs = len(env_fns)
+        ctx = mp.get_context(context)
+        probe = env_fns[0]()
+        self.agents = list(probe.possible_agents)
+        self.observation_spaces = {a: probe.observation_space(a) for a in self.agents}
+        self.action_spaces = {a: probe.action_space(a) for a in self.agents}
+        from ..spaces import space_shape
+
+        self._obs_shapes = {a: tuple(space_shape(self.observation_spaces[a])) for a in self.agents}
+        self._shms = {
+            a: ctx.Array("f", self.num_envs * _flat_size(s)) for a, s in self._obs_shapes.items()
+        }
+        self._error_queue = ctx.Queue()
+        self._pipes, self._procs = [], []
+        for i, fn in enumerate(env_fns):
+            parent, child = ctx.Pipe()
+            p = ctx.Process(
+                target=_pz_worker,
+                args=(i, fn, child, self._shms, self._obs_shapes, self._error_queue),
+                daemon=True,
+            )
+            p.start()
+            self._pipes.append(parent)
+            self._procs.append(p)
+
+    @property
+    def possible_agents(self):
+        return self.agents
+
+    def observation_space(self, agent):
+        return self.observation_spaces[agent]
+
+    def action_space(self, agent):
+        return self.action_spaces[agent]
+
+    def _obs_view(self) -> Dict[str, np.ndarray]:
+        out = {}
+        for a, shape in self._obs_shapes.items():
+            flat = np.frombuffer(self._shms[a].get_obj(), dtype=np.float32)
+            out[a] = flat.reshape(self.num_envs, *shape).copy()
+        return out
+
+    def _recv_all(self):
+        out = []
+        for pipe in self._pipes:
+            status, payload = pipe.recv()
+            if status == "error":
+                errs = []
+                while not self._error_queue.empty():
+                    errs.append(self._error_queue.get())
+                self.close()
+                raise RuntimeError(f"async PZ worker(s) crashed: {errs}")
+            out.append(payload)
+        return out
+
+    def reset(self, seed: Optional[int] = None):
+        for i, pipe in enumerate(self._pipes):
+            pipe.send(("reset", None if seed is None else seed + i))
+        self._recv_all()
+        return self._obs_view(), {}
+
+    def step(self, actions: Dict[str, np.ndarray]):
+        for i, pipe in enumerate(self._pipes):
+            per_env = {a: np.asarray(actions[a])[i] for a in actions}
+            per_env = {
+                a: (v.item() if np.ndim(v) == 0 else v) for a, v in per_env.items()
+            }
+            pipe.send(("step", per_env))
+        results = self._recv_all()
+        rewards = {
+            a: np.array([r[0].get(a, 0.0) for r in results], dtype=np.float32)
+            for a in self.agents
+        }
+        terms = {a: np.array([r[1].get(a, False) for r in results]) for a in self.agents}
+        truncs = {a: np.array([r[2].get(a, False) for r in results]) for a in self.agents}
+        return self._obs_view(), rewards, terms, truncs, {}
+
+    def close(self):
+        for pipe in self._pipes:
+            try:
+                pipe.send(("close", None))
+            except (BrokenPipeError, OSError):
+                pass
+        for p in self._procs:
+            p.join(timeout=5)
+            if p.is_alive():
+                p.terminate()
+
+
+# ---------------------------------------------------------------------------
+# Serial fallback (reference dummy_vec_env.py:31/:171)
+# ---------------------------------------------------------------------------
+
+class DummyVecEnv:
+    def __init__(self, env_fns: List[Callable]):
+        self.envs = [fn() for fn in env_fns]
+        self.num_envs = len(self.envs)
+        self.single_observation_space = self.envs[0].observation_space
+        self.single_action_space = self.envs[0].action_space
+
+    def reset(self, seed: Optional[int] = None):
+        obs = []
+        for i, env in enumerate(self.envs):
+            o, _ = env.reset(seed=None if seed is None else seed + i)
+            obs.append(o)
+        return np.stack(obs), {}
+
+    def step(self, actions):
+        obs, rewards, terms, truncs = [], [], [], []
+        final_obs = [None] * self.num_envs
+        for i, (env, a) in enumerate(zip(self.envs, np.asarray(actions))):
+            o, r, te, tr, _ = env.step(a.item() if np.ndim(a) == 0 else a)
+            if te or tr:
+                final_obs[i] = np.asarray(o)
+                o, _ = env.reset()
+            obs.append(o)
+            rewards.append(r)
+            terms.append(te)
+            truncs.append(tr)
+        info = {}
+        if any(f is not None for f in final_obs):
+            fo = np.stack([f if f is not None else o for f, o in zip(final_obs, obs)])
+            info["final_observation"] = fo
+        return (
+            np.stack(obs),
+            np.array(rewards, dtype=np.float32),
+            np.array(terms),
+            np.array(truncs),
+            info,
+        )
+
+    def close(self):
+        for env in self.envs:
+            if hasattr(env, "close"):
+                env.close()
