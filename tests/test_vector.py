@@ -26,6 +26,11 @@ class _ToySingleEnv:
         return obs, float(action), term, False, {}
 
 
+class _BrokenEnv(_ToySingleEnv):
+    def step(self, action):
+        raise ValueError("boom")
+
+
 class _ToyPZEnv:
     possible_agents = ["a0", "a1"]
 
@@ -102,11 +107,7 @@ class TestAsyncPZVecEnv:
     def test_worker_error_propagates(self):
         from agilerl_amd.vector import AsyncVectorEnv
 
-        class Broken(_ToySingleEnv):
-            def step(self, action):
-                raise ValueError("boom")
-
-        env = AsyncVectorEnv([Broken for _ in range(2)])
+        env = AsyncVectorEnv([_BrokenEnv for _ in range(2)])
         env.reset()
         with pytest.raises(RuntimeError, match="crashed"):
             env.step(np.zeros(2, dtype=int))
