@@ -2,6 +2,7 @@ from .base import VecEnv, BatchedVecEnv
 from .cartpole import CartPoleVecEnv
 from .lunar_lander import LunarLanderVecEnv
 from .pendulum import PendulumVecEnv
+from .visual import CatchPongVecEnv
 from .registry import ENV_REGISTRY, make_vect_envs, register_env
 from . import probe
 
@@ -11,6 +12,7 @@ __all__ = [
     "CartPoleVecEnv",
     "LunarLanderVecEnv",
     "PendulumVecEnv",
+    "CatchPongVecEnv",
     "ENV_REGISTRY",
     "make_vect_envs",
     "register_env",

@@ -13,6 +13,7 @@ from .base import VecEnv
 from .cartpole import CartPoleVecEnv
 from .lunar_lander import LunarLanderVecEnv
 from .pendulum import PendulumVecEnv
+from .visual import CatchPongVecEnv
 from .probe import (
     ConstantRewardEnv,
     ConstantRewardContActionsEnv,
@@ -31,6 +32,8 @@ ENV_REGISTRY: Dict[str, Callable[..., VecEnv]] = {
     "LunarLander-v2": LunarLanderVecEnv,
     "LunarLander-v3": LunarLanderVecEnv,
     "Pendulum-v1": PendulumVecEnv,
+    "CatchPong-v0": CatchPongVecEnv,
+    "PongLike-v0": CatchPongVecEnv,
     "probe/ConstantReward": ConstantRewardEnv,
     "probe/ObsDependentReward": ObsDependentRewardEnv,
     "probe/DiscountedReward": DiscountedRewardEnv,

@@ -54,9 +54,12 @@ def parse_args():
     p.add_argument("--learn-step", type=int, default=LEARN_STEP)
     p.add_argument("--pop-size", type=int, default=POP_SIZE)
     p.add_argument(
-        "--workload", choices=["ppo", "grpo"], default="ppo",
-        help="ppo = BASELINE headline (LunarLander pop=8); grpo = BASELINE "
-        "config 5 (Llama-3-8B random-init, DP over ranks, synthetic tokens)",
+        "--workload", choices=["ppo", "grpo", "dqn", "rainbow", "maddpg"], default="ppo",
+        help="BASELINE configs: ppo = headline (LunarLander pop=8); "
+        "dqn = config 1 (CartPole pop=1 CPU plumbing); rainbow = config 3 "
+        "(PER+noisy CNN on the Pong-like visual env, pop=4); maddpg = "
+        "config 4 (speaker_listener pop=8); grpo = config 5 (Llama-3-8B "
+        "random-init, DP, synthetic tokens)",
     )
     p.add_argument("--seq-len", type=int, default=1024, help="grpo: prompt+completion length")
     p.add_argument("--grpo-batch", type=int, default=32, help="grpo: sequences per step per rank")
@@ -295,12 +298,206 @@ def run_grpo(args):
         dist.destroy_process_group()
 
 
+def _emit(state, metric, value, unit, args, elapsed, extra_config, dtype="fp32",
+          scaling="strong", data="synthetic"):
+    if not state.is_main:
+        return
+    print(json.dumps({
+        "metric": metric, "value": value, "unit": unit,
+        "n_gpus": state.world_size, "steps": args.steps, "warmup": args.warmup,
+        "ms_per_step": elapsed / args.steps * 1000.0, "higher_is_better": True,
+        "scaling": scaling, "vs_baseline": None, "dtype": dtype, "data": data,
+        "config": extra_config,
+    }), flush=True)
+
+
+def _timed_loop(state, step_fn, args):
+    use_cuda = torch.cuda.is_available()
+    for _ in range(args.warmup):
+        step_fn()
+    barrier()
+    if use_cuda:
+        torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    local = 0
+    for _ in range(args.steps):
+        local += step_fn()
+    if use_cuda:
+        torch.cuda.synchronize()
+    barrier()
+    elapsed = time.perf_counter() - t0
+    if state.is_distributed:
+        import torch.distributed as dist
+
+        t = torch.tensor([elapsed, float(local)])
+        if state.backend == "nccl":
+            t = t.to(state.device)
+        dist.all_reduce(t[0:1], op=dist.ReduceOp.MAX)
+        dist.all_reduce(t[1:2], op=dist.ReduceOp.SUM)
+        elapsed, total = float(t[0]), float(t[1])
+    else:
+        total = float(local)
+    return elapsed, total
+
+
+def run_dqn(args):
+    """BASELINE config 1: DQN CartPole pop=1 on CPU vectorized envs."""
+    from agilerl_amd.algorithms import DQN
+    from agilerl_amd.components import ReplayBuffer
+    from agilerl_amd.envs import CartPoleVecEnv
+
+    state = DistributedState.get()
+    num_envs = min(args.num_envs or 64, 256)
+    env = CartPoleVecEnv(num_envs, seed=0)
+    agent = DQN(env.observation_space, env.action_space, batch_size=128, lr=1e-3,
+                device="cpu")
+    memory = ReplayBuffer(100_000)
+    obs, _ = env.reset()
+    eps = [1.0]
+
+    def step_fn():
+        nonlocal obs
+        steps = 0
+        for it in range(128):
+            action = agent.get_action(obs, epsilon=eps[0])
+            next_obs, reward, term, trunc, info = env.step(action)
+            memory.add(obs=obs, action=action, reward=reward, next_obs=next_obs,
+                       done=term.astype(np.float32))
+            obs = next_obs
+            eps[0] = max(0.05, eps[0] * 0.999)
+            steps += num_envs
+            if len(memory) >= 256 and it % agent.learn_step == 0:
+                agent.learn(memory.sample(agent.batch_size))
+        return steps
+
+    elapsed, total = _timed_loop(state, step_fn, args)
+    _emit(state, "env_steps_per_sec", total / elapsed, "steps/s", args, elapsed,
+          {"model": "DQN mlp[64,64]", "env": "CartPole-v1 (first-party)",
+           "global_batch": 128, "num_envs": num_envs, "pop_size": 1,
+           "parallelism": "cpu single-agent (plumbing check)"})
+
+
+def run_rainbow(args):
+    """BASELINE config 3 shape: Rainbow-DQN (PER + n-step + noisy + C51 HIP
+    kernels) on the Pong-like visual env, population across ranks."""
+    from agilerl_amd.algorithms import RainbowDQN
+    from agilerl_amd.components import PrioritizedReplayBuffer
+    from agilerl_amd.envs import CatchPongVecEnv
+
+    state = DistributedState.get()
+    device = state.device
+    pop_size = 4
+    num_envs = args.num_envs or (64 if torch.cuda.is_available() else 8)
+
+    def factory(index):
+        return RainbowDQN(
+            CatchPongVecEnv(1).single_observation_space,
+            CatchPongVecEnv(1).single_action_space,
+            index=index, batch_size=256, lr=1e-4, n_step=3,
+            net_config={"arch": "cnn", "channel_size": [32, 64, 64],
+                        "kernel_size": [8, 4, 3], "stride_size": [4, 2, 1]},
+            device=device,
+        )
+
+    pop = DistributedPopulation(factory, pop_size)
+    envs = {s: CatchPongVecEnv(num_envs, seed=100 + s) for s in pop.local_indices}
+    mems = {
+        s: PrioritizedReplayBuffer(200_000, n_step=3, gamma=0.99, device=device)
+        for s in pop.local_indices
+    }
+    obs_map = {s: envs[s].reset()[0] for s in pop.local_indices}
+
+    def step_fn():
+        steps = 0
+        for slot in pop.local_indices:
+            agent, env, mem = pop.agents[slot], envs[slot], mems[slot]
+            obs = obs_map[slot]
+            for it in range(32):
+                action = agent.get_action(obs)
+                next_obs, reward, term, trunc, info = env.step(action)
+                mem.add(obs=obs, action=action, reward=reward, next_obs=next_obs,
+                        done=term.astype(np.float32))
+                obs = next_obs
+                steps += num_envs
+                if len(mem) >= 2000 and it % agent.learn_step == 0:
+                    batch = mem.sample(agent.batch_size, beta=0.5)
+                    agent.learn(batch)
+                    mem.update_priorities(batch["idxs"], agent.last_td_errors)
+            obs_map[slot] = obs
+        return steps
+
+    elapsed, total = _timed_loop(state, step_fn, args)
+    _emit(state, "env_steps_per_sec", total / elapsed, "steps/s", args, elapsed,
+          {"model": "RainbowDQN cnn[32,64,64] C51 PER n-step3 noisy",
+           "env": "CatchPong-v0 (first-party Pong-like, 4x84x84 uint8)",
+           "global_batch": 256, "num_envs_per_agent": num_envs, "pop_size": pop_size,
+           "parallelism": f"population-parallel dp{state.world_size}"})
+
+
+def run_maddpg(args):
+    """BASELINE config 4: MADDPG speaker_listener pop=8, multi-agent replay."""
+    from agilerl_amd.algorithms import MADDPG
+    from agilerl_amd.components import ReplayBuffer
+    from agilerl_amd.envs.mpe import SpeakerListenerVecEnv
+
+    state = DistributedState.get()
+    device = state.device
+    pop_size = 8
+    num_envs = args.num_envs or 64
+    probe = SpeakerListenerVecEnv(1)
+
+    def factory(index):
+        return MADDPG(
+            probe.observation_spaces, probe.action_spaces, agent_ids=probe.agents,
+            index=index, batch_size=512, device=device,
+            net_config={"arch": "mlp", "hidden_size": [64, 64]},
+        )
+
+    pop = DistributedPopulation(factory, pop_size)
+    envs = {s: SpeakerListenerVecEnv(num_envs, seed=100 + s) for s in pop.local_indices}
+    mems = {s: ReplayBuffer(200_000, device=device) for s in pop.local_indices}
+    obs_map = {s: envs[s].reset()[0] for s in pop.local_indices}
+    agent_ids = probe.agents
+
+    def step_fn():
+        steps = 0
+        for slot in pop.local_indices:
+            agent, env, mem = pop.agents[slot], envs[slot], mems[slot]
+            obs = obs_map[slot]
+            for it in range(32):
+                env_actions, raw = agent.get_action(obs)
+                next_obs, rewards, term, trunc, info = env.step(env_actions)
+                mem.add(obs=obs, action=raw,
+                        reward={a: rewards[a] for a in agent_ids},
+                        next_obs=next_obs,
+                        done={a: term[a].astype(np.float32) for a in agent_ids})
+                obs = next_obs
+                steps += num_envs
+                if len(mem) >= 2000 and it % agent.learn_step == 0:
+                    agent.learn(mem.sample(agent.batch_size))
+            obs_map[slot] = obs
+        return steps
+
+    elapsed, total = _timed_loop(state, step_fn, args)
+    _emit(state, "env_steps_per_sec", total / elapsed, "steps/s", args, elapsed,
+          {"model": "MADDPG mlp[64,64] centralized critics",
+           "env": "simple_speaker_listener (first-party MPE)",
+           "global_batch": 512, "num_envs_per_agent": num_envs, "pop_size": pop_size,
+           "parallelism": f"population-parallel dp{state.world_size}"})
+
+
 def main():
     args = parse_args()
-    if args.num_envs is None:
+    if args.num_envs is None and args.workload == "ppo":
         args.num_envs = 16384 if torch.cuda.is_available() else NUM_ENVS
     if args.workload == "grpo":
         return run_grpo(args)
+    if args.workload == "dqn":
+        return run_dqn(args)
+    if args.workload == "rainbow":
+        return run_rainbow(args)
+    if args.workload == "maddpg":
+        return run_maddpg(args)
     runner = BenchRunner(args)
     state = runner.state
     use_cuda = torch.cuda.is_available()
