@@ -171,9 +171,11 @@ class MADDPG(MultiAgentRLAlgorithm):
         return torch.cat([d[aid].reshape(d[aid].shape[0], -1) for aid in self.agent_ids], dim=1)
 
     def _to_dev(self, d) -> Dict[str, torch.Tensor]:
-        return {
-            aid: torch.as_tensor(np.asarray(v)).float().to(self.device) for aid, v in d.items()
-        }
+        out = {}
+        for aid, v in d.items():
+            t = v if isinstance(v, torch.Tensor) else torch.as_tensor(np.asarray(v))
+            out[aid] = t.float().to(self.device)
+        return out
 
     def learn(self, experiences: Dict[str, Dict[str, torch.Tensor]]) -> float:
         obs = self._to_dev(experiences["obs"])

@@ -66,18 +66,25 @@ class MultiAgentVecEnv:
         self._elapsed[:] = 0
         return self._obs(), {}
 
+    TERMINATES_AT_LIMIT = False  # probe envs: episode END is a true terminal
+
     def step(self, actions: Dict[str, np.ndarray]):
         rewards = self._step_all(actions)
         self._elapsed += 1
-        trunc_arr = self._elapsed >= self.max_episode_steps
-        term = {a: np.zeros(self.num_envs, dtype=bool) for a in self.agents}
+        limit = self._elapsed >= self.max_episode_steps
+        if self.TERMINATES_AT_LIMIT:
+            term_arr, trunc_arr = limit, np.zeros(self.num_envs, dtype=bool)
+        else:
+            term_arr, trunc_arr = np.zeros(self.num_envs, dtype=bool), limit
+        term = {a: term_arr.copy() for a in self.agents}
         trunc = {a: trunc_arr.copy() for a in self.agents}
         obs = self._obs()
         info: Dict = {}
-        if trunc_arr.any():
+        done_arr = term_arr | trunc_arr
+        if done_arr.any():
             info["final_observation"] = {a: o.copy() for a, o in obs.items()}
-            self._reset_rows(trunc_arr)
-            self._elapsed[trunc_arr] = 0
+            self._reset_rows(done_arr)
+            self._elapsed[done_arr] = 0
             obs = self._obs()
         return obs, rewards, term, trunc, info
 

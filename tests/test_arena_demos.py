@@ -1,0 +1,94 @@
+"""Arena client + multi-agent probe tests."""
+
+import os
+
+import numpy as np
+import pytest
+
+from agilerl_amd.arena import ArenaClient, ArenaError
+
+
+class TestArenaClient:
+    def _manifest(self):
+        return {
+            "algorithm": {"name": "DQN", "hyperparameters": {"batch_size": 32, "lr": 1e-3}},
+            "environment": {"env_id": "CartPole-v1", "num_envs": 4},
+            "network": {"arch": "mlp", "encoder_config": {"hidden_size": [16]}},
+            "training": {"max_steps": 300, "pop_size": 1, "evo_steps": 150, "eval_loop": 1},
+        }
+
+    def test_requires_login(self, tmp_path):
+        client = ArenaClient(workspace=str(tmp_path))
+        with pytest.raises(ArenaError, match="not logged in"):
+            client.submit_experiment(self._manifest(), run=False)
+
+    def test_submit_and_status(self, tmp_path):
+        client = ArenaClient(workspace=str(tmp_path))
+        client.login()
+        handle = client.submit_experiment(self._manifest(), run=True)
+        assert handle.status == "completed"
+        assert client.experiment_status(handle.experiment_id)["status"] == "completed"
+        assert handle.experiment_id in client.list_experiments()
+        assert os.path.exists(os.path.join(handle.workspace, "manifest.yaml"))
+
+    def test_validate_environment(self, tmp_path):
+        client = ArenaClient(workspace=str(tmp_path))
+        report = client.validate_environment({"env_id": "CartPole-v1", "algorithm": "DQN"})
+        assert report["valid"]
+        report2 = client.validate_environment({"env_id": "NoSuchEnv", "algorithm": "DQN"})
+        assert report2["warnings"]
+
+    def test_resume(self, tmp_path):
+        client = ArenaClient(workspace=str(tmp_path))
+        client.login()
+        handle = client.submit_experiment(self._manifest(), run=False)
+        resumed = client.resume_experiment(handle.experiment_id)
+        assert resumed.status == "completed"
+
+
+class TestMAProbes:
+    def test_joint_action_env(self):
+        from agilerl_amd.envs.probe_ma import JointActionMAEnv
+
+        env = JointActionMAEnv(num_envs=4)
+        obs, _ = env.reset()
+        actions = {"agent_0": np.zeros(4, dtype=int), "agent_1": np.zeros(4, dtype=int)}
+        _, rewards, term, trunc, _ = env.step(actions)
+        np.testing.assert_allclose(rewards["agent_0"], 1.0)
+        actions["agent_1"] = np.ones(4, dtype=int)
+        env.reset()
+        _, rewards, _, _, _ = env.step(actions)
+        np.testing.assert_allclose(rewards["agent_0"], -1.0)
+
+    def test_maddpg_learns_joint_q(self):
+        """Centralized critic must learn Q(s, a0=0, a1=0)=1 vs mixed=-1."""
+        import torch
+
+        from agilerl_amd.algorithms import MADDPG
+        from agilerl_amd.components import ReplayBuffer
+        from agilerl_amd.envs.probe_ma import JointActionMAEnv
+
+        torch.manual_seed(0)
+        np.random.seed(0)
+        env = JointActionMAEnv(num_envs=8, seed=0)
+        agent = MADDPG(env.observation_spaces, env.action_spaces, agent_ids=env.agents,
+                       batch_size=64, lr_critic=1e-2, tau=0.1,
+                       net_config={"arch": "mlp", "hidden_size": [32]})
+        buf = ReplayBuffer(2000)
+        obs, _ = env.reset()
+        for _ in range(150):
+            acts = {a: np.random.randint(0, 2, 8) for a in env.agents}
+            raw = {a: np.eye(2, dtype=np.float32)[acts[a]] for a in env.agents}
+            next_obs, rewards, term, trunc, _ = env.step(acts)
+            buf.add(obs=obs, action=raw, reward={a: rewards[a] for a in env.agents},
+                    next_obs=next_obs, done={a: term[a].astype(np.float32) for a in env.agents})
+            obs = next_obs
+        for _ in range(300):
+            agent.learn(buf.sample(64))
+        joint_obs = torch.zeros(1, 2)
+        both_zero = torch.tensor([[1.0, 0.0, 1.0, 0.0]])
+        mixed = torch.tensor([[1.0, 0.0, 0.0, 1.0]])
+        q_good = float(agent.critics["agent_0"](joint_obs, both_zero))
+        q_bad = float(agent.critics["agent_0"](joint_obs, mixed))
+        assert q_good > 0.5
+        assert q_bad < 0.0
