@@ -139,6 +139,11 @@ class LLMAlgorithm(EvolvableAlgorithm):
 
         self.model = build_causal_lm(model, model_config, model_name_or_path, dtype, device)
         self.model.requires_grad_(False)
+        if str(device).startswith("cuda") and dtype == torch.bfloat16:
+            # fused CDNA4 RMSNorm (analog of the reference's Liger patches)
+            from ...architectures.llama_patches import apply_hip_kernels_to_llama
+
+            apply_hip_kernels_to_llama(self.model)
         if gradient_checkpointing and hasattr(self.model, "gradient_checkpointing_enable"):
             self.model.gradient_checkpointing_enable()
 

@@ -1,3 +1,6 @@
+from .packing import pack_padded_batch, unpack_values
+from .scheduler import WarmupCosineLR, create_warmup_cosine_scheduler
+from .offload import activation_offload
 from .lora import (
     LoraConfig,
     LoraLinear,
@@ -12,6 +15,11 @@ from .lora import (
 )
 
 __all__ = [
+    "pack_padded_batch",
+    "unpack_values",
+    "WarmupCosineLR",
+    "create_warmup_cosine_scheduler",
+    "activation_offload",
     "LoraConfig",
     "LoraLinear",
     "apply_lora",

@@ -1,0 +1,19 @@
+from .utils import (
+    make_vect_envs,
+    create_population,
+    save_population_checkpoint,
+    save_llm_checkpoint,
+    run_selection_and_mutation,
+    get_env_defined_actions,
+    observation_space_channels_to_first,
+)
+
+__all__ = [
+    "make_vect_envs",
+    "create_population",
+    "save_population_checkpoint",
+    "save_llm_checkpoint",
+    "run_selection_and_mutation",
+    "get_env_defined_actions",
+    "observation_space_channels_to_first",
+]

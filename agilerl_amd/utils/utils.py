@@ -1,0 +1,102 @@
+"""General utilities: population factories, checkpoint helpers, env helpers.
+
+Reference parity: ``agilerl/utils/utils.py`` — ``make_vect_envs`` :222
+(re-exported from envs), ``create_population`` :383,
+``save_population_checkpoint`` :1171, ``save_llm_checkpoint`` :1695,
+``run_selection_and_mutation`` :1248, ``get_env_defined_actions`` :1672.
+"""
+
+from __future__ import annotations
+
+from typing import Any, Dict, List, Optional, Tuple
+
+import numpy as np
+
+from ..envs.registry import make_vect_envs  # noqa: F401 (reference location parity)
+from ..models.manifest import ALGO_REGISTRY, algo_workload, resolve_algo_class
+from ..training.train_off_policy import save_population_checkpoint  # noqa: F401
+
+__all__ = [
+    "make_vect_envs",
+    "create_population",
+    "save_population_checkpoint",
+    "save_llm_checkpoint",
+    "run_selection_and_mutation",
+    "get_env_defined_actions",
+    "observation_space_channels_to_first",
+]
+
+
+def create_population(
+    algo: str,
+    observation_space=None,
+    action_space=None,
+    net_config: Optional[Dict[str, Any]] = None,
+    INIT_HP: Optional[Dict[str, Any]] = None,
+    population_size: int = 4,
+    device: str = "cpu",
+    agent_ids: Optional[List[str]] = None,
+    observation_spaces: Optional[Dict[str, Any]] = None,
+    action_spaces: Optional[Dict[str, Any]] = None,
+    **kwargs,
+) -> List:
+    """Build a population of ``population_size`` agents of the named algo.
+
+    Accepts the reference's calling shapes: single-agent (observation_space,
+    action_space), multi-agent (observation_spaces/action_spaces/agent_ids),
+    hyperparameters via ``INIT_HP`` or kwargs.
+    """
+    cls = resolve_algo_class(algo)
+    workload = algo_workload(algo)
+    hp = dict(INIT_HP or {})
+    hp.update(kwargs)
+    if net_config is not None:
+        hp.setdefault("net_config", net_config)
+    if workload.startswith("multi_agent"):
+        return cls.population(
+            population_size, observation_spaces or observation_space,
+            action_spaces or action_space, agent_ids=agent_ids, device=device, **hp,
+        )
+    if workload.startswith("llm"):
+        return cls.population(population_size, device=device, **hp)
+    return cls.population(population_size, observation_space, action_space, device=device, **hp)
+
+
+def save_llm_checkpoint(agent, path: str) -> None:
+    """Adapter-directory checkpoint for one LLM agent (reference :1695)."""
+    agent.save_checkpoint(path)
+
+
+def run_selection_and_mutation(
+    population: List,
+    tournament,
+    mutations,
+) -> Tuple[Any, List]:
+    """One evolution round: tournament select then mutate (reference :1248)."""
+    elite, new_pop = tournament.select(population)
+    new_pop = mutations.mutation(new_pop)
+    return elite, new_pop
+
+
+def get_env_defined_actions(info: Dict[str, Any], agents: Optional[List[str]] = None):
+    """Extract env-supplied action masks from step info (reference :1672)."""
+    if agents is not None:
+        masks = {}
+        for a in agents:
+            sub = info.get(a, {}) if isinstance(info.get(a), dict) else {}
+            mask = sub.get("action_mask", info.get("action_mask", {}).get(a) if isinstance(info.get("action_mask"), dict) else None)
+            masks[a] = mask
+        return masks if any(m is not None for m in masks.values()) else None
+    return info.get("action_mask")
+
+
+def observation_space_channels_to_first(space):
+    """(H, W, C) image Box -> (C, H, W) (reference algo_utils.py:555-646)."""
+    from ..spaces import Box
+
+    if isinstance(space, Box) and len(space.shape) == 3 and space.shape[-1] in (1, 3, 4):
+        h, w, c = space.shape
+        low = np.transpose(space.low, (2, 0, 1))
+        high = np.transpose(space.high, (2, 0, 1))
+        return Box(low, high, dtype=space.dtype)
+    return space

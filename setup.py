@@ -21,6 +21,7 @@ ext = CUDAExtension(
     sources=[
         "agilerl_amd/ops/csrc/rl_ops.hip",
         "agilerl_amd/ops/csrc/lm_ops.hip",
+        "agilerl_amd/ops/csrc/norm_ops.hip",
     ],
     extra_compile_args={
         "cxx": ["-O3", "-std=c++17"],

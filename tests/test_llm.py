@@ -376,3 +376,41 @@ class TestPPOLLMAndReinforce:
         # RLOO: a_0 = r_0 - mean(others) = 1 - 0 = 1
         assert adv[0].item() == pytest.approx(1.0, abs=1e-5)
         assert adv[1].item() == pytest.approx(-1.0 / 3.0, abs=1e-5)
+
+
+class TestLLMUtils:
+    def test_packing_roundtrip(self):
+        from agilerl_amd.llm import pack_padded_batch, unpack_values
+
+        ids = torch.tensor([[5, 6, 7, 0, 0], [8, 9, 0, 0, 0]])
+        am = (ids != 0).long()
+        pack = pack_padded_batch(ids, am)
+        assert pack["packed_ids"].tolist() == [[5, 6, 7, 8, 9]]
+        assert pack["position_ids"].tolist() == [[0, 1, 2, 0, 1]]
+        assert pack["cu_seqlens"].tolist() == [0, 3, 5]
+        vals = torch.tensor([1.0, 2.0, 3.0, 4.0, 5.0])
+        unpacked = unpack_values(vals, pack)
+        assert unpacked[0].tolist() == [1.0, 2.0, 3.0, 0.0, 0.0]
+        assert unpacked[1].tolist() == [4.0, 5.0, 0.0, 0.0, 0.0]
+
+    def test_warmup_cosine(self):
+        from agilerl_amd.llm import create_warmup_cosine_scheduler
+
+        opt = torch.optim.AdamW([torch.nn.Parameter(torch.zeros(1))], lr=1.0)
+        sched = create_warmup_cosine_scheduler(opt, total_steps=100, warmup_ratio=0.1)
+        lrs = []
+        for _ in range(100):
+            lrs.append(opt.param_groups[0]["lr"])
+            opt.step()
+            sched.step()
+        assert lrs[0] < 0.2           # warmup starts low
+        assert abs(lrs[10] - 1.0) < 0.05  # peak after warmup
+        assert lrs[-1] < 0.25          # decayed toward min ratio
+
+    def test_create_population_util(self):
+        from agilerl_amd.spaces import Box, Discrete
+        from agilerl_amd.utils import create_population
+
+        pop = create_population("DQN", Box(-1, 1, (4,)), Discrete(2),
+                                population_size=3, INIT_HP={"batch_size": 32})
+        assert len(pop) == 3 and pop[0].batch_size == 32

@@ -71,3 +71,42 @@ class TestGRPOGpu:
         for _ in range(10):
             last = agent.learn(env.sample())["loss"]
         assert last < first
+
+
+class TestHipRMSNorm:
+    def test_matches_eager(self):
+        from agilerl_amd.ops.rmsnorm import rms_norm
+
+        x = torch.randn(37, 4096, device=DEV, dtype=torch.bfloat16, requires_grad=True)
+        w = torch.randn(4096, device=DEV, dtype=torch.bfloat16, requires_grad=True)
+        y = rms_norm(x, w, 1e-6)
+        xf = x.float()
+        ref = (xf * torch.rsqrt(xf.pow(2).mean(-1, keepdim=True) + 1e-6) * w.float()).to(torch.bfloat16)
+        torch.testing.assert_close(y, ref, rtol=2e-2, atol=2e-2)
+
+        g = torch.randn_like(y)
+        gx, gw = torch.autograd.grad(y, [x, w], g, retain_graph=True)
+        x2 = x.detach().float().requires_grad_(True)
+        w2 = w.detach().float().requires_grad_(True)
+        ref2 = x2 * torch.rsqrt(x2.pow(2).mean(-1, keepdim=True) + 1e-6) * w2
+        rgx, rgw = torch.autograd.grad(ref2, [x2, w2], g.float())
+        torch.testing.assert_close(gx.float(), rgx, rtol=5e-2, atol=5e-2)
+        torch.testing.assert_close(gw.float(), rgw, rtol=5e-2, atol=2e-1)
+
+    def test_llama_patch(self):
+        from agilerl_amd.architectures import apply_hip_kernels_to_llama
+        from transformers import AutoConfig, AutoModelForCausalLM
+
+        cfg = AutoConfig.for_model("llama", vocab_size=256, hidden_size=128,
+                                   intermediate_size=256, num_hidden_layers=2,
+                                   num_attention_heads=4, num_key_value_heads=2,
+                                   max_position_embeddings=128, pad_token_id=0)
+        m = AutoModelForCausalLM.from_config(cfg, dtype=torch.bfloat16).to(DEV)
+        ids = torch.randint(0, 256, (2, 16), device=DEV)
+        with torch.no_grad():
+            ref = m(input_ids=ids).logits.float()
+        n = apply_hip_kernels_to_llama(m)
+        assert n >= 5  # per-layer input/post-attn + final norm
+        with torch.no_grad():
+            out = m(input_ids=ids).logits.float()
+        torch.testing.assert_close(out, ref, rtol=5e-2, atol=5e-1)
