@@ -47,6 +47,7 @@ class DQN(RLAlgorithm):
         gamma: float = 0.99,
         tau: float = 1e-3,
         double: bool = False,
+        cudagraphs: bool = False,
         latent_dim: int = 64,
         normalize_images: bool = True,
         device: str = "cpu",
@@ -65,9 +66,12 @@ class DQN(RLAlgorithm):
         self.gamma = float(gamma)
         self.tau = float(tau)
         self.double = bool(double)
+        self.cudagraphs = bool(cudagraphs)
         self.net_config = net_config
         self.latent_dim = latent_dim
         self.normalize_images = normalize_images
+        self._graph = None
+        self._graph_static = None
 
         self.actor = QNetwork(
             observation_space,
@@ -97,6 +101,8 @@ class DQN(RLAlgorithm):
         self.actor_target.load_state_dict(self.actor.state_dict())
         for p in self.actor_target.parameters():
             p.requires_grad = False
+        self._graph = None  # captured graph refers to the old modules
+        self._graph_static = None
 
     # ------------------------------------------------------------------
     def get_action(
@@ -126,7 +132,83 @@ class DQN(RLAlgorithm):
         return greedy
 
     # ------------------------------------------------------------------
+    # hipGraph-captured update (reference dqn.py:120/:219 `cudagraphs`)
+    # ------------------------------------------------------------------
+    def _update_body(self, obs, actions, rewards, next_obs, dones, loss_out):
+        with torch.no_grad():
+            q_next_target = self.actor_target(self.actor_target.preprocess(next_obs))
+            if self.double:
+                next_actions = self.actor(self.actor.preprocess(next_obs)).argmax(-1, keepdim=True)
+                q_next = q_next_target.gather(1, next_actions).squeeze(-1)
+            else:
+                q_next = q_next_target.max(dim=-1).values
+            target = rewards + (1.0 - dones) * self.gamma * q_next
+        q_pred = self.actor(self.actor.preprocess(obs)).gather(1, actions.unsqueeze(1)).squeeze(-1)
+        loss = F.huber_loss(q_pred, target)
+        self.optimizer.zero_grad(set_to_none=False)
+        loss.backward()
+        self.optimizer.step()
+        # graph-safe polyak: foreach lerp (stable param addresses)
+        torch._foreach_lerp_(
+            list(self.actor_target.parameters()), list(self.actor.parameters()), self.tau
+        )
+        loss_out.copy_(loss.detach())
+
+    def _graphed_learn(self, obs, actions, rewards, next_obs, dones) -> float:
+        B = actions.shape[0]
+        if self._graph_static is not None and self._graph_static["obs"].shape[0] != B:
+            self._graph = None
+        if self._graph is None:
+            # capturable optimizer state required for in-graph Adam steps
+            self.optimizer.optimizer = torch.optim.Adam(
+                [p for p in self.actor.parameters() if p.requires_grad],
+                lr=self.lr, capturable=True,
+            )
+            st = {
+                "obs": obs.clone(), "actions": actions.clone(), "rewards": rewards.clone(),
+                "next_obs": next_obs.clone(), "dones": dones.clone(),
+                "loss": torch.zeros((), device=obs.device),
+            }
+            self._graph_static = st
+            torch.cuda.synchronize()
+            s = torch.cuda.Stream()
+            s.wait_stream(torch.cuda.current_stream())
+            with torch.cuda.stream(s):
+                for _ in range(3):  # warmup
+                    self._update_body(st["obs"], st["actions"], st["rewards"],
+                                      st["next_obs"], st["dones"], st["loss"])
+            torch.cuda.current_stream().wait_stream(s)
+            torch.cuda.synchronize()
+            self._graph = torch.cuda.CUDAGraph()
+            with torch.cuda.graph(self._graph):
+                self._update_body(st["obs"], st["actions"], st["rewards"],
+                                  st["next_obs"], st["dones"], st["loss"])
+        st = self._graph_static
+        st["obs"].copy_(obs)
+        st["actions"].copy_(actions)
+        st["rewards"].copy_(rewards)
+        st["next_obs"].copy_(next_obs)
+        st["dones"].copy_(dones)
+        self._graph.replay()
+        return float(st["loss"])
+
+    # ------------------------------------------------------------------
     def learn(self, experiences: Dict[str, torch.Tensor]) -> float:
+        if (
+            self.cudagraphs
+            and torch.cuda.is_available()
+            and "weights" not in experiences
+            and "n_steps" not in experiences
+        ):
+            obs = experiences["obs"]
+            if isinstance(obs, torch.Tensor):
+                return self._graphed_learn(
+                    obs.to(self.device).float(),
+                    experiences["action"].to(self.device).long().reshape(-1),
+                    experiences["reward"].to(self.device).float().reshape(-1),
+                    experiences["next_obs"].to(self.device).float(),
+                    experiences["done"].to(self.device).float().reshape(-1),
+                )
         obs = experiences["obs"]
         actions = experiences["action"].to(self.device).long().reshape(-1)
         rewards = experiences["reward"].to(self.device).float().reshape(-1)

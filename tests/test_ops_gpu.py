@@ -203,3 +203,36 @@ class TestAgentOnGpu:
         obs, done, _ = collect_rollouts(agent, env, buf, 16)
         stats = agent.learn(buf)
         assert np.isfinite(stats["policy_loss"])
+
+
+class TestDQNGraph:
+    def test_graphed_update_matches_eager(self):
+        from agilerl_amd.algorithms.dqn import DQN
+        from agilerl_amd.spaces import Box, Discrete
+
+        torch.manual_seed(0)
+        obs_s, act_s = Box(-1, 1, (8,)), Discrete(4)
+        # identical init for both agents
+        torch.manual_seed(7)
+        eager = DQN(obs_s, act_s, device=DEV, lr=1e-3)
+        torch.manual_seed(7)
+        graphed = DQN(obs_s, act_s, device=DEV, lr=1e-3, cudagraphs=True)
+        for p, q in zip(eager.actor.parameters(), graphed.actor.parameters()):
+            torch.testing.assert_close(p, q)
+
+        batches = []
+        for _ in range(5):
+            batches.append({
+                "obs": torch.randn(64, 8, device=DEV),
+                "action": torch.randint(0, 4, (64,), device=DEV),
+                "reward": torch.randn(64, device=DEV),
+                "next_obs": torch.randn(64, 8, device=DEV),
+                "done": (torch.rand(64, device=DEV) < 0.2).float(),
+            })
+        for b in batches:
+            l1 = eager.learn(dict(b))
+            l2 = graphed.learn(dict(b))
+        for p, q in zip(eager.actor.parameters(), graphed.actor.parameters()):
+            torch.testing.assert_close(p, q, rtol=1e-3, atol=1e-4)
+        for p, q in zip(eager.actor_target.parameters(), graphed.actor_target.parameters()):
+            torch.testing.assert_close(p, q, rtol=1e-3, atol=1e-4)
