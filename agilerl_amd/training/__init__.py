@@ -5,6 +5,7 @@ from .train_multi_agent_off_policy import train_multi_agent_off_policy
 from .train_multi_agent_on_policy import train_multi_agent_on_policy
 from .train_bandits import train_bandits
 from .train_offline import train_offline, load_transitions_into_buffer
+from .train_distributed import train_on_policy_distributed
 
 __all__ = [
     "Trainer",
@@ -16,5 +17,6 @@ __all__ = [
     "train_bandits",
     "train_offline",
     "load_transitions_into_buffer",
+    "train_on_policy_distributed",
     "save_population_checkpoint",
 ]

@@ -127,3 +127,61 @@ def test_ddp_allreduce_gloo():
     for p in procs:
         p.join(timeout=180)
         assert p.exitcode == 0
+
+
+def _dist_train_worker(rank, world_size, port, results):
+    os.environ["RANK"] = str(rank)
+    os.environ["LOCAL_RANK"] = str(rank)
+    os.environ["WORLD_SIZE"] = str(world_size)
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+
+    import numpy as np
+
+    from agilerl_amd.algorithms.ppo import PPO
+    from agilerl_amd.envs import CartPoleVecEnv
+    from agilerl_amd.hpo import Mutations, TournamentSelection
+    from agilerl_amd.parallel import DistributedState
+    from agilerl_amd.training.train_distributed import train_on_policy_distributed
+
+    DistributedState.reset()
+    DistributedState.get()
+
+    def agent_factory(slot):
+        torch.manual_seed(slot)
+        env = CartPoleVecEnv(1)
+        return PPO(env.single_observation_space, env.single_action_space,
+                   index=slot, learn_step=32, batch_size=64,
+                   net_config={"arch": "mlp", "hidden_size": [32]})
+
+    def env_factory(slot):
+        return CartPoleVecEnv(8, seed=slot)
+
+    agents, hist = train_on_policy_distributed(
+        agent_factory, env_factory, pop_size=4,
+        max_steps=600, evo_steps=256,
+        tournament=TournamentSelection(2, True, rng=np.random.default_rng(0)),
+        mutation=Mutations(no_mutation=0.6, architecture=0.0, parameters=0.0,
+                           activation=0.0, rl_hp=0.4, rand_seed=rank),
+        use_graph=False, verbose=False,
+    )
+    results[rank] = (sorted(agents.keys()), len(hist))
+    torch.distributed.destroy_process_group()
+
+
+def test_train_on_policy_distributed_gloo():
+    port = _find_free_port()
+    ctx = mp.get_context("spawn")
+    with ctx.Manager() as mgr:
+        results = mgr.dict()
+        procs = [ctx.Process(target=_dist_train_worker, args=(r, 2, port, results))
+                 for r in range(2)]
+        for p in procs:
+            p.start()
+        for p in procs:
+            p.join(timeout=300)
+            assert p.exitcode == 0
+        # rank 0 owns even slots, rank 1 odd slots
+        assert results[0][0] == [0, 2]
+        assert results[1][0] == [1, 3]
+        assert results[0][1] >= 1
