@@ -118,6 +118,13 @@ class PPO(RLAlgorithm):
         self.register_optimizer(
             OptimizerConfig(name="optimizer", networks=["actor", "critic"], lr_name="lr")
         )
+        self.register_mutation_hook("_clear_learn_graph")
+        self._learn_graph = None
+        self._learn_static = None
+
+    def _clear_learn_graph(self) -> None:
+        self._learn_graph = None
+        self._learn_static = None
 
     # ------------------------------------------------------------------
     def get_action(
@@ -234,6 +241,15 @@ class PPO(RLAlgorithm):
         if self.recurrent and not isinstance(rollout, dict):
             return self._learn_recurrent(rollout)
         if isinstance(rollout, dict):
+            from ..spaces import Discrete as _Discrete
+
+            if (
+                torch.cuda.is_available()
+                and rollout["obs"].is_cuda
+                and isinstance(self.action_space, _Discrete)
+                and self.target_kl is None
+            ):
+                return self._graphed_flat_learn(rollout)
             minibatches = lambda: self._flat_minibatches(rollout)
         else:
             assert rollout.advantages is not None, "call compute_returns_and_advantages first"
@@ -286,6 +302,102 @@ class PPO(RLAlgorithm):
         for start in range(0, n, self.batch_size):
             sel = idx[start : start + self.batch_size]
             yield {k: v[sel] for k, v in flat.items()}
+
+    # ------------------------------------------------------------------
+    # hipGraph-captured update over flat rollouts (discrete actions).
+    # One graph replay per minibatch; stats read once per learn() call.
+    # ------------------------------------------------------------------
+    _GRAPH_KEYS = ("obs", "action", "log_prob", "advantages", "returns", "value")
+
+    def _graph_update_body(self, st: Dict[str, torch.Tensor]) -> None:
+        adv = st["advantages"]
+        if self.normalize_advantage:
+            adv = (adv - adv.mean()) / (adv.std() + 1e-8)
+        logits = self.actor(self.actor.preprocess(st["obs"]))
+        logp_all = torch.log_softmax(logits, dim=-1)
+        new_log_prob = logp_all.gather(1, st["action"].unsqueeze(1)).squeeze(1)
+        entropy = -(logp_all.exp() * logp_all).sum(-1)
+        values = self.critic(self.critic.preprocess(st["obs"])).reshape(-1)
+
+        log_ratio = new_log_prob - st["log_prob"]
+        ratio = log_ratio.exp()
+        pg1 = -adv * ratio
+        pg2 = -adv * ratio.clamp(1 - self.clip_coef, 1 + self.clip_coef)
+        policy_loss = torch.maximum(pg1, pg2).mean()
+        if self.clip_vloss:
+            v_clipped = st["value"] + (values - st["value"]).clamp(-self.clip_coef, self.clip_coef)
+            value_loss = 0.5 * torch.maximum(
+                (values - st["returns"]) ** 2, (v_clipped - st["returns"]) ** 2
+            ).mean()
+        else:
+            value_loss = 0.5 * ((values - st["returns"]) ** 2).mean()
+        entropy_loss = entropy.mean()
+        loss = policy_loss + self.vf_coef * value_loss - self.ent_coef * entropy_loss
+        self.optimizer.zero_grad(set_to_none=False)
+        loss.backward()
+        nn.utils.clip_grad_norm_(
+            [p for net in (self.actor, self.critic) for p in net.parameters()],
+            self.max_grad_norm, foreach=True,
+        )
+        self.optimizer.step()
+        st["stats"][0] += policy_loss.detach()
+        st["stats"][1] += value_loss.detach()
+        st["stats"][2] += entropy_loss.detach()
+        st["stats"][3] += ((ratio - 1) - log_ratio).mean().detach()
+
+    def _graphed_flat_learn(self, flat: Dict[str, torch.Tensor]) -> Dict[str, float]:
+        n = flat["advantages"].shape[0]
+        B = min(self.batch_size, n)
+        if self._learn_static is not None and self._learn_static["obs"].shape[0] != B:
+            self._clear_learn_graph()
+        if self._learn_graph is None:
+            self.optimizer.optimizer = torch.optim.Adam(
+                [p for net in (self.actor, self.critic) for p in net.parameters()
+                 if p.requires_grad],
+                lr=self.lr, capturable=True,
+            )
+            st = {
+                "obs": flat["obs"][:B].clone().float(),
+                "action": flat["action"][:B].clone().long(),
+                "log_prob": flat["log_prob"][:B].clone().float(),
+                "advantages": flat["advantages"][:B].clone().float(),
+                "returns": flat["returns"][:B].clone().float(),
+                "value": flat["value"][:B].clone().float(),
+                "stats": torch.zeros(4, device=flat["obs"].device),
+            }
+            self._learn_static = st
+            torch.cuda.synchronize()
+            s = torch.cuda.Stream()
+            s.wait_stream(torch.cuda.current_stream())
+            with torch.cuda.stream(s):
+                for _ in range(3):
+                    self._graph_update_body(st)
+            torch.cuda.current_stream().wait_stream(s)
+            torch.cuda.synchronize()
+            self._learn_graph = torch.cuda.CUDAGraph()
+            with torch.cuda.graph(self._learn_graph):
+                self._graph_update_body(st)
+        st = self._learn_static
+        st["stats"].zero_()
+        n_updates = 0
+        n_full = (n // B) * B
+        for _ in range(self.update_epochs):
+            perm = torch.randperm(n, device=st["obs"].device)
+            for start in range(0, n_full, B):
+                sel = perm[start : start + B]
+                for key in self._GRAPH_KEYS:
+                    st[key].copy_(
+                        flat[key].index_select(0, sel).to(st[key].dtype), non_blocking=True
+                    )
+                self._learn_graph.replay()
+                n_updates += 1
+        stats_v = (st["stats"] / max(n_updates, 1)).cpu()
+        return {
+            "policy_loss": float(stats_v[0]),
+            "value_loss": float(stats_v[1]),
+            "entropy": float(stats_v[2]),
+            "approx_kl": float(stats_v[3]),
+        }
 
     def _update_minibatch(self, mb: Dict[str, torch.Tensor]) -> Dict[str, float]:
         obs = mb["obs"]
