@@ -1,6 +1,7 @@
 from .packing import pack_padded_batch, unpack_values
 from .scheduler import WarmupCosineLR, create_warmup_cosine_scheduler
 from .offload import activation_offload
+from .chat import apply_chat_template
 from .lora import (
     LoraConfig,
     LoraLinear,
@@ -20,6 +21,7 @@ __all__ = [
     "WarmupCosineLR",
     "create_warmup_cosine_scheduler",
     "activation_offload",
+    "apply_chat_template",
     "LoraConfig",
     "LoraLinear",
     "apply_lora",

@@ -1,5 +1,6 @@
 from .agent import AgentWrapper, RSNorm, AsyncAgentsWrapper
 from .learning import Skill, BanditEnv
 from .make_evolvable import MakeEvolvable
+from .pettingzoo_wrappers import AutoResetParallelWrapper
 
-__all__ = ["AgentWrapper", "RSNorm", "AsyncAgentsWrapper", "Skill", "BanditEnv", "MakeEvolvable"]
+__all__ = ["AgentWrapper", "RSNorm", "AsyncAgentsWrapper", "Skill", "BanditEnv", "MakeEvolvable", "AutoResetParallelWrapper"]
