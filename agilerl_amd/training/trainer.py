@@ -96,7 +96,9 @@ class LocalTrainer(Trainer):
         m = self.manifest
         algo_cls = resolve_algo_class(m.algorithm.name)
         workload = algo_workload(m.algorithm.name)
-        hps = dict(m.algorithm.hyperparameters)
+        from ..models.algorithms import merged_hyperparameters
+
+        hps = merged_hyperparameters(m.algorithm.name, m.algorithm.hyperparameters)
         net = m.network
         net_config = dict(net.encoder_config) or None
         if net.arch and net_config is not None:
