@@ -269,6 +269,48 @@ class LLMAlgorithm(EvolvableAlgorithm):
         self.model.train()
         return out
 
+    # ------------------------------------------------------------------
+    # Cross-rank safety (reference base.py:5201-5222 / grpo.py:727-740)
+    # ------------------------------------------------------------------
+    def raise_if_loss_not_finite_on_any_rank(self, loss: torch.Tensor) -> None:
+        """All DP ranks must raise together, or the healthy ranks hang in
+        the next collective while one rank has already aborted."""
+        from ...parallel import DistributedState
+
+        state = DistributedState.get()
+        bad = (~torch.isfinite(loss.detach())).float().reshape(1)
+        if state.is_distributed:
+            import torch.distributed as dist
+
+            flag = bad.to(self.device) if state.backend == "nccl" else bad.cpu()
+            dist.all_reduce(flag, op=dist.ReduceOp.MAX)
+            bad = flag
+        if float(bad) > 0:
+            raise RuntimeError(
+                f"non-finite loss detected on at least one rank (local loss={float(loss)})"
+            )
+
+    def check_seq_len_agreement(self, seq_len: int) -> None:
+        """DP micro-batches must agree on T or the flat all-reduce deadlocks
+        with mismatched bucket sizes; fail fast with a clear error."""
+        from ...parallel import DistributedState
+
+        state = DistributedState.get()
+        if not state.is_distributed:
+            return
+        import torch.distributed as dist
+
+        t = torch.tensor([float(seq_len), -float(seq_len)])
+        if state.backend == "nccl":
+            t = t.to(self.device)
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        mx, mn = float(t[0]), -float(t[1])
+        if mx != mn:
+            raise RuntimeError(
+                f"sequence-length mismatch across ranks: min={int(mn)} max={int(mx)} "
+                f"(local={seq_len}); pad batches to a common length"
+            )
+
     def backward_and_step(self, loss: torch.Tensor, accumulate: bool = False) -> None:
         loss.backward()
         if accumulate:

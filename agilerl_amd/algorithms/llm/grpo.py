@@ -101,6 +101,7 @@ class GRPO(LLMAlgorithm):
         action_mask = experiences["action_mask"].to(self.device).float()
         rewards = experiences["rewards"].to(self.device).float()
 
+        self.check_seq_len_agreement(ids.shape[1])
         advantages = self._calculate_advantages(rewards)  # (B,)
         adv_tok = advantages.unsqueeze(1).expand_as(action_mask)
 
@@ -135,8 +136,7 @@ class GRPO(LLMAlgorithm):
                     clip_lo,
                     clip_hi,
                 )
-                if not torch.isfinite(loss):
-                    raise RuntimeError(f"non-finite GRPO loss: {loss}")
+                self.raise_if_loss_not_finite_on_any_rank(loss)
                 self.backward_and_step(loss)
                 with torch.no_grad():
                     ratio = (logp - old_logp[sel]).exp()

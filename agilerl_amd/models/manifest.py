@@ -187,6 +187,7 @@ class TrainingSpec(_Base):
     save_elite: bool = False
     elite_path: Optional[str] = None
     max_wall_seconds: Optional[float] = None
+    resume_from_checkpoint: Optional[str] = None
     seed: Optional[int] = None
 
 

@@ -112,8 +112,9 @@ class LocalTrainer(Trainer):
             kwargs["head_config"] = dict(net.head_config)
         kwargs["latent_dim"] = net.latent_dim
 
+        resume = m.training.resume_from_checkpoint
         if workload.startswith("multi_agent"):
-            return algo_cls.population(
+            pop = algo_cls.population(
                 m.training.pop_size,
                 env.observation_spaces,
                 env.action_spaces,
@@ -121,13 +122,31 @@ class LocalTrainer(Trainer):
                 device=self.device,
                 **kwargs,
             )
-        return algo_cls.population(
-            m.training.pop_size,
-            env.single_observation_space,
-            env.single_action_space,
-            device=self.device,
-            **kwargs,
-        )
+        else:
+            pop = algo_cls.population(
+                m.training.pop_size,
+                env.single_observation_space,
+                env.single_action_space,
+                device=self.device,
+                **kwargs,
+            )
+        if resume:
+            self._resume_population(pop, resume)
+        return pop
+
+    @staticmethod
+    def _resume_population(pop: List, base_path: str) -> None:
+        """Load per-agent checkpoint files written by
+        save_population_checkpoint (<base>_<i>.pt; reference
+        resume_from_checkpoint threading, base.py:263-275)."""
+        import os
+
+        base, ext = os.path.splitext(base_path)
+        ext = ext or ".pt"
+        for i, agent in enumerate(pop):
+            path = f"{base}_{i}{ext}"
+            if os.path.exists(path):
+                agent.load_checkpoint(path)
 
     def _make_buffer(self) -> ReplayBuffer:
         spec = self.manifest.replay_buffer

@@ -149,3 +149,26 @@ class TestBandits:
         }
         agents, _ = LocalTrainer.from_manifest(m).train()
         assert agents[0].algo == "NeuralUCB"
+
+
+class TestResume:
+    def test_resume_from_population_checkpoint(self, tmp_path):
+        ckpt = str(tmp_path / "pop.pt")
+        m = {
+            "algorithm": {"name": "DQN", "hyperparameters": {"batch_size": 32, "lr": 1e-3}},
+            "environment": {"env_id": "CartPole-v1", "num_envs": 4},
+            "network": {"arch": "mlp", "encoder_config": {"hidden_size": [16]}},
+            "training": {"max_steps": 800, "pop_size": 2, "evo_steps": 400,
+                         "eval_loop": 1, "checkpoint": 400, "checkpoint_path": ckpt},
+        }
+        agents1, _ = LocalTrainer.from_manifest(m).train()
+        import os
+
+        assert os.path.exists(str(tmp_path / "pop_0.pt"))
+        # resume: new trainer picks up saved weights and step counts
+        m2 = dict(m)
+        m2["training"] = dict(m["training"], resume_from_checkpoint=ckpt, max_steps=1200)
+        trainer2 = LocalTrainer.from_manifest(m2)
+        env = trainer2._make_env()
+        pop2 = trainer2._make_population(env)
+        assert pop2[0].steps[-1] > 0  # carried over from the checkpoint
