@@ -170,6 +170,10 @@ class DQN(RLAlgorithm):
                 "loss": torch.zeros((), device=obs.device),
             }
             self._graph_static = st
+            # warmup + capture run REAL updates; snapshot and restore so the
+            # graphed agent stays numerically identical to the eager path
+            saved_actor = [p.detach().clone() for p in self.actor.parameters()]
+            saved_target = [p.detach().clone() for p in self.actor_target.parameters()]
             torch.cuda.synchronize()
             s = torch.cuda.Stream()
             s.wait_stream(torch.cuda.current_stream())
@@ -183,6 +187,17 @@ class DQN(RLAlgorithm):
             with torch.cuda.graph(self._graph):
                 self._update_body(st["obs"], st["actions"], st["rewards"],
                                   st["next_obs"], st["dones"], st["loss"])
+            with torch.no_grad():
+                for p, sv in zip(self.actor.parameters(), saved_actor):
+                    p.copy_(sv)
+                for p, sv in zip(self.actor_target.parameters(), saved_target):
+                    p.copy_(sv)
+                # zero optimizer state IN PLACE (the graph holds these tensors)
+                for group_state in self.optimizer.optimizer.state.values():
+                    for v in group_state.values():
+                        if torch.is_tensor(v):
+                            v.zero_()
+            torch.cuda.synchronize()
         st = self._graph_static
         st["obs"].copy_(obs)
         st["actions"].copy_(actions)

@@ -366,6 +366,10 @@ class PPO(RLAlgorithm):
                 "stats": torch.zeros(4, device=flat["obs"].device),
             }
             self._learn_static = st
+            # warmup + capture run REAL updates; snapshot and restore so the
+            # first minibatch replays start from the pre-capture weights
+            nets = [p for net in (self.actor, self.critic) for p in net.parameters()]
+            saved = [p.detach().clone() for p in nets]
             torch.cuda.synchronize()
             s = torch.cuda.Stream()
             s.wait_stream(torch.cuda.current_stream())
@@ -377,6 +381,14 @@ class PPO(RLAlgorithm):
             self._learn_graph = torch.cuda.CUDAGraph()
             with torch.cuda.graph(self._learn_graph):
                 self._graph_update_body(st)
+            with torch.no_grad():
+                for p, sv in zip(nets, saved):
+                    p.copy_(sv)
+                for group_state in self.optimizer.optimizer.state.values():
+                    for v in group_state.values():
+                        if torch.is_tensor(v):
+                            v.zero_()
+            torch.cuda.synchronize()
         st = self._learn_static
         st["stats"].zero_()
         n_updates = 0
