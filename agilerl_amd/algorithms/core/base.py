@@ -197,6 +197,11 @@ class EvolvableAlgorithm(metaclass=AlgorithmMeta):
     # ------------------------------------------------------------------
     # Cloning (evolution)
     # ------------------------------------------------------------------
+    # Never cloned/checkpointed: captured hipGraphs are bound to ONE agent's
+    # live tensors — a copied graph handle replayed by an offspring reads
+    # freed/foreign memory (observed as an HSA hardware exception).
+    _GRAPH_ATTRS = ("_graph", "_graph_static", "_learn_graph", "_learn_static")
+
     def inspect_attributes(self, ignore: Tuple[str, ...] = ()) -> Dict[str, Any]:
         """Plain (non-network, non-optimizer) attributes to carry across clones."""
         skip = set(ignore) | {
@@ -206,11 +211,14 @@ class EvolvableAlgorithm(metaclass=AlgorithmMeta):
             "_wrapped",
             "accelerator",
         }
+        skip.update(self._GRAPH_ATTRS)
         skip.update(self.registry.all_network_names())
         skip.update(cfg.name for cfg in self.registry.optimizer_configs)
         out = {}
         for k, v in vars(self).items():
             if k in skip or isinstance(v, (nn.Module, OptimizerWrapper)):
+                continue
+            if type(v).__name__ == "CUDAGraph":
                 continue
             out[k] = v
         return out

@@ -151,3 +151,33 @@ class TestGraphCollector:
         flat2, _ = col.collect()
         st2 = agent.learn(flat2)
         assert np.isfinite(st2["policy_loss"])
+
+    @pytest.mark.gpu
+    def test_evolution_with_graphs_survives(self):
+        """Clone after graph capture must NOT inherit graph handles; training
+        continues through evolution rounds (regression: HSA exception from a
+        deep-copied CUDAGraph replay)."""
+        import numpy as np
+
+        from agilerl_amd.algorithms.ppo import PPO
+        from agilerl_amd.hpo import Mutations, TournamentSelection
+        from agilerl_amd.training.train_distributed import train_on_policy_distributed
+
+        def agent_factory(slot):
+            torch.manual_seed(slot)
+            env = LunarLanderTorchVecEnv(1, device="cuda:0")
+            return PPO(env.single_observation_space, env.single_action_space,
+                       index=slot, learn_step=32, batch_size=4096, device="cuda:0")
+
+        def env_factory(slot):
+            return LunarLanderTorchVecEnv(512, device="cuda:0", seed=slot)
+
+        agents, hist = train_on_policy_distributed(
+            agent_factory, env_factory, pop_size=2,
+            max_steps=100_000, evo_steps=33_000,
+            tournament=TournamentSelection(2, True, rng=np.random.default_rng(0)),
+            mutation=Mutations(no_mutation=0.3, architecture=0.3, parameters=0.2,
+                               activation=0.0, rl_hp=0.2, rand_seed=0),
+            verbose=False)
+        assert len(hist) >= 2  # survived at least one evolution + recapture
+        torch.cuda.synchronize()
