@@ -50,7 +50,13 @@ _LAZY = {
     "Population": "agilerl_amd.population",
     "TournamentSelection": "agilerl_amd.hpo.tournament",
     "Mutations": "agilerl_amd.hpo.mutation",
+    "MultiFrequencySelection": "agilerl_amd.hpo.multi_frequency",
     "make_vect_envs": "agilerl_amd.envs.registry",
+    "create_population": "agilerl_amd.utils.utils",
+    "TrainingManifest": "agilerl_amd.models.manifest",
+    "ArenaClient": "agilerl_amd.arena.client",
+    "DistributedPopulation": "agilerl_amd.parallel.population_runtime",
+    "DistributedState": "agilerl_amd.parallel.state",
 }
 
 
@@ -75,5 +81,11 @@ __all__ = [
     "Population",
     "TournamentSelection",
     "Mutations",
+    "MultiFrequencySelection",
     "make_vect_envs",
+    "create_population",
+    "TrainingManifest",
+    "ArenaClient",
+    "DistributedPopulation",
+    "DistributedState",
 ]

@@ -126,7 +126,10 @@ class RainbowDQN(RLAlgorithm):
         if action_mask is not None:
             mask_t = torch.as_tensor(np.asarray(action_mask), device=q.device, dtype=torch.bool)
             q = q.masked_fill(~mask_t, float("-inf"))
-        return q.argmax(dim=-1).cpu().numpy()
+        greedy = q.argmax(dim=-1)
+        if isinstance(obs, torch.Tensor) and obs.is_cuda:
+            return greedy  # device-native path (torch envs)
+        return greedy.cpu().numpy()
 
     # ------------------------------------------------------------------
     def learn(self, experiences: Dict[str, torch.Tensor]) -> float:

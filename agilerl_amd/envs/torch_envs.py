@@ -80,7 +80,8 @@ class TorchVecEnv:
         self._reset_rows(done)
         self._elapsed.copy_(torch.where(done, torch.zeros_like(self._elapsed), self._elapsed))
         self._ep_return.copy_(torch.where(done, torch.zeros_like(self._ep_return), self._ep_return))
-        obs = torch.where(done.unsqueeze(-1), self._obs(), obs)
+        done_b = done.view(-1, *([1] * (obs.dim() - 1)))
+        obs = torch.where(done_b, self._obs(), obs)
         return obs, reward, terminated, truncated, info
 
 
@@ -183,6 +184,87 @@ class LunarLanderTorchVecEnv(TorchVecEnv):
         reward = torch.where(crash, reward - 100.0, reward)
         reward = torch.where(landed, reward + 100.0, reward)
         return reward, terminated
+
+
+class CatchPongTorchVecEnv(TorchVecEnv):
+    """Device-resident port of ``CatchPongVecEnv`` (uint8 frame stacks stay
+    in HBM; the Rainbow pipeline never crosses PCIe for observations)."""
+
+    max_episode_steps = 2000
+    H = W = 84
+    FRAMES = 4
+    PADDLE_W = 12
+    BALL = 3
+
+    def __init__(self, num_envs: int = 1, device: str = "cuda", seed: Optional[int] = None):
+        super().__init__(num_envs, device, seed)
+        self.single_observation_space = Box(0, 255, (self.FRAMES, self.H, self.W), dtype=np.uint8)
+        self.single_action_space = Discrete(3)
+        N = self.num_envs
+        self.ball = torch.zeros(N, 3, device=device)  # x, y, vx
+        self.paddle_x = torch.zeros(N, device=device)
+        self.frames = torch.zeros(N, self.FRAMES, self.H, self.W, dtype=torch.uint8, device=device)
+        self._rows = torch.arange(N, device=device)
+        self._ball_off = torch.stack(
+            torch.meshgrid(torch.arange(self.BALL, device=device),
+                           torch.arange(self.BALL, device=device), indexing="ij"),
+            dim=-1,
+        ).reshape(-1, 2)  # (9, 2) dy,dx
+        self._pad_off = torch.arange(self.PADDLE_W, device=device)
+
+    def _spawn_ball(self, mask: torch.Tensor) -> None:
+        N = self.num_envs
+        new = torch.stack(
+            [self._rand(N, low=5.0, high=self.W - 5.0),
+             torch.full((N,), 2.0, device=self.device),
+             self._rand(N, low=-1.0, high=1.0)], dim=1)
+        self.ball.copy_(torch.where(mask.unsqueeze(1), new, self.ball))
+
+    def _reset_rows(self, mask: torch.Tensor) -> None:
+        self._spawn_ball(mask)
+        self.paddle_x.copy_(torch.where(mask, torch.full_like(self.paddle_x, self.W / 2), self.paddle_x))
+        self.frames.copy_(torch.where(mask.view(-1, 1, 1, 1), torch.zeros_like(self.frames), self.frames))
+        self._render()
+
+    def _render(self) -> None:
+        self.frames[:, :-1].copy_(self.frames[:, 1:].clone())
+        frame = torch.zeros(self.num_envs, self.H, self.W, dtype=torch.uint8, device=self.device)
+        bx = self.ball[:, 0].long().clamp(0, self.W - self.BALL)
+        by = self.ball[:, 1].long().clamp(0, self.H - self.BALL)
+        px = self.paddle_x.long().clamp(0, self.W - self.PADDLE_W)
+        rows = self._rows.view(-1, 1)
+        # ball: 9 pixels per env via flat index_put
+        yy = (by.view(-1, 1) + self._ball_off[:, 0].view(1, -1)).reshape(-1)
+        xx = (bx.view(-1, 1) + self._ball_off[:, 1].view(1, -1)).reshape(-1)
+        rr = rows.expand(-1, self._ball_off.shape[0]).reshape(-1)
+        frame[rr, yy, xx] = 255
+        cols = (px.view(-1, 1) + self._pad_off.view(1, -1))
+        frame[rows.expand_as(cols), self.H - 3, cols] = 180
+        frame[rows.expand_as(cols), self.H - 2, cols] = 180
+        self.frames[:, -1].copy_(frame)
+
+    def _obs(self) -> torch.Tensor:
+        return self.frames.clone()
+
+    def _step_all(self, actions: torch.Tensor):
+        a = actions.reshape(-1).long()
+        self.paddle_x.add_(torch.where(a == 1, -3.0, torch.where(a == 2, 3.0, torch.zeros_like(self.paddle_x))))
+        self.paddle_x.clamp_(0, self.W - self.PADDLE_W)
+        bx, by, bvx = self.ball.unbind(1)
+        by = by + 2.0
+        bx = bx + bvx
+        bounce = (bx <= 0) | (bx >= self.W - self.BALL)
+        bvx = torch.where(bounce, -bvx, bvx)
+        bx = bx.clamp(0, float(self.W - self.BALL))
+        self.ball.copy_(torch.stack([bx, by, bvx], dim=1))
+
+        at_bottom = by >= self.H - 5
+        caught = at_bottom & (bx + self.BALL >= self.paddle_x) & (bx <= self.paddle_x + self.PADDLE_W)
+        missed = at_bottom & ~caught
+        reward = torch.where(caught, 1.0, torch.where(missed, -1.0, torch.zeros_like(bx)))
+        self._spawn_ball(at_bottom)
+        self._render()
+        return reward, missed
 
 
 class CartPoleTorchVecEnv(TorchVecEnv):

@@ -140,7 +140,10 @@ class EvolvableNetwork(EvolvableModule):
         elif isinstance(space, MultiBinary):
             obs = obs.float()
         else:
-            obs = obs.float()
+            if isinstance(space, Box) and space.dtype == np.uint8 and not obs.is_floating_point():
+                obs = obs.float() / 255.0
+            else:
+                obs = obs.float()
             if obs.dim() == len(space.shape):
                 obs = obs.unsqueeze(0)
         return obs

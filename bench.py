@@ -400,9 +400,19 @@ def run_rainbow(args):
         )
 
     pop = DistributedPopulation(factory, pop_size)
-    envs = {s: CatchPongVecEnv(num_envs, seed=100 + s) for s in pop.local_indices}
+    use_torch_env = torch.cuda.is_available()
+    if use_torch_env:
+        from agilerl_amd.envs.torch_envs import CatchPongTorchVecEnv
+
+        envs = {s: CatchPongTorchVecEnv(num_envs, device=device, seed=100 + s)
+                for s in pop.local_indices}
+        storage = device  # replay lives in HBM (288 GB): no PCIe in the loop
+    else:
+        envs = {s: CatchPongVecEnv(num_envs, seed=100 + s) for s in pop.local_indices}
+        storage = None
     mems = {
-        s: PrioritizedReplayBuffer(200_000, n_step=3, gamma=0.99, device=device)
+        s: PrioritizedReplayBuffer(100_000, n_step=3, gamma=0.99, device=device,
+                                   storage_device=storage)
         for s in pop.local_indices
     }
     obs_map = {s: envs[s].reset()[0] for s in pop.local_indices}
@@ -415,8 +425,9 @@ def run_rainbow(args):
             for it in range(32):
                 action = agent.get_action(obs)
                 next_obs, reward, term, trunc, info = env.step(action)
+                done = term.float() if use_torch_env else term.astype(np.float32)
                 mem.add(obs=obs, action=action, reward=reward, next_obs=next_obs,
-                        done=term.astype(np.float32))
+                        done=done)
                 obs = next_obs
                 steps += num_envs
                 if len(mem) >= 2000 and it % agent.learn_step == 0:
