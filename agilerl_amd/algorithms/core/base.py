@@ -175,8 +175,17 @@ class EvolvableAlgorithm(metaclass=AlgorithmMeta):
         policy = self.policy_network
         result = policy.apply_mutation(method, **choices) or {}
         merged = {**choices, **result}
+        policy_enc = getattr(policy, "encoder", None)
         for name, net in self.evolvable_networks().items():
             if net is policy:
+                continue
+            shares_encoder = (
+                policy_enc is not None and getattr(net, "encoder", None) is policy_enc
+            )
+            if shares_encoder and method.startswith("encoder."):
+                continue  # the shared encoder object was already mutated once
+            if shares_encoder and method in ("add_latent_node", "remove_latent_node"):
+                net._resize_latent(policy.latent_dim, resize_encoder=False)
                 continue
             if method in net.mutation_methods:
                 net.apply_mutation(method, **merged)

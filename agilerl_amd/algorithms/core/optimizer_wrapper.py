@@ -41,9 +41,12 @@ class OptimizerWrapper:
 
     # ------------------------------------------------------------------
     def _params(self, networks: Iterable[nn.Module]):
-        params = []
+        params, seen = [], set()
         for net in networks:
-            params.extend(p for p in net.parameters() if p.requires_grad)
+            for p in net.parameters():
+                if p.requires_grad and id(p) not in seen:
+                    seen.add(id(p))
+                    params.append(p)
         return params
 
     def reinit(self, networks: List[nn.Module], lr: Optional[float] = None) -> None:

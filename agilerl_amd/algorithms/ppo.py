@@ -60,6 +60,7 @@ class PPO(RLAlgorithm):
         log_std_init: float = 0.0,
         latent_dim: int = 64,
         recurrent: bool = False,
+        share_encoders: bool = False,
         device: str = "cpu",
     ):
         super().__init__(
@@ -109,6 +110,9 @@ class PPO(RLAlgorithm):
             latent_dim=latent_dim,
             device=device,
         )
+        self.share_encoders = bool(share_encoders)
+        if self.share_encoders:
+            self.critic.encoder = self.actor.encoder  # one encoder, two heads
         self.optimizer = OptimizerWrapper(
             torch.optim.Adam, [self.actor, self.critic], lr=self.lr
         )
@@ -125,6 +129,17 @@ class PPO(RLAlgorithm):
     def _clear_learn_graph(self) -> None:
         self._learn_graph = None
         self._learn_static = None
+
+    def post_clone_hook(self, parent) -> None:
+        if getattr(self, "share_encoders", False):
+            self.critic.encoder = self.actor.encoder
+            self._reinit_optimizers()
+
+    def _apply_checkpoint(self, ckpt) -> None:
+        super()._apply_checkpoint(ckpt)
+        if getattr(self, "share_encoders", False):
+            self.critic.encoder = self.actor.encoder
+            self._reinit_optimizers()
 
     # ------------------------------------------------------------------
     def get_action(

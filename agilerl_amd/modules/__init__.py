@@ -24,6 +24,7 @@ from .simba import EvolvableSimBa
 from .multi_input import EvolvableMultiInput
 from .gpt import EvolvableGPT, CausalSelfAttention, GPTBlock
 from .bert import EvolvableBERT
+from .resnet import EvolvableResNet
 
 __all__ = [
     "EvolvableModule",
@@ -51,4 +52,5 @@ __all__ = [
     "CausalSelfAttention",
     "GPTBlock",
     "EvolvableBERT",
+    "EvolvableResNet",
 ]

@@ -66,6 +66,12 @@ def build_encoder(
             device=device,
             **{k: v for k, v in cfg.items() if k in ("mlp_config", "cnn_config")},
         )
+    if arch == "resnet":
+        from ..modules.resnet import EvolvableResNet
+
+        return EvolvableResNet(
+            input_shape=observation_space.shape, num_outputs=latent_dim, device=device, **cfg
+        )
     if arch == "cnn" or is_image_space(observation_space):
         return EvolvableCNN(
             input_shape=observation_space.shape, num_outputs=latent_dim, device=device, **cfg
@@ -206,15 +212,14 @@ class EvolvableNetwork(EvolvableModule):
         self._last_mutation = (name, result if isinstance(result, dict) else {})
         return result
 
-    def _resize_latent(self, new_dim: int) -> None:
+    def _resize_latent(self, new_dim: int, resize_encoder: bool = True) -> None:
         new_dim = int(np.clip(new_dim, self.MIN_LATENT, self.MAX_LATENT))
         if new_dim == self.latent_dim:
             return
         self.latent_dim = new_dim
-        # encoder output width
-        self.encoder.num_outputs = new_dim
-        self.encoder.recreate_network()
-        # head input width
+        if resize_encoder:  # skipped when the encoder is shared and already resized
+            self.encoder.num_outputs = new_dim
+            self.encoder.recreate_network()
         self.head_net.num_inputs = new_dim
         self.head_net.recreate_network()
 

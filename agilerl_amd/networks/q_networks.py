@@ -172,14 +172,15 @@ class ContinuousQNetwork(EvolvableNetwork):
             action = action.unsqueeze(0)
         return self.head_net(torch.cat([feats, action.reshape(feats.shape[0], -1)], dim=-1))
 
-    def _resize_latent(self, new_dim: int) -> None:
+    def _resize_latent(self, new_dim: int, resize_encoder: bool = True) -> None:
         import numpy as np
 
         new_dim = int(np.clip(new_dim, self.MIN_LATENT, self.MAX_LATENT))
         if new_dim == self.latent_dim:
             return
         self.latent_dim = new_dim
-        self.encoder.num_outputs = new_dim
-        self.encoder.recreate_network()
+        if resize_encoder:
+            self.encoder.num_outputs = new_dim
+            self.encoder.recreate_network()
         self.head_net.num_inputs = new_dim + self.action_dim
         self.head_net.recreate_network()

@@ -147,3 +147,43 @@ class TestCloneCheckpoint:
         agent.apply_architecture_mutation("encoder.add_layer")
         x = torch.randn(3, 4)
         assert torch.allclose(agent.actor(x), agent.actor_target(x))
+
+
+class TestSharedEncoders:
+    def test_share_and_mutate(self):
+        from agilerl_amd.algorithms.ppo import PPO
+        from agilerl_amd.spaces import Box, Discrete
+
+        a = PPO(Box(-1, 1, (6,)), Discrete(3), share_encoders=True)
+        assert a.critic.encoder is a.actor.encoder
+        h0 = sum(a.actor.encoder.hidden_size)
+        a.apply_architecture_mutation("encoder.add_node", numb_new_nodes=16)
+        # applied exactly once to the shared object
+        assert sum(a.actor.encoder.hidden_size) - h0 == 16
+        assert a.critic.encoder is a.actor.encoder
+        a.apply_architecture_mutation("add_latent_node", numb_new_nodes=16)
+        assert a.actor.latent_dim == a.critic.latent_dim
+        x = torch.randn(3, 6)
+        assert a.actor(x).shape == (3, 3) and a.critic(x).shape == (3, 1)
+
+    def test_clone_preserves_sharing(self):
+        from agilerl_amd.algorithms.ppo import PPO
+        from agilerl_amd.spaces import Box, Discrete
+
+        a = PPO(Box(-1, 1, (6,)), Discrete(3), share_encoders=True)
+        c = a.clone(1)
+        assert c.critic.encoder is c.actor.encoder
+        assert c.actor.encoder is not a.actor.encoder
+        x = torch.randn(3, 6)
+        assert torch.allclose(c.actor(x), a.actor(x))
+
+    def test_resnet_encoder_q(self):
+        from agilerl_amd.networks import QNetwork
+        from agilerl_amd.spaces import Box, Discrete
+
+        q = QNetwork(Box(0, 255, (3, 16, 16)), Discrete(4),
+                     encoder_config={"arch": "resnet", "channel_size": 16, "num_blocks": 1})
+        x = torch.randn(2, 3, 16, 16)
+        assert q(x).shape == (2, 4)
+        q.apply_mutation("encoder.add_block")
+        assert q(x).shape == (2, 4)
