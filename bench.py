@@ -387,7 +387,7 @@ def run_rainbow(args):
     state = DistributedState.get()
     device = state.device
     pop_size = 4
-    num_envs = args.num_envs or (512 if torch.cuda.is_available() else 8)
+    num_envs = args.num_envs or (2048 if torch.cuda.is_available() else 8)
 
     def factory(index):
         return RainbowDQN(
