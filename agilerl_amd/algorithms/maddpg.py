@@ -146,6 +146,8 @@ class MADDPG(MultiAgentRLAlgorithm):
     ) -> Tuple[Dict[str, np.ndarray], Dict[str, np.ndarray]]:
         """Returns (env_actions, raw_actions).  Raw actions are what the
         centralized critics consume (one-hot / continuous vectors)."""
+        first = obs[self.agent_ids[0]]
+        device_native = isinstance(first, torch.Tensor) and first.is_cuda
         env_actions, raw_actions = {}, {}
         with torch.no_grad():
             for aid in self.agent_ids:
@@ -155,15 +157,18 @@ class MADDPG(MultiAgentRLAlgorithm):
                 space = self.action_spaces[aid]
                 if isinstance(space, Discrete):
                     raw = out  # gumbel-softmax one-hot (train) / hard one-hot (eval)
-                    env_actions[aid] = raw.argmax(-1).cpu().numpy()
+                    act = raw.argmax(-1)
+                    env_actions[aid] = act if device_native else act.cpu().numpy()
+                    raw_actions[aid] = raw if device_native else raw.cpu().numpy()
                 else:
-                    raw = out
-                    a = out.cpu().numpy()
+                    a = out
                     if training and self.expl_noise > 0:
-                        a = a + np.random.normal(0, self.expl_noise, a.shape)
-                    env_actions[aid] = np.clip(a, space.low, space.high)
-                    raw = torch.as_tensor(env_actions[aid], device=out.device, dtype=out.dtype)
-                raw_actions[aid] = raw.cpu().numpy()
+                        a = a + torch.randn_like(a) * self.expl_noise
+                    low = torch.as_tensor(space.low, device=a.device, dtype=a.dtype)
+                    high = torch.as_tensor(space.high, device=a.device, dtype=a.dtype)
+                    a = a.clamp(low.min(), high.max())
+                    env_actions[aid] = a if device_native else a.cpu().numpy()
+                    raw_actions[aid] = env_actions[aid]
         return env_actions, raw_actions
 
     # ------------------------------------------------------------------

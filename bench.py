@@ -465,8 +465,18 @@ def run_maddpg(args):
         )
 
     pop = DistributedPopulation(factory, pop_size)
-    envs = {s: SpeakerListenerVecEnv(num_envs, seed=100 + s) for s in pop.local_indices}
-    mems = {s: ReplayBuffer(200_000, device=device) for s in pop.local_indices}
+    use_torch_env = torch.cuda.is_available()
+    if use_torch_env:
+        from agilerl_amd.envs.torch_mpe import SpeakerListenerTorchVecEnv
+
+        envs = {s: SpeakerListenerTorchVecEnv(num_envs, device=device, seed=100 + s)
+                for s in pop.local_indices}
+        storage = device
+    else:
+        envs = {s: SpeakerListenerVecEnv(num_envs, seed=100 + s) for s in pop.local_indices}
+        storage = None
+    mems = {s: ReplayBuffer(100_000, device=device, storage_device=storage)
+            for s in pop.local_indices}
     obs_map = {s: envs[s].reset()[0] for s in pop.local_indices}
     agent_ids = probe.agents
 
@@ -478,10 +488,13 @@ def run_maddpg(args):
             for it in range(32):
                 env_actions, raw = agent.get_action(obs)
                 next_obs, rewards, term, trunc, info = env.step(env_actions)
+                if use_torch_env:
+                    done = {a: term[a].float() for a in agent_ids}
+                else:
+                    done = {a: term[a].astype(np.float32) for a in agent_ids}
                 mem.add(obs=obs, action=raw,
                         reward={a: rewards[a] for a in agent_ids},
-                        next_obs=next_obs,
-                        done={a: term[a].astype(np.float32) for a in agent_ids})
+                        next_obs=next_obs, done=done)
                 obs = next_obs
                 steps += num_envs
                 if len(mem) >= 2000 and it % agent.learn_step == 0:
