@@ -454,7 +454,7 @@ def run_maddpg(args):
     state = DistributedState.get()
     device = state.device
     pop_size = 8
-    num_envs = args.num_envs or 64
+    num_envs = args.num_envs or (2048 if torch.cuda.is_available() else 64)
     probe = SpeakerListenerVecEnv(1)
 
     def factory(index):
