@@ -49,7 +49,7 @@ def parse_args():
     p.add_argument("--warmup", type=int, default=2)
     p.add_argument(
         "--num-envs", type=int, default=None,
-        help="envs per agent (default: 16384 on GPU via the hipGraph collector, 64 on CPU)",
+        help="envs per agent (default: 65536 on GPU via the hipGraph collector, 64 on CPU)",
     )
     p.add_argument("--learn-step", type=int, default=LEARN_STEP)
     p.add_argument("--pop-size", type=int, default=POP_SIZE)
@@ -513,7 +513,7 @@ def run_maddpg(args):
 def main():
     args = parse_args()
     if args.num_envs is None and args.workload == "ppo":
-        args.num_envs = 16384 if torch.cuda.is_available() else NUM_ENVS
+        args.num_envs = 65536 if torch.cuda.is_available() else NUM_ENVS
     if args.workload == "grpo":
         return run_grpo(args)
     if args.workload == "dqn":
