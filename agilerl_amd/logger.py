@@ -78,11 +78,17 @@ class CSVLogger(Logger):
 
 class TensorboardLogger(Logger):
     def __init__(self, log_dir: str = "runs"):
-        from torch.utils.tensorboard import SummaryWriter
+        try:
+            from torch.utils.tensorboard import SummaryWriter
 
-        self.writer = SummaryWriter(log_dir)
+            self.writer = SummaryWriter(log_dir)
+        except ImportError:
+            self.writer = None
+            print("tensorboard not installed; TensorboardLogger disabled", file=sys.stderr)
 
     def log_report(self, report: Dict[str, Any]) -> None:
+        if self.writer is None:
+            return
         step = report.get("global_step", 0)
         for k, v in report.items():
             if isinstance(v, (int, float)):
@@ -94,7 +100,8 @@ class TensorboardLogger(Logger):
                     self.writer.add_scalar(f"agent_{idx}/{k}", v, step)
 
     def close(self) -> None:
-        self.writer.close()
+        if self.writer is not None:
+            self.writer.close()
 
 
 class WandbLogger(Logger):

@@ -6,6 +6,8 @@ from .utils import (
     run_selection_and_mutation,
     get_env_defined_actions,
     observation_space_channels_to_first,
+    consolidate_mutations,
+    log_gpu_memory_snapshot,
 )
 
 __all__ = [
@@ -16,4 +18,6 @@ __all__ = [
     "run_selection_and_mutation",
     "get_env_defined_actions",
     "observation_space_channels_to_first",
+    "consolidate_mutations",
+    "log_gpu_memory_snapshot",
 ]

@@ -100,3 +100,29 @@ def observation_space_channels_to_first(space):
         high = np.transpose(space.high, (2, 0, 1))
         return Box(low, high, dtype=space.dtype)
     return space
+
+
+def consolidate_mutations(population: List) -> Dict[str, int]:
+    """Histogram of applied mutations across the population
+    (reference utils/utils.py:1726)."""
+    counts: Dict[str, int] = {}
+    for agent in population:
+        counts[agent.mut] = counts.get(agent.mut, 0) + 1
+    return counts
+
+
+def log_gpu_memory_snapshot(prefix: str = "") -> Dict[str, float]:
+    """GPU memory observability (reference llm_utils.py:1152
+    log_cuda_memory_snapshot).  Returns/prints allocated/reserved GB."""
+    import torch
+
+    if not torch.cuda.is_available():
+        return {}
+    out = {
+        "allocated_gb": torch.cuda.memory_allocated() / 1e9,
+        "reserved_gb": torch.cuda.memory_reserved() / 1e9,
+        "max_allocated_gb": torch.cuda.max_memory_allocated() / 1e9,
+    }
+    print(f"[gpu-mem]{(' ' + prefix) if prefix else ''} " +
+          " ".join(f"{k}={v:.2f}" for k, v in out.items()), flush=True)
+    return out
