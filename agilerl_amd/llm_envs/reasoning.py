@@ -140,3 +140,28 @@ class ReasoningGym(LLMEnvBase):
             float(self.reward_fn(c, a)) for c, a in zip(completions, self._batch_answers)
         ]
         return np.asarray(rewards, dtype=np.float32)
+
+    @classmethod
+    def from_dataset(
+        cls,
+        dataset,
+        reward_fn,
+        tokenizer,
+        prompt_key: str = "question",
+        answer_key: str = "answer",
+        **kwargs,
+    ) -> "ReasoningGym":
+        """Build from a HF ``datasets.Dataset``, a local saved-dataset path,
+        or a list of dicts (reference HuggingFaceGym, llm_envs/base.py:93;
+        offline only — no hub access)."""
+        if isinstance(dataset, str):
+            from datasets import load_from_disk
+
+            dataset = load_from_disk(dataset)
+        if hasattr(dataset, "column_names"):
+            prompts = list(dataset[prompt_key])
+            answers = list(dataset[answer_key]) if answer_key in dataset.column_names else None
+        else:
+            prompts = [row[prompt_key] for row in dataset]
+            answers = [row.get(answer_key) for row in dataset]
+        return cls(prompts, answers, reward_fn, tokenizer, **kwargs)
