@@ -6,6 +6,7 @@ HuggingFaceGym = ReasoningGym
 from .sft import SFTGym, SyntheticSFTGym
 from .preference import PreferenceGym, SyntheticPreferenceGym
 from .multiturn import MultiTurnTokenEnv, TokenGuessEnv, SyncMultiTurnVecEnv
+from .search import SearchTool, FormatRewardWrapper, extract_answer
 
 __all__ = [
     "LLMEnvBase",
@@ -20,4 +21,7 @@ __all__ = [
     "MultiTurnTokenEnv",
     "TokenGuessEnv",
     "SyncMultiTurnVecEnv",
+    "SearchTool",
+    "FormatRewardWrapper",
+    "extract_answer",
 ]

@@ -414,3 +414,57 @@ class TestLLMUtils:
         pop = create_population("DQN", Box(-1, 1, (4,)), Discrete(2),
                                 population_size=3, INIT_HP={"batch_size": 32})
         assert len(pop) == 3 and pop[0].batch_size == 32
+
+
+class TestSearchAndFormat:
+    def test_search_tool(self):
+        from agilerl_amd.llm_envs import SearchTool
+
+        tool = SearchTool(["the cat sat", "dogs bark loudly", "cats and dogs"])
+        hits = tool("cat dogs", k=2)
+        assert len(hits) >= 1
+        assert any("cat" in h or "dogs" in h for h in hits)
+
+    def test_format_reward(self):
+        from agilerl_amd.llm_envs import FormatRewardWrapper
+
+        base = lambda c, a: 1.0 if c == a else 0.0
+        fn = FormatRewardWrapper(base, format_bonus=0.1)
+        assert fn("<answer>42</answer>", "42") == pytest.approx(1.1)
+        assert fn("42", "42") == pytest.approx(1.0)  # fallback to raw
+        strict = FormatRewardWrapper(base, require_format=True)
+        assert strict("42", "42") == 0.0
+
+    def test_from_dataset(self):
+        from datasets import Dataset
+
+        from agilerl_amd.llm_envs import HuggingFaceGym, ReasoningGym
+
+        assert HuggingFaceGym is ReasoningGym
+
+        class TinyTok:
+            pad_token_id = 0
+            pad_token = "<pad>"
+            eos_token = "<eos>"
+            chat_template = None
+
+            def __call__(self, texts, **kw):
+                ids = [[(hash(w) % 99) + 1 for w in t.split()][:6] for t in texts]
+                L = max(len(i) for i in ids)
+                out = torch.zeros(len(ids), L, dtype=torch.long)
+                am = torch.zeros(len(ids), L, dtype=torch.long)
+                for j, i in enumerate(ids):
+                    out[j, L - len(i):] = torch.tensor(i)
+                    am[j, L - len(i):] = 1
+                return {"input_ids": out, "attention_mask": am}
+
+            def batch_decode(self, seqs, **kw):
+                return [" ".join(str(int(t)) for t in s if t) for s in seqs]
+
+        ds = Dataset.from_dict({"question": ["a b", "c d"], "answer": ["x", "y"]})
+        gym = ReasoningGym.from_dataset(ds, lambda c, a: 0.5, TinyTok(),
+                                        data_batch_size=2, group_size=2)
+        p = gym.reset()
+        assert p["input_ids"].shape[0] == 4
+        seqs = torch.cat([p["input_ids"], torch.randint(1, 99, (4, 2))], dim=1)
+        assert (gym.score(seqs) == 0.5).all()
