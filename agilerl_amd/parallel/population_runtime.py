@@ -30,7 +30,64 @@ from ..hpo.mutation import Mutations
 from ..hpo.tournament import TournamentSelection
 from .state import DistributedState, barrier
 
-__all__ = ["DistributedPopulation"]
+__all__ = ["DistributedPopulation", "adopt_agent_state"]
+
+
+@torch.no_grad()
+def adopt_agent_state(old, new) -> bool:
+    """Copy ``new``'s state INTO ``old`` in place when architectures match.
+
+    Evolution clones replace agent objects, which invalidates their
+    captured hipGraphs (graphs are bound to the old parameter tensors) and
+    forces expensive recapture every round.  When the offspring kept the
+    parent architecture (the common case — arch-mutation probability is a
+    fraction), adopting its weights/optimizer-state/attributes into the
+    existing object keeps every captured graph valid.
+
+    Returns False (caller must rebuild) when shapes differ or the learning
+    rate changed (graphed capturable-Adam bakes the lr at capture time).
+    """
+    if type(old) is not type(new):
+        return False
+    names = old.registry.all_network_names()
+    for name in names:
+        so = getattr(old, name).state_dict()
+        sn = getattr(new, name).state_dict()
+        if so.keys() != sn.keys():
+            return False
+        for k in so:
+            if so[k].shape != sn[k].shape or so[k].dtype != sn[k].dtype:
+                return False
+    lr_changed = any(
+        getattr(old, cfg.lr_name, None) != getattr(new, cfg.lr_name, None)
+        for cfg in old.registry.optimizer_configs
+    )
+    for name in names:
+        so = getattr(old, name).state_dict()
+        sn = getattr(new, name).state_dict()
+        for k in so:
+            so[k].copy_(sn[k])
+    # optimizer moments: copy in place (same param ordering by construction)
+    for cfg in old.registry.optimizer_configs:
+        oo = getattr(old, cfg.name).optimizer
+        no = getattr(new, cfg.name).optimizer
+        if isinstance(oo, dict) or isinstance(no, dict):
+            continue
+        o_state, n_state = list(oo.state.values()), list(no.state.values())
+        if len(o_state) == len(n_state):
+            for os_, ns_ in zip(o_state, n_state):
+                for key, v in os_.items():
+                    nv = ns_.get(key)
+                    if torch.is_tensor(v) and torch.is_tensor(nv) and v.shape == nv.shape:
+                        v.copy_(nv)
+    for k, v in new.inspect_attributes().items():
+        setattr(old, k, v)
+    if lr_changed:
+        for attr in getattr(old, "_GRAPH_ATTRS", ()):  # force recapture
+            if hasattr(old, attr):
+                setattr(old, attr, None)
+        old._reinit_optimizers()
+    return True
 
 
 class DistributedPopulation:

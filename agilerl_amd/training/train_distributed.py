@@ -152,8 +152,18 @@ def train_on_policy_distributed(
             break
 
         if tournament is not None:
+            from ..parallel.population_runtime import adopt_agent_state
+
+            old_agents = dict(pop.agents)
             pop.evolve(tournament, mutation)
             for slot in pop.local_indices:
+                old = old_agents.get(slot)
+                new = pop.agents[slot]
+                if old is not None and old is not new and adopt_agent_state(old, new):
+                    # same architecture: keep the old object so its captured
+                    # hipGraphs (collect + learn) survive the evolution round
+                    pop.agents[slot] = old
+                    continue
                 if slot not in envs:
                     init_slot(slot)
                 else:

@@ -174,9 +174,15 @@ class BenchRunner:
                 agent.fitness.append(float(np.mean(self.fit_window[slot])))
         self.step_count += 1
         if self.step_count % EVO_EVERY == 0:
+            from agilerl_amd.parallel.population_runtime import adopt_agent_state
+
+            old_agents = dict(self.pop.agents)
             self.pop.evolve(self.tournament, self.mutations)
-            # re-bind env/buffer state for (possibly new) local slots
             for slot in self.pop.local_indices:
+                old, new = old_agents.get(slot), self.pop.agents[slot]
+                if old is not None and old is not new and adopt_agent_state(old, new):
+                    self.pop.agents[slot] = old  # captured graphs stay valid
+                    continue
                 if slot not in self.envs:
                     self._init_slot(slot)
                 else:

@@ -185,3 +185,31 @@ def test_train_on_policy_distributed_gloo():
         assert results[0][0] == [0, 2]
         assert results[1][0] == [1, 3]
         assert results[0][1] >= 1
+
+
+def test_adopt_agent_state():
+    from agilerl_amd.algorithms.ppo import PPO
+    from agilerl_amd.parallel.population_runtime import adopt_agent_state
+    from agilerl_amd.spaces import Box, Discrete
+
+    torch.manual_seed(0)
+    a = PPO(Box(-1, 1, (4,)), Discrete(2))
+    b = a.clone(1)
+    with torch.no_grad():
+        for p in b.actor.parameters():
+            p.add_(1.0)
+    b.fitness = [5.0]
+    assert adopt_agent_state(a, b)
+    x = torch.randn(3, 4)
+    assert torch.allclose(a.actor(x), b.actor(x))
+    assert a.fitness == [5.0]
+    # architecture mismatch refuses
+    c = a.clone(2)
+    c.apply_architecture_mutation("encoder.add_node", numb_new_nodes=16)
+    assert not adopt_agent_state(a, c)
+    # lr change adopts but invalidates captured graphs
+    d = a.clone(3)
+    d.lr = 1e-5
+    a._learn_graph = object()
+    assert adopt_agent_state(a, d)
+    assert a._learn_graph is None and a.lr == 1e-5
