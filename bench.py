@@ -80,12 +80,26 @@ class BenchRunner:
         # minibatch sized to the rollout so big env counts keep the GPU in
         # few large GEMMs instead of hundreds of small update steps
         mb_size = max(2048, args.num_envs * args.learn_step // 16)
+        # HP-mutation bounds sized to THIS config: the generic defaults cap
+        # batch_size at 4096, so one rl_hp mutation would shred the 524k
+        # minibatch into 128x more update steps and the slowdown spreads
+        # through clones (observed: +7 s/step per evolution round)
+        from agilerl_amd.algorithms.core.registry import HyperparameterConfig, RLParameter
+
+        hp = HyperparameterConfig(
+            lr=RLParameter(min=1e-5, max=1e-2),
+            batch_size=RLParameter(min=mb_size // 4, max=mb_size * 2, dtype=int),
+            clip_coef=RLParameter(min=0.05, max=0.4),
+            ent_coef=RLParameter(min=1e-4, max=0.05),
+            update_epochs=RLParameter(min=1, max=4, dtype=int),
+        )
 
         def factory(index: int) -> PPO:
             return PPO(
                 observation_space=LunarLanderVecEnv(1).single_observation_space,
                 action_space=LunarLanderVecEnv(1).single_action_space,
                 index=index,
+                hp_config=hp,
                 learn_step=args.learn_step,
                 batch_size=mb_size,
                 lr=3e-4,
