@@ -334,8 +334,14 @@ class EvolvableAlgorithm(metaclass=AlgorithmMeta):
     def to_device(self, device: str) -> "EvolvableAlgorithm":
         self.device = device
         for name, net in self.evolvable_networks().items():
-            net.device = device
-            setattr(self, name, net.to(device))
+            net = net.to(device)
+            # every evolvable sub-module rebuilds onto ITS OWN device attr
+            # during mutations — refresh them all or post-move mutations
+            # would recreate layers on the old device
+            for mod in net.modules():
+                if hasattr(mod, "device"):
+                    mod.device = device
+            setattr(self, name, net)
         self._reinit_optimizers()
         return self
 

@@ -182,6 +182,10 @@ class DistributedPopulation:
             new_agents: Dict[int, object] = {}
             for slot, parent in enumerate(plan):
                 new_agents[slot] = parents[parent].clone(index=slot)
+            kept = set(id(a) for a in new_agents.values())
+            for agent in parents.values():  # free dropped LLM adapter slots
+                if id(agent) not in kept and hasattr(agent, "clean_up"):
+                    agent.clean_up()
             self.agents = new_agents
             self.local_indices = list(range(self.pop_size))
             return
@@ -211,5 +215,9 @@ class DistributedPopulation:
                 agent._apply_checkpoint(parent_ckpts[parent])
                 agent.index = slot
                 new_agents[slot] = agent
+        kept = set(id(a) for a in new_agents.values())
+        for agent in self.agents.values():  # free dropped LLM adapter slots
+            if id(agent) not in kept and hasattr(agent, "clean_up"):
+                agent.clean_up()
         self.agents = new_agents
         barrier()

@@ -187,3 +187,13 @@ class TestSharedEncoders:
         assert q(x).shape == (2, 4)
         q.apply_mutation("encoder.add_block")
         assert q(x).shape == (2, 4)
+
+
+class TestToDevice:
+    def test_device_attrs_refresh(self):
+        """Mutations after to_device must rebuild on the new device."""
+        agent = DQN(Box(-1, 1, (4,)), Discrete(2))
+        agent.to_device("cpu")  # round-trip on CPU still exercises the walk
+        assert agent.actor.encoder.device == "cpu"
+        agent.apply_architecture_mutation("encoder.add_node", numb_new_nodes=16)
+        assert next(agent.actor.parameters()).device.type == "cpu"

@@ -468,3 +468,29 @@ class TestSearchAndFormat:
         assert p["input_ids"].shape[0] == 4
         seqs = torch.cat([p["input_ids"], torch.randint(1, 99, (4, 2))], dim=1)
         assert (gym.score(seqs) == 0.5).all()
+
+
+class TestAdapterCleanupOnEvolve:
+    def test_distributed_population_frees_adapters(self):
+        from agilerl_amd.algorithms.llm.grpo import GRPO
+        from agilerl_amd.hpo import TournamentSelection
+        from agilerl_amd.llm.lora import iter_lora_modules
+        from agilerl_amd.parallel import DistributedPopulation
+
+        pop_agents = GRPO.population(3, model_config=dict(TINY), dtype=torch.float32,
+                                     lora_config={"r": 4}, group_size=2)
+        model = pop_agents[0].model
+        dp = DistributedPopulation.__new__(DistributedPopulation)
+        from agilerl_amd.parallel import DistributedState
+
+        dp.state = DistributedState(0, 0, 1, "none", "cpu")
+        dp.pop_size = 3
+        dp.agent_factory = None
+        dp.agents = {i: a for i, a in enumerate(pop_agents)}
+        dp.local_indices = [0, 1, 2]
+        dp.evo_step = 0
+        for i, a in dp.agents.items():
+            a.fitness.append(float(i))
+        dp.evolve(TournamentSelection(3, True, rng=__import__("numpy").random.default_rng(0)))
+        n_adapters = len(next(iter_lora_modules(model))[1].lora_A)
+        assert n_adapters == 3  # exactly pop_size slots remain (no leak)
