@@ -58,6 +58,7 @@ class MADDPG(MultiAgentRLAlgorithm):
         tau: float = 1e-2,
         expl_noise: float = 0.1,
         latent_dim: int = 64,
+        shared_agent_groups: Optional[List[List[str]]] = None,
         device: str = "cpu",
     ):
         super().__init__(
@@ -65,6 +66,7 @@ class MADDPG(MultiAgentRLAlgorithm):
             learn_step=learn_step, device=device,
             hp_config=hp_config or default_hp_config(), name="MADDPG",
         )
+        self.shared_agent_groups = shared_agent_groups
         self.batch_size = int(batch_size)
         self.lr_actor = float(lr_actor)
         self.lr_critic = float(lr_critic)
@@ -79,17 +81,24 @@ class MADDPG(MultiAgentRLAlgorithm):
         self.joint_action_dim = sum(self._raw_action_dim(sp) for sp in self.action_spaces.values())
         joint_space = Box(-np.inf, np.inf, (self.joint_obs_dim,))
 
-        self.actors = ModuleDict(
-            {
-                aid: DeterministicActor(
-                    self.observation_spaces[aid], self.action_spaces[aid],
+        # grouped-agent net sharing (reference base.py:2330): agents inside a
+        # group share ONE actor module object (spaces must match)
+        group_of = {}
+        for group in shared_agent_groups or []:
+            for aid in group:
+                group_of[aid] = group[0]
+        actor_modules: Dict[str, DeterministicActor] = {}
+        actors = {}
+        for aid in self.agent_ids:
+            leader = group_of.get(aid, aid)
+            if leader not in actor_modules:
+                actor_modules[leader] = DeterministicActor(
+                    self.observation_spaces[leader], self.action_spaces[leader],
                     encoder_config=net_config, head_config=head_config,
                     latent_dim=latent_dim, device=device,
                 )
-                for aid in self.agent_ids
-            },
-            device=device,
-        )
+            actors[aid] = actor_modules[leader]
+        self.actors = ModuleDict(actors, device=device)
         self.actor_targets = self.actors.clone()
         self.critics = ModuleDict(
             {

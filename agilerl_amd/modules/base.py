@@ -271,7 +271,11 @@ class ModuleDict(EvolvableModule):
 
     def apply_mutation(self, name: str, **choices) -> Optional[dict]:
         out: Optional[dict] = None
+        seen = set()  # grouped agents may share one module object
         for mod in self._modules_dict.values():
+            if id(mod) in seen:
+                continue
+            seen.add(id(mod))
             if out is None:
                 out = mod.apply_mutation(name, **choices)
                 if isinstance(out, dict):
@@ -282,7 +286,15 @@ class ModuleDict(EvolvableModule):
         return out
 
     def clone(self) -> "ModuleDict":
-        return ModuleDict({k: v.clone() for k, v in self._modules_dict.items()}, device=self.device)
+        # preserve object sharing: keys mapping to the same module keep
+        # mapping to ONE clone (grouped-agent net sharing)
+        clones: Dict[int, EvolvableModule] = {}
+        out = {}
+        for k, v in self._modules_dict.items():
+            if id(v) not in clones:
+                clones[id(v)] = v.clone()
+            out[k] = clones[id(v)]
+        return ModuleDict(out, device=self.device)
 
     @property
     def init_dict(self) -> Dict[str, Any]:

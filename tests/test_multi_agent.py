@@ -151,3 +151,23 @@ class TestIPPO:
             verbose=False,
         )
         assert len(hist) >= 1
+
+
+class TestGroupedAgents:
+    def test_shared_actor_group(self):
+        env = SimpleSpreadVecEnv(num_envs=2, seed=0)
+        agent = MADDPG(env.observation_spaces, env.action_spaces, agent_ids=env.agents,
+                       shared_agent_groups=[["agent_0", "agent_1"]],
+                       batch_size=16, net_config={"arch": "mlp", "hidden_size": [16]})
+        assert agent.actors["agent_0"] is agent.actors["agent_1"]
+        assert agent.actors["agent_2"] is not agent.actors["agent_0"]
+        # mutation applies once to the shared module
+        h0 = sum(agent.actors["agent_0"].encoder.hidden_size)
+        agent.apply_architecture_mutation("encoder.add_node", numb_new_nodes=16)
+        assert sum(agent.actors["agent_0"].encoder.hidden_size) - h0 == 16
+        # clone preserves sharing
+        clone = agent.clone(1)
+        assert clone.actors["agent_0"] is clone.actors["agent_1"]
+        obs, _ = env.reset()
+        ea, raw = agent.get_action(obs)
+        assert ea["agent_0"].shape == (2,)
