@@ -56,6 +56,7 @@ class IPPO(MultiAgentRLAlgorithm):
         update_epochs: int = 4,
         max_grad_norm: float = 0.5,
         latent_dim: int = 64,
+        shared_agent_groups: Optional[List[List[str]]] = None,
         device: str = "cpu",
     ):
         super().__init__(
@@ -74,28 +75,31 @@ class IPPO(MultiAgentRLAlgorithm):
         self.max_grad_norm = float(max_grad_norm)
         self.net_config = net_config
         self.latent_dim = latent_dim
+        self.shared_agent_groups = shared_agent_groups
 
-        self.actors = ModuleDict(
-            {
-                aid: StochasticActor(
-                    self.observation_spaces[aid], self.action_spaces[aid],
+        # homogeneous agent groups share one actor/critic (reference
+        # grouped-agent sharing, base.py:2330)
+        group_of = {}
+        for group in shared_agent_groups or []:
+            for aid in group:
+                group_of[aid] = group[0]
+        actor_mods, critic_mods, actors, critics = {}, {}, {}, {}
+        for aid in self.agent_ids:
+            leader = group_of.get(aid, aid)
+            if leader not in actor_mods:
+                actor_mods[leader] = StochasticActor(
+                    self.observation_spaces[leader], self.action_spaces[leader],
                     encoder_config=net_config, head_config=head_config,
                     latent_dim=latent_dim, device=device,
                 )
-                for aid in self.agent_ids
-            },
-            device=device,
-        )
-        self.critics = ModuleDict(
-            {
-                aid: ValueNetwork(
-                    self.observation_spaces[aid], encoder_config=net_config,
+                critic_mods[leader] = ValueNetwork(
+                    self.observation_spaces[leader], encoder_config=net_config,
                     head_config=head_config, latent_dim=latent_dim, device=device,
                 )
-                for aid in self.agent_ids
-            },
-            device=device,
-        )
+            actors[aid] = actor_mods[leader]
+            critics[aid] = critic_mods[leader]
+        self.actors = ModuleDict(actors, device=device)
+        self.critics = ModuleDict(critics, device=device)
         self.optimizer = OptimizerWrapper(
             torch.optim.Adam, [self.actors, self.critics], lr=self.lr
         )

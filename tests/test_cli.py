@@ -1,0 +1,76 @@
+"""CLI entry-point tests (subprocess)."""
+
+import json
+import os
+import subprocess
+import sys
+
+import pytest
+
+ROOT = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+
+
+def test_train_cli(tmp_path):
+    manifest = tmp_path / "m.yaml"
+    manifest.write_text(
+        """
+algorithm:
+  name: DQN
+  hyperparameters: {batch_size: 32, lr: 0.001}
+environment: {env_id: CartPole-v1, num_envs: 4}
+network:
+  arch: mlp
+  encoder_config: {hidden_size: [16]}
+training: {max_steps: 300, pop_size: 1, evo_steps: 150, eval_loop: 1}
+"""
+    )
+    out = subprocess.run(
+        [sys.executable, "-m", "agilerl_amd.train", str(manifest), "--device", "cpu",
+         "--csv", str(tmp_path / "log.csv")],
+        cwd=ROOT, capture_output=True, text=True, timeout=300,
+    )
+    assert out.returncode == 0, out.stderr[-2000:]
+    assert (tmp_path / "log.csv").exists()
+
+
+def test_arena_cli(tmp_path):
+    manifest = tmp_path / "m.yaml"
+    manifest.write_text(
+        """
+algorithm:
+  name: DQN
+  hyperparameters: {batch_size: 16}
+environment: {env_id: CartPole-v1, num_envs: 2}
+training: {max_steps: 100, pop_size: 1, evo_steps: 50, eval_loop: 1}
+"""
+    )
+    ws = str(tmp_path / "arena")
+    out = subprocess.run(
+        [sys.executable, "-m", "agilerl_amd.arena.cli", "--workspace", ws,
+         "submit", str(manifest)],
+        cwd=ROOT, capture_output=True, text=True, timeout=300,
+    )
+    assert out.returncode == 0, out.stderr[-2000:]
+    assert "completed" in out.stdout
+    out2 = subprocess.run(
+        [sys.executable, "-m", "agilerl_amd.arena.cli", "--workspace", ws, "list"],
+        cwd=ROOT, capture_output=True, text=True, timeout=60,
+    )
+    assert out2.stdout.strip().startswith("exp-")
+
+
+def test_bench_contract_cpu():
+    """bench.py emits exactly one JSON line with the driver-contract keys."""
+    out = subprocess.run(
+        [sys.executable, "bench.py", "--steps", "2", "--warmup", "1",
+         "--num-envs", "8", "--pop-size", "2"],
+        cwd=ROOT, capture_output=True, text=True, timeout=600,
+    )
+    assert out.returncode == 0, out.stderr[-2000:]
+    lines = [l for l in out.stdout.strip().splitlines() if l.startswith("{")]
+    assert len(lines) == 1
+    payload = json.loads(lines[0])
+    for key in ("metric", "value", "unit", "n_gpus", "steps", "warmup",
+                "ms_per_step", "higher_is_better", "scaling", "vs_baseline",
+                "dtype", "data", "config"):
+        assert key in payload, key

@@ -171,3 +171,15 @@ class TestGroupedAgents:
         obs, _ = env.reset()
         ea, raw = agent.get_action(obs)
         assert ea["agent_0"].shape == (2,)
+
+    def test_ippo_shared_groups(self):
+        env = SimpleSpreadVecEnv(num_envs=2, seed=0)
+        agent = IPPO(env.observation_spaces, env.action_spaces, agent_ids=env.agents,
+                     shared_agent_groups=[["agent_0", "agent_1", "agent_2"]],
+                     learn_step=8, batch_size=16,
+                     net_config={"arch": "mlp", "hidden_size": [16]})
+        assert agent.actors["agent_0"] is agent.actors["agent_2"]
+        assert agent.critics["agent_0"] is agent.critics["agent_1"]
+        obs, _ = env.reset()
+        ea, lp, v = agent.get_action(obs)
+        assert ea["agent_0"].shape == (2,)
