@@ -35,6 +35,13 @@ def _importable(name: str) -> bool:
 HAS_LLM_DEPENDENCIES = _importable("transformers")
 HAS_HIP_KERNELS = None  # resolved lazily below
 
+# Reference-familiar flags: these stacks are REPLACED by first-party
+# MI355X-native code (HIP kernels, RCCL layer, in-framework LoRA), so the
+# flags exist for API familiarity and are always False.
+HAS_VLLM = False
+HAS_DEEPSPEED = False
+HAS_LIGER_KERNEL = False
+
 
 class AgentType(str, Enum):
     RL = "rl"
@@ -76,6 +83,9 @@ __all__ = [
     "AgentType",
     "HAS_LLM_DEPENDENCIES",
     "HAS_HIP_KERNELS",
+    "HAS_VLLM",
+    "HAS_DEEPSPEED",
+    "HAS_LIGER_KERNEL",
     "LocalTrainer",
     "Trainer",
     "Population",

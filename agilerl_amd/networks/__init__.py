@@ -1,5 +1,5 @@
 from .base import EvolvableNetwork, build_encoder, get_default_encoder_config
-from .distributions import ActionDistribution
+from .distributions import ActionDistribution, EvolvableDistribution
 from .q_networks import QNetwork, RainbowQNetwork, ContinuousQNetwork
 from .actors import DeterministicActor, StochasticActor
 from .value_networks import ValueNetwork
@@ -9,6 +9,7 @@ __all__ = [
     "build_encoder",
     "get_default_encoder_config",
     "ActionDistribution",
+    "EvolvableDistribution",
     "QNetwork",
     "RainbowQNetwork",
     "ContinuousQNetwork",

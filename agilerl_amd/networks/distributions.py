@@ -15,7 +15,7 @@ from torch.distributions import Bernoulli, Categorical, Normal
 
 from ..spaces import Box, Discrete, MultiBinary, MultiDiscrete, Space
 
-__all__ = ["ActionDistribution"]
+__all__ = ["ActionDistribution", "EvolvableDistribution"]
 
 
 def _masked_logits(logits: torch.Tensor, mask: Optional[torch.Tensor]) -> torch.Tensor:
@@ -133,3 +133,7 @@ class ActionDistribution(nn.Module):
         if isinstance(dist, Bernoulli):
             return (dist.logits > 0).float()
         return dist.logits.argmax(-1)
+
+
+# reference class name (agilerl/networks/distributions.py:119)
+EvolvableDistribution = ActionDistribution

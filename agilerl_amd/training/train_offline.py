@@ -99,3 +99,13 @@ def train_offline(
         if max_wall_seconds is not None and _time.time() - t_start > max_wall_seconds:
             break
     return population.agents, fitness_history
+
+
+def save_transitions(dataset: Dict[str, np.ndarray], path: str) -> None:
+    """Persist a transition dataset as compressed npz (offline-RL format)."""
+    np.savez_compressed(path, **dataset)
+
+
+def load_transitions(path: str) -> Dict[str, np.ndarray]:
+    with np.load(path) as z:
+        return {k: z[k] for k in z.files}
