@@ -85,3 +85,110 @@ def test_ppo_mutations(method):
     collect_rollouts(agent, env, buf, 8)
     stats = agent.learn(buf)
     assert np.isfinite(stats["policy_loss"])
+
+
+# ---------------------------------------------------------------------------
+# multi-agent sweep: ModuleDict groups + shared-agent dedup under mutation
+# ---------------------------------------------------------------------------
+
+def _ma_agent(cls, shared=False):
+    from agilerl_amd.envs.mpe import SpeakerListenerVecEnv
+
+    env = SpeakerListenerVecEnv(num_envs=2, seed=0)
+    kw = {}
+    if shared:
+        kw["shared_agent_groups"] = [["speaker_0"], ["listener_0"]]
+    agent = cls(
+        env.observation_spaces, env.action_spaces, agent_ids=env.agents,
+        batch_size=16, net_config={"arch": "mlp", "hidden_size": [16]}, **kw,
+    )
+    return env, agent
+
+
+def _ma_offpolicy_batch(env, agent):
+    from agilerl_amd.components import ReplayBuffer
+
+    buf = ReplayBuffer(200)
+    obs, _ = env.reset()
+    for _ in range(24):
+        env_actions, raw = agent.get_action(obs)
+        next_obs, rewards, term, trunc, _ = env.step(env_actions)
+        buf.add(obs=obs, action=raw,
+                reward={a: rewards[a] for a in env.agents},
+                next_obs=next_obs,
+                done={a: term[a].astype(np.float32) for a in env.agents})
+        obs = next_obs
+    return buf.sample(16)
+
+
+MA_OFFPOLICY = []
+try:
+    from agilerl_amd.algorithms import MADDPG, MATD3
+
+    MA_OFFPOLICY = [MADDPG, MATD3]
+except ImportError:  # pragma: no cover
+    pass
+
+
+@pytest.mark.parametrize("cls", MA_OFFPOLICY, ids=lambda c: c.__name__)
+def test_multiagent_mutation_sweep(cls):
+    env, base = _ma_agent(cls)
+    for method in base.mutation_methods:
+        env, agent = _ma_agent(cls)
+        agent.apply_architecture_mutation(method)
+        # every target ModuleDict mirrors its eval ModuleDict per agent id
+        for group in agent.registry.groups:
+            ev = getattr(agent, group.eval_network)
+            for shared in group.shared_networks:
+                sh = getattr(agent, shared)
+                for aid in ev.keys():
+                    n_ev = sum(p.numel() for p in ev[aid].parameters())
+                    n_sh = sum(p.numel() for p in sh[aid].parameters())
+                    assert n_ev == n_sh, f"{cls.__name__}.{method}[{aid}]"
+        loss = agent.learn(_ma_offpolicy_batch(env, agent))
+        assert np.isfinite(loss)
+        clone = agent.clone(index=3)
+        obs, _ = env.reset()
+        a1, _ = agent.get_action(obs, training=False)
+        a2, _ = clone.get_action(obs, training=False)
+        for aid in env.agents:
+            np.testing.assert_array_equal(a1[aid], a2[aid])
+
+
+def test_ippo_mutation_sweep():
+    from agilerl_amd.algorithms import IPPO
+    from agilerl_amd.components import RolloutBuffer
+    from agilerl_amd.training.train_multi_agent_on_policy import _collect_ma_rollout
+
+    env, base = _ma_agent(IPPO)
+    for method in base.mutation_methods:
+        env, agent = _ma_agent(IPPO)
+        agent.learn_step = 8
+        agent.apply_architecture_mutation(method)
+        bufs = {
+            aid: RolloutBuffer(8, env.num_envs, gamma=agent.gamma,
+                               gae_lambda=agent.gae_lambda)
+            for aid in env.agents
+        }
+        _collect_ma_rollout(agent, env, bufs, 8, None)
+        stats = agent.learn(bufs)
+        assert np.isfinite(stats["policy_loss"])
+
+
+def test_maddpg_shared_groups_stay_shared_after_mutation():
+    from agilerl_amd.algorithms import IPPO
+
+    env, agent = _ma_agent(IPPO, shared=False)
+    # same-space agents sharing one module: build a 2-listener env stand-in
+    obs_sp = {"a_0": env.observation_spaces["listener_0"],
+              "a_1": env.observation_spaces["listener_0"]}
+    act_sp = {"a_0": env.action_spaces["listener_0"],
+              "a_1": env.action_spaces["listener_0"]}
+    agent = IPPO(obs_sp, act_sp, agent_ids=["a_0", "a_1"],
+                 shared_agent_groups=[["a_0", "a_1"]],
+                 net_config={"arch": "mlp", "hidden_size": [16]})
+    assert agent.actors["a_0"] is agent.actors["a_1"]
+    for method in agent.mutation_methods:
+        agent.apply_architecture_mutation(method)
+        assert agent.actors["a_0"] is agent.actors["a_1"], method
+        assert agent.critics["a_0"] is agent.critics["a_1"], method
