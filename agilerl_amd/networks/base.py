@@ -125,6 +125,18 @@ class EvolvableNetwork(EvolvableModule):
         """Space-aware preprocessing (one-hot for Discrete etc.)."""
         space = self.observation_space
         if isinstance(space, (DictSpace, TupleSpace)):
+            dev = self.device if isinstance(self.device, torch.device) else torch.device(self.device)
+
+            def _leaf(x):
+                if not isinstance(x, torch.Tensor):
+                    x = torch.as_tensor(np.asarray(x))
+                x = x.to(dev)
+                return x.float() if not x.is_floating_point() else x
+
+            if isinstance(obs, dict):
+                return {k: _leaf(v) for k, v in obs.items()}
+            if isinstance(obs, (tuple, list)):
+                return type(obs)(_leaf(v) for v in obs)
             return obs
         if not isinstance(obs, torch.Tensor):
             obs = torch.as_tensor(np.asarray(obs))
