@@ -99,6 +99,9 @@ class ReplayBuffer:
                 data = data.to_dict()
         else:
             data = kwargs
+        for key in ("obs", "next_obs"):
+            if isinstance(data.get(key), (tuple, list)):
+                data[key] = {str(i): v for i, v in enumerate(data[key])}
         data, batch = self._coerce(data)
         if self._storage is None:
             self.num_envs = batch

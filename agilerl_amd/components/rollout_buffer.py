@@ -82,6 +82,10 @@ class RolloutBuffer:
         hidden_state=None,
         **extras,
     ) -> None:
+        if isinstance(obs, (tuple, list)):
+            # tuple observation spaces are stored as index-keyed dicts; the
+            # multi-input encoder accepts either form
+            obs = {str(i): v for i, v in enumerate(obs)}
         data = {
             "obs": obs,
             "action": action,

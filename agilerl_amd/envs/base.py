@@ -55,6 +55,15 @@ class VecEnv:
         return None
 
 
+def _copy_obs(obs):
+    """Deep-ish copy of a (possibly dict/tuple structured) observation."""
+    if isinstance(obs, dict):
+        return {k: _copy_obs(v) for k, v in obs.items()}
+    if isinstance(obs, (tuple, list)):
+        return type(obs)(_copy_obs(v) for v in obs)
+    return obs.copy() if hasattr(obs, "copy") else obs
+
+
 class BatchedVecEnv(VecEnv):
     """Base for natively-batched numpy envs with auto-reset.
 
@@ -107,7 +116,7 @@ class BatchedVecEnv(VecEnv):
         self._ep_return += reward
         if done.any():
             info = dict(info)
-            info["final_observation"] = obs.copy()
+            info["final_observation"] = _copy_obs(obs)
             info["episode_return"] = self._ep_return[done].copy()
             self._reset_rows(done)
             self._elapsed[done] = 0
