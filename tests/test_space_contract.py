@@ -221,3 +221,28 @@ def test_ppo_uint8_image_observations():
     obs, done, _ = collect_rollouts(agent, env, buf, 4)
     stats = agent.learn(buf)
     assert np.isfinite(stats["policy_loss"])
+
+
+def test_dqn_action_mask_respected():
+    torch.manual_seed(0), np.random.seed(0)
+    agent = DQN(Box(-1.0, 1.0, (3,)), Discrete(4), net_config=dict(NET))
+    obs = np.random.randn(16, 3).astype(np.float32)
+    mask = np.zeros((16, 4), dtype=bool)
+    mask[:, 2] = True  # only action 2 legal
+    for eps in (0.0, 1.0):
+        actions = agent.get_action(obs, epsilon=eps, action_mask=mask)
+        assert (actions == 2).all(), f"epsilon={eps}: {actions}"
+
+
+def test_ppo_action_mask_respected_and_stored():
+    torch.manual_seed(0), np.random.seed(0)
+    agent = PPO(Box(-1.0, 1.0, (3,)), Discrete(4), net_config=dict(NET),
+                learn_step=8, batch_size=16)
+    obs = np.random.randn(8, 3).astype(np.float32)
+    mask = np.zeros((8, 4), dtype=bool)
+    mask[:, 1] = True
+    for _ in range(5):
+        action, logp, ent, value = agent.get_action(obs, action_mask=mask)
+        assert (action == 1).all()
+    det = agent.get_action(obs, action_mask=mask, training=False)
+    assert (det == 1).all()
