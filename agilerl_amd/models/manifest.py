@@ -154,6 +154,9 @@ class MutationSpec(_Base):
     activation_selection: List[str] = Field(default_factory=lambda: ["ReLU", "ELU", "GELU"])
     mutate_elite: bool = True
     rand_seed: Optional[int] = None
+    # per-hyperparameter mutation bounds: {hp_name: {min: .., max: ..}}
+    # (reference manifest.py mutation.rl_hp_selection -> RLParameter ranges)
+    rl_hp_selection: Dict[str, Dict[str, float]] = Field(default_factory=dict)
 
 
 class TournamentSelectionSpec(_Base):

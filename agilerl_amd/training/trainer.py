@@ -132,7 +132,28 @@ class LocalTrainer(Trainer):
             )
         if resume:
             self._resume_population(pop, resume)
+        self._apply_hp_bounds(pop)
         return pop
+
+    def _apply_hp_bounds(self, pop: List) -> None:
+        """Override per-HP mutation bounds from manifest
+        mutation.rl_hp_selection (reference manifest rl_hp_selection ->
+        RLParameter ranges, registry.py:134)."""
+        bounds = getattr(self.manifest.mutation, "rl_hp_selection", None)
+        if not bounds:
+            return
+        for agent in pop:
+            hp = getattr(agent, "hp_config", None)
+            if hp is None:
+                continue
+            for name, rng in bounds.items():
+                param = hp.config.get(name) if hasattr(hp, "config") else None
+                if param is None:
+                    continue
+                if "min" in rng:
+                    param.min = type(param.min)(rng["min"])
+                if "max" in rng:
+                    param.max = type(param.max)(rng["max"])
 
     @staticmethod
     def _resume_population(pop: List, base_path: str) -> None:

@@ -172,3 +172,25 @@ class TestResume:
         env = trainer2._make_env()
         pop2 = trainer2._make_population(env)
         assert pop2[0].steps[-1] > 0  # carried over from the checkpoint
+
+
+def test_rl_hp_selection_bounds_applied():
+    """manifest mutation.rl_hp_selection overrides per-HP mutation ranges."""
+    from agilerl_amd.training import LocalTrainer
+
+    manifest = {
+        "algorithm": {"name": "DQN", "hyperparameters": {"batch_size": 32}},
+        "environment": {"type": "gym", "env_id": "CartPole-v1", "num_envs": 4},
+        "network": {"arch": "mlp", "encoder_config": {"hidden_size": [16]}},
+        "mutation": {"rl_hp_selection": {"lr": {"min": 1e-4, "max": 1e-3},
+                                         "batch_size": {"min": 16, "max": 64}}},
+        "training": {"max_steps": 100, "pop_size": 2, "evo_steps": 50},
+    }
+    trainer = LocalTrainer.from_manifest(manifest)
+    env = trainer._make_env()
+    pop = trainer._make_population(env)
+    for agent in pop:
+        lr = agent.hp_config.config["lr"]
+        assert lr.min == pytest.approx(1e-4) and lr.max == pytest.approx(1e-3)
+        bs = agent.hp_config.config["batch_size"]
+        assert bs.min == 16 and bs.max == 64 and isinstance(bs.min, int)
