@@ -213,3 +213,48 @@ def test_adopt_agent_state():
     a._learn_graph = object()
     assert adopt_agent_state(a, d)
     assert a._learn_graph is None and a.lr == 1e-5
+
+
+def _nonfinite_worker(rank, world_size, port, results):
+    os.environ["RANK"] = str(rank)
+    os.environ["LOCAL_RANK"] = str(rank)
+    os.environ["WORLD_SIZE"] = str(world_size)
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+
+    from agilerl_amd.parallel import DistributedState
+
+    DistributedState.reset()
+    DistributedState.get()
+
+    class _Carrier:
+        device = "cpu"
+        from agilerl_amd.algorithms.llm.base import LLMAlgorithm
+
+        raise_if_loss_not_finite_on_any_rank = LLMAlgorithm.raise_if_loss_not_finite_on_any_rank
+
+    carrier = _Carrier()
+    # only rank 1's local loss is NaN, but BOTH ranks must raise
+    loss = torch.tensor(float("nan") if rank == 1 else 0.5)
+    try:
+        carrier.raise_if_loss_not_finite_on_any_rank(loss)
+        results[rank] = "no_raise"
+    except RuntimeError:
+        results[rank] = "raised"
+    torch.distributed.barrier()
+    torch.distributed.destroy_process_group()
+
+
+def test_nonfinite_loss_aborts_all_ranks():
+    """coordinated abort — a NaN on one rank raises on every
+    rank, so no healthy rank hangs in the next collective."""
+    ctx = mp.get_context("spawn")
+    results = ctx.Manager().dict()
+    port = _find_free_port()
+    procs = [ctx.Process(target=_nonfinite_worker, args=(r, 2, port, results)) for r in range(2)]
+    for p in procs:
+        p.start()
+    for p in procs:
+        p.join(120)
+        assert p.exitcode == 0
+    assert results[0] == "raised" and results[1] == "raised"
