@@ -8,11 +8,12 @@ from __future__ import annotations
 
 import csv
 import os
+import warnings
 import sys
 import time
 from typing import Any, Dict, List, Optional
 
-__all__ = ["Logger", "StdOutLogger", "CSVLogger", "TensorboardLogger", "WandbLogger", "make_loggers"]
+__all__ = ["Logger", "StdOutLogger", "CSVLogger", "TensorboardLogger", "WandbLogger", "PrometheusLogger", "make_loggers"]
 
 
 class Logger:
@@ -74,6 +75,54 @@ class CSVLogger(Logger):
     def close(self) -> None:
         if self._file:
             self._file.close()
+
+
+class PrometheusLogger(Logger):
+    """Exposes population metrics as Prometheus gauges on an HTTP endpoint.
+
+    Production-serving observability sink (prometheus_client is a hard
+    dependency of the serving stack, optional here — degrades to a no-op
+    with a warning when unavailable).  Scalar report fields become
+    ``agilerl_<name>`` gauges; per-agent scalars become
+    ``agilerl_agent_<name>{agent="<idx>"}``.
+    """
+
+    def __init__(self, port: int = 9300, addr: str = "127.0.0.1", start_server: bool = True):
+        try:
+            from prometheus_client import Gauge, start_http_server
+        except ImportError:  # pragma: no cover
+            warnings.warn("prometheus_client not installed; PrometheusLogger disabled")
+            self._gauge_cls = None
+            return
+        self._gauge_cls = Gauge
+        self._gauges = {}
+        self._agent_gauges = {}
+        if start_server:
+            start_http_server(port, addr=addr)
+
+    @staticmethod
+    def _sanitize(name: str) -> str:
+        return "".join(c if c.isalnum() or c == "_" else "_" for c in name)
+
+    def log_report(self, report: Dict[str, Any]) -> None:
+        if self._gauge_cls is None:
+            return
+        for key, val in report.items():
+            if isinstance(val, bool) or not isinstance(val, (int, float)):
+                continue
+            name = "agilerl_" + self._sanitize(key)
+            if name not in self._gauges:
+                self._gauges[name] = self._gauge_cls(name, key)
+            self._gauges[name].set(float(val))
+        for snap in report.get("population", []):
+            idx = str(snap.get("index", "?"))
+            for key, val in snap.items():
+                if isinstance(val, bool) or not isinstance(val, (int, float)):
+                    continue
+                name = "agilerl_agent_" + self._sanitize(key)
+                if name not in self._agent_gauges:
+                    self._agent_gauges[name] = self._gauge_cls(name, key, ["agent"])
+                self._agent_gauges[name].labels(agent=idx).set(float(val))
 
 
 class TensorboardLogger(Logger):

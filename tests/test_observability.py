@@ -109,3 +109,21 @@ class TestGradBucketer:
         loss.backward()
         b.finalize()  # no-op, must not raise
         assert net.weight.grad is not None
+
+
+def test_prometheus_logger_gauges():
+    pytest.importorskip("prometheus_client")
+    from prometheus_client import REGISTRY
+
+    from agilerl_amd.logger import PrometheusLogger
+
+    logger = PrometheusLogger(start_server=False)
+    logger.log_report({
+        "global_step": 1000, "mean_steps_per_sec": 123.0, "best_fitness": 42.0,
+        "population": [{"index": 0, "fitness": 42.0}, {"index": 1, "fitness": 7.0}],
+    })
+    assert REGISTRY.get_sample_value("agilerl_global_step") == 1000
+    assert REGISTRY.get_sample_value("agilerl_agent_fitness", {"agent": "1"}) == 7.0
+    # repeated reports reuse gauges (no duplicate-registration error)
+    logger.log_report({"global_step": 2000, "population": []})
+    assert REGISTRY.get_sample_value("agilerl_global_step") == 2000
