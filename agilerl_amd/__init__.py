@@ -59,6 +59,8 @@ _LAZY = {
     "Mutations": "agilerl_amd.hpo.mutation",
     "MultiFrequencySelection": "agilerl_amd.hpo.multi_frequency",
     "make_vect_envs": "agilerl_amd.envs.registry",
+    "create_app": "agilerl_amd.serve",
+    "load_agent": "agilerl_amd.serve",
     "create_population": "agilerl_amd.utils.utils",
     "TrainingManifest": "agilerl_amd.models.manifest",
     "ArenaClient": "agilerl_amd.arena.client",
@@ -93,6 +95,8 @@ __all__ = [
     "Mutations",
     "MultiFrequencySelection",
     "make_vect_envs",
+    "create_app",
+    "load_agent",
     "create_population",
     "TrainingManifest",
     "ArenaClient",

@@ -1,0 +1,58 @@
+"""HTTP agent-serving tests (FastAPI TestClient, no real sockets)."""
+
+import numpy as np
+import pytest
+
+fastapi = pytest.importorskip("fastapi")
+from fastapi.testclient import TestClient
+
+from agilerl_amd.algorithms import DQN, PPO
+from agilerl_amd.serve import create_app, load_agent
+from agilerl_amd.spaces import Box, Discrete
+
+NET = {"arch": "mlp", "hidden_size": [16]}
+
+
+@pytest.fixture
+def dqn_checkpoint(tmp_path):
+    agent = DQN(Box(-1.0, 1.0, (4,)), Discrete(2), net_config=dict(NET))
+    agent.fitness.append(123.0)
+    path = str(tmp_path / "dqn.pt")
+    agent.save_checkpoint(path)
+    return path
+
+
+def test_serve_roundtrip(dqn_checkpoint):
+    agent = load_agent(dqn_checkpoint)
+    client = TestClient(create_app(agent))
+    assert client.get("/healthz").json()["status"] == "ok"
+    info = client.get("/info").json()
+    assert info["algo"] == "DQN"
+    assert info["action_space"]["n"] == 2
+    assert info["fitness_tail"][-1] == 123.0
+
+    r = client.post("/predict", json={"obs": np.random.randn(3, 4).tolist()})
+    assert r.status_code == 200
+    actions = r.json()["action"]
+    assert len(actions) == 3 and all(a in (0, 1) for a in actions)
+
+    # single-observation mode
+    r = client.post("/predict", json={"obs": [0.1, 0.2, 0.3, 0.4], "batch": False})
+    assert r.json()["action"] in (0, 1)
+
+    # bad shape -> 400, not 500
+    r = client.post("/predict", json={"obs": [[1.0, 2.0]]})
+    assert r.status_code == 400
+
+
+def test_serve_reload(dqn_checkpoint, tmp_path):
+    agent = load_agent(dqn_checkpoint)
+    client = TestClient(create_app(agent))
+    ppo = PPO(Box(-1.0, 1.0, (4,)), Discrete(2), net_config=dict(NET))
+    ppo_path = str(tmp_path / "ppo.pt")
+    ppo.save_checkpoint(ppo_path)
+    r = client.post("/reload", json={"path": ppo_path})
+    assert r.json() == {"status": "reloaded", "algo": "PPO"}
+    assert client.get("/info").json()["algo"] == "PPO"
+    r = client.post("/reload", json={"path": str(tmp_path / "missing.pt")})
+    assert r.status_code == 400
