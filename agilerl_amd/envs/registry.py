@@ -12,6 +12,7 @@ from typing import Callable, Dict, Optional, Union
 from .base import VecEnv
 from .cartpole import CartPoleVecEnv
 from .lunar_lander import LunarLanderVecEnv
+from .classic_control import AcrobotVecEnv, MountainCarContinuousVecEnv, MountainCarVecEnv
 from .pendulum import PendulumVecEnv
 from .visual import CatchPongVecEnv
 from .probe import (
@@ -32,6 +33,9 @@ ENV_REGISTRY: Dict[str, Callable[..., VecEnv]] = {
     "LunarLander-v2": LunarLanderVecEnv,
     "LunarLander-v3": LunarLanderVecEnv,
     "Pendulum-v1": PendulumVecEnv,
+    "MountainCar-v0": MountainCarVecEnv,
+    "MountainCarContinuous-v0": MountainCarContinuousVecEnv,
+    "Acrobot-v1": AcrobotVecEnv,
     "CatchPong-v0": CatchPongVecEnv,
     "PongLike-v0": CatchPongVecEnv,
     "probe/ConstantReward": ConstantRewardEnv,

@@ -80,3 +80,56 @@ def test_registry():
     assert env.num_envs == 2
     with pytest.raises(KeyError):
         make_vect_envs("NoSuchEnv-v0")
+
+
+class TestClassicControl:
+    def test_mountain_car_api_and_goal(self):
+        from agilerl_amd.envs import MountainCarVecEnv
+
+        env = MountainCarVecEnv(num_envs=8, seed=0)
+        obs, _ = env.reset()
+        assert obs.shape == (8, 2)
+        # always-push-right from the left slope cannot exceed bounds
+        for _ in range(50):
+            obs, r, term, trunc, _ = env.step(np.full(8, 2))
+            assert (r == -1.0).all()
+            assert (obs[:, 0] >= env.MIN_POS).all() and (obs[:, 0] <= env.MAX_POS).all()
+        # a car placed just below the goal with max speed terminates
+        env.pos[:] = 0.49
+        env.vel[:] = env.MAX_SPEED
+        obs, r, term, trunc, _ = env.step(np.full(8, 2))
+        assert term.all()
+
+    def test_mountain_car_continuous_reward(self):
+        from agilerl_amd.envs import MountainCarContinuousVecEnv
+
+        env = MountainCarContinuousVecEnv(num_envs=4, seed=0)
+        env.reset()
+        _, r, term, _, _ = env.step(np.full((4, 1), 1.0))
+        assert not term.any()
+        np.testing.assert_allclose(r, -0.1, atol=1e-6)  # action cost only
+        env.pos[:] = 0.449
+        env.vel[:] = env.MAX_SPEED
+        _, r, term, _, _ = env.step(np.zeros((4, 1)))
+        assert term.all() and (r >= 99.0).all()
+
+    def test_acrobot_energy_conserving_dynamics(self):
+        from agilerl_amd.envs import AcrobotVecEnv
+
+        env = AcrobotVecEnv(num_envs=8, seed=0)
+        obs, _ = env.reset()
+        assert obs.shape == (8, 6)
+        # obs invariant: first four entries are cos/sin pairs
+        for _ in range(25):
+            obs, r, term, trunc, _ = env.step(np.random.randint(0, 3, 8))
+            np.testing.assert_allclose(obs[:, 0] ** 2 + obs[:, 1] ** 2, 1.0, atol=1e-5)
+            np.testing.assert_allclose(obs[:, 2] ** 2 + obs[:, 3] ** 2, 1.0, atol=1e-5)
+            assert ((r == -1.0) | (r == 0.0)).all()
+
+    def test_registry_entries(self):
+        from agilerl_amd.envs import make_vect_envs
+
+        for env_id in ("MountainCar-v0", "MountainCarContinuous-v0", "Acrobot-v1"):
+            env = make_vect_envs(env_id, num_envs=2, seed=0)
+            obs, _ = env.reset()
+            assert obs.shape[0] == 2
