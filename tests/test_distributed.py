@@ -258,3 +258,47 @@ def test_nonfinite_loss_aborts_all_ranks():
         p.join(120)
         assert p.exitcode == 0
     assert results[0] == "raised" and results[1] == "raised"
+
+
+def _seqlen_worker(rank, world_size, port, results):
+    os.environ["RANK"] = str(rank)
+    os.environ["LOCAL_RANK"] = str(rank)
+    os.environ["WORLD_SIZE"] = str(world_size)
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+
+    from agilerl_amd.parallel import DistributedState
+
+    DistributedState.reset()
+    DistributedState.get()
+
+    class _Carrier:
+        device = "cpu"
+        from agilerl_amd.algorithms.llm.base import LLMAlgorithm
+
+        check_seq_len_agreement = LLMAlgorithm.check_seq_len_agreement
+
+    carrier = _Carrier()
+    # agreeing lengths pass on both ranks
+    carrier.check_seq_len_agreement(128)
+    # mismatched lengths raise on both ranks
+    try:
+        carrier.check_seq_len_agreement(128 + rank)
+        results[rank] = "no_raise"
+    except RuntimeError as e:
+        results[rank] = "raised" if "mismatch" in str(e) else f"wrong: {e}"
+    torch.distributed.barrier()
+    torch.distributed.destroy_process_group()
+
+
+def test_seq_len_mismatch_raises_all_ranks():
+    ctx = mp.get_context("spawn")
+    results = ctx.Manager().dict()
+    port = _find_free_port()
+    procs = [ctx.Process(target=_seqlen_worker, args=(r, 2, port, results)) for r in range(2)]
+    for p in procs:
+        p.start()
+    for p in procs:
+        p.join(120)
+        assert p.exitcode == 0
+    assert results[0] == "raised" and results[1] == "raised"
