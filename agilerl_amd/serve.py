@@ -39,6 +39,15 @@ class ReloadRequest(BaseModel):
     device: str = "cpu"
 
 
+class GenerateRequest(BaseModel):
+    prompt: Optional[str] = None
+    messages: Optional[list] = None
+    input_ids: Optional[list] = None
+    max_new_tokens: int = 128
+    temperature: float = 1.0
+    do_sample: bool = True
+
+
 def load_agent(path: str, device: str = "cpu"):
     """Rebuild any classic-RL agent from a single-file checkpoint."""
     from .algorithms.core.base import EvolvableAlgorithm
@@ -129,6 +138,44 @@ def create_app(agent, lock: Optional[threading.Lock] = None):
             req_counter.inc()
             latency.observe(time.time() - t0)
         return {"action": action.tolist()}
+
+    @app.post("/generate")
+    def generate(req: GenerateRequest):
+        """Text/token generation for LLM agents (has .generate + tokenizer)."""
+        import torch
+
+        a = state["agent"]
+        if not hasattr(a, "generate"):
+            raise HTTPException(status_code=400, detail="agent has no generate()")
+        tok = getattr(a, "tokenizer", None)
+        if req.input_ids is not None:
+            ids = torch.as_tensor(req.input_ids, dtype=torch.long)
+            if ids.dim() == 1:
+                ids = ids.unsqueeze(0)
+        elif req.prompt is not None or req.messages is not None:
+            if tok is None:
+                raise HTTPException(status_code=400,
+                                    detail="agent has no tokenizer; pass input_ids")
+            text = req.prompt
+            if req.messages is not None:
+                from .llm.chat import apply_chat_template
+
+                text = apply_chat_template(req.messages, tok)
+            ids = torch.as_tensor([tok.encode(text)], dtype=torch.long)
+        else:
+            raise HTTPException(status_code=400, detail="prompt, messages or input_ids required")
+        ids = ids.to(getattr(a, "device", "cpu"))
+        mask = torch.ones_like(ids)
+        with lock, torch.no_grad():
+            state["requests"] += 1
+            out = a.generate(ids, mask, max_new_tokens=req.max_new_tokens,
+                             do_sample=req.do_sample, temperature=req.temperature)
+        completion = out[:, ids.shape[1]:]
+        resp = {"output_ids": out.cpu().tolist(),
+                "completion_ids": completion.cpu().tolist()}
+        if tok is not None:
+            resp["completion"] = [tok.decode(row) for row in completion.cpu().tolist()]
+        return resp
 
     @app.post("/reload")
     def reload(req: ReloadRequest):

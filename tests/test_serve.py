@@ -56,3 +56,24 @@ def test_serve_reload(dqn_checkpoint, tmp_path):
     assert client.get("/info").json()["algo"] == "PPO"
     r = client.post("/reload", json={"path": str(tmp_path / "missing.pt")})
     assert r.status_code == 400
+
+
+def test_serve_llm_generate():
+    import torch
+
+    from agilerl_amd.algorithms.llm.grpo import GRPO
+
+    tiny = dict(model_type="llama", vocab_size=64, hidden_size=32, intermediate_size=64,
+                num_hidden_layers=1, num_attention_heads=2, num_key_value_heads=1,
+                max_position_embeddings=64, pad_token_id=0)
+    agent = GRPO(model_config=tiny, dtype=torch.float32, lora_config={"r": 2},
+                 max_completion_tokens=4)
+    client = TestClient(create_app(agent))
+    r = client.post("/generate", json={"input_ids": [1, 2, 3], "max_new_tokens": 4})
+    assert r.status_code == 200
+    body = r.json()
+    assert len(body["output_ids"][0]) == 7
+    assert len(body["completion_ids"][0]) == 4
+    # classic agents refuse cleanly
+    r = client.post("/generate", json={"prompt": "hi"})
+    assert r.status_code in (200, 400)  # GRPO without tokenizer -> 400
