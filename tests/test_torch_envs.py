@@ -181,3 +181,54 @@ class TestGraphCollector:
             verbose=False)
         assert len(hist) >= 2  # survived at least one evolution + recapture
         torch.cuda.synchronize()
+
+
+class TestTorchMPEOnCPU:
+    """Torch MPE envs run on any device; CPU here, GPU in bench."""
+
+    def test_speaker_listener_contract(self):
+        import torch
+
+        from agilerl_amd.envs.torch_mpe import SpeakerListenerTorchVecEnv
+
+        env = SpeakerListenerTorchVecEnv(num_envs=4, device="cpu", seed=0)
+        obs, _ = env.reset()
+        assert obs["speaker_0"].shape == (4, 3)
+        assert obs["listener_0"].shape == (4, 11)
+        actions = {"speaker_0": torch.randint(0, 3, (4,)),
+                   "listener_0": torch.randint(0, 5, (4,))}
+        obs, rewards, term, trunc, _ = env.step(actions)
+        assert rewards["speaker_0"].shape == (4,)
+        torch.testing.assert_close(rewards["speaker_0"], rewards["listener_0"])
+
+    def test_spread_contract(self):
+        import torch
+
+        from agilerl_amd.envs.torch_mpe import SimpleSpreadTorchVecEnv
+
+        env = SimpleSpreadTorchVecEnv(num_envs=4, device="cpu", seed=0)
+        obs, _ = env.reset()
+        truncated_seen = False
+        for _ in range(26):
+            actions = {a: torch.randint(0, 5, (4,)) for a in env.agents}
+            obs, rewards, term, trunc, _ = env.step(actions)
+            truncated_seen = truncated_seen or bool(trunc["agent_0"].any())
+        assert truncated_seen  # auto-reset fired within max_episode_steps+1
+
+
+def test_activation_offload_noop_paths():
+    import torch
+
+    from agilerl_amd.llm.offload import activation_offload
+
+    x = torch.randn(4, 4, requires_grad=True)
+    with activation_offload(enabled=False):
+        y = (x * 2).sum()
+    y.backward()
+    assert torch.allclose(x.grad, torch.full_like(x, 2.0))
+    x.grad = None
+    # enabled without cuda degrades to a no-op instead of failing
+    with activation_offload(enabled=True):
+        y = (x * 3).sum()
+    y.backward()
+    assert torch.allclose(x.grad, torch.full_like(x, 3.0))
