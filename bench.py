@@ -235,7 +235,7 @@ class GrpoBenchRunner:
             model_config=dict(cfg),
             dtype=dtype,
             lora_config={"r": 16, "lora_alpha": 32},
-            group_size=8,
+            group_size=min(8, max(1, args.grpo_batch)),
             micro_batch_size=4 if args.model_size == "8b" else 8,
             update_epochs=1,
             beta=0.04,

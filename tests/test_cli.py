@@ -74,3 +74,33 @@ def test_bench_contract_cpu():
                 "ms_per_step", "higher_is_better", "scaling", "vs_baseline",
                 "dtype", "data", "config"):
         assert key in payload, key
+
+
+@pytest.mark.slow
+@pytest.mark.parametrize("workload", ["dqn", "rainbow", "maddpg"])
+def test_bench_other_workloads_cpu(workload):
+    """Every BASELINE workload's bench path emits a valid contract line on CPU."""
+    out = subprocess.run(
+        [sys.executable, "bench.py", "--workload", workload, "--steps", "2",
+         "--warmup", "1", "--num-envs", "8", "--pop-size", "2"],
+        cwd=ROOT, capture_output=True, text=True, timeout=600,
+    )
+    assert out.returncode == 0, out.stderr[-2000:]
+    lines = [l for l in out.stdout.strip().splitlines() if l.startswith("{")]
+    assert len(lines) == 1
+    payload = json.loads(lines[0])
+    assert payload["value"] > 0
+    assert payload["config"]
+
+
+@pytest.mark.slow
+def test_bench_grpo_tiny_cpu():
+    out = subprocess.run(
+        [sys.executable, "bench.py", "--workload", "grpo", "--model-size", "tiny",
+         "--steps", "2", "--warmup", "1", "--seq-len", "64", "--grpo-batch", "4"],
+        cwd=ROOT, capture_output=True, text=True, timeout=900,
+    )
+    assert out.returncode == 0, out.stderr[-2000:]
+    lines = [l for l in out.stdout.strip().splitlines() if l.startswith("{")]
+    payload = json.loads(lines[0])
+    assert payload["unit"].startswith("tokens")
