@@ -186,6 +186,7 @@ def make_loggers(
     csv_path: Optional[str] = None,
     tensorboard_dir: Optional[str] = None,
     wandb_project: Optional[str] = None,
+    prometheus_port: Optional[int] = None,
 ) -> List[Logger]:
     loggers: List[Logger] = []
     if stdout:
@@ -196,4 +197,6 @@ def make_loggers(
         loggers.append(TensorboardLogger(tensorboard_dir))
     if wandb_project:
         loggers.append(WandbLogger(wandb_project))
+    if prometheus_port:
+        loggers.append(PrometheusLogger(port=prometheus_port))
     return loggers

@@ -20,6 +20,8 @@ def parse_args(argv=None):
     p.add_argument("--wb", action="store_true", help="log to wandb if available")
     p.add_argument("--csv", default=None, help="CSV log path")
     p.add_argument("--tensorboard", default=None, help="TensorBoard log dir")
+    p.add_argument("--prometheus-port", type=int, default=None,
+                   help="export population metrics as Prometheus gauges")
     return p.parse_args(argv)
 
 
@@ -35,6 +37,7 @@ def main(argv=None):
         csv_path=args.csv,
         tensorboard_dir=args.tensorboard,
         wandb_project="agilerl-amd" if args.wb else None,
+        prometheus_port=args.prometheus_port,
     )
     trainer = LocalTrainer.from_manifest(args.manifest, device=device, loggers=loggers)
     if args.checkpoint_steps is not None:
