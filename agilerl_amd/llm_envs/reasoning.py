@@ -88,9 +88,23 @@ class ReasoningGym(LLMEnvBase):
         apply_chat_template: bool = True,
         system_prompt: Optional[str] = None,
         seed: Optional[int] = None,
+        shard_across_ranks: bool = True,
     ):
-        self.prompts = list(prompts)
-        self.answers = list(answers) if answers is not None else [None] * len(prompts)
+        prompts = list(prompts)
+        answers = list(answers) if answers is not None else [None] * len(prompts)
+        if shard_across_ranks:
+            # per-rank data sharding (reference: accelerate-prepared
+            # DataLoader split, llm_envs/base.py:156) — each DP rank sees a
+            # disjoint prompt slice so the population never re-scores the
+            # same prompt twice in one pass
+            from ..parallel import DistributedState
+
+            state = DistributedState.get()
+            if state.world_size > 1 and len(prompts) >= state.world_size:
+                prompts = prompts[state.rank :: state.world_size]
+                answers = answers[state.rank :: state.world_size]
+        self.prompts = prompts
+        self.answers = answers
         self.reward_fn = reward_fn
         self.tokenizer = tokenizer
         self.data_batch_size = int(data_batch_size)

@@ -83,3 +83,23 @@ def barrier() -> None:
             dist.barrier(device_ids=[torch.cuda.current_device()])
         else:
             dist.barrier()
+
+
+def aggregate_metrics_across_ranks(stats: dict) -> dict:
+    """Mean of scalar metrics over all DP ranks (reference
+    aggregate_metrics_across_gpus, llm_utils.py:569).  No-op when not
+    distributed; non-scalar values pass through from the local rank."""
+    state = DistributedState.get()
+    if not state.is_distributed:
+        return stats
+    keys = sorted(k for k, v in stats.items() if isinstance(v, (int, float)) and not isinstance(v, bool))
+    if not keys:
+        return stats
+    t = torch.tensor([float(stats[k]) for k in keys], dtype=torch.float64)
+    if state.backend == "nccl":
+        t = t.to(state.device)
+    dist.all_reduce(t, op=dist.ReduceOp.SUM)
+    t = (t / state.world_size).cpu()
+    out = dict(stats)
+    out.update({k: float(t[i]) for i, k in enumerate(keys)})
+    return out

@@ -60,8 +60,11 @@ def finetune_llm_reasoning(
                 stats_acc.append(stats)
                 agent.steps[-1] += 1
             metrics.finalize_training_step(evo_steps)
-            for key in stats_acc[0]:
-                metrics.log(key, float(np.mean([s[key] for s in stats_acc])))
+            mean_stats = {k: float(np.mean([s[k] for s in stats_acc])) for k in stats_acc[0]}
+            from ...parallel import aggregate_metrics_across_ranks
+
+            for key, val in aggregate_metrics_across_ranks(mean_stats).items():
+                metrics.log(key, val)
             fitness = agent.test(env, loop=eval_loop)
             metrics.log_fitness(fitness)
 

@@ -302,3 +302,58 @@ def test_seq_len_mismatch_raises_all_ranks():
         p.join(120)
         assert p.exitcode == 0
     assert results[0] == "raised" and results[1] == "raised"
+
+
+def _metrics_shard_worker(rank, world_size, port, results):
+    os.environ["RANK"] = str(rank)
+    os.environ["LOCAL_RANK"] = str(rank)
+    os.environ["WORLD_SIZE"] = str(world_size)
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+
+    from agilerl_amd.parallel import DistributedState, aggregate_metrics_across_ranks
+
+    DistributedState.reset()
+    DistributedState.get()
+
+    # cross-rank metric mean
+    stats = aggregate_metrics_across_ranks({"loss": float(rank), "note": "x"})
+    results[f"loss_{rank}"] = stats["loss"]
+    results[f"note_{rank}"] = stats["note"]
+
+    # per-rank prompt sharding: disjoint slices
+    class _Tok:
+        pad_token_id = 0
+        eos_token = "<eos>"
+        pad_token = "<pad>"
+        chat_template = None
+
+        def __call__(self, texts, **kw):
+            import torch as _t
+
+            ids = _t.ones(len(texts), 4, dtype=_t.long)
+            return {"input_ids": ids, "attention_mask": _t.ones_like(ids)}
+
+    from agilerl_amd.llm_envs import ReasoningGym
+
+    gym = ReasoningGym([f"p{i}" for i in range(10)], list(range(10)),
+                       lambda c, a: 0.0, _Tok(), data_batch_size=2)
+    results[f"prompts_{rank}"] = tuple(gym.prompts)
+    torch.distributed.barrier()
+    torch.distributed.destroy_process_group()
+
+
+def test_metric_aggregation_and_data_sharding():
+    ctx = mp.get_context("spawn")
+    results = ctx.Manager().dict()
+    port = _find_free_port()
+    procs = [ctx.Process(target=_metrics_shard_worker, args=(r, 2, port, results)) for r in range(2)]
+    for p in procs:
+        p.start()
+    for p in procs:
+        p.join(120)
+        assert p.exitcode == 0
+    assert results["loss_0"] == results["loss_1"] == 0.5  # mean of 0 and 1
+    assert results["note_0"] == "x"
+    assert set(results["prompts_0"]).isdisjoint(results["prompts_1"])
+    assert len(results["prompts_0"]) + len(results["prompts_1"]) == 10
