@@ -14,7 +14,7 @@ import torch.nn as nn
 
 from ..ops.rmsnorm import HipRMSNorm
 
-__all__ = ["apply_hip_kernels_to_llama"]
+__all__ = ["apply_hip_kernels_to_llama", "patch_llama_swiglu"]
 
 
 def apply_hip_kernels_to_llama(model: nn.Module) -> int:
@@ -32,4 +32,26 @@ def apply_hip_kernels_to_llama(model: nn.Module) -> int:
                 new.weight.requires_grad = child.weight.requires_grad
                 setattr(parent, child_name, new)
                 patched += 1
+    return patched
+
+
+def patch_llama_swiglu(model: nn.Module) -> int:
+    """Swap each Llama MLP's ``act_fn(gate) * up`` for the fused SwiGLU
+    kernel (ops/csrc/act_ops.hip).  Opt-in: call after model creation.
+    Returns the number of MLP modules patched."""
+    from ..ops.swiglu import swiglu
+
+    patched = 0
+    for module in model.modules():
+        if (
+            type(module).__name__.endswith("MLP")
+            and hasattr(module, "gate_proj")
+            and hasattr(module, "up_proj")
+            and hasattr(module, "down_proj")
+        ):
+            def fused_forward(x, _m=module):
+                return _m.down_proj(swiglu(_m.gate_proj(x), _m.up_proj(x)))
+
+            module.forward = fused_forward
+            patched += 1
     return patched

@@ -292,3 +292,5 @@ def masked_mean(x: torch.Tensor, mask: torch.Tensor, dim: Optional[int] = None, 
 def masked_sum(x: torch.Tensor, mask: torch.Tensor, dim: Optional[int] = None):
     mask = mask.to(x.dtype)
     return (x * mask).sum() if dim is None else (x * mask).sum(dim=dim)
+
+from .swiglu import swiglu  # noqa: F401
