@@ -140,10 +140,15 @@ class LLMAlgorithm(EvolvableAlgorithm):
         self.model = build_causal_lm(model, model_config, model_name_or_path, dtype, device)
         self.model.requires_grad_(False)
         if str(device).startswith("cuda") and dtype == torch.bfloat16:
-            # fused CDNA4 RMSNorm (analog of the reference's Liger patches)
-            from ...architectures.llama_patches import apply_hip_kernels_to_llama
+            # fused CDNA4 RMSNorm + SwiGLU (analog of the reference's Liger
+            # patches); GEMMs stay on hipBLASLt
+            from ...architectures.llama_patches import (
+                apply_hip_kernels_to_llama,
+                patch_llama_swiglu,
+            )
 
             apply_hip_kernels_to_llama(self.model)
+            patch_llama_swiglu(self.model)
         if gradient_checkpointing and hasattr(self.model, "gradient_checkpointing_enable"):
             self.model.gradient_checkpointing_enable()
 
