@@ -11,6 +11,8 @@ from .neural_ucb import NeuralUCB
 from .neural_ts import NeuralTS
 
 __all__ = [
+    "ILQL",
+    "BC_LM",
     "DQN",
     "RainbowDQN",
     "CQN",
@@ -23,3 +25,6 @@ __all__ = [
     "NeuralUCB",
     "NeuralTS",
 ]
+
+from .ilql import ILQL  # noqa: E402,F401
+from .bc_lm import BC_LM  # noqa: E402,F401

@@ -294,3 +294,4 @@ def masked_sum(x: torch.Tensor, mask: torch.Tensor, dim: Optional[int] = None):
     return (x * mask).sum() if dim is None else (x * mask).sum(dim=dim)
 
 from .swiglu import swiglu  # noqa: F401
+from .rmsnorm import rms_norm, HipRMSNorm  # noqa: F401

@@ -12,11 +12,14 @@ from typing import Any, Dict
 from .reasoning import finetune_llm_reasoning
 from .sft import finetune_llm_sft
 from .preference import finetune_llm_preference
+from .multiturn import finetune_llm_multiturn, rollout_multiturn
 
 __all__ = [
     "finetune_llm_reasoning",
     "finetune_llm_sft",
     "finetune_llm_preference",
+    "finetune_llm_multiturn",
+    "rollout_multiturn",
     "run_llm_workload",
 ]
 
