@@ -8,7 +8,7 @@ to GPU boxes with the repo snapshot).
 
 import os
 
-from setuptools import setup
+from setuptools import find_packages, setup
 
 os.environ.setdefault("PYTORCH_ROCM_ARCH", "gfx950")
 
@@ -33,7 +33,7 @@ ext = CUDAExtension(
 setup(
     name="agilerl-amd",
     version="0.1.0",
-    packages=["agilerl_amd"],
+    packages=find_packages(include=["agilerl_amd", "agilerl_amd.*"]),
     ext_modules=[ext],
     cmdclass={"build_ext": BuildExtension.with_options(no_python_abi_suffix=False)},
 )
