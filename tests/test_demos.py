@@ -14,6 +14,7 @@ DEMOS = [
     ("demos/multi_agent/demo_multi_agent.py", ["--max-steps", "2000"]),
     ("demos/bandits/demo_bandit.py", ["--max-steps", "400"]),
     ("demos/llm/demo_llm_finetuning.py", ["--iterations", "2"]),
+    ("demos/llm/demo_multiturn.py", ["--iterations", "2"]),
 ]
 
 
