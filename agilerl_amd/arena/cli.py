@@ -29,6 +29,15 @@ def main(argv=None):
     st = sub.add_parser("status")
     st.add_argument("experiment_id")
     sub.add_parser("list")
+    du = sub.add_parser("dataset-upload")
+    du.add_argument("path")
+    du.add_argument("--name", default=None)
+    sub.add_parser("dataset-list")
+    dp = sub.add_parser("deploy")
+    dp.add_argument("experiment_id")
+    dp.add_argument("--checkpoint", default=None)
+    dp.add_argument("--port", type=int, default=8000)
+    dp.add_argument("--host", default="127.0.0.1")
     args = p.parse_args(argv)
 
     client = ArenaClient(workspace=args.workspace)
@@ -56,6 +65,18 @@ def main(argv=None):
         print(json.dumps(client.experiment_status(args.experiment_id)))
     elif args.cmd == "list":
         print("\n".join(client.list_experiments()))
+    elif args.cmd == "dataset-upload":
+        print(client.upload_dataset(args.path, name=args.name))
+    elif args.cmd == "dataset-list":
+        print("\n".join(client.list_datasets()))
+    elif args.cmd == "deploy":
+        dep = client.deploy(args.experiment_id, checkpoint=args.checkpoint)
+        print(json.dumps(dep.info()))
+        import uvicorn
+
+        from ..serve import create_app
+
+        uvicorn.run(create_app(dep.agent), host=args.host, port=args.port)
 
 
 if __name__ == "__main__":

@@ -40,6 +40,29 @@ class ExperimentHandle:
     workspace: Optional[str] = None
 
 
+@dataclass
+class Deployment:
+    """In-process deployment handle (the local analog of an Arena
+    inference endpoint)."""
+
+    experiment_id: str
+    checkpoint: str
+    agent: Any
+
+    def predict(self, obs):
+        import numpy as np
+
+        action = self.agent.get_action(np.asarray(obs, dtype=np.float32), training=False)
+        return action[0] if isinstance(action, tuple) else action
+
+    def info(self) -> Dict[str, Any]:
+        return {
+            "experiment_id": self.experiment_id,
+            "checkpoint": self.checkpoint,
+            "algo": type(self.agent).__name__,
+        }
+
+
 class ArenaClient:
     def __init__(
         self,
@@ -155,6 +178,66 @@ class ArenaClient:
         return sorted(
             f for f in os.listdir(exp_dir) if f.endswith(".pt") or f.startswith("ckpt")
         )
+
+    # ------------------------------------------------------------------
+    # Datasets (reference arena dataset upload/list; local workspace copy)
+    # ------------------------------------------------------------------
+    def upload_dataset(self, path: str, name: Optional[str] = None) -> str:
+        """Register a local dataset file (npz/json/arrow dir) in the
+        workspace; returns the dataset id."""
+        self._require_auth()
+        if not os.path.exists(path):
+            raise ArenaError(f"dataset path not found: {path}")
+        name = name or os.path.basename(path).split(".")[0]
+        ds_id = f"ds-{name}"
+        ds_dir = os.path.join(self.workspace, "datasets")
+        os.makedirs(ds_dir, exist_ok=True)
+        import shutil
+
+        dest = os.path.join(ds_dir, ds_id + os.path.splitext(path)[1])
+        if os.path.isdir(path):
+            dest = os.path.join(ds_dir, ds_id)
+            if not os.path.exists(dest):
+                shutil.copytree(path, dest)
+        else:
+            shutil.copy(path, dest)
+        self._write(os.path.join("datasets", ds_id + ".json"),
+                    {"id": ds_id, "source": os.path.abspath(path),
+                     "stored": dest, "uploaded_at": time.time()})
+        return ds_id
+
+    def list_datasets(self) -> List[str]:
+        ds_dir = os.path.join(self.workspace, "datasets")
+        if not os.path.isdir(ds_dir):
+            return []
+        return sorted(f[:-5] for f in os.listdir(ds_dir) if f.endswith(".json"))
+
+    def dataset_path(self, ds_id: str) -> str:
+        meta = os.path.join(self.workspace, "datasets", ds_id + ".json")
+        if not os.path.exists(meta):
+            raise ArenaError(f"unknown dataset {ds_id}")
+        with open(meta) as f:
+            return json.load(f)["stored"]
+
+    # ------------------------------------------------------------------
+    # Deploy / infer (reference arena deploy+inference endpoints; local
+    # backend wraps the in-process serving app — see agilerl_amd/serve.py)
+    # ------------------------------------------------------------------
+    def deploy(self, experiment_id: str, checkpoint: Optional[str] = None):
+        """Load a checkpoint from the experiment dir and return a
+        Deployment with .predict()/.info(); host it over HTTP with
+        `python -m agilerl_amd.serve <ckpt>`."""
+        self._require_auth()
+        exp_dir = os.path.join(self.workspace, experiment_id)
+        ckpts = self.list_checkpoints(experiment_id)
+        if checkpoint is None:
+            if not ckpts:
+                raise ArenaError(f"no checkpoints in {experiment_id}")
+            checkpoint = ckpts[-1]
+        from ..serve import load_agent
+
+        agent = load_agent(os.path.join(exp_dir, checkpoint))
+        return Deployment(experiment_id, checkpoint, agent)
 
     # ------------------------------------------------------------------
     def _write(self, rel: str, payload: Dict[str, Any]) -> None:

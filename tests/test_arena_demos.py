@@ -46,6 +46,43 @@ class TestArenaClient:
         assert resumed.status == "completed"
 
 
+class TestArenaDatasetsAndDeploy:
+    def test_dataset_roundtrip(self, tmp_path):
+        import numpy as np
+
+        client = ArenaClient(workspace=str(tmp_path / "ws"))
+        client.login()
+        ds_file = tmp_path / "transitions.npz"
+        np.savez(ds_file, observations=np.zeros((4, 2), dtype=np.float32))
+        ds_id = client.upload_dataset(str(ds_file))
+        assert ds_id in client.list_datasets()
+        stored = client.dataset_path(ds_id)
+        with np.load(stored) as z:
+            assert z["observations"].shape == (4, 2)
+        with pytest.raises(ArenaError):
+            client.dataset_path("ds-missing")
+
+    def test_deploy_and_predict(self, tmp_path):
+        import numpy as np
+
+        from agilerl_amd.algorithms import DQN
+        from agilerl_amd.spaces import Box, Discrete
+
+        client = ArenaClient(workspace=str(tmp_path))
+        client.login()
+        exp_dir = tmp_path / "exp-abc"
+        exp_dir.mkdir()
+        agent = DQN(Box(-1.0, 1.0, (4,)), Discrete(2),
+                    net_config={"arch": "mlp", "hidden_size": [16]})
+        agent.save_checkpoint(str(exp_dir / "ckpt_final.pt"))
+        dep = client.deploy("exp-abc")
+        assert dep.info()["algo"] == "DQN"
+        actions = dep.predict(np.zeros((3, 4), dtype=np.float32))
+        assert len(actions) == 3
+        with pytest.raises(ArenaError):
+            client.deploy("exp-nope")
+
+
 class TestMAProbes:
     def test_joint_action_env(self):
         from agilerl_amd.envs.probe_ma import JointActionMAEnv
