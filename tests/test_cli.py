@@ -104,3 +104,14 @@ def test_bench_grpo_tiny_cpu():
     lines = [l for l in out.stdout.strip().splitlines() if l.startswith("{")]
     payload = json.loads(lines[0])
     assert payload["unit"].startswith("tokens")
+
+
+def test_kernel_isa_report_tool():
+    """The committed ISA table can be regenerated from the shipped .so."""
+    out = subprocess.run(
+        [sys.executable, os.path.join(ROOT, "tools", "kernel_isa_report.py")],
+        capture_output=True, text=True, timeout=120,
+    )
+    assert out.returncode == 0, out.stderr[-1000:]
+    assert "rmsnorm_fwd_kernel" in out.stdout
+    assert "swiglu_fwd_kernel" in out.stdout
