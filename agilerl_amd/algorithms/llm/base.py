@@ -243,6 +243,61 @@ class LLMAlgorithm(EvolvableAlgorithm):
             self._activate("self")
         return lp
 
+    def compute_logprobs_packed(
+        self,
+        input_ids: torch.Tensor,
+        attention_mask: torch.Tensor,
+        adapter: Optional[str] = "self",
+        with_grad: bool = False,
+        chunk_rows: Optional[int] = None,
+    ) -> torch.Tensor:
+        """Padding-free variant of compute_logprobs (SURVEY §2.8 long-context
+        strategy; reference llm_packing.py): real tokens are concatenated
+        into one row with per-sequence position ids and a block-diagonal
+        causal mask, so the decoder and the chunked lm_head GEMMs process
+        only real tokens.  Returns (B, T-1) with zeros at pad positions —
+        numerically equivalent to compute_logprobs on the real positions
+        (RoPE attention depends only on position differences, so the
+        per-sequence position restart is offset-invariant)."""
+        from ...llm.packing import pack_padded_batch
+
+        pack = pack_padded_batch(input_ids, attention_mask)
+        packed, pos, cu = pack["packed_ids"], pack["position_ids"], pack["cu_seqlens"]
+        N = packed.shape[1]
+        lengths = cu[1:] - cu[:-1]
+        seq_id = torch.repeat_interleave(
+            torch.arange(lengths.numel(), device=packed.device), lengths
+        )
+        same = seq_id.unsqueeze(0) == seq_id.unsqueeze(1)
+        causal = torch.arange(N, device=packed.device).unsqueeze(0) <= torch.arange(
+            N, device=packed.device
+        ).unsqueeze(1)
+        dtype = next(self.model.parameters()).dtype
+        neg = torch.finfo(dtype).min
+        mask4d = torch.where(same & causal, 0.0, neg).to(dtype)[None, None]
+
+        self._activate(adapter)
+        ctx = torch.enable_grad() if with_grad else torch.no_grad()
+        with ctx:
+            hidden = self._decoder()(
+                input_ids=packed, position_ids=pos, attention_mask=mask4d
+            ).last_hidden_state[0]  # (N, H)
+            # positions whose next token is real and in the same sequence
+            k = torch.arange(N - 1, device=packed.device)
+            valid = seq_id[k] == seq_id[k + 1]
+            vk = k[valid]
+            lp_valid = fused_linear_logprobs(
+                hidden[vk], self._lm_head_weight(), packed[0, vk + 1],
+                temperature=self.temperature, chunk_rows=chunk_rows,
+            )
+            B, T = input_ids.shape
+            out = lp_valid.new_zeros(B * T)
+            out[pack["indices"][vk]] = lp_valid
+            out = out.reshape(B, T)[:, :-1]
+        if adapter != "self":
+            self._activate("self")
+        return out
+
     @torch.no_grad()
     def generate(
         self,

@@ -56,6 +56,7 @@ class GRPO(LLMAlgorithm):
         max_completion_tokens: int = 256,
         dtype: torch.dtype = torch.bfloat16,
         gradient_checkpointing: bool = False,
+        use_packing: bool = False,
         device: str = "cpu",
     ):
         super().__init__(
@@ -68,6 +69,9 @@ class GRPO(LLMAlgorithm):
         )
         self.group_size = int(group_size)
         self.update_epochs = int(update_epochs)
+        # padding-free grad/old-policy passes (compute_logprobs_packed);
+        # opt-in until validated at 8-GPU scale
+        self.use_packing = bool(use_packing)
         self.clip_coef = float(clip_coef)
         self.clip_coef_lower = float(clip_coef_lower) if clip_coef_lower is not None else None
         self.beta = float(beta)  # k3 KL coefficient
@@ -107,15 +111,16 @@ class GRPO(LLMAlgorithm):
 
         B = ids.shape[0]
         mb = max(self.micro_batch_size, 1)
+        logprob_fn = self.compute_logprobs_packed if self.use_packing else self.compute_logprobs
 
         # old-policy + reference logprobs (no grad, micro-batched)
         old_logp = torch.empty(action_mask.shape, device=self.device)
         ref_logp = torch.empty_like(old_logp) if self.beta > 0 else None
         for s in range(0, B, mb):
             e = min(s + mb, B)
-            old_logp[s:e] = self.compute_logprobs(ids[s:e], attention_mask[s:e], adapter="self")
+            old_logp[s:e] = logprob_fn(ids[s:e], attention_mask[s:e], adapter="self")
             if ref_logp is not None:
-                ref_logp[s:e] = self.compute_logprobs(ids[s:e], attention_mask[s:e], adapter=None)
+                ref_logp[s:e] = logprob_fn(ids[s:e], attention_mask[s:e], adapter=None)
 
         clip_hi = 1.0 + self.clip_coef
         clip_lo = 1.0 - (self.clip_coef_lower if self.clip_coef_lower is not None else self.clip_coef)
@@ -126,7 +131,7 @@ class GRPO(LLMAlgorithm):
             perm = torch.randperm(B, device=self.device)
             for s in range(0, B, mb):
                 sel = perm[s : s + mb]
-                logp = self.compute_logprobs(ids[sel], attention_mask[sel], with_grad=True)
+                logp = logprob_fn(ids[sel], attention_mask[sel], with_grad=True)
                 loss = self._policy_loss(
                     logp,
                     old_logp[sel],

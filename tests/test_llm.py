@@ -494,3 +494,40 @@ class TestAdapterCleanupOnEvolve:
         dp.evolve(TournamentSelection(3, True, rng=__import__("numpy").random.default_rng(0)))
         n_adapters = len(next(iter_lora_modules(model))[1].lora_A)
         assert n_adapters == 3  # exactly pop_size slots remain (no leak)
+
+
+class TestPacking:
+    def test_packed_logprobs_match_padded(self):
+        torch.manual_seed(0)
+        from agilerl_amd.algorithms.llm.grpo import GRPO
+
+        agent = tiny_agent(GRPO)
+        B, T = 3, 12
+        ids = torch.randint(1, 128, (B, T))
+        mask = torch.ones(B, T, dtype=torch.long)
+        mask[0, :5] = 0
+        ids[0, :5] = 0
+        mask[2, :2] = 0
+        ids[2, :2] = 0
+        lp = agent.compute_logprobs(ids, mask)
+        lpp = agent.compute_logprobs_packed(ids, mask)
+        real = (mask[:, 1:] * mask[:, :-1]).bool()
+        torch.testing.assert_close(lp[real], lpp[real], atol=1e-4, rtol=1e-4)
+        # pad targets are zeroed in the packed output
+        assert (lpp[~real] == 0).all()
+
+    def test_grpo_learn_with_packing(self):
+        torch.manual_seed(0)
+        from agilerl_amd.algorithms.llm.grpo import GRPO
+        from agilerl_amd.llm_envs import TokenReasoningGym, make_grpo_experiences
+
+        agent = tiny_agent(GRPO, group_size=4, lr=1e-3, beta=0.04,
+                           max_completion_tokens=6, use_packing=True)
+        env = TokenReasoningGym(vocab_size=128, prompt_len=6, data_batch_size=2,
+                                group_size=4, seed=0)
+        prompts = env.reset()
+        seqs = agent.get_action(prompts)
+        exp = make_grpo_experiences(env, seqs, env.score(seqs))
+        stats = agent.learn(exp)
+        assert np.isfinite(stats["loss"])
+        assert stats["kl"] >= -1e-5
