@@ -4,7 +4,7 @@ from .train_on_policy import train_on_policy
 from .train_multi_agent_off_policy import train_multi_agent_off_policy
 from .train_multi_agent_on_policy import train_multi_agent_on_policy
 from .train_bandits import train_bandits
-from .train_offline import train_offline, load_transitions_into_buffer, save_transitions, load_transitions
+from .train_offline import train_offline, load_transitions_into_buffer, save_transitions, load_transitions, collect_transitions
 from .train_distributed import train_on_policy_distributed
 
 __all__ = [

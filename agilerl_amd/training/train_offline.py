@@ -109,3 +109,38 @@ def save_transitions(dataset: Dict[str, np.ndarray], path: str) -> None:
 def load_transitions(path: str) -> Dict[str, np.ndarray]:
     with np.load(path) as z:
         return {k: z[k] for k in z.files}
+
+
+def collect_transitions(env, agent=None, steps: int = 1000,
+                        epsilon: float = 1.0) -> Dict[str, np.ndarray]:
+    """Roll a (possibly random) policy to build an offline dataset.
+
+    agent=None or epsilon=1.0 gives uniform-random behavior; otherwise
+    epsilon-greedy around ``agent.get_action``.  Returns minari-style
+    arrays consumable by load_transitions_into_buffer / save_transitions.
+    """
+    obs_l, act_l, rew_l, next_l, term_l = [], [], [], [], []
+    obs, _ = env.reset()
+    for _ in range(steps):
+        if agent is None or np.random.rand() < epsilon:
+            action = np.array(
+                [env.action_space.sample() for _ in range(env.num_envs)]
+            ) if hasattr(env.action_space, "sample") else np.random.randint(
+                0, env.action_space.n, env.num_envs
+            )
+        else:
+            action = agent.get_action(obs, training=False)
+        next_obs, reward, term, trunc, _ = env.step(action)
+        obs_l.append(np.asarray(obs))
+        act_l.append(np.asarray(action))
+        rew_l.append(np.asarray(reward, dtype=np.float32))
+        next_l.append(np.asarray(next_obs))
+        term_l.append(np.asarray(term, dtype=np.float32))
+        obs = next_obs
+    return {
+        "observations": np.concatenate(obs_l),
+        "actions": np.concatenate(act_l),
+        "rewards": np.concatenate(rew_l),
+        "next_observations": np.concatenate(next_l),
+        "terminals": np.concatenate(term_l),
+    }

@@ -26,3 +26,20 @@ def test_capability_flags():
     assert pkg.HAS_DEEPSPEED is False
     assert pkg.HAS_LIGER_KERNEL is False
     assert isinstance(pkg.HAS_LLM_DEPENDENCIES, bool)
+
+
+def test_collect_transitions_random_policy():
+    import numpy as np
+
+    from agilerl_amd.envs import CartPoleVecEnv
+    from agilerl_amd.training import collect_transitions, load_transitions_into_buffer
+    from agilerl_amd.components import ReplayBuffer
+
+    env = CartPoleVecEnv(num_envs=4, seed=0)
+    ds = collect_transitions(env, steps=25)
+    assert ds["observations"].shape == (100, 4)
+    assert set(ds) == {"observations", "actions", "rewards",
+                       "next_observations", "terminals"}
+    buf = ReplayBuffer(200)
+    load_transitions_into_buffer(ds, buf)
+    assert len(buf) == 100
