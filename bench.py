@@ -65,6 +65,7 @@ def parse_args():
     p.add_argument("--grpo-batch", type=int, default=32, help="grpo: sequences per step per rank")
     p.add_argument("--model-size", choices=["8b", "tiny"], default="8b")
     p.add_argument("--no-graph", action="store_true", help="disable hipGraph collector")
+    p.add_argument("--packing", action="store_true", help="grpo: padding-free packed logprob passes")
     return p.parse_args()
 
 
@@ -241,6 +242,7 @@ class GrpoBenchRunner:
             beta=0.04,
             lr=5e-6,
             gradient_checkpointing=args.model_size == "8b",
+            use_packing=args.packing,
             device=self.device,
         )
         self.vocab = cfg["vocab_size"]
