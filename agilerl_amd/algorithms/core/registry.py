@@ -76,7 +76,7 @@ class RLParameter:
         self.categorical = categorical
 
     def mutate(self, value, rng: Optional[np.random.Generator] = None):
-        rng = rng or np.random.default_rng()
+        rng = rng or np.random.default_rng(np.random.randint(0, 2**31 - 1))
         if self.categorical is not None:
             return self.categorical[int(rng.integers(len(self.categorical)))]
         factor = self.grow_factor if rng.random() < 0.5 else self.shrink_factor
@@ -101,7 +101,7 @@ class HyperparameterConfig:
     def sample(self, rng: Optional[np.random.Generator] = None) -> Optional[str]:
         if not self.config:
             return None
-        rng = rng or np.random.default_rng()
+        rng = rng or np.random.default_rng(np.random.randint(0, 2**31 - 1))
         return self.names()[int(rng.integers(len(self.config)))]
 
     def __getitem__(self, name: str) -> RLParameter:

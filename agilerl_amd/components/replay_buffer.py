@@ -47,6 +47,7 @@ class ReplayBuffer:
         device: str = "cpu",
         storage_device: Optional[str] = None,
         pin_memory: bool = True,
+        seed: Optional[int] = None,
     ):
         self.max_size = int(max_size)
         self.device = device
@@ -56,7 +57,11 @@ class ReplayBuffer:
         self._ptr = 0
         self._size = 0
         self.num_envs = 1
-        self._rng = np.random.default_rng()
+        # unseeded buffers derive their stream from the global numpy RNG so
+        # np.random.seed(...) upstream makes the whole run reproducible
+        if seed is None:
+            seed = int(np.random.randint(0, 2**31 - 1))
+        self._rng = np.random.default_rng(seed)
 
     def __len__(self) -> int:
         return self._size
@@ -147,8 +152,9 @@ class MultiStepReplayBuffer(ReplayBuffer):
         device: str = "cpu",
         storage_device: Optional[str] = None,
         pin_memory: bool = True,
+        seed: Optional[int] = None,
     ):
-        super().__init__(max_size, device, storage_device, pin_memory)
+        super().__init__(max_size, device, storage_device, pin_memory, seed=seed)
         self.n_step = int(n_step)
         self.gamma = float(gamma)
 
@@ -196,8 +202,9 @@ class PrioritizedReplayBuffer(ReplayBuffer):
         pin_memory: bool = True,
         n_step: int = 1,
         gamma: float = 0.99,
+        seed: Optional[int] = None,
     ):
-        super().__init__(max_size, device, storage_device, pin_memory)
+        super().__init__(max_size, device, storage_device, pin_memory, seed=seed)
         self.alpha = float(alpha)
         self.n_step = int(n_step)
         self.gamma = float(gamma)

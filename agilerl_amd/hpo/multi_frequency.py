@@ -29,7 +29,7 @@ class MultiFrequencySelection:
         self.frequencies = tuple(int(f) for f in frequencies)
         self.elitism = elitism
         self.migration_fraction = migration_fraction
-        self.rng = rng or np.random.default_rng()
+        self.rng = rng or np.random.default_rng(np.random.randint(0, 2**31 - 1))
         self.generation = 0
 
     # ------------------------------------------------------------------

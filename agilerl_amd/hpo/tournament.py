@@ -29,7 +29,7 @@ class TournamentSelection:
         self.elitism = bool(elitism)
         self.population_size = population_size
         self.eval_loop = eval_loop
-        self.rng = rng or np.random.default_rng()
+        self.rng = rng or np.random.default_rng(np.random.randint(0, 2**31 - 1))
 
     # ------------------------------------------------------------------
     def _fitnesses(self, population) -> np.ndarray:
