@@ -235,7 +235,20 @@ class ModuleDict(EvolvableModule):
 
     def __init__(self, modules: Optional[Dict[str, EvolvableModule]] = None, device: str = "cpu"):
         super().__init__(device)
-        self._modules_dict = nn.ModuleDict(modules or {})
+        built = {}
+        for key, val in (modules or {}).items():
+            if isinstance(val, dict) and "shared_with" in val:
+                # identity-sharing marker (grouped agents share one module)
+                val = built[val["shared_with"]]
+            elif isinstance(val, dict) and "module_cls" in val:
+                # self-describing child spec from init_dict (checkpoint rebuild)
+                cls = val["module_cls"]
+                init = dict(val["init_dict"])
+                if "device" in init:
+                    init["device"] = device
+                val = cls(**init)
+            built[key] = val
+        self._modules_dict = nn.ModuleDict(built)
 
     def __getitem__(self, key: str) -> EvolvableModule:
         return self._modules_dict[key]
@@ -298,7 +311,15 @@ class ModuleDict(EvolvableModule):
 
     @property
     def init_dict(self) -> Dict[str, Any]:
-        return {"modules": {k: v.init_dict for k, v in self._modules_dict.items()}, "device": self.device}
+        specs: Dict[str, Any] = {}
+        seen: Dict[int, str] = {}
+        for k, v in self._modules_dict.items():
+            if id(v) in seen:
+                specs[k] = {"shared_with": seen[id(v)]}
+            else:
+                seen[id(v)] = k
+                specs[k] = {"module_cls": type(v), "init_dict": v.init_dict}
+        return {"modules": specs, "device": self.device}
 
 
 # ---------------------------------------------------------------------------
