@@ -2,6 +2,8 @@ from .packing import pack_padded_batch, unpack_values
 from .scheduler import WarmupCosineLR, create_warmup_cosine_scheduler
 from .offload import activation_offload
 from .chat import apply_chat_template
+from .paged_cache import PagedKVCache
+from .decode_engine import DecodeEngine
 from .lora import (
     LoraConfig,
     LoraLinear,
@@ -22,6 +24,8 @@ __all__ = [
     "create_warmup_cosine_scheduler",
     "activation_offload",
     "apply_chat_template",
+    "PagedKVCache",
+    "DecodeEngine",
     "LoraConfig",
     "LoraLinear",
     "apply_lora",

@@ -1,0 +1,249 @@
+"""Continuous-batching decode engine over the paged KV cache (eager mode).
+
+Round-2 decode-engine milestone 1 (docs/design/round2_perf_plan.md):
+scheduling and storage are final — sequences are admitted the moment a
+slot frees, KV lives in pages — while the per-step attention reads go
+through an eager gather into a transformers ``DynamicCache`` (round 2
+swaps that for a paged-attention HIP kernel and a graph-captured step).
+
+Parity contract (tested): greedy decode through the engine matches
+``model.generate(do_sample=False)`` sequence-for-sequence, including
+ragged admission mid-flight.
+"""
+
+from __future__ import annotations
+
+from dataclasses import dataclass, field
+from typing import Callable, Dict, List, Optional, Tuple
+
+import torch
+
+from .paged_cache import PagedKVCache
+
+__all__ = ["DecodeEngine", "SequenceState"]
+
+
+@dataclass
+class SequenceState:
+    seq_id: int
+    prompt_ids: torch.Tensor  # (T,)
+    max_new_tokens: int
+    temperature: float = 0.0  # 0 => greedy
+    adapter: Optional[str] = None
+    generated: List[int] = field(default_factory=list)
+    prefilled: bool = False
+    done: bool = False
+
+    @property
+    def length(self) -> int:
+        return self.prompt_ids.numel() + len(self.generated)
+
+    @property
+    def last_token(self) -> int:
+        return self.generated[-1] if self.generated else int(self.prompt_ids[-1])
+
+    def output_ids(self) -> torch.Tensor:
+        gen = torch.tensor(self.generated, dtype=torch.long, device=self.prompt_ids.device)
+        return torch.cat([self.prompt_ids, gen])
+
+
+class DecodeEngine:
+    """Batch-of-one-token step loop with continuous admission.
+
+    ``submit()`` any time; ``step()`` prefills newly admitted sequences
+    and decodes one token for every active one; finished sequences are
+    returned and their pages freed.  ``max_batch`` bounds concurrent
+    sequences; excess submissions queue.
+    """
+
+    def __init__(
+        self,
+        model,
+        max_batch: int = 64,
+        num_pages: int = 512,
+        page_size: int = 16,
+        eos_token_id: Optional[int] = None,
+        set_adapter_fn: Optional[Callable[[Optional[str]], None]] = None,
+    ):
+        self.model = model
+        cfg = model.config
+        n_kv = getattr(cfg, "num_key_value_heads", cfg.num_attention_heads)
+        head_dim = getattr(cfg, "head_dim", cfg.hidden_size // cfg.num_attention_heads)
+        dtype = next(model.parameters()).dtype
+        device = str(next(model.parameters()).device)
+        self.cache = PagedKVCache(
+            cfg.num_hidden_layers, n_kv, head_dim,
+            num_pages=num_pages, page_size=page_size, dtype=dtype, device=device,
+        )
+        self.device = device
+        self.max_batch = max_batch
+        self.eos_token_id = eos_token_id
+        self.set_adapter_fn = set_adapter_fn
+        self._next_id = 0
+        self._reserved: Dict[int, int] = {}  # seq_id -> pages promised
+        self.active: Dict[int, SequenceState] = {}
+        self.waiting: List[SequenceState] = []
+        self.steps_run = 0
+
+    # ------------------------------------------------------------------
+    def submit(
+        self,
+        prompt_ids,
+        max_new_tokens: int = 32,
+        temperature: float = 0.0,
+        adapter: Optional[str] = None,
+    ) -> int:
+        ids = torch.as_tensor(prompt_ids, dtype=torch.long, device=self.device).reshape(-1)
+        seq = SequenceState(self._next_id, ids, max_new_tokens, temperature, adapter)
+        self._next_id += 1
+        self.waiting.append(seq)
+        return seq.seq_id
+
+    @property
+    def num_active(self) -> int:
+        return len(self.active)
+
+    def has_work(self) -> bool:
+        return bool(self.active or self.waiting)
+
+    # ------------------------------------------------------------------
+    def _available_pages(self) -> int:
+        outstanding = sum(
+            max(0, need - len(self.cache.page_tables.get(sid, ())))
+            for sid, need in self._reserved.items()
+        )
+        return self.cache.free_pages - outstanding
+
+    def _admit(self) -> None:
+        while self.waiting and len(self.active) < self.max_batch:
+            seq = self.waiting[0]
+            need = self.cache.pages_for(seq.prompt_ids.numel() + seq.max_new_tokens)
+            if need > self._available_pages():
+                break  # admit in order; wait for pages
+            self.waiting.pop(0)
+            self.cache.alloc(seq.seq_id)
+            self._reserved[seq.seq_id] = need
+            self.active[seq.seq_id] = seq
+
+    def _set_adapter(self, adapter: Optional[str]) -> None:
+        if self.set_adapter_fn is not None:
+            self.set_adapter_fn(adapter)
+
+    @torch.no_grad()
+    def _prefill(self, seqs: List[SequenceState]) -> None:
+        """Forward full prompts (left-padded batch), store KV pages, and
+        emit each sequence's first generated token."""
+        from transformers.cache_utils import DynamicCache
+
+        L = max(s.prompt_ids.numel() for s in seqs)
+        B = len(seqs)
+        ids = torch.zeros((B, L), dtype=torch.long, device=self.device)
+        mask = torch.zeros((B, L), dtype=torch.long, device=self.device)
+        for i, s in enumerate(seqs):
+            n = s.prompt_ids.numel()
+            ids[i, L - n :] = s.prompt_ids
+            mask[i, L - n :] = 1
+        pos = (mask.cumsum(-1) - 1).clamp(min=0)
+        cache = DynamicCache()
+        out = self.model(
+            input_ids=ids, attention_mask=mask, position_ids=pos,
+            past_key_values=cache, use_cache=True,
+        )
+        for i, s in enumerate(seqs):
+            n = s.prompt_ids.numel()
+            layers_k, layers_v = [], []
+            for layer in range(self.cache.num_layers):
+                k_l, v_l = self._layer_kv(out.past_key_values, layer)
+                layers_k.append(k_l[i, :, L - n :].permute(1, 0, 2))  # (T, n_kv, D)
+                layers_v.append(v_l[i, :, L - n :].permute(1, 0, 2))
+            self.cache.append(s.seq_id, torch.stack(layers_k), torch.stack(layers_v))
+            s.prefilled = True
+            logits = out.logits[i, -1]
+            s.generated.append(self._select(logits, s))
+
+    @staticmethod
+    def _layer_kv(cache, layer: int) -> Tuple[torch.Tensor, torch.Tensor]:
+        lay = cache.layers[layer]
+        return lay.keys, lay.values
+
+    def _select(self, logits: torch.Tensor, seq: SequenceState) -> int:
+        if seq.temperature and seq.temperature > 0:
+            probs = torch.softmax(logits / seq.temperature, dim=-1)
+            return int(torch.multinomial(probs, 1))
+        return int(logits.argmax())
+
+    @torch.no_grad()
+    def _decode(self, seqs: List[SequenceState]) -> None:
+        """One token for every sequence: gather pages -> DynamicCache ->
+        single batched forward at the shared end position."""
+        from transformers.cache_utils import DynamicCache
+
+        seq_ids = [s.seq_id for s in seqs]
+        k, v, mask = self.cache.gather(seq_ids)
+        legacy = tuple((k[l], v[l]) for l in range(self.cache.num_layers))
+        cache = DynamicCache(config=self.model.config)
+        for l, (kl, vl) in enumerate(legacy):
+            cache.update(kl, vl, l)
+        ids = torch.tensor([[s.last_token] for s in seqs], dtype=torch.long, device=self.device)
+        full_mask = torch.cat(
+            [mask, torch.ones((len(seqs), 1), dtype=torch.long, device=self.device)], dim=1
+        )
+        pos = torch.tensor([[self.cache.lengths[s.seq_id]] for s in seqs],
+                           dtype=torch.long, device=self.device)
+        out = self.model(
+            input_ids=ids, attention_mask=full_mask, position_ids=pos,
+            past_key_values=cache, use_cache=True,
+        )
+        for i, s in enumerate(seqs):
+            layers_k, layers_v = [], []
+            for layer in range(self.cache.num_layers):
+                k_l, v_l = self._layer_kv(out.past_key_values, layer)
+                layers_k.append(k_l[i, :, -1:].permute(1, 0, 2))
+                layers_v.append(v_l[i, :, -1:].permute(1, 0, 2))
+            self.cache.append(s.seq_id, torch.stack(layers_k), torch.stack(layers_v))
+            s.generated.append(self._select(out.logits[i, -1], s))
+
+    # ------------------------------------------------------------------
+    def step(self) -> List[Tuple[int, torch.Tensor]]:
+        """Admit, prefill, decode one token each, retire finished.
+
+        Returns [(seq_id, full_ids)] for sequences that finished this step.
+        """
+        self._admit()
+        if not self.active:
+            return []
+        self.steps_run += 1
+        by_adapter: Dict[Optional[str], List[SequenceState]] = {}
+        for s in self.active.values():
+            by_adapter.setdefault(s.adapter, []).append(s)
+        for adapter, seqs in by_adapter.items():
+            self._set_adapter(adapter)
+            new = [s for s in seqs if not s.prefilled]
+            old = [s for s in seqs if s.prefilled]
+            if new:
+                # the token the last prefill/decode emitted has no KV yet;
+                # it becomes this step's decode input for old sequences
+                self._prefill(new)
+            if old:
+                self._decode(old)
+        finished = []
+        for s in list(self.active.values()):
+            hit_eos = self.eos_token_id is not None and s.generated and \
+                s.generated[-1] == self.eos_token_id
+            if len(s.generated) >= s.max_new_tokens or hit_eos:
+                s.done = True
+                finished.append((s.seq_id, s.output_ids()))
+                self.cache.free(s.seq_id)
+                self._reserved.pop(s.seq_id, None)
+                del self.active[s.seq_id]
+        return finished
+
+    def run_all(self, max_steps: int = 10_000) -> Dict[int, torch.Tensor]:
+        """Drive until every submitted sequence finishes."""
+        results: Dict[int, torch.Tensor] = {}
+        for _ in range(max_steps):
+            if not self.has_work():
+                break
+            for seq_id, ids in self.step():
+                results[seq_id] = ids
+        return results
