@@ -219,6 +219,7 @@ class EvolvableAlgorithm(metaclass=AlgorithmMeta):
             "_init_args",
             "_wrapped",
             "accelerator",
+            "_decode_engine",  # paged-KV pools + model ref; rebuilt lazily
         }
         skip.update(self._GRAPH_ATTRS)
         skip.update(self.registry.all_network_names())

@@ -299,6 +299,55 @@ class LLMAlgorithm(EvolvableAlgorithm):
         return out
 
     @torch.no_grad()
+    def generate_paged(
+        self,
+        input_ids: torch.Tensor,
+        attention_mask: torch.Tensor,
+        max_new_tokens: Optional[int] = None,
+        do_sample: bool = True,
+        temperature: Optional[float] = None,
+    ) -> torch.Tensor:
+        """Generation through the continuous-batching paged-KV engine
+        (llm/decode_engine.py).  Same (B, P+C) rectangular contract as
+        ``generate``; completions are right-padded with pad_token_id."""
+        from ...llm.decode_engine import DecodeEngine
+
+        C = max_new_tokens or self.max_completion_tokens
+        B, P = input_ids.shape
+        pad_id = getattr(self.model.config, "pad_token_id", 0) or 0
+        tokens_needed = int(attention_mask.sum()) + B * C
+        engine = getattr(self, "_decode_engine", None)
+        page_size = 16
+        if engine is None or engine.cache.num_pages * page_size < tokens_needed:
+            engine = DecodeEngine(
+                self.model,
+                max_batch=max(B, 8),
+                num_pages=max(64, 2 * (tokens_needed // page_size + 1)),
+                page_size=page_size,
+                eos_token_id=getattr(self.model.config, "eos_token_id", None),
+                set_adapter_fn=self._activate,
+            )
+            self._decode_engine = engine
+        engine.max_batch = max(engine.max_batch, B)
+        self.model.eval()
+        temp = temperature if temperature is not None else self.temperature
+        sids = []
+        for i in range(B):
+            row_mask = attention_mask[i].bool()
+            sids.append(engine.submit(
+                input_ids[i][row_mask], max_new_tokens=C,
+                temperature=temp if do_sample else 0.0, adapter="self",
+            ))
+        results = engine.run_all()
+        self.model.train()
+        out = torch.full((B, P + C), pad_id, dtype=torch.long, device=self.device)
+        out[:, :P] = input_ids
+        for i, sid in enumerate(sids):
+            comp = results[sid][int(attention_mask[i].sum()):]
+            out[i, P : P + comp.numel()] = comp.to(self.device)
+        return out
+
+    @torch.no_grad()
     def generate(
         self,
         input_ids: torch.Tensor,

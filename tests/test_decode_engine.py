@@ -104,3 +104,40 @@ class TestDecodeEngine:
         engine.submit(torch.randint(1, 64, (4,)), max_new_tokens=4, temperature=1.0)
         results = engine.run_all()
         assert len(results) == 1
+
+
+class TestGRPOPagedGeneration:
+    def test_generate_paged_matches_generate_greedy(self):
+        from agilerl_amd.algorithms.llm.grpo import GRPO
+
+        tiny = dict(model_type="llama", vocab_size=64, hidden_size=32,
+                    intermediate_size=64, num_hidden_layers=1,
+                    num_attention_heads=2, num_key_value_heads=1,
+                    max_position_embeddings=128, pad_token_id=0)
+        torch.manual_seed(0)
+        agent = GRPO(model_config=tiny, dtype=torch.float32, lora_config={"r": 2},
+                     max_completion_tokens=6)
+        ids = torch.randint(1, 64, (3, 5))
+        ids[0, :2] = 0  # left-padded row
+        mask = (ids != 0).long()
+        ref = agent.generate(ids, mask, do_sample=False)
+        paged = agent.generate_paged(ids, mask, do_sample=False)
+        torch.testing.assert_close(paged, ref)
+
+    def test_generate_paged_sampling_shape(self):
+        from agilerl_amd.algorithms.llm.grpo import GRPO
+
+        tiny = dict(model_type="llama", vocab_size=64, hidden_size=32,
+                    intermediate_size=64, num_hidden_layers=1,
+                    num_attention_heads=2, num_key_value_heads=1,
+                    max_position_embeddings=128, pad_token_id=0)
+        agent = GRPO(model_config=tiny, dtype=torch.float32, lora_config={"r": 2},
+                     max_completion_tokens=4)
+        ids = torch.randint(1, 64, (4, 5))
+        mask = torch.ones_like(ids)
+        out = agent.generate_paged(ids, mask, do_sample=True)
+        assert out.shape == (4, 9)
+        # engine reused on second call
+        eng = agent._decode_engine
+        agent.generate_paged(ids, mask, do_sample=True)
+        assert agent._decode_engine is eng
