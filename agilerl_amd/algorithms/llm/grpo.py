@@ -57,6 +57,7 @@ class GRPO(LLMAlgorithm):
         dtype: torch.dtype = torch.bfloat16,
         gradient_checkpointing: bool = False,
         use_packing: bool = False,
+        generation: str = "hf",
         device: str = "cpu",
     ):
         super().__init__(
@@ -72,6 +73,11 @@ class GRPO(LLMAlgorithm):
         # padding-free grad/old-policy passes (compute_logprobs_packed);
         # opt-in until validated at 8-GPU scale
         self.use_packing = bool(use_packing)
+        # "hf" = model.generate (default); "paged" = continuous-batching
+        # paged-KV engine (llm/decode_engine.py, greedy-parity tested)
+        if generation not in ("hf", "paged"):
+            raise ValueError(f"generation must be 'hf' or 'paged', got {generation!r}")
+        self.generation = generation
         self.clip_coef = float(clip_coef)
         self.clip_coef_lower = float(clip_coef_lower) if clip_coef_lower is not None else None
         self.beta = float(beta)  # k3 KL coefficient
@@ -85,6 +91,8 @@ class GRPO(LLMAlgorithm):
         (B, P+C)."""
         input_ids = prompts["input_ids"].to(self.device)
         attention_mask = prompts["attention_mask"].to(self.device)
+        if self.generation == "paged":
+            return self.generate_paged(input_ids, attention_mask, do_sample=training)
         return self.generate(input_ids, attention_mask, do_sample=training)
 
     # ------------------------------------------------------------------

@@ -141,3 +141,25 @@ class TestGRPOPagedGeneration:
         eng = agent._decode_engine
         agent.generate_paged(ids, mask, do_sample=True)
         assert agent._decode_engine is eng
+
+
+def test_grpo_full_loop_with_paged_generation():
+    import numpy as np
+
+    from agilerl_amd.algorithms.llm.grpo import GRPO
+    from agilerl_amd.llm_envs import TokenReasoningGym, make_grpo_experiences
+
+    tiny = dict(model_type="llama", vocab_size=64, hidden_size=32,
+                intermediate_size=64, num_hidden_layers=1,
+                num_attention_heads=2, num_key_value_heads=1,
+                max_position_embeddings=128, pad_token_id=0)
+    torch.manual_seed(0)
+    agent = GRPO(model_config=tiny, dtype=torch.float32, lora_config={"r": 2},
+                 group_size=2, max_completion_tokens=4, generation="paged")
+    env = TokenReasoningGym(vocab_size=64, prompt_len=4, data_batch_size=2,
+                            group_size=2, seed=0)
+    prompts = env.reset()
+    seqs = agent.get_action(prompts)
+    assert seqs.shape == (4, 8)
+    stats = agent.learn(make_grpo_experiences(env, seqs, env.score(seqs)))
+    assert np.isfinite(stats["loss"])
