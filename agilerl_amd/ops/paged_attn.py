@@ -1,0 +1,60 @@
+"""Paged-attention decode: HIP kernel dispatch + eager reference.
+
+Round-2 decode-engine milestone 2 (docs/design/round2_perf_plan.md §2).
+The eager reference is also the CPU implementation the decode engine can
+call; the HIP kernel reads K/V straight from the page pool through the
+page tables (no gather).
+"""
+
+from __future__ import annotations
+
+import math
+
+import torch
+
+from .backend import extension, use_hip
+
+__all__ = ["paged_attention_decode"]
+
+
+def _eager_reference(q, k_pool, v_pool, page_table, lengths, scale):
+    B, Hq, D = q.shape
+    S, Hkv = k_pool.shape[1], k_pool.shape[2]
+    gqa = Hq // Hkv
+    out = torch.empty(B, Hq, D, dtype=torch.float32, device=q.device)
+    qf = q.float()
+    for b in range(B):
+        n = int(lengths[b])
+        pages = page_table[b, : (n + S - 1) // S].long()
+        k = k_pool[pages].reshape(-1, Hkv, D)[:n].float()  # (n, Hkv, D)
+        v = v_pool[pages].reshape(-1, Hkv, D)[:n].float()
+        for hk in range(Hkv):
+            for g in range(gqa):
+                h = hk * gqa + g
+                scores = (k[:, hk] @ qf[b, h]) * scale  # (n,)
+                p = torch.softmax(scores, dim=0)
+                out[b, h] = p @ v[:, hk]
+    return out
+
+
+def paged_attention_decode(
+    q: torch.Tensor,
+    k_pool: torch.Tensor,
+    v_pool: torch.Tensor,
+    page_table: torch.Tensor,
+    lengths: torch.Tensor,
+    scale: float = None,
+) -> torch.Tensor:
+    """Single-token decode attention over paged K/V.
+
+    q: (B, Hq, D); k_pool/v_pool: (num_pages, page_size, Hkv, D);
+    page_table: (B, max_pages) int; lengths: (B,) int.
+    Returns (B, Hq, D) fp32.
+    """
+    if scale is None:
+        scale = 1.0 / math.sqrt(q.shape[-1])
+    ext = extension()
+    if use_hip(q) and ext is not None and hasattr(ext, "paged_attn_decode") \
+            and q.dtype == torch.bfloat16:
+        return ext.paged_attn_decode(q, k_pool, v_pool, page_table, lengths, float(scale))
+    return _eager_reference(q, k_pool, v_pool, page_table, lengths, scale)

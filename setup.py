@@ -23,6 +23,7 @@ ext = CUDAExtension(
         "agilerl_amd/ops/csrc/lm_ops.hip",
         "agilerl_amd/ops/csrc/norm_ops.hip",
         "agilerl_amd/ops/csrc/act_ops.hip",
+        "agilerl_amd/ops/csrc/paged_attn.hip",
     ],
     extra_compile_args={
         "cxx": ["-O3", "-std=c++17"],

@@ -281,3 +281,24 @@ def test_llama_swiglu_patch_matches_unpatched():
     with torch.no_grad():
         after = model(ids).logits.float()
     torch.testing.assert_close(after, before, atol=5e-2, rtol=5e-2)
+
+
+@pytest.mark.gpu
+def test_paged_attn_decode_matches_eager():
+    """HIP paged-attention decode vs the fp32 eager reference: random
+    pools, ragged lengths, GQA 4:1 and 1:1, head dims 64/128."""
+    from agilerl_amd.ops.paged_attn import _eager_reference, paged_attention_decode
+
+    torch.manual_seed(0)
+    for (Hq, Hkv, D) in ((8, 2, 128), (4, 4, 64)):
+        B, S, P = 3, 16, 32
+        q = torch.randn(B, Hq, D, device="cuda", dtype=torch.bfloat16)
+        kp = torch.randn(P, S, Hkv, D, device="cuda", dtype=torch.bfloat16)
+        vp = torch.randn(P, S, Hkv, D, device="cuda", dtype=torch.bfloat16)
+        lengths = torch.tensor([37, 5, 64], device="cuda", dtype=torch.int32)
+        max_pages = 5
+        table = torch.randperm(P, device="cuda")[: B * max_pages].reshape(B, max_pages).to(torch.int32)
+        out = paged_attention_decode(q, kp, vp, table, lengths)
+        ref = _eager_reference(q.float(), kp.float(), vp.float(), table, lengths,
+                               1.0 / D ** 0.5)
+        torch.testing.assert_close(out, ref, atol=2e-2, rtol=2e-2)
