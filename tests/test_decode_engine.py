@@ -189,3 +189,35 @@ def test_paged_attention_eager_reference_cpu():
             hk = h // gqa
             att = torch.softmax((k[:, hk] @ q[b, h]) / math.sqrt(D), 0)
             torch.testing.assert_close(out[b, h], att @ v[:, hk], atol=1e-5, rtol=1e-5)
+
+
+def test_engine_multi_adapter_population_sharing():
+    """One engine serves a whole GRPO population: per-sequence adapters
+    are grouped per step and activated through set_adapter_fn."""
+    from agilerl_amd.algorithms.llm.grpo import GRPO
+
+    tiny = dict(model_type="llama", vocab_size=64, hidden_size=32,
+                intermediate_size=64, num_hidden_layers=1,
+                num_attention_heads=2, num_key_value_heads=1,
+                max_position_embeddings=128, pad_token_id=0)
+    torch.manual_seed(0)
+    pop = GRPO.population(2, model_config=tiny, dtype=torch.float32,
+                          lora_config={"r": 2}, max_completion_tokens=4)
+    a0, a1 = pop
+    assert a0.model is a1.model
+    activations = []
+
+    def set_adapter(name):
+        activations.append(name)
+        from agilerl_amd.llm import set_active_adapter
+
+        set_active_adapter(a0.model, name)
+
+    engine = DecodeEngine(a0.model, num_pages=64, page_size=4,
+                          set_adapter_fn=set_adapter)
+    p = torch.randint(1, 64, (4,))
+    s0 = engine.submit(p, max_new_tokens=3, adapter=a0.adapter_name)
+    s1 = engine.submit(p, max_new_tokens=3, adapter=a1.adapter_name)
+    results = engine.run_all()
+    assert set(results) == {s0, s1}
+    assert a0.adapter_name in activations and a1.adapter_name in activations
