@@ -32,6 +32,11 @@ class QNetwork(EvolvableNetwork):
         latent_dim: int = 64,
         device: str = "cpu",
     ):
+        if not hasattr(action_space, "n"):
+            raise TypeError(
+                f"QNetwork requires a discrete action space, got "
+                f"{type(action_space).__name__} (use DDPG/TD3/PPO for Box actions)"
+            )
         self.action_space = action_space
         super().__init__(
             observation_space,
