@@ -74,6 +74,8 @@ def algo_workload(name: str) -> str:
 def resolve_algo_class(name: str):
     import importlib
 
+    if name not in ALGO_REGISTRY:
+        raise KeyError(f"Unknown algorithm '{name}'. Registered: {sorted(ALGO_REGISTRY)}")
     path = ALGO_REGISTRY[name]["cls"]
     module, cls = path.rsplit(".", 1)
     return getattr(importlib.import_module(module), cls)
