@@ -84,6 +84,9 @@ class DecodeEngine:
         self.active: Dict[int, SequenceState] = {}
         self.waiting: List[SequenceState] = []
         self.steps_run = 0
+        self.tokens_generated = 0
+        self.sequences_finished = 0
+        self._t_start = None
 
     # ------------------------------------------------------------------
     def submit(
@@ -209,9 +212,13 @@ class DecodeEngine:
 
         Returns [(seq_id, full_ids)] for sequences that finished this step.
         """
+        import time as _time
+
         self._admit()
         if not self.active:
             return []
+        if self._t_start is None:
+            self._t_start = _time.time()
         self.steps_run += 1
         by_adapter: Dict[Optional[str], List[SequenceState]] = {}
         for s in self.active.values():
@@ -232,11 +239,28 @@ class DecodeEngine:
                 s.generated[-1] == self.eos_token_id
             if len(s.generated) >= s.max_new_tokens or hit_eos:
                 s.done = True
+                self.sequences_finished += 1
+                self.tokens_generated += len(s.generated)
                 finished.append((s.seq_id, s.output_ids()))
                 self.cache.free(s.seq_id)
                 self._reserved.pop(s.seq_id, None)
                 del self.active[s.seq_id]
         return finished
+
+    def stats(self) -> Dict[str, float]:
+        """Decode observability: steps, finished sequences, tokens/s."""
+        import time as _time
+
+        elapsed = (_time.time() - self._t_start) if self._t_start else 0.0
+        return {
+            "steps": self.steps_run,
+            "active": len(self.active),
+            "waiting": len(self.waiting),
+            "sequences_finished": self.sequences_finished,
+            "tokens_generated": self.tokens_generated,
+            "tokens_per_sec": self.tokens_generated / elapsed if elapsed > 0 else 0.0,
+            "free_pages": self.cache.free_pages,
+        }
 
     def run_all(self, max_steps: int = 10_000) -> Dict[int, torch.Tensor]:
         """Drive until every submitted sequence finishes."""

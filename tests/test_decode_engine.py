@@ -221,3 +221,15 @@ def test_engine_multi_adapter_population_sharing():
     results = engine.run_all()
     assert set(results) == {s0, s1}
     assert a0.adapter_name in activations and a1.adapter_name in activations
+
+
+def test_engine_stats():
+    model = tiny_model()
+    engine = DecodeEngine(model, num_pages=32, page_size=4)
+    engine.submit(torch.randint(1, 64, (4,)), max_new_tokens=3)
+    engine.run_all()
+    st = engine.stats()
+    assert st["sequences_finished"] == 1
+    assert st["tokens_generated"] == 3
+    assert st["free_pages"] == 32
+    assert st["tokens_per_sec"] >= 0
