@@ -247,6 +247,19 @@ class DecodeEngine:
                 del self.active[s.seq_id]
         return finished
 
+    def cancel(self, seq_id: int) -> bool:
+        """Abort a sequence (waiting or active); frees its pages."""
+        for i, w in enumerate(self.waiting):
+            if w.seq_id == seq_id:
+                self.waiting.pop(i)
+                return True
+        if seq_id in self.active:
+            self.cache.free(seq_id)
+            self._reserved.pop(seq_id, None)
+            del self.active[seq_id]
+            return True
+        return False
+
     def stats(self) -> Dict[str, float]:
         """Decode observability: steps, finished sequences, tokens/s."""
         import time as _time

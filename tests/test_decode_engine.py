@@ -233,3 +233,16 @@ def test_engine_stats():
     assert st["tokens_generated"] == 3
     assert st["free_pages"] == 32
     assert st["tokens_per_sec"] >= 0
+
+
+def test_engine_cancel():
+    model = tiny_model()
+    engine = DecodeEngine(model, num_pages=32, page_size=4)
+    s1 = engine.submit(torch.randint(1, 64, (4,)), max_new_tokens=8)
+    s2 = engine.submit(torch.randint(1, 64, (4,)), max_new_tokens=8)
+    engine.step()  # both active
+    assert engine.cancel(s1)
+    results = engine.run_all()
+    assert s1 not in results and s2 in results
+    assert engine.cache.free_pages == 32
+    assert not engine.cancel(999)
