@@ -183,3 +183,33 @@ class TestGroupedAgents:
         obs, _ = env.reset()
         ea, lp, v = agent.get_action(obs)
         assert ea["agent_0"].shape == (2,)
+
+
+class TestContinuousMPE:
+    def test_continuous_speaker_listener_with_maddpg(self):
+        """continuous_actions=True: Box action spaces end to end through
+        MADDPG collect+learn (the reference's default MADDPG setting)."""
+        import torch
+
+        from agilerl_amd.spaces import Box
+
+        env = SpeakerListenerVecEnv(num_envs=4, seed=0, continuous_actions=True)
+        for aid in env.agents:
+            assert isinstance(env.action_spaces[aid], Box)
+        pop = MADDPG.population(
+            1, env.observation_spaces, env.action_spaces, agent_ids=env.agents,
+            batch_size=16, net_config={"arch": "mlp", "hidden_size": [16]},
+        )
+        agent = pop[0]
+        obs, _ = env.reset()
+        buf = ReplayBuffer(300)
+        for _ in range(30):
+            env_actions, raw = agent.get_action(obs)
+            next_obs, rewards, term, trunc, _ = env.step(env_actions)
+            buf.add(obs=obs, action=raw,
+                    reward={a: rewards[a] for a in env.agents},
+                    next_obs=next_obs,
+                    done={a: term[a].astype(np.float32) for a in env.agents})
+            obs = next_obs
+        loss = agent.learn(buf.sample(16))
+        assert np.isfinite(loss)
