@@ -35,6 +35,7 @@ class SFT(LLMAlgorithm):
         max_grad_norm: float = 1.0,
         dtype: torch.dtype = torch.bfloat16,
         gradient_checkpointing: bool = False,
+        use_packing: bool = False,
         device: str = "cpu",
     ):
         super().__init__(
@@ -44,6 +45,8 @@ class SFT(LLMAlgorithm):
             dtype=dtype, gradient_checkpointing=gradient_checkpointing, device=device,
             name="SFT",
         )
+        # padding-free grad pass (compute_logprobs_packed); opt-in as in GRPO
+        self.use_packing = bool(use_packing)
 
     def learn(self, experiences: Dict[str, Any]) -> Dict[str, float]:
         """experiences: ids (B, T), attention_mask (B, T), action_mask
@@ -60,7 +63,8 @@ class SFT(LLMAlgorithm):
         total_loss, n = 0.0, 0
         for s in range(0, B, mb):
             e = min(s + mb, B)
-            logp = self.compute_logprobs(ids[s:e], attention_mask[s:e], with_grad=True)
+            fn = self.compute_logprobs_packed if self.use_packing else self.compute_logprobs
+            logp = fn(ids[s:e], attention_mask[s:e], with_grad=True)
             loss = -ops.masked_mean(logp, action_mask[s:e])
             self.backward_and_step(loss)
             total_loss += float(loss.detach())

@@ -531,3 +531,24 @@ class TestPacking:
         stats = agent.learn(exp)
         assert np.isfinite(stats["loss"])
         assert stats["kl"] >= -1e-5
+
+
+def test_sft_packed_matches_padded_loss():
+    torch.manual_seed(0)
+    from agilerl_amd.algorithms.llm.sft import SFT
+
+    def mk(**kw):
+        return tiny_agent(SFT, lr=1e-3, **kw)
+
+    a1, a2 = mk(), mk(use_packing=True)
+    B, T = 4, 12
+    ids = torch.randint(1, 128, (B, T))
+    mask = torch.ones(B, T, dtype=torch.long)
+    mask[0, :5] = 0
+    ids[0, :5] = 0
+    amask = torch.zeros(B, T - 1)
+    amask[:, 6:] = 1.0
+    amask[0, : 5] = 0.0
+    l1 = a1.learn({"ids": ids, "attention_mask": mask, "action_mask": amask})
+    l2 = a2.learn({"ids": ids, "attention_mask": mask, "action_mask": amask})
+    assert abs(l1["loss"] - l2["loss"]) < 1e-4
