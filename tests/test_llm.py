@@ -538,6 +538,7 @@ def test_sft_packed_matches_padded_loss():
     from agilerl_amd.algorithms.llm.sft import SFT
 
     def mk(**kw):
+        torch.manual_seed(0)
         return tiny_agent(SFT, lr=1e-3, **kw)
 
     a1, a2 = mk(), mk(use_packing=True)
