@@ -38,6 +38,7 @@ class DPO(LLMAlgorithm):
         max_grad_norm: float = 1.0,
         dtype: torch.dtype = torch.bfloat16,
         gradient_checkpointing: bool = False,
+        use_packing: bool = False,
         device: str = "cpu",
     ):
         super().__init__(
@@ -49,9 +50,12 @@ class DPO(LLMAlgorithm):
         )
         self.beta = float(beta)
         self.label_smoothing = float(label_smoothing)
+        # padding-free logprob passes (compute_logprobs_packed); opt-in
+        self.use_packing = bool(use_packing)
 
     def _seq_logp(self, ids, attention_mask, action_mask, with_grad: bool, adapter="self"):
-        logp = self.compute_logprobs(ids, attention_mask, adapter=adapter, with_grad=with_grad)
+        fn = self.compute_logprobs_packed if self.use_packing else self.compute_logprobs
+        logp = fn(ids, attention_mask, adapter=adapter, with_grad=with_grad)
         return (logp * action_mask).sum(dim=1)
 
     def learn(self, experiences: Dict[str, Any]) -> Dict[str, float]:

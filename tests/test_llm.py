@@ -553,3 +553,29 @@ def test_sft_packed_matches_padded_loss():
     l1 = a1.learn({"ids": ids, "attention_mask": mask, "action_mask": amask})
     l2 = a2.learn({"ids": ids, "attention_mask": mask, "action_mask": amask})
     assert abs(l1["loss"] - l2["loss"]) < 1e-4
+
+
+def test_dpo_packed_matches_padded_loss():
+    from agilerl_amd.algorithms.llm.dpo import DPO
+
+    def mk(**kw):
+        torch.manual_seed(0)
+        return tiny_agent(DPO, lr=1e-3, beta=0.1, **kw)
+
+    a1, a2 = mk(), mk(use_packing=True)
+    B, T = 4, 10
+    c_ids = torch.randint(1, 128, (B, T))
+    r_ids = torch.randint(1, 128, (B, T))
+    c_am = torch.ones(B, T, dtype=torch.long)
+    c_am[1, :3] = 0
+    c_ids[1, :3] = 0
+    r_am = torch.ones(B, T, dtype=torch.long)
+    cm = torch.zeros(B, T - 1); cm[:, 4:] = 1.0
+    rm = torch.zeros(B, T - 1); rm[:, 4:] = 1.0
+    exp = {"chosen_ids": c_ids, "rejected_ids": r_ids,
+           "chosen_attention_mask": c_am, "rejected_attention_mask": r_am,
+           "chosen_mask": cm, "rejected_mask": rm}
+    l1 = a1.learn(dict(exp))
+    l2 = a2.learn(dict(exp))
+    assert abs(l1["loss"] - l2["loss"]) < 1e-4
+    assert abs(l1["margin"] - l2["margin"]) < 1e-3
