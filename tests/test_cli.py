@@ -115,3 +115,24 @@ def test_kernel_isa_report_tool():
     assert out.returncode == 0, out.stderr[-1000:]
     assert "rmsnorm_fwd_kernel" in out.stdout
     assert "swiglu_fwd_kernel" in out.stdout
+
+
+@pytest.mark.slow
+def test_bench_distributed_two_ranks_cpu():
+    """Exactly the driver's multi-GPU launch shape, on CPU gloo: torchrun
+    2 ranks, whole-job aggregate from rank 0 only."""
+    env = dict(os.environ)
+    env.update({"MASTER_ADDR": "127.0.0.1"})
+    out = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+         "--master-port", "29517", "bench.py", "--gpus", "2",
+         "--steps", "2", "--warmup", "1", "--num-envs", "8", "--pop-size", "2"],
+        cwd=ROOT, capture_output=True, text=True, timeout=900, env=env,
+    )
+    assert out.returncode == 0, out.stderr[-3000:]
+    lines = [l for l in out.stdout.strip().splitlines() if l.startswith("{")]
+    assert len(lines) == 1, out.stdout[-2000:]
+    payload = json.loads(lines[0])
+    assert payload["n_gpus"] == 2
+    assert payload["value"] > 0
