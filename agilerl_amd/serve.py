@@ -158,9 +158,15 @@ def create_app(agent, lock: Optional[threading.Lock] = None):
                                     detail="agent has no tokenizer; pass input_ids")
             text = req.prompt
             if req.messages is not None:
-                from .llm.chat import apply_chat_template
-
-                text = apply_chat_template(req.messages, tok)
+                if getattr(tok, "chat_template", None):
+                    text = tok.apply_chat_template(
+                        req.messages, tokenize=False, add_generation_prompt=True
+                    )
+                else:
+                    text = "\n".join(
+                        f"{m.get('role', 'user').capitalize()}: {m.get('content', '')}"
+                        for m in req.messages
+                    ) + "\nAssistant:"
             ids = torch.as_tensor([tok.encode(text)], dtype=torch.long)
         else:
             raise HTTPException(status_code=400, detail="prompt, messages or input_ids required")

@@ -77,3 +77,35 @@ def test_serve_llm_generate():
     # classic agents refuse cleanly
     r = client.post("/generate", json={"prompt": "hi"})
     assert r.status_code in (200, 400)  # GRPO without tokenizer -> 400
+
+
+def test_serve_llm_generate_with_messages():
+    import torch
+
+    from agilerl_amd.algorithms.llm.grpo import GRPO
+
+    class WordTok:
+        chat_template = None
+        pad_token_id = 0
+
+        def encode(self, text):
+            return [(hash(w) % 60) + 1 for w in text.split()][:16]
+
+        def decode(self, ids):
+            return " ".join(f"t{int(i)}" for i in ids)
+
+    tiny = dict(model_type="llama", vocab_size=64, hidden_size=32,
+                intermediate_size=64, num_hidden_layers=1,
+                num_attention_heads=2, num_key_value_heads=1,
+                max_position_embeddings=128, pad_token_id=0)
+    agent = GRPO(model_config=tiny, dtype=torch.float32, lora_config={"r": 2},
+                 max_completion_tokens=4, tokenizer=WordTok())
+    client = TestClient(create_app(agent))
+    r = client.post("/generate", json={
+        "messages": [{"role": "user", "content": "hello there"}],
+        "max_new_tokens": 3,
+    })
+    assert r.status_code == 200, r.text
+    body = r.json()
+    assert len(body["completion_ids"][0]) == 3
+    assert isinstance(body["completion"][0], str)
