@@ -5,7 +5,8 @@ from .reasoning import ReasoningGym, TokenReasoningGym
 HuggingFaceGym = ReasoningGym
 from .sft import SFTGym, SyntheticSFTGym
 from .preference import PreferenceGym, SyntheticPreferenceGym
-from .multiturn import MultiTurnTokenEnv, TokenGuessEnv, SyncMultiTurnVecEnv
+from .multiturn import (MultiTurnTokenEnv, TokenGuessEnv, SyncMultiTurnVecEnv,
+                        TextMultiTurnEnv, SearchQAEnv)
 from .search import SearchTool, FormatRewardWrapper, extract_answer
 
 __all__ = [
@@ -21,6 +22,8 @@ __all__ = [
     "MultiTurnTokenEnv",
     "TokenGuessEnv",
     "SyncMultiTurnVecEnv",
+    "TextMultiTurnEnv",
+    "SearchQAEnv",
     "SearchTool",
     "FormatRewardWrapper",
     "extract_answer",

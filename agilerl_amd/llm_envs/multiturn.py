@@ -17,7 +17,7 @@ from typing import Any, Callable, Dict, List, Optional, Tuple
 import numpy as np
 import torch
 
-__all__ = ["MultiTurnTokenEnv", "TokenGuessEnv", "SyncMultiTurnVecEnv"]
+__all__ = ["MultiTurnTokenEnv", "TokenGuessEnv", "SyncMultiTurnVecEnv", "TextMultiTurnEnv", "SearchQAEnv"]
 
 
 class MultiTurnTokenEnv:
@@ -183,3 +183,94 @@ class SyncMultiTurnVecEnv:
             "turn_ids": turn_ids,
             "rewards": torch.tensor(self._rewards, dtype=torch.float32),
         }
+
+
+class TextMultiTurnEnv(MultiTurnTokenEnv):
+    """Tokenizer boundary for text-level multi-turn tasks.
+
+    Reference parity: ``agilerl/llm_envs/token_observation.py:43``
+    (TokenObservationWrapper) — the conversation lives as text turns; the
+    tokenizer converts at the env boundary so the vec env / algorithms
+    stay purely token-space.  Subclasses implement ``initial_message``
+    and ``respond_text``; feedback is rendered as a new user turn through
+    the chat template (or a plain role-prefixed format without one).
+    """
+
+    def __init__(self, tokenizer, system_prompt: Optional[str] = None,
+                 max_prompt_tokens: int = 512):
+        self.tokenizer = tokenizer
+        self.system_prompt = system_prompt
+        self.max_prompt_tokens = max_prompt_tokens
+
+    # -- subclass hooks -------------------------------------------------
+    def initial_message(self, rng: np.random.Generator) -> str:
+        raise NotImplementedError
+
+    def respond_text(
+        self, turn: int, completion_text: str, rng: np.random.Generator
+    ) -> Tuple[str, float, bool]:
+        """Returns (feedback_text, reward, done); feedback ignored when done."""
+        raise NotImplementedError
+
+    # -- token boundary -------------------------------------------------
+    def _encode(self, text: str) -> List[int]:
+        ids = self.tokenizer.encode(text)
+        return list(ids[-self.max_prompt_tokens:])
+
+    def initial_prompt(self, rng: np.random.Generator) -> List[int]:
+        from ..llm.chat import apply_chat_template
+
+        text = apply_chat_template(
+            self.tokenizer, self.initial_message(rng), system_prompt=self.system_prompt
+        )
+        return self._encode(text)
+
+    def respond(self, turn: int, completion: List[int], rng: np.random.Generator):
+        completion_text = self.tokenizer.decode(completion)
+        feedback, reward, done = self.respond_text(turn, completion_text, rng)
+        if done:
+            return [], reward, True
+        return self._encode(f"\nUser: {feedback}\nAssistant:"), reward, False
+
+
+class SearchQAEnv(TextMultiTurnEnv):
+    """Tool-use QA over an offline corpus (reference search.py flow):
+    turn 0 the model may emit ``<tool>query</tool>`` calls; feedback is
+    the retrieved documents; the final turn is scored by
+    ``<answer>...</answer>`` match against the gold answer, with a small
+    format bonus for issuing a well-formed tool call."""
+
+    def __init__(self, tokenizer, documents, questions, answers,
+                 max_turns: int = 2, tool_k: int = 2, **kw):
+        super().__init__(tokenizer, **kw)
+        from .search import SearchTool
+
+        self.tool = SearchTool(documents)
+        self.questions = list(questions)
+        self.answers = list(answers)
+        self.max_turns = max_turns
+        self.tool_k = tool_k
+        self._answer = None
+
+    def initial_message(self, rng: np.random.Generator) -> str:
+        i = int(rng.integers(0, len(self.questions)))
+        self._answer = str(self.answers[i])
+        return (
+            f"{self.questions[i]}\n"
+            "Use <tool>query</tool> to search, then answer with <answer>...</answer>."
+        )
+
+    def respond_text(self, turn, completion_text, rng):
+        from .search import extract_answer, parse_tool_calls
+
+        final = turn + 1 >= self.max_turns
+        answer = extract_answer(completion_text)
+        if answer is not None or final:
+            correct = answer is not None and self._answer.lower() in answer.lower()
+            return "", (1.0 if correct else 0.0), True
+        calls = parse_tool_calls(completion_text)
+        if not calls:
+            return "No tool call found. Answer with <answer>...</answer>.", 0.0, False
+        docs = self.tool(calls[0], k=self.tool_k)
+        joined = "\n".join(docs) if docs else "(no results)"
+        return f"Search results:\n{joined}", 0.05, False

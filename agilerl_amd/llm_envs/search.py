@@ -64,3 +64,12 @@ class FormatRewardWrapper:
                 return 0.0
             return float(self.reward_fn(completion, answer))
         return float(self.reward_fn(extracted, answer)) + self.format_bonus
+
+
+TOOL_CALL_RE = re.compile(r"<tool>(.*?)</tool>", re.DOTALL)
+
+
+def parse_tool_calls(text: str) -> List[str]:
+    """Extract ``<tool>query</tool>`` invocations from a completion
+    (reference search-env tool-call convention)."""
+    return [m.strip() for m in TOOL_CALL_RE.findall(text)]

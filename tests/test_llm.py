@@ -579,3 +579,84 @@ def test_dpo_packed_matches_padded_loss():
     l2 = a2.learn(dict(exp))
     assert abs(l1["loss"] - l2["loss"]) < 1e-4
     assert abs(l1["margin"] - l2["margin"]) < 1e-3
+
+
+class _WordTok:
+    """Reversible word-level tokenizer for text-env tests (offline)."""
+
+    chat_template = None
+    pad_token_id = 0
+
+    def __init__(self):
+        self.vocab = {}
+        self.words = []
+
+    def encode(self, text):
+        out = []
+        for w in text.split():
+            if w not in self.vocab:
+                self.vocab[w] = len(self.words) + 1
+                self.words.append(w)
+            out.append(self.vocab[w])
+        return out
+
+    def decode(self, ids):
+        return " ".join(
+            self.words[int(i) - 1] if 0 < int(i) <= len(self.words) else "?"
+            for i in ids
+        )
+
+
+class TestTextMultiTurn:
+    def test_tool_call_parsing(self):
+        from agilerl_amd.llm_envs.search import parse_tool_calls
+
+        calls = parse_tool_calls("think <tool>capital of france</tool> more <tool>x</tool>")
+        assert calls == ["capital of france", "x"]
+        assert parse_tool_calls("no calls here") == []
+
+    def test_search_qa_env_flow(self):
+        import numpy as np
+
+        from agilerl_amd.llm_envs import SearchQAEnv
+
+        tok = _WordTok()
+        env = SearchQAEnv(
+            tok,
+            documents=["paris is the capital of france", "berlin is in germany"],
+            questions=["what is the capital of france"],
+            answers=["paris"],
+            max_turns=2,
+        )
+        rng = np.random.default_rng(0)
+        prompt = env.initial_prompt(rng)
+        assert "what is the capital" in tok.decode(prompt)  # chat-template framing around it
+        # turn 0: a tool call gets search results + small format bonus
+        call = tok.encode("<tool>capital france</tool>")
+        feedback, r, done = env.respond(0, call, rng)
+        assert not done and r == 0.05
+        assert "paris" in tok.decode(feedback)
+        # final turn: correct answer scores 1
+        ans = tok.encode("<answer>paris</answer>")
+        _, r, done = env.respond(1, ans, rng)
+        assert done and r == 1.0
+        # wrong answer scores 0
+        env.initial_prompt(rng)
+        _, r, done = env.respond(1, tok.encode("<answer>berlin</answer>"), rng)
+        assert done and r == 0.0
+
+    def test_text_env_through_sync_vec_env(self):
+        import numpy as np
+
+        from agilerl_amd.llm_envs import SearchQAEnv, SyncMultiTurnVecEnv
+
+        tok = _WordTok()
+        # pre-seed vocab so the env's encodes are stable across factories
+        factory = lambda: SearchQAEnv(
+            tok, documents=["paris is the capital of france"],
+            questions=["capital of france ?"], answers=["paris"], max_turns=2,
+        )
+        venv = SyncMultiTurnVecEnv(factory, data_batch_size=2, group_size=2,
+                                   max_turns=2, seed=0)
+        obs = venv.reset()
+        assert obs["input_ids"].shape[0] == 4
