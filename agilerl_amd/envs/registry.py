@@ -14,7 +14,7 @@ from .cartpole import CartPoleVecEnv
 from .lunar_lander import LunarLanderVecEnv
 from .classic_control import AcrobotVecEnv, MountainCarContinuousVecEnv, MountainCarVecEnv
 from .pendulum import PendulumVecEnv
-from .visual import CatchPongVecEnv
+from .visual import BreakoutLiteVecEnv, CatchPongVecEnv
 from .probe import (
     ConstantRewardEnv,
     ConstantRewardContActionsEnv,
@@ -37,6 +37,7 @@ ENV_REGISTRY: Dict[str, Callable[..., VecEnv]] = {
     "MountainCarContinuous-v0": MountainCarContinuousVecEnv,
     "Acrobot-v1": AcrobotVecEnv,
     "CatchPong-v0": CatchPongVecEnv,
+    "BreakoutLite-v0": BreakoutLiteVecEnv,
     "PongLike-v0": CatchPongVecEnv,
     "probe/ConstantReward": ConstantRewardEnv,
     "probe/ObsDependentReward": ObsDependentRewardEnv,

@@ -16,7 +16,7 @@ import numpy as np
 from ..spaces import Box, Discrete
 from .base import BatchedVecEnv
 
-__all__ = ["CatchPongVecEnv"]
+__all__ = ["CatchPongVecEnv", "BreakoutLiteVecEnv"]
 
 
 class CatchPongVecEnv(BatchedVecEnv):
@@ -93,3 +93,103 @@ class CatchPongVecEnv(BatchedVecEnv):
             self._spawn_ball(at_bottom)
         self._render()
         return reward, missed, None
+
+
+class BreakoutLiteVecEnv(BatchedVecEnv):
+    """Breakout-shaped visual env: paddle, ball, 4x8 brick wall; +1 per
+    brick, episode ends on ball drop or wall cleared.  Same (4, 84, 84)
+    uint8 frame-stack contract as CatchPong — a second Atari-like task so
+    Rainbow/visual-PPO coverage is not single-game."""
+
+    max_episode_steps = 2000
+
+    H = W = 84
+    FRAMES = 4
+    PADDLE_W = 14
+    BALL = 3
+    ROWS, COLS = 4, 8
+    BRICK_H = 4
+
+    def __init__(self, num_envs: int = 1, seed: Optional[int] = None):
+        super().__init__(num_envs, seed)
+        self.single_observation_space = Box(
+            0, 255, (self.FRAMES, self.H, self.W), dtype=np.uint8
+        )
+        self.single_action_space = Discrete(3)  # noop / left / right
+        N = self.num_envs
+        self.ball_x = np.zeros(N)
+        self.ball_y = np.zeros(N)
+        self.ball_vx = np.zeros(N)
+        self.ball_vy = np.zeros(N)
+        self.paddle_x = np.zeros(N)
+        self.bricks = np.ones((N, self.ROWS, self.COLS), dtype=bool)
+        self.frames = np.zeros((N, self.FRAMES, self.H, self.W), dtype=np.uint8)
+
+    def _reset_rows(self, mask: np.ndarray) -> None:
+        n = int(mask.sum())
+        self.ball_x[mask] = self.rng.uniform(10, self.W - 10, n)
+        self.ball_y[mask] = self.H / 2
+        self.ball_vx[mask] = self.rng.choice([-1.2, 1.2], n)
+        self.ball_vy[mask] = -1.5
+        self.paddle_x[mask] = self.W / 2
+        self.bricks[mask] = True
+        self.frames[mask] = 0
+        self._render()
+
+    def _obs(self) -> np.ndarray:
+        return self.frames.copy()
+
+    def _brick_cell(self):
+        bw = self.W // self.COLS
+        col = (self.ball_x // bw).astype(int).clip(0, self.COLS - 1)
+        row = ((self.ball_y - 6) // self.BRICK_H).astype(int)
+        return row, col
+
+    def _render(self) -> None:
+        self.frames[:, :-1] = self.frames[:, 1:]
+        frame = np.zeros((self.num_envs, self.H, self.W), dtype=np.uint8)
+        bw = self.W // self.COLS
+        for r in range(self.ROWS):
+            y0 = 6 + r * self.BRICK_H
+            for c in range(self.COLS):
+                live = self.bricks[:, r, c]
+                frame[live, y0 : y0 + self.BRICK_H - 1, c * bw : (c + 1) * bw - 1] = 128
+        bx = self.ball_x.astype(int).clip(0, self.W - self.BALL)
+        by = self.ball_y.astype(int).clip(0, self.H - self.BALL)
+        for i in range(self.num_envs):
+            frame[i, by[i] : by[i] + self.BALL, bx[i] : bx[i] + self.BALL] = 255
+            p = int(np.clip(self.paddle_x[i] - self.PADDLE_W // 2, 0, self.W - self.PADDLE_W))
+            frame[i, self.H - 3 : self.H - 1, p : p + self.PADDLE_W] = 255
+        self.frames[:, -1] = frame
+
+    def _step_all(self, actions):
+        a = np.asarray(actions).reshape(-1)
+        self.paddle_x += np.where(a == 1, -2.5, 0.0) + np.where(a == 2, 2.5, 0.0)
+        self.paddle_x = np.clip(self.paddle_x, self.PADDLE_W / 2, self.W - self.PADDLE_W / 2)
+        self.ball_x += self.ball_vx
+        self.ball_y += self.ball_vy
+        # walls
+        hit_side = (self.ball_x <= 0) | (self.ball_x >= self.W - self.BALL)
+        self.ball_vx = np.where(hit_side, -self.ball_vx, self.ball_vx)
+        self.ball_x = np.clip(self.ball_x, 0, self.W - self.BALL)
+        hit_top = self.ball_y <= 0
+        self.ball_vy = np.where(hit_top, -self.ball_vy, self.ball_vy)
+        self.ball_y = np.clip(self.ball_y, 0, None)
+        # bricks
+        reward = np.zeros(self.num_envs, dtype=np.float32)
+        row, col = self._brick_cell()
+        in_wall = (row >= 0) & (row < self.ROWS) & (self.ball_vy < 0)
+        for i in np.flatnonzero(in_wall):
+            if self.bricks[i, row[i], col[i]]:
+                self.bricks[i, row[i], col[i]] = False
+                self.ball_vy[i] = -self.ball_vy[i]
+                reward[i] = 1.0
+        # paddle bounce
+        at_paddle = self.ball_y >= self.H - 5
+        caught = at_paddle & (np.abs(self.ball_x + self.BALL / 2 - self.paddle_x) <= self.PADDLE_W / 2 + 1)
+        self.ball_vy = np.where(caught & (self.ball_vy > 0), -self.ball_vy, self.ball_vy)
+        dropped = (self.ball_y >= self.H - 2) & ~caught
+        cleared = ~self.bricks.any(axis=(1, 2))
+        terminated = dropped | cleared
+        self._render()
+        return reward, terminated, None

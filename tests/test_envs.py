@@ -133,3 +133,20 @@ class TestClassicControl:
             env = make_vect_envs(env_id, num_envs=2, seed=0)
             obs, _ = env.reset()
             assert obs.shape[0] == 2
+
+
+class TestBreakoutLite:
+    def test_contract_and_brick_rewards(self):
+        from agilerl_amd.envs.visual import BreakoutLiteVecEnv
+
+        env = BreakoutLiteVecEnv(num_envs=4, seed=0)
+        obs, _ = env.reset()
+        assert obs.shape == (4, 4, 84, 84) and obs.dtype == np.uint8
+        total_r = 0.0
+        for _ in range(400):
+            obs, r, term, trunc, _ = env.step(np.random.randint(0, 3, 4))
+            assert (r >= 0).all()
+            total_r += r.sum()
+        # ball physics must hit some bricks within 400 random steps
+        assert total_r > 0
+        assert "BreakoutLite-v0" in __import__("agilerl_amd.envs", fromlist=["ENV_REGISTRY"]).ENV_REGISTRY
