@@ -141,3 +141,33 @@ class TestCQN:
             loss = agent.learn(buf.sample(32))
         assert np.isfinite(loss)
         assert agent.algo == "CQN"
+
+
+def test_rainbow_cnn_on_visual_env():
+    """Rainbow + PER + CNN encoder through a real uint8 visual env
+    (BreakoutLite) — the Atari-like BASELINE config shape on CPU."""
+    from agilerl_amd.algorithms import RainbowDQN
+    from agilerl_amd.components import PrioritizedReplayBuffer
+    from agilerl_amd.envs.visual import BreakoutLiteVecEnv
+
+    torch.manual_seed(0), np.random.seed(0)
+    env = BreakoutLiteVecEnv(num_envs=2, seed=0)
+    agent = RainbowDQN(
+        env.observation_space, env.action_space,
+        net_config={"arch": "cnn", "channel_size": [8, 8],
+                    "kernel_size": [8, 4], "stride_size": [4, 2]},
+        batch_size=8, n_step=3,
+    )
+    buf = PrioritizedReplayBuffer(200, n_step=3)
+    obs, _ = env.reset()
+    for _ in range(12):
+        action = agent.get_action(obs)
+        next_obs, r, term, trunc, _ = env.step(action)
+        buf.add(obs=obs, action=action, reward=r, next_obs=next_obs,
+                done=term.astype(np.float32))
+        obs = next_obs
+    for _ in range(2):
+        batch = buf.sample(8, beta=0.4)
+        loss = agent.learn(batch)
+        buf.update_priorities(batch["idxs"], agent.last_td_errors)
+        assert np.isfinite(loss)
