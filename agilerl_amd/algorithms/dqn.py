@@ -50,6 +50,7 @@ class DQN(RLAlgorithm):
         cudagraphs: bool = False,
         latent_dim: int = 64,
         normalize_images: bool = True,
+        actor_network=None,
         device: str = "cpu",
     ):
         super().__init__(
@@ -73,14 +74,20 @@ class DQN(RLAlgorithm):
         self._graph = None
         self._graph_static = None
 
-        self.actor = QNetwork(
-            observation_space,
-            action_space,
-            encoder_config=net_config,
-            head_config=head_config,
-            latent_dim=latent_dim,
-            device=device,
-        )
+        if actor_network is not None:
+            # user-supplied policy net (reference dqn.py:117 actor_network)
+            from ..networks.base import CustomNetworkAdapter
+
+            self.actor = CustomNetworkAdapter(actor_network, observation_space, device=device)
+        else:
+            self.actor = QNetwork(
+                observation_space,
+                action_space,
+                encoder_config=net_config,
+                head_config=head_config,
+                latent_dim=latent_dim,
+                device=device,
+            )
         self.actor_target = self.actor.clone()
         for p in self.actor_target.parameters():
             p.requires_grad = False

@@ -1,4 +1,5 @@
-from .base import EvolvableNetwork, build_encoder, get_default_encoder_config
+from .base import (EvolvableNetwork, build_encoder, get_default_encoder_config,
+                   preprocess_observation, CustomNetworkAdapter)
 from .distributions import ActionDistribution, EvolvableDistribution
 from .q_networks import QNetwork, RainbowQNetwork, ContinuousQNetwork
 from .actors import DeterministicActor, StochasticActor
@@ -8,6 +9,8 @@ __all__ = [
     "EvolvableNetwork",
     "build_encoder",
     "get_default_encoder_config",
+    "preprocess_observation",
+    "CustomNetworkAdapter",
     "ActionDistribution",
     "EvolvableDistribution",
     "QNetwork",

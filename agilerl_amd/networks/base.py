@@ -24,7 +24,8 @@ from ..modules.multi_input import EvolvableMultiInput
 from ..modules.simba import EvolvableSimBa
 from ..spaces import Box, DictSpace, Discrete, MultiBinary, MultiDiscrete, Space, TupleSpace, flatdim, is_image_space
 
-__all__ = ["EvolvableNetwork", "build_encoder", "get_default_encoder_config"]
+__all__ = ["EvolvableNetwork", "build_encoder", "get_default_encoder_config",
+           "preprocess_observation", "CustomNetworkAdapter"]
 
 
 def get_default_encoder_config(observation_space: Space, simba: bool = False) -> Dict[str, Any]:
@@ -84,6 +85,53 @@ def build_encoder(
     return EvolvableMLP(num_inputs=num_inputs, num_outputs=latent_dim, device=device, **cfg)
 
 
+def preprocess_observation(obs, space: Space, device) -> Any:
+    """Space-aware observation preprocessing (one-hot Discrete/MultiDiscrete,
+    uint8 /255, dict/tuple leaf conversion).  Shared by EvolvableNetwork and
+    CustomNetworkAdapter."""
+    if isinstance(space, (DictSpace, TupleSpace)):
+        dev = device if isinstance(device, torch.device) else torch.device(device)
+
+        def _leaf(x):
+            if not isinstance(x, torch.Tensor):
+                x = torch.as_tensor(np.asarray(x))
+            x = x.to(dev)
+            return x.float() if not x.is_floating_point() else x
+
+        if isinstance(obs, dict):
+            return {k: _leaf(v) for k, v in obs.items()}
+        if isinstance(obs, (tuple, list)):
+            return type(obs)(_leaf(v) for v in obs)
+        return obs
+    if not isinstance(obs, torch.Tensor):
+        obs = torch.as_tensor(np.asarray(obs))
+    obs = obs.to(device if isinstance(device, torch.device) else torch.device(device))
+    if isinstance(space, Discrete):
+        if obs.dim() == 0:
+            obs = obs.unsqueeze(0)
+        if obs.dim() == 1 and obs.dtype in (torch.int32, torch.int64):
+            obs = torch.nn.functional.one_hot(obs.long(), space.n).float()
+        elif obs.shape[-1] != space.n:
+            obs = torch.nn.functional.one_hot(obs.long().reshape(-1), space.n).float()
+    elif isinstance(space, MultiDiscrete):
+        if obs.shape[-1] == len(space.nvec) and obs.dtype in (torch.int32, torch.int64):
+            hots = [
+                torch.nn.functional.one_hot(obs[..., i].long(), int(n)).float()
+                for i, n in enumerate(space.nvec)
+            ]
+            obs = torch.cat(hots, dim=-1)
+    elif isinstance(space, MultiBinary):
+        obs = obs.float()
+    else:
+        if isinstance(space, Box) and space.dtype == np.uint8 and not obs.is_floating_point():
+            obs = obs.float() / 255.0
+        else:
+            obs = obs.float()
+        if obs.dim() == len(space.shape):
+            obs = obs.unsqueeze(0)
+    return obs
+
+
 class EvolvableNetwork(EvolvableModule):
     """Encoder -> latent -> head network with a namespaced mutation surface."""
 
@@ -123,48 +171,7 @@ class EvolvableNetwork(EvolvableModule):
 
     def preprocess(self, obs):
         """Space-aware preprocessing (one-hot for Discrete etc.)."""
-        space = self.observation_space
-        if isinstance(space, (DictSpace, TupleSpace)):
-            dev = self.device if isinstance(self.device, torch.device) else torch.device(self.device)
-
-            def _leaf(x):
-                if not isinstance(x, torch.Tensor):
-                    x = torch.as_tensor(np.asarray(x))
-                x = x.to(dev)
-                return x.float() if not x.is_floating_point() else x
-
-            if isinstance(obs, dict):
-                return {k: _leaf(v) for k, v in obs.items()}
-            if isinstance(obs, (tuple, list)):
-                return type(obs)(_leaf(v) for v in obs)
-            return obs
-        if not isinstance(obs, torch.Tensor):
-            obs = torch.as_tensor(np.asarray(obs))
-        obs = obs.to(self.device if isinstance(self.device, torch.device) else torch.device(self.device))
-        if isinstance(space, Discrete):
-            if obs.dim() == 0:
-                obs = obs.unsqueeze(0)
-            if obs.dim() == 1 and obs.dtype in (torch.int32, torch.int64):
-                obs = torch.nn.functional.one_hot(obs.long(), space.n).float()
-            elif obs.shape[-1] != space.n:
-                obs = torch.nn.functional.one_hot(obs.long().reshape(-1), space.n).float()
-        elif isinstance(space, MultiDiscrete):
-            if obs.shape[-1] == len(space.nvec) and obs.dtype in (torch.int32, torch.int64):
-                hots = [
-                    torch.nn.functional.one_hot(obs[..., i].long(), int(n)).float()
-                    for i, n in enumerate(space.nvec)
-                ]
-                obs = torch.cat(hots, dim=-1)
-        elif isinstance(space, MultiBinary):
-            obs = obs.float()
-        else:
-            if isinstance(space, Box) and space.dtype == np.uint8 and not obs.is_floating_point():
-                obs = obs.float() / 255.0
-            else:
-                obs = obs.float()
-            if obs.dim() == len(space.shape):
-                obs = obs.unsqueeze(0)
-        return obs
+        return preprocess_observation(obs, self.observation_space, self.device)
 
     def forward(self, obs) -> torch.Tensor:
         return self.head_net(self.extract_features(obs))
@@ -287,3 +294,44 @@ class EvolvableNetwork(EvolvableModule):
         base["head_config"] = self._live_head_config()
         base["latent_dim"] = self.latent_dim
         return base
+
+
+class CustomNetworkAdapter(EvolvableModule):
+    """Wraps a user-provided network as an algorithm policy/value net
+    (reference ``actor_network=``/``critic_network=`` constructor support,
+    agilerl/algorithms/dqn.py:117): adds space-aware ``preprocess`` and a
+    ``net.``-namespaced mutation passthrough.  Raw ``nn.Module``s are
+    converted through :func:`wrappers.MakeEvolvable` first."""
+
+    def __init__(self, net, observation_space: Space, device: str = "cpu"):
+        super().__init__(device)
+        if not isinstance(net, EvolvableModule):
+            raise TypeError(
+                "custom networks must be EvolvableModule — wrap plain "
+                "nn.Modules with agilerl_amd.wrappers.MakeEvolvable first"
+            )
+        self.observation_space = observation_space
+        self.net = net.to(device)
+
+    def forward(self, obs) -> torch.Tensor:
+        return self.net(obs)
+
+    def preprocess(self, obs):
+        return preprocess_observation(obs, self.observation_space, self.device)
+
+    @property
+    def mutation_methods(self) -> List[str]:
+        return [f"net.{m}" for m in self.net.mutation_methods]
+
+    def get_mutation_methods(self):
+        return {
+            f"net.{m}": t for m, t in self.net.get_mutation_methods().items()
+        }
+
+    def apply_mutation(self, name: str, **choices):
+        if name.startswith("net."):
+            result = self.net.apply_mutation(name[len("net."):], **choices)
+        else:
+            result = super().apply_mutation(name, **choices)
+        self._last_mutation = (name, result if isinstance(result, dict) else {})
+        return result
