@@ -36,13 +36,14 @@ class CQN(DQN):
         double: bool = True,
         cql_alpha: float = 1.0,
         latent_dim: int = 64,
+        actor_network=None,
         device: str = "cpu",
     ):
         super().__init__(
             observation_space, action_space, index=index, hp_config=hp_config,
             net_config=net_config, head_config=head_config, batch_size=batch_size,
             lr=lr, learn_step=learn_step, gamma=gamma, tau=tau, double=double,
-            latent_dim=latent_dim, device=device,
+            latent_dim=latent_dim, actor_network=actor_network, device=device,
         )
         self.algo = "CQN"
         self.cql_alpha = float(cql_alpha)
