@@ -55,6 +55,7 @@ class DDPG(RLAlgorithm):
         theta: float = 0.15,
         dt: float = 1e-2,
         latent_dim: int = 64,
+        actor_network=None,
         device: str = "cpu",
     ):
         super().__init__(
@@ -77,10 +78,17 @@ class DDPG(RLAlgorithm):
         self._learn_counter = 0
         self._ou_state: Optional[np.ndarray] = None
 
-        self.actor = DeterministicActor(
-            observation_space, action_space, encoder_config=net_config,
-            head_config=head_config, latent_dim=latent_dim, device=device,
-        )
+        if actor_network is not None:
+            # user-supplied policy net (reference ddpg.py actor_network);
+            # must map preprocessed obs -> in-range actions
+            from ..networks.base import CustomNetworkAdapter
+
+            self.actor = CustomNetworkAdapter(actor_network, observation_space, device=device)
+        else:
+            self.actor = DeterministicActor(
+                observation_space, action_space, encoder_config=net_config,
+                head_config=head_config, latent_dim=latent_dim, device=device,
+            )
         self.actor_target = self.actor.clone()
         self.critic = ContinuousQNetwork(
             observation_space, action_space, encoder_config=net_config,
