@@ -43,6 +43,7 @@ class TD3(DDPG):
         O_U_noise: bool = True,
         expl_noise: float = 0.1,
         latent_dim: int = 64,
+        actor_network=None,
         device: str = "cpu",
     ):
         super().__init__(
@@ -50,7 +51,8 @@ class TD3(DDPG):
             net_config=net_config, head_config=head_config, batch_size=batch_size,
             lr_actor=lr_actor, lr_critic=lr_critic, learn_step=learn_step,
             gamma=gamma, tau=tau, policy_freq=policy_freq, O_U_noise=O_U_noise,
-            expl_noise=expl_noise, latent_dim=latent_dim, device=device,
+            expl_noise=expl_noise, latent_dim=latent_dim,
+            actor_network=actor_network, device=device,
         )
         self.algo = "TD3"
         self.policy_noise = policy_noise
