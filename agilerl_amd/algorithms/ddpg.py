@@ -83,7 +83,8 @@ class DDPG(RLAlgorithm):
             # must map preprocessed obs -> in-range actions
             from ..networks.base import CustomNetworkAdapter
 
-            self.actor = CustomNetworkAdapter(actor_network, observation_space, device=device)
+            self.actor = CustomNetworkAdapter(actor_network, observation_space,
+                                              action_space=action_space, device=device)
         else:
             self.actor = DeterministicActor(
                 observation_space, action_space, encoder_config=net_config,

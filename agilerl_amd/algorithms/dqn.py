@@ -78,7 +78,8 @@ class DQN(RLAlgorithm):
             # user-supplied policy net (reference dqn.py:117 actor_network)
             from ..networks.base import CustomNetworkAdapter
 
-            self.actor = CustomNetworkAdapter(actor_network, observation_space, device=device)
+            self.actor = CustomNetworkAdapter(actor_network, observation_space,
+                                              action_space=action_space, device=device)
         else:
             self.actor = QNetwork(
                 observation_space,

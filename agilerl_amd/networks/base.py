@@ -303,7 +303,8 @@ class CustomNetworkAdapter(EvolvableModule):
     ``net.``-namespaced mutation passthrough.  Raw ``nn.Module``s are
     converted through :func:`wrappers.MakeEvolvable` first."""
 
-    def __init__(self, net, observation_space: Space, device: str = "cpu"):
+    def __init__(self, net, observation_space: Space, action_space: Optional[Space] = None,
+                 device: str = "cpu"):
         super().__init__(device)
         if not isinstance(net, EvolvableModule):
             raise TypeError(
@@ -311,7 +312,11 @@ class CustomNetworkAdapter(EvolvableModule):
                 "nn.Modules with agilerl_amd.wrappers.MakeEvolvable first"
             )
         self.observation_space = observation_space
+        self.action_space = action_space
         self.net = net.to(device)
+        if isinstance(action_space, Box):
+            self.register_buffer("action_low", torch.as_tensor(action_space.low, dtype=torch.float32).to(device))
+            self.register_buffer("action_high", torch.as_tensor(action_space.high, dtype=torch.float32).to(device))
 
     def forward(self, obs) -> torch.Tensor:
         return self.net(obs)
