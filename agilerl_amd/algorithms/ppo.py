@@ -61,6 +61,7 @@ class PPO(RLAlgorithm):
         latent_dim: int = 64,
         recurrent: bool = False,
         share_encoders: bool = False,
+        actor_network=None,
         device: str = "cpu",
     ):
         super().__init__(
@@ -94,15 +95,24 @@ class PPO(RLAlgorithm):
             net_config.setdefault("hidden_state_size", 64)
             self.net_config = net_config
 
-        self.actor = StochasticActor(
-            observation_space,
-            action_space,
-            encoder_config=net_config,
-            head_config=head_config,
-            latent_dim=latent_dim,
-            log_std_init=log_std_init,
-            device=device,
-        )
+        if actor_network is not None:
+            # user-supplied policy net (reference ppo.py actor_network):
+            # maps preprocessed obs -> distribution head outputs
+            from ..networks.base import CustomStochasticAdapter
+
+            self.actor = CustomStochasticAdapter(
+                actor_network, observation_space, action_space, device=device
+            )
+        else:
+            self.actor = StochasticActor(
+                observation_space,
+                action_space,
+                encoder_config=net_config,
+                head_config=head_config,
+                latent_dim=latent_dim,
+                log_std_init=log_std_init,
+                device=device,
+            )
         self.critic = ValueNetwork(
             observation_space,
             encoder_config=net_config,
@@ -110,6 +120,8 @@ class PPO(RLAlgorithm):
             latent_dim=latent_dim,
             device=device,
         )
+        if actor_network is not None and share_encoders:
+            raise ValueError("share_encoders is not supported with a custom actor_network")
         self.share_encoders = bool(share_encoders)
         if self.share_encoders:
             self.critic.encoder = self.actor.encoder  # one encoder, two heads

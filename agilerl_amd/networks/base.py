@@ -25,7 +25,7 @@ from ..modules.simba import EvolvableSimBa
 from ..spaces import Box, DictSpace, Discrete, MultiBinary, MultiDiscrete, Space, TupleSpace, flatdim, is_image_space
 
 __all__ = ["EvolvableNetwork", "build_encoder", "get_default_encoder_config",
-           "preprocess_observation", "CustomNetworkAdapter"]
+           "preprocess_observation", "CustomNetworkAdapter", "CustomStochasticAdapter"]
 
 
 def get_default_encoder_config(observation_space: Space, simba: bool = False) -> Dict[str, Any]:
@@ -340,3 +340,26 @@ class CustomNetworkAdapter(EvolvableModule):
             result = super().apply_mutation(name, **choices)
         self._last_mutation = (name, result if isinstance(result, dict) else {})
         return result
+
+
+class CustomStochasticAdapter(CustomNetworkAdapter):
+    """Custom net as a stochastic policy (reference PPO ``actor_network=``):
+    the user net maps preprocessed obs -> distribution head outputs
+    (logits for Discrete/MultiDiscrete/MultiBinary, means for Box); an
+    :class:`networks.distributions.ActionDistribution` sits on top."""
+
+    def __init__(self, net, observation_space: Space, action_space: Space,
+                 device: str = "cpu"):
+        super().__init__(net, observation_space, action_space=action_space, device=device)
+        from .distributions import ActionDistribution
+
+        self.dist_layer = ActionDistribution(action_space).to(device)
+
+    def sample(self, obs, action_mask=None):
+        return self.dist_layer.sample(self.net(obs), action_mask)
+
+    def evaluate_actions(self, obs, actions, action_mask=None):
+        return self.dist_layer.log_prob_entropy(self.net(obs), actions, action_mask)
+
+    def deterministic_action(self, obs, action_mask=None):
+        return self.dist_layer.mode(self.net(obs), action_mask)

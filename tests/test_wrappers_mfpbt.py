@@ -243,3 +243,42 @@ class TestCustomActorNetwork:
             n_a = sum(p.numel() for p in agent.actor.parameters())
             n_t = sum(p.numel() for p in agent.actor_target.parameters())
             assert n_a == n_t, method
+
+    def test_ppo_custom_actor_collect_learn_mutate(self):
+        from agilerl_amd.algorithms import PPO
+        from agilerl_amd.components import RolloutBuffer
+        from agilerl_amd.envs import CartPoleVecEnv
+        from agilerl_amd.rollouts.on_policy import collect_rollouts
+
+        torch.manual_seed(0), np.random.seed(0)
+        env = CartPoleVecEnv(4, seed=0)
+        net = torch.nn.Sequential(torch.nn.Linear(4, 16), torch.nn.ReLU(),
+                                  torch.nn.Linear(16, 2))
+        agent = PPO(env.observation_space, env.action_space,
+                    actor_network=MakeEvolvable(net, torch.randn(1, 4)),
+                    learn_step=8, batch_size=16)
+        buf = RolloutBuffer(8, 4, gamma=agent.gamma, gae_lambda=agent.gae_lambda)
+        collect_rollouts(agent, env, buf, 8)
+        assert np.isfinite(agent.learn(buf)["policy_loss"])
+        for method in [m for m in agent.mutation_methods if m.startswith("net.")]:
+            agent.apply_architecture_mutation(method)
+        collect_rollouts(agent, env, buf, 8)
+        assert np.isfinite(agent.learn(buf)["policy_loss"])
+        with pytest.raises(ValueError, match="share_encoders"):
+            PPO(env.observation_space, env.action_space, share_encoders=True,
+                actor_network=MakeEvolvable(net, torch.randn(1, 4)))
+
+    def test_td3_custom_actor_learns(self):
+        from agilerl_amd.algorithms import TD3
+        from agilerl_amd.spaces import Box
+
+        torch.manual_seed(0), np.random.seed(0)
+        net = torch.nn.Sequential(torch.nn.Linear(4, 16), torch.nn.ReLU(),
+                                  torch.nn.Linear(16, 2), torch.nn.Tanh())
+        agent = TD3(Box(-1.0, 1.0, (4,)), Box(-1.0, 1.0, (2,)),
+                    actor_network=MakeEvolvable(net, torch.randn(1, 4)),
+                    batch_size=16)
+        batch = {"obs": torch.randn(16, 4), "action": torch.rand(16, 2) * 2 - 1,
+                 "reward": torch.randn(16), "next_obs": torch.randn(16, 4),
+                 "done": torch.zeros(16)}
+        assert np.isfinite(agent.learn(dict(batch)))
