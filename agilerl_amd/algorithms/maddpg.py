@@ -59,6 +59,7 @@ class MADDPG(MultiAgentRLAlgorithm):
         expl_noise: float = 0.1,
         latent_dim: int = 64,
         shared_agent_groups: Optional[List[List[str]]] = None,
+        actor_networks: Optional[Dict[str, Any]] = None,
         device: str = "cpu",
     ):
         super().__init__(
@@ -92,11 +93,21 @@ class MADDPG(MultiAgentRLAlgorithm):
         for aid in self.agent_ids:
             leader = group_of.get(aid, aid)
             if leader not in actor_modules:
-                actor_modules[leader] = DeterministicActor(
-                    self.observation_spaces[leader], self.action_spaces[leader],
-                    encoder_config=net_config, head_config=head_config,
-                    latent_dim=latent_dim, device=device,
-                )
+                if actor_networks is not None and leader in actor_networks:
+                    # user-supplied per-agent policy net (reference
+                    # maddpg actor_networks constructor arg)
+                    from ..networks.base import CustomNetworkAdapter
+
+                    actor_modules[leader] = CustomNetworkAdapter(
+                        actor_networks[leader], self.observation_spaces[leader],
+                        action_space=self.action_spaces[leader], device=device,
+                    )
+                else:
+                    actor_modules[leader] = DeterministicActor(
+                        self.observation_spaces[leader], self.action_spaces[leader],
+                        encoder_config=net_config, head_config=head_config,
+                        latent_dim=latent_dim, device=device,
+                    )
             actors[aid] = actor_modules[leader]
         self.actors = ModuleDict(actors, device=device)
         self.actor_targets = self.actors.clone()
