@@ -41,6 +41,7 @@ class MATD3(MADDPG):
         policy_freq: int = 2,
         expl_noise: float = 0.1,
         latent_dim: int = 64,
+        actor_networks: Optional[Dict[str, Any]] = None,
         device: str = "cpu",
     ):
         super().__init__(
@@ -48,7 +49,7 @@ class MATD3(MADDPG):
             hp_config=hp_config, net_config=net_config, head_config=head_config,
             batch_size=batch_size, lr_actor=lr_actor, lr_critic=lr_critic,
             learn_step=learn_step, gamma=gamma, tau=tau, expl_noise=expl_noise,
-            latent_dim=latent_dim, device=device,
+            latent_dim=latent_dim, actor_networks=actor_networks, device=device,
         )
         self.algo = "MATD3"
         self.policy_freq = int(policy_freq)
