@@ -57,6 +57,7 @@ class IPPO(MultiAgentRLAlgorithm):
         max_grad_norm: float = 0.5,
         latent_dim: int = 64,
         shared_agent_groups: Optional[List[List[str]]] = None,
+        actor_networks: Optional[Dict[str, Any]] = None,
         device: str = "cpu",
     ):
         super().__init__(
@@ -87,11 +88,19 @@ class IPPO(MultiAgentRLAlgorithm):
         for aid in self.agent_ids:
             leader = group_of.get(aid, aid)
             if leader not in actor_mods:
-                actor_mods[leader] = StochasticActor(
-                    self.observation_spaces[leader], self.action_spaces[leader],
-                    encoder_config=net_config, head_config=head_config,
-                    latent_dim=latent_dim, device=device,
-                )
+                if actor_networks is not None and leader in actor_networks:
+                    from ..networks.base import CustomStochasticAdapter
+
+                    actor_mods[leader] = CustomStochasticAdapter(
+                        actor_networks[leader], self.observation_spaces[leader],
+                        self.action_spaces[leader], device=device,
+                    )
+                else:
+                    actor_mods[leader] = StochasticActor(
+                        self.observation_spaces[leader], self.action_spaces[leader],
+                        encoder_config=net_config, head_config=head_config,
+                        latent_dim=latent_dim, device=device,
+                    )
                 critic_mods[leader] = ValueNetwork(
                     self.observation_spaces[leader], encoder_config=net_config,
                     head_config=head_config, latent_dim=latent_dim, device=device,
