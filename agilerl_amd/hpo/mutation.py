@@ -76,6 +76,12 @@ class Mutations:
                 else:
                     self.rl_hyperparam_mutation(agent)
             except Exception as e:  # mutation must never kill training
+                import warnings
+
+                warnings.warn(
+                    f"mutation failed on agent {getattr(agent, 'index', '?')}: "
+                    f"{type(e).__name__}: {e}", RuntimeWarning,
+                )
                 agent.mut = f"Failed({type(e).__name__})"
         return population
 
