@@ -20,6 +20,24 @@ from ..population import Population
 __all__ = ["train_off_policy"]
 
 
+def _merge_final_obs(next_obs, final_obs, mask):
+    """Bootstrap targets use the pre-reset observation for done rows;
+    handles Dict/Tuple observation structures."""
+    if isinstance(next_obs, dict):
+        out = {k: v.copy() for k, v in next_obs.items()}
+        for k in out:
+            out[k][mask] = final_obs[k][mask]
+        return out
+    if isinstance(next_obs, (tuple, list)):
+        out = [v.copy() for v in next_obs]
+        for o, f in zip(out, final_obs):
+            o[mask] = f[mask]
+        return type(next_obs)(out)
+    out = next_obs.copy()
+    out[mask] = final_obs[mask]
+    return out
+
+
 def train_off_policy(
     env,
     env_name: str,
@@ -78,8 +96,7 @@ def train_off_policy(
                 store_next = next_obs
                 done_any = term | trunc
                 if done_any.any() and "final_observation" in info:
-                    store_next = next_obs.copy()
-                    store_next[done_any] = info["final_observation"][done_any]
+                    store_next = _merge_final_obs(next_obs, info["final_observation"], done_any)
                 memory.add(
                     obs=obs,
                     action=action,

@@ -246,3 +246,21 @@ def test_ppo_action_mask_respected_and_stored():
         assert (action == 1).all()
     det = agent.get_action(obs, action_mask=mask, training=False)
     assert (det == 1).all()
+
+
+def test_dict_obs_through_train_off_policy():
+    """The full off-policy loop (auto-reset bootstrap merge included)
+    handles Dict observation envs."""
+    from agilerl_amd.components import ReplayBuffer
+    from agilerl_amd.training import train_off_policy
+
+    torch.manual_seed(0), np.random.seed(0)
+    env = DictObsEnv(num_envs=4, seed=0)
+    agent = DQN(env.observation_space, env.action_space,
+                net_config={"arch": "multi_input", "hidden_size": [16]},
+                batch_size=32, lr=1e-2)
+    agents, hist = train_off_policy(
+        env, "dictobs", "DQN", [agent], ReplayBuffer(500),
+        max_steps=400, evo_steps=200, eval_loop=1, verbose=False,
+    )
+    assert np.isfinite(agents[0].fitness[-1])
