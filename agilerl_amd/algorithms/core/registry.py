@@ -8,7 +8,7 @@ MutationRegistry :440.
 from __future__ import annotations
 
 from dataclasses import dataclass, field
-from typing import Any, Callable, Dict, List, Optional, Type
+from typing import Any, Dict, List, Optional
 
 import numpy as np
 

@@ -8,9 +8,8 @@ re-implementation on this framework's :class:`EvolvableGPT`.
 
 from __future__ import annotations
 
-from typing import Dict, Optional, Tuple
+from typing import Dict
 
-import numpy as np
 import torch
 import torch.nn as nn
 import torch.nn.functional as F

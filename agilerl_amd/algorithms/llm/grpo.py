@@ -16,15 +16,15 @@ Data flow per learn() call (one prompt batch of B*G completions):
 
 from __future__ import annotations
 
-from typing import Any, Dict, List, Optional
+from typing import Any, Dict, Optional
 
 import numpy as np
 import torch
 
 from ... import ops
 from ...ops.grpo_loss import grpo_policy_loss
-from ..core.registry import HyperparameterConfig, RLParameter
-from .base import LLMAlgorithm, default_hp_config
+from ..core.registry import HyperparameterConfig
+from .base import LLMAlgorithm
 
 __all__ = ["GRPO"]
 

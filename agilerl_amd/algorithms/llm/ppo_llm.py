@@ -14,9 +14,6 @@ import numpy as np
 import torch
 import torch.nn as nn
 
-from ...ops.grpo_loss import grpo_policy_loss
-from ..core.registry import HyperparameterConfig
-from .base import LLMAlgorithm
 from .grpo import GRPO
 
 __all__ = ["PPOLLM"]

@@ -15,7 +15,7 @@ import torch
 import torch.nn.functional as F
 
 from ..networks.base import EvolvableNetwork
-from ..spaces import Box, Discrete, Space
+from ..spaces import Discrete, Space
 from .core.base import RLAlgorithm
 from .core.optimizer_wrapper import OptimizerWrapper
 from .core.registry import HyperparameterConfig, NetworkGroup, OptimizerConfig, RLParameter

@@ -8,7 +8,7 @@ HIP kernels via ``agilerl_amd.ops``.
 
 from __future__ import annotations
 
-from typing import Any, Dict, Optional, Tuple
+from typing import Any, Dict, Optional
 
 import numpy as np
 import torch
@@ -17,7 +17,7 @@ import torch.nn as nn
 from ..components.rollout_buffer import RolloutBuffer
 from ..networks.actors import StochasticActor
 from ..networks.value_networks import ValueNetwork
-from ..spaces import Box, Space
+from ..spaces import Space
 from .core.base import RLAlgorithm
 from .core.optimizer_wrapper import OptimizerWrapper
 from .core.registry import HyperparameterConfig, NetworkGroup, OptimizerConfig, RLParameter

@@ -7,17 +7,15 @@ from __future__ import annotations
 
 from typing import Any, Dict, Optional
 
-import numpy as np
 import torch
 import torch.nn.functional as F
 
 from .. import ops
-from ..networks.actors import DeterministicActor
 from ..networks.q_networks import ContinuousQNetwork
 from ..spaces import Box, Space
 from .core.registry import HyperparameterConfig, NetworkGroup, OptimizerConfig
 from .core.optimizer_wrapper import OptimizerWrapper
-from .ddpg import DDPG, default_hp_config
+from .ddpg import DDPG
 
 __all__ = ["TD3"]
 

@@ -19,7 +19,7 @@ import numpy as np
 import torch
 
 from .. import ops
-from .data import to_tensor, tree_index, tree_map
+from .data import to_tensor, tree_map
 
 __all__ = ["RolloutBuffer"]
 

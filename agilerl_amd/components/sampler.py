@@ -5,10 +5,9 @@ Reference parity: ``agilerl/components/sampler.py:21``.
 
 from __future__ import annotations
 
-from typing import Any, Dict, Optional
+from typing import Optional
 
 from .replay_buffer import MultiStepReplayBuffer, PrioritizedReplayBuffer, ReplayBuffer
-from .rollout_buffer import RolloutBuffer
 
 __all__ = ["Sampler"]
 

@@ -7,7 +7,7 @@ All N instances integrate in one vectorized step.
 
 from __future__ import annotations
 
-from typing import Optional, Tuple
+from typing import Optional
 
 import numpy as np
 

@@ -7,11 +7,11 @@ centralized-critic learning).  Batched dict API like the MPE envs.
 
 from __future__ import annotations
 
-from typing import Dict, List, Optional
+from typing import Optional
 
 import numpy as np
 
-from ..spaces import Box, Discrete, Space
+from ..spaces import Box, Discrete
 from .mpe import MultiAgentVecEnv
 
 __all__ = ["ConstantRewardMAEnv", "FixedObsPolicyMAEnv", "JointActionMAEnv"]

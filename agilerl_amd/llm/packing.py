@@ -9,7 +9,7 @@ chunked GEMMs see dense rows (SURVEY §2.8 long-context strategy).
 
 from __future__ import annotations
 
-from typing import Dict, Tuple
+from typing import Dict
 
 import torch
 

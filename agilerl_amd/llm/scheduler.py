@@ -7,7 +7,6 @@ Reference parity: ``agilerl/utils/algo_utils.py``
 from __future__ import annotations
 
 import math
-from typing import Optional
 
 import torch
 

@@ -10,7 +10,7 @@ GRPO-family ``learn()`` consumes.
 
 from __future__ import annotations
 
-from typing import Any, Callable, Dict, List, Optional
+from typing import Dict
 
 import numpy as np
 import torch

@@ -9,7 +9,7 @@ structure (e.g. ``<answer>...</answer>`` tags).
 from __future__ import annotations
 
 import re
-from typing import Callable, Dict, List, Optional, Sequence
+from typing import Callable, List, Optional, Sequence
 
 import numpy as np
 

@@ -10,7 +10,6 @@ import csv
 import os
 import warnings
 import sys
-import time
 from typing import Any, Dict, List, Optional
 
 __all__ = ["Logger", "StdOutLogger", "CSVLogger", "TensorboardLogger", "WandbLogger", "PrometheusLogger", "make_loggers"]

@@ -11,7 +11,7 @@ reference path; the op layer dispatches to the CDNA4 kernel on gfx950.
 from __future__ import annotations
 
 import math
-from typing import Callable, Optional
+from typing import Optional
 
 import torch
 import torch.nn as nn

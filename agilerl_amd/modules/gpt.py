@@ -9,8 +9,7 @@ preservation.
 
 from __future__ import annotations
 
-import math
-from typing import Optional, Tuple
+from typing import Optional
 
 import numpy as np
 import torch

@@ -13,7 +13,7 @@ from typing import Dict, Optional, Union
 import torch
 import torch.nn as nn
 
-from ..spaces import Box, DictSpace, Space, TupleSpace, flatdim, is_image_space
+from ..spaces import DictSpace, Space, TupleSpace, flatdim, is_image_space
 from .base import EvolvableModule, MutationType, mutation, preserve_parameters
 from .cnn import EvolvableCNN
 from .mlp import EvolvableMLP

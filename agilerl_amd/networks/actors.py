@@ -9,10 +9,9 @@ from __future__ import annotations
 
 from typing import Any, Dict, Optional, Tuple
 
-import numpy as np
 import torch
 
-from ..spaces import Box, Discrete, MultiBinary, MultiDiscrete, Space, flatdim
+from ..spaces import Box, Discrete, Space, flatdim
 from .base import EvolvableNetwork
 from .distributions import ActionDistribution
 

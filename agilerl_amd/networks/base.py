@@ -11,7 +11,7 @@ every network of the same group (target nets, shared encoders).
 
 from __future__ import annotations
 
-from typing import Any, Dict, List, Optional, Union
+from typing import Any, Dict, List, Optional
 
 import numpy as np
 import torch

@@ -19,7 +19,7 @@ Workspace is bounded to (chunk_rows, V) regardless of sequence length.
 
 from __future__ import annotations
 
-from typing import Optional, Tuple
+from typing import Optional
 
 import torch
 

@@ -10,7 +10,7 @@ logprobs is a single multiply.
 
 from __future__ import annotations
 
-from typing import Optional, Tuple
+from typing import Optional
 
 import torch
 

@@ -13,7 +13,7 @@ before ``optimizer.step()``.
 
 from __future__ import annotations
 
-from typing import List, Optional
+from typing import List
 
 import torch
 import torch.distributed as dist

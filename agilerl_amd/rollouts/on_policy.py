@@ -10,7 +10,7 @@ buffer).  Truncation bootstrapping uses ``info["final_observation"]``.
 
 from __future__ import annotations
 
-from typing import Any, Dict, Optional, Tuple
+from typing import Dict, Optional, Tuple
 
 import numpy as np
 import torch

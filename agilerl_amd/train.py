@@ -6,7 +6,6 @@ Reference parity: ``agilerl/train.py:30-60``.
 from __future__ import annotations
 
 import argparse
-import sys
 
 import torch
 

@@ -9,7 +9,6 @@ import time as _time
 from typing import Dict, List, Optional
 
 import numpy as np
-import torch
 
 from ..components.rollout_buffer import RolloutBuffer
 from ..hpo.mutation import Mutations

@@ -7,10 +7,9 @@ Reference parity: ``agilerl/training/train_off_policy.py:121``
 from __future__ import annotations
 
 import os
-from typing import Any, Dict, List, Optional
+from typing import Dict, List, Optional
 
 import numpy as np
-import torch
 
 from ..components.replay_buffer import PrioritizedReplayBuffer, ReplayBuffer
 from ..hpo.mutation import Mutations

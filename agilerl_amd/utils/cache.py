@@ -7,7 +7,7 @@ ILQL/BC_LM generation).  Per-layer preallocated K/V rings sized to
 
 from __future__ import annotations
 
-from typing import List, Optional, Tuple
+from typing import List, Tuple
 
 import torch
 

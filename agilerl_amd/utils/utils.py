@@ -13,7 +13,7 @@ from typing import Any, Dict, List, Optional, Tuple
 import numpy as np
 
 from ..envs.registry import make_vect_envs  # noqa: F401 (reference location parity)
-from ..models.manifest import ALGO_REGISTRY, algo_workload, resolve_algo_class
+from ..models.manifest import algo_workload, resolve_algo_class
 from ..training.train_off_policy import save_population_checkpoint  # noqa: F401
 
 __all__ = [

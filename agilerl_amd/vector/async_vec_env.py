@@ -16,7 +16,7 @@ from __future__ import annotations
 
 import multiprocessing as mp
 import traceback
-from typing import Any, Callable, Dict, List, Optional, Tuple
+from typing import Any, Callable, Dict, List, Optional
 
 import numpy as np
 

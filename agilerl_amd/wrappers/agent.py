@@ -7,7 +7,7 @@ AsyncAgentsWrapper :631 (masks inactive agents in async multi-agent envs).
 
 from __future__ import annotations
 
-from typing import Any, Dict, Optional
+from typing import Any, Dict
 
 import numpy as np
 import torch

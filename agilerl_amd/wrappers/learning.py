@@ -6,9 +6,7 @@ reward shaping wrapper), BanditEnv :66 (re-exported from envs.bandit).
 
 from __future__ import annotations
 
-from typing import Optional
 
-import numpy as np
 
 from ..envs.bandit import BanditEnv  # noqa: F401  (reference places it here)
 

@@ -10,12 +10,12 @@ back to :class:`EvolvableWrapper` (no architecture mutations).
 
 from __future__ import annotations
 
-from typing import List, Optional
+from typing import List
 
 import torch
 import torch.nn as nn
 
-from ..modules.base import EvolvableWrapper, preserve_parameters
+from ..modules.base import EvolvableWrapper
 from ..modules.mlp import EvolvableMLP
 
 __all__ = ["MakeEvolvable"]
