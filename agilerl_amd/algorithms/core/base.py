@@ -327,6 +327,13 @@ class EvolvableAlgorithm(metaclass=AlgorithmMeta):
         init_args["device"] = device
         agent = algo_cls(**init_args)
         agent._apply_checkpoint(ckpt)
+        wrapper = ckpt.get("wrapper")
+        if wrapper and wrapper.get("cls") == "RSNorm":
+            from ...wrappers.agent import RSNorm
+
+            wrapped = RSNorm(agent)
+            wrapped.load_wrapper_state(wrapper)
+            return wrapped
         return agent
 
     # ------------------------------------------------------------------

@@ -119,6 +119,37 @@ class RSNorm(AgentWrapper):
                     experiences[key] = self._norm(experiences[key])
         return self.agent.learn(experiences, *args, **kwargs)
 
+    # -- persistence: normalizer stats ride inside the checkpoint --------
+    def wrapper_state(self) -> dict:
+        return {
+            "cls": type(self).__name__,
+            "mean": self.rms.mean.copy(),
+            "var": self.rms.var.copy(),
+            "count": float(self.rms.count),
+            "clip": self.clip,
+        }
+
+    def load_wrapper_state(self, state: dict) -> None:
+        self.rms.mean[...] = state["mean"]
+        self.rms.var[...] = state["var"]
+        self.rms.count = state["count"]
+        object.__setattr__(self, "clip", state.get("clip", self.clip))
+
+    def save_checkpoint(self, path: str) -> None:
+        import dill
+
+        ckpt = self.agent.get_checkpoint_dict()
+        ckpt["wrapper"] = self.wrapper_state()
+        torch.save(ckpt, path, pickle_module=dill)
+
+    def load_checkpoint(self, path: str) -> None:
+        import dill
+
+        ckpt = torch.load(path, pickle_module=dill, weights_only=False, map_location="cpu")
+        self.agent._apply_checkpoint(ckpt)
+        if "wrapper" in ckpt:
+            self.load_wrapper_state(ckpt["wrapper"])
+
     def test(self, env, *args, **kwargs):
         # evaluation uses frozen stats via a normalized-view env
         wrapper_self = self

@@ -343,3 +343,24 @@ class TestCustomActorNetwork:
                 for aid in env.agents}
         _collect_ma_rollout(agent, env, bufs, 8, None)
         assert np.isfinite(agent.learn(bufs)["policy_loss"])
+
+    def test_rsnorm_stats_survive_checkpoint(self, tmp_path):
+        from agilerl_amd.algorithms import DQN
+        from agilerl_amd.algorithms.core.base import EvolvableAlgorithm
+        from agilerl_amd.spaces import Box, Discrete
+
+        torch.manual_seed(0), np.random.seed(0)
+        agent = RSNorm(DQN(Box(-1.0, 1.0, (4,)), Discrete(2),
+                           net_config={"arch": "mlp", "hidden_size": [8]}))
+        agent.get_action(np.random.randn(64, 4).astype(np.float32) * 5 + 3)
+        path = str(tmp_path / "a.pt")
+        agent.save_checkpoint(path)
+        back = EvolvableAlgorithm.load(path)
+        assert isinstance(back, RSNorm)
+        np.testing.assert_allclose(back.rms.mean, agent.rms.mean)
+        np.testing.assert_allclose(back.rms.var, agent.rms.var)
+        # normalized predictions identical
+        obs = np.random.randn(3, 4).astype(np.float32) * 5 + 3
+        np.testing.assert_array_equal(
+            agent.get_action(obs, training=False), back.get_action(obs, training=False)
+        )
