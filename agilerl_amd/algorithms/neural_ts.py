@@ -10,14 +10,41 @@ from __future__ import annotations
 import numpy as np
 import torch
 
+from typing import Any, Dict, Optional
+
+from ..spaces import Discrete, Space
+from .core.registry import HyperparameterConfig
 from .neural_ucb import NeuralUCB
 
 __all__ = ["NeuralTS"]
 
 
 class NeuralTS(NeuralUCB):
-    def __init__(self, *args, **kwargs):
-        super().__init__(*args, **kwargs)
+    # explicit signature (not *args) so AlgorithmMeta captures init args —
+    # clone()/checkpoint/evolution rebuild from them
+    def __init__(
+        self,
+        observation_space: Space,
+        action_space: Discrete,
+        index: int = 0,
+        hp_config: Optional[HyperparameterConfig] = None,
+        net_config: Optional[Dict[str, Any]] = None,
+        head_config: Optional[Dict[str, Any]] = None,
+        batch_size: int = 64,
+        lr: float = 1e-3,
+        learn_step: int = 2,
+        gamma: float = 1.0,
+        lamb: float = 1.0,
+        reg: float = 0.000625,
+        latent_dim: int = 64,
+        device: str = "cpu",
+    ):
+        super().__init__(
+            observation_space, action_space, index=index, hp_config=hp_config,
+            net_config=net_config, head_config=head_config, batch_size=batch_size,
+            lr=lr, learn_step=learn_step, gamma=gamma, lamb=lamb, reg=reg,
+            latent_dim=latent_dim, device=device,
+        )
         self.algo = "NeuralTS"
 
     def get_action(self, context, training: bool = True, **kwargs) -> int:

@@ -192,3 +192,28 @@ def test_maddpg_shared_groups_stay_shared_after_mutation():
         agent.apply_architecture_mutation(method)
         assert agent.actors["a_0"] is agent.actors["a_1"], method
         assert agent.critics["a_0"] is agent.critics["a_1"], method
+
+
+@pytest.mark.parametrize("cls_name", ["NeuralUCB", "NeuralTS"])
+def test_bandit_mutation_sweep(cls_name):
+    """Neural bandits: every mutation keeps the confidence machinery and
+    learn path functional (reference mutation.py:1196 grad reinit)."""
+    import agilerl_amd.algorithms as algos
+
+    cls = getattr(algos, cls_name)
+    ctx_dim = 8
+    base = cls(Box(-1.0, 1.0, (ctx_dim,)), Discrete(4),
+               net_config={"arch": "mlp", "hidden_size": [16]})
+    for method in base.mutation_methods:
+        agent = cls(Box(-1.0, 1.0, (ctx_dim,)), Discrete(4),
+                    net_config={"arch": "mlp", "hidden_size": [16]})
+        agent.apply_architecture_mutation(method)
+        context = np.random.randn(4, ctx_dim).astype(np.float32)
+        arm = agent.get_action(context)
+        assert 0 <= arm < 4
+        batch = {"obs": torch.randn(16, ctx_dim), "reward": torch.rand(16)}
+        loss = agent.learn(batch)
+        assert np.isfinite(loss), method
+        clone = agent.clone(index=3)
+        x = torch.randn(2, ctx_dim)
+        torch.testing.assert_close(agent.actor(x), clone.actor(x))
