@@ -150,8 +150,8 @@ class TestGrpoLoss:
         mask = torch.tensor([[1.0, 1, 1, 1], [1, 0, 0, 0]])
         l_tok = grpo_policy_loss(logp, old, adv, mask, loss_norm="token")
         l_seq = grpo_policy_loss(logp, old, adv, mask, loss_norm="sequence")
-        assert float(l_tok) == pytest.approx(-1.0)
-        assert float(l_seq) == pytest.approx(-1.0)
+        assert float(l_tok.detach()) == pytest.approx(-1.0)
+        assert float(l_seq.detach()) == pytest.approx(-1.0)
 
 
 class TestGRPOAgent:
