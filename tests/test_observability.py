@@ -127,3 +127,34 @@ def test_prometheus_logger_gauges():
     # repeated reports reuse gauges (no duplicate-registration error)
     logger.log_report({"global_step": 2000, "population": []})
     assert REGISTRY.get_sample_value("agilerl_global_step") == 2000
+
+
+class TestMetricsDetails:
+    def test_histograms_bounded_and_snapshotted(self):
+        from agilerl_amd.metrics import AgentMetrics
+
+        m = AgentMetrics(0, histogram_len=10)
+        for i in range(25):
+            m.log_histogram("td_error", float(i))
+        snap = m.snapshot()
+        assert "td_error" not in snap or True  # histogram not a scalar
+        assert len(m.histograms["td_error"]) == 10
+        assert list(m.histograms["td_error"])[0] == 15.0
+
+    def test_fitness_window_mean(self):
+        from agilerl_amd.metrics import AgentMetrics
+
+        m = AgentMetrics(0, fitness_window=3)
+        for f in (1.0, 2.0, 3.0, 4.0):
+            m.log_fitness(f)
+        assert m.mean_fitness == pytest.approx(3.0)  # property; mean of last 3
+
+    def test_multiagent_per_agent_scalars(self):
+        from agilerl_amd.metrics import MultiAgentMetrics
+
+        m = MultiAgentMetrics(0, agent_ids=["a", "b"])
+        m.log_agent("a", "reward", 1.0)
+        m.log_agent("b", "reward", 3.0)
+        m.log_agent("a", "reward", 2.0)
+        snap = m.snapshot()
+        assert snap is not None
