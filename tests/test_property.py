@@ -122,3 +122,36 @@ def test_c51_projection_is_distribution(batch, atoms, gamma, seed):
     assert out.shape == (batch, atoms)
     assert (out >= -1e-6).all()
     torch.testing.assert_close(out.sum(-1), torch.ones(batch), atol=1e-4, rtol=1e-4)
+
+
+@settings(**COMMON)
+@given(seed=st.integers(0, 10_000))
+def test_space_samples_in_bounds(seed):
+    from agilerl_amd.spaces import (Box, Discrete, MultiBinary, MultiDiscrete,
+                                    DictSpace, TupleSpace, flatdim)
+
+    rng = np.random.default_rng(seed)
+    lo, hi = sorted(rng.uniform(-5, 5, 2))
+    spaces = [
+        Box(lo, hi + 1e-3, (3,)),
+        Discrete(int(rng.integers(2, 10))),
+        MultiDiscrete([2, 3, 4]),
+        MultiBinary(5),
+    ]
+    for sp in spaces:
+        s = sp.sample()
+        if isinstance(sp, Box):
+            assert (np.asarray(s) >= sp.low - 1e-6).all()
+            assert (np.asarray(s) <= sp.high + 1e-6).all()
+        elif isinstance(sp, Discrete):
+            assert 0 <= s < sp.n
+        elif isinstance(sp, MultiDiscrete):
+            assert all(0 <= v < n for v, n in zip(s, sp.nvec))
+        else:
+            assert set(np.asarray(s).ravel()) <= {0, 1}
+        assert flatdim(sp) > 0
+    d = DictSpace({"a": spaces[0], "b": spaces[1]})
+    t = TupleSpace((spaces[0], spaces[3]))
+    assert flatdim(d) == flatdim(spaces[0]) + flatdim(spaces[1])
+    assert set(d.sample()) == {"a", "b"}
+    assert len(t.sample()) == 2
