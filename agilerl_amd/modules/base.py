@@ -282,6 +282,13 @@ class ModuleDict(EvolvableModule):
         first = next(iter(self._modules_dict.values()), None)
         return first.mutation_methods if first is not None else []
 
+    def get_mutation_methods(self) -> Dict[str, MutationType]:
+        # delegate to the first member (all members expose the same surface);
+        # without this override the base-class type() lookup fails and
+        # hpo.Mutations silently skips architecture mutations for MA agents
+        first = next(iter(self._modules_dict.values()), None)
+        return first.get_mutation_methods() if first is not None else {}
+
     def apply_mutation(self, name: str, **choices) -> Optional[dict]:
         out: Optional[dict] = None
         seen = set()  # grouped agents may share one module object
