@@ -125,7 +125,8 @@ class TestMAProbes:
         joint_obs = torch.zeros(1, 2)
         both_zero = torch.tensor([[1.0, 0.0, 1.0, 0.0]])
         mixed = torch.tensor([[1.0, 0.0, 0.0, 1.0]])
-        q_good = float(agent.critics["agent_0"](joint_obs, both_zero))
-        q_bad = float(agent.critics["agent_0"](joint_obs, mixed))
+        with torch.no_grad():
+            q_good = float(agent.critics["agent_0"](joint_obs, both_zero))
+            q_bad = float(agent.critics["agent_0"](joint_obs, mixed))
         assert q_good > 0.5
         assert q_bad < 0.0
