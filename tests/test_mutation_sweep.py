@@ -219,17 +219,19 @@ def test_bandit_mutation_sweep(cls_name):
         torch.testing.assert_close(agent.actor(x), clone.actor(x))
 
 
-def test_hpo_mutations_actually_apply_to_multiagent():
+@pytest.mark.parametrize("cls_name", ["MADDPG", "MATD3", "IPPO"])
+def test_hpo_mutations_actually_apply_to_multiagent(cls_name):
     """Regression: hpo.Mutations must reach multi-agent ModuleDicts — the
     type-level lookup in get_mutation_methods used to fail and silently
     skip every architecture mutation (surfaced as Failed(AttributeError))."""
-    from agilerl_amd.algorithms import MADDPG
+    import agilerl_amd.algorithms as algos
     from agilerl_amd.envs.mpe import SpeakerListenerVecEnv
     from agilerl_amd.hpo import Mutations
 
+    cls = getattr(algos, cls_name)
     torch.manual_seed(0), np.random.seed(0)
     env = SpeakerListenerVecEnv(num_envs=2, seed=0)
-    pop = MADDPG.population(4, env.observation_spaces, env.action_spaces,
+    pop = cls.population(4, env.observation_spaces, env.action_spaces,
                             agent_ids=env.agents,
                             net_config={"arch": "mlp", "hidden_size": [16]})
     muts = Mutations(no_mutation=0.0, architecture=1.0, parameters=0.0,
