@@ -246,3 +246,19 @@ def test_engine_cancel():
     assert s1 not in results and s2 in results
     assert engine.cache.free_pages == 32
     assert not engine.cancel(999)
+
+
+def test_eos_stops_sequence_early():
+    model = tiny_model()
+    torch.manual_seed(4)
+    prompt = torch.randint(1, 64, (5,))
+    # discover the greedy first token, then make it the EOS
+    probe = DecodeEngine(model, num_pages=32, page_size=4)
+    sid = probe.submit(prompt, max_new_tokens=1)
+    first = int(probe.run_all()[sid][-1])
+
+    engine = DecodeEngine(model, num_pages=32, page_size=4, eos_token_id=first)
+    sid = engine.submit(prompt, max_new_tokens=8)
+    out = engine.run_all()[sid]
+    assert out.numel() == prompt.numel() + 1  # stopped at EOS, not max tokens
+    assert int(out[-1]) == first
