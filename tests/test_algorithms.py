@@ -3,7 +3,6 @@
 import os
 
 import numpy as np
-import pytest
 import torch
 
 from agilerl_amd.algorithms.dqn import DQN

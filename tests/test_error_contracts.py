@@ -1,10 +1,9 @@
 """Error contracts: invalid configurations fail fast with clear messages."""
 
-import numpy as np
 import pytest
 import torch
 
-from agilerl_amd.spaces import Box, Discrete
+from agilerl_amd.spaces import Box
 
 NET = {"arch": "mlp", "hidden_size": [16]}
 

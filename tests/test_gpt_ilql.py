@@ -1,7 +1,6 @@
 """EvolvableGPT / EvolvableBERT / ILQL / BC_LM tests."""
 
 import numpy as np
-import pytest
 import torch
 
 from agilerl_amd.algorithms.bc_lm import BC_LM

@@ -1,7 +1,6 @@
 """Tournament + mutation engine tests."""
 
 import numpy as np
-import pytest
 import torch
 
 from agilerl_amd.algorithms.dqn import DQN

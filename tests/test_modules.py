@@ -1,7 +1,6 @@
 """Module-layer tests: mutations, parameter preservation, cloning."""
 
 import numpy as np
-import pytest
 import torch
 
 from agilerl_amd.modules import (

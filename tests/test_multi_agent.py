@@ -1,7 +1,6 @@
 """Multi-agent algorithm + env tests."""
 
 import numpy as np
-import pytest
 import torch
 
 from agilerl_amd.algorithms import IPPO, MADDPG, MATD3

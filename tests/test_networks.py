@@ -1,7 +1,5 @@
 """Network-layer tests."""
 
-import numpy as np
-import pytest
 import torch
 
 from agilerl_amd.networks import (

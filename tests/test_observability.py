@@ -3,7 +3,6 @@
 import csv
 import os
 
-import numpy as np
 import pytest
 import torch
 

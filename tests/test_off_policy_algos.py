@@ -1,7 +1,6 @@
 """Rainbow / DDPG / TD3 / CQN tests (probe convergence + mechanics)."""
 
 import numpy as np
-import pytest
 import torch
 
 from agilerl_amd.algorithms import CQN, DDPG, TD3, RainbowDQN

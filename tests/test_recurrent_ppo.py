@@ -1,7 +1,6 @@
 """Recurrent PPO (LSTM + BPTT) tests."""
 
 import numpy as np
-import pytest
 import torch
 
 from agilerl_amd.algorithms.ppo import PPO

@@ -1,7 +1,6 @@
 """First-party space primitive tests."""
 
 import numpy as np
-import pytest
 
 from agilerl_amd.spaces import (
     Box,

@@ -1,7 +1,5 @@
 """KV-cache and sampling-utility tests (legacy offline-RL support stack)."""
 
-import numpy as np
-import pytest
 import torch
 
 from agilerl_amd.utils.cache import Cache

@@ -9,7 +9,7 @@ from agilerl_amd.envs import CartPoleVecEnv
 from agilerl_amd.envs.mpe import SpeakerListenerVecEnv
 from agilerl_amd.hpo import MultiFrequencySelection, Mutations, TournamentSelection
 from agilerl_amd.spaces import Box, Discrete
-from agilerl_amd.wrappers import AgentWrapper, MakeEvolvable, RSNorm, Skill
+from agilerl_amd.wrappers import MakeEvolvable, RSNorm, Skill
 
 
 class TestRSNorm:
