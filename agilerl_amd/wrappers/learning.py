@@ -6,8 +6,6 @@ reward shaping wrapper), BanditEnv :66 (re-exported from envs.bandit).
 
 from __future__ import annotations
 
-
-
 from ..envs.bandit import BanditEnv  # noqa: F401  (reference places it here)
 
 __all__ = ["Skill", "BanditEnv"]
