@@ -122,6 +122,8 @@ class PPO(RLAlgorithm):
         )
         if actor_network is not None and share_encoders:
             raise ValueError("share_encoders is not supported with a custom actor_network")
+        if actor_network is not None and recurrent:
+            raise ValueError("recurrent=True is not supported with a custom actor_network")
         self.share_encoders = bool(share_encoders)
         if self.share_encoders:
             self.critic.encoder = self.actor.encoder  # one encoder, two heads
