@@ -172,10 +172,12 @@ def create_app(agent, lock: Optional[threading.Lock] = None):
             raise HTTPException(status_code=400, detail="prompt, messages or input_ids required")
         ids = ids.to(getattr(a, "device", "cpu"))
         mask = torch.ones_like(ids)
+        gen = a.generate_paged if getattr(a, "generation", "hf") == "paged" and \
+            hasattr(a, "generate_paged") else a.generate
         with lock, torch.no_grad():
             state["requests"] += 1
-            out = a.generate(ids, mask, max_new_tokens=req.max_new_tokens,
-                             do_sample=req.do_sample, temperature=req.temperature)
+            out = gen(ids, mask, max_new_tokens=req.max_new_tokens,
+                      do_sample=req.do_sample, temperature=req.temperature)
         completion = out[:, ids.shape[1]:]
         resp = {"output_ids": out.cpu().tolist(),
                 "completion_ids": completion.cpu().tolist()}

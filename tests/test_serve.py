@@ -109,3 +109,22 @@ def test_serve_llm_generate_with_messages():
     body = r.json()
     assert len(body["completion_ids"][0]) == 3
     assert isinstance(body["completion"][0], str)
+
+
+def test_serve_generate_uses_paged_engine_when_configured():
+    import torch
+
+    from agilerl_amd.algorithms.llm.grpo import GRPO
+
+    tiny = dict(model_type="llama", vocab_size=64, hidden_size=32,
+                intermediate_size=64, num_hidden_layers=1,
+                num_attention_heads=2, num_key_value_heads=1,
+                max_position_embeddings=128, pad_token_id=0)
+    agent = GRPO(model_config=tiny, dtype=torch.float32, lora_config={"r": 2},
+                 max_completion_tokens=4, generation="paged")
+    client = TestClient(create_app(agent))
+    r = client.post("/generate", json={"input_ids": [1, 2, 3], "max_new_tokens": 3,
+                                       "do_sample": False})
+    assert r.status_code == 200
+    assert len(r.json()["completion_ids"][0]) == 3
+    assert agent._decode_engine is not None  # the paged engine actually ran
