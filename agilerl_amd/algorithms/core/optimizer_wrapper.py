@@ -53,6 +53,7 @@ class OptimizerWrapper:
         """(Re)build the underlying optimizer(s) over the given networks."""
         if lr is not None:
             self.lr = lr
+        self._networks = list(networks)
         if self.multiagent and len(networks) >= 1 and isinstance(networks[0], ModuleDict):
             keys = list(networks[0].keys())
             self.optimizer = {}
@@ -75,6 +76,15 @@ class OptimizerWrapper:
             self.optimizer.zero_grad(set_to_none=set_to_none)
 
     def step(self) -> None:
+        # Data-parallel correctness: when wrap_models attached a GradBucketer
+        # (parallel/ddp.py), its async bucket allreduces must be drained and
+        # scattered back BEFORE the optimizer consumes the grads — RL nets
+        # rarely reach the auto-flush threshold, so without this the ranks
+        # would silently diverge.
+        for net in getattr(self, "_networks", []):
+            bucketer = getattr(net, "_grad_bucketer", None)
+            if bucketer is not None:
+                bucketer.finalize()
         if isinstance(self.optimizer, dict):
             for opt in self.optimizer.values():
                 opt.step()

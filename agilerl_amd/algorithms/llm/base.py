@@ -443,6 +443,11 @@ class LLMAlgorithm(EvolvableAlgorithm):
         import copy as _copy
 
         clone = _copy.copy(self)  # shallow: shares the base model
+        # The decode engine is bound to the PARENT's _activate (its
+        # set_adapter_fn) — sharing it would make the clone's paged
+        # generation sample under the parent's LoRA adapter.  Drop it so the
+        # clone lazily rebuilds an engine bound to its own _activate.
+        clone.__dict__.pop("_decode_engine", None)
         clone.registry = self.registry
         new_index = self.index if index is None else index
         clone.index = new_index

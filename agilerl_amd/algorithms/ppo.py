@@ -380,6 +380,8 @@ class PPO(RLAlgorithm):
         if self._learn_static is not None and self._learn_static["obs"].shape[0] != B:
             self._clear_learn_graph()
         if self._learn_graph is None:
+            prev_opt = self.optimizer.optimizer
+            prev_state = dict(prev_opt.state) if prev_opt is not None else {}
             self.optimizer.optimizer = torch.optim.Adam(
                 [p for net in (self.actor, self.critic) for p in net.parameters()
                  if p.requires_grad],
@@ -413,9 +415,21 @@ class PPO(RLAlgorithm):
             with torch.no_grad():
                 for p, sv in zip(nets, saved):
                     p.copy_(sv)
-                for group_state in self.optimizer.optimizer.state.values():
-                    for v in group_state.values():
-                        if torch.is_tensor(v):
+                # restore optimizer moments: carry over any pre-existing Adam
+                # state (checkpoint-loaded or accumulated eagerly) into the
+                # capturable optimizer's state tensors; params without prior
+                # state start from zeroed moments
+                for p, group_state in self.optimizer.optimizer.state.items():
+                    prior = prev_state.get(p)
+                    for k, v in group_state.items():
+                        if not torch.is_tensor(v):
+                            continue
+                        pv = prior.get(k) if prior else None
+                        if pv is not None and torch.is_tensor(pv) and pv.shape == v.shape:
+                            v.copy_(pv.to(v.device, v.dtype))
+                        elif pv is not None and k == "step" and v.numel() == 1:
+                            v.fill_(float(pv))
+                        else:
                             v.zero_()
             torch.cuda.synchronize()
         st = self._learn_static

@@ -115,16 +115,16 @@ class RolloutBuffer:
     # ------------------------------------------------------------------
     @torch.no_grad()
     def compute_returns_and_advantages(self, last_value, last_done=None) -> None:
+        # ``last_done`` is accepted for backward compatibility but unused:
+        # dones[t] = done-after-step-t already cuts the final step's bootstrap.
         T = len(self)
         sd = self._storage
         last_value = to_tensor(last_value).to(self.device).float().reshape(-1)
-        if last_done is not None:
-            last_done = to_tensor(last_done).to(self.device).float().reshape(-1)
         values = sd["value"][:T].reshape(T, self.num_envs)
         rewards = sd["reward"][:T].reshape(T, self.num_envs)
         dones = sd["done"][:T].reshape(T, self.num_envs).float()
         adv, ret = ops.gae_scan(
-            rewards, values, dones, last_value, self.gamma, self.gae_lambda, last_done
+            rewards, values, dones, last_value, self.gamma, self.gae_lambda
         )
         self.advantages = adv
         self.returns = ret

@@ -14,7 +14,9 @@ Kernel inventory (reference call sites in SURVEY.md §2.9):
 - :func:`masked_mean` — token-masked mean reduction.
 - :func:`fused_linear_logprobs` — chunked lm_head logprob (fwd+bwd).
 - :func:`grpo_token_loss` — fused token-masked GRPO/CISPO surrogate.
-- :func:`segtree_*` — GPU sum/min segment tree for PER.
+
+PER's GPU segment tree lives in :mod:`agilerl_amd.components.segment_tree`
+(``segtree_update`` / ``segtree_sample`` kernels in ``csrc/rl_ops.hip``).
 """
 
 from __future__ import annotations
@@ -126,12 +128,14 @@ def gae_scan(
     last_value: torch.Tensor,
     gamma: float,
     gae_lambda: float,
-    last_done: Optional[torch.Tensor] = None,
 ) -> Tuple[torch.Tensor, torch.Tensor]:
     """Generalized advantage estimation over a (T, N) rollout.
 
-    ``dones[t]`` marks episode termination *after* step ``t`` (so the value
-    bootstrap through t+1 is cut).  Returns ``(advantages, returns)`` with
+    ``dones[t]`` marks episode termination *after* step ``t`` — the
+    convention every collector in this package stores — so step ``t``'s
+    value bootstrap through ``V(s_{t+1})`` AND its lambda carry from
+    ``adv[t+1]`` are both cut by ``dones[t]``: a terminal step's advantage
+    is exactly ``r_t - V(s_t)``.  Returns ``(advantages, returns)`` with
     ``returns = advantages + values``.
 
     GPU path: one HIP kernel, one wavefront-lane per env column, the T-loop
@@ -140,14 +144,11 @@ def gae_scan(
     """
     ext = extension()
     if use_hip(rewards) and ext is not None and hasattr(ext, "gae_scan"):
-        if last_done is None:
-            last_done = torch.zeros_like(last_value)
         adv = ext.gae_scan(
             rewards.float().contiguous(),
             values.float().contiguous(),
             dones.float().contiguous(),
             last_value.float().contiguous(),
-            last_done.float().contiguous(),
             float(gamma),
             float(gae_lambda),
         )
@@ -160,15 +161,12 @@ def gae_scan(
     adv = torch.zeros_like(rewards)
     next_adv = torch.zeros_like(last_value.float())
     next_value = last_value.float()
-    next_not_done = (
-        1.0 - last_done.float() if last_done is not None else torch.ones_like(next_value)
-    )
     for t in range(T - 1, -1, -1):
-        delta = rewards[t] + gamma * next_value * next_not_done - values[t]
-        next_adv = delta + gamma * gae_lambda * next_not_done * next_adv
+        nd = not_done[t]
+        delta = rewards[t] + gamma * next_value * nd - values[t]
+        next_adv = delta + gamma * gae_lambda * nd * next_adv
         adv[t] = next_adv
         next_value = values[t]
-        next_not_done = not_done[t]
     return adv, adv + values
 
 

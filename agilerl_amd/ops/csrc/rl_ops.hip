@@ -30,33 +30,34 @@ static inline int grid_1d(long n, int block) {
 // ---------------------------------------------------------------------------
 // GAE reverse scan: one lane per env column, T-loop in-register.
 // rewards/values/dones: (T, N) row-major; advantages out: (T, N).
+// dones[t] marks termination AFTER step t (the collectors' convention), so
+// step t's bootstrap through V(s_{t+1}) and its lambda carry are both cut by
+// dones[t] itself.
 // ---------------------------------------------------------------------------
 __global__ void gae_scan_kernel(
     const float* __restrict__ rewards,
     const float* __restrict__ values,
     const float* __restrict__ dones,
     const float* __restrict__ last_value,
-    const float* __restrict__ last_done,
     float* __restrict__ adv,
     int T, int N, float gamma, float lam) {
   int n = blockIdx.x * blockDim.x + threadIdx.x;
   if (n >= N) return;
   float next_adv = 0.f;
   float next_val = last_value[n];
-  float next_nd = 1.f - last_done[n];
   for (int t = T - 1; t >= 0; --t) {
     long i = (long)t * N + n;
-    float delta = rewards[i] + gamma * next_val * next_nd - values[i];
-    next_adv = delta + gamma * lam * next_nd * next_adv;
+    float nd = 1.f - dones[i];
+    float delta = rewards[i] + gamma * next_val * nd - values[i];
+    next_adv = delta + gamma * lam * nd * next_adv;
     adv[i] = next_adv;
     next_val = values[i];
-    next_nd = 1.f - dones[i];
   }
 }
 
 torch::Tensor gae_scan(
     torch::Tensor rewards, torch::Tensor values, torch::Tensor dones,
-    torch::Tensor last_value, torch::Tensor last_done,
+    torch::Tensor last_value,
     double gamma, double lam) {
   CHECK_GPU(rewards); CHECK_CONTIG(rewards);
   int T = rewards.size(0), N = rewards.size(1);
@@ -66,7 +67,7 @@ torch::Tensor gae_scan(
   hipStream_t stream = at::hip::getCurrentHIPStream();
   hipLaunchKernelGGL(gae_scan_kernel, dim3(grid), dim3(block), 0, stream,
       rewards.data_ptr<float>(), values.data_ptr<float>(), dones.data_ptr<float>(),
-      last_value.data_ptr<float>(), last_done.data_ptr<float>(),
+      last_value.data_ptr<float>(),
       adv.data_ptr<float>(), T, N, (float)gamma, (float)lam);
   return adv;
 }

@@ -136,7 +136,7 @@ class GraphedPPOCollector:
         last_value = agent.critic(agent.critic.preprocess(self.obs)).squeeze(-1)
         adv, ret = ops.gae_scan(
             self.storage["reward"], self.storage["value"], self.storage["done"],
-            last_value, agent.gamma, agent.gae_lambda, self.done,
+            last_value, agent.gamma, agent.gae_lambda,
         )
         T, N = self.storage["reward"].shape
         # clone: the storage is reused by the next collect; returning views

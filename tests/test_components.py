@@ -149,18 +149,41 @@ class TestGAE:
         last_value = torch.randn(N)
         gamma, lam = 0.99, 0.95
         adv, ret = ops.gae_scan(rewards, values, dones, last_value, gamma, lam)
-        # manual per-env python scan
+        # manual per-env python scan: dones[t] (done-after-step-t) cuts both
+        # step t's bootstrap and its lambda carry
         for n in range(N):
-            next_adv, next_val, next_nd = 0.0, last_value[n].item(), 1.0
+            next_adv, next_val = 0.0, last_value[n].item()
             expected = np.zeros(T)
             for t in reversed(range(T)):
                 nd = 1.0 - dones[t, n].item()
-                delta = rewards[t, n].item() + gamma * next_val * next_nd - values[t, n].item()
-                next_adv = delta + gamma * lam * next_nd * next_adv
+                delta = rewards[t, n].item() + gamma * next_val * nd - values[t, n].item()
+                next_adv = delta + gamma * lam * nd * next_adv
                 expected[t] = next_adv
                 next_val = values[t, n].item()
-                next_nd = nd
             np.testing.assert_allclose(adv[:, n].numpy(), expected, rtol=1e-4, atol=1e-5)
+
+    def test_terminal_step_advantage_is_r_minus_v(self):
+        """A mid-rollout terminal step must not bootstrap the next episode's
+        value: adv[t_term] == r[t_term] - V(s_t_term) exactly."""
+        T, N = 8, 2
+        torch.manual_seed(1)
+        rewards = torch.randn(T, N)
+        values = torch.randn(T, N)
+        dones = torch.zeros(T, N)
+        dones[3, 0] = 1.0
+        dones[5, 1] = 1.0
+        last_value = torch.randn(N)
+        adv, _ = ops.gae_scan(rewards, values, dones, last_value, 0.99, 0.95)
+        assert adv[3, 0].item() == pytest.approx(
+            (rewards[3, 0] - values[3, 0]).item(), abs=1e-6
+        )
+        assert adv[5, 1].item() == pytest.approx(
+            (rewards[5, 1] - values[5, 1]).item(), abs=1e-6
+        )
+        # and the step BEFORE the terminal still carries through it normally
+        delta2 = rewards[2, 0] + 0.99 * values[3, 0] - values[2, 0]
+        expect2 = delta2 + 0.99 * 0.95 * adv[3, 0]
+        assert adv[2, 0].item() == pytest.approx(expect2.item(), abs=1e-5)
 
     def test_rollout_buffer_end_to_end(self):
         buf = RolloutBuffer(capacity=8, num_envs=4, gamma=0.99, gae_lambda=0.95)

@@ -262,3 +262,30 @@ def test_eos_stops_sequence_early():
     out = engine.run_all()[sid]
     assert out.numel() == prompt.numel() + 1  # stopped at EOS, not max tokens
     assert int(out[-1]) == first
+
+
+class TestCloneDecodeEngineIsolation:
+    def test_clone_does_not_share_parent_engine(self):
+        """copy.copy shared _decode_engine whose set_adapter_fn was
+        bound to the PARENT's _activate — a clone generating paged would
+        sample under the parent's adapter.  The clone must drop the engine
+        and lazily rebuild one bound to its own _activate."""
+        from agilerl_amd.algorithms.llm.grpo import GRPO
+
+        tiny = dict(model_type="llama", vocab_size=64, hidden_size=32,
+                    intermediate_size=64, num_hidden_layers=1,
+                    num_attention_heads=2, num_key_value_heads=1,
+                    max_position_embeddings=128, pad_token_id=0)
+        torch.manual_seed(0)
+        parent = GRPO(model_config=tiny, dtype=torch.float32, lora_config={"r": 2},
+                      max_completion_tokens=4)
+        ids = torch.randint(1, 64, (2, 5))
+        mask = torch.ones_like(ids)
+        parent.generate_paged(ids, mask, do_sample=False)
+        assert parent._decode_engine is not None
+        clone = parent.clone(index=7)
+        assert getattr(clone, "_decode_engine", None) is None
+        # clone's lazily-built engine must activate the CLONE's adapter
+        clone.generate_paged(ids, mask, do_sample=False)
+        assert clone._decode_engine is not parent._decode_engine
+        assert clone._decode_engine.set_adapter_fn.__self__ is clone

@@ -356,3 +356,57 @@ def test_metric_aggregation_and_data_sharding():
     assert results["note_0"] == "x"
     assert set(results["prompts_0"]).isdisjoint(results["prompts_1"])
     assert len(results["prompts_0"]) + len(results["prompts_1"]) == 10
+
+
+def _wrap_models_worker(rank, world_size, port):
+    os.environ["RANK"] = str(rank)
+    os.environ["LOCAL_RANK"] = str(rank)
+    os.environ["WORLD_SIZE"] = str(world_size)
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+
+    from agilerl_amd.algorithms.dqn import DQN
+    from agilerl_amd.parallel import DistributedState, broadcast_module
+    from agilerl_amd.spaces import Box, Discrete
+
+    DistributedState.reset()
+    DistributedState.get()
+
+    torch.manual_seed(0)
+    agent = DQN(Box(-1, 1, (4,)), Discrete(2), batch_size=8)
+    broadcast_module(agent.actor, src=0)
+    broadcast_module(agent.actor_target, src=0)
+    agent.wrap_models()
+
+    # DIFFERENT experiences per rank: without gradient sync the ranks diverge
+    g = torch.Generator().manual_seed(rank + 1)
+    batch = {
+        "obs": torch.randn(8, 4, generator=g),
+        "action": torch.randint(0, 2, (8, 1), generator=g),
+        "reward": torch.randn(8, 1, generator=g),
+        "next_obs": torch.randn(8, 4, generator=g),
+        "done": torch.zeros(8, 1),
+    }
+    for _ in range(3):
+        agent.learn(batch)
+
+    # parameters must be bitwise-identical across ranks after synced steps
+    for p in agent.actor.parameters():
+        gather = [torch.zeros_like(p.data) for _ in range(world_size)]
+        torch.distributed.all_gather(gather, p.data)
+        assert torch.allclose(gather[0], gather[1], atol=1e-7), "ranks diverged"
+    torch.distributed.destroy_process_group()
+
+
+def test_wrap_models_learn_syncs_gradients_gloo():
+    """wrap_models attached bucketer hooks but no RL learn path
+    drained them before optimizer.step() — ranks silently diverged.  The
+    OptimizerWrapper now finalizes bucketers inside step()."""
+    port = _find_free_port()
+    ctx = mp.get_context("spawn")
+    procs = [ctx.Process(target=_wrap_models_worker, args=(r, 2, port)) for r in range(2)]
+    for p in procs:
+        p.start()
+    for p in procs:
+        p.join(timeout=180)
+        assert p.exitcode == 0

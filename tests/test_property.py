@@ -59,16 +59,17 @@ def test_gae_scan_matches_naive_recursion(t, n, gamma, lam, seed):
     dones = (torch.rand(t, n, generator=g) < 0.2).float()
     last_value = torch.randn(n, generator=g)
     adv, ret = ops.gae_scan(rewards, values, dones, last_value, gamma, lam)
-    # naive per-env python recursion
+    # naive per-env python recursion (dones[i] = done-after-step-i cuts both
+    # the bootstrap and the lambda carry of step i)
     for env in range(n):
-        next_adv, next_val, next_nd = 0.0, float(last_value[env]), 1.0
+        next_adv, next_val = 0.0, float(last_value[env])
         expect = np.zeros(t)
         for i in range(t - 1, -1, -1):
-            delta = float(rewards[i, env]) + gamma * next_val * next_nd - float(values[i, env])
-            next_adv = delta + gamma * lam * next_nd * next_adv
+            nd = 1.0 - float(dones[i, env])
+            delta = float(rewards[i, env]) + gamma * next_val * nd - float(values[i, env])
+            next_adv = delta + gamma * lam * nd * next_adv
             expect[i] = next_adv
             next_val = float(values[i, env])
-            next_nd = 1.0 - float(dones[i, env])
         np.testing.assert_allclose(adv[:, env].numpy(), expect, rtol=1e-4, atol=1e-4)
     torch.testing.assert_close(ret, adv + values)
 
