@@ -190,19 +190,32 @@ class DistributedPopulation:
             self.local_indices = list(range(self.pop_size))
             return
 
-        # Cross-rank: broadcast each *distinct* parent checkpoint once from
-        # its owner, then every rank materializes its own slots.
+        # Cross-rank: broadcast each *distinct* parent once from its owner.
+        # Weights/optimizer moments travel as flat per-dtype tensor
+        # broadcasts (RCCL over xGMI, no pickle round-trip); only the tiny
+        # architecture/attribute skeleton is pickled (flat_transfer.py —
+        # replaces the reference's full-dict broadcast_object_list,
+        # hpo/tournament.py:179).
+        from .flat_transfer import broadcast_checkpoint
+
         needed_parents = sorted(set(plan))
         parent_ckpts: Dict[int, dict] = {}
         for parent in needed_parents:
             src = self.owner(parent)
-            if self.state.rank == src:
-                ckpt = self.agents[parent].get_checkpoint_dict()
-                obj = [ckpt]
-            else:
-                obj = [None]
-            dist.broadcast_object_list(obj, src=src)
-            parent_ckpts[parent] = obj[0]
+            # the plan is identical on every rank, so all ranks agree on
+            # which parents actually cross rank boundaries — purely-local
+            # clones skip the collective entirely
+            if all(self.owner(s) == src for s, p in enumerate(plan) if p == parent):
+                continue
+            ckpt = (
+                self.agents[parent].get_checkpoint_dict()
+                if self.state.rank == src
+                else None
+            )
+            parent_ckpts[parent] = broadcast_checkpoint(
+                ckpt, src, self.state.rank,
+                device=self.state.device, backend=self.state.backend,
+            )
 
         new_agents: Dict[int, object] = {}
         for slot, parent in enumerate(plan):

@@ -95,10 +95,13 @@ class BenchRunner:
             update_epochs=RLParameter(min=1, max=4, dtype=int),
         )
 
+        probe_env = LunarLanderVecEnv(1)
+        obs_space, act_space = probe_env.single_observation_space, probe_env.single_action_space
+
         def factory(index: int) -> PPO:
             return PPO(
-                observation_space=LunarLanderVecEnv(1).single_observation_space,
-                action_space=LunarLanderVecEnv(1).single_action_space,
+                observation_space=obs_space,
+                action_space=act_space,
                 index=index,
                 hp_config=hp,
                 learn_step=args.learn_step,
