@@ -224,17 +224,30 @@ class PrioritizedReplayBuffer(ReplayBuffer):
         self.min_tree.update(idx, prio)
 
     def sample(self, batch_size: int, beta: float = 0.4):
-        total = self.sum_tree.sum(0, self._size)
-        device = self.sum_tree.device
-        # stratified prefix sampling
-        bounds = torch.linspace(0, total, batch_size + 1, device=device)
-        u = bounds[:-1] + torch.rand(batch_size, device=device) * (bounds[1:] - bounds[:-1])
-        idx = self.sum_tree.retrieve(u).clamp_(max=self._size - 1)
+        from ..ops.backend import extension, use_hip
 
-        p = self.sum_tree.get(idx) / max(total, 1e-12)
-        p_min = self.min_tree.min() / max(total, 1e-12)
-        max_weight = (p_min * self._size) ** (-beta) if p_min > 0 else 1.0
-        weights = ((p * self._size).clamp(min=1e-12) ** (-beta)) / max_weight
+        device = self.sum_tree.device
+        ext = extension()
+        if use_hip(self.sum_tree.tree) and ext is not None and hasattr(ext, "per_sample"):
+            # fused HIP path: stratified prefixes + LDS descent + IS weights
+            # in ONE launch; total and p_min are read from the tree roots on
+            # device (zero host syncs in the priority-sample math)
+            rand01 = torch.rand(batch_size, device=device)
+            idx, weights = ext.per_sample(
+                self.sum_tree.tree, self.min_tree.tree, rand01,
+                int(self._size), float(beta),
+            )
+        else:
+            total = self.sum_tree.sum(0, self._size)
+            # stratified prefix sampling
+            bounds = torch.linspace(0, total, batch_size + 1, device=device)
+            u = bounds[:-1] + torch.rand(batch_size, device=device) * (bounds[1:] - bounds[:-1])
+            idx = self.sum_tree.retrieve(u).clamp_(max=self._size - 1)
+
+            p = self.sum_tree.get(idx) / max(total, 1e-12)
+            p_min = self.min_tree.min() / max(total, 1e-12)
+            max_weight = (p_min * self._size) ** (-beta) if p_min > 0 else 1.0
+            weights = ((p * self._size).clamp(min=1e-12) ** (-beta)) / max_weight
 
         if self.n_step > 1:
             batch = self._nstep_gather(idx.cpu())

@@ -16,11 +16,16 @@ from typing import Optional
 
 import torch
 
+from ..ops.backend import extension, use_hip
+
 __all__ = ["SumSegmentTree", "MinSegmentTree"]
+
+_OP_SUM, _OP_MIN = 0, 1
 
 
 class _SegmentTree:
     neutral: float = 0.0
+    _hip_op: int = _OP_SUM
 
     def __init__(self, capacity: int, device: str = "cpu"):
         # round up to a power of two for a perfect binary tree
@@ -38,9 +43,19 @@ class _SegmentTree:
         """Batched leaf update + path propagation to the root.
 
         Duplicate indices take the last value (scatter semantics).
+
+        GPU path: one HIP launch (``segtree_update``) — scatter + per-level
+        barriered propagation inside a single workgroup (csrc/rl_ops.hip).
         """
-        idx = idx.to(self.device).long() + self.capacity
+        idx = idx.to(self.device).long()
         values = values.to(self.device).float()
+        ext = extension()
+        if use_hip(self.tree) and ext is not None and hasattr(ext, "segtree_update"):
+            ext.segtree_update(
+                self.tree, idx.contiguous(), values.contiguous(), self._hip_op
+            )
+            return
+        idx = idx + self.capacity
         self.tree[idx] = values
         # propagate: level by level, recompute affected parents up to the root
         nodes = torch.unique(idx >> 1)
@@ -61,6 +76,7 @@ class _SegmentTree:
 
 class SumSegmentTree(_SegmentTree):
     neutral = 0.0
+    _hip_op = _OP_SUM
 
     def _combine(self, a, b):
         return a + b
@@ -68,9 +84,16 @@ class SumSegmentTree(_SegmentTree):
     @torch.no_grad()
     def retrieve(self, prefix: torch.Tensor) -> torch.Tensor:
         """Batched prefix-sum descent: for each p find leaf i with
-        cumsum[:i] <= p < cumsum[:i+1].  Fully vectorized (O(B log N) torch
-        ops, device-resident)."""
-        prefix = prefix.to(self.device).float().clone()
+        cumsum[:i] <= p < cumsum[:i+1].
+
+        GPU path: one lane per prefix with the top tree levels staged in
+        LDS (``segtree_retrieve``, csrc/rl_ops.hip); CPU path is a
+        vectorized O(B log N) torch descent."""
+        prefix = prefix.to(self.device).float()
+        ext = extension()
+        if use_hip(self.tree) and ext is not None and hasattr(ext, "segtree_retrieve"):
+            return ext.segtree_retrieve(self.tree, prefix.contiguous())
+        prefix = prefix.clone()
         idx = torch.ones_like(prefix, dtype=torch.long)
         depth = int(math.log2(self.capacity))
         for _ in range(depth):
@@ -101,6 +124,7 @@ class SumSegmentTree(_SegmentTree):
 
 class MinSegmentTree(_SegmentTree):
     neutral = float("inf")
+    _hip_op = _OP_MIN
 
     def _combine(self, a, b):
         return torch.minimum(a, b)

@@ -15,8 +15,10 @@ Kernel inventory (reference call sites in SURVEY.md §2.9):
 - :func:`fused_linear_logprobs` — chunked lm_head logprob (fwd+bwd).
 - :func:`grpo_token_loss` — fused token-masked GRPO/CISPO surrogate.
 
-PER's GPU segment tree lives in :mod:`agilerl_amd.components.segment_tree`
-(``segtree_update`` / ``segtree_sample`` kernels in ``csrc/rl_ops.hip``).
+PER's GPU segment trees dispatch from
+:mod:`agilerl_amd.components.segment_tree` to the ``segtree_update`` /
+``segtree_retrieve`` / fused ``per_sample`` kernels in ``csrc/rl_ops.hip``
+(single-workgroup barriered propagation; LDS-staged prefix descent).
 """
 
 from __future__ import annotations

@@ -310,6 +310,155 @@ torch::Tensor group_advantage(torch::Tensor rewards, long group_size, bool scale
 }
 
 // ---------------------------------------------------------------------------
+// PER segment trees (reference CPU counterpart
+// agilerl/components/segment_tree.py:119-222).  Tree layout: flat
+// (2*capacity) array, root at 1, leaves at [capacity, 2*capacity).
+//
+// update: one workgroup, barrier per level — every thread scatters its
+// leaves, then recomputes the ancestors of its leaves level by level.
+// Shared ancestors are recomputed redundantly by several threads, but each
+// write is a full recompute from already-final children, so racing writes
+// store identical bits (benign).  Duplicate leaf indices resolve
+// nondeterministically (same as the torch scatter path).
+//
+// descent: one lane per prefix value; the top levels of the tree are
+// staged into LDS once per workgroup (32 KB = top 13 levels) so the
+// latency-bound top of every descent hits LDS instead of HBM.
+// ---------------------------------------------------------------------------
+
+#define SEG_OP_SUM 0
+#define SEG_OP_MIN 1
+#define SEG_LDS_N 8192  // staged floats (32 KB LDS): nodes [0, 8192)
+
+__global__ void segtree_update_kernel(
+    float* __restrict__ tree,
+    const long* __restrict__ idx,     // (B,) leaf indices in [0, capacity)
+    const float* __restrict__ values, // (B,)
+    long B, int capacity, int depth, int op) {
+  int tid = threadIdx.x;
+  int nthreads = blockDim.x;
+  // scatter leaves
+  for (long b = tid; b < B; b += nthreads) {
+    tree[capacity + idx[b]] = values[b];
+  }
+  __syncthreads();
+  // propagate: level k recomputes each updated leaf's ancestor at that level
+  for (int k = 1; k <= depth; ++k) {
+    for (long b = tid; b < B; b += nthreads) {
+      long node = (capacity + idx[b]) >> k;
+      float l = tree[2 * node], r = tree[2 * node + 1];
+      tree[node] = (op == SEG_OP_MIN) ? fminf(l, r) : (l + r);
+    }
+    __syncthreads();
+  }
+}
+
+void segtree_update(torch::Tensor tree, torch::Tensor idx, torch::Tensor values, long op) {
+  CHECK_GPU(tree); CHECK_CONTIG(tree);
+  CHECK_GPU(idx); CHECK_CONTIG(idx);
+  int capacity = tree.size(0) / 2;
+  int depth = 0;  // log2(capacity): leaf node (capacity+i) >> depth == 1
+  while ((1 << depth) < capacity) ++depth;
+  long B = idx.numel();
+  if (B == 0) return;
+  hipStream_t stream = at::hip::getCurrentHIPStream();
+  hipLaunchKernelGGL(segtree_update_kernel, dim3(1), dim3(1024), 0, stream,
+      tree.data_ptr<float>(), idx.data_ptr<long>(), values.data_ptr<float>(),
+      B, capacity, depth, (int)op);
+}
+
+__device__ inline long segtree_descend_one(
+    const float* __restrict__ tree, const float* __restrict__ lds,
+    int lds_n, float p, int capacity, int depth) {
+  long node = 1;
+  for (int d = 0; d < depth; ++d) {
+    long left = 2 * node;
+    float lv = (left + 1 < lds_n) ? lds[left] : tree[left];
+    if (p >= lv) { p -= lv; node = left + 1; } else { node = left; }
+  }
+  return node - capacity;
+}
+
+__global__ void segtree_retrieve_kernel(
+    const float* __restrict__ tree,
+    const float* __restrict__ prefix,
+    long* __restrict__ out_idx,
+    long B, int capacity, int depth) {
+  __shared__ float lds[SEG_LDS_N];
+  int lds_n = min(2 * capacity, SEG_LDS_N);
+  for (int i = threadIdx.x; i < lds_n; i += blockDim.x) lds[i] = tree[i];
+  __syncthreads();
+  long b = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  for (; b < B; b += (long)gridDim.x * blockDim.x) {
+    out_idx[b] = segtree_descend_one(tree, lds, lds_n, prefix[b], capacity, depth);
+  }
+}
+
+torch::Tensor segtree_retrieve(torch::Tensor tree, torch::Tensor prefix) {
+  CHECK_GPU(tree); CHECK_CONTIG(tree); CHECK_GPU(prefix); CHECK_CONTIG(prefix);
+  int capacity = tree.size(0) / 2;
+  int depth = 0;
+  while ((1 << depth) < capacity) ++depth;
+  long B = prefix.numel();
+  auto out = torch::empty({B}, prefix.options().dtype(torch::kLong));
+  if (B == 0) return out;
+  int block = 256;
+  hipStream_t stream = at::hip::getCurrentHIPStream();
+  hipLaunchKernelGGL(segtree_retrieve_kernel, dim3(grid_1d(B, block)), dim3(block), 0, stream,
+      tree.data_ptr<float>(), prefix.data_ptr<float>(), out.data_ptr<long>(),
+      B, capacity, depth);
+  return out;
+}
+
+// Fused PER sample: stratified prefixes from uniform jitter, LDS descent,
+// leaf gather, IS-weight computation — one launch, zero host syncs
+// (total and p_min are read from the tree roots on device).
+__global__ void per_sample_kernel(
+    const float* __restrict__ sum_tree,
+    const float* __restrict__ min_tree,
+    const float* __restrict__ rand01,  // (B,) uniform jitter
+    long* __restrict__ out_idx,
+    float* __restrict__ out_w,
+    long B, int capacity, int depth, long size, float beta) {
+  __shared__ float lds[SEG_LDS_N];
+  int lds_n = min(2 * capacity, SEG_LDS_N);
+  for (int i = threadIdx.x; i < lds_n; i += blockDim.x) lds[i] = sum_tree[i];
+  __syncthreads();
+  float total = fmaxf(sum_tree[1], 1e-12f);
+  float p_min = min_tree[1] / total;
+  float max_w = (p_min > 0.f) ? powf(p_min * (float)size, -beta) : 1.f;
+  long b = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  for (; b < B; b += (long)gridDim.x * blockDim.x) {
+    float u = ((float)b + rand01[b]) / (float)B * total;  // stratified
+    long leaf = segtree_descend_one(sum_tree, lds, lds_n, u, capacity, depth);
+    if (leaf > size - 1) leaf = size - 1;
+    out_idx[b] = leaf;
+    float p = sum_tree[capacity + leaf] / total;
+    float pw = fmaxf(p * (float)size, 1e-12f);
+    out_w[b] = powf(pw, -beta) / max_w;
+  }
+}
+
+std::vector<torch::Tensor> per_sample(
+    torch::Tensor sum_tree, torch::Tensor min_tree, torch::Tensor rand01,
+    long size, double beta) {
+  CHECK_GPU(sum_tree); CHECK_CONTIG(sum_tree);
+  int capacity = sum_tree.size(0) / 2;
+  int depth = 0;
+  while ((1 << depth) < capacity) ++depth;
+  long B = rand01.numel();
+  auto idx = torch::empty({B}, rand01.options().dtype(torch::kLong));
+  auto w = torch::empty({B}, rand01.options());
+  int block = 256;
+  hipStream_t stream = at::hip::getCurrentHIPStream();
+  hipLaunchKernelGGL(per_sample_kernel, dim3(grid_1d(B, block)), dim3(block), 0, stream,
+      sum_tree.data_ptr<float>(), min_tree.data_ptr<float>(),
+      rand01.data_ptr<float>(), idx.data_ptr<long>(), w.data_ptr<float>(),
+      B, capacity, depth, size, (float)beta);
+  return {idx, w};
+}
+
+// ---------------------------------------------------------------------------
 
 void init_lm_ops(pybind11::module_& m);    // lm_ops.hip
 void init_norm_ops(pybind11::module_& m);  // norm_ops.hip
@@ -327,4 +476,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("polyak_", &polyak_, "fused multi-tensor polyak update");
   m.def("noisy_linear_fwd", &noisy_linear_fwd, "fused NoisyLinear forward");
   m.def("group_advantage", &group_advantage, "GRPO group-relative advantage");
+  m.def("segtree_update", &segtree_update, "PER segment-tree batched update+propagate");
+  m.def("segtree_retrieve", &segtree_retrieve, "PER sum-tree prefix descent (LDS-staged)");
+  m.def("per_sample", &per_sample, "fused PER stratified sample + IS weights");
 }
