@@ -64,6 +64,7 @@ class DecodeEngine:
         page_size: int = 16,
         eos_token_id: Optional[int] = None,
         set_adapter_fn: Optional[Callable[[Optional[str]], None]] = None,
+        use_paged_attention: Optional[bool] = None,
     ):
         self.model = model
         cfg = model.config
@@ -76,6 +77,18 @@ class DecodeEngine:
             num_pages=num_pages, page_size=page_size, dtype=dtype, device=device,
         )
         self.device = device
+        # paged-attention decode path: per-layer step reading K/V straight
+        # from the pool via the HIP kernel (auto: on for Llama-family on
+        # GPU; CPU keeps the gather path unless explicitly requested)
+        from .paged_llama import PagedLlamaDecodeRunner, is_paged_decodable
+
+        if use_paged_attention is None:
+            use_paged_attention = device.startswith("cuda") and is_paged_decodable(model)
+        self._paged_runner = (
+            PagedLlamaDecodeRunner(model, self.cache)
+            if use_paged_attention and is_paged_decodable(model)
+            else None
+        )
         self.max_batch = max_batch
         self.eos_token_id = eos_token_id
         self.set_adapter_fn = set_adapter_fn
@@ -176,9 +189,30 @@ class DecodeEngine:
         return int(logits.argmax())
 
     @torch.no_grad()
+    def _decode_paged(self, seqs: List[SequenceState]) -> None:
+        """One token for every sequence through the paged-attention path:
+        no gather — Q/K/V projections for the single new token, K/V
+        appended to the pool in place, attention reads pages directly."""
+        runner = self._paged_runner
+        seq_ids = [s.seq_id for s in seqs]
+        for sid in seq_ids:  # page for the incoming token
+            self.cache._ensure_capacity(sid, 1)
+        table = runner.build_table(seq_ids, self.device)
+        tokens = torch.tensor([s.last_token for s in seqs], dtype=torch.long,
+                              device=self.device)
+        positions = torch.tensor([self.cache.lengths[sid] for sid in seq_ids],
+                                 dtype=torch.long, device=self.device)
+        logits = runner.decode_step(tokens, positions, table)
+        for i, (s, sid) in enumerate(zip(seqs, seq_ids)):
+            self.cache.lengths[sid] += 1
+            s.generated.append(self._select(logits[i], s))
+
+    @torch.no_grad()
     def _decode(self, seqs: List[SequenceState]) -> None:
         """One token for every sequence: gather pages -> DynamicCache ->
         single batched forward at the shared end position."""
+        if self._paged_runner is not None:
+            return self._decode_paged(seqs)
         from transformers.cache_utils import DynamicCache
 
         seq_ids = [s.seq_id for s in seqs]

@@ -54,7 +54,8 @@ def parse_args():
     p.add_argument("--learn-step", type=int, default=LEARN_STEP)
     p.add_argument("--pop-size", type=int, default=POP_SIZE)
     p.add_argument(
-        "--workload", choices=["ppo", "grpo", "dqn", "rainbow", "maddpg"], default="ppo",
+        "--workload",
+        choices=["ppo", "grpo", "grpo-rollout", "dqn", "rainbow", "maddpg"], default="ppo",
         help="BASELINE configs: ppo = headline (LunarLander pop=8); "
         "dqn = config 1 (CartPole pop=1 CPU plumbing); rainbow = config 3 "
         "(PER+noisy CNN on the Pong-like visual env, pop=4); maddpg = "
@@ -266,12 +267,50 @@ class GrpoBenchRunner:
         return int(batch["ids"].numel())
 
 
-def run_grpo(args):
-    runner = GrpoBenchRunner(args)
+class GrpoRolloutBenchRunner(GrpoBenchRunner):
+    """grpo-rollout: generation INSIDE the timed region — the dominant RFT
+    cost the reference covers with colocated vLLM.  Each step: paged-engine
+    generation of C completion tokens per sequence, then the GRPO learn
+    phase on the generated batch."""
+
+    def __init__(self, args):
+        super().__init__(args)
+        self.agent.generation = "paged"
+        self.prompt_len = max(8, args.seq_len // 2)
+        self.completion_len = args.seq_len - self.prompt_len
+        self.agent.max_completion_tokens = self.completion_len
+        self.gen_tokens = 0
+        self.gen_seconds = 0.0
+
+    def bench_step(self) -> int:
+        import time as _t
+
+        B, P = self.args.grpo_batch, self.prompt_len
+        ids = torch.from_numpy(np.random.randint(1, self.vocab, (B, P))).to(self.device)
+        mask = torch.ones_like(ids)
+        t0 = _t.perf_counter()
+        seqs = self.agent.get_action({"input_ids": ids, "attention_mask": mask})
+        if torch.cuda.is_available():
+            torch.cuda.synchronize()
+        self.gen_seconds += _t.perf_counter() - t0
+        self.gen_tokens += B * self.completion_len
+        T = seqs.shape[1]
+        pos = torch.arange(T - 1, device=self.device).unsqueeze(0)
+        action_mask = (pos + 1 >= P).float().expand(B, T - 1)
+        rewards = torch.from_numpy(np.random.rand(B).astype(np.float32))
+        self.agent.learn({"ids": seqs, "action_mask": action_mask, "rewards": rewards})
+        return int(seqs.numel())
+
+
+def run_grpo(args, rollout: bool = False):
+    runner = GrpoRolloutBenchRunner(args) if rollout else GrpoBenchRunner(args)
     state = runner.state
     use_cuda = torch.cuda.is_available()
     for _ in range(args.warmup):
         runner.bench_step()
+    if rollout:  # generation stats restart with the timed region
+        runner.gen_tokens = 0
+        runner.gen_seconds = 0.0
     barrier()
     if use_cuda:
         torch.cuda.synchronize()
@@ -295,8 +334,20 @@ def run_grpo(args):
     else:
         total_tokens = float(local_tokens)
     if state.is_main:
+        config = {
+            "model": "Llama-3-8B (random init)" if args.model_size == "8b" else "llama-tiny",
+            "global_batch": args.grpo_batch * state.world_size,
+            "seq_len": args.seq_len,
+            "parallelism": f"dp{state.world_size} (adapter-grad RCCL allreduce)",
+            "algo": "GRPO group_size=8 beta=0.04 LoRA r=16",
+        }
+        if rollout:
+            config["phase"] = "generate(paged engine)+learn end-to-end"
+            if runner.gen_seconds > 0:
+                config["generation_tokens_per_sec"] = runner.gen_tokens / runner.gen_seconds
+                config["generation_fraction_of_step"] = runner.gen_seconds / elapsed
         result = {
-            "metric": "train_tokens_per_sec",
+            "metric": "rollout_tokens_per_sec" if rollout else "train_tokens_per_sec",
             "value": total_tokens / elapsed,
             "unit": "tokens/s",
             "n_gpus": state.world_size,
@@ -308,13 +359,7 @@ def run_grpo(args):
             "vs_baseline": None,
             "dtype": "bf16" if use_cuda else "fp32",
             "data": "synthetic prompt/response tokens, random-init weights",
-            "config": {
-                "model": "Llama-3-8B (random init)" if args.model_size == "8b" else "llama-tiny",
-                "global_batch": args.grpo_batch * state.world_size,
-                "seq_len": args.seq_len,
-                "parallelism": f"dp{state.world_size} (adapter-grad RCCL allreduce)",
-                "algo": "GRPO group_size=8 beta=0.04 LoRA r=16",
-            },
+            "config": config,
         }
         print(json.dumps(result), flush=True)
     if state.is_distributed:
@@ -541,6 +586,8 @@ def main():
         args.num_envs = 65536 if torch.cuda.is_available() else NUM_ENVS
     if args.workload == "grpo":
         return run_grpo(args)
+    if args.workload == "grpo-rollout":
+        return run_grpo(args, rollout=True)
     if args.workload == "dqn":
         return run_dqn(args)
     if args.workload == "rainbow":

@@ -289,3 +289,49 @@ class TestCloneDecodeEngineIsolation:
         clone.generate_paged(ids, mask, do_sample=False)
         assert clone._decode_engine is not parent._decode_engine
         assert clone._decode_engine.set_adapter_fn.__self__ is clone
+
+
+class TestPagedAttentionDecodePath:
+    """The paged-attention decode path (per-layer step + pool-direct
+    attention) must match model.generate greedy exactly — run on CPU via
+    the eager reference attention, so the runner math (RoPE, norms,
+    residuals, pool writes) is pinned without a GPU."""
+
+    def _model(self, seed=0):
+        import torch as _t
+        from transformers import AutoConfig, AutoModelForCausalLM
+
+        cfg = AutoConfig.for_model(
+            "llama", vocab_size=64, hidden_size=32, intermediate_size=64,
+            num_hidden_layers=2, num_attention_heads=2, num_key_value_heads=1,
+            max_position_embeddings=128, pad_token_id=0)
+        _t.manual_seed(seed)
+        return AutoModelForCausalLM.from_config(cfg)
+
+    def test_greedy_parity_vs_generate(self):
+        model = self._model()
+        engine = DecodeEngine(model, num_pages=64, page_size=4,
+                              use_paged_attention=True)
+        assert engine._paged_runner is not None
+        torch.manual_seed(1)
+        prompts = [torch.randint(1, 64, (n,)) for n in (7, 3, 5)]
+        sids = [engine.submit(p, max_new_tokens=8) for p in prompts]
+        results = engine.run_all()
+        for p, sid in zip(prompts, sids):
+            ref = model.generate(
+                p.unsqueeze(0), max_new_tokens=8, do_sample=False,
+                pad_token_id=0)
+            torch.testing.assert_close(results[sid], ref[0])
+
+    def test_ragged_admission_parity(self):
+        model = self._model(seed=3)
+        engine = DecodeEngine(model, max_batch=2, num_pages=64, page_size=4,
+                              use_paged_attention=True)
+        torch.manual_seed(2)
+        prompts = [torch.randint(1, 64, (n,)) for n in (6, 4, 9, 2)]
+        sids = [engine.submit(p, max_new_tokens=5) for p in prompts]
+        results = engine.run_all()
+        for p, sid in zip(prompts, sids):
+            ref = model.generate(p.unsqueeze(0), max_new_tokens=5,
+                                 do_sample=False, pad_token_id=0)
+            torch.testing.assert_close(results[sid], ref[0])
