@@ -168,6 +168,8 @@ class DQN(RLAlgorithm):
             self._graph = None
         if self._graph is None:
             # capturable optimizer state required for in-graph Adam steps
+            prev_opt = self.optimizer.optimizer
+            prev_state = dict(prev_opt.state) if prev_opt is not None else {}
             self.optimizer.optimizer = torch.optim.Adam(
                 [p for p in self.actor.parameters() if p.requires_grad],
                 lr=self.lr, capturable=True,
@@ -200,10 +202,19 @@ class DQN(RLAlgorithm):
                     p.copy_(sv)
                 for p, sv in zip(self.actor_target.parameters(), saved_target):
                     p.copy_(sv)
-                # zero optimizer state IN PLACE (the graph holds these tensors)
-                for group_state in self.optimizer.optimizer.state.values():
-                    for v in group_state.values():
-                        if torch.is_tensor(v):
+                # restore optimizer state IN PLACE (the graph holds these
+                # tensors): carry over pre-existing Adam moments, else zero
+                for p, group_state in self.optimizer.optimizer.state.items():
+                    prior = prev_state.get(p)
+                    for k, v in group_state.items():
+                        if not torch.is_tensor(v):
+                            continue
+                        pv = prior.get(k) if prior else None
+                        if pv is not None and torch.is_tensor(pv) and pv.shape == v.shape:
+                            v.copy_(pv.to(v.device, v.dtype))
+                        elif pv is not None and k == "step" and v.numel() == 1:
+                            v.fill_(float(pv))
+                        else:
                             v.zero_()
             torch.cuda.synchronize()
         st = self._graph_static
