@@ -89,6 +89,16 @@ class DecodeEngine:
             if use_paged_attention and is_paged_decodable(model)
             else None
         )
+        # hipGraph-captured decode (GPU): one graph per adapter over static
+        # slot buffers; a reserved scratch page absorbs zombie-row writes
+        self._graph_decoders: Dict[Optional[str], object] = {}
+        self._scratch_page: Optional[int] = None
+        self._use_decode_graph = (
+            self._paged_runner is not None and device.startswith("cuda")
+            and torch.cuda.is_available()
+        )
+        if self._use_decode_graph and self.cache._free:
+            self._scratch_page = self.cache._free.pop()
         self.max_batch = max_batch
         self.eos_token_id = eos_token_id
         self.set_adapter_fn = set_adapter_fn
@@ -165,6 +175,7 @@ class DecodeEngine:
             input_ids=ids, attention_mask=mask, position_ids=pos,
             past_key_values=cache, use_cache=True,
         )
+        sel = self._select_batch(out.logits[:, -1], seqs).tolist()
         for i, s in enumerate(seqs):
             n = s.prompt_ids.numel()
             layers_k, layers_v = [], []
@@ -174,8 +185,7 @@ class DecodeEngine:
                 layers_v.append(v_l[i, :, L - n :].permute(1, 0, 2))
             self.cache.append(s.seq_id, torch.stack(layers_k), torch.stack(layers_v))
             s.prefilled = True
-            logits = out.logits[i, -1]
-            s.generated.append(self._select(logits, s))
+            s.generated.append(int(sel[i]))
 
     @staticmethod
     def _layer_kv(cache, layer: int) -> Tuple[torch.Tensor, torch.Tensor]:
@@ -188,31 +198,84 @@ class DecodeEngine:
             return int(torch.multinomial(probs, 1))
         return int(logits.argmax())
 
+    def _select_batch(self, logits: torch.Tensor, seqs: List[SequenceState]) -> torch.Tensor:
+        """One device-side selection for the whole batch (a single host
+        sync at .tolist() time instead of one per sequence)."""
+        temps = [float(s.temperature or 0.0) for s in seqs]
+        greedy = logits.argmax(-1)
+        if not any(t > 0 for t in temps):
+            return greedy
+        t = torch.tensor(
+            [tt if tt > 0 else 1.0 for tt in temps], device=logits.device
+        )
+        probs = torch.softmax(logits.float() / t.unsqueeze(1), dim=-1)
+        sampled = torch.multinomial(probs, 1).squeeze(1)
+        use = torch.tensor([tt > 0 for tt in temps], device=logits.device)
+        return torch.where(use, sampled, greedy)
+
     @torch.no_grad()
-    def _decode_paged(self, seqs: List[SequenceState]) -> None:
+    def _decode_paged(self, seqs: List[SequenceState], adapter=None) -> None:
         """One token for every sequence through the paged-attention path:
         no gather — Q/K/V projections for the single new token, K/V
-        appended to the pool in place, attention reads pages directly."""
+        appended to the pool in place, attention reads pages directly.
+        On GPU the whole step replays a captured hipGraph."""
         runner = self._paged_runner
         seq_ids = [s.seq_id for s in seqs]
         for sid in seq_ids:  # page for the incoming token
             self.cache._ensure_capacity(sid, 1)
-        table = runner.build_table(seq_ids, self.device)
-        tokens = torch.tensor([s.last_token for s in seqs], dtype=torch.long,
-                              device=self.device)
-        positions = torch.tensor([self.cache.lengths[sid] for sid in seq_ids],
-                                 dtype=torch.long, device=self.device)
-        logits = runner.decode_step(tokens, positions, table)
-        for i, (s, sid) in enumerate(zip(seqs, seq_ids)):
+
+        dec = self._graph_decoder(adapter, len(seqs))
+        if dec is not None:
+            for s, sid in zip(seqs, seq_ids):
+                if sid not in dec.slot_of:
+                    slot = dec.acquire_slot(sid)
+                    dec.load_row(slot, s.last_token, self.cache.lengths[sid],
+                                 self.cache.page_tables[sid])
+                else:
+                    dec.sync_row_pages(dec.slot_of[sid], self.cache.page_tables[sid])
+            all_logits = dec.replay()
+            slot_t = torch.tensor([dec.slot_of[sid] for sid in seq_ids],
+                                  dtype=torch.long, device=self.device)
+            sel = self._select_batch(all_logits[slot_t], seqs)
+            dec.tokens[slot_t] = sel
+            dec.positions[slot_t] += 1
+            sel_host = sel.tolist()
+        else:
+            table = runner.build_table(seq_ids, self.device)
+            tokens = torch.tensor([s.last_token for s in seqs], dtype=torch.long,
+                                  device=self.device)
+            positions = torch.tensor([self.cache.lengths[sid] for sid in seq_ids],
+                                     dtype=torch.long, device=self.device)
+            logits = runner.decode_step(tokens, positions, table)
+            sel_host = self._select_batch(logits, seqs).tolist()
+        for s, sid, tok in zip(seqs, seq_ids, sel_host):
             self.cache.lengths[sid] += 1
-            s.generated.append(self._select(logits[i], s))
+            s.generated.append(int(tok))
+
+    def _graph_decoder(self, adapter, batch_size: int):
+        """Per-adapter graphed decoder (adapter routing is python control
+        flow inside LoraLinear, so it bakes at capture time)."""
+        if not self._use_decode_graph or self._scratch_page is None:
+            return None
+        from .paged_llama import GraphedPagedDecoder
+
+        dec = self._graph_decoders.get(adapter)
+        needed = max(self.max_batch, batch_size)
+        if dec is not None and dec.max_batch < needed:
+            dec = None  # batch outgrew the captured statics: recapture
+        if dec is None:
+            dec = GraphedPagedDecoder(
+                self._paged_runner, needed, self._scratch_page, self.device
+            )
+            self._graph_decoders[adapter] = dec
+        return dec
 
     @torch.no_grad()
-    def _decode(self, seqs: List[SequenceState]) -> None:
+    def _decode(self, seqs: List[SequenceState], adapter=None) -> None:
         """One token for every sequence: gather pages -> DynamicCache ->
         single batched forward at the shared end position."""
         if self._paged_runner is not None:
-            return self._decode_paged(seqs)
+            return self._decode_paged(seqs, adapter)
         from transformers.cache_utils import DynamicCache
 
         seq_ids = [s.seq_id for s in seqs]
@@ -266,7 +329,7 @@ class DecodeEngine:
                 # it becomes this step's decode input for old sequences
                 self._prefill(new)
             if old:
-                self._decode(old)
+                self._decode(old, adapter)
         finished = []
         for s in list(self.active.values()):
             hit_eos = self.eos_token_id is not None and s.generated and \
@@ -276,10 +339,15 @@ class DecodeEngine:
                 self.sequences_finished += 1
                 self.tokens_generated += len(s.generated)
                 finished.append((s.seq_id, s.output_ids()))
-                self.cache.free(s.seq_id)
-                self._reserved.pop(s.seq_id, None)
-                del self.active[s.seq_id]
+                self._release_seq(s.seq_id)
         return finished
+
+    def _release_seq(self, seq_id: int) -> None:
+        self.cache.free(seq_id)
+        self._reserved.pop(seq_id, None)
+        self.active.pop(seq_id, None)
+        for dec in self._graph_decoders.values():
+            dec.release_slot(seq_id)
 
     def cancel(self, seq_id: int) -> bool:
         """Abort a sequence (waiting or active); frees its pages."""
@@ -288,9 +356,7 @@ class DecodeEngine:
                 self.waiting.pop(i)
                 return True
         if seq_id in self.active:
-            self.cache.free(seq_id)
-            self._reserved.pop(seq_id, None)
-            del self.active[seq_id]
+            self._release_seq(seq_id)
             return True
         return False
 
