@@ -298,9 +298,11 @@ torch::Tensor paged_attn_decode(
   if (fd_ok) {
     int log2S = 0;
     while ((1 << log2S) < S) ++log2S;
-    // enough blocks to fill 256 CUs; bounded by ~64 tokens per split
+    // the kernel is latency-bound per wave, so oversubscribe: target ~4096
+    // blocks (16 waves per SIMD across 256 CUs), bounded by ~64 tokens per
+    // split so short sequences don't shred into empty splits
     long bh = (long)B * Hkv;
-    int splits = (int)std::min<long>(FD_MAX_SPLITS, std::max<long>(1, 1024 / bh));
+    int splits = (int)std::min<long>(FD_MAX_SPLITS, std::max<long>(1, 4096 / bh));
     splits = (int)std::min<long>(splits,
         std::max<long>(1, ((long)max_pages * S + 63) / 64));
     auto ws_acc = torch::empty({(long)B * Hq * splits * D},
