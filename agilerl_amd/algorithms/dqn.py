@@ -156,10 +156,14 @@ class DQN(RLAlgorithm):
         self.optimizer.zero_grad(set_to_none=False)
         loss.backward()
         self.optimizer.step()
-        # graph-safe polyak: foreach lerp (stable param addresses)
-        torch._foreach_lerp_(
-            list(self.actor_target.parameters()), list(self.actor.parameters()), self.tau
-        )
+        # graph-safe polyak: foreach lerp (stable param addresses).  MUST be
+        # no-grad: an in-place lerp from grad-requiring actor params would
+        # silently make the target params non-leaf graph nodes (leaking the
+        # autograd graph and breaking later requires_grad mutation).
+        with torch.no_grad():
+            torch._foreach_lerp_(
+                list(self.actor_target.parameters()), list(self.actor.parameters()), self.tau
+            )
         loss_out.copy_(loss.detach())
 
     def _graphed_learn(self, obs, actions, rewards, next_obs, dones) -> float:

@@ -342,9 +342,20 @@ class LLMAlgorithm(EvolvableAlgorithm):
         self.model.train()
         out = torch.full((B, P + C), pad_id, dtype=torch.long, device=self.device)
         out[:, :P] = input_ids
+        # behavior-policy logprobs on the TARGET grid (B, P+C-1): position j
+        # holds the sampling logprob of token out[:, j+1]; zero elsewhere
+        # (the reference keeps vLLM sampling logprobs the same way)
+        samp = torch.zeros((B, P + C - 1), dtype=torch.float32, device=self.device)
         for i, sid in enumerate(sids):
             comp = results[sid][int(attention_mask[i].sum()):]
             out[i, P : P + comp.numel()] = comp.to(self.device)
+            lps = engine.finished_logps.get(sid)
+            if lps:
+                n = min(len(lps), C)
+                samp[i, P - 1 : P - 1 + n] = torch.tensor(
+                    lps[:n], dtype=torch.float32, device=self.device
+                )
+        self.last_sampling_logps = samp
         return out
 
     @torch.no_grad()
@@ -358,6 +369,7 @@ class LLMAlgorithm(EvolvableAlgorithm):
     ) -> torch.Tensor:
         """Batched KV-cached generation with the agent's adapter active."""
         self._activate("self")
+        self.last_sampling_logps = None  # only the paged engine captures these
         self.model.eval()
         pad_id = getattr(self.model.config, "pad_token_id", None)
         if pad_id is None:

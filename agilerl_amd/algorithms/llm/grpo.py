@@ -51,6 +51,11 @@ class GRPO(LLMAlgorithm):
         beta: float = 0.0,
         scale_rewards: bool = True,
         loss_norm: str = "token",
+        importance_sampling_level: str = "token",
+        advantage_level: str = "trajectory",
+        sampling_is_correction: bool = False,
+        sampling_is_cap: float = 2.0,
+        grad_accumulation_steps: int = 1,
         max_grad_norm: float = 1.0,
         temperature: float = 1.0,
         max_completion_tokens: int = 256,
@@ -82,7 +87,30 @@ class GRPO(LLMAlgorithm):
         self.clip_coef_lower = float(clip_coef_lower) if clip_coef_lower is not None else None
         self.beta = float(beta)  # k3 KL coefficient
         self.scale_rewards = bool(scale_rewards)
+        if loss_norm not in ("token", "sequence", "accumulation_window"):
+            raise ValueError(f"unknown loss_norm {loss_norm!r}")
         self.loss_norm = loss_norm
+        # importance-sampling pooling level (reference grpo.py:1848-1903):
+        # "token" (GRPO) / "turn" (per-turn pooled ratio) / "trajectory"
+        # (GSPO).  Subclass default via SEQUENCE_LEVEL_IS for back-compat.
+        if importance_sampling_level not in ("token", "turn", "trajectory"):
+            raise ValueError(
+                f"unknown importance_sampling_level {importance_sampling_level!r}"
+            )
+        if self.SEQUENCE_LEVEL_IS and importance_sampling_level == "token":
+            importance_sampling_level = "trajectory"
+        self.importance_sampling_level = importance_sampling_level
+        # advantage granularity: "trajectory" (group-relative over final
+        # rewards) or "turn" (group-relative per turn over turn_rewards,
+        # broadcast to that turn's tokens — reference grpo.py:1250-1379)
+        if advantage_level not in ("trajectory", "turn"):
+            raise ValueError(f"unknown advantage_level {advantage_level!r}")
+        self.advantage_level = advantage_level
+        # truncated-IS correction against the decode engine's sampling
+        # logprobs (the vLLM-IS analog, reference grpo.py:2500-2510)
+        self.sampling_is_correction = bool(sampling_is_correction)
+        self.sampling_is_cap = float(sampling_is_cap)
+        self.grad_accumulation_steps = max(1, int(grad_accumulation_steps))
 
     # ------------------------------------------------------------------
     def get_action(self, prompts: Dict[str, torch.Tensor], training: bool = True) -> torch.Tensor:
@@ -112,10 +140,28 @@ class GRPO(LLMAlgorithm):
         attention_mask = attention_mask.to(self.device)
         action_mask = experiences["action_mask"].to(self.device).float()
         rewards = experiences["rewards"].to(self.device).float()
+        turn_ids = experiences.get("turn_ids")
+        if turn_ids is not None:
+            turn_ids = turn_ids.to(self.device)
+        sampling_logps = experiences.get("sampling_logps")
+        if sampling_logps is not None and self.sampling_is_correction:
+            sampling_logps = sampling_logps.to(self.device).float()
+        else:
+            sampling_logps = None
+        turn_rewards = experiences.get("turn_rewards")
 
         self.check_seq_len_agreement(ids.shape[1])
-        advantages = self._calculate_advantages(rewards)  # (B,)
-        adv_tok = advantages.unsqueeze(1).expand_as(action_mask)
+        if (
+            self.advantage_level == "turn"
+            and turn_rewards is not None
+            and turn_ids is not None
+        ):
+            adv_tok = self._turn_advantages(
+                turn_rewards.to(self.device).float(), turn_ids, action_mask
+            )
+        else:
+            advantages = self._calculate_advantages(rewards)  # (B,)
+            adv_tok = advantages.unsqueeze(1).expand_as(action_mask)
 
         B = ids.shape[0]
         mb = max(self.micro_batch_size, 1)
@@ -135,10 +181,21 @@ class GRPO(LLMAlgorithm):
 
         stats = {"loss": 0.0, "kl": 0.0, "clip_frac": 0.0}
         n_updates = 0
+        accum = self.grad_accumulation_steps
         for _ in range(self.update_epochs):
             perm = torch.randperm(B, device=self.device)
-            for s in range(0, B, mb):
+            starts = list(range(0, B, mb))
+            # accumulation windows: optimizer steps every `accum`
+            # micro-batches; "accumulation_window" loss_norm divides every
+            # micro-batch by the WINDOW's action-token count (reference
+            # grpo.py:1619-1694 loss-norm windows)
+            for i, s in enumerate(starts):
                 sel = perm[s : s + mb]
+                window_denom = None
+                if self.loss_norm == "accumulation_window":
+                    w0 = (i // accum) * accum
+                    wsel = perm[starts[w0] : starts[w0] + mb * accum]
+                    window_denom = float(action_mask[wsel].sum().clamp(min=1.0))
                 logp = logprob_fn(ids[sel], attention_mask[sel], with_grad=True)
                 loss = self._policy_loss(
                     logp,
@@ -148,9 +205,13 @@ class GRPO(LLMAlgorithm):
                     ref_logp[sel] if ref_logp is not None else None,
                     clip_lo,
                     clip_hi,
+                    turn_ids=turn_ids[sel] if turn_ids is not None else None,
+                    sampling_logp=sampling_logps[sel] if sampling_logps is not None else None,
+                    denom_tokens=window_denom,
                 )
                 self.raise_if_loss_not_finite_on_any_rank(loss)
-                self.backward_and_step(loss)
+                accumulate = (i % accum) != accum - 1 and i != len(starts) - 1
+                self.backward_and_step(loss, accumulate=accumulate)
                 with torch.no_grad():
                     ratio = (logp - old_logp[sel]).exp()
                     m = action_mask[sel].bool()
@@ -169,13 +230,37 @@ class GRPO(LLMAlgorithm):
         return stats
 
     # ------------------------------------------------------------------
-    def _policy_loss(self, logp, old_logp, adv_tok, mask, ref_logp, clip_lo, clip_hi):
-        """Token-level surrogate (GRPO / CISPO) via the fused HIP kernel;
-        GSPO overrides with the sequence-level pooled-IS objective."""
+    def _turn_advantages(
+        self,
+        turn_rewards: torch.Tensor,  # (B, K) per-turn rewards
+        turn_ids: torch.Tensor,      # (B, T) turn index per target (-1 pad)
+        action_mask: torch.Tensor,   # (B, T)
+    ) -> torch.Tensor:
+        """Group-relative advantage per (group, turn), broadcast to each
+        turn's tokens (reference grpo.py:1250-1379)."""
+        B, K = turn_rewards.shape
+        g = turn_rewards.view(-1, self.group_size, K)
+        adv = g - g.mean(dim=1, keepdim=True)
+        if self.scale_rewards:
+            adv = adv / (g.std(dim=1, keepdim=True) + 1e-8)
+        adv = adv.view(B, K)
+        safe = turn_ids.clamp(min=0, max=K - 1).long()
+        return adv.gather(1, safe) * (turn_ids >= 0).float() * action_mask
+
+    def _policy_loss(self, logp, old_logp, adv_tok, mask, ref_logp, clip_lo,
+                     clip_hi, turn_ids=None, sampling_logp=None,
+                     denom_tokens=None):
+        """Surrogate at the configured IS level: token level (GRPO/CISPO)
+        runs the fused HIP kernel; turn/trajectory pooling and the
+        sampling-IS correction take the eager path (pooling is cheap; the
+        logprob computation upstream stays fused either way)."""
         return grpo_policy_loss(
             logp, old_logp, adv_tok, mask, ref_logp=ref_logp,
             clip_lo=clip_lo, clip_hi=clip_hi, kl_coef=self.beta,
             cispo=self.CISPO, loss_norm=self.loss_norm,
+            level=self.importance_sampling_level, turn_ids=turn_ids,
+            sampling_logp=sampling_logp, sampling_cap=self.sampling_is_cap,
+            denom_tokens=denom_tokens,
         )
 
     # ------------------------------------------------------------------

@@ -31,6 +31,9 @@ class SequenceState:
     temperature: float = 0.0  # 0 => greedy
     adapter: Optional[str] = None
     generated: List[int] = field(default_factory=list)
+    # behavior-policy logprob of each generated token (under the sampling
+    # temperature) — the decode-engine analog of vLLM sampling logprobs
+    gen_logps: List[float] = field(default_factory=list)
     prefilled: bool = False
     done: bool = False
 
@@ -106,6 +109,7 @@ class DecodeEngine:
         self._reserved: Dict[int, int] = {}  # seq_id -> pages promised
         self.active: Dict[int, SequenceState] = {}
         self.waiting: List[SequenceState] = []
+        self.finished_logps: Dict[int, List[float]] = {}  # sampling logprobs
         self.steps_run = 0
         self.tokens_generated = 0
         self.sequences_finished = 0
@@ -175,7 +179,9 @@ class DecodeEngine:
             input_ids=ids, attention_mask=mask, position_ids=pos,
             past_key_values=cache, use_cache=True,
         )
-        sel = self._select_batch(out.logits[:, -1], seqs).tolist()
+        sel_t = self._select_batch(out.logits[:, -1], seqs)
+        lps = self._chosen_logps(out.logits[:, -1], sel_t, seqs).tolist()
+        sel = sel_t.tolist()
         for i, s in enumerate(seqs):
             n = s.prompt_ids.numel()
             layers_k, layers_v = [], []
@@ -186,6 +192,7 @@ class DecodeEngine:
             self.cache.append(s.seq_id, torch.stack(layers_k), torch.stack(layers_v))
             s.prefilled = True
             s.generated.append(int(sel[i]))
+            s.gen_logps.append(float(lps[i]))
 
     @staticmethod
     def _layer_kv(cache, layer: int) -> Tuple[torch.Tensor, torch.Tensor]:
@@ -213,6 +220,19 @@ class DecodeEngine:
         use = torch.tensor([tt > 0 for tt in temps], device=logits.device)
         return torch.where(use, sampled, greedy)
 
+    @staticmethod
+    def _chosen_logps(logits: torch.Tensor, chosen: torch.Tensor,
+                      seqs: List[SequenceState]) -> torch.Tensor:
+        """Behavior-policy logprob of the chosen tokens (temperature-scaled
+        for sampling rows, plain softmax for greedy rows)."""
+        t = torch.tensor(
+            [float(s.temperature) if s.temperature and s.temperature > 0 else 1.0
+             for s in seqs],
+            device=logits.device,
+        )
+        lsm = torch.log_softmax(logits.float() / t.unsqueeze(1), dim=-1)
+        return lsm.gather(1, chosen.unsqueeze(1)).squeeze(1)
+
     @torch.no_grad()
     def _decode_paged(self, seqs: List[SequenceState], adapter=None) -> None:
         """One token for every sequence through the paged-attention path:
@@ -236,9 +256,11 @@ class DecodeEngine:
             all_logits = dec.replay()
             slot_t = torch.tensor([dec.slot_of[sid] for sid in seq_ids],
                                   dtype=torch.long, device=self.device)
-            sel = self._select_batch(all_logits[slot_t], seqs)
+            row_logits = all_logits[slot_t]
+            sel = self._select_batch(row_logits, seqs)
             dec.tokens[slot_t] = sel
             dec.positions[slot_t] += 1
+            lps_host = self._chosen_logps(row_logits, sel, seqs).tolist()
             sel_host = sel.tolist()
         else:
             table = runner.build_table(seq_ids, self.device)
@@ -247,10 +269,13 @@ class DecodeEngine:
             positions = torch.tensor([self.cache.lengths[sid] for sid in seq_ids],
                                      dtype=torch.long, device=self.device)
             logits = runner.decode_step(tokens, positions, table)
-            sel_host = self._select_batch(logits, seqs).tolist()
-        for s, sid, tok in zip(seqs, seq_ids, sel_host):
+            sel = self._select_batch(logits, seqs)
+            lps_host = self._chosen_logps(logits, sel, seqs).tolist()
+            sel_host = sel.tolist()
+        for s, sid, tok, lp in zip(seqs, seq_ids, sel_host, lps_host):
             self.cache.lengths[sid] += 1
             s.generated.append(int(tok))
+            s.gen_logps.append(float(lp))
 
     def _graph_decoder(self, adapter, batch_size: int):
         """Per-adapter graphed decoder (adapter routing is python control
@@ -294,6 +319,9 @@ class DecodeEngine:
             input_ids=ids, attention_mask=full_mask, position_ids=pos,
             past_key_values=cache, use_cache=True,
         )
+        sel_t = self._select_batch(out.logits[:, -1], seqs)
+        lps = self._chosen_logps(out.logits[:, -1], sel_t, seqs).tolist()
+        sel = sel_t.tolist()
         for i, s in enumerate(seqs):
             layers_k, layers_v = [], []
             for layer in range(self.cache.num_layers):
@@ -301,7 +329,8 @@ class DecodeEngine:
                 layers_k.append(k_l[i, :, -1:].permute(1, 0, 2))
                 layers_v.append(v_l[i, :, -1:].permute(1, 0, 2))
             self.cache.append(s.seq_id, torch.stack(layers_k), torch.stack(layers_v))
-            s.generated.append(self._select(out.logits[i, -1], s))
+            s.generated.append(int(sel[i]))
+            s.gen_logps.append(float(lps[i]))
 
     # ------------------------------------------------------------------
     def step(self) -> List[Tuple[int, torch.Tensor]]:
@@ -339,6 +368,7 @@ class DecodeEngine:
                 self.sequences_finished += 1
                 self.tokens_generated += len(s.generated)
                 finished.append((s.seq_id, s.output_ids()))
+                self.finished_logps[s.seq_id] = list(s.gen_logps)
                 self._release_seq(s.seq_id)
         return finished
 

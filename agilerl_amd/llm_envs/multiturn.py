@@ -82,8 +82,10 @@ class SyncMultiTurnVecEnv:
         self.rng = np.random.default_rng(seed)
         self.n_traj = data_batch_size * group_size
         self._envs: List[MultiTurnTokenEnv] = []
-        self._segments: List[List[Tuple[List[int], bool, int]]] = []  # (tokens, is_action, turn)
+        # (tokens, is_action, turn, sampling_logps-or-None)
+        self._segments: List[List[Tuple[List[int], bool, int, Optional[List[float]]]]] = []
         self._rewards: List[float] = []
+        self._turn_rewards: List[List[float]] = []
         self._done: List[bool] = []
         self._turn = 0
         self.prompt_len = 0
@@ -93,6 +95,7 @@ class SyncMultiTurnVecEnv:
         self._envs = []
         self._segments = []
         self._rewards = [0.0] * self.n_traj
+        self._turn_rewards = [[0.0] * self.max_turns for _ in range(self.n_traj)]
         self._done = [False] * self.n_traj
         self._turn = 0
         for b in range(self.data_batch_size):
@@ -103,12 +106,12 @@ class SyncMultiTurnVecEnv:
 
                 env_g = _copy.deepcopy(env)
                 self._envs.append(env_g)
-                self._segments.append([(list(prompt), False, -1)])
+                self._segments.append([(list(prompt), False, -1, None)])
         return self._current_prompts()
 
     def _history(self, i: int) -> List[int]:
         out: List[int] = []
-        for tokens, _is_action, _turn in self._segments[i]:
+        for tokens, _is_action, _turn, _lps in self._segments[i]:
             out.extend(tokens)
         return out
 
@@ -128,18 +131,30 @@ class SyncMultiTurnVecEnv:
         return all(self._done)
 
     # ------------------------------------------------------------------
-    def step(self, sequences: torch.Tensor) -> Tuple[Optional[Dict[str, torch.Tensor]], bool]:
-        """``sequences``: (n_traj, P + C) from generate on the last prompts."""
+    def step(
+        self,
+        sequences: torch.Tensor,
+        sampling_logps: Optional[torch.Tensor] = None,
+    ) -> Tuple[Optional[Dict[str, torch.Tensor]], bool]:
+        """``sequences``: (n_traj, P + C) from generate on the last prompts;
+        ``sampling_logps``: optional (n_traj, P+C-1) behavior-policy
+        logprobs on the target grid (reference sync_vec_env.py:239 accepts
+        captured vLLM sampling logprobs the same way)."""
         completions = sequences[:, self.prompt_len :].cpu()
+        P = self.prompt_len
         for i in range(self.n_traj):
             if self._done[i]:
                 continue
             comp = [int(t) for t in completions[i] if int(t) != self.pad_token_id]
-            self._segments[i].append((comp, True, self._turn))
+            lps = None
+            if sampling_logps is not None:
+                lps = [float(x) for x in sampling_logps[i, P - 1 : P - 1 + len(comp)]]
+            self._segments[i].append((comp, True, self._turn, lps))
             feedback, reward, done = self._envs[i].respond(self._turn, comp, self.rng)
             self._rewards[i] += float(reward)
+            self._turn_rewards[i][self._turn] += float(reward)
             if feedback:
-                self._segments[i].append((list(feedback), False, self._turn))
+                self._segments[i].append((list(feedback), False, self._turn, None))
             self._done[i] = done
         self._turn += 1
         if self.all_done or self._turn >= self.max_turns:
@@ -151,28 +166,36 @@ class SyncMultiTurnVecEnv:
     def get_trajectories(self) -> Dict[str, torch.Tensor]:
         """Right-padded full trajectories with completion-token action masks
         and per-target turn ids (-1 for non-action targets)."""
-        rows, masks, turns = [], [], []
+        rows, masks, turns, samp_rows = [], [], [], []
         for segs in self._segments:
             ids: List[int] = []
             act: List[float] = []
             trn: List[int] = []
-            for tokens, is_action, turn in segs:
+            smp: List[float] = []
+            for tokens, is_action, turn, lps in segs:
                 ids.extend(tokens)
                 act.extend([1.0 if is_action else 0.0] * len(tokens))
                 trn.extend([turn] * len(tokens))
+                if is_action and lps is not None and len(lps) == len(tokens):
+                    smp.extend(lps)
+                else:
+                    smp.extend([0.0] * len(tokens))
             rows.append(ids)
             masks.append(act)
             turns.append(trn)
+            samp_rows.append(smp)
         T = max(len(r) for r in rows)
         ids_t = torch.full((self.n_traj, T), self.pad_token_id, dtype=torch.long)
         am_t = torch.zeros((self.n_traj, T), dtype=torch.long)
         act_t = torch.zeros((self.n_traj, T), dtype=torch.float32)
         turn_t = torch.full((self.n_traj, T), -1, dtype=torch.long)
-        for i, (r, m, tr) in enumerate(zip(rows, masks, turns)):
+        samp_t = torch.zeros((self.n_traj, T), dtype=torch.float32)
+        for i, (r, m, tr, sm) in enumerate(zip(rows, masks, turns, samp_rows)):
             ids_t[i, : len(r)] = torch.tensor(r)
             am_t[i, : len(r)] = 1
             act_t[i, : len(m)] = torch.tensor(m)
             turn_t[i, : len(tr)] = torch.tensor(tr)
+            samp_t[i, : len(sm)] = torch.tensor(sm)
         # action_mask over TARGET positions j (predicting ids[:, j+1])
         action_mask = act_t[:, 1:]
         turn_ids = turn_t[:, 1:]
@@ -182,6 +205,8 @@ class SyncMultiTurnVecEnv:
             "action_mask": action_mask,
             "turn_ids": turn_ids,
             "rewards": torch.tensor(self._rewards, dtype=torch.float32),
+            "turn_rewards": torch.tensor(self._turn_rewards, dtype=torch.float32),
+            "sampling_logps": samp_t[:, 1:],
         }
 
 

@@ -109,13 +109,21 @@ paged_attn_split_kernel(
     for (int e = 0; e < 8; ++e) acc[g][e] = 0.f;
   }
 
-  for (int t = s0 + grp; t < s1; t += G) {
+  // software pipeline: issue the NEXT token's K/V loads before the current
+  // token's shuffle-reduce/exp chain, hiding HBM latency behind ALU work
+  auto load_kv = [&](int t, float* kf, float* vf) {
     int page = (t >> log2S) < n_table ? lds_table[t >> log2S]
                                       : table[(long)b * max_pages + (t >> log2S)];
     long base = (((long)page * S + (t & (S - 1))) * Hkv + hk) * D + p * 8;
-    float kf[8], vf[8];
     load_bf16x8(k_pool + base, kf);
     load_bf16x8(v_pool + base, vf);
+  };
+  float kf[8], vf[8];
+  int t0 = s0 + grp;
+  if (t0 < s1) load_kv(t0, kf, vf);
+  for (int t = t0; t < s1; t += G) {
+    float kf2[8], vf2[8];
+    if (t + G < s1) load_kv(t + G, kf2, vf2);
     for (int g = 0; g < gqa; ++g) {
       float part = 0.f;
       #pragma unroll
@@ -131,6 +139,8 @@ paged_attn_split_kernel(
       for (int e = 0; e < 8; ++e) acc[g][e] = acc[g][e] * corr + pr * vf[e];
       m[g] = m_new;
     }
+    #pragma unroll
+    for (int e = 0; e < 8; ++e) { kf[e] = kf2[e]; vf[e] = vf2[e]; }
   }
 
   // combine groups within the wave: xor offsets lpt..WAVE/2 keep the
@@ -298,11 +308,16 @@ torch::Tensor paged_attn_decode(
   if (fd_ok) {
     int log2S = 0;
     while ((1 << log2S) < S) ++log2S;
-    // the kernel is latency-bound per wave, so oversubscribe: target ~4096
-    // blocks (16 waves per SIMD across 256 CUs), bounded by ~64 tokens per
-    // split so short sequences don't shred into empty splits
+    // block-count target: enough (b, hk, split) blocks to cover the 1024
+    // wave slots with latency-hiding headroom, bounded by ~64 tokens per
+    // split so short sequences don't shred into empty splits.  Tunable for
+    // sweeps via AGILERL_PA_SPLIT_TARGET.
+    static long target = [] {
+      const char* env = std::getenv("AGILERL_PA_SPLIT_TARGET");
+      return env ? std::atol(env) : 1024L;
+    }();
     long bh = (long)B * Hkv;
-    int splits = (int)std::min<long>(FD_MAX_SPLITS, std::max<long>(1, 4096 / bh));
+    int splits = (int)std::min<long>(FD_MAX_SPLITS, std::max<long>(1, target / bh));
     splits = (int)std::min<long>(splits,
         std::max<long>(1, ((long)max_pages * S + 63) / 64));
     auto ws_acc = torch::empty({(long)B * Hq * splits * D},

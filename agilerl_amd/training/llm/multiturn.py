@@ -22,12 +22,22 @@ __all__ = ["finetune_llm_multiturn"]
 
 
 def rollout_multiturn(agent, env):
-    """One synchronized multi-turn episode batch -> trajectory experiences."""
+    """One synchronized multi-turn episode batch -> trajectory experiences.
+
+    When the agent generates through the paged engine it exposes the
+    behavior-policy sampling logprobs of the completions
+    (``last_sampling_logps``); they thread into the trajectories for the
+    truncated-IS correction (reference collect_rollouts_llm ->
+    sync_vec_env.step(completion_ids, sampling_logps))."""
     prompts = env.reset()
     done = False
     while not done:
         sequences = agent.get_action(prompts, training=True)
-        prompts, done = env.step(sequences)
+        samp = getattr(agent, "last_sampling_logps", None)
+        try:
+            prompts, done = env.step(sequences, sampling_logps=samp)
+        except TypeError:  # env without sampling-logp support
+            prompts, done = env.step(sequences)
     return env.get_trajectories()
 
 

@@ -77,8 +77,11 @@ class TestPPOGraphMutationSweep:
         eager.learn(dict(flat))
         _assert_params_close(graphed, eager)
 
-        # identical architecture mutation on both
+        # identical architecture mutation on both (new-node init draws RNG,
+        # so reseed per agent for identical fresh weights)
         for agent in (graphed, eager):
+            torch.manual_seed(123)
+            torch.cuda.manual_seed_all(123)
             agent.apply_architecture_mutation("encoder.add_node", numb_new_nodes=16)
         assert graphed._learn_graph is None, "mutation hook must clear the graph"
         flat2 = _flat_rollout(self.N, 8, 4, seed=2)
@@ -184,6 +187,8 @@ class TestDQNGraphMutationSweep:
             torch.testing.assert_close(pa, pb, rtol=2e-3, atol=2e-4)
 
         for agent in (graphed, eager):
+            torch.manual_seed(321)
+            torch.cuda.manual_seed_all(321)
             agent.apply_architecture_mutation("encoder.add_layer")
         for seed in (3, 4):
             graphed.learn(self._batch(seed))

@@ -660,3 +660,156 @@ class TestTextMultiTurn:
                                    max_turns=2, seed=0)
         obs = venv.reset()
         assert obs["input_ids"].shape[0] == 4
+
+
+class TestGrpoParityDeepening:
+    """Round-2 parity: IS levels, loss-norm variants, turn advantages,
+    sampling-IS correction (reference grpo.py:1848-1903, 1619-1694,
+    2500-2510, 1250-1379)."""
+
+    def _tiny_agent(self, **kw):
+        from agilerl_amd.algorithms.llm.grpo import GRPO
+
+        tiny = dict(model_type="llama", vocab_size=64, hidden_size=32,
+                    intermediate_size=64, num_hidden_layers=1,
+                    num_attention_heads=2, num_key_value_heads=1,
+                    max_position_embeddings=128, pad_token_id=0)
+        torch.manual_seed(0)
+        kw.setdefault("micro_batch_size", 2)
+        return GRPO(model_config=tiny, dtype=torch.float32, lora_config={"r": 2},
+                    group_size=2, **kw)
+
+    def _batch(self, B=4, T=12, turns=False):
+        g = torch.Generator().manual_seed(3)
+        ids = torch.randint(1, 64, (B, T), generator=g)
+        pos = torch.arange(T - 1).unsqueeze(0)
+        action_mask = (pos + 1 >= T // 2).float().expand(B, T - 1)
+        out = {"ids": ids, "action_mask": action_mask,
+               "rewards": torch.rand(B, generator=g)}
+        if turns:
+            # two turns over the completion half
+            tid = torch.full((B, T - 1), -1, dtype=torch.long)
+            half = (T - 1 + T // 2) // 2
+            tid[:, T // 2 - 1: half] = 0
+            tid[:, half:] = 1
+            out["turn_ids"] = tid
+            out["turn_rewards"] = torch.rand(B, 2, generator=g)
+        return out
+
+    def test_is_levels_run_and_differ(self):
+        losses = {}
+        for level in ("token", "turn", "trajectory"):
+            agent = self._tiny_agent(importance_sampling_level=level,
+                                     update_epochs=2)
+            stats = agent.learn(self._batch(turns=True))
+            losses[level] = stats["loss"]
+            assert np.isfinite(stats["loss"])
+        # epoch 2 ratios != 1, so pooled levels give different losses
+        assert losses["token"] != losses["trajectory"]
+
+    def test_gspo_uses_trajectory_level(self):
+        from agilerl_amd.algorithms.llm.gspo import GSPO
+
+        tiny = dict(model_type="llama", vocab_size=64, hidden_size=32,
+                    intermediate_size=64, num_hidden_layers=1,
+                    num_attention_heads=2, num_key_value_heads=1,
+                    max_position_embeddings=128, pad_token_id=0)
+        agent = GSPO(model_config=tiny, dtype=torch.float32, lora_config={"r": 2},
+                     group_size=2)
+        assert agent.importance_sampling_level == "trajectory"
+        stats = agent.learn(self._batch())
+        assert np.isfinite(stats["loss"])
+
+    def test_turn_level_advantages(self):
+        agent = self._tiny_agent(advantage_level="turn")
+        batch = self._batch(turns=True)
+        adv = agent._turn_advantages(
+            batch["turn_rewards"], batch["turn_ids"], batch["action_mask"])
+        assert adv.shape == batch["action_mask"].shape
+        # non-action targets carry zero advantage
+        assert (adv * (1 - batch["action_mask"])).abs().sum() == 0
+        # group-centered: within each prompt group and turn, advantages sum to ~0
+        B = adv.shape[0]
+        tr = batch["turn_rewards"].view(-1, 2, 2)
+        centered = tr - tr.mean(dim=1, keepdim=True)
+        assert abs(float(centered.view(B, 2).sum())) < 1e-4
+        stats = agent.learn(batch)
+        assert np.isfinite(stats["loss"])
+
+    def test_sampling_is_correction_changes_loss(self):
+        from agilerl_amd.ops.grpo_loss import grpo_policy_loss
+
+        g = torch.Generator().manual_seed(1)
+        B, T = 4, 10
+        logp = torch.randn(B, T, generator=g) * 0.1 - 2
+        old = logp + torch.randn(B, T, generator=g) * 0.05
+        samp = old + torch.randn(B, T, generator=g) * 0.5
+        adv = torch.randn(B, T, generator=g)
+        mask = torch.ones(B, T)
+        base = grpo_policy_loss(logp, old, adv, mask)
+        corrected = grpo_policy_loss(logp, old, adv, mask, sampling_logp=samp,
+                                     sampling_cap=2.0)
+        assert float(base) != float(corrected)
+        # cap: with sampling == old the ratio is 1 -> identical loss
+        same = grpo_policy_loss(logp, old, adv, mask, sampling_logp=old)
+        torch.testing.assert_close(same, base)
+
+    def test_pool_log_ratio_turn_vs_trajectory(self):
+        from agilerl_amd.ops.grpo_loss import pool_log_ratio
+
+        g = torch.Generator().manual_seed(2)
+        lr = torch.randn(2, 8, generator=g)
+        mask = torch.ones(2, 8)
+        traj = pool_log_ratio(lr, mask, None, "trajectory")
+        torch.testing.assert_close(traj.squeeze(1), lr.mean(dim=1))
+        tid = torch.tensor([[0, 0, 0, 0, 1, 1, 1, 1]] * 2)
+        turn = pool_log_ratio(lr, mask, tid, "turn")
+        torch.testing.assert_close(turn[:, 0], lr[:, :4].mean(dim=1))
+        torch.testing.assert_close(turn[:, 5], lr[:, 4:].mean(dim=1))
+
+    def test_accumulation_window_norm(self):
+        agent = self._tiny_agent(loss_norm="accumulation_window",
+                                 grad_accumulation_steps=2, micro_batch_size=2)
+        stats = agent.learn(self._batch(B=4))
+        assert np.isfinite(stats["loss"])
+
+    def test_multiturn_env_emits_turn_rewards_and_sampling_logps(self):
+        from agilerl_amd.llm_envs.multiturn import SyncMultiTurnVecEnv, TokenGuessEnv
+
+        env = SyncMultiTurnVecEnv(lambda: TokenGuessEnv(vocab_size=32),
+                                  data_batch_size=2, group_size=2, max_turns=2)
+        prompts = env.reset()
+        P = prompts["input_ids"].shape[1]
+        done = False
+        while not done:
+            B, Pn = prompts["input_ids"].shape
+            seqs = torch.cat([prompts["input_ids"],
+                              torch.randint(1, 32, (B, 4))], dim=1)
+            samp = torch.full((B, seqs.shape[1] - 1), -1.5)
+            prompts, done = env.step(seqs, sampling_logps=samp)
+        traj = env.get_trajectories()
+        assert traj["turn_rewards"].shape == (4, 2)
+        torch.testing.assert_close(
+            traj["turn_rewards"].sum(dim=1), traj["rewards"])
+        # sampling logps present exactly on action targets
+        sl = traj["sampling_logps"]
+        am = traj["action_mask"]
+        assert sl.shape == am.shape
+        assert ((sl != 0).float() * (1 - am)).sum() == 0
+        assert float((sl * am).sum()) != 0.0
+
+    def test_multiturn_learn_with_sampling_is(self):
+        from agilerl_amd.llm_envs.multiturn import SyncMultiTurnVecEnv, TokenGuessEnv
+        from agilerl_amd.training.llm.multiturn import rollout_multiturn
+
+        agent = self._tiny_agent(sampling_is_correction=True,
+                                 advantage_level="turn",
+                                 importance_sampling_level="turn",
+                                 generation="paged", max_completion_tokens=4)
+        env = SyncMultiTurnVecEnv(lambda: TokenGuessEnv(vocab_size=60),
+                                  data_batch_size=2, group_size=2, max_turns=2)
+        traj = rollout_multiturn(agent, env)
+        assert "sampling_logps" in traj
+        assert float(traj["sampling_logps"].abs().sum()) > 0  # engine captured
+        stats = agent.learn(traj)
+        assert np.isfinite(stats["loss"])

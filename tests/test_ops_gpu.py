@@ -474,6 +474,7 @@ class TestPagedEngineGpu:
         seq = engine.active[sid]
         prefix = prompt.tolist()
         for tok in forced:
+            engine.cache._ensure_capacity(sid, 1)  # page for the new token
             table = runner.build_table([sid], DEV)
             pos = torch.tensor([engine.cache.lengths[sid]], device=DEV)
             logits = runner.decode_step(
