@@ -300,9 +300,22 @@ class EvolvableAlgorithm(metaclass=AlgorithmMeta):
         torch.save(self.get_checkpoint_dict(), path, pickle_module=dill)
 
     def load_checkpoint(self, path: str) -> None:
-        import dill
+        """Load this package's OR a reference-written checkpoint file.
 
-        ckpt = torch.load(path, pickle_module=dill, weights_only=False, map_location="cpu")
+        Reference format (``agilerl/algorithms/core/base.py:315-372``):
+        flat attribute dict + ``network_info`` with ``{name}_cls/
+        {name}_init_dict/{name}_state_dict`` keys, pickled classes from
+        ``agilerl.*``/``gymnasium.*`` — translated by utils/ref_compat.
+        """
+        from ...utils.ref_compat import (
+            convert_reference_checkpoint,
+            is_reference_layout,
+            load_checkpoint_file,
+        )
+
+        ckpt = load_checkpoint_file(path)
+        if is_reference_layout(ckpt):
+            ckpt = convert_reference_checkpoint(ckpt, device=self.device)
         self._apply_checkpoint(ckpt)
 
     def _apply_checkpoint(self, ckpt: Dict[str, Any]) -> None:
@@ -313,17 +326,44 @@ class EvolvableAlgorithm(metaclass=AlgorithmMeta):
             if hasattr(self, opt_name):
                 getattr(self, opt_name).load_state_dict(opt_state)
         for k, v in ckpt["attributes"].items():
-            if k == "device":  # receiving agent keeps its own device
-                continue
+            if k == "device" or k.startswith("_ref_compat"):
+                continue  # receiving agent keeps its own device
             setattr(self, k, v)
 
     @classmethod
     def load(cls, path: str, device: str = "cpu") -> "EvolvableAlgorithm":
-        import dill
+        from ...utils.ref_compat import (
+            convert_reference_checkpoint,
+            is_reference_layout,
+            load_checkpoint_file,
+        )
 
-        ckpt = torch.load(path, pickle_module=dill, weights_only=False, map_location="cpu")
-        algo_cls = ckpt.get("algo_cls", cls)
-        init_args = dict(ckpt["init_args"])
+        ckpt = load_checkpoint_file(path)
+        if is_reference_layout(ckpt):
+            ckpt = convert_reference_checkpoint(ckpt, device=device)
+            algo_cls = ckpt.get("attributes", {}).get("algo_cls")
+            if algo_cls is None:
+                algo_name = ckpt.get("attributes", {}).get("algo")
+                if isinstance(algo_name, str):
+                    from ...utils.ref_compat import _NAME_MAP
+                    import importlib as _il
+
+                    mod = _NAME_MAP.get(algo_name)
+                    algo_cls = getattr(_il.import_module(mod), algo_name) if mod else cls
+                else:
+                    algo_cls = cls
+            # reference checkpoints carry constructor kwargs as flat
+            # attributes; filter against the target constructor
+            import inspect as _inspect
+
+            params = _inspect.signature(algo_cls.__init__).parameters
+            init_args = {
+                k: v for k, v in ckpt["attributes"].items()
+                if k in params and k != "self"
+            }
+        else:
+            algo_cls = ckpt.get("algo_cls", cls)
+            init_args = dict(ckpt["init_args"])
         init_args["device"] = device
         agent = algo_cls(**init_args)
         agent._apply_checkpoint(ckpt)
