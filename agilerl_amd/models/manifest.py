@@ -62,6 +62,9 @@ ALGO_REGISTRY: Dict[str, Dict[str, str]] = {
     "LLM_PPO": {"cls": "agilerl_amd.algorithms.llm.ppo_llm.PPOLLM", "workload": "llm_reasoning"},
     "ReinforceLLM": {"cls": "agilerl_amd.algorithms.llm.reinforce_llm.ReinforceLLM", "workload": "llm_reasoning"},
     "LLM_REINFORCE": {"cls": "agilerl_amd.algorithms.llm.reinforce_llm.ReinforceLLM", "workload": "llm_reasoning"},
+    # reference manifest spellings (configs/training/llm_finetuning/*.yaml)
+    "LLMPPO": {"cls": "agilerl_amd.algorithms.llm.ppo_llm.PPOLLM", "workload": "llm_reasoning"},
+    "LLMREINFORCE": {"cls": "agilerl_amd.algorithms.llm.reinforce_llm.ReinforceLLM", "workload": "llm_reasoning"},
 }
 
 

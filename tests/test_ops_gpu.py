@@ -519,3 +519,94 @@ class TestPagedEngineGpu:
             outs.append([results[s].cpu() for s in sids])
         for a, b in zip(outs[0], outs[1]):
             torch.testing.assert_close(a, b)
+
+
+class TestKernelEdgeCases:
+    """non-divisible vocab, huge row counts,
+    empty/zero masks, non-contiguous inputs, bf16 GAE paths."""
+
+    def test_row_lse_gather_non_divisible_vocab(self):
+        ext = extension()
+        for V in (63, 1001, 32013, 127999):
+            rows = 17
+            logits = torch.randn(rows, V, device=DEV)
+            targets = torch.randint(0, V, (rows,), device=DEV)
+            lp, lse = ext.row_lse_gather(logits, targets, 1.0)
+            ref_lse = torch.logsumexp(logits, dim=-1)
+            ref_lp = logits.gather(1, targets.unsqueeze(1)).squeeze(1) - ref_lse
+            torch.testing.assert_close(lse, ref_lse, rtol=1e-4, atol=1e-4,
+                                       msg=lambda m: f"V={V}: {m}")
+            torch.testing.assert_close(lp, ref_lp, rtol=1e-4, atol=1e-4)
+
+    def test_fused_logprobs_many_rows(self):
+        # >64k rows exercises the chunk loop and grid clamps
+        from agilerl_amd.ops.fused_logprobs import fused_linear_logprobs
+
+        N, H, V = 70_000, 64, 1024
+        h = torch.randn(N, H, device=DEV)
+        w = torch.randn(V, H, device=DEV)
+        t = torch.randint(0, V, (N,), device=DEV)
+        out = fused_linear_logprobs(h, w, t)
+        assert out.shape == (N,)
+        # spot-check a slice against eager
+        sl = slice(65_530, 65_600)
+        logits = h[sl] @ w.t()
+        ref = logits.gather(1, t[sl].unsqueeze(1)).squeeze(1) - torch.logsumexp(logits, -1)
+        torch.testing.assert_close(out[sl], ref, rtol=1e-3, atol=1e-3)
+
+    def test_grpo_loss_empty_mask(self):
+        from agilerl_amd.ops.grpo_loss import grpo_policy_loss
+
+        B, T = 4, 16
+        logp = (torch.randn(B, T, device=DEV) * 0.1).requires_grad_(True)
+        old = logp.detach() + 0.05
+        adv = torch.randn(B, T, device=DEV)
+        mask = torch.zeros(B, T, device=DEV)  # nothing unmasked
+        loss = grpo_policy_loss(logp, old, adv, mask)
+        assert torch.isfinite(loss)
+        assert float(loss) == 0.0
+        loss.backward()
+        assert torch.isfinite(logp.grad).all()
+        assert float(logp.grad.abs().sum()) == 0.0
+
+    def test_gae_scan_non_contiguous_and_bf16(self):
+        T, N = 32, 65
+        base = torch.randn(T, N * 2, device=DEV)
+        rewards = base[:, ::2]  # non-contiguous view
+        assert not rewards.is_contiguous()
+        values = torch.randn(T, N, device=DEV)
+        dones = (torch.rand(T, N, device=DEV) < 0.1).float()
+        last_value = torch.randn(N, device=DEV)
+        adv, ret = ops.gae_scan(rewards, values, dones, last_value, 0.99, 0.95)
+        adv_ref, _ = ops.gae_scan(rewards.cpu(), values.cpu(), dones.cpu(),
+                                  last_value.cpu(), 0.99, 0.95)
+        torch.testing.assert_close(adv.cpu(), adv_ref, rtol=1e-4, atol=1e-5)
+        # bf16 rollout tensors: dispatcher casts to fp32 for the scan
+        adv_bf, ret_bf = ops.gae_scan(
+            rewards.bfloat16(), values.bfloat16(), dones.bfloat16(),
+            last_value.bfloat16(), 0.99, 0.95)
+        assert adv_bf.dtype == torch.float32
+        torch.testing.assert_close(adv_bf.cpu(), adv_ref, rtol=2e-2, atol=2e-2)
+
+    def test_c51_non_contiguous_rewards(self):
+        B, A = 33, 51
+        dist = torch.softmax(torch.randn(B, A, device=DEV), -1)
+        rewards2 = torch.randn(B, 2, device=DEV)
+        rewards = rewards2[:, 0]  # stride-2 view
+        dones = torch.zeros(B, device=DEV)
+        support = torch.linspace(-10, 10, A, device=DEV)
+        out = ops.c51_project(dist, rewards, dones, support, 0.99, -10, 10)
+        ref = ops.c51_project(dist.cpu(), rewards.cpu(), dones.cpu(),
+                              support.cpu(), 0.99, -10, 10)
+        torch.testing.assert_close(out.cpu(), ref, rtol=1e-4, atol=1e-5)
+        torch.testing.assert_close(out.sum(-1).cpu(), torch.ones(B), rtol=1e-4, atol=1e-4)
+
+    def test_group_advantage_single_group_and_group_of_one(self):
+        r = torch.randn(8, device=DEV)
+        # one big group
+        out = ops.group_advantage(r, 8)
+        ref = ops.group_advantage(r.cpu(), 8)
+        torch.testing.assert_close(out.cpu(), ref, rtol=1e-4, atol=1e-5)
+        # degenerate group_size=1: zero advantages, no NaN from std
+        out1 = ops.group_advantage(r, 1)
+        assert torch.isfinite(out1).all()
