@@ -1,8 +1,5 @@
-from .base import LLMEnvBase, make_grpo_experiences
+from .base import HuggingFaceGym, LLMEnvBase, make_grpo_experiences
 from .reasoning import ReasoningGym, TokenReasoningGym
-
-# reference name for the dataset-backed gym (agilerl/llm_envs/base.py:93)
-HuggingFaceGym = ReasoningGym
 from .sft import SFTGym, SyntheticSFTGym
 from .preference import PreferenceGym, SyntheticPreferenceGym
 from .multiturn import (MultiTurnTokenEnv, TokenGuessEnv, SyncMultiTurnVecEnv,

@@ -813,3 +813,110 @@ class TestGrpoParityDeepening:
         assert float(traj["sampling_logps"].abs().sum()) > 0  # engine captured
         stats = agent.learn(traj)
         assert np.isfinite(stats["loss"])
+
+
+def _word_tokenizer(corpus):
+    """Real PreTrainedTokenizerFast trained offline on the corpus."""
+    from tokenizers import Tokenizer, models, pre_tokenizers, trainers
+    from transformers import PreTrainedTokenizerFast
+
+    tok = Tokenizer(models.WordLevel(unk_token="<unk>"))
+    tok.pre_tokenizer = pre_tokenizers.Whitespace()
+    tok.train_from_iterator(
+        corpus, trainers.WordLevelTrainer(special_tokens=["<pad>", "<unk>", "<eos>"])
+    )
+    return PreTrainedTokenizerFast(
+        tokenizer_object=tok, pad_token="<pad>", eos_token="<eos>", unk_token="<unk>"
+    )
+
+
+class TestHuggingFaceGym:
+    """Dataset-backed gym (reference HuggingFaceGym, llm_envs/base.py:93):
+    real datasets.Dataset + real fast tokenizer, epoch dataloaders,
+    train/test split, evaluation mode."""
+
+    def _gym(self, **kw):
+        from datasets import Dataset
+
+        from agilerl_amd.llm_envs import HuggingFaceGym
+
+        train = Dataset.from_dict({
+            "question": [f"what is {i} plus {i}" for i in range(10)],
+            "answer": [str(2 * i) for i in range(10)],
+        })
+        test = Dataset.from_dict({
+            "question": ["what is one plus one"], "answer": ["2"],
+        })
+        corpus = list(train["question"]) + list(train["answer"])
+        tokenizer = _word_tokenizer(corpus)
+
+        def reward(completion, answer):
+            return float(answer in completion)
+
+        kw.setdefault("data_batch_size", 4)
+        kw.setdefault("group_size", 2)
+        return HuggingFaceGym(train, test, tokenizer, reward, **kw), tokenizer
+
+    def test_epoch_iteration_and_shapes(self):
+        gym, tokenizer = self._gym()
+        batch = gym.reset()
+        assert batch["input_ids"].shape[0] == 4 * 2  # batch x group
+        assert batch["attention_mask"].shape == batch["input_ids"].shape
+        # epochs advance after consuming the shard
+        for _ in range(6):
+            gym.reset()
+        assert gym.num_epochs >= 2
+        assert gym.dataset_size == {"train": 10, "test": 1}
+
+    def test_score_with_real_tokenizer_round_trip(self):
+        gym, tokenizer = self._gym(group_size=1, data_batch_size=2)
+        batch = gym.reset()
+        P = batch["input_ids"].shape[1]
+        # append the CORRECT answers as completion tokens
+        answers = gym._batch_answers
+        comp = tokenizer(list(answers), return_tensors="pt", padding=True)["input_ids"]
+        seqs = torch.cat([batch["input_ids"], comp], dim=1)
+        rewards = gym.score(seqs)
+        assert rewards.shape == (2,)
+        assert (rewards == 1.0).all()  # decoded completions contain answers
+
+    def test_evaluation_mode_uses_test_split(self):
+        gym, _ = self._gym(data_batch_size=1, group_size=1)
+        gym.eval(True)
+        gym.reset()
+        assert gym._batch_answers == ["2"]
+        gym.eval(False)
+        gym.reset()
+        assert gym.evaluation_mode is False
+
+    def test_reset_dataloaders_restarts_epochs(self):
+        gym, _ = self._gym()
+        for _ in range(8):
+            gym.reset()
+        assert gym.num_epochs > 0
+        gym.reset(reset_dataloaders=True)
+        assert gym.num_epochs == 0
+
+    def test_grpo_trains_on_hf_gym(self):
+        """End-to-end: tiny llama + real tokenizer vocabulary + dataset gym
+        -> generate + score + learn (real-tokenizer flow, VERDICT r1 #5)."""
+        from agilerl_amd.algorithms.llm.grpo import GRPO
+        from agilerl_amd.llm_envs.base import make_grpo_experiences
+
+        gym, tokenizer = self._gym(group_size=2, data_batch_size=2)
+        vocab = tokenizer.vocab_size + 8
+        tiny = dict(model_type="llama", vocab_size=vocab, hidden_size=32,
+                    intermediate_size=64, num_hidden_layers=1,
+                    num_attention_heads=2, num_key_value_heads=1,
+                    max_position_embeddings=128,
+                    pad_token_id=tokenizer.pad_token_id)
+        agent = GRPO(model_config=tiny, dtype=torch.float32, lora_config={"r": 2},
+                     group_size=2, micro_batch_size=2, max_completion_tokens=4,
+                     tokenizer=tokenizer)
+        prompts = gym.reset()
+        seqs = agent.get_action(prompts)
+        rewards = gym.score(seqs)
+        exp = make_grpo_experiences(gym, seqs, rewards,
+                                    pad_token_id=tokenizer.pad_token_id)
+        stats = agent.learn(exp)
+        assert np.isfinite(stats["loss"])
