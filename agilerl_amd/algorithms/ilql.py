@@ -17,7 +17,7 @@ import torch.nn.functional as F
 from .. import ops
 from ..modules.gpt import EvolvableGPT
 
-__all__ = ["ILQL"]
+__all__ = ["ILQL", "ILQL_Policy", "ILQL_Evaluator"]
 
 
 class ILQL(nn.Module):
@@ -130,9 +130,23 @@ class ILQL(nn.Module):
     # ------------------------------------------------------------------
     @torch.no_grad()
     def generate(
-        self, idx: torch.Tensor, max_new_tokens: int, beta: float = 1.0, temperature: float = 1.0
+        self,
+        idx: torch.Tensor,
+        max_new_tokens: int,
+        beta: float = 1.0,
+        temperature: float = 1.0,
+        top_k=None,
+        top_p=None,
+        eos_token_id=None,
+        greedy: bool = False,
     ) -> torch.Tensor:
-        """Decode with Q-V perturbed logits: logits + beta * (Q - V)."""
+        """Decode with Q-V perturbed logits: ``logits/T + beta * (Q - V)``
+        (reference ilql.py:933-1093 beta-perturbed sampling), with top-k /
+        nucleus processing and EOS early stop."""
+        from ..utils.sampling import process_logits
+
+        B = idx.shape[0]
+        done = torch.zeros(B, dtype=torch.bool, device=idx.device)
         for _ in range(max_new_tokens):
             ctx = idx[:, -self.gpt.max_positions :]
             hidden = self.gpt.transformer_forward(ctx)
@@ -141,6 +155,84 @@ class ILQL(nn.Module):
             q = torch.minimum(self.q1_head(h_last), self.q2_head(h_last))
             v = self.v_head(h_last)
             perturbed = logits / max(temperature, 1e-6) + beta * (q - v)
-            probs = F.softmax(perturbed, dim=-1)
-            idx = torch.cat([idx, torch.multinomial(probs, 1)], dim=1)
+            perturbed = process_logits(perturbed, 1.0, top_k, top_p)
+            if greedy:
+                nxt = perturbed.argmax(-1, keepdim=True)
+            else:
+                nxt = torch.multinomial(F.softmax(perturbed, dim=-1), 1)
+            if eos_token_id is not None:
+                nxt = torch.where(done.unsqueeze(1), torch.full_like(nxt, eos_token_id), nxt)
+                done = done | (nxt.squeeze(1) == eos_token_id)
+            idx = torch.cat([idx, nxt], dim=1)
+            if eos_token_id is not None and bool(done.all()):
+                break
         return idx
+
+
+class ILQL_Policy:
+    """Acting interface over a token env (reference ilql.py:1335 /
+    :2067 ``act``): history tokens in, beta-perturbed completion out."""
+
+    def __init__(self, model: "ILQL", beta: float = 1.0, max_new_tokens: int = 32,
+                 temperature: float = 1.0, top_k=None, top_p=None,
+                 eos_token_id=None, greedy: bool = False):
+        self.model = model
+        self.beta = beta
+        self.max_new_tokens = max_new_tokens
+        self.temperature = temperature
+        self.top_k = top_k
+        self.top_p = top_p
+        self.eos_token_id = eos_token_id
+        self.greedy = greedy
+
+    @torch.no_grad()
+    def act(self, history_tokens: torch.Tensor) -> torch.Tensor:
+        """(B, T) context -> (B, C) generated completion tokens."""
+        if history_tokens.dim() == 1:
+            history_tokens = history_tokens.unsqueeze(0)
+        T = history_tokens.shape[1]
+        full = self.model.generate(
+            history_tokens.to(self.model.device), self.max_new_tokens,
+            beta=self.beta, temperature=self.temperature,
+            top_k=self.top_k, top_p=self.top_p,
+            eos_token_id=self.eos_token_id, greedy=self.greedy,
+        )
+        return full[:, T:]
+
+
+class ILQL_Evaluator:
+    """Rollout scoring for ILQL policies (reference ilql.py:2089).
+
+    ``env`` contract: ``reset() -> (B, T) prompt tokens``;
+    ``score(full_sequences) -> (B,) rewards`` (same shape family as the
+    LLM gyms).  Also reports the policy's value estimates on the prompts.
+    """
+
+    def __init__(self, env, n_batches: int = 1):
+        self.env = env
+        self.n_batches = n_batches
+
+    @torch.no_grad()
+    def evaluate(self, policy: ILQL_Policy) -> Dict[str, float]:
+        import numpy as np
+
+        rewards, values = [], []
+        for _ in range(self.n_batches):
+            prompts = self.env.reset()
+            ids = prompts["input_ids"] if isinstance(prompts, dict) else prompts
+            comp = policy.act(ids)
+            full = torch.cat([ids.to(comp.device), comp], dim=1)
+            r = self.env.score(full)
+            rewards.extend(np.asarray(r, dtype=np.float32).tolist())
+            hidden = policy.model.gpt.transformer_forward(
+                ids.to(policy.model.device)
+            )
+            values.extend(
+                policy.model.v_head(hidden[:, -1]).squeeze(-1).cpu().tolist()
+            )
+        return {
+            "mean_reward": float(np.mean(rewards)),
+            "std_reward": float(np.std(rewards)),
+            "mean_value": float(np.mean(values)),
+            "n": len(rewards),
+        }

@@ -1,6 +1,7 @@
 """EvolvableGPT / EvolvableBERT / ILQL / BC_LM tests."""
 
 import numpy as np
+import pytest
 import torch
 
 from agilerl_amd.algorithms.bc_lm import BC_LM
@@ -107,3 +108,87 @@ class TestBCLM:
         agent = BC_LM(vocab_size=64, n_layer=1, n_head=2, n_embd=16, max_positions=32)
         out = agent.generate(torch.randint(1, 64, (2, 4)), 4, top_p=0.9, top_k=10)
         assert out.shape == (2, 8)
+
+
+class TestSamplingUtils:
+    def test_top_k_masks_tail(self):
+        from agilerl_amd.utils.sampling import process_logits
+
+        logits = torch.tensor([[1.0, 5.0, 3.0, 0.5, 4.0]])
+        out = process_logits(logits, top_k=2)
+        keep = torch.isfinite(out[0])
+        assert keep.tolist() == [False, True, False, False, True]
+
+    def test_top_p_keeps_nucleus(self):
+        from agilerl_amd.utils.sampling import process_logits
+
+        logits = torch.log(torch.tensor([[0.5, 0.3, 0.15, 0.05]]))
+        out = process_logits(logits, top_p=0.7)
+        keep = torch.isfinite(out[0])
+        # 0.5 alone < 0.7, so 0.3 is kept too; 0.15/0.05 dropped
+        assert keep.tolist() == [True, True, False, False]
+        # softmax renormalizes over the kept set
+        probs = torch.softmax(out, dim=-1)[0]
+        assert probs[0].item() == pytest.approx(0.5 / 0.8, rel=1e-5)
+
+    def test_sample_respects_mask(self):
+        from agilerl_amd.utils.sampling import sample_from_logits
+
+        torch.manual_seed(0)
+        logits = torch.tensor([[10.0, -1.0, -1.0, -1.0]]).expand(64, 4)
+        out = sample_from_logits(logits, top_k=1)
+        assert (out == 0).all()
+
+
+class TestILQLPolicyEvaluator:
+    def _model(self):
+        torch.manual_seed(0)
+        from agilerl_amd.algorithms.ilql import ILQL
+
+        return ILQL(vocab_size=32, n_layer=1, n_head=2, n_embd=32,
+                    max_positions=64)
+
+    def test_policy_act_and_eos_stop(self):
+        from agilerl_amd.algorithms.ilql import ILQL_Policy
+
+        model = self._model()
+        pol = ILQL_Policy(model, beta=0.5, max_new_tokens=6, eos_token_id=3,
+                          greedy=True)
+        ctx = torch.randint(4, 32, (2, 5))
+        comp = pol.act(ctx)
+        assert comp.shape[0] == 2 and 1 <= comp.shape[1] <= 6
+        # after an EOS, subsequent tokens are EOS (frozen rows)
+        for row in comp:
+            hit = (row == 3).nonzero()
+            if hit.numel():
+                assert (row[int(hit[0]):] == 3).all()
+
+    def test_evaluator_scores_token_env(self):
+        from agilerl_amd.algorithms.ilql import ILQL_Evaluator, ILQL_Policy
+
+        class CountTargetEnv:
+            def reset(self):
+                return torch.randint(4, 32, (3, 4))
+
+            def score(self, full):
+                return (full[:, 4:] == 7).float().mean(dim=1).numpy()
+
+        model = self._model()
+        pol = ILQL_Policy(model, max_new_tokens=5)
+        ev = ILQL_Evaluator(CountTargetEnv(), n_batches=2)
+        stats = ev.evaluate(pol)
+        assert set(stats) == {"mean_reward", "std_reward", "mean_value", "n"}
+        assert stats["n"] == 6
+        assert 0.0 <= stats["mean_reward"] <= 1.0
+
+    def test_beta_changes_decode_distribution(self):
+        """beta=0 ignores Q-V; large beta shifts decoding toward high-Q
+        tokens — distributions must differ."""
+        model = self._model()
+        torch.manual_seed(1)
+        ctx = torch.randint(4, 32, (8, 6))
+        torch.manual_seed(2)
+        out0 = model.generate(ctx, 8, beta=0.0)
+        torch.manual_seed(2)
+        outb = model.generate(ctx, 8, beta=50.0)
+        assert not torch.equal(out0, outb)
