@@ -283,6 +283,7 @@ class GrpoRolloutBenchRunner(GrpoBenchRunner):
         self.gen_seconds = 0.0
 
     def bench_step(self) -> int:
+        import sys as _sys
         import time as _t
 
         B, P = self.args.grpo_batch, self.prompt_len
@@ -294,6 +295,12 @@ class GrpoRolloutBenchRunner(GrpoBenchRunner):
             torch.cuda.synchronize()
         self.gen_seconds += _t.perf_counter() - t0
         self.gen_tokens += B * self.completion_len
+        eng = getattr(self.agent, "_decode_engine", None)
+        if eng is not None and not getattr(self, "_graph_reported", False):
+            self._graph_reported = True
+            graphs = {k: d.graph is not None for k, d in eng._graph_decoders.items()}
+            print(f"[bench] decode graphs: {graphs} use_graph={eng._use_decode_graph}",
+                  file=_sys.stderr, flush=True)
         T = seqs.shape[1]
         pos = torch.arange(T - 1, device=self.device).unsqueeze(0)
         action_mask = (pos + 1 >= P).float().expand(B, T - 1)
