@@ -64,30 +64,101 @@ class Deployment:
 
 
 class ArenaClient:
+    """Two backends behind one API (reference client.py:128):
+
+    - ``base_url=None``: LOCAL workspace store — experiments run through
+      LocalTrainer in-process.
+    - ``base_url="http://..."``: HTTP transport to an
+      :mod:`agilerl_amd.arena.service` deployment (httpx); ``http_client``
+      injects a preconfigured client (e.g. an in-process ASGI transport
+      in tests).
+    """
+
     def __init__(
         self,
         base_url: Optional[str] = None,
         workspace: str = ".arena",
         api_key: Optional[str] = None,
+        http_client=None,
     ):
-        self.base_url = base_url
+        self.base_url = base_url.rstrip("/") if base_url else None
         self.workspace = workspace
         self.api_key = api_key or os.environ.get("ARENA_API_KEY")
         self._logged_in = False
+        self._stream_handler = None
+        self._http = http_client
+        if self.base_url is not None and self._http is None:
+            import httpx
+
+            self._http = httpx.Client(base_url=self.base_url, timeout=60.0)
         os.makedirs(workspace, exist_ok=True)
+
+    # ------------------------------------------------------------------
+    @property
+    def remote(self) -> bool:
+        return self.base_url is not None or self._http is not None
+
+    def _get(self, path: str, **kw):
+        r = self._http.get(path, **kw)
+        if r.status_code >= 400:
+            raise ArenaError(f"GET {path} -> {r.status_code}: {r.text[:300]}")
+        return r.json()
+
+    def _post(self, path: str, **kw):
+        r = self._http.post(path, **kw)
+        if r.status_code >= 400:
+            raise ArenaError(f"POST {path} -> {r.status_code}: {r.text[:300]}")
+        return r.json()
 
     # ------------------------------------------------------------------
     # Auth (device-flow shape; offline backend auto-authorizes)
     # ------------------------------------------------------------------
     def login(self, interactive: bool = True) -> bool:
-        if self.base_url is not None:
-            raise ArenaError(
-                "remote Arena transport is not available in this offline build; "
-                "use the local workspace backend (base_url=None)"
-            )
+        if self.remote:
+            tok = self._post("/auth/device")
+            self.api_key = tok["access_token"]
+            self._logged_in = True
+            return True
         self._logged_in = True
         self._write("auth.json", {"logged_in_at": time.time()})
         return True
+
+    # ------------------------------------------------------------------
+    # Streaming (reference client.py:342 set_stream_handler /
+    # _open_stream over NDJSON)
+    # ------------------------------------------------------------------
+    def set_stream_handler(self, handler) -> None:
+        """Register a callback invoked for each StreamEvent while
+        streaming experiment progress."""
+        self._stream_handler = handler
+
+    def stream_experiment(self, experiment_id: str, follow: float = 0.0):
+        """Yield StreamEvents for an experiment (remote backend); also
+        dispatches to the registered stream handler."""
+        from .stream import NDJsonStream
+
+        if not self.remote:
+            raise ArenaError("streaming requires the HTTP backend")
+        self._require_auth()
+        with self._http.stream(
+            "GET", f"/experiments/{experiment_id}/stream", params={"follow": follow}
+        ) as r:
+            if r.status_code >= 400:
+                raise ArenaError(f"stream -> {r.status_code}")
+            for event in NDJsonStream(r.iter_lines()):
+                if self._stream_handler is not None:
+                    self._stream_handler(event)
+                yield event
+
+    def wait_for_completion(self, experiment_id: str, timeout: float = 300.0,
+                            poll: float = 0.2) -> Dict[str, Any]:
+        deadline = time.time() + timeout
+        while time.time() < deadline:
+            st = self.experiment_status(experiment_id)
+            if st.get("status") in ("completed", "failed"):
+                return st
+            time.sleep(poll)
+        raise ArenaError(f"experiment {experiment_id} did not finish in {timeout}s")
 
     def _require_auth(self):
         if not self._logged_in:
@@ -98,6 +169,8 @@ class ArenaClient:
     # ------------------------------------------------------------------
     def validate_environment(self, env_spec: Dict[str, Any]) -> Dict[str, Any]:
         """Structural validation matching the reference's pre-submission check."""
+        if self.remote:
+            return self._post("/environments/validate", json=env_spec)
         manifest = TrainingManifest.model_validate(
             {"algorithm": {"name": env_spec.get("algorithm", "DQN")}, "environment": env_spec}
         )
@@ -124,6 +197,12 @@ class ArenaClient:
         self._require_auth()
         if isinstance(manifest, dict):
             manifest = TrainingManifest.model_validate(manifest)
+        if self.remote:
+            out = self._post("/experiments", json={
+                "manifest": manifest.model_dump(mode="json"), "device": device,
+            })
+            return ExperimentHandle(out["experiment_id"], status=out["status"],
+                                    manifest=manifest.model_dump())
         exp_id = f"exp-{uuid.uuid4().hex[:12]}"
         exp_dir = os.path.join(self.workspace, exp_id)
         os.makedirs(exp_dir, exist_ok=True)
@@ -158,6 +237,8 @@ class ArenaClient:
         return self.submit_experiment(manifest, run=True, device=device)
 
     def experiment_status(self, experiment_id: str) -> Dict[str, Any]:
+        if self.remote:
+            return self._get(f"/experiments/{experiment_id}")
         path = os.path.join(self.workspace, experiment_id, "status.json")
         if not os.path.exists(path):
             raise ArenaError(f"unknown experiment {experiment_id}")
@@ -165,6 +246,8 @@ class ArenaClient:
             return json.load(f)
 
     def list_experiments(self) -> List[str]:
+        if self.remote:
+            return self._get("/experiments")["experiments"]
         return sorted(
             d for d in os.listdir(self.workspace)
             if d.startswith("exp-") and os.path.isdir(os.path.join(self.workspace, d))
@@ -172,6 +255,8 @@ class ArenaClient:
 
     def list_checkpoints(self, experiment_id: str) -> List[str]:
         self._require_auth()
+        if self.remote:
+            return self._get(f"/experiments/{experiment_id}/checkpoints")["checkpoints"]
         exp_dir = os.path.join(self.workspace, experiment_id)
         if not os.path.isdir(exp_dir):
             raise ArenaError(f"unknown experiment {experiment_id}")
@@ -189,6 +274,16 @@ class ArenaClient:
         if not os.path.exists(path):
             raise ArenaError(f"dataset path not found: {path}")
         name = name or os.path.basename(path).split(".")[0]
+        if self.remote:
+            if os.path.isdir(path):
+                raise ArenaError("remote dataset upload takes a single file")
+            with open(path, "rb") as f:
+                out = self._post(
+                    "/datasets",
+                    params={"name": name, "ext": os.path.splitext(path)[1]},
+                    content=f.read(),
+                )
+            return out["dataset_id"]
         ds_id = f"ds-{name}"
         ds_dir = os.path.join(self.workspace, "datasets")
         os.makedirs(ds_dir, exist_ok=True)
@@ -207,6 +302,8 @@ class ArenaClient:
         return ds_id
 
     def list_datasets(self) -> List[str]:
+        if self.remote:
+            return self._get("/datasets")["datasets"]
         ds_dir = os.path.join(self.workspace, "datasets")
         if not os.path.isdir(ds_dir):
             return []
@@ -228,6 +325,24 @@ class ArenaClient:
         Deployment with .predict()/.info(); host it over HTTP with
         `python -m agilerl_amd.serve <ckpt>`."""
         self._require_auth()
+        if self.remote:
+            # fetch the checkpoint bytes into the local workspace first
+            ckpts = self.list_checkpoints(experiment_id)
+            if checkpoint is None:
+                if not ckpts:
+                    raise ArenaError(f"no checkpoints in {experiment_id}")
+                checkpoint = ckpts[-1]
+            r = self._http.get(f"/experiments/{experiment_id}/checkpoints/{checkpoint}")
+            if r.status_code >= 400:
+                raise ArenaError(f"checkpoint fetch -> {r.status_code}")
+            local_dir = os.path.join(self.workspace, experiment_id)
+            os.makedirs(local_dir, exist_ok=True)
+            local_path = os.path.join(local_dir, checkpoint)
+            with open(local_path, "wb") as f:
+                f.write(r.content)
+            from ..serve import load_agent
+
+            return Deployment(experiment_id, checkpoint, load_agent(local_path))
         exp_dir = os.path.join(self.workspace, experiment_id)
         ckpts = self.list_checkpoints(experiment_id)
         if checkpoint is None:

@@ -217,6 +217,9 @@ class TrainingManifest(_Base):
 
     def env_spec(self) -> EnvSpec:
         env = dict(self.environment)
+        # reference manifests key the env as `name:` (configs/training/*)
+        if "env_id" not in env and "name" in env:
+            env["env_id"] = env.pop("name")
         etype = env.get("type")
         if etype is None:
             workload = algo_workload(self.algorithm.name)

@@ -160,6 +160,7 @@ class TestReferenceManifests:
                 doc = _yaml.safe_load(open(f))
                 manifest = TrainingManifest.model_validate(doc)
                 resolve_algo_class(manifest.algorithm.name)
+                manifest.env_spec()  # `name:`-keyed env sections must resolve
             except Exception as e:  # noqa: BLE001 - collect all failures
                 failures.append(f"{f}: {e}")
         assert not failures, "\n".join(failures)
