@@ -246,6 +246,14 @@ class DecodeEngine:
 
         dec = self._graph_decoder(adapter, len(seqs))
         if dec is not None:
+            # split sizing: the longest length any active sequence can reach
+            hint = max(
+                self.cache.lengths[sid] + s.max_new_tokens - len(s.generated)
+                for s, sid in zip(seqs, seq_ids)
+            )
+            if hint > dec.len_hint:
+                dec.len_hint = int(hint)
+                dec.graph = None  # recapture with the wider split config
             for s, sid in zip(seqs, seq_ids):
                 if sid not in dec.slot_of:
                     slot = dec.acquire_slot(sid)
@@ -268,7 +276,8 @@ class DecodeEngine:
                                   device=self.device)
             positions = torch.tensor([self.cache.lengths[sid] for sid in seq_ids],
                                      dtype=torch.long, device=self.device)
-            logits = runner.decode_step(tokens, positions, table)
+            hint = int(max(self.cache.lengths[sid] for sid in seq_ids)) + 1
+            logits = runner.decode_step(tokens, positions, table, max_len_hint=hint)
             sel = self._select_batch(logits, seqs)
             lps_host = self._chosen_logps(logits, sel, seqs).tolist()
             sel_host = sel.tolist()

@@ -60,6 +60,7 @@ class PagedLlamaDecodeRunner:
         tokens: torch.Tensor,      # (B,) last emitted token per sequence
         positions: torch.Tensor,   # (B,) position of the NEW token (== length)
         table: torch.Tensor,       # (B, max_pages) int32 page table
+        max_len_hint: int = 0,     # host-side length upper bound (split sizing)
     ) -> torch.Tensor:             # (B, V) next-token logits
         from transformers.models.llama.modeling_llama import apply_rotary_pos_emb
 
@@ -99,6 +100,7 @@ class PagedLlamaDecodeRunner:
             attn_out = paged_attention_decode(
                 q.to(pool_dtype) if pool_dtype == torch.bfloat16 else q,
                 cache.k_pool[li], cache.v_pool[li], table, lengths, self.scale,
+                max_len_hint=max_len_hint,
             )
             attn_out = attn_out.to(x.dtype).view(B, 1, Hq * D)
             x = x + attn.o_proj(attn_out)
@@ -154,6 +156,7 @@ class GraphedPagedDecoder:
         self.table = torch.full((B, maxp), scratch_page, dtype=torch.int32, device=device)
         self.logits: torch.Tensor = None
         self.graph = None
+        self.len_hint = 0              # max expected sequence length (splits)
         self.slot_of = {}              # seq_id -> row
         self._free_slots = list(range(B - 1, -1, -1))
         self._pages_synced = [0] * B   # table entries filled per row
@@ -202,12 +205,14 @@ class GraphedPagedDecoder:
         s.wait_stream(torch.cuda.current_stream())
         with torch.cuda.stream(s):
             for _ in range(3):  # warmup allocations on a side stream
-                self.runner.decode_step(self.tokens, self.positions, self.table)
+                self.runner.decode_step(self.tokens, self.positions, self.table,
+                                        max_len_hint=self.len_hint)
         torch.cuda.current_stream().wait_stream(s)
         torch.cuda.synchronize()
         self.graph = torch.cuda.CUDAGraph()
         with torch.cuda.graph(self.graph):
-            self.logits = self.runner.decode_step(self.tokens, self.positions, self.table)
+            self.logits = self.runner.decode_step(
+                self.tokens, self.positions, self.table, max_len_hint=self.len_hint)
         torch.cuda.synchronize()
 
     @torch.no_grad()

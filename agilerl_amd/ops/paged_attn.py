@@ -44,11 +44,15 @@ def paged_attention_decode(
     page_table: torch.Tensor,
     lengths: torch.Tensor,
     scale: float = None,
+    max_len_hint: int = 0,
 ) -> torch.Tensor:
     """Single-token decode attention over paged K/V.
 
     q: (B, Hq, D); k_pool/v_pool: (num_pages, page_size, Hkv, D);
     page_table: (B, max_pages) int; lengths: (B,) int.
+    ``max_len_hint``: host-side upper bound on the batch's sequence
+    lengths; sizes the flash-decoding split count (lengths are device-
+    resident, and the table width badly overestimates short decodes).
     Returns (B, Hq, D) fp32.
     """
     if scale is None:
@@ -56,5 +60,6 @@ def paged_attention_decode(
     ext = extension()
     if use_hip(q) and ext is not None and hasattr(ext, "paged_attn_decode") \
             and q.dtype == torch.bfloat16:
-        return ext.paged_attn_decode(q, k_pool, v_pool, page_table, lengths, float(scale))
+        return ext.paged_attn_decode(q, k_pool, v_pool, page_table, lengths,
+                                     float(scale), int(max_len_hint))
     return _eager_reference(q, k_pool, v_pool, page_table, lengths, scale)

@@ -77,12 +77,14 @@ class TestPPOGraphMutationSweep:
         eager.learn(dict(flat))
         _assert_params_close(graphed, eager)
 
-        # identical architecture mutation on both (new-node init draws RNG,
-        # so reseed per agent for identical fresh weights)
+        # identical architecture mutation on both: pin BOTH the sampled
+        # layer choice (add_node draws it from a non-torch RNG) and the
+        # torch seed (new-node weight init)
         for agent in (graphed, eager):
             torch.manual_seed(123)
             torch.cuda.manual_seed_all(123)
-            agent.apply_architecture_mutation("encoder.add_node", numb_new_nodes=16)
+            agent.apply_architecture_mutation(
+                "encoder.add_node", hidden_layer=0, numb_new_nodes=16)
         assert graphed._learn_graph is None, "mutation hook must clear the graph"
         flat2 = _flat_rollout(self.N, 8, 4, seed=2)
         graphed.learn(dict(flat2))   # recapture with the mutated architecture
@@ -190,6 +192,8 @@ class TestDQNGraphMutationSweep:
             torch.manual_seed(321)
             torch.cuda.manual_seed_all(321)
             agent.apply_architecture_mutation("encoder.add_layer")
+        assert [tuple(p.shape) for p in graphed.actor.parameters()] == \
+            [tuple(p.shape) for p in eager.actor.parameters()]
         for seed in (3, 4):
             graphed.learn(self._batch(seed))
             eager.learn(self._batch(seed))

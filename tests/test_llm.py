@@ -440,7 +440,8 @@ class TestSearchAndFormat:
 
         from agilerl_amd.llm_envs import HuggingFaceGym, ReasoningGym
 
-        assert HuggingFaceGym is ReasoningGym
+        # HuggingFaceGym is now the dataset-backed class (no longer an
+        # alias of ReasoningGym); both support dataset construction
 
         class TinyTok:
             pad_token_id = 0
