@@ -19,10 +19,42 @@ from __future__ import annotations
 from typing import List
 
 import torch
+import torch.nn.functional as F
 
+from ..ops.backend import extension, use_hip
 from ..ops.paged_attn import paged_attention_decode
 
 __all__ = ["PagedLlamaDecodeRunner", "is_paged_decodable"]
+
+
+def _fast_linear(module, x2d: torch.Tensor) -> torch.Tensor:
+    """Decode-shape projection: route the big weight stream through the
+    skinny-GEMM kernel (ops/csrc/skinny_gemm.hip) instead of hipBLASLt's
+    tile kernels (~0.6-1 TB/s at M<=16), keeping any LoRA delta as the
+    small eager side-path exactly like LoraLinear.forward."""
+    ext = extension()
+    eligible = (
+        ext is not None
+        and hasattr(ext, "skinny_gemm")
+        and use_hip(x2d)
+        and x2d.dtype == torch.bfloat16
+        and x2d.shape[0] <= 16
+        and x2d.shape[1] % 8 == 0
+    )
+    base = getattr(module, "base", module)  # LoraLinear wraps .base
+    if (
+        not eligible
+        or not isinstance(base, torch.nn.Linear)
+        or base.bias is not None
+        or base.weight.dtype != torch.bfloat16
+    ):
+        return module(x2d)
+    out = ext.skinny_gemm(x2d, base.weight)
+    name = getattr(module, "active_adapter", None)
+    if name is not None and name in getattr(module, "lora_A", ()):
+        h = module.dropout(x2d) if getattr(module, "dropout", None) is not None else x2d
+        out = out + F.linear(F.linear(h, module.lora_A[name]), module.lora_B[name]) * module.scaling
+    return out
 
 
 def is_paged_decodable(model) -> bool:
@@ -86,9 +118,10 @@ class PagedLlamaDecodeRunner:
         for li, layer in enumerate(base.layers):
             h = layer.input_layernorm(x)
             attn = layer.self_attn
-            q = attn.q_proj(h).view(B, Hq, D)
-            k = attn.k_proj(h).view(B, Hkv, D)
-            v = attn.v_proj(h).view(B, Hkv, D)
+            h2 = h.view(B, -1)
+            q = _fast_linear(attn.q_proj, h2).view(B, Hq, D)
+            k = _fast_linear(attn.k_proj, h2).view(B, Hkv, D)
+            v = _fast_linear(attn.v_proj, h2).view(B, Hkv, D)
             qr, kr = apply_rotary_pos_emb(
                 q.unsqueeze(2), k.unsqueeze(2), cos, sin, unsqueeze_dim=1
             )
@@ -102,11 +135,23 @@ class PagedLlamaDecodeRunner:
                 cache.k_pool[li], cache.v_pool[li], table, lengths, self.scale,
                 max_len_hint=max_len_hint,
             )
-            attn_out = attn_out.to(x.dtype).view(B, 1, Hq * D)
-            x = x + attn.o_proj(attn_out)
-            x = x + layer.mlp(layer.post_attention_layernorm(x))
+            attn_out = attn_out.to(x.dtype).view(B, Hq * D)
+            x = x + _fast_linear(attn.o_proj, attn_out).view(B, 1, -1)
+            x = x + self._mlp(layer.mlp, layer.post_attention_layernorm(x), B)
         x = base.norm(x)
-        return model.lm_head(x[:, -1])
+        return _fast_linear(model.lm_head, x[:, -1])
+
+    def _mlp(self, mlp, h: torch.Tensor, B: int) -> torch.Tensor:
+        """Llama MLP with skinny-GEMM projections + fused SwiGLU when the
+        module has the standard gate/up/down layout."""
+        if all(hasattr(mlp, a) for a in ("gate_proj", "up_proj", "down_proj")):
+            from ..ops.swiglu import swiglu
+
+            h2 = h.view(B, -1)
+            gate = _fast_linear(mlp.gate_proj, h2)
+            up = _fast_linear(mlp.up_proj, h2)
+            return _fast_linear(mlp.down_proj, swiglu(gate, up)).view(B, 1, -1)
+        return mlp(h)
 
     # ------------------------------------------------------------------
     def build_table(self, seq_ids: List[int], device: str) -> torch.Tensor:

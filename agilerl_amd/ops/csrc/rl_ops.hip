@@ -464,12 +464,14 @@ void init_lm_ops(pybind11::module_& m);    // lm_ops.hip
 void init_norm_ops(pybind11::module_& m);  // norm_ops.hip
 void init_act_ops(pybind11::module_& m);   // act_ops.hip
 void init_paged_attn(pybind11::module_& m);  // paged_attn.hip
+void init_skinny_gemm(pybind11::module_& m);  // skinny_gemm.hip
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   init_lm_ops(m);
   init_norm_ops(m);
   init_act_ops(m);
   init_paged_attn(m);
+  init_skinny_gemm(m);
   m.def("gae_scan", &gae_scan, "GAE reverse scan (T,N)");
   m.def("nstep_scan", &nstep_scan, "n-step returns over sampled windows");
   m.def("c51_project", &c51_project, "C51 categorical projection");

@@ -89,6 +89,27 @@ def main():
     rows.append(("paged_attn_decode", f"B{B} Hq{Hq} len{int(lengths[0])}",
                  timeit(lambda: paged_attention_decode(q, kp, vp, table, lengths),
                         args.iters, cuda)))
+    rows.append(("paged_attn(hint)", f"B{B} Hq{Hq} len{int(lengths[0])}",
+                 timeit(lambda: paged_attention_decode(
+                     q, kp, vp, table, lengths, max_len_hint=int(lengths[0])),
+                        args.iters, cuda)))
+
+    # skinny decode GEMM vs hipBLASLt at M=8 (Llama-3-8B projection shapes)
+    if cuda:
+        from agilerl_amd.ops.backend import extension as _ext
+
+        ext = _ext()
+        for (M, N, K) in [(8, 4096, 4096), (8, 1024, 4096), (8, 14336, 4096),
+                          (8, 4096, 14336), (8, 128256, 4096)]:
+            x = torch.randn(M, K, device=dev, dtype=torch.bfloat16)
+            w = torch.randn(N, K, device=dev, dtype=torch.bfloat16)
+            gb = N * K * 2 / 1e9
+            ms_sg = timeit(lambda: ext.skinny_gemm(x, w), args.iters, True)
+            ms_bl = timeit(lambda: x @ w.t(), args.iters, True)
+            rows.append((f"skinny_gemm", f"({M},{K})x({N},{K})T {gb:.2f}GB",
+                         ms_sg))
+            rows.append((f"skinny/blaslt", f"-> {gb/ms_sg*1000:.0f} vs {gb/ms_bl*1000:.0f} GB/s",
+                         ms_bl))
 
     print(f"{'op':<22} {'shape':<28} {'ms':>8}")
     for name, shape, ms in rows:

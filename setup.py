@@ -20,6 +20,7 @@ ext = CUDAExtension(
     name="agilerl_amd.ops._hip_ops",
     sources=[
         "agilerl_amd/ops/csrc/rl_ops.hip",
+        "agilerl_amd/ops/csrc/skinny_gemm.hip",
         "agilerl_amd/ops/csrc/lm_ops.hip",
         "agilerl_amd/ops/csrc/norm_ops.hip",
         "agilerl_amd/ops/csrc/act_ops.hip",

@@ -610,3 +610,51 @@ class TestKernelEdgeCases:
         # degenerate group_size=1: zero advantages, no NaN from std
         out1 = ops.group_advantage(r, 1)
         assert torch.isfinite(out1).all()
+
+
+class TestSkinnyGemm:
+    """Decode-shape GEMM kernel vs torch matmul."""
+
+    @pytest.mark.parametrize("M,N,K", [
+        (1, 4096, 4096), (8, 1024, 4096), (8, 4096, 4096),
+        (16, 14336, 4096), (8, 4096, 14336), (3, 129, 64),
+    ])
+    def test_matches_matmul(self, M, N, K):
+        ext = extension()
+        g = torch.Generator().manual_seed(M + N + K)
+        x = (torch.randn(M, K, generator=g) * 0.2).bfloat16().to(DEV)
+        w = (torch.randn(N, K, generator=g) * 0.2).bfloat16().to(DEV)
+        got = ext.skinny_gemm(x, w)
+        want = (x.float() @ w.float().t())
+        torch.testing.assert_close(got.float(), want, rtol=2e-2, atol=2e-1)
+
+    def test_decode_runner_uses_it(self):
+        """_fast_linear must route through the kernel (same numerics as the
+        module) for bf16 M<=16 bias-free projections."""
+        import torch.nn as nn
+
+        from agilerl_amd.llm.paged_llama import _fast_linear
+
+        lin = nn.Linear(4096, 1024, bias=False).bfloat16().to(DEV)
+        x = torch.randn(8, 4096).bfloat16().to(DEV)
+        out = _fast_linear(lin, x)
+        ref = lin(x)
+        torch.testing.assert_close(out.float(), ref.float(), rtol=2e-2, atol=2e-1)
+
+    def test_lora_delta_preserved(self):
+        import torch.nn as nn
+
+        from agilerl_amd.llm.lora import LoraConfig, LoraLinear
+        from agilerl_amd.llm.paged_llama import _fast_linear
+
+        base = nn.Linear(512, 256, bias=False).bfloat16().to(DEV)
+        mod = LoraLinear(base, LoraConfig(r=8, lora_alpha=16), adapters=("self",))
+        mod = mod.to(DEV).bfloat16()
+        with torch.no_grad():
+            mod.lora_B["self"].normal_(0, 0.05)  # nonzero delta
+        mod.active_adapter = "self"
+        x = torch.randn(4, 512).bfloat16().to(DEV)
+        out = _fast_linear(mod, x)
+        ref = mod(x)
+        torch.testing.assert_close(out.float(), ref.float(), rtol=3e-2, atol=3e-1)
+        assert not torch.allclose(out.float(), base(x).float(), rtol=1e-3, atol=1e-3)
