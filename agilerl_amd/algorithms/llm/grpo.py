@@ -61,7 +61,7 @@ class GRPO(LLMAlgorithm):
         max_completion_tokens: int = 256,
         dtype: torch.dtype = torch.bfloat16,
         gradient_checkpointing: bool = False,
-        use_packing: bool = False,
+        use_packing="auto",
         generation: str = "hf",
         device: str = "cpu",
     ):
@@ -75,9 +75,13 @@ class GRPO(LLMAlgorithm):
         )
         self.group_size = int(group_size)
         self.update_epochs = int(update_epochs)
-        # padding-free grad/old-policy passes (compute_logprobs_packed);
-        # opt-in until validated at 8-GPU scale
-        self.use_packing = bool(use_packing)
+        # padding-free grad/old-policy passes (compute_logprobs_packed):
+        # True / False / "auto" (default) — auto packs whenever the batch
+        # carries meaningful padding (>10% pad tokens), since packing only
+        # pays when there are pads to skip (GPU-equivalence tested)
+        if use_packing not in (True, False, "auto"):
+            raise ValueError(f"use_packing must be bool or 'auto', got {use_packing!r}")
+        self.use_packing = use_packing
         # "hf" = model.generate (default); "paged" = continuous-batching
         # paged-KV engine (llm/decode_engine.py, greedy-parity tested)
         if generation not in ("hf", "paged"):
@@ -165,7 +169,11 @@ class GRPO(LLMAlgorithm):
 
         B = ids.shape[0]
         mb = max(self.micro_batch_size, 1)
-        logprob_fn = self.compute_logprobs_packed if self.use_packing else self.compute_logprobs
+        packing = self.use_packing
+        if packing == "auto":
+            pad_frac = 1.0 - attention_mask.float().mean().item()
+            packing = pad_frac > 0.10
+        logprob_fn = self.compute_logprobs_packed if packing else self.compute_logprobs
 
         # old-policy + reference logprobs (no grad, micro-batched)
         old_logp = torch.empty(action_mask.shape, device=self.device)

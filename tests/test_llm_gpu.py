@@ -110,3 +110,52 @@ class TestHipRMSNorm:
         with torch.no_grad():
             out = m(input_ids=ids).logits.float()
         torch.testing.assert_close(out, ref, rtol=5e-2, atol=5e-1)
+
+
+class TestPackingEquivalenceGpu:
+    def test_packed_matches_padded_bf16(self):
+        """compute_logprobs_packed == compute_logprobs on real positions for
+        a ragged bf16 batch on GPU (packing is default-'auto' now)."""
+        from agilerl_amd.algorithms.llm.grpo import GRPO
+
+        tiny = dict(model_type="llama", vocab_size=128, hidden_size=64,
+                    intermediate_size=128, num_hidden_layers=2,
+                    num_attention_heads=4, num_key_value_heads=2,
+                    max_position_embeddings=256, pad_token_id=0)
+        torch.manual_seed(0)
+        agent = GRPO(model_config=tiny, dtype=torch.bfloat16,
+                     lora_config={"r": 4}, device="cuda:0")
+        B, T = 6, 24
+        g = torch.Generator().manual_seed(1)
+        ids = torch.randint(1, 128, (B, T), generator=g).to("cuda:0")
+        mask = torch.ones_like(ids)
+        for i, pad in enumerate([0, 3, 7, 11, 5, 16]):  # ragged left pads
+            ids[i, :pad] = 0
+            mask[i, :pad] = 0
+        lp_pad = agent.compute_logprobs(ids, mask)
+        lp_pack = agent.compute_logprobs_packed(ids, mask)
+        real = mask[:, 1:].bool() & mask[:, :-1].bool()
+        torch.testing.assert_close(
+            lp_pack[real].float(), lp_pad[real].float(), rtol=5e-2, atol=5e-1)
+
+    def test_auto_packing_selects_by_pad_fraction(self):
+        from agilerl_amd.algorithms.llm.grpo import GRPO
+
+        tiny = dict(model_type="llama", vocab_size=64, hidden_size=32,
+                    intermediate_size=64, num_hidden_layers=1,
+                    num_attention_heads=2, num_key_value_heads=1,
+                    max_position_embeddings=128, pad_token_id=0)
+        agent = GRPO(model_config=tiny, dtype=torch.bfloat16,
+                     lora_config={"r": 2}, device="cuda:0", group_size=2)
+        assert agent.use_packing == "auto"
+        # heavily padded batch learns fine through the packed path
+        ids = torch.randint(1, 64, (4, 24)).to("cuda:0")
+        mask = torch.ones_like(ids)
+        mask[:, :14] = 0
+        ids[:, :14] = 0
+        pos = torch.arange(23, device="cuda:0").unsqueeze(0)
+        action_mask = ((pos + 1 >= 18) & (mask[:, 1:].bool())).float().expand(4, 23)
+        stats = agent.learn({"ids": ids, "attention_mask": mask,
+                             "action_mask": action_mask,
+                             "rewards": torch.rand(4)})
+        assert np.isfinite(stats["loss"])
