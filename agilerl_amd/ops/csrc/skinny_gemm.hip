@@ -35,32 +35,43 @@ __device__ inline void sg_load8(const bf16* __restrict__ p, float* f) {
   for (int i = 0; i < 8; ++i) f[i] = __uint_as_float(((unsigned int)v[i]) << 16);
 }
 
+template <int MT>
 __global__ void __launch_bounds__(WAVE * SG_WAVES)
 skinny_gemm_kernel(
-    const bf16* __restrict__ x,   // (M, K)
+    const bf16* __restrict__ x,   // (MT, K)
     const bf16* __restrict__ w,   // (N, K)
-    bf16* __restrict__ out,       // (M, N)
-    int M, long N, int K) {
+    bf16* __restrict__ out,       // (MT, N)
+    long N, int K) {
   int wave = threadIdx.x / WAVE;
   int lane = threadIdx.x % WAVE;
   long n0 = ((long)blockIdx.x * SG_WAVES + wave) * SG_ROWS;
   if (n0 >= N) return;
-  int rows = (int)min((long)SG_ROWS, N - n0);
+  bool full = (n0 + SG_ROWS) <= N;
 
-  float acc[SG_MAX_M][SG_ROWS];
+  // compile-time MT => acc and the x slice stay in registers (a runtime
+  // M loop indexes the array dynamically and spills to scratch — measured
+  // 315 GB/s before templating)
+  float acc[MT][SG_ROWS];
   #pragma unroll
-  for (int m = 0; m < SG_MAX_M; ++m)
+  for (int m = 0; m < MT; ++m)
     #pragma unroll
     for (int r = 0; r < SG_ROWS; ++r) acc[m][r] = 0.f;
 
   for (int k0 = lane * 8; k0 < K; k0 += WAVE * 8) {
     float wf[SG_ROWS][8];
-    for (int r = 0; r < rows; ++r)
-      sg_load8(w + (n0 + r) * K + k0, wf[r]);
-    for (int m = 0; m < M; ++m) {
+    if (full) {
+      #pragma unroll
+      for (int r = 0; r < SG_ROWS; ++r) sg_load8(w + (n0 + r) * K + k0, wf[r]);
+    } else {
+      for (int r = 0; r < SG_ROWS; ++r)
+        if (n0 + r < N) sg_load8(w + (n0 + r) * K + k0, wf[r]);
+    }
+    #pragma unroll
+    for (int m = 0; m < MT; ++m) {
       float xf[8];
       sg_load8(x + (long)m * K + k0, xf);
-      for (int r = 0; r < rows; ++r) {
+      #pragma unroll
+      for (int r = 0; r < SG_ROWS; ++r) {
         float part = 0.f;
         #pragma unroll
         for (int e = 0; e < 8; ++e) part += xf[e] * wf[r][e];
@@ -69,13 +80,14 @@ skinny_gemm_kernel(
     }
   }
 
-  // reduce each (m, r) partial across the wave
-  for (int m = 0; m < M; ++m) {
-    for (int r = 0; r < rows; ++r) {
+  #pragma unroll
+  for (int m = 0; m < MT; ++m) {
+    #pragma unroll
+    for (int r = 0; r < SG_ROWS; ++r) {
       float v = acc[m][r];
       #pragma unroll
       for (int off = WAVE / 2; off > 0; off >>= 1) v += __shfl_xor(v, off);
-      if (lane == 0) out[(long)m * N + n0 + r] = __float2bfloat16(v);
+      if (lane == 0 && n0 + r < N) out[(long)m * N + n0 + r] = __float2bfloat16(v);
     }
   }
 }
@@ -90,17 +102,25 @@ torch::Tensor skinny_gemm(torch::Tensor x, torch::Tensor w) {
   int M = xc.size(0), K = xc.size(1);
   long N = wc.size(0);
   TORCH_CHECK(wc.size(1) == K, "skinny_gemm: K mismatch");
-  TORCH_CHECK(M <= SG_MAX_M, "skinny_gemm: M must be <= 16");
+  TORCH_CHECK(M >= 1 && M <= SG_MAX_M, "skinny_gemm: M must be in [1, 16]");
   TORCH_CHECK(K % 8 == 0, "skinny_gemm: K must be a multiple of 8");
   auto out = torch::empty({M, N}, xc.options());
   long rows_per_block = (long)SG_WAVES * SG_ROWS;
   long grid = (N + rows_per_block - 1) / rows_per_block;
   hipStream_t stream = at::hip::getCurrentHIPStream();
-  hipLaunchKernelGGL(skinny_gemm_kernel, dim3((unsigned)grid),
-      dim3(WAVE * SG_WAVES), 0, stream,
-      reinterpret_cast<const bf16*>(xc.data_ptr()),
-      reinterpret_cast<const bf16*>(wc.data_ptr()),
-      reinterpret_cast<bf16*>(out.data_ptr()), M, N, K);
+  const bf16* xp = reinterpret_cast<const bf16*>(xc.data_ptr());
+  const bf16* wp = reinterpret_cast<const bf16*>(wc.data_ptr());
+  bf16* op = reinterpret_cast<bf16*>(out.data_ptr());
+  dim3 g((unsigned)grid), b(WAVE * SG_WAVES);
+  switch (M) {
+#define SG_CASE(MT) case MT: \
+    hipLaunchKernelGGL(skinny_gemm_kernel<MT>, g, b, 0, stream, xp, wp, op, N, K); break;
+    SG_CASE(1) SG_CASE(2) SG_CASE(3) SG_CASE(4) SG_CASE(5) SG_CASE(6)
+    SG_CASE(7) SG_CASE(8) SG_CASE(9) SG_CASE(10) SG_CASE(11) SG_CASE(12)
+    SG_CASE(13) SG_CASE(14) SG_CASE(15) SG_CASE(16)
+#undef SG_CASE
+    default: TORCH_CHECK(false, "skinny_gemm: unsupported M");
+  }
   return out;
 }
 
