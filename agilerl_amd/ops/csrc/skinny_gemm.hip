@@ -42,22 +42,23 @@ skinny_gemm_kernel(
     const bf16* __restrict__ w,   // (N, K)
     bf16* __restrict__ out,       // (MT, N)
     long N, int K) {
+  // One block owns SG_ROWS consecutive W rows; all 256 lanes stride K
+  // together (the whole block is one wide K-reader), so small-N shapes
+  // still launch N/4 blocks and the per-wave dependent-load chain is a
+  // quarter as long as a wave-per-rows layout.
   int wave = threadIdx.x / WAVE;
   int lane = threadIdx.x % WAVE;
-  long n0 = ((long)blockIdx.x * SG_WAVES + wave) * SG_ROWS;
+  long n0 = (long)blockIdx.x * SG_ROWS;
   if (n0 >= N) return;
   bool full = (n0 + SG_ROWS) <= N;
 
-  // compile-time MT => acc and the x slice stay in registers (a runtime
-  // M loop indexes the array dynamically and spills to scratch — measured
-  // 315 GB/s before templating)
   float acc[MT][SG_ROWS];
   #pragma unroll
   for (int m = 0; m < MT; ++m)
     #pragma unroll
     for (int r = 0; r < SG_ROWS; ++r) acc[m][r] = 0.f;
 
-  for (int k0 = lane * 8; k0 < K; k0 += WAVE * 8) {
+  for (int k0 = (int)threadIdx.x * 8; k0 < K; k0 += WAVE * SG_WAVES * 8) {
     float wf[SG_ROWS][8];
     if (full) {
       #pragma unroll
@@ -80,15 +81,29 @@ skinny_gemm_kernel(
     }
   }
 
+  // wave-level shuffle reduce, then cross-wave LDS combine
+  __shared__ float lds[SG_WAVES][MT > 0 ? MT : 1][SG_ROWS];
   #pragma unroll
-  for (int m = 0; m < MT; ++m) {
+  for (int m = 0; m < MT; ++m)
     #pragma unroll
     for (int r = 0; r < SG_ROWS; ++r) {
       float v = acc[m][r];
       #pragma unroll
       for (int off = WAVE / 2; off > 0; off >>= 1) v += __shfl_xor(v, off);
-      if (lane == 0 && n0 + r < N) out[(long)m * N + n0 + r] = __float2bfloat16(v);
+      if (lane == 0) lds[wave][m][r] = v;
     }
+  __syncthreads();
+  if (wave == 0 && lane == 0) {
+    #pragma unroll
+    for (int m = 0; m < MT; ++m)
+      #pragma unroll
+      for (int r = 0; r < SG_ROWS; ++r) {
+        if (n0 + r >= N) continue;
+        float v = 0.f;
+        #pragma unroll
+        for (int s = 0; s < SG_WAVES; ++s) v += lds[s][m][r];
+        out[(long)m * N + n0 + r] = __float2bfloat16(v);
+      }
   }
 }
 
@@ -105,8 +120,7 @@ torch::Tensor skinny_gemm(torch::Tensor x, torch::Tensor w) {
   TORCH_CHECK(M >= 1 && M <= SG_MAX_M, "skinny_gemm: M must be in [1, 16]");
   TORCH_CHECK(K % 8 == 0, "skinny_gemm: K must be a multiple of 8");
   auto out = torch::empty({M, N}, xc.options());
-  long rows_per_block = (long)SG_WAVES * SG_ROWS;
-  long grid = (N + rows_per_block - 1) / rows_per_block;
+  long grid = (N + SG_ROWS - 1) / SG_ROWS;
   hipStream_t stream = at::hip::getCurrentHIPStream();
   const bf16* xp = reinterpret_cast<const bf16*>(xc.data_ptr());
   const bf16* wp = reinterpret_cast<const bf16*>(wc.data_ptr());
