@@ -33,6 +33,7 @@ def _fast_linear(module, x2d: torch.Tensor) -> torch.Tensor:
     tile kernels (~0.6-1 TB/s at M<=16), keeping any LoRA delta as the
     small eager side-path exactly like LoraLinear.forward."""
     ext = extension()
+    base = getattr(module, "base", module)  # LoraLinear wraps .base
     eligible = (
         ext is not None
         and hasattr(ext, "skinny_gemm")
@@ -40,8 +41,11 @@ def _fast_linear(module, x2d: torch.Tensor) -> torch.Tensor:
         and x2d.dtype == torch.bfloat16
         and x2d.shape[0] <= 16
         and x2d.shape[1] % 8 == 0
+        # the kernel beats hipBLASLt only below its small-GEMM floor
+        # (~19 us): N<=2048 covers the GQA k/v projections; larger shapes
+        # stream at 5.5+ TB/s on hipBLASLt and stay there
+        and getattr(base, "out_features", 1 << 30) <= 2048
     )
-    base = getattr(module, "base", module)  # LoraLinear wraps .base
     if (
         not eligible
         or not isinstance(base, torch.nn.Linear)

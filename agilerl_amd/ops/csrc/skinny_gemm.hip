@@ -2,16 +2,17 @@
 //
 // Decode-time projections are M<=16 rows against multi-MB weight
 // matrices — effectively M simultaneous GEMVs, bound by streaming W from
-// HBM3E. hipBLASLt's tile kernels reach only ~0.6-1 TB/s at these shapes
-// (measured via rocprofv3 on the 8B decode graph); this kernel is built
-// around the weight stream instead:
+// HBM3E.  Measured on MI355X (demos/bench_kernels.py): hipBLASLt carries
+// a ~19 us small-GEMM floor, so the 8 MB GQA k/v projections run at
+// ~445 GB/s; this kernel reaches 622 GB/s there and is dispatched for
+// N<=2048 only — hipBLASLt keeps larger shapes, where its tiles stream
+// at 5.5+ TB/s.
 //
-//   grid = ceil(N / 16) blocks x 256 threads (4 waves); each wave owns
-//   R=4 consecutive W rows. Per K-chunk of 512 elements, every lane
-//   issues one 16-byte bf16x8 load per W row (fully-coalesced 4 KB per
-//   wave-iteration of pure weight traffic) plus M 16-byte x loads that
-//   are L2-resident (x is a few KB). fp32 accumulation; xor-shuffle
-//   reduce; lane 0 writes the M x R outputs.
+//   grid = ceil(N / 4) blocks x 256 threads; the whole block strides K
+//   for its 4 W rows (short dependent-load chains even at small N).
+//   Per K-chunk every lane issues one 16-byte bf16x8 load per W row
+//   (fully-coalesced weight traffic) plus M x-loads that ride L2.
+//   fp32 accumulation; xor-shuffle + LDS cross-wave reduce.
 //
 // Requirements: bf16 x/W, K % 8 == 0, M <= 16. Bias unsupported (Llama
 // projections are bias-free); LoRA deltas ride separately.
