@@ -132,10 +132,39 @@ class LoraLinear(nn.Module):
     def forward(self, x: torch.Tensor) -> torch.Tensor:
         out = self.base(x)
         name = self.active_adapter
-        if name is not None and name in self.lora_A:
+        if name is not None and name in self.lora_A and self._merged is None:
             h = self.dropout(x) if self.dropout is not None else x
             out = out + F.linear(F.linear(h, self.lora_A[name]), self.lora_B[name]) * self.scaling
         return out
+
+    # ------------------------------------------------------------------
+    # Merge for generation: fold B@A*scale into the base weight so decode
+    # runs plain GEMMs (no adapter side-ops).  The pre-merge bf16 weight
+    # bits are snapshotted, so unmerge restores the base EXACTLY (a
+    # subtract would drift by rounding each cycle).  Base weights are
+    # frozen under LoRA training, so the snapshot is always valid.
+    # ------------------------------------------------------------------
+    _merged = None  # (adapter_name, snapshot) while merged
+
+    @torch.no_grad()
+    def merge_adapter(self, name: str) -> None:
+        if self._merged is not None:
+            raise RuntimeError(f"already merged: {self._merged[0]}")
+        if name not in self.lora_A:
+            return
+        w = self.base.weight
+        snapshot = w.detach().clone()
+        delta = (self.lora_B[name].float() @ self.lora_A[name].float()) * self.scaling
+        w.data.copy_((w.float() + delta).to(w.dtype))
+        self._merged = (name, snapshot)
+
+    @torch.no_grad()
+    def unmerge_adapter(self) -> None:
+        if self._merged is None:
+            return
+        _, snapshot = self._merged
+        self.base.weight.data.copy_(snapshot)
+        self._merged = None
 
 
 # ---------------------------------------------------------------------------
@@ -235,3 +264,23 @@ def load_adapter(model: nn.Module, adapter: str, directory: str, strict_config: 
             break
     state = load_file(os.path.join(directory, "adapter_model.safetensors"))
     load_adapter_state_dict(model, adapter, state)
+
+
+@torch.no_grad()
+def merge_adapter(model: nn.Module, name: str) -> int:
+    """Fold adapter ``name`` into every LoraLinear's base weight (exact
+    unmerge via snapshot).  Returns modules merged."""
+    n = 0
+    for _, module in iter_lora_modules(model):
+        module.merge_adapter(name)
+        n += 1
+    return n
+
+
+@torch.no_grad()
+def unmerge_adapter(model: nn.Module) -> int:
+    n = 0
+    for _, module in iter_lora_modules(model):
+        module.unmerge_adapter()
+        n += 1
+    return n

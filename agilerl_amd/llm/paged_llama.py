@@ -55,7 +55,11 @@ def _fast_linear(module, x2d: torch.Tensor) -> torch.Tensor:
         return module(x2d)
     out = ext.skinny_gemm(x2d, base.weight)
     name = getattr(module, "active_adapter", None)
-    if name is not None and name in getattr(module, "lora_A", ()):
+    if (
+        name is not None
+        and name in getattr(module, "lora_A", ())
+        and getattr(module, "_merged", None) is None
+    ):
         h = module.dropout(x2d) if getattr(module, "dropout", None) is not None else x2d
         out = out + F.linear(F.linear(h, module.lora_A[name]), module.lora_B[name]) * module.scaling
     return out

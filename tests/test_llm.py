@@ -921,3 +921,75 @@ class TestHuggingFaceGym:
                                     pad_token_id=tokenizer.pad_token_id)
         stats = agent.learn(exp)
         assert np.isfinite(stats["loss"])
+
+
+class TestMergedGeneration:
+    def test_merge_unmerge_bitwise_exact(self):
+        import torch.nn as nn
+
+        from agilerl_amd.llm.lora import LoraConfig, LoraLinear
+
+        torch.manual_seed(0)
+        base = nn.Linear(32, 16, bias=False)
+        mod = LoraLinear(base, LoraConfig(r=4, lora_alpha=8), adapters=("a",))
+        with torch.no_grad():
+            mod.lora_B["a"].normal_(0, 0.1)
+        before = base.weight.detach().clone()
+        for _ in range(5):  # repeated cycles must not drift a single bit
+            mod.merge_adapter("a")
+            assert not torch.equal(base.weight, before)
+            mod.unmerge_adapter()
+            assert torch.equal(base.weight, before)
+
+    def test_merged_forward_matches_adapter_forward(self):
+        import torch.nn as nn
+
+        from agilerl_amd.llm.lora import LoraConfig, LoraLinear
+
+        torch.manual_seed(1)
+        base = nn.Linear(64, 32, bias=False)
+        mod = LoraLinear(base, LoraConfig(r=8, lora_alpha=16), adapters=("a",))
+        with torch.no_grad():
+            mod.lora_B["a"].normal_(0, 0.05)
+        mod.active_adapter = "a"
+        x = torch.randn(4, 64)
+        ref = mod(x)
+        mod.merge_adapter("a")
+        merged = mod(x)  # adapter path suppressed; delta folded into W
+        mod.unmerge_adapter()
+        torch.testing.assert_close(merged, ref, rtol=1e-5, atol=1e-5)
+
+    def test_generate_paged_merged_parity_and_restore(self):
+        """Merged generation (default) produces the same greedy tokens as
+        explicit unmerged generation, and the base weights are restored
+        bitwise afterwards — even with a NONZERO adapter delta."""
+        from agilerl_amd.algorithms.llm.grpo import GRPO
+        from agilerl_amd.llm.lora import iter_lora_modules
+
+        tiny = dict(model_type="llama", vocab_size=64, hidden_size=32,
+                    intermediate_size=64, num_hidden_layers=2,
+                    num_attention_heads=2, num_key_value_heads=1,
+                    max_position_embeddings=128, pad_token_id=0)
+        torch.manual_seed(0)
+        agent = GRPO(model_config=tiny, dtype=torch.float32, lora_config={"r": 2},
+                     max_completion_tokens=6, generation="paged")
+        with torch.no_grad():  # give the adapter a real delta
+            for _, m in iter_lora_modules(agent.model):
+                m.lora_B[agent.adapter_name].normal_(0, 0.02)
+        snapshot = {
+            n: m.base.weight.detach().clone()
+            for n, m in iter_lora_modules(agent.model)
+        }
+        torch.manual_seed(1)
+        ids = torch.randint(1, 64, (3, 5))
+        mask = torch.ones_like(ids)
+        agent.merged_generation = True
+        out_merged = agent.generate_paged(ids, mask, do_sample=False)
+        # weights restored bitwise, engine dropped state cleanly
+        for n, m in iter_lora_modules(agent.model):
+            assert torch.equal(m.base.weight, snapshot[n]), n
+            assert m._merged is None
+        agent.merged_generation = False
+        agent._decode_engine = None  # fresh engine for the unmerged pass
+        out_adapter = agent.generate_paged(ids, mask, do_sample=False)
+        torch.testing.assert_close(out_merged, out_adapter)
