@@ -74,3 +74,27 @@ def test_bench_grpo_tiny_n2_gloo():
     assert result["n_gpus"] == 2
     assert result["value"] > 0
     assert result["config"]["global_batch"] == 16
+
+
+@pytest.mark.slow
+def test_bench_rainbow_n2_gloo():
+    result = _run_bench([
+        "--workload", "rainbow", "--steps", "2", "--warmup", "1",
+        "--num-envs", "8",
+    ])
+    assert result["metric"] == "env_steps_per_sec"
+    assert result["n_gpus"] == 2
+    assert result["value"] > 0
+    assert result["config"]["pop_size"] == 4
+
+
+@pytest.mark.slow
+def test_bench_maddpg_n2_gloo():
+    result = _run_bench([
+        "--workload", "maddpg", "--steps", "2", "--warmup", "1",
+        "--num-envs", "8",
+    ])
+    assert result["metric"] == "env_steps_per_sec"
+    assert result["n_gpus"] == 2
+    assert result["config"]["pop_size"] == 8
+    assert result["value"] > 0
