@@ -46,20 +46,26 @@ def run_llm_workload(trainer, workload: str):
 
     env_kwargs = dict(spec.env_kwargs)
     if workload == "llm_reasoning":
-        from ...llm_envs.reasoning import TokenReasoningGym
-
         reward_fn = None
         if spec.reward_fn:
             mod, fn = spec.reward_fn.rsplit(".", 1)
             reward_fn = getattr(importlib.import_module(mod), fn)
-        env = TokenReasoningGym(
-            vocab_size=vocab_size,
-            prompt_len=spec.max_prompt_tokens,
-            data_batch_size=spec.data_batch_size,
-            group_size=spec.group_size,
-            reward_fn=reward_fn,
-            **env_kwargs,
-        )
+        if spec.dataset:
+            # dataset-backed gym (reference HuggingFaceGym flow): a local
+            # saved `datasets` directory + the agent's tokenizer (or a
+            # tokenizer_path in env_kwargs)
+            env = _dataset_reasoning_gym(spec, pop[0], reward_fn, env_kwargs)
+        else:
+            from ...llm_envs.reasoning import TokenReasoningGym
+
+            env = TokenReasoningGym(
+                vocab_size=vocab_size,
+                prompt_len=spec.max_prompt_tokens,
+                data_batch_size=spec.data_batch_size,
+                group_size=spec.group_size,
+                reward_fn=reward_fn,
+                **env_kwargs,
+            )
         loop = finetune_llm_reasoning
     elif workload == "llm_sft":
         from ...llm_envs.sft import SyntheticSFTGym
@@ -99,4 +105,48 @@ def run_llm_workload(trainer, workload: str):
         mutation=mutations,
         loggers=trainer.loggers,
         max_wall_seconds=t.max_wall_seconds,
+    )
+
+
+def _dataset_reasoning_gym(spec, agent, reward_fn, env_kwargs):
+    """Build a HuggingFaceGym from a manifest `dataset:` entry.
+
+    ``dataset``: path to a `datasets.save_to_disk` directory (or a
+    DatasetDict dir with train/test splits).  Tokenizer resolution order:
+    ``env_kwargs["tokenizer_path"]`` -> the agent's tokenizer.
+    """
+    from datasets import load_from_disk
+
+    from ...llm_envs.base import HuggingFaceGym
+
+    if reward_fn is None:
+        raise ValueError(
+            "a dataset-backed reasoning env needs `reward_fn` in the manifest "
+            "(dotted import path of reward_fn(completion_text, answer))"
+        )
+    kwargs = dict(env_kwargs)
+    tok_path = kwargs.pop("tokenizer_path", None)
+    if tok_path:
+        from transformers import AutoTokenizer
+
+        tokenizer = AutoTokenizer.from_pretrained(tok_path)
+    else:
+        tokenizer = getattr(agent, "tokenizer", None)
+    if tokenizer is None:
+        raise ValueError(
+            "a dataset-backed reasoning env needs a tokenizer: pass "
+            "`tokenizer_path` in environment.env_kwargs or give the "
+            "algorithm a tokenizer"
+        )
+    ds = load_from_disk(spec.dataset)
+    if hasattr(ds, "keys") and "train" in ds:  # DatasetDict
+        train, test = ds["train"], ds.get("test", ds["train"])
+    else:
+        train = test = ds
+    return HuggingFaceGym(
+        train, test, tokenizer, reward_fn,
+        data_batch_size=spec.data_batch_size,
+        group_size=spec.group_size,
+        max_prompt_tokens=spec.max_prompt_tokens,
+        **kwargs,
     )

@@ -993,3 +993,63 @@ class TestMergedGeneration:
         agent._decode_engine = None  # fresh engine for the unmerged pass
         out_adapter = agent.generate_paged(ids, mask, do_sample=False)
         torch.testing.assert_close(out_merged, out_adapter)
+
+
+def _contains_answer_reward(completion, answer):
+    return float(str(answer) in completion)
+
+
+class TestDatasetManifestFlow:
+    def test_trainer_runs_dataset_backed_reasoning(self, tmp_path):
+        """Manifest `dataset:` + `reward_fn:` + `tokenizer_path` ->
+        LocalTrainer builds a HuggingFaceGym and trains GRPO end to end
+        (the reference HuggingFaceGym flow, llm_envs/base.py:93)."""
+        from datasets import Dataset, DatasetDict
+
+        from agilerl_amd.models.manifest import TrainingManifest
+        from agilerl_amd.training.trainer import LocalTrainer
+
+        train = Dataset.from_dict({
+            "question": [f"what is {i} plus {i}" for i in range(8)],
+            "answer": [str(2 * i) for i in range(8)],
+        })
+        dd = DatasetDict({"train": train, "test": train.select(range(2))})
+        ds_dir = tmp_path / "ds"
+        dd.save_to_disk(str(ds_dir))
+        tok = _word_tokenizer(list(train["question"]) + list(train["answer"]))
+        tok_dir = tmp_path / "tok"
+        tok.save_pretrained(str(tok_dir))
+
+        vocab = tok.vocab_size + 8
+        manifest = TrainingManifest.model_validate({
+            "algorithm": {
+                "name": "GRPO",
+                "hyperparameters": {
+                    "model_config": {
+                        "model_type": "llama", "vocab_size": vocab,
+                        "hidden_size": 32, "intermediate_size": 64,
+                        "num_hidden_layers": 1, "num_attention_heads": 2,
+                        "num_key_value_heads": 1,
+                        "max_position_embeddings": 128,
+                        "pad_token_id": tok.pad_token_id,
+                    },
+                    "lora_config": {"r": 2},
+                    "dtype": "float32",
+                    "group_size": 2,
+                    "micro_batch_size": 2,
+                    "max_completion_tokens": 4,
+                },
+            },
+            "environment": {
+                "type": "llm", "env_type": "reasoning",
+                "dataset": str(ds_dir),
+                "reward_fn": "tests.test_llm._contains_answer_reward",
+                "data_batch_size": 2, "group_size": 2,
+                "max_prompt_tokens": 16,
+                "env_kwargs": {"tokenizer_path": str(tok_dir)},
+            },
+            "training": {"max_steps": 2, "pop_size": 2, "evo_steps": 1},
+        })
+        trainer = LocalTrainer(manifest, device="cpu")
+        results = trainer.train()
+        assert results is not None
