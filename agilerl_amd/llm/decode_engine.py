@@ -182,14 +182,18 @@ class DecodeEngine:
         sel_t = self._select_batch(out.logits[:, -1], seqs)
         lps = self._chosen_logps(out.logits[:, -1], sel_t, seqs).tolist()
         sel = sel_t.tolist()
+        # batched pool write: (L, B, n_kv, T, D) stacked once, four tensor
+        # ops per layer (vs a python append per sequence per layer)
+        k_all = torch.stack([
+            self._layer_kv(out.past_key_values, layer)[0]
+            for layer in range(self.cache.num_layers)
+        ])
+        v_all = torch.stack([
+            self._layer_kv(out.past_key_values, layer)[1]
+            for layer in range(self.cache.num_layers)
+        ])
+        self.cache.append_prefill([s.seq_id for s in seqs], k_all, v_all, mask)
         for i, s in enumerate(seqs):
-            n = s.prompt_ids.numel()
-            layers_k, layers_v = [], []
-            for layer in range(self.cache.num_layers):
-                k_l, v_l = self._layer_kv(out.past_key_values, layer)
-                layers_k.append(k_l[i, :, L - n :].permute(1, 0, 2))  # (T, n_kv, D)
-                layers_v.append(v_l[i, :, L - n :].permute(1, 0, 2))
-            self.cache.append(s.seq_id, torch.stack(layers_k), torch.stack(layers_v))
             s.prefilled = True
             s.generated.append(int(sel[i]))
             s.gen_logps.append(float(lps[i]))

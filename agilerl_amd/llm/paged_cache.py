@@ -89,6 +89,43 @@ class PagedKVCache:
             done += n
         self.lengths[seq_id] = pos + T
 
+    @torch.no_grad()
+    def append_prefill(self, seq_ids, k, v, mask) -> None:
+        """Batched first-append for freshly-admitted sequences.
+
+        k/v: (num_layers, B, n_kv, T, D) as produced by the prefill
+        forward; mask: (B, T) with 1 on real (right-aligned) positions.
+        Four tensor ops per layer instead of a python stack+append per
+        sequence (the per-step decode path appends in-kernel already).
+        Sequences must have length 0 (just allocated).
+        """
+        L = self.num_layers
+        S = self.page_size
+        lengths = mask.sum(dim=1).tolist()
+        slot_rows = []
+        for sid, n in zip(seq_ids, lengths):
+            n = int(n)
+            if self.lengths[sid] != 0:
+                raise RuntimeError("append_prefill requires fresh sequences")
+            self._ensure_capacity(sid, n)
+            pos = torch.arange(n)
+            pages = torch.tensor(self.page_tables[sid], dtype=torch.long)[
+                torch.div(pos, S, rounding_mode="floor")
+            ]
+            slot_rows.append(pages * S + pos % S)
+        slots = torch.cat(slot_rows).to(self.device) if slot_rows else None
+        if slots is None or slots.numel() == 0:
+            return
+        valid = mask.bool()
+        flat_k = self.k_pool.view(L, self.num_pages * S, self.num_kv_heads, self.head_dim)
+        flat_v = self.v_pool.view(L, self.num_pages * S, self.num_kv_heads, self.head_dim)
+        for layer in range(L):
+            # (B, n_kv, T, D) -> (B, T, n_kv, D) -> (total, n_kv, D)
+            flat_k[layer][slots] = k[layer].permute(0, 2, 1, 3)[valid].to(self.k_pool.dtype)
+            flat_v[layer][slots] = v[layer].permute(0, 2, 1, 3)[valid].to(self.v_pool.dtype)
+        for sid, n in zip(seq_ids, lengths):
+            self.lengths[sid] = int(n)
+
     def gather(
         self, seq_ids: List[int], pad_to: Optional[int] = None
     ) -> Tuple[torch.Tensor, torch.Tensor, torch.Tensor]:
