@@ -33,6 +33,7 @@ def create_mlp(
     layer_norm: bool = False,
     output_layernorm: bool = False,
     init_layers: bool = True,
+    output_vanish: bool = False,
 ) -> nn.Sequential:
     """Plain (non-evolvable) MLP builder used for network heads."""
 
@@ -52,7 +53,13 @@ def create_mlp(
         if layer_norm:
             layers.append(nn.LayerNorm(sizes[i + 1]))
         layers.append(get_activation(activation))
-    layers.append(linear(sizes[-1], output_size))
+    out_layer = linear(sizes[-1], output_size)
+    if output_vanish and not noisy:
+        with torch.no_grad():
+            out_layer.weight.mul_(0.1)
+            if out_layer.bias is not None:
+                out_layer.bias.mul_(0.1)
+    layers.append(out_layer)
     if output_layernorm:
         layers.append(nn.LayerNorm(output_size))
     if output_activation is not None:
@@ -79,6 +86,7 @@ class EvolvableMLP(EvolvableModule):
         noisy: bool = False,
         noise_std: float = 0.5,
         init_layers: bool = True,
+        output_vanish: bool = False,
         device: str = "cpu",
     ):
         super().__init__(device)
@@ -96,6 +104,9 @@ class EvolvableMLP(EvolvableModule):
         self.noisy = noisy
         self.noise_std = noise_std
         self.init_layers = init_layers
+        # reference-parity knob (agilerl mlp config): scale the final
+        # layer's weights/bias by 0.1 so initial outputs start near zero
+        self.output_vanish = output_vanish
 
         self.model = self._build().to(device)
 
@@ -112,6 +123,7 @@ class EvolvableMLP(EvolvableModule):
             layer_norm=self.layer_norm,
             output_layernorm=self.output_layernorm,
             init_layers=self.init_layers,
+            output_vanish=self.output_vanish,
         )
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
