@@ -658,3 +658,33 @@ class TestSkinnyGemm:
         ref = mod(x)
         torch.testing.assert_close(out.float(), ref.float(), rtol=3e-2, atol=3e-1)
         assert not torch.allclose(out.float(), base(x).float(), rtol=1e-3, atol=1e-3)
+
+
+class TestKernelSanitizer:
+    """SURVEY 5.2 analog of compute-sanitizer CI: re-run the kernel
+    numerics suite with serialized kernel launches (AMD_SERIALIZE_KERNEL=3
+    forces a sync after every launch, so async faults surface at the
+    guilty kernel instead of a later sync point)."""
+
+    def test_kernels_under_serialized_launches(self):
+        import os
+        import subprocess
+        import sys
+
+        env = dict(os.environ)
+        env["AMD_SERIALIZE_KERNEL"] = "3"
+        targets = [
+            "tests/test_ops_gpu.py::TestGaeScan",
+            "tests/test_ops_gpu.py::TestNStep",
+            "tests/test_ops_gpu.py::TestC51",
+            "tests/test_ops_gpu.py::TestGroupAdvantage",
+            "tests/test_ops_gpu.py::TestSegmentTreeKernels",
+            "tests/test_ops_gpu.py::TestSkinnyGemm",
+            "tests/test_ops_gpu.py::TestPagedAttnKernel",
+        ]
+        out = subprocess.run(
+            [sys.executable, "-m", "pytest", *targets, "-q", "-m", "gpu",
+             "-p", "no:cacheprovider"],
+            capture_output=True, text=True, timeout=600, env=env,
+        )
+        assert out.returncode == 0, out.stdout[-2000:] + out.stderr[-1000:]
