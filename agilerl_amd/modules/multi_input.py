@@ -29,6 +29,7 @@ class EvolvableMultiInput(EvolvableModule):
         latent_dim: int = 64,
         mlp_config: Optional[dict] = None,
         cnn_config: Optional[dict] = None,
+        sub_configs: Optional[Dict[str, dict]] = None,
         device: str = "cpu",
     ):
         super().__init__(device)
@@ -37,6 +38,11 @@ class EvolvableMultiInput(EvolvableModule):
         self.latent_dim = int(latent_dim)
         self.mlp_config = dict(mlp_config or {})
         self.cnn_config = dict(cnn_config or {})
+        # per-key overrides: mutations let sub-encoders diverge (each
+        # samples its own layer/channel choice), so clones/checkpoints
+        # rebuild from the LIVE per-encoder shapes via the `sub_configs`
+        # property, not the shared construction-time configs
+        ctor_sub = {k: dict(v) for k, v in (sub_configs or {}).items()}
 
         if isinstance(observation_space, DictSpace):
             items = list(observation_space.spaces.items())
@@ -47,19 +53,36 @@ class EvolvableMultiInput(EvolvableModule):
 
         encoders = {}
         for key, space in items:
-            encoders[key] = self._make_encoder(space)
+            encoders[key] = self._make_encoder(space, ctor_sub.get(key))
         self.encoders = nn.ModuleDict(encoders)
         self.head = nn.Linear(self.latent_dim * len(items), self.num_outputs).to(device)
 
-    def _make_encoder(self, space: Space) -> EvolvableModule:
+    @property
+    def sub_configs(self) -> Dict[str, dict]:
+        """Live per-encoder architecture (feeds init_dict -> clone)."""
+        out: Dict[str, dict] = {}
+        for key, enc in self.encoders.items():
+            if isinstance(enc, EvolvableCNN):
+                out[key] = {
+                    "channel_size": list(enc.channel_size),
+                    "kernel_size": list(enc.kernel_size),
+                    "stride_size": list(enc.stride_size),
+                }
+            else:
+                out[key] = {"hidden_size": list(enc.hidden_size)}
+        return out
+
+    def _make_encoder(self, space: Space, override: Optional[dict] = None) -> EvolvableModule:
         if is_image_space(space):
             cfg = {"channel_size": [32, 32], "kernel_size": [3, 3], "stride_size": [2, 2]}
             cfg.update(self.cnn_config)
+            cfg.update(override or {})
             return EvolvableCNN(
                 input_shape=space.shape, num_outputs=self.latent_dim, device=self.device, **cfg
             )
         cfg = {"hidden_size": [64]}
         cfg.update(self.mlp_config)
+        cfg.update(override or {})
         return EvolvableMLP(
             num_inputs=flatdim(space), num_outputs=self.latent_dim, device=self.device, **cfg
         )
@@ -88,25 +111,37 @@ class EvolvableMultiInput(EvolvableModule):
         self.head = new_head
 
     @mutation(MutationType.NODE)
-    def add_node(self, numb_new_nodes: Optional[int] = None) -> dict:
-        out = {}
-        for enc in self.encoders.values():
-            method = "add_node" if hasattr(enc, "add_node") else "add_channel"
-            r = enc.apply_mutation(method, **out) or {}
-            nn_key = "numb_new_nodes" if "numb_new_nodes" in r else None
-            if nn_key and "numb_new_nodes" not in out:
-                out["numb_new_nodes"] = r["numb_new_nodes"]
-        return {"numb_new_nodes": out.get("numb_new_nodes", numb_new_nodes)}
+    def add_node(
+        self, per_encoder: Optional[dict] = None, numb_new_nodes: Optional[int] = None
+    ) -> dict:
+        return self._fanout("add_node", "add_channel", per_encoder, numb_new_nodes)
 
     @mutation(MutationType.NODE)
-    def remove_node(self, numb_new_nodes: Optional[int] = None) -> dict:
-        out = {}
-        for enc in self.encoders.values():
-            method = "remove_node" if hasattr(enc, "remove_node") else "remove_channel"
-            r = enc.apply_mutation(method, **out) or {}
-            if "numb_new_nodes" in r and "numb_new_nodes" not in out:
-                out["numb_new_nodes"] = r["numb_new_nodes"]
-        return {"numb_new_nodes": out.get("numb_new_nodes", numb_new_nodes)}
+    def remove_node(
+        self, per_encoder: Optional[dict] = None, numb_new_nodes: Optional[int] = None
+    ) -> dict:
+        return self._fanout("remove_node", "remove_channel", per_encoder, numb_new_nodes)
+
+    def _fanout(self, mlp_method, cnn_method, per_encoder, numb_new_nodes) -> dict:
+        """Apply a node mutation to every sub-encoder, capturing EVERY
+        sampled choice per encoder key so group replay (targets/critics)
+        reproduces the exact same shapes — forwarding only the node count
+        let each replayed sub-encoder re-sample its layer/channel choice
+        and desynchronize from the policy (caught by the arch sweep)."""
+        per_encoder = {k: dict(v) for k, v in (per_encoder or {}).items()}
+        for key, enc in self.encoders.items():
+            method = mlp_method if hasattr(enc, mlp_method) else cnn_method
+            choices = per_encoder.get(key)
+            if choices is None:
+                choices = {}
+                if numb_new_nodes is not None:
+                    count_key = (
+                        "numb_new_nodes" if method.endswith("node") else "numb_new_channels"
+                    )
+                    choices[count_key] = numb_new_nodes
+            r = enc.apply_mutation(method, **choices) or {}
+            per_encoder[key] = r
+        return {"per_encoder": per_encoder}
 
     @mutation(MutationType.ACTIVATION)
     def change_activation(self, activation: Optional[str] = None, output: bool = False) -> dict:

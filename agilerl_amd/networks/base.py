@@ -65,7 +65,10 @@ def build_encoder(
             observation_space=observation_space,
             num_outputs=latent_dim,
             device=device,
-            **{k: v for k, v in cfg.items() if k in ("mlp_config", "cnn_config")},
+            # sub_configs carries the live (possibly mutated) per-encoder
+            # shapes so clones/checkpoints rebuild exactly
+            **{k: v for k, v in cfg.items()
+               if k in ("mlp_config", "cnn_config", "sub_configs")},
         )
     if arch == "resnet":
         from ..modules.resnet import EvolvableResNet

@@ -240,3 +240,86 @@ def test_hpo_mutations_actually_apply_to_multiagent(cls_name):
     applied = [a.mut for a in pop]
     assert not any(m.startswith("Failed") for m in applied), applied
     assert any(m != "None" for m in applied), applied
+
+
+class TestArchTypeMutationSweep:
+    """Architecture mutations across every encoder family (the reference
+    mutation suite exercises CNN/MultiInput/SimBa too, not just MLP)."""
+
+    def _agent_for(self, arch):
+        from agilerl_amd.algorithms.dqn import DQN
+        from agilerl_amd.spaces import Box, DictSpace, Discrete
+
+        torch.manual_seed(0)
+        if arch == "cnn":
+            space = Box(0, 255, (3, 16, 16))
+            cfg = {"arch": "cnn", "channel_size": [8, 8], "kernel_size": [3, 3],
+                   "stride_size": [1, 1]}
+        elif arch == "simba":
+            space = Box(-1, 1, (8,))
+            cfg = {"arch": "simba", "hidden_size": 32, "num_blocks": 1}
+        elif arch == "multi_input":
+            space = DictSpace({"vec": Box(-1, 1, (6,)), "img": Box(0, 255, (1, 8, 8))})
+            cfg = {"arch": "multi_input"}
+        else:
+            space = Box(-1, 1, (8,))
+            cfg = {"arch": "mlp", "hidden_size": [16, 16]}
+        return DQN(space, Discrete(3), net_config=cfg, batch_size=8), space
+
+    def _batch(self, space):
+        from agilerl_amd.spaces import DictSpace
+
+        def sample(n):
+            if isinstance(space, DictSpace):
+                return {k: torch.as_tensor(
+                    np.stack([space.spaces[k].sample() for _ in range(n)])
+                ).float() for k in space.spaces}
+            return torch.as_tensor(
+                np.stack([space.sample() for _ in range(n)])).float()
+
+        return {
+            "obs": sample(8),
+            "action": torch.randint(0, 3, (8, 1)),
+            "reward": torch.randn(8, 1),
+            "next_obs": sample(8),
+            "done": torch.zeros(8, 1),
+        }
+
+    @pytest.mark.parametrize("arch", ["mlp", "cnn", "simba", "multi_input"])
+    def test_every_architecture_mutation_keeps_agent_trainable(self, arch):
+        agent, space = self._agent_for(arch)
+        methods = [m for m in agent.mutation_methods if "activation" not in m]
+        assert methods, f"{arch}: no architecture mutations exposed"
+        applied = 0
+        for method in methods:
+            agent.apply_architecture_mutation(method)
+            applied += 1
+            loss = agent.learn(self._batch(space))
+            assert np.isfinite(loss)
+            # targets must track the mutated eval net structurally
+            sd_a = agent.actor.state_dict()
+            sd_t = agent.actor_target.state_dict()
+            assert set(sd_a.keys()) == set(sd_t.keys())
+            for k in sd_a:
+                assert sd_a[k].shape == sd_t[k].shape, (arch, method, k)
+        assert applied == len(methods)
+        # clone + checkpoint survive the fully-mutated architecture
+        clone = agent.clone(5)
+        x = self._batch(space)["obs"]
+        if isinstance(x, dict):
+            torch.testing.assert_close(clone.actor(x), agent.actor(x))
+        else:
+            torch.testing.assert_close(clone.actor(x), agent.actor(x))
+
+    def test_lstm_network_mutations(self):
+        from agilerl_amd.modules.lstm import EvolvableLSTM
+
+        torch.manual_seed(1)
+        lstm = EvolvableLSTM(input_size=6, num_outputs=4, hidden_state_size=16)
+        for method in list(lstm.mutation_methods):
+            if "activation" in method:
+                continue
+            lstm.apply_mutation(method)
+            out = lstm(torch.randn(3, 5, 6))
+            assert out.shape[-1] == 4
+            assert torch.isfinite(out).all()
