@@ -323,3 +323,69 @@ class TestArchTypeMutationSweep:
             out = lstm(torch.randn(3, 5, 6))
             assert out.shape[-1] == 4
             assert torch.isfinite(out).all()
+
+
+class TestArchSweepOtherAlgos:
+    """The replay-desync bug class hits any algorithm with target nets —
+    pin Rainbow(cnn) and DDPG(multi_input) too."""
+
+    def test_rainbow_cnn_mutations(self):
+        from agilerl_amd.algorithms.dqn_rainbow import RainbowDQN
+        from agilerl_amd.spaces import Box, Discrete
+
+        torch.manual_seed(0)
+        agent = RainbowDQN(
+            Box(0, 255, (3, 16, 16)), Discrete(4), batch_size=8, n_step=1,
+            net_config={"arch": "cnn", "channel_size": [8, 8],
+                        "kernel_size": [3, 3], "stride_size": [1, 1]},
+        )
+        batch = {
+            "obs": torch.rand(8, 3, 16, 16) * 255,
+            "action": torch.randint(0, 4, (8, 1)),
+            "reward": torch.randn(8, 1),
+            "next_obs": torch.rand(8, 3, 16, 16) * 255,
+            "done": torch.zeros(8, 1),
+            "weights": torch.ones(8),
+            "idxs": torch.arange(8),
+        }
+        for method in [m for m in agent.mutation_methods if "activation" not in m]:
+            agent.apply_architecture_mutation(method)
+            loss = agent.learn(batch)
+            assert np.isfinite(float(loss if not isinstance(loss, tuple) else loss[0]))
+            sd_a, sd_t = agent.actor.state_dict(), agent.actor_target.state_dict()
+            for k in sd_a:
+                assert sd_a[k].shape == sd_t[k].shape, (method, k)
+        clone = agent.clone(1)
+        x = torch.rand(4, 3, 16, 16) * 255
+        torch.testing.assert_close(clone.actor(x), agent.actor(x))
+
+    def test_ddpg_multi_input_mutations(self):
+        from agilerl_amd.algorithms.ddpg import DDPG
+        from agilerl_amd.spaces import Box, DictSpace
+
+        torch.manual_seed(1)
+        space = DictSpace({"vec": Box(-1, 1, (5,)), "img": Box(0, 255, (1, 8, 8))})
+        agent = DDPG(space, Box(-1, 1, (2,)), batch_size=8,
+                     net_config={"arch": "multi_input"})
+
+        def obs(n):
+            return {"vec": torch.randn(n, 5), "img": torch.rand(n, 1, 8, 8) * 255}
+
+        batch = {
+            "obs": obs(8), "action": torch.rand(8, 2) * 2 - 1,
+            "reward": torch.randn(8, 1), "next_obs": obs(8),
+            "done": torch.zeros(8, 1),
+        }
+        for method in [m for m in agent.mutation_methods if "activation" not in m]:
+            agent.apply_architecture_mutation(method)
+            out = agent.learn(batch)
+            vals = out if isinstance(out, tuple) else (out,)
+            assert all(np.isfinite(float(v)) for v in vals if v is not None)
+            for name, tgt in (("actor", "actor_target"), ("critic", "critic_target")):
+                sd_a = getattr(agent, name).state_dict()
+                sd_t = getattr(agent, tgt).state_dict()
+                for k in sd_a:
+                    assert sd_a[k].shape == sd_t[k].shape, (name, method, k)
+        clone = agent.clone(2)
+        o = obs(4)
+        torch.testing.assert_close(clone.actor(o), agent.actor(o))
