@@ -62,6 +62,11 @@ class PPOLLM(GRPO):
         )
         action_mask = experiences["action_mask"].to(self.device).float()
         rewards = experiences["rewards"].to(self.device).float()
+        sampling_logps = experiences.get("sampling_logps")
+        if sampling_logps is not None and self.sampling_is_correction:
+            sampling_logps = sampling_logps.to(self.device).float()
+        else:
+            sampling_logps = None
 
         B = ids.shape[0]
         mb = max(self.micro_batch_size, 1)
@@ -94,6 +99,7 @@ class PPOLLM(GRPO):
                 loss = self._policy_loss(
                     logp, old_logp[sel], adv_tok[sel], action_mask[sel],
                     ref_logp[sel] if ref_logp is not None else None, clip_lo, clip_hi,
+                    sampling_logp=sampling_logps[sel] if sampling_logps is not None else None,
                 )
                 self.backward_and_step(loss)
                 # value head regression toward realized reward

@@ -1053,3 +1053,24 @@ class TestDatasetManifestFlow:
         trainer = LocalTrainer(manifest, device="cpu")
         results = trainer.train()
         assert results is not None
+
+
+class TestPpoLlmSamplingIs:
+    def test_ppollm_consumes_sampling_logps(self):
+        from agilerl_amd.algorithms.llm.ppo_llm import PPOLLM
+
+        tiny = dict(model_type="llama", vocab_size=64, hidden_size=32,
+                    intermediate_size=64, num_hidden_layers=1,
+                    num_attention_heads=2, num_key_value_heads=1,
+                    max_position_embeddings=128, pad_token_id=0)
+        torch.manual_seed(0)
+        agent = PPOLLM(model_config=tiny, dtype=torch.float32,
+                       lora_config={"r": 2}, group_size=2, micro_batch_size=2,
+                       sampling_is_correction=True)
+        g = torch.Generator().manual_seed(1)
+        ids = torch.randint(1, 64, (4, 12), generator=g)
+        pos = torch.arange(11).unsqueeze(0)
+        am = (pos + 1 >= 6).float().expand(4, 11)
+        base = {"ids": ids, "action_mask": am, "rewards": torch.rand(4, generator=g)}
+        stats = agent.learn({**base, "sampling_logps": torch.randn(4, 11, generator=g) - 2})
+        assert np.isfinite(stats["loss"])
