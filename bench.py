@@ -68,8 +68,10 @@ def parse_args():
     p.add_argument("--no-graph", action="store_true", help="disable hipGraph collector")
     p.add_argument("--packing", action="store_true", help="grpo: padding-free packed logprob passes")
     p.add_argument("--micro-batch", type=int, default=None, help="grpo: micro batch size per backward")
-    p.add_argument("--no-grad-ckpt", action="store_true",
-                   help="grpo: disable gradient checkpointing (288 GB fits 8B seq-1k activations)")
+    p.add_argument("--grad-ckpt", action="store_true",
+                   help="grpo: enable gradient checkpointing (default OFF: 288 GB "
+                   "fits 8B seq-1k activations at micro-batch 16; measured +36%%)")
+    p.add_argument("--no-grad-ckpt", action="store_true", help=argparse.SUPPRESS)
     return p.parse_args()
 
 
@@ -244,11 +246,11 @@ class GrpoBenchRunner:
             dtype=dtype,
             lora_config={"r": 16, "lora_alpha": 32},
             group_size=min(8, max(1, args.grpo_batch)),
-            micro_batch_size=args.micro_batch or (4 if args.model_size == "8b" else 8),
+            micro_batch_size=args.micro_batch or (16 if args.model_size == "8b" else 8),
             update_epochs=1,
             beta=0.04,
             lr=5e-6,
-            gradient_checkpointing=(args.model_size == "8b") and not args.no_grad_ckpt,
+            gradient_checkpointing=bool(args.grad_ckpt),
             use_packing=args.packing,
             device=self.device,
         )
