@@ -33,7 +33,6 @@ recorded list in ``ckpt["_ref_compat_dropped"]``.
 from __future__ import annotations
 
 import importlib
-import io
 from typing import Any, Dict, List, Tuple
 
 import numpy as np
@@ -103,11 +102,6 @@ def _resolve(name: str, ref_module: str):
             f"utils/ref_compat._NAME_MAP if it should load)"
         )
     return getattr(importlib.import_module(target), name)
-
-
-class _SpaceShim:
-    """Absorbs a gymnasium space's pickled state and rebuilds the
-    equivalent first-party space on demand."""
 
 
 def _gym_space_to_ours(obj: Any) -> Any:
