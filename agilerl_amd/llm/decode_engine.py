@@ -1,14 +1,21 @@
-"""Continuous-batching decode engine over the paged KV cache (eager mode).
+"""Continuous-batching decode engine over the paged KV cache.
 
-Round-2 decode-engine milestone 1 (docs/design/round2_perf_plan.md):
-scheduling and storage are final — sequences are admitted the moment a
-slot frees, KV lives in pages — while the per-step attention reads go
-through an eager gather into a transformers ``DynamicCache`` (round 2
-swaps that for a paged-attention HIP kernel and a graph-captured step).
+Sequences are admitted the moment pages free up, KV lives in fixed-size
+pages behind per-sequence page tables, and finished sequences retire
+individually.  On GPU the decode step is fully first-party: a per-layer
+forward writes K/V straight into the pool and attends through the
+flash-decoding ``paged_attn`` HIP kernel (no gather, no DynamicCache),
+the whole step replays as a hipGraph per adapter over static slot
+buffers (``llm/paged_llama.py``), token selection is batched (one device
+op + one host sync per step), and every generated token's behavior-
+policy logprob is captured (the vLLM sampling-logprob analog).  The
+eager gather + ``DynamicCache`` path remains as the CPU/fallback
+implementation and the parity reference.
 
-Parity contract (tested): greedy decode through the engine matches
-``model.generate(do_sample=False)`` sequence-for-sequence, including
-ragged admission mid-flight.
+Parity contracts (tested): greedy decode through the engine matches
+``model.generate(do_sample=False)`` sequence-for-sequence including
+ragged admission; graphed and ungraphed engines produce bitwise-equal
+sequences under slot churn.
 """
 
 from __future__ import annotations

@@ -1,10 +1,11 @@
 """Paged KV cache: fixed-size HBM pages + per-sequence page tables.
 
-Round-2 decode-engine milestone 1 (docs/design/round2_perf_plan.md):
-the storage layer is final — pages in a preallocated pool, sequences own
-page lists — while reads go through an eager gather (round 2 replaces
-the gather with a paged-attention kernel that walks page tables
-directly).  Reference analog: vLLM's PagedAttention block manager
+Pages live in a preallocated pool; sequences own page lists.  The GPU
+decode path reads the pool DIRECTLY through the page tables inside the
+``paged_attn`` HIP kernel and appends each new token's K/V in place
+(``llm/paged_llama.py``); prefill appends are batched tensor writes
+(:meth:`append_prefill`).  :meth:`gather` remains for the CPU/fallback
+decode path.  Reference analog: vLLM's PagedAttention block manager
 (the reference colocates vLLM for generation; here it is first-party).
 """
 

@@ -1,9 +1,9 @@
 """Paged-attention decode: HIP kernel dispatch + eager reference.
 
-Round-2 decode-engine milestone 2 (docs/design/round2_perf_plan.md §2).
-The eager reference is also the CPU implementation the decode engine can
-call; the HIP kernel reads K/V straight from the page pool through the
-page tables (no gather).
+The flash-decoding kernel (csrc/paged_attn.hip) reads K/V straight from
+the page pool through the page tables (no gather); the eager reference
+below is the CPU implementation and the numerics oracle the GPU tests
+pin the kernel against.
 """
 
 from __future__ import annotations
