@@ -298,24 +298,9 @@ class LLMAlgorithm(EvolvableAlgorithm):
             self._activate("self")
         return out
 
-    @torch.no_grad()
-    def generate_paged(
-        self,
-        input_ids: torch.Tensor,
-        attention_mask: torch.Tensor,
-        max_new_tokens: Optional[int] = None,
-        do_sample: bool = True,
-        temperature: Optional[float] = None,
-    ) -> torch.Tensor:
-        """Generation through the continuous-batching paged-KV engine
-        (llm/decode_engine.py).  Same (B, P+C) rectangular contract as
-        ``generate``; completions are right-padded with pad_token_id."""
+    def _ensure_engine(self, tokens_needed: int, B: int):
         from ...llm.decode_engine import DecodeEngine
 
-        C = max_new_tokens or self.max_completion_tokens
-        B, P = input_ids.shape
-        pad_id = getattr(self.model.config, "pad_token_id", 0) or 0
-        tokens_needed = int(attention_mask.sum()) + B * C
         engine = getattr(self, "_decode_engine", None)
         page_size = 16
         if engine is None or engine.cache.num_pages * page_size < tokens_needed:
@@ -329,6 +314,65 @@ class LLMAlgorithm(EvolvableAlgorithm):
             )
             self._decode_engine = engine
         engine.max_batch = max(engine.max_batch, B)
+        return engine
+
+    @torch.no_grad()
+    def stream_generate(
+        self,
+        input_ids: torch.Tensor,
+        attention_mask: torch.Tensor,
+        max_new_tokens: Optional[int] = None,
+        do_sample: bool = True,
+        temperature: Optional[float] = None,
+    ):
+        """Yield completion token ids one step at a time (single prompt)
+        through the paged engine — the serving /generate/stream source."""
+        C = max_new_tokens or self.max_completion_tokens
+        ids = input_ids.reshape(-1)
+        engine = self._ensure_engine(int(ids.numel()) + C, 1)
+        self.model.eval()
+        temp = temperature if temperature is not None else self.temperature
+        sid = engine.submit(
+            ids, max_new_tokens=C,
+            temperature=temp if do_sample else 0.0, adapter="self",
+        )
+        try:
+            seen = 0
+            done = False
+            while engine.has_work() and not done:
+                finished = engine.step()
+                done = any(fsid == sid for fsid, _ in finished)
+                seq = engine.active.get(sid)
+                gen = (
+                    seq.generated if seq is not None
+                    else next(ids_ for fsid, ids_ in finished if fsid == sid)[
+                        input_ids.reshape(-1).numel():
+                    ].tolist()
+                )
+                for tok in list(gen)[seen:]:
+                    yield int(tok)
+                seen = len(gen)
+        finally:
+            engine.cancel(sid)
+            self.model.train()
+
+    @torch.no_grad()
+    def generate_paged(
+        self,
+        input_ids: torch.Tensor,
+        attention_mask: torch.Tensor,
+        max_new_tokens: Optional[int] = None,
+        do_sample: bool = True,
+        temperature: Optional[float] = None,
+    ) -> torch.Tensor:
+        """Generation through the continuous-batching paged-KV engine
+        (llm/decode_engine.py).  Same (B, P+C) rectangular contract as
+        ``generate``; completions are right-padded with pad_token_id."""
+        C = max_new_tokens or self.max_completion_tokens
+        B, P = input_ids.shape
+        pad_id = getattr(self.model.config, "pad_token_id", 0) or 0
+        tokens_needed = int(attention_mask.sum()) + B * C
+        engine = self._ensure_engine(tokens_needed, B)
         self.model.eval()
         temp = temperature if temperature is not None else self.temperature
         # Merged generation: fold this agent's LoRA delta into the base

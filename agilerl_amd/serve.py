@@ -185,6 +185,49 @@ def create_app(agent, lock: Optional[threading.Lock] = None):
             resp["completion"] = [tok.decode(row) for row in completion.cpu().tolist()]
         return resp
 
+    @app.post("/generate/stream")
+    def generate_stream(req: GenerateRequest):
+        """NDJSON token stream: one {"token_id", "text"} line per generated
+        token, then a final {"done": true, "completion": ...} line."""
+        import json as _json
+
+        import torch
+        from fastapi.responses import StreamingResponse
+
+        a = state["agent"]
+        if not hasattr(a, "stream_generate"):
+            raise HTTPException(status_code=400, detail="agent cannot stream")
+        tok = getattr(a, "tokenizer", None)
+        if req.input_ids is not None:
+            ids = torch.as_tensor(req.input_ids, dtype=torch.long).reshape(-1)
+        elif req.prompt is not None and tok is not None:
+            ids = torch.as_tensor(tok.encode(req.prompt), dtype=torch.long)
+        else:
+            raise HTTPException(status_code=400,
+                                detail="input_ids (or prompt + tokenizer) required")
+        ids = ids.to(getattr(a, "device", "cpu"))
+        mask = torch.ones_like(ids)
+
+        def event_lines():
+            produced = []
+            with lock, torch.no_grad():
+                state["requests"] += 1
+                for t in a.stream_generate(
+                    ids, mask, max_new_tokens=req.max_new_tokens,
+                    do_sample=req.do_sample, temperature=req.temperature,
+                ):
+                    produced.append(t)
+                    event = {"token_id": t}
+                    if tok is not None:
+                        event["text"] = tok.decode([t])
+                    yield _json.dumps(event) + "\n"
+            final = {"done": True, "completion_ids": produced}
+            if tok is not None:
+                final["completion"] = tok.decode(produced)
+            yield _json.dumps(final) + "\n"
+
+        return StreamingResponse(event_lines(), media_type="application/x-ndjson")
+
     @app.post("/reload")
     def reload(req: ReloadRequest):
         try:

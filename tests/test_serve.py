@@ -128,3 +128,55 @@ def test_serve_generate_uses_paged_engine_when_configured():
     assert r.status_code == 200
     assert len(r.json()["completion_ids"][0]) == 3
     assert agent._decode_engine is not None  # the paged engine actually ran
+
+
+class TestStreamingGenerate:
+    def test_stream_generate_tokens_match_paged(self):
+        import torch
+        """stream_generate yields exactly the tokens generate_paged
+        produces for the same greedy prompt."""
+        from agilerl_amd.algorithms.llm.grpo import GRPO
+
+        tiny = dict(model_type="llama", vocab_size=64, hidden_size=32,
+                    intermediate_size=64, num_hidden_layers=1,
+                    num_attention_heads=2, num_key_value_heads=1,
+                    max_position_embeddings=128, pad_token_id=0)
+        torch.manual_seed(0)
+        agent = GRPO(model_config=tiny, dtype=torch.float32, lora_config={"r": 2},
+                     max_completion_tokens=6, generation="paged")
+        torch.manual_seed(1)
+        ids = torch.randint(1, 64, (1, 5))
+        mask = torch.ones_like(ids)
+        full = agent.generate_paged(ids, mask, do_sample=False)
+        want = full[0, 5:].tolist()
+        agent._decode_engine = None  # independent engine for the stream
+        got = list(agent.stream_generate(ids[0], mask[0], do_sample=False))
+        assert got == want
+
+    def test_http_stream_endpoint_ndjson(self):
+        import json
+
+        import torch
+
+        from starlette.testclient import TestClient
+
+        from agilerl_amd.algorithms.llm.grpo import GRPO
+        from agilerl_amd.serve import create_app
+
+        tiny = dict(model_type="llama", vocab_size=64, hidden_size=32,
+                    intermediate_size=64, num_hidden_layers=1,
+                    num_attention_heads=2, num_key_value_heads=1,
+                    max_position_embeddings=128, pad_token_id=0)
+        torch.manual_seed(0)
+        agent = GRPO(model_config=tiny, dtype=torch.float32, lora_config={"r": 2},
+                     max_completion_tokens=5, generation="paged")
+        client = TestClient(create_app(agent))
+        with client.stream("POST", "/generate/stream", json={
+            "input_ids": [3, 7, 11], "max_new_tokens": 5, "do_sample": False,
+        }) as r:
+            assert r.status_code == 200
+            lines = [json.loads(l) for l in r.iter_lines() if l.strip()]
+        assert lines[-1]["done"] is True
+        token_events = [l for l in lines[:-1] if "token_id" in l]
+        assert len(token_events) == 5
+        assert lines[-1]["completion_ids"] == [e["token_id"] for e in token_events]
