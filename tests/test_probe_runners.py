@@ -61,3 +61,35 @@ def test_grpo_constant_target_reward_improves():
                             group_size=8, reward_fn=reward_fn, seed=0)
     hist = check_llm_policy_with_probe_env(agent, env, iterations=25)
     assert np.mean(hist[-8:]) > np.mean(hist[:3]) + 0.02, hist
+
+
+def test_continuous_q_learning_runner_ddpg():
+    from agilerl_amd.algorithms import DDPG
+    from agilerl_amd.envs.probe import (
+        ConstantRewardContActionsEnv,
+        check_policy_q_learning_with_probe_env,
+    )
+
+    np.random.seed(0), torch.manual_seed(0)
+    agent = check_policy_q_learning_with_probe_env(
+        ConstantRewardContActionsEnv(num_envs=4), DDPG,
+        dict(lr_actor=1e-3, lr_critic=1e-2, tau=0.1, batch_size=64,
+             net_config=dict(NET)),
+    )
+    assert agent is not None
+
+
+def test_continuous_q_learning_runner_td3():
+    from agilerl_amd.algorithms import TD3
+    from agilerl_amd.envs.probe import (
+        ConstantRewardContActionsEnv,
+        check_policy_q_learning_with_probe_env,
+    )
+
+    np.random.seed(1), torch.manual_seed(1)
+    agent = check_policy_q_learning_with_probe_env(
+        ConstantRewardContActionsEnv(num_envs=4), TD3,
+        dict(lr_actor=1e-3, lr_critic=1e-2, tau=0.1, batch_size=64,
+             net_config=dict(NET)),
+    )
+    assert agent is not None
