@@ -151,10 +151,39 @@ class MutationProbabilities(_Base):
     activation: float = 0.0
     rl_hp: float = 0.2
 
+    @model_validator(mode="before")
+    @classmethod
+    def _reference_aliases(cls, data):
+        # reference manifests spell these no_mut / arch_mut / params_mut /
+        # act_mut / rl_hp_mut (configs/training/*); without the aliases
+        # they validated but silently fell back to defaults
+        if isinstance(data, dict):
+            alias = {"no_mut": "no_mutation", "arch_mut": "architecture",
+                     "params_mut": "parameters", "act_mut": "activation",
+                     "rl_hp_mut": "rl_hp"}
+            data = dict(data)
+            for ref, ours in alias.items():
+                if ref in data and ours not in data:
+                    data[ours] = data.pop(ref)
+        return data
+
 
 class MutationSpec(_Base):
     probabilities: MutationProbabilities = Field(default_factory=MutationProbabilities)
     new_layer_prob: float = 0.2
+
+    @model_validator(mode="before")
+    @classmethod
+    def _hoist_new_layer(cls, data):
+        # the reference nests new_layer under probabilities
+        if isinstance(data, dict):
+            probs = data.get("probabilities")
+            if isinstance(probs, dict) and "new_layer" in probs:
+                data = dict(data)
+                probs = dict(probs)
+                data.setdefault("new_layer_prob", probs.pop("new_layer"))
+                data["probabilities"] = probs
+        return data
     mutation_sd: float = 0.1
     activation_selection: List[str] = Field(default_factory=lambda: ["ReLU", "ELU", "GELU"])
     mutate_elite: bool = True
@@ -189,6 +218,14 @@ class TrainingSpec(_Base):
     eps_end: float = 0.05
     eps_decay: float = 0.995
     target: Optional[float] = None
+
+    @model_validator(mode="before")
+    @classmethod
+    def _reference_aliases(cls, data):
+        if isinstance(data, dict) and "target_score" in data and "target" not in data:
+            data = dict(data)
+            data["target"] = data.pop("target_score")
+        return data
     checkpoint: Optional[int] = None
     checkpoint_path: Optional[str] = None
     overwrite_checkpoints: bool = True

@@ -202,3 +202,23 @@ class TestReferenceConfigEndToEnd:
         trainer = LocalTrainer(manifest, device="cpu")
         results = trainer.train()
         assert results is not None
+
+
+class TestReferenceFieldAliases:
+    def test_reference_mutation_and_target_fields_apply(self):
+        """Reference spellings (no_mut/arch_mut/.../new_layer/target_score)
+        must MAP, not silently fall back to defaults."""
+        import yaml as _yaml
+
+        from agilerl_amd.models.manifest import TrainingManifest
+
+        doc = _yaml.safe_load(open("/root/reference/configs/training/dqn/dqn.yaml"))
+        m = TrainingManifest.model_validate(doc)
+        p = m.mutation.probabilities
+        assert (p.no_mutation, p.architecture, p.parameters, p.activation,
+                p.rl_hp) == (0.4, 0.2, 0.2, 0.2, 0.2)
+        assert m.mutation.new_layer_prob == 0.2
+        assert m.mutation.mutation_sd == 0.1
+        assert m.training.target == 200.0
+        # rl_hp_selection bounds flow too
+        assert m.mutation.rl_hp_selection["lr"]["max"] == 0.01
