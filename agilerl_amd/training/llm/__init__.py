@@ -30,6 +30,9 @@ def run_llm_workload(trainer, workload: str):
 
     from ...models.manifest import resolve_algo_class
 
+    import inspect
+    import warnings
+
     m = trainer.manifest
     spec = m.env_spec()
     algo_cls = resolve_algo_class(m.algorithm.name)
@@ -39,9 +42,27 @@ def run_llm_workload(trainer, workload: str):
         for k in ("model_config", "model_name_or_path", "lora_config")
         if k in hps
     }
+    # reference manifests put prompts-per-step under algorithm.batch_size
+    # and carry vLLM/deepspeed-era fields this build designs out; filter
+    # to the local constructor surface with an explicit warning so
+    # reference YAMLs construct instead of TypeError-ing
+    batch_override = hps.pop("batch_size", None)
+    accepted = set(inspect.signature(algo_cls.__init__).parameters)
+    dropped = sorted(k for k in hps if k not in accepted)
+    for k in dropped:
+        hps.pop(k)
+    if dropped:
+        warnings.warn(
+            f"{m.algorithm.name}: dropping reference-only hyperparameters "
+            f"{dropped} (no equivalent on this build — vLLM/DeepSpeed era "
+            "fields are designed out; see COMPONENTS.md §2.2)",
+            RuntimeWarning,
+        )
     pop = algo_cls.population(
         m.training.pop_size, device=trainer.device, **model_kwargs, **hps
     )
+    if batch_override and not spec.env_kwargs.get("data_batch_size"):
+        spec.data_batch_size = int(batch_override)
     vocab_size = pop[0].model.config.vocab_size
 
     env_kwargs = dict(spec.env_kwargs)
@@ -114,17 +135,39 @@ def _dataset_reasoning_gym(spec, agent, reward_fn, env_kwargs):
     ``dataset``: path to a `datasets.save_to_disk` directory (or a
     DatasetDict dir with train/test splits).  Tokenizer resolution order:
     ``env_kwargs["tokenizer_path"]`` -> the agent's tokenizer.
+
+    Reference-manifest conveniences (configs/training/llm_finetuning/*):
+    ``columns: {<dataset_col>: question|answer}`` maps dataset columns,
+    ``reward_file_path`` + ``reward_fn_name`` load the reward from a
+    python file, ``train_test_split`` splits a single dataset.
     """
     from datasets import load_from_disk
 
     from ...llm_envs.base import HuggingFaceGym
 
     if reward_fn is None:
+        path = getattr(spec, "reward_file_path", None)
+        fn_name = getattr(spec, "reward_fn_name", None)
+        if path and fn_name:
+            import importlib.util
+
+            spec_mod = importlib.util.spec_from_file_location("manifest_reward", path)
+            mod = importlib.util.module_from_spec(spec_mod)
+            spec_mod.loader.exec_module(mod)
+            reward_fn = getattr(mod, fn_name)
+    if reward_fn is None:
         raise ValueError(
             "a dataset-backed reasoning env needs `reward_fn` in the manifest "
-            "(dotted import path of reward_fn(completion_text, answer))"
+            "(dotted import path of reward_fn(completion_text, answer)) or "
+            "reward_file_path + reward_fn_name"
         )
     kwargs = dict(env_kwargs)
+    cols = getattr(spec, "columns", None) or {}
+    for ds_col, role in cols.items():
+        if role == "question":
+            kwargs.setdefault("prompt_key", ds_col)
+        elif role == "answer":
+            kwargs.setdefault("answer_key", ds_col)
     tok_path = kwargs.pop("tokenizer_path", None)
     if tok_path:
         from transformers import AutoTokenizer
@@ -142,7 +185,13 @@ def _dataset_reasoning_gym(spec, agent, reward_fn, env_kwargs):
     if hasattr(ds, "keys") and "train" in ds:  # DatasetDict
         train, test = ds["train"], ds.get("test", ds["train"])
     else:
-        train = test = ds
+        frac = float(getattr(spec, "train_test_split", 1.0) or 1.0)
+        if 0.0 < frac < 1.0:
+            n_train = max(1, int(len(ds) * frac))
+            train = ds.select(range(n_train))
+            test = ds.select(range(n_train, len(ds))) if n_train < len(ds) else train
+        else:
+            train = test = ds
     return HuggingFaceGym(
         train, test, tokenizer, reward_fn,
         data_batch_size=spec.data_batch_size,
