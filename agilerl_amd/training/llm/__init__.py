@@ -42,6 +42,20 @@ def run_llm_workload(trainer, workload: str):
         for k in ("model_config", "model_name_or_path", "lora_config")
         if k in hps
     }
+    # reference manifests carry the model under the NETWORK section
+    # (network.pretrained_model_name_or_path / lora_config with lora_r)
+    net = getattr(m, "network", None)
+    if net is not None:
+        pretrained = getattr(net, "pretrained_model_name_or_path", None)
+        if pretrained and "model_config" not in model_kwargs \
+                and "model_name_or_path" not in model_kwargs:
+            model_kwargs["model_name_or_path"] = pretrained
+        net_lora = getattr(net, "lora_config", None)
+        if net_lora and "lora_config" not in model_kwargs:
+            net_lora = dict(net_lora)
+            if "lora_r" in net_lora:
+                net_lora["r"] = net_lora.pop("lora_r")
+            model_kwargs["lora_config"] = net_lora
     # reference manifests put prompts-per-step under algorithm.batch_size
     # and carry vLLM/deepspeed-era fields this build designs out; filter
     # to the local constructor surface with an explicit warning so
@@ -66,6 +80,18 @@ def run_llm_workload(trainer, workload: str):
     vocab_size = pop[0].model.config.vocab_size
 
     env_kwargs = dict(spec.env_kwargs)
+    # the reference routes by env_type, not algorithm: a GRPO-family agent
+    # with environment.env_type=multiturn trains through the multi-turn loop
+    if getattr(spec, "env_type", None) == "multiturn" and workload == "llm_reasoning":
+        env = _multiturn_env(spec, pop[0])
+        t = m.training
+        tournament, mutations = trainer._make_hpo()
+        return finetune_llm_multiturn(
+            env, pop, max_steps=t.max_steps, evo_steps=t.evo_steps,
+            eval_loop=t.eval_loop, target=t.target, tournament=tournament,
+            mutation=mutations, loggers=trainer.loggers,
+            max_wall_seconds=t.max_wall_seconds,
+        )
     if workload == "llm_reasoning":
         reward_fn = None
         if spec.reward_fn:
@@ -197,5 +223,39 @@ def _dataset_reasoning_gym(spec, agent, reward_fn, env_kwargs):
         data_batch_size=spec.data_batch_size,
         group_size=spec.group_size,
         max_prompt_tokens=spec.max_prompt_tokens,
+        **kwargs,
+    )
+
+
+def _multiturn_env(spec, agent):
+    """Map a reference multiturn env_name onto the first-party game envs.
+
+    ``game:GuessTheNumber-*`` (the reference's GEM game) maps to
+    :class:`TokenGuessEnv`; anything else needs ``env_kwargs.env_factory``
+    (dotted path to a zero-arg factory).
+    """
+    import importlib
+
+    from ...llm_envs.multiturn import SyncMultiTurnVecEnv, TokenGuessEnv
+
+    kwargs = dict(spec.env_kwargs)
+    name = getattr(spec, "env_name", "") or kwargs.pop("env_name", "")
+    factory_path = kwargs.pop("env_factory", None)
+    vocab = agent.model.config.vocab_size
+    if factory_path:
+        mod, fn = factory_path.rsplit(".", 1)
+        factory = getattr(importlib.import_module(mod), fn)
+    elif "GuessTheNumber" in str(name):
+        factory = lambda: TokenGuessEnv(vocab_size=vocab)  # noqa: E731
+    else:
+        raise ValueError(
+            f"no multiturn env mapping for {name!r}; set "
+            "environment.env_kwargs.env_factory to a factory import path"
+        )
+    return SyncMultiTurnVecEnv(
+        factory,
+        data_batch_size=spec.data_batch_size,
+        group_size=spec.group_size,
+        max_turns=kwargs.pop("max_turns", 2),
         **kwargs,
     )
