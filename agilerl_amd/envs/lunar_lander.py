@@ -19,7 +19,7 @@ import numpy as np
 from ..spaces import Box, Discrete
 from .base import BatchedVecEnv
 
-__all__ = ["LunarLanderVecEnv"]
+__all__ = ["LunarLanderVecEnv", "LunarLanderContinuousVecEnv"]
 
 
 class LunarLanderVecEnv(BatchedVecEnv):
@@ -125,6 +125,65 @@ class LunarLanderVecEnv(BatchedVecEnv):
         reward -= (0.30 * main + 0.03 * (left + right)).astype(np.float32)
 
         # now damp ground-contact velocities for rows that keep running
+        vy = np.where(grounded & (vy < 0), 0.0, vy)
+        vx = np.where(grounded, vx * 0.8, vx)
+        vang = np.where(grounded, vang * 0.5, vang)
+        self.state = np.stack([px, py, vx, vy, ang, vang], axis=1)
+
+        crash = (grounded & (~upright | (impact_speed > 1.5))) | (np.abs(px) > self.X_WORLD)
+        landed = grounded & upright & (impact_speed <= 1.5) & (np.abs(vang) < 0.3)
+        terminated = crash | landed
+        reward = np.where(crash, reward - 100.0, reward)
+        reward = np.where(landed, reward + 100.0, reward)
+        return reward.astype(np.float32), terminated, None
+
+
+class LunarLanderContinuousVecEnv(LunarLanderVecEnv):
+    """Continuous-action variant (LunarLanderContinuous-v2/-v3 interface):
+    action (2,) in [-1, 1] — [main throttle, lateral].  Standard mapping:
+    the main engine fires for a0 > 0 with power 0.5 + 0.5*a0; the lateral
+    thruster fires when |a1| > 0.5 (sign picks the side), matching the
+    classic env's control contract on the same first-party dynamics."""
+
+    def __init__(self, num_envs: int = 1, seed: Optional[int] = None):
+        super().__init__(num_envs, seed)
+        self.single_action_space = Box(-1.0, 1.0, (2,))
+
+    def _step_all(self, actions: np.ndarray):
+        a = np.asarray(actions, dtype=np.float64).reshape(self.num_envs, 2)
+        a = np.clip(a, -1.0, 1.0)
+        px, py, vx, vy, ang, vang = self.state.T
+
+        main = np.where(a[:, 0] > 0.0, 0.5 + 0.5 * a[:, 0], 0.0)
+        lateral = np.where(np.abs(a[:, 1]) > 0.5, a[:, 1], 0.0)
+        right = np.clip(lateral, 0.0, 1.0)
+        left = np.clip(-lateral, 0.0, 1.0)
+
+        ax = -np.sin(ang) * self.MAIN_ACCEL * main + (right - left) * self.SIDE_ACCEL * np.cos(ang)
+        ay = np.cos(ang) * self.MAIN_ACCEL * main - self.GRAVITY + (right - left) * self.SIDE_ACCEL * np.sin(ang)
+        aang = (left - right) * self.SIDE_TORQUE
+
+        vx = vx + ax * self.DT
+        vy = vy + ay * self.DT
+        vang = (vang + aang * self.DT) * self.ANGLE_DAMP
+        px = px + vx * self.DT
+        py = py + vy * self.DT
+        ang = ang + vang * self.DT
+
+        on_ground = py <= self.LEG_Y
+        upright = np.abs(ang) < 0.4
+        self.legs[:, 0] = (on_ground & upright).astype(np.float64)
+        self.legs[:, 1] = (on_ground & upright).astype(np.float64)
+        py = np.maximum(py, 0.0)
+        grounded = py <= 0.0 + 1e-9
+        impact_speed = np.sqrt(vx**2 + vy**2)
+
+        self.state = np.stack([px, py, vx, vy, ang, vang], axis=1)
+        shaping = self._shaping()
+        reward = (shaping - self.prev_shaping).astype(np.float32)
+        self.prev_shaping = shaping
+        reward -= (0.30 * main + 0.03 * np.abs(lateral)).astype(np.float32)
+
         vy = np.where(grounded & (vy < 0), 0.0, vy)
         vx = np.where(grounded, vx * 0.8, vx)
         vang = np.where(grounded, vang * 0.5, vang)

@@ -11,7 +11,7 @@ from typing import Callable, Dict, Optional, Union
 
 from .base import VecEnv
 from .cartpole import CartPoleVecEnv
-from .lunar_lander import LunarLanderVecEnv
+from .lunar_lander import LunarLanderContinuousVecEnv, LunarLanderVecEnv
 from .classic_control import AcrobotVecEnv, MountainCarContinuousVecEnv, MountainCarVecEnv
 from .pendulum import PendulumVecEnv
 from .visual import BreakoutLiteVecEnv, CatchPongVecEnv
@@ -32,6 +32,8 @@ ENV_REGISTRY: Dict[str, Callable[..., VecEnv]] = {
     "CartPole-v0": CartPoleVecEnv,
     "LunarLander-v2": LunarLanderVecEnv,
     "LunarLander-v3": LunarLanderVecEnv,
+    "LunarLanderContinuous-v2": LunarLanderContinuousVecEnv,
+    "LunarLanderContinuous-v3": LunarLanderContinuousVecEnv,
     "Pendulum-v1": PendulumVecEnv,
     "MountainCar-v0": MountainCarVecEnv,
     "MountainCarContinuous-v0": MountainCarContinuousVecEnv,

@@ -27,6 +27,8 @@ from ..models.manifest import (
 __all__ = ["Trainer", "LocalTrainer"]
 
 PZ_REGISTRY = {
+    # reference manifests may prefix the module (mpe2.simple_...); the
+    # lookup strips any dotted prefix
     "simple_speaker_listener_v4": "agilerl_amd.envs.mpe.SpeakerListenerVecEnv",
     "simple_speaker_listener": "agilerl_amd.envs.mpe.SpeakerListenerVecEnv",
     "simple_spread_v3": "agilerl_amd.envs.mpe.SimpleSpreadVecEnv",
@@ -78,7 +80,8 @@ class LocalTrainer(Trainer):
                                   seed=self.manifest.training.seed,
                                   **getattr(spec, "env_kwargs", {}))
         if spec.type == "pettingzoo":
-            env_path = PZ_REGISTRY.get(spec.env_id)
+            pz_id = spec.env_id.rsplit(".", 1)[-1]  # strip mpe2./pettingzoo. prefixes
+            env_path = PZ_REGISTRY.get(pz_id)
             if env_path is None:
                 raise KeyError(
                     f"Unknown multi-agent env '{spec.env_id}'. Known: {sorted(PZ_REGISTRY)}"

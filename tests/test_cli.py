@@ -181,7 +181,10 @@ class TestReferenceConfigEndToEnd:
     WORK here."""
 
     @pytest.mark.slow
-    @pytest.mark.parametrize("rel", ["dqn/dqn.yaml", "ppo/ppo.yaml"])
+    @pytest.mark.parametrize("rel", [
+        "dqn/dqn.yaml", "ppo/ppo.yaml", "dqn/dqn_rainbow.yaml",
+        "td3.yaml", "ddpg/ddpg.yaml", "multi_agent/maddpg.yaml",
+    ])
     def test_reference_yaml_trains(self, rel):
         import yaml as _yaml
 

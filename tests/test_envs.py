@@ -150,3 +150,46 @@ class TestBreakoutLite:
         # ball physics must hit some bricks within 400 random steps
         assert total_r > 0
         assert "BreakoutLite-v0" in __import__("agilerl_amd.envs", fromlist=["ENV_REGISTRY"]).ENV_REGISTRY
+
+
+class TestLunarLanderContinuous:
+    def test_contract_and_throttle_mapping(self):
+        from agilerl_amd.envs.lunar_lander import LunarLanderContinuousVecEnv
+
+        env = LunarLanderContinuousVecEnv(num_envs=4, seed=0)
+        obs, _ = env.reset()
+        assert obs.shape == (4, 8)
+        assert env.single_action_space.shape == (2,)
+        # full main throttle decelerates the fall faster than no throttle
+        import copy
+
+        env2 = LunarLanderContinuousVecEnv(num_envs=4, seed=0)
+        env2.reset()
+        env2.state = env.state.copy()
+        env2.prev_shaping = env.prev_shaping.copy()
+        for _ in range(10):
+            o1, *_r1 = env.step(np.tile([1.0, 0.0], (4, 1)))    # full main
+            o2, *_r2 = env2.step(np.tile([-1.0, 0.0], (4, 1)))  # engines off
+        assert (env.state[:, 3] > env2.state[:, 3]).all()  # vy higher w/ thrust
+
+    def test_lateral_spins(self):
+        from agilerl_amd.envs.lunar_lander import LunarLanderContinuousVecEnv
+
+        env = LunarLanderContinuousVecEnv(num_envs=2, seed=1)
+        env.reset()
+        env.state[:, 5] = 0.0  # zero angular velocity
+        env.step(np.tile([0.0, 1.0], (2, 1)))
+        spin_right = env.state[:, 5].copy()
+        env.state[:, 5] = 0.0
+        env.step(np.tile([0.0, -1.0], (2, 1)))
+        spin_left = env.state[:, 5].copy()
+        assert (spin_right < 0).all() and (spin_left > 0).all()
+
+    def test_dead_zone(self):
+        from agilerl_amd.envs.lunar_lander import LunarLanderContinuousVecEnv
+
+        env = LunarLanderContinuousVecEnv(num_envs=2, seed=2)
+        env.reset()
+        env.state[:, 5] = 0.0
+        env.step(np.tile([0.0, 0.3], (2, 1)))  # below the 0.5 threshold
+        assert np.allclose(env.state[:, 5], 0.0, atol=1e-9)
