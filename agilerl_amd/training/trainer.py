@@ -9,6 +9,7 @@ the per-workload loop :768-870).
 from __future__ import annotations
 
 import importlib
+import os
 from typing import Any, Dict, List, Optional, Union
 
 import numpy as np
@@ -253,7 +254,38 @@ class LocalTrainer(Trainer):
             loggers=self.loggers,
             max_wall_seconds=t.max_wall_seconds,
         )
-        if workload == "off_policy" or workload == "offline":
+        if workload == "offline":
+            # reference CQN flow: learn from a stored transition dataset
+            # (minari analog); environment.dataset_path points at an npz
+            # written by training.train_offline.save_transitions.  With no
+            # dataset the loop collects a random-policy dataset first
+            # (offline-on-synthetic, matching the bench/test fixtures).
+            from .train_offline import (
+                collect_transitions,
+                load_transitions,
+                train_offline,
+            )
+
+            ds_path = m.environment.get("dataset_path") or m.environment.get(
+                "minari_dataset_id"
+            )
+            if ds_path and os.path.exists(str(ds_path)):
+                dataset = load_transitions(str(ds_path))
+            else:
+                if ds_path:
+                    import warnings
+
+                    warnings.warn(
+                        f"offline dataset '{ds_path}' not found locally "
+                        "(no hub access); collecting a random-policy dataset",
+                        RuntimeWarning,
+                    )
+                dataset = collect_transitions(env, steps=2000)
+            return train_offline(
+                env, m.environment.get("env_id", "env"), dataset,
+                m.algorithm.name, pop, self._make_buffer(), **common,
+            )
+        if workload == "off_policy":
             from .train_off_policy import train_off_policy
 
             return train_off_policy(
