@@ -326,6 +326,26 @@ class LocalTrainer(Trainer):
         spec = self.manifest.env_spec()
         from ..envs.bandit import SyntheticBanditEnv
 
+        # reference bandit manifests point at labelled-dataset CSVs
+        # (environment.features/targets — configs/training/bandit/*)
+        features = getattr(spec, "features", None)
+        targets = getattr(spec, "targets", None)
+        if features and targets:
+            import numpy as _np
+
+            from ..envs.bandit import BanditEnv
+
+            def _load_csv(path):
+                return _np.genfromtxt(path, delimiter=",", skip_header=0)
+
+            f = _load_csv(str(features))
+            t = _load_csv(str(targets))
+            if f.ndim == 2 and _np.isnan(f[0]).any():  # header row
+                f = f[1:]
+            t = t.reshape(-1)
+            if t.size and _np.isnan(t.astype(float, copy=False)[0]):
+                t = t[1:]
+            return BanditEnv(f, t)
         if spec.context_dim is not None:
             return SyntheticBanditEnv(
                 context_dim=spec.context_dim, num_arms=spec.num_arms or 4,

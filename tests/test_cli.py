@@ -226,3 +226,25 @@ class TestReferenceFieldAliases:
         assert m.training.target == 200.0
         # rl_hp_selection bounds flow too
         assert m.mutation.rl_hp_selection["lr"]["max"] == 0.01
+
+
+class TestReferenceBanditConfig:
+    def test_reference_neural_ucb_yaml_with_reference_csvs(self):
+        """The reference bandit manifest + the reference's own IRIS CSVs
+        train through the labelled-dataset BanditEnv."""
+        import yaml as _yaml
+
+        from agilerl_amd.models.manifest import TrainingManifest
+        from agilerl_amd.training.trainer import LocalTrainer
+
+        path = "/root/reference/configs/training/bandit/neural_ucb.yaml"
+        feats = "/root/reference/tests/data/iris_features.csv"
+        if not (os.path.exists(path) and os.path.exists(feats)):
+            pytest.skip("reference assets absent")
+        doc = _yaml.safe_load(open(path))
+        doc["environment"]["features"] = feats
+        doc["environment"]["targets"] = "/root/reference/tests/data/iris_targets.csv"
+        doc["training"].update({"max_steps": 200, "pop_size": 2, "evo_steps": 100})
+        m = TrainingManifest.model_validate(doc)
+        out = LocalTrainer(m, device="cpu").train()
+        assert out is not None
