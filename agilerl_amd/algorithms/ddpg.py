@@ -55,6 +55,7 @@ class DDPG(RLAlgorithm):
         theta: float = 0.15,
         dt: float = 1e-2,
         latent_dim: int = 64,
+        share_encoders: bool = False,
         actor_network=None,
         device: str = "cpu",
     ):
@@ -100,6 +101,19 @@ class DDPG(RLAlgorithm):
             for p in net.parameters():
                 p.requires_grad = False
 
+        # reference ddpg.py:315: critic's encoder is pinned to a detached copy
+        # of the actor's (trained only through the actor loss); re-pinned after
+        # every mutation via the hook below
+        self.share_encoders = bool(share_encoders)
+        if self.share_encoders:
+            if not (hasattr(self.actor, "encoder") and hasattr(self.critic, "encoder")):
+                import warnings
+
+                warnings.warn("share_encoders disabled: actor/critic has no encoder")
+                self.share_encoders = False
+            else:
+                self.share_encoder_parameters()
+
         self.actor_optimizer = OptimizerWrapper(torch.optim.Adam, [self.actor], lr=self.lr_actor)
         self.critic_optimizer = OptimizerWrapper(torch.optim.Adam, [self.critic], lr=self.lr_critic)
 
@@ -116,6 +130,8 @@ class DDPG(RLAlgorithm):
             OptimizerConfig(name="critic_optimizer", networks=["critic"], lr_name="lr_critic")
         )
         self.register_mutation_hook("_sync_targets_after_mutation")
+        if self.share_encoders:
+            self.register_mutation_hook("share_encoder_parameters")
 
     def _sync_targets_after_mutation(self) -> None:
         self.actor_target.load_state_dict(self.actor.state_dict())
@@ -123,6 +139,17 @@ class DDPG(RLAlgorithm):
         for net in (self.actor_target, self.critic_target):
             for p in net.parameters():
                 p.requires_grad = False
+
+    def share_encoder_parameters(self) -> None:
+        """Copy the actor's encoder weights into the critic (+ target) and
+        freeze them there, so only the actor loss trains the encoder."""
+        state = {k: v.detach().clone() for k, v in self.actor.encoder.state_dict().items()}
+        for net in (self.critic, self.critic_target):
+            enc = getattr(net, "encoder", None)
+            if enc is not None:
+                enc.load_state_dict(state)
+        for p in self.critic.encoder.parameters():
+            p.requires_grad = False
 
     # ------------------------------------------------------------------
     @property

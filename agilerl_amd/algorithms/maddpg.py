@@ -56,7 +56,11 @@ class MADDPG(MultiAgentRLAlgorithm):
         learn_step: int = 5,
         gamma: float = 0.95,
         tau: float = 1e-2,
+        O_U_noise: bool = True,
         expl_noise: float = 0.1,
+        mean_noise: float = 0.0,
+        theta: float = 0.15,
+        dt: float = 1e-2,
         latent_dim: int = 64,
         shared_agent_groups: Optional[List[List[str]]] = None,
         actor_networks: Optional[Dict[str, Any]] = None,
@@ -73,7 +77,13 @@ class MADDPG(MultiAgentRLAlgorithm):
         self.lr_critic = float(lr_critic)
         self.gamma = float(gamma)
         self.tau = float(tau)
+        self.O_U_noise = bool(O_U_noise)
         self.expl_noise = float(expl_noise)
+        self.mean_noise = float(mean_noise)
+        self.theta = float(theta)
+        self.dt = float(dt)
+        # per-agent Ornstein-Uhlenbeck state (reference maddpg.py:134 O_U_noise)
+        self._ou_state: Dict[str, torch.Tensor] = {}
         self.net_config = net_config
         self.latent_dim = latent_dim
 
@@ -161,6 +171,29 @@ class MADDPG(MultiAgentRLAlgorithm):
                 p.requires_grad = False
 
     # ------------------------------------------------------------------
+    def _sample_noise(self, aid: str, like: torch.Tensor) -> torch.Tensor:
+        """OU (stateful, mean-reverting) or Gaussian exploration noise for one
+        agent.  Reference maddpg.py O_U_noise/mean_noise/theta/dt."""
+        if not self.O_U_noise:
+            return torch.randn_like(like) * self.expl_noise + self.mean_noise
+        state = self._ou_state.get(aid)
+        if state is None or state.shape != like.shape or state.device != like.device:
+            state = torch.zeros_like(like)
+        dx = self.theta * (self.mean_noise - state) * self.dt + (
+            self.expl_noise * (self.dt ** 0.5)
+        ) * torch.randn_like(like)
+        state = state + dx
+        self._ou_state[aid] = state
+        return state
+
+    def reset_action_noise(self, indices=None) -> None:
+        """Reset OU state (per finished env row, or all)."""
+        for state in self._ou_state.values():
+            if indices is None:
+                state.zero_()
+            else:
+                state[np.asarray(indices)] = 0.0
+
     def get_action(
         self, obs: Dict[str, np.ndarray], training: bool = True, **kwargs
     ) -> Tuple[Dict[str, np.ndarray], Dict[str, np.ndarray]]:
@@ -182,8 +215,8 @@ class MADDPG(MultiAgentRLAlgorithm):
                     raw_actions[aid] = raw if device_native else raw.cpu().numpy()
                 else:
                     a = out
-                    if training and self.expl_noise > 0:
-                        a = a + torch.randn_like(a) * self.expl_noise
+                    if training and (self.expl_noise > 0 or self.mean_noise != 0):
+                        a = a + self._sample_noise(aid, a)
                     low = torch.as_tensor(space.low, device=a.device, dtype=a.dtype)
                     high = torch.as_tensor(space.high, device=a.device, dtype=a.dtype)
                     a = a.clamp(low.min(), high.max())

@@ -39,7 +39,11 @@ class MATD3(MADDPG):
         gamma: float = 0.95,
         tau: float = 1e-2,
         policy_freq: int = 2,
+        O_U_noise: bool = True,
         expl_noise: float = 0.1,
+        mean_noise: float = 0.0,
+        theta: float = 0.15,
+        dt: float = 1e-2,
         latent_dim: int = 64,
         actor_networks: Optional[Dict[str, Any]] = None,
         device: str = "cpu",
@@ -48,7 +52,8 @@ class MATD3(MADDPG):
             observation_spaces, action_spaces, agent_ids=agent_ids, index=index,
             hp_config=hp_config, net_config=net_config, head_config=head_config,
             batch_size=batch_size, lr_actor=lr_actor, lr_critic=lr_critic,
-            learn_step=learn_step, gamma=gamma, tau=tau, expl_noise=expl_noise,
+            learn_step=learn_step, gamma=gamma, tau=tau, O_U_noise=O_U_noise,
+            expl_noise=expl_noise, mean_noise=mean_noise, theta=theta, dt=dt,
             latent_dim=latent_dim, actor_networks=actor_networks, device=device,
         )
         self.algo = "MATD3"

@@ -58,6 +58,7 @@ class PPO(RLAlgorithm):
         target_kl: Optional[float] = None,
         normalize_advantage: bool = True,
         log_std_init: float = 0.0,
+        action_std_init: Optional[float] = None,
         latent_dim: int = 64,
         recurrent: bool = False,
         share_encoders: bool = False,
@@ -85,6 +86,12 @@ class PPO(RLAlgorithm):
         self.max_grad_norm = float(max_grad_norm)
         self.target_kl = target_kl
         self.normalize_advantage = normalize_advantage
+        if action_std_init is not None:
+            # reference ppo.py:143 `action_std_init` — despite the name it is
+            # the initial LOG standard deviation (docstring ppo.py:279)
+            log_std_init = float(action_std_init)
+        self.action_std_init = action_std_init
+        self.log_std_init = float(log_std_init)
         self.net_config = net_config
         self.latent_dim = latent_dim
         self.recurrent = recurrent

@@ -40,7 +40,11 @@ class TD3(DDPG):
         noise_clip: float = 0.5,
         O_U_noise: bool = True,
         expl_noise: float = 0.1,
+        mean_noise: float = 0.0,
+        theta: float = 0.15,
+        dt: float = 1e-2,
         latent_dim: int = 64,
+        share_encoders: bool = False,
         actor_network=None,
         device: str = "cpu",
     ):
@@ -49,7 +53,8 @@ class TD3(DDPG):
             net_config=net_config, head_config=head_config, batch_size=batch_size,
             lr_actor=lr_actor, lr_critic=lr_critic, learn_step=learn_step,
             gamma=gamma, tau=tau, policy_freq=policy_freq, O_U_noise=O_U_noise,
-            expl_noise=expl_noise, latent_dim=latent_dim,
+            expl_noise=expl_noise, mean_noise=mean_noise, theta=theta, dt=dt,
+            latent_dim=latent_dim, share_encoders=share_encoders,
             actor_network=actor_network, device=device,
         )
         self.algo = "TD3"
@@ -74,10 +79,24 @@ class TD3(DDPG):
             OptimizerConfig(name="critic_2_optimizer", networks=["critic_2"], lr_name="lr_critic")
         )
         self.register_mutation_hook("_sync_twin_after_mutation")
+        if self.share_encoders:
+            self.share_encoder_parameters()  # now also pins critic_2
 
     def _sync_twin_after_mutation(self) -> None:
         self.critic_2_target.load_state_dict(self.critic_2.state_dict())
         for p in self.critic_2_target.parameters():
+            p.requires_grad = False
+
+    def share_encoder_parameters(self) -> None:
+        super().share_encoder_parameters()
+        if not hasattr(self, "critic_2"):  # called from DDPG.__init__ pre-twin
+            return
+        state = {k: v.detach().clone() for k, v in self.actor.encoder.state_dict().items()}
+        for net in (self.critic_2, self.critic_2_target):
+            enc = getattr(net, "encoder", None)
+            if enc is not None:
+                enc.load_state_dict(state)
+        for p in self.critic_2.encoder.parameters():
             p.requires_grad = False
 
     # ------------------------------------------------------------------

@@ -102,7 +102,14 @@ class LocalTrainer(Trainer):
         workload = algo_workload(m.algorithm.name)
         from ..models.algorithms import merged_hyperparameters
 
-        hps = merged_hyperparameters(m.algorithm.name, m.algorithm.hyperparameters)
+        # reference manifests put algorithm kwargs at the TOP level of the
+        # algorithm section (batch_size/lr/recurrent/... — configs/
+        # training/*); fold extras under the nested hyperparameters dict
+        extra = dict(getattr(m.algorithm, "model_extra", None) or {})
+        extra.pop("name", None)
+        hps = merged_hyperparameters(
+            m.algorithm.name, {**extra, **m.algorithm.hyperparameters}
+        )
         net = m.network
         net_config = dict(net.encoder_config) or None
         if net.arch and net_config is not None:
@@ -110,6 +117,12 @@ class LocalTrainer(Trainer):
         elif net.arch:
             net_config = {"arch": net.arch}
         kwargs: Dict[str, Any] = dict(hps)
+        if hps.get("recurrent") and net_config is not None:
+            # reference recurrent manifests give LSTM encoder fields
+            # without an arch tag (ppo_recurrent.yaml)
+            net_config.setdefault("arch", "lstm")
+        elif hps.get("recurrent"):
+            net_config = {"arch": "lstm"}
         if net_config:
             kwargs["net_config"] = net_config
         if net.head_config:
@@ -189,10 +202,24 @@ class LocalTrainer(Trainer):
 
     def _make_hpo(self):
         m = self.manifest
-        tournament = TournamentSelection(
-            tournament_size=m.selection_strategy.tournament_size,
-            elitism=m.selection_strategy.elitism,
-        )
+        strat = getattr(m.selection_strategy, "strategy", None)
+        if strat == "multi_frequency":
+            # reference MF-PBT section (configs/training/*_mfpbt.yaml):
+            # evolution_frequency_ratios map to subpopulation frequencies
+            from ..hpo.multi_frequency import MultiFrequencySelection
+
+            ratios = getattr(
+                m.selection_strategy, "evolution_frequency_ratios", None
+            ) or (1, 2, 4)
+            tournament = MultiFrequencySelection(
+                frequencies=tuple(int(r) for r in ratios),
+                elitism=m.selection_strategy.elitism,
+            )
+        else:
+            tournament = TournamentSelection(
+                tournament_size=m.selection_strategy.tournament_size,
+                elitism=m.selection_strategy.elitism,
+            )
         p = m.mutation.probabilities
         mutations = Mutations(
             no_mutation=p.no_mutation,

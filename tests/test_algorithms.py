@@ -176,6 +176,43 @@ class TestSharedEncoders:
         x = torch.randn(3, 6)
         assert torch.allclose(c.actor(x), a.actor(x))
 
+    def test_ddpg_pinned_encoder(self):
+        """DDPG share_encoders pins a detached copy of the actor's encoder
+        into the critic (+ targets) — reference ddpg.py:315 semantics: the
+        critic loss never trains the encoder, re-pinned after mutations."""
+        from agilerl_amd.algorithms.ddpg import DDPG
+
+        a = DDPG(Box(-1, 1, (5,)), Box(-1, 1, (2,)), share_encoders=True)
+        sd_a = a.actor.encoder.state_dict()
+        sd_c = a.critic.encoder.state_dict()
+        for k in sd_a:
+            assert torch.equal(sd_a[k], sd_c[k])
+        assert all(not p.requires_grad for p in a.critic.encoder.parameters())
+        # mutation hook re-pins after architecture change
+        a.apply_architecture_mutation("encoder.add_node", numb_new_nodes=8, hidden_layer=0)
+        a.mutation_hook()
+        for k, v in a.actor.encoder.state_dict().items():
+            assert torch.equal(v, a.critic.encoder.state_dict()[k])
+        assert all(not p.requires_grad for p in a.critic.encoder.parameters())
+
+    def test_td3_pinned_encoder_covers_twin(self):
+        from agilerl_amd.algorithms.td3 import TD3
+
+        a = TD3(Box(-1, 1, (5,)), Box(-1, 1, (2,)), share_encoders=True)
+        for k, v in a.actor.encoder.state_dict().items():
+            assert torch.equal(v, a.critic_2.encoder.state_dict()[k])
+        assert all(not p.requires_grad for p in a.critic_2.encoder.parameters())
+
+    def test_ppo_action_std_init_alias(self):
+        """Reference ppo.py:143 action_std_init is the initial LOG std."""
+        from agilerl_amd.algorithms.ppo import PPO
+
+        a = PPO(Box(-1, 1, (4,)), Box(-1, 1, (2,)), action_std_init=0.4)
+        log_std = a.actor.dist_layer.log_std
+        assert torch.allclose(log_std, torch.full_like(log_std, 0.4))
+        c = a.clone(1)
+        assert torch.allclose(c.actor.dist_layer.log_std, log_std)
+
     def test_resnet_encoder_q(self):
         from agilerl_amd.networks import QNetwork
         from agilerl_amd.spaces import Box, Discrete

@@ -184,7 +184,7 @@ class TestReferenceConfigEndToEnd:
     @pytest.mark.parametrize("rel", [
         "dqn/dqn.yaml", "ppo/ppo.yaml", "dqn/dqn_rainbow.yaml",
         "td3.yaml", "ddpg/ddpg.yaml", "multi_agent/maddpg.yaml",
-        "cqn.yaml",
+        "cqn.yaml", "dqn/dqn_mfpbt.yaml", "ppo/ppo_recurrent.yaml",
     ])
     def test_reference_yaml_trains(self, rel):
         import yaml as _yaml

@@ -62,6 +62,32 @@ class TestMADDPG:
         loss = agent.learn(buf.sample(32))
         assert np.isfinite(loss)
 
+    def test_ou_noise_stateful_and_resettable(self):
+        """Reference maddpg.py:134 O_U_noise: mean-reverting stateful noise
+        per continuous agent, zeroed by reset_action_noise."""
+        from agilerl_amd.spaces import Box
+
+        obs_sp = {a: Box(-1, 1, (4,)) for a in ("a0", "a1")}
+        act_sp = {a: Box(-1, 1, (2,)) for a in ("a0", "a1")}
+        agent = MADDPG(obs_sp, act_sp, agent_ids=["a0", "a1"],
+                       O_U_noise=True, expl_noise=0.3,
+                       net_config={"arch": "mlp", "hidden_size": [16]})
+        obs = {a: np.random.randn(3, 4).astype(np.float32) for a in ("a0", "a1")}
+        torch.manual_seed(0)
+        agent.get_action(obs, training=True)
+        assert set(agent._ou_state) == {"a0", "a1"}
+        s1 = agent._ou_state["a0"].clone()
+        agent.get_action(obs, training=True)
+        assert not torch.equal(agent._ou_state["a0"], s1)  # state evolves
+        agent.reset_action_noise()
+        assert torch.all(agent._ou_state["a0"] == 0)
+        # Gaussian mode keeps no state
+        a2 = MADDPG(obs_sp, act_sp, agent_ids=["a0", "a1"],
+                    O_U_noise=False, mean_noise=0.05,
+                    net_config={"arch": "mlp", "hidden_size": [16]})
+        a2.get_action(obs, training=True)
+        assert not a2._ou_state
+
     def test_clone_and_mutation(self):
         env, pop = _sl_pop(MADDPG, 1)
         agent = pop[0]
