@@ -54,6 +54,7 @@ class AgentType(str, Enum):
 _LAZY = {
     "LocalTrainer": "agilerl_amd.training.trainer",
     "Trainer": "agilerl_amd.training.trainer",
+    "ArenaTrainer": "agilerl_amd.training.trainer",
     "Population": "agilerl_amd.population",
     "TournamentSelection": "agilerl_amd.hpo.tournament",
     "Mutations": "agilerl_amd.hpo.mutation",
@@ -90,6 +91,7 @@ __all__ = [
     "HAS_LIGER_KERNEL",
     "LocalTrainer",
     "Trainer",
+    "ArenaTrainer",
     "Population",
     "TournamentSelection",
     "Mutations",

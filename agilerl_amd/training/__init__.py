@@ -1,4 +1,4 @@
-from .trainer import Trainer, LocalTrainer
+from .trainer import Trainer, LocalTrainer, ArenaTrainer
 from .train_off_policy import train_off_policy, save_population_checkpoint
 from .train_on_policy import train_on_policy
 from .train_multi_agent_off_policy import train_multi_agent_off_policy
@@ -10,6 +10,7 @@ from .train_distributed import train_on_policy_distributed
 __all__ = [
     "Trainer",
     "LocalTrainer",
+    "ArenaTrainer",
     "train_off_policy",
     "train_on_policy",
     "train_multi_agent_off_policy",

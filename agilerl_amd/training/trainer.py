@@ -392,3 +392,71 @@ class LocalTrainer(Trainer):
         from .llm import run_llm_workload
 
         return run_llm_workload(self, workload)
+
+
+class ArenaTrainer(Trainer):
+    """Submits training jobs to an Arena service instead of running locally.
+
+    Reference parity: ``agilerl/training/trainer.py:872`` (ArenaTrainer) —
+    same surface: construct from a manifest (``from_manifest``) or sections,
+    ``train()`` submits via an :class:`~agilerl_amd.arena.ArenaClient`
+    (local workspace store or a remote :mod:`agilerl_amd.arena.service`
+    endpoint), plus resume / listing / streaming helpers.  Arena runs
+    tournament selection only, so MF-PBT manifests are rejected
+    (reference trainer.py:914-921).
+    """
+
+    def __init__(
+        self,
+        manifest: TrainingManifest,
+        device: str = "cpu",
+        loggers: Optional[List] = None,
+        client=None,
+        api_key: Optional[str] = None,
+        base_url: Optional[str] = None,
+    ):
+        super().__init__(manifest, device=device, loggers=loggers)
+        strategy = getattr(manifest.selection_strategy, "strategy", None)
+        if strategy == "multi_frequency":
+            raise ValueError(
+                "ArenaTrainer only supports tournament selection: MF-PBT is "
+                "not available on Arena. Use LocalTrainer to run MF-PBT."
+            )
+        if client is None:
+            from ..arena import ArenaClient
+
+            client = ArenaClient(api_key=api_key, base_url=base_url)
+        self._client = client
+
+    @property
+    def client(self):
+        return self._client
+
+    def train(self, run: bool = True, experiment_name: Optional[str] = None):
+        """Submit the manifest; returns an ExperimentHandle (reference
+        ArenaTrainer.train returns the Arena API response)."""
+        handle = self._client.submit_experiment(
+            self.manifest, run=run, device=self.device
+        )
+        if experiment_name:
+            handle.name = experiment_name
+        return handle
+
+    def resume_from_checkpoint(self, experiment_id: str,
+                               max_steps: Optional[int] = None):
+        if max_steps is not None:
+            # bump the budget before resubmitting (reference passes max_steps)
+            self.manifest.training.max_steps = int(max_steps)
+        return self._client.resume_experiment(experiment_id, device=self.device)
+
+    def list_experiments(self):
+        return self._client.list_experiments()
+
+    def list_checkpoints(self, experiment_id: str):
+        return self._client.list_checkpoints(experiment_id)
+
+    def stream(self, experiment_id: str, follow: float = 0.0):
+        return self._client.stream_experiment(experiment_id, follow=follow)
+
+    def wait(self, experiment_id: str, timeout: float = 300.0):
+        return self._client.wait_for_completion(experiment_id, timeout=timeout)

@@ -46,6 +46,40 @@ class TestArenaClient:
         assert resumed.status == "completed"
 
 
+class TestArenaTrainer:
+    """Reference trainer.py:872 ArenaTrainer: manifest -> submit via client."""
+
+    def _manifest(self):
+        return {
+            "algorithm": {"name": "DQN", "hyperparameters": {"batch_size": 32, "lr": 1e-3}},
+            "environment": {"env_id": "CartPole-v1", "num_envs": 4},
+            "network": {"arch": "mlp", "encoder_config": {"hidden_size": [16]}},
+            "training": {"max_steps": 300, "pop_size": 1, "evo_steps": 150, "eval_loop": 1},
+        }
+
+    def test_train_submits_and_completes(self, tmp_path):
+        from agilerl_amd.training import ArenaTrainer
+
+        client = ArenaClient(workspace=str(tmp_path))
+        client.login()
+        trainer = ArenaTrainer.from_manifest(self._manifest(), client=client)
+        handle = trainer.train()
+        assert handle.status == "completed"
+        assert handle.experiment_id in trainer.list_experiments()
+        # resume re-runs from the stored manifest
+        resumed = trainer.resume_from_checkpoint(handle.experiment_id, max_steps=300)
+        assert resumed.status == "completed"
+
+    def test_rejects_mfpbt(self, tmp_path):
+        from agilerl_amd.training import ArenaTrainer
+
+        m = self._manifest()
+        m["selection_strategy"] = {"strategy": "multi_frequency",
+                                   "evolution_frequency_ratios": [1, 2]}
+        with pytest.raises(ValueError, match="MF-PBT"):
+            ArenaTrainer.from_manifest(m, client=ArenaClient(workspace=str(tmp_path)))
+
+
 class TestArenaDatasetsAndDeploy:
     def test_dataset_roundtrip(self, tmp_path):
         import numpy as np
