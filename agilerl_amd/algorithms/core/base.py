@@ -103,6 +103,58 @@ class EvolvableAlgorithm(metaclass=AlgorithmMeta):
         self._wrapped: Dict[str, nn.Module] = {}
 
     # ------------------------------------------------------------------
+    # Reference-API compatibility kwargs
+    # ------------------------------------------------------------------
+    _COMPAT_KWARGS = ("accelerator", "wrap", "mut", "normalize_images", "torch_compiler")
+
+    def _accept_compat_kwargs(self, **kw) -> None:
+        """Accept the reference constructor kwargs that have no MI355X-native
+        role (reference base.py:819-845 Accelerate wrap, torch.compile).
+
+        - ``accelerator``/``wrap``: HF-Accelerate integration — replaced here
+          by one-process-per-GPU RCCL (``agilerl_amd.parallel``); warned if set.
+        - ``torch_compiler``: hot paths run hand-written HIP kernels and
+          hipGraph capture instead of a tracing compiler; warned if set.
+        - ``mut``: last-mutation tag (stored; the HPO loop overwrites it).
+        - ``normalize_images``: uint8 image obs are always scaled by 1/255
+          (warned when explicitly disabled).
+        Unknown keys raise TypeError exactly like a plain signature would.
+        """
+        import warnings
+
+        unknown = set(kw) - set(self._COMPAT_KWARGS)
+        if unknown:
+            raise TypeError(
+                f"{type(self).__name__}.__init__() got an unexpected keyword "
+                f"argument '{sorted(unknown)[0]}'"
+            )
+        if kw.get("accelerator") is not None:
+            warnings.warn(
+                "`accelerator` is accepted for reference-API compatibility but "
+                "ignored: use agilerl_amd.parallel (one process per GPU over "
+                "RCCL) for distributed training.",
+                RuntimeWarning,
+            )
+            if hasattr(self, "_init_args"):
+                self._init_args["accelerator"] = None  # keep clones picklable
+        self.accelerator = None
+        if kw.get("torch_compiler"):
+            warnings.warn(
+                "`torch_compiler` ignored: hot paths use hand-written HIP "
+                "kernels and hipGraph capture.",
+                RuntimeWarning,
+            )
+        if kw.get("mut") is not None:
+            self.mut = kw["mut"]
+        self.normalize_images = bool(kw.get("normalize_images", True))
+        if not self.normalize_images:
+            warnings.warn(
+                "normalize_images=False ignored: uint8 image observations are "
+                "always scaled by 1/255.",
+                RuntimeWarning,
+            )
+
+    # ------------------------------------------------------------------
     # Registry
     # ------------------------------------------------------------------
     def register_network_group(self, group: NetworkGroup) -> None:

@@ -38,12 +38,14 @@ class CQN(DQN):
         latent_dim: int = 64,
         actor_network=None,
         device: str = "cpu",
+        **kwargs,
     ):
         super().__init__(
             observation_space, action_space, index=index, hp_config=hp_config,
             net_config=net_config, head_config=head_config, batch_size=batch_size,
             lr=lr, learn_step=learn_step, gamma=gamma, tau=tau, double=double,
             latent_dim=latent_dim, actor_network=actor_network, device=device,
+            **kwargs,
         )
         self.algo = "CQN"
         self.cql_alpha = float(cql_alpha)

@@ -54,15 +54,19 @@ class DDPG(RLAlgorithm):
         mean_noise: float = 0.0,
         theta: float = 0.15,
         dt: float = 1e-2,
+        vect_noise_dim: int = 1,
         latent_dim: int = 64,
         share_encoders: bool = False,
         actor_network=None,
+        critic_network=None,
         device: str = "cpu",
+        **kwargs,
     ):
         super().__init__(
             observation_space, action_space, index=index, learn_step=learn_step,
             device=device, hp_config=hp_config or default_hp_config(), name="DDPG",
         )
+        self._accept_compat_kwargs(**kwargs)
         self.batch_size = int(batch_size)
         self.lr_actor = float(lr_actor)
         self.lr_critic = float(lr_critic)
@@ -77,7 +81,12 @@ class DDPG(RLAlgorithm):
         self.net_config = net_config
         self.latent_dim = latent_dim
         self._learn_counter = 0
-        self._ou_state: Optional[np.ndarray] = None
+        # reference ddpg.py:119 vect_noise_dim: per-env OU noise rows,
+        # presized here and auto-resized on the first batched get_action
+        self.vect_noise_dim = int(vect_noise_dim)
+        self._ou_state: Optional[np.ndarray] = (
+            np.zeros((self.vect_noise_dim, self.action_dim)) if vect_noise_dim > 1 else None
+        )
 
         if actor_network is not None:
             # user-supplied policy net (reference ddpg.py actor_network);
@@ -92,10 +101,17 @@ class DDPG(RLAlgorithm):
                 head_config=head_config, latent_dim=latent_dim, device=device,
             )
         self.actor_target = self.actor.clone()
-        self.critic = ContinuousQNetwork(
-            observation_space, action_space, encoder_config=net_config,
-            head_config=head_config, latent_dim=latent_dim, device=device,
-        )
+        if critic_network is not None:
+            # user-supplied (state, action) -> Q net (reference ddpg.py:136)
+            from ..networks.base import CustomQAdapter
+
+            self.critic = CustomQAdapter(critic_network, observation_space,
+                                         action_space=action_space, device=device)
+        else:
+            self.critic = ContinuousQNetwork(
+                observation_space, action_space, encoder_config=net_config,
+                head_config=head_config, latent_dim=latent_dim, device=device,
+            )
         self.critic_target = self.critic.clone()
         for net in (self.actor_target, self.critic_target):
             for p in net.parameters():

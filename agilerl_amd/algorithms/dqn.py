@@ -52,6 +52,7 @@ class DQN(RLAlgorithm):
         normalize_images: bool = True,
         actor_network=None,
         device: str = "cpu",
+        **kwargs,
     ):
         super().__init__(
             observation_space,
@@ -62,6 +63,7 @@ class DQN(RLAlgorithm):
             hp_config=hp_config or default_hp_config(),
             name="DQN",
         )
+        self._accept_compat_kwargs(**kwargs)
         self.batch_size = int(batch_size)
         self.lr = float(lr)
         self.gamma = float(gamma)

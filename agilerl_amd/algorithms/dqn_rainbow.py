@@ -51,8 +51,11 @@ class RainbowDQN(RLAlgorithm):
         v_max: float = 10.0,
         n_step: int = 3,
         noise_std: float = 0.5,
+        combined_reward: bool = False,
         latent_dim: int = 64,
+        actor_network=None,
         device: str = "cpu",
+        **kwargs,
     ):
         super().__init__(
             observation_space,
@@ -63,6 +66,7 @@ class RainbowDQN(RLAlgorithm):
             hp_config=hp_config or default_hp_config(),
             name="RainbowDQN",
         )
+        self._accept_compat_kwargs(**kwargs)
         self.batch_size = int(batch_size)
         self.lr = float(lr)
         self.gamma = float(gamma)
@@ -74,22 +78,36 @@ class RainbowDQN(RLAlgorithm):
         self.v_max = float(v_max)
         self.n_step = int(n_step)
         self.noise_std = noise_std
+        # reference dqn_rainbow.py:124: sum the 1-step and n-step losses
+        # (requires the buffer to emit the *_1step fields; the off-policy
+        # loop passes include_one_step when this flag is set)
+        self.combined_reward = bool(combined_reward)
         self.net_config = net_config
         self.latent_dim = latent_dim
         self.last_td_errors: Optional[torch.Tensor] = None
 
-        self.actor = RainbowQNetwork(
-            observation_space,
-            action_space,
-            encoder_config=net_config,
-            head_config=head_config,
-            latent_dim=latent_dim,
-            num_atoms=num_atoms,
-            v_min=v_min,
-            v_max=v_max,
-            noise_std=noise_std,
-            device=device,
-        )
+        if actor_network is not None:
+            # user-supplied distributional Q net: must expose .dist()/.support
+            # like RainbowQNetwork (reference dqn_rainbow.py:125)
+            if not (hasattr(actor_network, "dist") and hasattr(actor_network, "support")):
+                raise TypeError(
+                    "RainbowDQN actor_network must expose .dist(obs) and "
+                    ".support (see networks.RainbowQNetwork)"
+                )
+            self.actor = actor_network.to(device)
+        else:
+            self.actor = RainbowQNetwork(
+                observation_space,
+                action_space,
+                encoder_config=net_config,
+                head_config=head_config,
+                latent_dim=latent_dim,
+                num_atoms=num_atoms,
+                v_min=v_min,
+                v_max=v_max,
+                noise_std=noise_std,
+                device=device,
+            )
         self.actor_target = self.actor.clone()
         for p in self.actor_target.parameters():
             p.requires_grad = False
@@ -166,6 +184,22 @@ class RainbowDQN(RLAlgorithm):
         dist = self.actor.dist(self.actor.preprocess(obs))
         log_p = torch.log(dist[torch.arange(B, device=dist.device), actions])
         elementwise_loss = -(target_dist * log_p).sum(-1)
+
+        if self.combined_reward and "reward_1step" in experiences:
+            # reference dqn_rainbow.py:453: add the 1-step loss for the same
+            # anchors to the n-step loss (1-step fields come from the buffer's
+            # include_one_step sampling mode)
+            r1 = experiences["reward_1step"].to(self.device).float().reshape(-1)
+            no1 = experiences["next_obs_1step"]
+            d1 = experiences["done_1step"].to(self.device).float().reshape(-1)
+            with torch.no_grad():
+                nq1 = self.actor(self.actor.preprocess(no1))
+                na1 = nq1.argmax(dim=-1)
+                nd1 = self.actor_target.dist(self.actor_target.preprocess(no1))
+                nd1 = nd1[torch.arange(B, device=nd1.device), na1]
+                target_1 = self._project(nd1, r1, d1, torch.full_like(r1, self.gamma))
+            elementwise_loss = elementwise_loss - (target_1 * log_p).sum(-1)
+
         if weights is not None:
             loss = (elementwise_loss * weights.to(self.device).reshape(-1)).mean()
         else:

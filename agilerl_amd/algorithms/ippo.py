@@ -55,16 +55,22 @@ class IPPO(MultiAgentRLAlgorithm):
         vf_coef: float = 0.5,
         update_epochs: int = 4,
         max_grad_norm: float = 0.5,
+        target_kl: Optional[float] = None,
+        action_std_init: Optional[float] = None,
+        action_batch_size: Optional[int] = None,
         latent_dim: int = 64,
         shared_agent_groups: Optional[List[List[str]]] = None,
         actor_networks: Optional[Dict[str, Any]] = None,
+        critic_networks: Optional[Dict[str, Any]] = None,
         device: str = "cpu",
+        **kwargs,
     ):
         super().__init__(
             observation_spaces, action_spaces, agent_ids=agent_ids, index=index,
             learn_step=learn_step, device=device,
             hp_config=hp_config or default_hp_config(), name="IPPO",
         )
+        self._accept_compat_kwargs(**kwargs)
         self.batch_size = int(batch_size)
         self.lr = float(lr)
         self.gamma = float(gamma)
@@ -74,6 +80,12 @@ class IPPO(MultiAgentRLAlgorithm):
         self.vf_coef = float(vf_coef)
         self.update_epochs = int(update_epochs)
         self.max_grad_norm = float(max_grad_norm)
+        # reference ippo.py: target_kl early-stop, action_std_init (initial
+        # LOG std, ppo.py:279 semantics), action_batch_size (chunked
+        # get_action for very wide vector envs)
+        self.target_kl = target_kl
+        self.action_std_init = action_std_init
+        self.action_batch_size = action_batch_size
         self.net_config = net_config
         self.latent_dim = latent_dim
         self.shared_agent_groups = shared_agent_groups
@@ -100,11 +112,20 @@ class IPPO(MultiAgentRLAlgorithm):
                         self.observation_spaces[leader], self.action_spaces[leader],
                         encoder_config=net_config, head_config=head_config,
                         latent_dim=latent_dim, device=device,
+                        log_std_init=float(action_std_init or 0.0),
                     )
-                critic_mods[leader] = ValueNetwork(
-                    self.observation_spaces[leader], encoder_config=net_config,
-                    head_config=head_config, latent_dim=latent_dim, device=device,
-                )
+                if critic_networks is not None and leader in critic_networks:
+                    from ..networks.base import CustomNetworkAdapter
+
+                    critic_mods[leader] = CustomNetworkAdapter(
+                        critic_networks[leader], self.observation_spaces[leader],
+                        device=device,
+                    )
+                else:
+                    critic_mods[leader] = ValueNetwork(
+                        self.observation_spaces[leader], encoder_config=net_config,
+                        head_config=head_config, latent_dim=latent_dim, device=device,
+                    )
             actors[aid] = actor_mods[leader]
             critics[aid] = critic_mods[leader]
         self.actors = ModuleDict(actors, device=device)
@@ -129,7 +150,15 @@ class IPPO(MultiAgentRLAlgorithm):
                 if not training:
                     env_actions[aid] = actor.deterministic_action(pre).cpu().numpy()
                     continue
-                a, lp, _ = actor.sample(pre)
+                abs_ = self.action_batch_size
+                if abs_ is not None and pre.shape[0] > abs_:
+                    # reference action_batch_size: chunk very wide vector
+                    # envs through the actor to bound activation memory
+                    outs = [actor.sample(pre[i:i + abs_]) for i in range(0, pre.shape[0], abs_)]
+                    a = torch.cat([o[0] for o in outs])
+                    lp = torch.cat([o[1] for o in outs])
+                else:
+                    a, lp, _ = actor.sample(pre)
                 env_actions[aid] = a.cpu().numpy()
                 log_probs[aid] = lp
                 values[aid] = self.critics[aid](self.critics[aid].preprocess(obs[aid])).squeeze(-1)
@@ -149,6 +178,7 @@ class IPPO(MultiAgentRLAlgorithm):
         stats = {"policy_loss": 0.0, "value_loss": 0.0, "entropy": 0.0, "approx_kl": 0.0}
         n = 0
         for _ in range(self.update_epochs):
+            epoch_kl, epoch_mbs = 0.0, 0
             iters = {aid: buffers[aid].get_minibatches(self.batch_size) for aid in self.agent_ids}
             while True:
                 mbs = {}
@@ -175,7 +205,10 @@ class IPPO(MultiAgentRLAlgorithm):
                     vloss = 0.5 * ((v - mb["returns"].reshape(-1)) ** 2).mean()
                     loss = loss + pg + self.vf_coef * vloss - self.ent_coef * ent.mean()
                     with torch.no_grad():
-                        stats["approx_kl"] += float((ratio - 1 - ratio.log()).mean())
+                        kl = float((ratio - 1 - ratio.log()).mean())
+                        stats["approx_kl"] += kl
+                        epoch_kl += kl
+                        epoch_mbs += 1
                     stats["policy_loss"] += float(pg.detach())
                     stats["value_loss"] += float(vloss.detach())
                     stats["entropy"] += float(ent.mean().detach())
@@ -187,6 +220,12 @@ class IPPO(MultiAgentRLAlgorithm):
                 )
                 self.optimizer.step()
                 n += len(mbs)
+            if (
+                self.target_kl is not None
+                and epoch_mbs
+                and epoch_kl / epoch_mbs > self.target_kl
+            ):
+                break  # reference PPO-family target_kl early stop
         if n:
             stats = {k: v / n for k, v in stats.items()}
         return stats

@@ -61,16 +61,20 @@ class MADDPG(MultiAgentRLAlgorithm):
         mean_noise: float = 0.0,
         theta: float = 0.15,
         dt: float = 1e-2,
+        vect_noise_dim: int = 1,
         latent_dim: int = 64,
         shared_agent_groups: Optional[List[List[str]]] = None,
         actor_networks: Optional[Dict[str, Any]] = None,
+        critic_networks: Optional[Dict[str, Any]] = None,
         device: str = "cpu",
+        **kwargs,
     ):
         super().__init__(
             observation_spaces, action_spaces, agent_ids=agent_ids, index=index,
             learn_step=learn_step, device=device,
             hp_config=hp_config or default_hp_config(), name="MADDPG",
         )
+        self._accept_compat_kwargs(**kwargs)
         self.shared_agent_groups = shared_agent_groups
         self.batch_size = int(batch_size)
         self.lr_actor = float(lr_actor)
@@ -82,6 +86,7 @@ class MADDPG(MultiAgentRLAlgorithm):
         self.mean_noise = float(mean_noise)
         self.theta = float(theta)
         self.dt = float(dt)
+        self.vect_noise_dim = int(vect_noise_dim)
         # per-agent Ornstein-Uhlenbeck state (reference maddpg.py:134 O_U_noise)
         self._ou_state: Dict[str, torch.Tensor] = {}
         self.net_config = net_config
@@ -121,15 +126,25 @@ class MADDPG(MultiAgentRLAlgorithm):
             actors[aid] = actor_modules[leader]
         self.actors = ModuleDict(actors, device=device)
         self.actor_targets = self.actors.clone()
-        self.critics = ModuleDict(
-            {
-                aid: ContinuousQNetwork(
-                    joint_space, Box(-1.0, 1.0, (self.joint_action_dim,)),
-                    encoder_config=net_config, head_config=head_config,
-                    latent_dim=latent_dim, action_dim=self.joint_action_dim, device=device,
+        def _make_critic(aid):
+            if critic_networks is not None and aid in critic_networks:
+                # user-supplied centralized critic: (joint obs, joint action)
+                # -> Q (reference maddpg.py critic_networks)
+                from ..networks.base import CustomQAdapter
+
+                return CustomQAdapter(
+                    critic_networks[aid], joint_space,
+                    action_space=Box(-1.0, 1.0, (self.joint_action_dim,)),
+                    device=device,
                 )
-                for aid in self.agent_ids
-            },
+            return ContinuousQNetwork(
+                joint_space, Box(-1.0, 1.0, (self.joint_action_dim,)),
+                encoder_config=net_config, head_config=head_config,
+                latent_dim=latent_dim, action_dim=self.joint_action_dim, device=device,
+            )
+
+        self.critics = ModuleDict(
+            {aid: _make_critic(aid) for aid in self.agent_ids},
             device=device,
         )
         self.critic_targets = self.critics.clone()

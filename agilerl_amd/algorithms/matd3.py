@@ -46,30 +46,55 @@ class MATD3(MADDPG):
         dt: float = 1e-2,
         latent_dim: int = 64,
         actor_networks: Optional[Dict[str, Any]] = None,
+        critic_networks: Optional[Dict[str, Any]] = None,
         device: str = "cpu",
+        **kwargs,
     ):
+        # critic_networks: {agent_id: (critic_1, critic_2)} custom twin
+        # centralized critics (reference matd3.py critic_networks)
+        first = second = None
+        if critic_networks is not None:
+            bad = [a for a, v in critic_networks.items()
+                   if not (isinstance(v, (list, tuple)) and len(v) == 2)]
+            if bad:
+                raise ValueError(
+                    f"MATD3 critic_networks values must be (critic_1, critic_2) pairs; got bad entries for {bad}"
+                )
+            first = {a: v[0] for a, v in critic_networks.items()}
+            second = {a: v[1] for a, v in critic_networks.items()}
         super().__init__(
             observation_spaces, action_spaces, agent_ids=agent_ids, index=index,
             hp_config=hp_config, net_config=net_config, head_config=head_config,
             batch_size=batch_size, lr_actor=lr_actor, lr_critic=lr_critic,
             learn_step=learn_step, gamma=gamma, tau=tau, O_U_noise=O_U_noise,
             expl_noise=expl_noise, mean_noise=mean_noise, theta=theta, dt=dt,
-            latent_dim=latent_dim, actor_networks=actor_networks, device=device,
+            latent_dim=latent_dim, actor_networks=actor_networks,
+            critic_networks=first, device=device,
+            **kwargs,
         )
         self.algo = "MATD3"
         self.policy_freq = int(policy_freq)
         self._learn_counter = 0
 
         joint_space = Box(-np.inf, np.inf, (self.joint_obs_dim,))
-        self.critics_2 = ModuleDict(
-            {
-                aid: ContinuousQNetwork(
-                    joint_space, Box(-1.0, 1.0, (self.joint_action_dim,)),
-                    encoder_config=net_config, head_config=head_config,
-                    latent_dim=latent_dim, action_dim=self.joint_action_dim, device=device,
+
+        def _make_critic_2(aid):
+            if second is not None and aid in second:
+                from ..networks.base import CustomQAdapter
+
+                return CustomQAdapter(
+                    second[aid], joint_space,
+                    action_space=Box(-1.0, 1.0, (self.joint_action_dim,)),
+                    device=device,
                 )
-                for aid in self.agent_ids
-            },
+            return ContinuousQNetwork(
+                joint_space, Box(-1.0, 1.0, (self.joint_action_dim,)),
+                encoder_config=net_config, head_config=head_config,
+                latent_dim=latent_dim, action_dim=self.joint_action_dim, device=device,
+            )
+
+        self.critics_2 = ModuleDict(
+            {aid: _make_critic_2(aid) for aid in self.agent_ids},
             device=device,
         )
         self.critic_2_targets = self.critics_2.clone()

@@ -38,12 +38,14 @@ class NeuralTS(NeuralUCB):
         reg: float = 0.000625,
         latent_dim: int = 64,
         device: str = "cpu",
+        **kwargs,
     ):
         super().__init__(
             observation_space, action_space, index=index, hp_config=hp_config,
             net_config=net_config, head_config=head_config, batch_size=batch_size,
             lr=lr, learn_step=learn_step, gamma=gamma, lamb=lamb, reg=reg,
             latent_dim=latent_dim, device=device,
+            **kwargs,
         )
         self.algo = "NeuralTS"
 

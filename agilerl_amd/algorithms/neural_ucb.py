@@ -47,12 +47,15 @@ class NeuralUCB(RLAlgorithm):
         lamb: float = 1.0,
         reg: float = 0.000625,
         latent_dim: int = 64,
+        actor_network=None,
         device: str = "cpu",
+        **kwargs,
     ):
         super().__init__(
             observation_space, action_space, index=index, learn_step=learn_step,
             device=device, hp_config=hp_config or default_hp_config(), name="NeuralUCB",
         )
+        self._accept_compat_kwargs(**kwargs)
         self.batch_size = int(batch_size)
         self.lr = float(lr)
         self.gamma = float(gamma)
@@ -61,10 +64,18 @@ class NeuralUCB(RLAlgorithm):
         self.net_config = net_config
         self.latent_dim = latent_dim
 
-        self.actor = EvolvableNetwork(
-            observation_space, num_outputs=1, encoder_config=net_config,
-            head_config=head_config, latent_dim=latent_dim, device=device,
-        )
+        if actor_network is not None:
+            # user-supplied scorer: preprocessed context -> (B,1) reward
+            # estimate (reference neural_ucb_bandit.py actor_network)
+            from ..networks.base import CustomNetworkAdapter
+
+            self.actor = CustomNetworkAdapter(actor_network, observation_space,
+                                              device=device)
+        else:
+            self.actor = EvolvableNetwork(
+                observation_space, num_outputs=1, encoder_config=net_config,
+                head_config=head_config, latent_dim=latent_dim, device=device,
+            )
         self.optimizer = OptimizerWrapper(torch.optim.Adam, [self.actor], lr=self.lr)
         self.register_network_group(NetworkGroup(eval_network="actor", policy=True))
         self.register_optimizer(OptimizerConfig(name="optimizer", networks=["actor"], lr_name="lr"))

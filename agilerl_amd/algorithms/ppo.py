@@ -61,9 +61,15 @@ class PPO(RLAlgorithm):
         action_std_init: Optional[float] = None,
         latent_dim: int = 64,
         recurrent: bool = False,
+        num_envs: int = 1,
+        max_seq_len: Optional[int] = None,
+        bptt_sequence_type: str = "chunked",
+        rollout_buffer_config: Optional[Dict[str, Any]] = None,
         share_encoders: bool = False,
         actor_network=None,
+        critic_network=None,
         device: str = "cpu",
+        **kwargs,
     ):
         super().__init__(
             observation_space,
@@ -74,6 +80,7 @@ class PPO(RLAlgorithm):
             hp_config=hp_config or default_hp_config(),
             name="PPO",
         )
+        self._accept_compat_kwargs(**kwargs)
         self.batch_size = int(batch_size)
         self.lr = float(lr)
         self.gamma = float(gamma)
@@ -95,7 +102,18 @@ class PPO(RLAlgorithm):
         self.net_config = net_config
         self.latent_dim = latent_dim
         self.recurrent = recurrent
-        self.sequence_length = 16
+        # reference ppo.py:154-161: num_envs sizes the rollout buffer;
+        # max_seq_len = truncated-BPTT window; bptt_sequence_type picks the
+        # window stride (typing.py:473); rollout_buffer_config is forwarded
+        # to RolloutBuffer by the training loop
+        self.num_envs = int(num_envs)
+        self.sequence_length = int(max_seq_len) if max_seq_len else 16
+        if bptt_sequence_type not in ("chunked", "maximum", "fifty_percent_overlap"):
+            raise ValueError(
+                "bptt_sequence_type must be chunked | maximum | fifty_percent_overlap"
+            )
+        self.bptt_sequence_type = bptt_sequence_type
+        self.rollout_buffer_config = dict(rollout_buffer_config or {})
         if recurrent and (net_config is None or net_config.get("arch") != "lstm"):
             net_config = dict(net_config or {})
             net_config["arch"] = "lstm"
@@ -120,15 +138,23 @@ class PPO(RLAlgorithm):
                 log_std_init=log_std_init,
                 device=device,
             )
-        self.critic = ValueNetwork(
-            observation_space,
-            encoder_config=net_config,
-            head_config=head_config,
-            latent_dim=latent_dim,
-            device=device,
-        )
-        if actor_network is not None and share_encoders:
-            raise ValueError("share_encoders is not supported with a custom actor_network")
+        if critic_network is not None:
+            # user-supplied value net: preprocessed obs -> (B,1) value
+            # (reference ppo.py critic_network)
+            from ..networks.base import CustomNetworkAdapter
+
+            self.critic = CustomNetworkAdapter(critic_network, observation_space,
+                                               device=device)
+        else:
+            self.critic = ValueNetwork(
+                observation_space,
+                encoder_config=net_config,
+                head_config=head_config,
+                latent_dim=latent_dim,
+                device=device,
+            )
+        if (actor_network is not None or critic_network is not None) and share_encoders:
+            raise ValueError("share_encoders is not supported with custom networks")
         if actor_network is not None and recurrent:
             raise ValueError("recurrent=True is not supported with a custom actor_network")
         self.share_encoders = bool(share_encoders)
@@ -215,7 +241,9 @@ class PPO(RLAlgorithm):
         L = self.sequence_length
         for _ in range(self.update_epochs):
             seq_bs = max(self.batch_size // L, 1)
-            for mb in rollout.get_sequence_minibatches(L, seq_bs):
+            for mb in rollout.get_sequence_minibatches(
+                L, seq_bs, sequence_type=self.bptt_sequence_type
+            ):
                 obs_seq = mb["obs"].float()  # (B, L, F)
                 B = obs_seq.shape[0]
                 hs = mb["hidden_state"]

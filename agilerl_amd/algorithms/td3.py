@@ -46,8 +46,12 @@ class TD3(DDPG):
         latent_dim: int = 64,
         share_encoders: bool = False,
         actor_network=None,
+        critic_networks=None,
         device: str = "cpu",
+        **kwargs,
     ):
+        if critic_networks is not None and len(critic_networks) != 2:
+            raise ValueError("TD3 critic_networks must be a list of two nets")
         super().__init__(
             observation_space, action_space, index=index, hp_config=hp_config,
             net_config=net_config, head_config=head_config, batch_size=batch_size,
@@ -55,17 +59,26 @@ class TD3(DDPG):
             gamma=gamma, tau=tau, policy_freq=policy_freq, O_U_noise=O_U_noise,
             expl_noise=expl_noise, mean_noise=mean_noise, theta=theta, dt=dt,
             latent_dim=latent_dim, share_encoders=share_encoders,
-            actor_network=actor_network, device=device,
+            actor_network=actor_network,
+            critic_network=None if critic_networks is None else critic_networks[0],
+            device=device,
+            **kwargs,
         )
         self.algo = "TD3"
         self.policy_noise = policy_noise
         self.noise_clip = noise_clip
 
         # second critic (twin)
-        self.critic_2 = ContinuousQNetwork(
-            observation_space, action_space, encoder_config=net_config,
-            head_config=head_config, latent_dim=latent_dim, device=device,
-        )
+        if critic_networks is not None:
+            from ..networks.base import CustomQAdapter
+
+            self.critic_2 = CustomQAdapter(critic_networks[1], observation_space,
+                                           action_space=action_space, device=device)
+        else:
+            self.critic_2 = ContinuousQNetwork(
+                observation_space, action_space, encoder_config=net_config,
+                head_config=head_config, latent_dim=latent_dim, device=device,
+            )
         self.critic_2_target = self.critic_2.clone()
         for p in self.critic_2_target.parameters():
             p.requires_grad = False

@@ -162,7 +162,8 @@ class MultiStepReplayBuffer(ReplayBuffer):
         """Number of sampleable entries (oldest-first age order)."""
         return self._size - (self.n_step - 1) * self.num_envs
 
-    def sample(self, batch_size: int, return_idx: bool = False):
+    def sample(self, batch_size: int, return_idx: bool = False,
+               include_one_step: bool = False):
         span = self._valid_span()
         if span <= 0:
             raise RuntimeError("MultiStepReplayBuffer: not enough data for n-step windows")
@@ -187,6 +188,14 @@ class MultiStepReplayBuffer(ReplayBuffer):
             "done": sd["done"][last].to(self.device, non_blocking=True),
             "n_steps": steps.to(self.device),
         }
+        if include_one_step:
+            # 1-step view of the same anchors (reference dqn_rainbow.py:124
+            # combined_reward: sum of 1-step and n-step losses)
+            batch["reward_1step"] = sd["reward"][flat].to(self.device, non_blocking=True)
+            batch["next_obs_1step"] = tree_map(
+                lambda t: t[flat].to(self.device, non_blocking=True), sd["next_obs"]
+            )
+            batch["done_1step"] = sd["done"][flat].to(self.device, non_blocking=True)
         return (batch, flat) if return_idx else batch
 
 
@@ -223,7 +232,8 @@ class PrioritizedReplayBuffer(ReplayBuffer):
         self.sum_tree.update(idx, prio)
         self.min_tree.update(idx, prio)
 
-    def sample(self, batch_size: int, beta: float = 0.4):
+    def sample(self, batch_size: int, beta: float = 0.4,
+               include_one_step: bool = False):
         from ..ops.backend import extension, use_hip
 
         device = self.sum_tree.device
@@ -250,14 +260,15 @@ class PrioritizedReplayBuffer(ReplayBuffer):
             weights = ((p * self._size).clamp(min=1e-12) ** (-beta)) / max_weight
 
         if self.n_step > 1:
-            batch = self._nstep_gather(idx.cpu())
+            batch = self._nstep_gather(idx.cpu(), include_one_step=include_one_step)
         else:
             batch = self._gather(idx.cpu())
         batch["weights"] = weights.to(self.device)
         batch["idxs"] = idx.to(self.device)
         return batch
 
-    def _nstep_gather(self, idx: torch.Tensor) -> Dict[str, Any]:
+    def _nstep_gather(self, idx: torch.Tensor,
+                      include_one_step: bool = False) -> Dict[str, Any]:
         """n-step windows anchored at ``idx`` (clamped away from the write head)."""
         N = self.num_envs
         oldest = self._ptr if self._size == self.max_size else 0
@@ -271,7 +282,7 @@ class PrioritizedReplayBuffer(ReplayBuffer):
         dones_w = sd["done"][window].float().to(self.device)
         returns, steps = ops.nstep_scan(rewards_w, dones_w, self.gamma)
         last = window.gather(1, (steps.long().clamp(min=1) - 1).cpu().unsqueeze(1)).squeeze(1)
-        return {
+        out = {
             "obs": tree_map(lambda t: t[flat].to(self.device, non_blocking=True), sd["obs"]),
             "action": sd["action"][flat].to(self.device, non_blocking=True),
             "reward": returns.to(self.device),
@@ -279,6 +290,13 @@ class PrioritizedReplayBuffer(ReplayBuffer):
             "done": sd["done"][last].to(self.device, non_blocking=True),
             "n_steps": steps.to(self.device),
         }
+        if include_one_step:
+            out["reward_1step"] = sd["reward"][flat].to(self.device, non_blocking=True)
+            out["next_obs_1step"] = tree_map(
+                lambda t: t[flat].to(self.device, non_blocking=True), sd["next_obs"]
+            )
+            out["done_1step"] = sd["done"][flat].to(self.device, non_blocking=True)
+        return out
 
     @torch.no_grad()
     def update_priorities(self, idx: torch.Tensor, priorities: torch.Tensor) -> None:

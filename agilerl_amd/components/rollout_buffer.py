@@ -150,13 +150,23 @@ class RolloutBuffer:
             yield tree_map(lambda t: t[mb_idx], flat)
 
     def get_sequence_minibatches(
-        self, seq_len: int, batch_size: int, shuffle: bool = True
+        self, seq_len: int, batch_size: int, shuffle: bool = True,
+        sequence_type: str = "chunked",
     ) -> Iterator[Dict[str, torch.Tensor]]:
-        """(B, L, ...) contiguous same-env sequences for BPTT (recurrent PPO)."""
+        """(B, L, ...) contiguous same-env sequences for BPTT (recurrent PPO).
+
+        ``sequence_type`` (reference typing.py:473 BPTTSequenceType) sets the
+        start stride: "chunked" = non-overlapping (stride L), "maximum" = all
+        overlapping windows (stride 1), "fifty_percent_overlap" = stride L/2.
+        """
+        stride = {"chunked": seq_len, "maximum": 1,
+                  "fifty_percent_overlap": max(seq_len // 2, 1)}.get(sequence_type)
+        if stride is None:
+            raise ValueError(f"unknown BPTT sequence_type '{sequence_type}'")
         T = len(self)
         starts: List[Tuple[int, int]] = []
         for env in range(self.num_envs):
-            for t0 in range(0, T - seq_len + 1, seq_len):
+            for t0 in range(0, T - seq_len + 1, stride):
                 starts.append((t0, env))
         order = np.random.permutation(len(starts)) if shuffle else np.arange(len(starts))
         flatstore = self._storage

@@ -25,7 +25,8 @@ from ..modules.simba import EvolvableSimBa
 from ..spaces import Box, DictSpace, Discrete, MultiBinary, MultiDiscrete, Space, TupleSpace, flatdim, is_image_space
 
 __all__ = ["EvolvableNetwork", "build_encoder", "get_default_encoder_config",
-           "preprocess_observation", "CustomNetworkAdapter", "CustomStochasticAdapter"]
+           "preprocess_observation", "CustomNetworkAdapter",
+           "CustomStochasticAdapter", "CustomQAdapter"]
 
 
 def get_default_encoder_config(observation_space: Space, simba: bool = False) -> Dict[str, Any]:
@@ -343,6 +344,33 @@ class CustomNetworkAdapter(EvolvableModule):
             result = super().apply_mutation(name, **choices)
         self._last_mutation = (name, result if isinstance(result, dict) else {})
         return result
+
+
+class CustomQAdapter(CustomNetworkAdapter):
+    """Custom net as a (state, action) -> Q critic (reference DDPG/TD3
+    ``critic_network=``, ddpg.py:136).  The user net either takes
+    ``forward(obs, action)`` directly, or takes a single concatenated
+    ``[obs_flat, action_flat]`` tensor (MakeEvolvable-style critics)."""
+
+    def __init__(self, net, observation_space: Space, action_space: Space,
+                 device: str = "cpu"):
+        super().__init__(net, observation_space, action_space=action_space, device=device)
+        import inspect as _inspect
+
+        try:
+            n_params = len(_inspect.signature(net.forward).parameters)
+        except (TypeError, ValueError):
+            n_params = 1
+        self._two_arg = n_params >= 2
+
+    def forward(self, obs, action: torch.Tensor) -> torch.Tensor:
+        action = action.to(self.device).float()
+        if action.dim() == 1:
+            action = action.unsqueeze(0)
+        if self._two_arg:
+            return self.net(obs, action)
+        flat_obs = obs.reshape(obs.shape[0], -1) if isinstance(obs, torch.Tensor) else obs
+        return self.net(torch.cat([flat_obs, action.reshape(action.shape[0], -1)], dim=-1))
 
 
 class CustomStochasticAdapter(CustomNetworkAdapter):

@@ -113,9 +113,14 @@ def train_off_policy(
                     len(memory) >= max(agent.batch_size, learning_delay)
                     and it % agent.learn_step == 0
                 ):
+                    combined = bool(getattr(agent, "combined_reward", False)) and \
+                        getattr(memory, "n_step", 1) > 1
                     if per:
                         beta = per_beta_start + (1.0 - per_beta_start) * frac
-                        batch = memory.sample(agent.batch_size, beta=beta)
+                        batch = memory.sample(agent.batch_size, beta=beta,
+                                              include_one_step=combined)
+                    elif combined:
+                        batch = memory.sample(agent.batch_size, include_one_step=True)
                     else:
                         batch = memory.sample(agent.batch_size)
                     loss = agent.learn(batch)

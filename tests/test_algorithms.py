@@ -3,6 +3,7 @@
 import os
 
 import numpy as np
+import pytest
 import torch
 
 from agilerl_amd.algorithms.dqn import DQN
@@ -233,3 +234,95 @@ class TestToDevice:
         assert agent.actor.encoder.device == "cpu"
         agent.apply_architecture_mutation("encoder.add_node", numb_new_nodes=16)
         assert next(agent.actor.parameters()).device.type == "cpu"
+
+
+class TestReferenceConstructorSurface:
+    """Reference-API constructor kwargs: compat layer + custom critics
+    (reference ddpg.py:136 critic_network, dqn_rainbow.py:124
+    combined_reward, ppo.py:154-161 recurrent buffer kwargs)."""
+
+    def test_compat_kwargs_accept_and_warn(self):
+        import warnings
+
+        with warnings.catch_warnings(record=True) as w:
+            warnings.simplefilter("always")
+            a = DQN(Box(-1, 1, (4,)), Discrete(2),
+                    accelerator=object(), wrap=True, mut="param",
+                    normalize_images=False, torch_compiler="default")
+        assert a.mut == "param" and a.accelerator is None
+        msgs = " ".join(str(x.message) for x in w)
+        assert "accelerator" in msgs and "torch_compiler" in msgs
+        with pytest.raises(TypeError, match="bogus"):
+            DQN(Box(-1, 1, (4,)), Discrete(2), bogus=1)
+
+    def test_ddpg_td3_custom_critics(self):
+        from agilerl_amd.algorithms.td3 import TD3
+        from agilerl_amd.modules.mlp import EvolvableMLP
+
+        obs_s, act_s = Box(-1, 1, (4,)), Box(-1, 1, (2,))
+        mk = lambda: EvolvableMLP(num_inputs=6, num_outputs=1, hidden_size=[16])
+        t = TD3(obs_s, act_s, critic_networks=[mk(), mk()])
+        x = torch.randn(3, 4)
+        a = torch.randn(3, 2)
+        assert t.critic(t.critic.preprocess(x), a).shape == (3, 1)
+        assert t.critic_2(t.critic_2.preprocess(x), a).shape == (3, 1)
+        batch = {
+            "obs": torch.randn(8, 4), "action": torch.randn(8, 2),
+            "reward": torch.randn(8), "next_obs": torch.randn(8, 4),
+            "done": torch.zeros(8),
+        }
+        assert np.isfinite(t.learn(batch))
+        with pytest.raises(ValueError, match="two"):
+            TD3(obs_s, act_s, critic_networks=[mk()])
+
+    def test_rainbow_combined_reward_learns(self):
+        from agilerl_amd.algorithms.dqn_rainbow import RainbowDQN
+        from agilerl_amd.components import MultiStepReplayBuffer
+
+        agent = RainbowDQN(Box(-1, 1, (4,)), Discrete(2), combined_reward=True,
+                           batch_size=16)
+        buf = MultiStepReplayBuffer(500, n_step=3, gamma=0.99)
+        obs = np.random.randn(4, 4).astype(np.float32)
+        for _ in range(40):
+            nxt = np.random.randn(4, 4).astype(np.float32)
+            buf.add(obs=obs, action=np.random.randint(0, 2, (4,)),
+                    reward=np.random.randn(4).astype(np.float32),
+                    next_obs=nxt, done=np.zeros(4, np.float32))
+            obs = nxt
+        batch = buf.sample(16, include_one_step=True)
+        assert {"reward_1step", "next_obs_1step", "done_1step"} <= set(batch)
+        assert np.isfinite(agent.learn(batch))
+
+    def test_ppo_bptt_sequence_types(self):
+        from agilerl_amd.components.rollout_buffer import RolloutBuffer
+
+        buf = RolloutBuffer(32, num_envs=2)
+        for _ in range(32):
+            buf.add(obs=np.random.randn(2, 3).astype(np.float32),
+                    action=np.random.randn(2, 1).astype(np.float32),
+                    reward=np.zeros(2, np.float32), done=np.zeros(2, np.float32),
+                    value=np.zeros(2, np.float32), log_prob=np.zeros(2, np.float32))
+        buf.compute_returns_and_advantages(np.zeros(2, np.float32))
+        count = lambda st: sum(mb["obs"].shape[0] for mb in
+                               buf.get_sequence_minibatches(8, 64, sequence_type=st))
+        n_chunk, n_half, n_max = count("chunked"), count("fifty_percent_overlap"), count("maximum")
+        assert n_chunk == (32 // 8) * 2
+        assert n_max == (32 - 8 + 1) * 2
+        assert n_chunk < n_half < n_max
+        with pytest.raises(ValueError, match="sequence_type"):
+            next(buf.get_sequence_minibatches(8, 4, sequence_type="bogus"))
+        # PPO validates the kwarg too and honours max_seq_len
+        a = PPO(Box(-1, 1, (4,)), Discrete(2), recurrent=True, max_seq_len=8,
+                bptt_sequence_type="fifty_percent_overlap")
+        assert a.sequence_length == 8
+        with pytest.raises(ValueError, match="bptt_sequence_type"):
+            PPO(Box(-1, 1, (4,)), Discrete(2), bptt_sequence_type="bogus")
+
+    def test_bandit_custom_scorer(self):
+        from agilerl_amd.algorithms.neural_ts import NeuralTS
+        from agilerl_amd.modules.mlp import EvolvableMLP
+
+        net = EvolvableMLP(num_inputs=5, num_outputs=1, hidden_size=[8])
+        b = NeuralTS(Box(-1, 1, (5,)), Discrete(3), actor_network=net)
+        ctx = np.random.randn(3, 5).astype(np.float32)
+        assert 0 <= int(b.get_action(ctx)) < 3
