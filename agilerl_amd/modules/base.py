@@ -116,9 +116,16 @@ class ModuleMeta(type):
 class EvolvableModule(nn.Module, metaclass=ModuleMeta):
     """Base class for all evolvable building blocks."""
 
-    def __init__(self, device: str = "cpu"):
+    def __init__(self, device: str = "cpu", name: Optional[str] = None,
+                 random_seed: Optional[int] = None):
         super().__init__()
         self.device = device
+        # reference modules/base.py: `name` labels the module in mutation
+        # logs; `random_seed` makes weight init reproducible
+        self.name = name or type(self).__name__.lower()
+        self.random_seed = random_seed
+        if random_seed is not None:
+            torch.manual_seed(int(random_seed))
         self._last_mutation: Optional[Tuple[str, dict]] = None
 
     # ------------------------------------------------------------------

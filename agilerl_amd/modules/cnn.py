@@ -35,9 +35,23 @@ class EvolvableCNN(EvolvableModule):
         min_channel_size: int = 16,
         max_channel_size: int = 256,
         layer_norm: bool = False,
+        block_type: str = "Conv2d",
+        sample_input: Optional[torch.Tensor] = None,
+        init_layers: bool = True,
         device: str = "cpu",
+        name: Optional[str] = None,
+        random_seed: Optional[int] = None,
     ):
-        super().__init__(device)
+        super().__init__(device, name=name, random_seed=random_seed)
+        # reference cnn.py:15 BlockType: Conv1d/2d/3d selected by the obs
+        # rank; sample_input overrides the flat-dim inference for exotic
+        # shapes; init_layers toggles orthogonal-style re-init (our layers
+        # use PyTorch defaults already, kept for API compat)
+        if block_type not in ("Conv1d", "Conv2d", "Conv3d"):
+            raise ValueError("block_type must be Conv1d | Conv2d | Conv3d")
+        self.block_type = block_type
+        self.sample_input = sample_input
+        self.init_layers = bool(init_layers)
         self.input_shape = tuple(input_shape)
         self.num_outputs = int(num_outputs)
         self.channel_size = list(channel_size) if channel_size is not None else [32, 32]
@@ -53,16 +67,25 @@ class EvolvableCNN(EvolvableModule):
 
         self.model = self._build().to(device)
 
+    @property
+    def _conv_cls(self):
+        return {"Conv1d": nn.Conv1d, "Conv2d": nn.Conv2d, "Conv3d": nn.Conv3d}[self.block_type]
+
     def _conv_output_dim(self, conv: nn.Sequential) -> int:
         with torch.no_grad():
-            dummy = torch.zeros(1, *self.input_shape)
+            if self.sample_input is not None:
+                dummy = self.sample_input.to("cpu").float()
+                if dummy.dim() == len(self.input_shape):
+                    dummy = dummy.unsqueeze(0)
+            else:
+                dummy = torch.zeros(1, *self.input_shape)
             return int(np.prod(conv(dummy).shape[1:]))
 
     def _build(self) -> nn.Sequential:
         layers: List[nn.Module] = []
         in_ch = self.input_shape[0]
         for out_ch, k, s in zip(self.channel_size, self.kernel_size, self.stride_size):
-            layers.append(nn.Conv2d(in_ch, out_ch, k, s, padding=k // 2))
+            layers.append(self._conv_cls(in_ch, out_ch, k, s, padding=k // 2))
             if self.layer_norm:
                 layers.append(nn.GroupNorm(1, out_ch))
             layers.append(get_activation(self.activation))
@@ -75,7 +98,7 @@ class EvolvableCNN(EvolvableModule):
         return nn.Sequential(*head)
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
-        if x.dim() == 3:
+        if x.dim() == len(self.input_shape):
             x = x.unsqueeze(0)
         if not x.is_floating_point():
             x = x.float() / 255.0

@@ -27,9 +27,15 @@ class EvolvableLSTM(EvolvableModule):
         max_hidden_state_size: int = 500,
         min_layers: int = 1,
         max_layers: int = 3,
+        dropout: float = 0.0,
+        output_activation: Optional[str] = None,
         device: str = "cpu",
+        name: Optional[str] = None,
+        random_seed: Optional[int] = None,
     ):
-        super().__init__(device)
+        super().__init__(device, name=name, random_seed=random_seed)
+        self.dropout = float(dropout)
+        self.output_activation = output_activation
         self.input_size = int(input_size)
         self.num_outputs = int(num_outputs)
         self.hidden_state_size = int(hidden_state_size)
@@ -41,10 +47,15 @@ class EvolvableLSTM(EvolvableModule):
 
         self.lstm = self._build_lstm().to(device)
         self.proj = nn.Linear(self.hidden_state_size, self.num_outputs).to(device)
+        from .components import get_activation
+
+        self.out_act = get_activation(output_activation).to(device)
 
     def _build_lstm(self) -> nn.LSTM:
         return nn.LSTM(
-            self.input_size, self.hidden_state_size, self.num_layers, batch_first=True
+            self.input_size, self.hidden_state_size, self.num_layers,
+            batch_first=True,
+            dropout=self.dropout if self.num_layers > 1 else 0.0,
         )
 
     def forward(
@@ -64,7 +75,7 @@ class EvolvableLSTM(EvolvableModule):
         self, x: torch.Tensor, hidden: Optional[Tuple[torch.Tensor, torch.Tensor]] = None
     ) -> Tuple[torch.Tensor, Tuple[torch.Tensor, torch.Tensor]]:
         out, hidden = self.lstm(x.float(), hidden)
-        return self.proj(out), hidden
+        return self.out_act(self.proj(out)), hidden
 
     def initial_hidden(self, batch_size: int) -> Tuple[torch.Tensor, torch.Tensor]:
         h = torch.zeros(self.num_layers, batch_size, self.hidden_state_size, device=self.device)
@@ -77,7 +88,7 @@ class EvolvableLSTM(EvolvableModule):
         if hidden is None:
             hidden = self.initial_hidden(x.shape[0])
         out, new_hidden = self.lstm(x.float().unsqueeze(1), hidden)
-        return self.proj(out[:, -1]), new_hidden
+        return self.out_act(self.proj(out[:, -1])), new_hidden
 
     @property
     def output_size(self) -> int:

@@ -87,9 +87,16 @@ class EvolvableMLP(EvolvableModule):
         noise_std: float = 0.5,
         init_layers: bool = True,
         output_vanish: bool = False,
+        new_gelu: bool = False,
         device: str = "cpu",
+        name: Optional[str] = None,
+        random_seed: Optional[int] = None,
     ):
-        super().__init__(device)
+        super().__init__(device, name=name, random_seed=random_seed)
+        if new_gelu:
+            # reference mlp.py new_gelu: tanh-approx GELU activations
+            activation = "NewGELU"
+        self.new_gelu = bool(new_gelu)
         self.num_inputs = int(num_inputs)
         self.num_outputs = int(num_outputs)
         self.hidden_size = list(hidden_size) if hidden_size is not None else [64, 64]

@@ -30,9 +30,32 @@ class EvolvableMultiInput(EvolvableModule):
         mlp_config: Optional[dict] = None,
         cnn_config: Optional[dict] = None,
         sub_configs: Optional[Dict[str, dict]] = None,
+        init_dicts: Optional[Dict[str, dict]] = None,
+        vector_space_mlp: bool = True,
+        output_activation: Optional[str] = None,
+        output_layernorm: bool = False,
+        min_latent_dim: int = 8,
+        max_latent_dim: int = 128,
         device: str = "cpu",
+        name: Optional[str] = None,
+        random_seed: Optional[int] = None,
     ):
-        super().__init__(device)
+        super().__init__(device, name=name, random_seed=random_seed)
+        # reference multi_input.py:122: vector_space_mlp=False passes vector
+        # sub-spaces through a raw flatten (no learned encoder); our default
+        # keeps the MLP (documented divergence - richer and mutable).
+        # init_dicts are the reference's per-key construction overrides and
+        # merge into sub_configs; min/max_latent_dim bound latent mutations.
+        self.vector_space_mlp = bool(vector_space_mlp)
+        self.output_activation = output_activation
+        self.output_layernorm = bool(output_layernorm)
+        self.min_latent_dim = int(min_latent_dim)
+        self.max_latent_dim = int(max_latent_dim)
+        if init_dicts:
+            merged = {k: dict(v) for k, v in (sub_configs or {}).items()}
+            for k, v in init_dicts.items():
+                merged.setdefault(k, {}).update(v)
+            sub_configs = merged
         self.observation_space = observation_space
         self.num_outputs = int(num_outputs)
         self.latent_dim = int(latent_dim)
@@ -55,7 +78,17 @@ class EvolvableMultiInput(EvolvableModule):
         for key, space in items:
             encoders[key] = self._make_encoder(space, ctor_sub.get(key))
         self.encoders = nn.ModuleDict(encoders)
-        self.head = nn.Linear(self.latent_dim * len(items), self.num_outputs).to(device)
+        self.head = self._make_head(len(items))
+
+    def _make_head(self, n_items: int) -> nn.Module:
+        from .components import get_activation
+
+        layers = [nn.Linear(self.latent_dim * n_items, self.num_outputs)]
+        if self.output_layernorm:
+            layers.append(nn.LayerNorm(self.num_outputs))
+        if self.output_activation:
+            layers.append(get_activation(self.output_activation))
+        return (layers[0] if len(layers) == 1 else nn.Sequential(*layers)).to(self.device)
 
     @property
     def sub_configs(self) -> Dict[str, dict]:
@@ -68,8 +101,10 @@ class EvolvableMultiInput(EvolvableModule):
                     "kernel_size": list(enc.kernel_size),
                     "stride_size": list(enc.stride_size),
                 }
-            else:
+            elif isinstance(enc, EvolvableMLP):
                 out[key] = {"hidden_size": list(enc.hidden_size)}
+            else:  # _FlattenEncoder has no mutable shape
+                out[key] = {}
         return out
 
     def _make_encoder(self, space: Space, override: Optional[dict] = None) -> EvolvableModule:
@@ -80,6 +115,8 @@ class EvolvableMultiInput(EvolvableModule):
             return EvolvableCNN(
                 input_shape=space.shape, num_outputs=self.latent_dim, device=self.device, **cfg
             )
+        if not self.vector_space_mlp:
+            return _FlattenEncoder(flatdim(space), self.latent_dim, self.device)
         cfg = {"hidden_size": [64]}
         cfg.update(self.mlp_config)
         cfg.update(override or {})
@@ -106,7 +143,7 @@ class EvolvableMultiInput(EvolvableModule):
     def recreate_network(self) -> None:
         for enc in self.encoders.values():
             enc.recreate_network()
-        new_head = nn.Linear(self.latent_dim * len(self._keys), self.num_outputs).to(self.device)
+        new_head = self._make_head(len(self._keys))
         preserve_parameters(self.head, new_head)
         self.head = new_head
 
@@ -153,3 +190,32 @@ class EvolvableMultiInput(EvolvableModule):
                 ) or r
                 activation = r.get("activation", activation)
         return {"activation": activation, "output": output}
+
+
+class _FlattenEncoder(EvolvableModule):
+    """vector_space_mlp=False sub-encoder: flatten + single linear projection
+    to the shared latent width (no hidden layers, no mutations) — the
+    MI355X analog of the reference's raw vector concat
+    (multi_input.py:91-94)."""
+
+    def __init__(self, num_inputs: int, num_outputs: int, device: str = "cpu"):
+        super().__init__(device)
+        self.num_inputs = int(num_inputs)
+        self.num_outputs = int(num_outputs)
+        self.proj = nn.Linear(self.num_inputs, self.num_outputs).to(device)
+
+    def forward(self, x: torch.Tensor) -> torch.Tensor:
+        if not isinstance(x, torch.Tensor):
+            x = torch.as_tensor(x, device=self.device)
+        x = x.float().reshape(x.shape[0], -1) if x.dim() > 1 else x.float().unsqueeze(0)
+        return self.proj(x)
+
+    @property
+    def output_size(self) -> int:
+        return self.num_outputs
+
+    def reset_noise(self) -> None:
+        pass
+
+    def recreate_network(self) -> None:
+        pass

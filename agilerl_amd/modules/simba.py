@@ -29,9 +29,13 @@ class EvolvableSimBa(EvolvableModule):
         min_mlp_nodes: int = 16,
         max_mlp_nodes: int = 500,
         scale_factor: int = 4,
+        output_activation: Optional[str] = None,
         device: str = "cpu",
+        name: Optional[str] = None,
+        random_seed: Optional[int] = None,
     ):
-        super().__init__(device)
+        super().__init__(device, name=name, random_seed=random_seed)
+        self.output_activation = output_activation
         self.num_inputs = int(num_inputs)
         self.num_outputs = int(num_outputs)
         self.hidden_size = int(hidden_size)
@@ -50,6 +54,10 @@ class EvolvableSimBa(EvolvableModule):
             blocks.append(SimbaResidualBlock(self.hidden_size, self.scale_factor))
         blocks.append(nn.LayerNorm(self.hidden_size))
         blocks.append(nn.Linear(self.hidden_size, self.num_outputs))
+        if self.output_activation:
+            from .components import get_activation
+
+            blocks.append(get_activation(self.output_activation))
         return nn.Sequential(*blocks)
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:

@@ -61,6 +61,22 @@ def build_encoder(
     if arch is None:
         arch = get_default_encoder_config(observation_space)["arch"]
 
+    def _fit(module_cls, config):
+        """Keep only kwargs the module accepts; warn about the rest so
+        reference net_configs with extra fields load instead of crashing."""
+        import inspect
+        import warnings
+
+        params = inspect.signature(module_cls.__init__).parameters
+        dropped = sorted(k for k in config if k not in params)
+        if dropped:
+            warnings.warn(
+                f"{module_cls.__name__} ignores unsupported net_config "
+                f"fields {dropped}",
+                RuntimeWarning,
+            )
+        return {k: v for k, v in config.items() if k in params}
+
     if isinstance(observation_space, (DictSpace, TupleSpace)) or arch == "multi_input":
         return EvolvableMultiInput(
             observation_space=observation_space,
@@ -68,25 +84,29 @@ def build_encoder(
             device=device,
             # sub_configs carries the live (possibly mutated) per-encoder
             # shapes so clones/checkpoints rebuild exactly
-            **{k: v for k, v in cfg.items()
-               if k in ("mlp_config", "cnn_config", "sub_configs")},
+            **_fit(EvolvableMultiInput, cfg),
         )
     if arch == "resnet":
         from ..modules.resnet import EvolvableResNet
 
         return EvolvableResNet(
-            input_shape=observation_space.shape, num_outputs=latent_dim, device=device, **cfg
+            input_shape=observation_space.shape, num_outputs=latent_dim,
+            device=device, **_fit(EvolvableResNet, cfg)
         )
     if arch == "cnn" or is_image_space(observation_space):
         return EvolvableCNN(
-            input_shape=observation_space.shape, num_outputs=latent_dim, device=device, **cfg
+            input_shape=observation_space.shape, num_outputs=latent_dim,
+            device=device, **_fit(EvolvableCNN, cfg)
         )
     num_inputs = flatdim(observation_space)
     if arch == "simba":
-        return EvolvableSimBa(num_inputs=num_inputs, num_outputs=latent_dim, device=device, **cfg)
+        return EvolvableSimBa(num_inputs=num_inputs, num_outputs=latent_dim,
+                              device=device, **_fit(EvolvableSimBa, cfg))
     if arch == "lstm":
-        return EvolvableLSTM(input_size=num_inputs, num_outputs=latent_dim, device=device, **cfg)
-    return EvolvableMLP(num_inputs=num_inputs, num_outputs=latent_dim, device=device, **cfg)
+        return EvolvableLSTM(input_size=num_inputs, num_outputs=latent_dim,
+                             device=device, **_fit(EvolvableLSTM, cfg))
+    return EvolvableMLP(num_inputs=num_inputs, num_outputs=latent_dim,
+                        device=device, **_fit(EvolvableMLP, cfg))
 
 
 def preprocess_observation(obs, space: Space, device) -> Any:

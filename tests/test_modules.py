@@ -140,3 +140,83 @@ def test_preserve_parameters_slices():
     preserve_parameters(old, new)
     assert torch.equal(new.weight[:16], old.weight)
     assert torch.equal(new.bias[:16], old.bias)
+
+
+class TestReferenceModuleSurface:
+    """Reference module constructor kwargs added for parity (reference
+    modules/mlp.py new_gelu, cnn.py:15 BlockType, lstm dropout,
+    multi_input.py:122 vector_space_mlp/init_dicts)."""
+
+    def test_random_seed_reproducible_init(self):
+        import torch
+
+        from agilerl_amd.modules.mlp import EvolvableMLP
+
+        a = EvolvableMLP(num_inputs=4, num_outputs=2, hidden_size=[8], random_seed=11)
+        b = EvolvableMLP(num_inputs=4, num_outputs=2, hidden_size=[8], random_seed=11)
+        x = torch.randn(3, 4)
+        assert torch.allclose(a(x), b(x))
+        assert a.name == "evolvablemlp"
+        assert EvolvableMLP(num_inputs=4, num_outputs=2, name="pi").name == "pi"
+
+    def test_cnn_conv1d_conv3d(self):
+        import torch
+
+        from agilerl_amd.modules.cnn import EvolvableCNN
+
+        c3 = EvolvableCNN(input_shape=(1, 4, 10, 10), num_outputs=6, block_type="Conv3d",
+                          channel_size=[8], kernel_size=[3], stride_size=[1])
+        assert c3(torch.randn(2, 1, 4, 10, 10)).shape == (2, 6)
+        c3.apply_mutation("add_channel")
+        assert c3(torch.randn(2, 1, 4, 10, 10)).shape == (2, 6)
+        assert c3.clone()(torch.randn(2, 1, 4, 10, 10)).shape == (2, 6)
+        c1 = EvolvableCNN(input_shape=(2, 24), num_outputs=4, block_type="Conv1d",
+                          channel_size=[8], kernel_size=[3], stride_size=[1])
+        assert c1(torch.randn(3, 2, 24)).shape == (3, 4)
+        import pytest as _pytest
+        with _pytest.raises(ValueError, match="block_type"):
+            EvolvableCNN(input_shape=(3, 8, 8), num_outputs=2, block_type="Conv4d")
+
+    def test_lstm_dropout_and_output_activation(self):
+        import torch
+
+        from agilerl_amd.modules.lstm import EvolvableLSTM
+
+        l = EvolvableLSTM(input_size=4, num_outputs=3, num_layers=2, dropout=0.25,
+                          output_activation="Tanh")
+        assert l.lstm.dropout == 0.25
+        out, _ = l.step(torch.randn(2, 4))
+        assert out.abs().max() <= 1.0
+        c = l.clone()
+        assert c.lstm.dropout == 0.25
+
+    def test_multi_input_vector_space_and_init_dicts(self):
+        import torch
+
+        from agilerl_amd.modules.multi_input import EvolvableMultiInput, _FlattenEncoder
+        from agilerl_amd.spaces import Box, DictSpace
+
+        sp = DictSpace({"v": Box(-1, 1, (5,)), "img": Box(0, 255, (3, 8, 8))})
+        m = EvolvableMultiInput(sp, num_outputs=10, vector_space_mlp=False,
+                                output_activation="Tanh", output_layernorm=True,
+                                init_dicts={"img": {"channel_size": [16]}})
+        assert isinstance(m.encoders["v"], _FlattenEncoder)
+        assert m.encoders["img"].channel_size == [16]
+        obs = {"v": torch.randn(4, 5), "img": torch.randn(4, 3, 8, 8)}
+        out = m(obs)
+        assert out.shape == (4, 10) and out.abs().max() <= 1.0
+        c = m.clone()
+        assert torch.allclose(c(obs), out)
+
+    def test_build_encoder_filters_unknown_fields(self):
+        import warnings
+
+        from agilerl_amd.networks.base import build_encoder
+        from agilerl_amd.spaces import Box
+
+        with warnings.catch_warnings(record=True) as w:
+            warnings.simplefilter("always")
+            enc = build_encoder(Box(-1, 1, (4,)), 16,
+                                {"arch": "mlp", "hidden_size": [8], "exotic": 1})
+        assert type(enc).__name__ == "EvolvableMLP"
+        assert any("exotic" in str(x.message) for x in w)
