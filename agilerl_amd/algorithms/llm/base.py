@@ -437,16 +437,18 @@ class LLMAlgorithm(EvolvableAlgorithm):
             pad_id = getattr(self.model.config, "eos_token_id", 0) or 0
             if isinstance(pad_id, (list, tuple)):
                 pad_id = pad_id[0]
+        gen_kwargs = dict(getattr(self, "sampling_kwargs", {}) or {})
+        gen_kwargs.setdefault("top_p", 1.0)
+        gen_kwargs.setdefault("top_k", 0)
         out = self.model.generate(
             input_ids=input_ids,
             attention_mask=attention_mask,
             max_new_tokens=max_new_tokens or self.max_completion_tokens,
             do_sample=do_sample,
             temperature=temperature or self.temperature,
-            top_p=1.0,
-            top_k=0,
             pad_token_id=pad_id,
             use_cache=True,
+            **gen_kwargs,
         )
         self.model.train()
         return out

@@ -63,8 +63,82 @@ class GRPO(LLMAlgorithm):
         gradient_checkpointing: bool = False,
         use_packing="auto",
         generation: str = "hf",
+        loss_type: str = "grpo",
+        adv_norm: Optional[str] = None,
+        whiten_advantages: bool = False,
+        adv_clip_range: Optional[float] = None,
+        filter_zero_adv: bool = False,
+        adv_filter_eps: float = 0.0,
+        use_kl_advantage_shaping: bool = False,
+        top_k: Optional[int] = None,
+        top_p: Optional[float] = None,
+        min_p: Optional[float] = None,
+        repetition_penalty: Optional[float] = None,
+        min_output_tokens: Optional[int] = None,
+        seed: Optional[int] = None,
+        chunk_rows: Optional[int] = None,
+        cosine_lr_schedule_config=None,
+        activation_offload: bool = False,
+        pad_token_id: Optional[int] = None,
+        use_liger_loss: bool = True,
         device: str = "cpu",
+        **kwargs,
     ):
+        # ------------------------------------------------------------------
+        # Reference kwarg aliases (reference grpo.py:441-539) — accepted so
+        # code written against the reference constructs GRPO unchanged
+        # ------------------------------------------------------------------
+        if "model_name" in kwargs:
+            model_name_or_path = model_name_or_path or kwargs.pop("model_name")
+        if "use_sequence_packing" in kwargs:
+            use_packing = kwargs.pop("use_sequence_packing")
+        if "vllm_importance_sampling_correction" in kwargs:
+            sampling_is_correction = kwargs.pop("vllm_importance_sampling_correction")
+        if "vllm_importance_sampling_cap" in kwargs:
+            sampling_is_cap = kwargs.pop("vllm_importance_sampling_cap")
+        if "micro_batch_size_per_gpu" in kwargs:
+            micro_batch_size = kwargs.pop("micro_batch_size_per_gpu")
+        if "action_granularity" in kwargs:
+            importance_sampling_level = kwargs.pop("action_granularity")
+        if "advantage_granularity" in kwargs:
+            advantage_level = kwargs.pop("advantage_granularity")
+        if "max_output_tokens" in kwargs:
+            max_completion_tokens = kwargs.pop("max_output_tokens")
+        if kwargs.pop("use_vllm", False) or kwargs.pop("vllm_config", None) is not None:
+            import warnings
+
+            warnings.warn(
+                "use_vllm/vllm_config map to the native paged-KV decode "
+                "engine on MI355X (generation='paged'); vLLM itself is not "
+                "used.",
+                RuntimeWarning,
+            )
+            generation = "paged"
+        # infra kwargs with no MI355X role: warn-and-ignore (288 GB HBM
+        # removes the memory gymnastics; no bitsandbytes/DeepSpeed/vLLM)
+        _ignored = [k for k in (
+            "quantization_config", "use_memory_efficient_params",
+            "reduce_memory_peak", "calc_position_embeddings",
+            "cast_logprobs_to_fp32", "hf_generate_chunk_size",
+            "lora_target_scope", "use_separate_reference_adapter",
+            "turn_advantage_trajectory_fallback", "clone", "actor_network",
+            "max_model_len", "batch_size", "mini_batch_size", "pad_token",
+        ) if kwargs.pop(k, None) is not None]
+        if _ignored:
+            import warnings
+
+            warnings.warn(
+                f"GRPO ignores reference-only kwargs {_ignored} (see "
+                "docs/llm_finetuning.md for the MI355X equivalents)",
+                RuntimeWarning,
+            )
+        if adv_norm is not None:
+            # reference grpo.py:453 adv_norm: "mean_std" | "mean_only"
+            if adv_norm not in ("mean_std", "mean_only"):
+                raise ValueError("adv_norm must be 'mean_std' or 'mean_only'")
+            scale_rewards = adv_norm == "mean_std"
+        if seed is not None:
+            torch.manual_seed(int(seed))
         super().__init__(
             model=model, model_config=model_config, model_name_or_path=model_name_or_path,
             tokenizer=tokenizer, index=index, hp_config=hp_config, lora_config=lora_config,
@@ -73,6 +147,47 @@ class GRPO(LLMAlgorithm):
             dtype=dtype, gradient_checkpointing=gradient_checkpointing, device=device,
             name=type(self).__name__,
         )
+        self._accept_compat_kwargs(**kwargs)
+        if loss_type not in ("grpo", "gspo", "cispo"):
+            raise ValueError("loss_type must be grpo | gspo | cispo")
+        self.loss_type = loss_type
+        if loss_type == "cispo":
+            self.CISPO = True
+        elif loss_type == "gspo":
+            importance_sampling_level = "trajectory"
+        self.adv_norm = adv_norm or ("mean_std" if scale_rewards else "mean_only")
+        self.whiten_advantages = bool(whiten_advantages)
+        self.adv_clip_range = adv_clip_range
+        self.filter_zero_adv = bool(filter_zero_adv)
+        self.adv_filter_eps = float(adv_filter_eps)
+        self.use_kl_advantage_shaping = bool(use_kl_advantage_shaping)
+        self.seed = seed
+        self.chunk_rows = chunk_rows
+        self.activation_offload = bool(activation_offload)
+        self.use_liger_loss = bool(use_liger_loss)  # False forces the eager loss path
+        # sampling controls forwarded to generation (reference top_k/top_p/
+        # min_p/repetition_penalty/min_output_tokens)
+        self.sampling_kwargs = {
+            k: v for k, v in {
+                "top_k": top_k, "top_p": top_p, "min_p": min_p,
+                "repetition_penalty": repetition_penalty,
+                "min_new_tokens": min_output_tokens,
+            }.items() if v is not None
+        }
+        if pad_token_id is not None and self.tokenizer is not None:
+            self.tokenizer.pad_token_id = pad_token_id
+        self.lr_scheduler = None
+        if cosine_lr_schedule_config is not None:
+            from ...llm.scheduler import create_warmup_cosine_scheduler
+
+            cfg = cosine_lr_schedule_config
+            total = int(cfg.get("num_epochs", 1) if isinstance(cfg, dict)
+                        else getattr(cfg, "num_epochs", 1))
+            warm = float(cfg.get("warmup_proportion", 0.03) if isinstance(cfg, dict)
+                         else getattr(cfg, "warmup_proportion", 0.03))
+            self.lr_scheduler = create_warmup_cosine_scheduler(
+                self.optimizer, total_steps=max(total, 1), warmup_ratio=warm
+            )
         self.group_size = int(group_size)
         self.update_epochs = int(update_epochs)
         # padding-free grad/old-policy passes (compute_logprobs_packed):
@@ -166,6 +281,20 @@ class GRPO(LLMAlgorithm):
         else:
             advantages = self._calculate_advantages(rewards)  # (B,)
             adv_tok = advantages.unsqueeze(1).expand_as(action_mask)
+        if self.filter_zero_adv:
+            # reference grpo.py:461: drop samples with ~zero advantage
+            samp_adv = (adv_tok * action_mask).sum(1) / action_mask.sum(1).clamp(min=1)
+            action_mask = action_mask * (samp_adv.abs() > self.adv_filter_eps).float().unsqueeze(1)
+        if self.whiten_advantages:
+            m = action_mask.bool()
+            vals = adv_tok[m]
+            if vals.numel() > 1:
+                adv_tok = torch.where(
+                    m, (adv_tok - vals.mean()) / (vals.std() + 1e-8),
+                    torch.zeros_like(adv_tok),
+                )
+        if self.adv_clip_range is not None:
+            adv_tok = adv_tok.clamp(-self.adv_clip_range, self.adv_clip_range)
 
         B = ids.shape[0]
         mb = max(self.micro_batch_size, 1)
@@ -173,7 +302,11 @@ class GRPO(LLMAlgorithm):
         if packing == "auto":
             pad_frac = 1.0 - attention_mask.float().mean().item()
             packing = pad_frac > 0.10
-        logprob_fn = self.compute_logprobs_packed if packing else self.compute_logprobs
+        _logprob_fn = self.compute_logprobs_packed if packing else self.compute_logprobs
+
+        def logprob_fn(*a, **kw):
+            kw.setdefault("chunk_rows", self.chunk_rows)
+            return _logprob_fn(*a, **kw)
 
         # old-policy + reference logprobs (no grad, micro-batched)
         old_logp = torch.empty(action_mask.shape, device=self.device)
@@ -183,6 +316,13 @@ class GRPO(LLMAlgorithm):
             old_logp[s:e] = logprob_fn(ids[s:e], attention_mask[s:e], adapter="self")
             if ref_logp is not None:
                 ref_logp[s:e] = logprob_fn(ids[s:e], attention_mask[s:e], adapter=None)
+        if self.use_kl_advantage_shaping and ref_logp is not None:
+            # ART-style zero-mean KL shaping (reference grpo.py:1488-1503):
+            # k3 KL of the pre-update policy vs the reference adapter
+            d = ref_logp - old_logp
+            kl = (d.exp() - d - 1) * action_mask
+            avg = kl.sum(-1, keepdim=True) / action_mask.sum(-1, keepdim=True).clamp(min=1.0)
+            adv_tok = adv_tok + self.beta * (avg - kl)
 
         clip_hi = 1.0 + self.clip_coef
         clip_lo = 1.0 - (self.clip_coef_lower if self.clip_coef_lower is not None else self.clip_coef)
@@ -204,7 +344,10 @@ class GRPO(LLMAlgorithm):
                     w0 = (i // accum) * accum
                     wsel = perm[starts[w0] : starts[w0] + mb * accum]
                     window_denom = float(action_mask[wsel].sum().clamp(min=1.0))
-                logp = logprob_fn(ids[sel], attention_mask[sel], with_grad=True)
+                from ...llm.offload import activation_offload
+
+                with activation_offload(self.activation_offload):
+                    logp = logprob_fn(ids[sel], attention_mask[sel], with_grad=True)
                 loss = self._policy_loss(
                     logp,
                     old_logp[sel],
@@ -232,6 +375,8 @@ class GRPO(LLMAlgorithm):
                         d = ref_logp[sel] - logp
                         stats["kl"] += float(ops.masked_mean(d.exp() - d - 1, m))
                 n_updates += 1
+        if self.lr_scheduler is not None:
+            self.lr_scheduler.step()
         if n_updates:
             stats = {k: v / n_updates for k, v in stats.items()}
         stats["mean_reward"] = float(rewards.mean())
@@ -269,6 +414,7 @@ class GRPO(LLMAlgorithm):
             level=self.importance_sampling_level, turn_ids=turn_ids,
             sampling_logp=sampling_logp, sampling_cap=self.sampling_is_cap,
             denom_tokens=denom_tokens,
+            force_eager=not getattr(self, "use_liger_loss", True),
         )
 
     # ------------------------------------------------------------------

@@ -103,6 +103,7 @@ def grpo_policy_loss(
     sampling_logp: Optional[torch.Tensor] = None,
     sampling_cap: float = 2.0,
     denom_tokens: Optional[float] = None,
+    force_eager: bool = False,
 ) -> torch.Tensor:
     """Scalar policy loss over token tensors.
 
@@ -138,7 +139,7 @@ def grpo_policy_loss(
 
     ext = extension()
     fused_ok = level == "token" and sampling_logp is None
-    if fused_ok and use_hip(logp) and ext is not None and logp.is_cuda:
+    if fused_ok and not force_eager and use_hip(logp) and ext is not None and logp.is_cuda:
         return _FusedGrpoLoss.apply(
             flat(logp), flat(old_logp),
             flat(ref_logp) if ref_logp is not None else None,

@@ -1182,3 +1182,78 @@ class TestReferenceMultiturnManifest:
         agent = trainer.__dict__.get("_population", None)
         # indirect check: the run completed through the multiturn loop with
         # the reference's mutation/selection sections applied
+
+
+class TestGrpoReferenceSurface(TestGrpoParityDeepening):
+    """Reference grpo.py:441-539 constructor surface: aliases, advantage
+    post-processing, loss_type routing, schedules."""
+
+    def test_aliases_map_to_native_options(self):
+        import warnings
+
+        with warnings.catch_warnings(record=True):
+            warnings.simplefilter("always")
+            a = self._tiny_agent(
+                use_sequence_packing=False,
+                vllm_importance_sampling_correction=True,
+                vllm_importance_sampling_cap=3.0,
+                micro_batch_size_per_gpu=1,
+                action_granularity="turn",
+                advantage_granularity="turn",
+                max_output_tokens=33,
+            )
+        assert a.use_packing is False
+        assert a.sampling_is_correction is True and a.sampling_is_cap == 3.0
+        assert a.micro_batch_size == 1
+        assert a.importance_sampling_level == "turn"
+        assert a.advantage_level == "turn"
+        assert a.max_completion_tokens == 33
+
+    def test_loss_type_routing(self):
+        assert self._tiny_agent(loss_type="cispo").CISPO is True
+        assert self._tiny_agent(loss_type="gspo").importance_sampling_level == "trajectory"
+        with pytest.raises(ValueError, match="loss_type"):
+            self._tiny_agent(loss_type="bogus")
+
+    def test_advantage_postprocessing(self):
+        a = self._tiny_agent(whiten_advantages=True, adv_clip_range=0.5,
+                             filter_zero_adv=True, adv_filter_eps=1e-6)
+        stats = a.learn(self._batch())
+        assert np.isfinite(stats["loss"])
+        b = self._tiny_agent(adv_norm="mean_only")
+        assert b.scale_rewards is False
+        with pytest.raises(ValueError, match="adv_norm"):
+            self._tiny_agent(adv_norm="bogus")
+
+    def test_kl_advantage_shaping_needs_beta(self):
+        a = self._tiny_agent(use_kl_advantage_shaping=True, beta=0.04)
+        stats = a.learn(self._batch())
+        assert np.isfinite(stats["loss"]) and stats["kl"] >= 0
+
+    def test_cosine_schedule_steps(self):
+        a = self._tiny_agent(
+            cosine_lr_schedule_config={"num_epochs": 4, "warmup_proportion": 0.25})
+        lr0 = a.optimizer.param_groups[0]["lr"]
+        a.learn(self._batch())
+        lr1 = a.optimizer.param_groups[0]["lr"]
+        assert lr1 != lr0  # warmup/cosine moved the lr
+
+    def test_ignored_infra_kwargs_warn_not_crash(self):
+        import warnings
+
+        with warnings.catch_warnings(record=True) as w:
+            warnings.simplefilter("always")
+            a = self._tiny_agent(quantization_config={"bits": 4},
+                                 reduce_memory_peak=True,
+                                 lora_target_scope="all-linear")
+        assert any("reference-only" in str(x.message) for x in w)
+        assert np.isfinite(a.learn(self._batch())["loss"])
+
+    def test_use_liger_loss_false_matches_fused_default(self):
+        torch.manual_seed(0)
+        a = self._tiny_agent(use_liger_loss=False)
+        torch.manual_seed(0)
+        b = self._tiny_agent()
+        batch = self._batch()
+        sa, sb = a.learn(batch), b.learn(batch)
+        assert abs(sa["loss"] - sb["loss"]) < 1e-5  # CPU: both eager anyway
