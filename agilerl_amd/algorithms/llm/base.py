@@ -495,6 +495,47 @@ class LLMAlgorithm(EvolvableAlgorithm):
                 f"(local={seq_len}); pad batches to a common length"
             )
 
+    # ------------------------------------------------------------------
+    # Reference-API kwargs shared by the LLM algorithms
+    # ------------------------------------------------------------------
+    _REF_IGNORED_KWARGS = (
+        "quantization_config", "use_memory_efficient_params",
+        "reduce_memory_peak", "calc_position_embeddings",
+        "cast_logprobs_to_fp32", "hf_generate_chunk_size",
+        "lora_target_scope", "use_separate_reference_adapter", "clone",
+        "actor_network", "max_model_len", "batch_size", "mini_batch_size",
+        "pad_token",
+    )
+
+    @classmethod
+    def _resolve_reference_llm_kwargs(cls, kwargs: Dict[str, Any]) -> Dict[str, Any]:
+        """Pop reference-spelling kwargs from ``kwargs`` and return the
+        recognized values under their native names.  Infra-only keys
+        (quantization, vLLM memory gymnastics — unnecessary on 288 GB
+        HBM3E) warn and are dropped; anything left in ``kwargs`` falls
+        through to ``_accept_compat_kwargs`` (TypeError on true typos)."""
+        import warnings
+
+        out: Dict[str, Any] = {}
+        for ref, ours in (("model_name", "model_name_or_path"),
+                          ("micro_batch_size_per_gpu", "micro_batch_size"),
+                          ("use_sequence_packing", "use_packing"),
+                          ("max_output_tokens", "max_completion_tokens")):
+            if ref in kwargs:
+                out[ours] = kwargs.pop(ref)
+        for k in ("seed", "chunk_rows", "pad_token_id", "activation_offload",
+                  "use_liger_loss", "update_epochs", "nll_alpha"):
+            if k in kwargs:
+                out[k] = kwargs.pop(k)
+        ignored = [k for k in cls._REF_IGNORED_KWARGS if kwargs.pop(k, None) is not None]
+        if ignored:
+            warnings.warn(
+                f"{cls.__name__} ignores reference-only kwargs {ignored} "
+                "(see docs/llm_finetuning.md for the MI355X equivalents)",
+                RuntimeWarning,
+            )
+        return out
+
     def backward_and_step(self, loss: torch.Tensor, accumulate: bool = False) -> None:
         loss.backward()
         if accumulate:

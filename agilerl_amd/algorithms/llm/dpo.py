@@ -39,8 +39,25 @@ class DPO(LLMAlgorithm):
         dtype: torch.dtype = torch.bfloat16,
         gradient_checkpointing: bool = False,
         use_packing: bool = False,
+        nll_alpha: float = 0.0,
+        seed=None,
+        chunk_rows=None,
+        activation_offload: bool = False,
+        pad_token_id=None,
         device: str = "cpu",
+        **kwargs,
     ):
+        ref = self._resolve_reference_llm_kwargs(kwargs)
+        model_name_or_path = ref.get("model_name_or_path", model_name_or_path)
+        micro_batch_size = ref.get("micro_batch_size", micro_batch_size)
+        use_packing = ref.get("use_packing", use_packing)
+        nll_alpha = ref.get("nll_alpha", nll_alpha)
+        seed = ref.get("seed", seed)
+        chunk_rows = ref.get("chunk_rows", chunk_rows)
+        activation_offload = ref.get("activation_offload", activation_offload)
+        pad_token_id = ref.get("pad_token_id", pad_token_id)
+        if seed is not None:
+            torch.manual_seed(int(seed))
         super().__init__(
             model=model, model_config=model_config, model_name_or_path=model_name_or_path,
             tokenizer=tokenizer, index=index, hp_config=hp_config, lora_config=lora_config,
@@ -48,14 +65,27 @@ class DPO(LLMAlgorithm):
             dtype=dtype, gradient_checkpointing=gradient_checkpointing, device=device,
             name="DPO",
         )
+        self._accept_compat_kwargs(**kwargs)
         self.beta = float(beta)
         self.label_smoothing = float(label_smoothing)
+        # reference fused_loss.py:740 LigerDPOWithAlpha: nll_alpha adds an
+        # SFT-style NLL term on the chosen completions to the DPO loss
+        self.nll_alpha = float(nll_alpha)
+        self.seed = seed
+        self.chunk_rows = chunk_rows
+        self.activation_offload = bool(activation_offload)
         # padding-free logprob passes (compute_logprobs_packed); opt-in
         self.use_packing = bool(use_packing)
+        if pad_token_id is not None and self.tokenizer is not None:
+            self.tokenizer.pad_token_id = pad_token_id
 
     def _seq_logp(self, ids, attention_mask, action_mask, with_grad: bool, adapter="self"):
+        from ...llm.offload import activation_offload
+
         fn = self.compute_logprobs_packed if self.use_packing else self.compute_logprobs
-        logp = fn(ids, attention_mask, adapter=adapter, with_grad=with_grad)
+        with activation_offload(self.activation_offload and with_grad):
+            logp = fn(ids, attention_mask, adapter=adapter, with_grad=with_grad,
+                      chunk_rows=self.chunk_rows)
         return (logp * action_mask).sum(dim=1)
 
     def learn(self, experiences: Dict[str, Any]) -> Dict[str, float]:
@@ -85,6 +115,11 @@ class DPO(LLMAlgorithm):
                 -F.logsigmoid(logits) * (1 - self.label_smoothing)
                 - F.logsigmoid(-logits) * self.label_smoothing
             ).mean()
+            if self.nll_alpha > 0:
+                # reference LigerDPOWithAlpha (fused_loss.py:740): SFT-style
+                # NLL on the chosen completions stabilizes DPO
+                nll = -(pol_c / c_mask[s:e].sum(dim=1).clamp(min=1.0)).mean()
+                loss = loss + self.nll_alpha * nll
             self.backward_and_step(loss)
             stats["loss"] += float(loss.detach())
             stats["margin"] += float(logits.detach().mean())

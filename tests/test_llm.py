@@ -1257,3 +1257,60 @@ class TestGrpoReferenceSurface(TestGrpoParityDeepening):
         batch = self._batch()
         sa, sb = a.learn(batch), b.learn(batch)
         assert abs(sa["loss"] - sb["loss"]) < 1e-5  # CPU: both eager anyway
+
+
+class TestSftDpoReferenceSurface:
+    """Shared reference-kwarg resolver on SFT/DPO (reference sft.py /
+    dpo.py constructor surfaces)."""
+
+    def _tiny_cfg(self):
+        return dict(model_type="llama", vocab_size=64, hidden_size=32,
+                    intermediate_size=64, num_hidden_layers=1,
+                    num_attention_heads=2, num_key_value_heads=1,
+                    max_position_embeddings=128, pad_token_id=0)
+
+    def test_sft_update_epochs_and_aliases(self):
+        import warnings
+
+        from agilerl_amd.algorithms.llm.sft import SFT
+
+        with warnings.catch_warnings(record=True) as w:
+            warnings.simplefilter("always")
+            a = SFT(model_config=self._tiny_cfg(), dtype=torch.float32,
+                    lora_config={"r": 2}, model_name=None,
+                    micro_batch_size_per_gpu=2, update_epochs=2, seed=3,
+                    chunk_rows=64, quantization_config={"bits": 8})
+        assert a.update_epochs == 2 and a.micro_batch_size == 2
+        assert any("reference-only" in str(x.message) for x in w)
+        g = torch.Generator().manual_seed(0)
+        ids = torch.randint(1, 64, (4, 10), generator=g)
+        mask = torch.ones(4, 9)
+        out = a.learn({"ids": ids, "action_mask": mask})
+        assert np.isfinite(out["loss"])
+
+    def test_dpo_nll_alpha_changes_loss(self):
+        from agilerl_amd.algorithms.llm.dpo import DPO
+
+        g = torch.Generator().manual_seed(1)
+        batch = {
+            "chosen_ids": torch.randint(1, 64, (4, 10), generator=g),
+            "rejected_ids": torch.randint(1, 64, (4, 10), generator=g),
+            "chosen_mask": torch.ones(4, 9),
+            "rejected_mask": torch.ones(4, 9),
+        }
+        torch.manual_seed(0)
+        a = DPO(model_config=self._tiny_cfg(), dtype=torch.float32,
+                lora_config={"r": 2}, nll_alpha=0.0)
+        torch.manual_seed(0)
+        b = DPO(model_config=self._tiny_cfg(), dtype=torch.float32,
+                lora_config={"r": 2}, nll_alpha=0.5)
+        la = a.learn({k: v.clone() for k, v in batch.items()})["loss"]
+        lb = b.learn({k: v.clone() for k, v in batch.items()})["loss"]
+        assert abs(la - lb) > 1e-6  # NLL term moved the loss
+
+    def test_unknown_kwarg_still_raises(self):
+        from agilerl_amd.algorithms.llm.sft import SFT
+
+        with pytest.raises(TypeError, match="bogus"):
+            SFT(model_config=self._tiny_cfg(), dtype=torch.float32,
+                lora_config={"r": 2}, bogus_kwarg=1)
