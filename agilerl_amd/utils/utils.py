@@ -24,6 +24,10 @@ __all__ = [
     "run_selection_and_mutation",
     "get_env_defined_actions",
     "observation_space_channels_to_first",
+    "make_multi_agent_vect_envs",
+    "make_skill_vect_envs",
+    "calculate_vectorized_scores",
+    "print_hyperparams",
 ]
 
 
@@ -126,3 +130,81 @@ def log_gpu_memory_snapshot(prefix: str = "") -> Dict[str, float]:
     print(f"[gpu-mem]{(' ' + prefix) if prefix else ''} " +
           " ".join(f"{k}={v:.2f}" for k, v in out.items()), flush=True)
     return out
+
+
+def make_multi_agent_vect_envs(env, num_envs: int = 1, *, extra_wrappers=None,
+                               **env_kwargs):
+    """Async-vectorize a PettingZoo-style parallel-env factory
+    (reference utils.py:275): ``env`` is a callable returning one parallel
+    env; workers run in shared-memory subprocesses
+    (:class:`agilerl_amd.vector.AsyncPettingZooVecEnv`)."""
+    from ..vector.async_vec_env import AsyncPettingZooVecEnv
+
+    factory = env
+    if extra_wrappers:
+        def factory(**kw):
+            e = env(**kw)
+            for wrapper_cls in extra_wrappers:
+                e = wrapper_cls(e)
+            return e
+
+    return AsyncPettingZooVecEnv([
+        (lambda: factory(**env_kwargs)) for _ in range(num_envs)
+    ])
+
+
+def make_skill_vect_envs(env_id, skill, num_envs: int = 1, seed=None, **env_kwargs):
+    """Vectorized env with a curriculum :class:`~agilerl_amd.wrappers.Skill`
+    applied (reference utils.py:308).  Works with the first-party batched
+    envs: the Skill wraps the whole vec env (its reward hook sees batched
+    rewards)."""
+    from ..envs.registry import make_vect_envs
+
+    return skill(make_vect_envs(env_id, num_envs=num_envs, seed=seed, **env_kwargs))
+
+
+def calculate_vectorized_scores(rewards, terminations, include_unterminated: bool = False,
+                                only_first_episode: bool = True):
+    """Episode returns from (num_envs, T) reward/termination arrays
+    (reference utils.py:1588): segments each env's reward stream at its
+    termination points."""
+    import numpy as np
+
+    rewards = np.asarray(rewards)
+    terminations = np.asarray(terminations)
+    out = []
+    for env_idx in range(rewards.shape[0]):
+        term_idx = np.flatnonzero(terminations[env_idx] == 1)
+        if term_idx.size == 0:
+            out.append(float(rewards[env_idx].sum()))
+            continue
+        start = 0
+        for t in term_idx:
+            out.append(float(rewards[env_idx, start:t + 1].sum()))
+            if only_first_episode:
+                break
+            start = t + 1
+        if not only_first_episode and include_unterminated and start < rewards.shape[1]:
+            out.append(float(rewards[env_idx, start:].sum()))
+    return out
+
+
+def print_hyperparams(pop) -> None:
+    """Print each agent's mutable hyperparameters and recent fitness
+    (reference utils.py:1651)."""
+    import numpy as np
+
+    for agent in pop:
+        mean_fitness = (
+            float(np.mean(agent.fitness[-5:])) if agent.fitness else float("nan")
+        )
+        attrs = agent.inspect_attributes()
+        hps = {k: attrs[k] for k in getattr(agent.hp_config, "names", lambda: [])()
+               if k in attrs} or {
+            k: v for k, v in attrs.items()
+            if isinstance(v, (int, float)) and not k.startswith("_")
+        }
+        print(f"Agent ID: {agent.index}  |  Mean 5 Fitness: {mean_fitness:.2f}  |  "
+              f"Last mutation: {getattr(agent, 'mut', 'None')}")
+        for k in sorted(hps):
+            print(f"  {k}: {hps[k]}")

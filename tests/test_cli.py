@@ -248,3 +248,41 @@ class TestReferenceBanditConfig:
         m = TrainingManifest.model_validate(doc)
         out = LocalTrainer(m, device="cpu").train()
         assert out is not None
+
+
+class TestUtilsParityFunctions:
+    """Reference utils.py:275/:308/:1588/:1651 convenience functions."""
+
+    def test_calculate_vectorized_scores(self):
+        import numpy as np
+
+        from agilerl_amd.utils.utils import calculate_vectorized_scores
+
+        r = np.array([[1, 1, 1, 1], [2, 2, 2, 2]], dtype=float)
+        t = np.array([[0, 1, 0, 1], [0, 0, 0, 0]])
+        assert calculate_vectorized_scores(r, t) == [2.0, 8.0]
+        assert calculate_vectorized_scores(r, t, only_first_episode=False) == [2.0, 2.0, 8.0]
+
+    def test_make_skill_vect_envs_reward_hook(self):
+        from agilerl_amd.utils.utils import make_skill_vect_envs
+        from agilerl_amd.wrappers.learning import Skill
+
+        class Doubler(Skill):
+            def skill_reward(self, obs, reward, terminated, truncated, info):
+                return reward * 2, terminated, truncated
+
+        env = make_skill_vect_envs("CartPole-v1", Doubler, num_envs=2, seed=0)
+        env.reset()
+        import numpy as np
+
+        _, reward, _, _, _ = env.step(np.zeros(2, dtype=np.int64))
+        assert (reward == 2.0).all()
+
+    def test_print_hyperparams_runs(self, capsys):
+        from agilerl_amd.algorithms.dqn import DQN
+        from agilerl_amd.spaces import Box, Discrete
+        from agilerl_amd.utils.utils import print_hyperparams
+
+        print_hyperparams([DQN(Box(-1, 1, (4,)), Discrete(2))])
+        out = capsys.readouterr().out
+        assert "Agent ID: 0" in out and "lr" in out
