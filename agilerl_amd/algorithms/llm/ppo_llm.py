@@ -22,8 +22,28 @@ __all__ = ["PPOLLM"]
 class PPOLLM(GRPO):
     """GRPO machinery with a value-head baseline instead of group centering."""
 
-    def __init__(self, *args, vf_coef: float = 0.5, value_head_lr: Optional[float] = None, **kwargs):
+    def __init__(self, *args, vf_coef: float = 0.5, value_head_lr: Optional[float] = None,
+                 gamma: float = 1.0, gae_lambda: float = 1.0, **kwargs):
         kwargs.setdefault("group_size", 1)
+        self.gamma = float(gamma)
+        self.gae_lambda = float(gae_lambda)
+        # reference ppo_llm.py multi-turn kwargs: turn_ratio_pooling pools
+        # the IS ratio per turn (our importance_sampling_level="turn");
+        # turn_level_clip clips at that pooled level (implied by the level);
+        # turn_value_reduction is vLLM-era value bookkeeping (ignored)
+        if kwargs.pop("turn_ratio_pooling", False) or kwargs.pop("turn_level_clip", False):
+            kwargs.setdefault("importance_sampling_level", "turn")
+        if kwargs.pop("turn_value_reduction", None) is not None:
+            import warnings
+
+            warnings.warn("turn_value_reduction ignored: the value head pools "
+                          "at the last prompt token", RuntimeWarning)
+        if "lr_actor" in kwargs:
+            kwargs.setdefault("lr", kwargs.pop("lr_actor"))
+        if "lr_critic" in kwargs:
+            value_head_lr = value_head_lr or kwargs.pop("lr_critic")
+        else:
+            kwargs.pop("lr_critic", None)
         super().__init__(*args, **kwargs)
         self.algo = "PPOLLM"
         self.vf_coef = float(vf_coef)
@@ -54,6 +74,48 @@ class PPOLLM(GRPO):
         # placeholder — learn() computes advantage with the value baseline
         return rewards
 
+    def _turn_values(self, ids, attention_mask, turn_ids, with_grad: bool = False,
+                     num_turns=None):
+        """V per turn: value head on the hidden state at each turn's last
+        token (reference ppo_llm.py turn-level critic).  turn_ids indexes
+        TARGETS (position j predicts ids[j+1]), so turn k's last hidden
+        state sits at ids position (last j with turn_ids==k) + 1."""
+        self._activate("self")
+        B = ids.shape[0]
+        K = int(num_turns) if num_turns is not None else int(turn_ids.max().item()) + 1
+        ctx = torch.enable_grad() if with_grad else torch.no_grad()
+        with ctx:
+            hidden = self._decoder()(
+                input_ids=ids, attention_mask=attention_mask
+            ).last_hidden_state
+            T1 = turn_ids.shape[1]
+            pos = torch.arange(T1, device=ids.device).view(1, 1, T1)
+            is_turn = turn_ids.unsqueeze(1) == torch.arange(K, device=ids.device).view(1, K, 1)
+            # (B, K): last target index of each turn (-1 if unplayed)
+            last = torch.where(is_turn, pos, torch.full_like(pos, -1)).amax(dim=2)
+            played = last >= 0
+            gather_pos = (last.clamp(min=0) + 1).clamp(max=ids.shape[1] - 1)
+            pooled = hidden[torch.arange(B, device=ids.device).unsqueeze(1), gather_pos]
+            vals = self.value_head(pooled.float()).squeeze(-1)
+            return vals * played.float(), played
+
+    def _turn_gae(self, turn_rewards, turn_values, played):
+        """GAE across turns (reference ppo_llm.py:1141): gamma discounts
+        BETWEEN turns, not within; unplayed turns carry nothing."""
+        B, K = turn_rewards.shape
+        n_turns = played.float().sum(dim=1)  # per-sample played count
+        adv = torch.zeros_like(turn_rewards)
+        last_gae = torch.zeros(B, device=turn_rewards.device)
+        for t in reversed(range(K)):
+            is_last = (t >= n_turns - 1)
+            next_v = turn_values[:, t + 1] if t < K - 1 else torch.zeros(B, device=turn_rewards.device)
+            next_v = torch.where(is_last, torch.zeros_like(next_v), next_v)
+            delta = turn_rewards[:, t] + self.gamma * next_v - turn_values[:, t]
+            has = (n_turns > t).float()
+            last_gae = (delta + self.gamma * self.gae_lambda * last_gae) * has
+            adv[:, t] = last_gae
+        return adv
+
     def learn(self, experiences: Dict[str, Any]) -> Dict[str, float]:
         ids = experiences["ids"].to(self.device)
         attention_mask = experiences.get("attention_mask")
@@ -68,6 +130,13 @@ class PPOLLM(GRPO):
         else:
             sampling_logps = None
 
+        turn_ids = experiences.get("turn_ids")
+        turn_rewards = experiences.get("turn_rewards")
+        multi_turn = turn_ids is not None and turn_rewards is not None
+        if multi_turn:
+            turn_ids = turn_ids.to(self.device)
+            turn_rewards = turn_rewards.to(self.device).float()
+
         B = ids.shape[0]
         mb = max(self.micro_batch_size, 1)
 
@@ -76,17 +145,43 @@ class PPOLLM(GRPO):
             values = torch.empty(B, device=self.device)
             old_logp = torch.empty(action_mask.shape, device=self.device)
             ref_logp = torch.empty_like(old_logp) if self.beta > 0 else None
+            tv = played = None
+            if multi_turn:
+                K = int(turn_ids.max().item()) + 1
+                tv = torch.empty(B, K, device=self.device)
+                played = torch.empty(B, K, dtype=torch.bool, device=self.device)
             for s in range(0, B, mb):
                 e = min(s + mb, B)
-                values[s:e] = self._sequence_values(ids[s:e], attention_mask[s:e])
+                if multi_turn:
+                    tv[s:e], played[s:e] = self._turn_values(
+                        ids[s:e], attention_mask[s:e], turn_ids[s:e], num_turns=K
+                    )
+                else:
+                    values[s:e] = self._sequence_values(ids[s:e], attention_mask[s:e])
                 old_logp[s:e] = self.compute_logprobs(ids[s:e], attention_mask[s:e])
                 if ref_logp is not None:
                     ref_logp[s:e] = self.compute_logprobs(ids[s:e], attention_mask[s:e], adapter=None)
 
-        advantages = rewards - values
-        if advantages.numel() > 1:
-            advantages = (advantages - advantages.mean()) / (advantages.std() + 1e-8)
-        adv_tok = advantages.unsqueeze(1).expand_as(action_mask)
+        if multi_turn:
+            # reference ppo_llm.py:1141 turn-level GAE: each turn is one RL
+            # action; gamma/gae_lambda act across turns
+            K = tv.shape[1]
+            tr = turn_rewards[:, :K] if turn_rewards.shape[1] >= K else torch.nn.functional.pad(
+                turn_rewards, (0, K - turn_rewards.shape[1]))
+            turn_adv = self._turn_gae(tr, tv, played)
+            turn_returns = turn_adv + tv
+            flat = turn_adv[played]
+            if flat.numel() > 1:
+                turn_adv = torch.where(
+                    played, (turn_adv - flat.mean()) / (flat.std() + 1e-8),
+                    torch.zeros_like(turn_adv))
+            safe = turn_ids.clamp(min=0, max=K - 1).long()
+            adv_tok = turn_adv.gather(1, safe) * (turn_ids >= 0).float() * action_mask
+        else:
+            advantages = rewards - values
+            if advantages.numel() > 1:
+                advantages = (advantages - advantages.mean()) / (advantages.std() + 1e-8)
+            adv_tok = advantages.unsqueeze(1).expand_as(action_mask)
 
         clip_hi, clip_lo = 1.0 + self.clip_coef, 1.0 - self.clip_coef
         stats = {"loss": 0.0, "value_loss": 0.0}
@@ -102,9 +197,20 @@ class PPOLLM(GRPO):
                     sampling_logp=sampling_logps[sel] if sampling_logps is not None else None,
                 )
                 self.backward_and_step(loss)
-                # value head regression toward realized reward
-                v = self._sequence_values(ids[sel], attention_mask[sel], with_grad=True)
-                v_loss = self.vf_coef * ((v - rewards[sel]) ** 2).mean()
+                # value head regression toward realized reward / turn returns
+                if multi_turn:
+                    v_sel, p_sel = self._turn_values(
+                        ids[sel], attention_mask[sel], turn_ids[sel], with_grad=True,
+                        num_turns=turn_returns.shape[1],
+                    )
+                    tgt = turn_returns[sel]
+                    pf = p_sel.float()
+                    v_loss = self.vf_coef * (
+                        ((v_sel - tgt) ** 2 * pf).sum() / pf.sum().clamp(min=1.0)
+                    )
+                else:
+                    v = self._sequence_values(ids[sel], attention_mask[sel], with_grad=True)
+                    v_loss = self.vf_coef * ((v - rewards[sel]) ** 2).mean()
                 self.value_optimizer.zero_grad()
                 v_loss.backward()
                 self.value_optimizer.step()
