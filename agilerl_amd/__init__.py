@@ -33,6 +33,7 @@ def _importable(name: str) -> bool:
 
 
 HAS_LLM_DEPENDENCIES = _importable("transformers")
+HAS_ARENA_DEPENDENCIES = _importable("fastapi") and _importable("httpx")
 HAS_HIP_KERNELS = None  # resolved lazily below
 
 # Reference-familiar flags: these stacks are REPLACED by first-party
@@ -85,6 +86,7 @@ __all__ = [
     "__version__",
     "AgentType",
     "HAS_LLM_DEPENDENCIES",
+    "HAS_ARENA_DEPENDENCIES",
     "HAS_HIP_KERNELS",
     "HAS_VLLM",
     "HAS_DEEPSPEED",

@@ -48,11 +48,16 @@ class ReplayBuffer:
         storage_device: Optional[str] = None,
         pin_memory: bool = True,
         seed: Optional[int] = None,
+        dtype: Optional[torch.dtype] = None,
     ):
         self.max_size = int(max_size)
         self.device = device
         self.storage_device = storage_device or "cpu"
         self.pin_memory = pin_memory
+        # reference components/replay_buffer.py `dtype`: storage dtype for
+        # float data (e.g. torch.float16 halves resident bytes); learners
+        # upcast on consumption
+        self.dtype = dtype
         self._storage: Optional[Dict[str, Any]] = None
         self._ptr = 0
         self._size = 0
@@ -90,6 +95,10 @@ class ReplayBuffer:
             batch = ref.shape[0]
         # float64 -> float32 (envs often emit float64 rewards)
         data = tree_map(lambda t: t.float() if t.dtype == torch.float64 else t, data)
+        if self.dtype is not None:
+            data = tree_map(
+                lambda t: t.to(self.dtype) if t.is_floating_point() else t, data
+            )
         return data, batch
 
     @torch.no_grad()

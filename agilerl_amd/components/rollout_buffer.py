@@ -33,6 +33,13 @@ class RolloutBuffer:
         gamma: float = 0.99,
         gae_lambda: float = 0.95,
         recurrent: bool = False,
+        observation_space=None,
+        action_space=None,
+        use_gae: bool = True,
+        wrap_at_capacity: bool = False,
+        max_seq_len: Optional[int] = None,
+        bptt_sequence_type: str = "chunked",
+        hidden_state_architecture=None,
     ):
         self.capacity = int(capacity)
         self.num_envs = int(num_envs)
@@ -40,6 +47,18 @@ class RolloutBuffer:
         self.gamma = float(gamma)
         self.gae_lambda = float(gae_lambda)
         self.recurrent = recurrent
+        # reference rollout_buffer.py constructor surface: spaces are
+        # informational (storage shapes come from the first add), use_gae
+        # toggles the lambda-weighted scan vs plain discounted returns,
+        # wrap_at_capacity keeps adding as a ring instead of raising,
+        # max_seq_len/bptt_sequence_type are the BPTT slicing defaults
+        self.observation_space = observation_space
+        self.action_space = action_space
+        self.use_gae = bool(use_gae)
+        self.wrap_at_capacity = bool(wrap_at_capacity)
+        self.max_seq_len = max_seq_len
+        self.bptt_sequence_type = bptt_sequence_type
+        self.hidden_state_architecture = hidden_state_architecture
 
         self._storage: Optional[Dict[str, Any]] = None
         self.pos = 0
@@ -106,6 +125,13 @@ class RolloutBuffer:
         data = tree_map(lambda t: t.to(self.device), data)
         if self._storage is None:
             self._init_storage(data)
+        if self.pos >= self.capacity:
+            if not self.wrap_at_capacity:
+                raise RuntimeError(
+                    f"RolloutBuffer full ({self.capacity} steps); pass "
+                    "wrap_at_capacity=True to overwrite as a ring"
+                )
+            self.pos = 0
         for key, val in data.items():
             _write(self._storage[key], val, self.pos)
         self.pos += 1
@@ -123,9 +149,14 @@ class RolloutBuffer:
         values = sd["value"][:T].reshape(T, self.num_envs)
         rewards = sd["reward"][:T].reshape(T, self.num_envs)
         dones = sd["done"][:T].reshape(T, self.num_envs).float()
+        lam = self.gae_lambda if self.use_gae else 1.0
         adv, ret = ops.gae_scan(
-            rewards, values, dones, last_value, self.gamma, self.gae_lambda
+            rewards, values, dones, last_value, self.gamma, lam
         )
+        if not self.use_gae:
+            # plain discounted returns: lambda=1 GAE gives ret = discounted
+            # rewards-to-go; advantage = ret - V
+            adv = ret - values
         self.advantages = adv
         self.returns = ret
 
@@ -150,8 +181,8 @@ class RolloutBuffer:
             yield tree_map(lambda t: t[mb_idx], flat)
 
     def get_sequence_minibatches(
-        self, seq_len: int, batch_size: int, shuffle: bool = True,
-        sequence_type: str = "chunked",
+        self, seq_len: Optional[int] = None, batch_size: int = 64,
+        shuffle: bool = True, sequence_type: Optional[str] = None,
     ) -> Iterator[Dict[str, torch.Tensor]]:
         """(B, L, ...) contiguous same-env sequences for BPTT (recurrent PPO).
 
@@ -159,6 +190,8 @@ class RolloutBuffer:
         start stride: "chunked" = non-overlapping (stride L), "maximum" = all
         overlapping windows (stride 1), "fifty_percent_overlap" = stride L/2.
         """
+        seq_len = int(seq_len if seq_len is not None else (self.max_seq_len or 16))
+        sequence_type = sequence_type or self.bptt_sequence_type
         stride = {"chunked": seq_len, "maximum": 1,
                   "fifty_percent_overlap": max(seq_len // 2, 1)}.get(sequence_type)
         if stride is None:

@@ -25,7 +25,34 @@ class MultiFrequencySelection:
         elitism: bool = True,
         migration_fraction: float = 0.25,
         rng: Optional[np.random.Generator] = None,
+        evolution_frequency_ratios: Optional[Sequence[int]] = None,
+        population_size: Optional[int] = None,
+        n_subpopulations: Optional[int] = None,
+        n_winners: Optional[int] = None,
+        n_losers: Optional[int] = None,
+        n_survivors: Optional[int] = None,
+        n_open_for_migration: Optional[int] = None,
+        seed: Optional[int] = None,
     ):
+        # reference hpo/multi_frequency.py spellings: frequency ratios are
+        # the same thing as `frequencies`; the n_* counts fold into the
+        # migration fraction (winners migrate into open slots)
+        if evolution_frequency_ratios is not None:
+            frequencies = evolution_frequency_ratios
+        if n_subpopulations is not None and len(tuple(frequencies)) != int(n_subpopulations):
+            raise ValueError(
+                f"n_subpopulations={n_subpopulations} does not match "
+                f"{len(tuple(frequencies))} frequencies"
+            )
+        self.population_size = population_size
+        if n_winners is not None and n_open_for_migration:
+            migration_fraction = min(1.0, n_winners / max(n_open_for_migration, 1))
+        self.n_winners = n_winners
+        self.n_losers = n_losers
+        self.n_survivors = n_survivors
+        self.n_open_for_migration = n_open_for_migration
+        if seed is not None and rng is None:
+            rng = np.random.default_rng(seed)
         self.frequencies = tuple(int(f) for f in frequencies)
         self.elitism = elitism
         self.migration_fraction = migration_fraction

@@ -35,7 +35,16 @@ class Mutations:
         mutate_elite: bool = True,
         rand_seed: Optional[int] = None,
         device: str = "cpu",
+        accelerator=None,
     ):
+        if accelerator is not None:
+            import warnings
+
+            warnings.warn(
+                "Mutations ignores `accelerator` (reference Accelerate-era "
+                "kwarg): distributed mutation consensus runs over RCCL via "
+                "agilerl_amd.parallel.", RuntimeWarning,
+            )
         self.no_mutation = no_mutation
         self.architecture = architecture
         self.new_layer_prob = new_layer_prob

@@ -89,38 +89,78 @@ class RSNorm(AgentWrapper):
     transitions are raw; normalization applied on the way in).
     """
 
-    def __init__(self, agent, epsilon: float = 1e-4, clip: float = 10.0):
+    def __init__(self, agent, epsilon: float = 1e-4, clip: float = 10.0,
+                 norm_obs_keys=None):
         super().__init__(agent)
         space = agent.observation_space
         from ..spaces import DictSpace, TupleSpace, space_shape
 
-        if isinstance(space, (DictSpace, TupleSpace)):
-            raise TypeError("RSNorm supports flat observation spaces")
-        object.__setattr__(self, "rms", RunningMeanStd(space_shape(space), epsilon))
+        if isinstance(space, TupleSpace):
+            raise TypeError("RSNorm supports flat or Dict observation spaces")
+        if isinstance(space, DictSpace):
+            # reference wrappers/agent.py norm_obs_keys: normalize only the
+            # listed Dict-obs keys (default: every key)
+            keys = list(norm_obs_keys) if norm_obs_keys is not None else list(space.spaces)
+            unknown = [k for k in keys if k not in space.spaces]
+            if unknown:
+                raise KeyError(f"norm_obs_keys {unknown} not in observation space")
+            object.__setattr__(self, "norm_obs_keys", keys)
+            object.__setattr__(self, "rms", {
+                k: RunningMeanStd(space_shape(space.spaces[k]), epsilon) for k in keys
+            })
+        else:
+            object.__setattr__(self, "norm_obs_keys", None)
+            object.__setattr__(self, "rms", RunningMeanStd(space_shape(space), epsilon))
         object.__setattr__(self, "clip", clip)
 
     def _norm(self, obs):
+        if isinstance(self.rms, dict):
+            out = dict(obs)
+            for k in self.norm_obs_keys:
+                out[k] = self._norm_one(out[k], self.rms[k])
+            return out
+        return self._norm_one(obs, self.rms)
+
+    def _norm_one(self, obs, rms):
         if isinstance(obs, torch.Tensor):
-            mean = torch.as_tensor(self.rms.mean, dtype=obs.dtype, device=obs.device)
-            std = torch.as_tensor(np.sqrt(self.rms.var + 1e-8), dtype=obs.dtype, device=obs.device)
+            mean = torch.as_tensor(rms.mean, dtype=obs.dtype, device=obs.device)
+            std = torch.as_tensor(np.sqrt(rms.var + 1e-8), dtype=obs.dtype, device=obs.device)
             return ((obs - mean) / std).clamp(-self.clip, self.clip)
-        return np.clip(self.rms.normalize(obs), -self.clip, self.clip).astype(np.float32)
+        return np.clip(rms.normalize(obs), -self.clip, self.clip).astype(np.float32)
 
     def get_action(self, obs, *args, training: bool = True, **kwargs):
         if training:
-            self.rms.update(obs if not isinstance(obs, torch.Tensor) else obs.cpu().numpy())
+            if isinstance(self.rms, dict):
+                for k in self.norm_obs_keys:
+                    v = obs[k]
+                    self.rms[k].update(v if not isinstance(v, torch.Tensor) else v.cpu().numpy())
+            else:
+                self.rms.update(obs if not isinstance(obs, torch.Tensor) else obs.cpu().numpy())
         return self.agent.get_action(self._norm(obs), *args, training=training, **kwargs)
 
     def learn(self, experiences, *args, **kwargs):
         if isinstance(experiences, dict):
             experiences = dict(experiences)
             for key in ("obs", "next_obs"):
-                if key in experiences and not isinstance(experiences[key], dict):
+                if key not in experiences:
+                    continue
+                if isinstance(experiences[key], dict):
+                    if isinstance(self.rms, dict):
+                        experiences[key] = self._norm(experiences[key])
+                else:
                     experiences[key] = self._norm(experiences[key])
         return self.agent.learn(experiences, *args, **kwargs)
 
     # -- persistence: normalizer stats ride inside the checkpoint --------
     def wrapper_state(self) -> dict:
+        if isinstance(self.rms, dict):
+            return {
+                "cls": type(self).__name__,
+                "keys": list(self.norm_obs_keys),
+                "per_key": {k: {"mean": r.mean.copy(), "var": r.var.copy(),
+                                "count": float(r.count)} for k, r in self.rms.items()},
+                "clip": self.clip,
+            }
         return {
             "cls": type(self).__name__,
             "mean": self.rms.mean.copy(),
@@ -130,9 +170,15 @@ class RSNorm(AgentWrapper):
         }
 
     def load_wrapper_state(self, state: dict) -> None:
-        self.rms.mean[...] = state["mean"]
-        self.rms.var[...] = state["var"]
-        self.rms.count = state["count"]
+        if "per_key" in state:
+            for k, st in state["per_key"].items():
+                self.rms[k].mean[...] = st["mean"]
+                self.rms[k].var[...] = st["var"]
+                self.rms[k].count = st["count"]
+        else:
+            self.rms.mean[...] = state["mean"]
+            self.rms.var[...] = state["var"]
+            self.rms.count = state["count"]
         object.__setattr__(self, "clip", state.get("clip", self.clip))
 
     def save_checkpoint(self, path: str) -> None:

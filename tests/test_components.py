@@ -233,3 +233,84 @@ class TestGroupAdvantage:
         adv = ops.group_advantage(r, group_size=3, scale=False)
         assert adv[:3].sum() == pytest.approx(0.0, abs=1e-5)
         assert adv[3:].sum() == pytest.approx(0.0, abs=1e-5)
+
+
+class TestReferenceBufferSurface:
+    """Reference buffer constructor kwargs (components/replay_buffer.py
+    dtype, rollout_buffer.py use_gae/wrap_at_capacity)."""
+
+    def test_replay_buffer_storage_dtype(self):
+        import numpy as np
+        import torch
+
+        from agilerl_amd.components import ReplayBuffer
+
+        b = ReplayBuffer(100, dtype=torch.float16)
+        b.add(obs=np.random.randn(4, 3).astype(np.float32),
+              action=np.zeros(4, np.int64), reward=np.ones(4, np.float32),
+              next_obs=np.random.randn(4, 3).astype(np.float32),
+              done=np.zeros(4, np.float32))
+        s = b.sample(2)
+        assert s["obs"].dtype == torch.float16
+        assert s["action"].dtype == torch.int64  # ints untouched
+
+    def test_rollout_use_gae_false_discounted_returns(self):
+        import numpy as np
+        import torch
+
+        from agilerl_amd.components.rollout_buffer import RolloutBuffer
+
+        buf = RolloutBuffer(3, num_envs=1, gamma=0.5, use_gae=False)
+        for r in (1.0, 1.0, 1.0):
+            buf.add(obs=np.zeros((1, 2), np.float32), action=np.zeros((1,), np.int64),
+                    reward=np.array([r], np.float32), done=np.zeros(1, np.float32),
+                    value=np.array([0.25], np.float32), log_prob=np.zeros(1, np.float32))
+        buf.compute_returns_and_advantages(np.zeros(1, np.float32))
+        # rewards-to-go with gamma 0.5: [1.75, 1.5, 1.0]
+        assert torch.allclose(buf.returns.reshape(-1),
+                              torch.tensor([1.75, 1.5, 1.0]))
+        assert torch.allclose(buf.advantages, buf.returns - 0.25)
+
+    def test_rollout_wrap_at_capacity(self):
+        import numpy as np
+        import pytest
+
+        from agilerl_amd.components.rollout_buffer import RolloutBuffer
+
+        def one(buf, r):
+            buf.add(obs=np.zeros((1, 2), np.float32), action=np.zeros((1,), np.int64),
+                    reward=np.array([r], np.float32), done=np.zeros(1, np.float32),
+                    value=np.zeros(1, np.float32), log_prob=np.zeros(1, np.float32))
+
+        strict = RolloutBuffer(2, num_envs=1)
+        one(strict, 1.0); one(strict, 2.0)
+        with pytest.raises(RuntimeError, match="wrap_at_capacity"):
+            one(strict, 3.0)
+        ring = RolloutBuffer(2, num_envs=1, wrap_at_capacity=True)
+        one(ring, 1.0); one(ring, 2.0); one(ring, 3.0)  # overwrites slot 0
+        assert float(ring._storage["reward"][0]) == 3.0
+
+    def test_rsnorm_dict_norm_obs_keys(self):
+        import numpy as np
+
+        from agilerl_amd.spaces import Box, DictSpace
+        from agilerl_amd.wrappers.agent import RSNorm
+
+        class FakeAgent:
+            observation_space = DictSpace({"v": Box(-1, 1, (3,)), "img": Box(0, 255, (2, 2))})
+
+            def get_action(self, obs, training=True):
+                return obs
+
+        w = RSNorm(FakeAgent(), norm_obs_keys=["v"])
+        obs = {"v": np.random.randn(4, 3).astype(np.float32) * 5 + 2,
+               "img": np.zeros((4, 2, 2), np.float32)}
+        for _ in range(50):
+            w.get_action(obs)
+        out = w.get_action(obs, training=False)
+        assert abs(out["v"].mean()) < 1.0      # normalized
+        assert (out["img"] == 0).all()         # untouched key
+        st = w.wrapper_state()
+        w2 = RSNorm(FakeAgent(), norm_obs_keys=["v"])
+        w2.load_wrapper_state(st)
+        assert np.allclose(w2.rms["v"].mean, w.rms["v"].mean)
