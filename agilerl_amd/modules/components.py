@@ -65,7 +65,8 @@ class NoisyLinear(nn.Module):
     ``f(x) = sign(x) * sqrt(|x|)``.
     """
 
-    def __init__(self, in_features: int, out_features: int, std_init: float = 0.5):
+    def __init__(self, in_features: int, out_features: int, std_init: float = 0.5,
+                 device=None):
         super().__init__()
         self.in_features = in_features
         self.out_features = out_features
@@ -80,6 +81,8 @@ class NoisyLinear(nn.Module):
 
         self.reset_parameters()
         self.reset_noise()
+        if device is not None:
+            self.to(device)
 
     def reset_parameters(self) -> None:
         bound = 1.0 / math.sqrt(self.in_features)

@@ -45,14 +45,19 @@ class CausalSelfAttention(nn.Module):
 
 
 class GPTBlock(nn.Module):
-    def __init__(self, n_embd: int, n_head: int, dropout: float = 0.0, causal: bool = True):
+    def __init__(self, n_embd: int, n_head: int, dropout: float = 0.0, causal: bool = True,
+                 dim_feedfwd: int = 0, activation: str = "NewGELU",
+                 layer_norm_eps: float = 1e-5, bias: bool = True):
         super().__init__()
-        self.ln1 = nn.LayerNorm(n_embd)
+        from .components import get_activation
+
+        ffwd = dim_feedfwd or 4 * n_embd
+        self.ln1 = nn.LayerNorm(n_embd, eps=layer_norm_eps, bias=bias)
         self.attn = CausalSelfAttention(n_embd, n_head, dropout)
-        self.ln2 = nn.LayerNorm(n_embd)
+        self.ln2 = nn.LayerNorm(n_embd, eps=layer_norm_eps, bias=bias)
         self.mlp = nn.Sequential(
-            nn.Linear(n_embd, 4 * n_embd), NewGELU(), nn.Linear(4 * n_embd, n_embd),
-            nn.Dropout(dropout),
+            nn.Linear(n_embd, ffwd, bias=bias), get_activation(activation),
+            nn.Linear(ffwd, n_embd, bias=bias), nn.Dropout(dropout),
         )
         self.causal = causal
 
@@ -72,12 +77,26 @@ class EvolvableGPT(EvolvableModule):
         n_head: int = 4,
         n_embd: int = 128,
         max_positions: int = 512,
+        block_size: Optional[int] = None,
         dropout: float = 0.0,
+        dim_feedfwd: int = 0,
+        activation: str = "NewGELU",
+        layer_norm_eps: float = 1e-5,
+        bias: bool = True,
         min_layers: int = 1,
         max_layers: int = 12,
         device: str = "cpu",
+        name: Optional[str] = None,
+        random_seed: Optional[int] = None,
     ):
-        super().__init__(device)
+        super().__init__(device, name=name, random_seed=random_seed)
+        # reference modules/gpt.py spells the context window `block_size`
+        if block_size is not None:
+            max_positions = int(block_size)
+        self.dim_feedfwd = int(dim_feedfwd)
+        self.activation = activation
+        self.layer_norm_eps = float(layer_norm_eps)
+        self.bias = bool(bias)
         self.vocab_size = int(vocab_size)
         self.n_layer = int(n_layer)
         self.n_head = int(n_head)
@@ -94,10 +113,12 @@ class EvolvableGPT(EvolvableModule):
                 wte=nn.Embedding(self.vocab_size, self.n_embd),
                 wpe=nn.Embedding(self.max_positions, self.n_embd),
                 blocks=nn.ModuleList(
-                    GPTBlock(self.n_embd, self.n_head, self.dropout, self.CAUSAL)
+                    GPTBlock(self.n_embd, self.n_head, self.dropout, self.CAUSAL,
+                             dim_feedfwd=self.dim_feedfwd, activation=self.activation,
+                             layer_norm_eps=self.layer_norm_eps, bias=self.bias)
                     for _ in range(self.n_layer)
                 ),
-                ln_f=nn.LayerNorm(self.n_embd),
+                ln_f=nn.LayerNorm(self.n_embd, eps=self.layer_norm_eps, bias=self.bias),
                 head=nn.Linear(self.n_embd, self.vocab_size, bias=False),
             )
         )

@@ -28,9 +28,19 @@ class EvolvableResNet(EvolvableModule):
         max_blocks: int = 6,
         min_channel_size: int = 16,
         max_channel_size: int = 256,
+        kernel_size: int = 3,
+        stride_size: int = 1,
+        scale_factor: int = 4,
+        output_activation: Optional[str] = None,
         device: str = "cpu",
+        name: Optional[str] = None,
+        random_seed: Optional[int] = None,
     ):
-        super().__init__(device)
+        super().__init__(device, name=name, random_seed=random_seed)
+        self.kernel_size = int(kernel_size)
+        self.stride_size = int(stride_size)
+        self.scale_factor = int(scale_factor)
+        self.output_activation = output_activation
         self.input_shape = tuple(input_shape)
         self.num_outputs = int(num_outputs)
         self.channel_size = int(channel_size)
@@ -52,6 +62,10 @@ class EvolvableResNet(EvolvableModule):
         layers.append(nn.AdaptiveAvgPool2d((4, 4)))
         layers.append(nn.Flatten())
         layers.append(nn.Linear(self.channel_size * 16, self.num_outputs))
+        if self.output_activation:
+            from .components import get_activation
+
+            layers.append(get_activation(self.output_activation))
         return nn.Sequential(*layers)
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
