@@ -192,3 +192,47 @@ class TestILQLPolicyEvaluator:
         torch.manual_seed(2)
         outb = model.generate(ctx, 8, beta=50.0)
         assert not torch.equal(out0, outb)
+
+
+class TestSeq2SeqBERT:
+    """Reference modules/bert.py:63 end-to-end encoder-decoder surface."""
+
+    def _s2s(self):
+        return EvolvableBERT(encoder_layers=[64, 64], decoder_layers=[64],
+                             src_vocab_size=50, tgt_vocab_size=40, d_model=32,
+                             n_head=4, max_positions=32)
+
+    def test_forward_and_loss(self):
+        m = self._s2s()
+        src = torch.randint(0, 50, (2, 10))
+        tgt = torch.randint(0, 40, (2, 7))
+        assert m(src, tgt).shape == (2, 7, 40)
+        _, loss = m(src, tgt, targets=tgt)
+        loss.backward()
+        assert torch.isfinite(loss)
+
+    def test_decoder_causal_encoder_bidirectional(self):
+        m = self._s2s()
+        m.eval()
+        src = torch.randint(0, 50, (1, 10))
+        tgt = torch.randint(0, 40, (1, 6))
+        t2 = tgt.clone()
+        t2[0, -1] = (t2[0, -1] + 1) % 40
+        a, b = m(src, tgt), m(src, t2)
+        assert torch.allclose(a[0, :4], b[0, :4], atol=1e-5)   # causal decoder
+        s2 = src.clone()
+        s2[0, -1] = (s2[0, -1] + 1) % 50
+        c = m(s2, tgt)
+        assert not torch.allclose(a[0, 0], c[0, 0])  # encoder sees all of src
+
+    def test_layer_mutations_preserve_function_shape(self):
+        m = self._s2s()
+        src = torch.randint(0, 50, (2, 8))
+        tgt = torch.randint(0, 40, (2, 5))
+        for mut in ("add_encoder_layer", "add_decoder_layer", "add_node",
+                    "remove_encoder_layer"):
+            m.apply_mutation(mut)
+            assert m(src, tgt).shape == (2, 5, 40)
+        c = m.clone()
+        m.eval(), c.eval()
+        torch.testing.assert_close(c(src, tgt), m(src, tgt))
