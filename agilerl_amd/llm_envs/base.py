@@ -92,7 +92,29 @@ class HuggingFaceGym(LLMEnvBase):
         system_prompt=None,
         shard_across_ranks: bool = True,
         seed: int = 42,
+        conversation_template=None,
+        data_batch_size_per_gpu: int = None,
+        max_context_length: int = None,
+        min_completion_length: int = None,
+        accelerator=None,
     ):
+        # reference llm_envs/base.py:103-106 spellings: per-GPU batch is the
+        # same thing here (one process per GPU); conversation_template is a
+        # list of {role, content} messages prefixed before each prompt;
+        # max_context_length caps the tokenized prompt
+        if data_batch_size_per_gpu is not None:
+            data_batch_size = data_batch_size_per_gpu
+        if max_context_length is not None:
+            max_prompt_tokens = int(max_context_length)
+        self.conversation_template = conversation_template
+        self.min_completion_length = min_completion_length
+        if accelerator is not None:
+            import warnings
+
+            warnings.warn(
+                "HuggingFaceGym ignores `accelerator`: rank sharding reads "
+                "torch.distributed directly.", RuntimeWarning,
+            )
         self.tokenizer = tokenizer
         self.reward_fn = reward_fn
         self.prompt_key = prompt_key

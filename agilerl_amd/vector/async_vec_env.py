@@ -103,7 +103,8 @@ def _pz_worker(idx, env_fn, pipe, shms, obs_shapes, error_queue):
 # ---------------------------------------------------------------------------
 
 class AsyncVectorEnv:
-    def __init__(self, env_fns: List[Callable], context: str = "spawn"):
+    def __init__(self, env_fns: List[Callable], context: str = "spawn", copy: bool = True):
+        self.copy = bool(copy)
         self.num_envs = len(env_fns)
         ctx = mp.get_context(context)
         probe = env_fns[0]()
@@ -187,7 +188,8 @@ class AsyncVectorEnv:
 # ---------------------------------------------------------------------------
 
 class AsyncPettingZooVecEnv:
-    def __init__(self, env_fns: List[Callable], context: str = "spawn"):
+    def __init__(self, env_fns: List[Callable], context: str = "spawn", copy: bool = True):
+        self.copy = bool(copy)
         self.num_envs = len(env_fns)
         ctx = mp.get_context(context)
         probe = env_fns[0]()
