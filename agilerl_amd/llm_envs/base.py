@@ -199,12 +199,24 @@ class HuggingFaceGym(LLMEnvBase):
     def _render(self, prompt: str) -> str:
         if self.apply_chat_template and getattr(self.tokenizer, "chat_template", None):
             messages = []
-            if self.system_prompt:
-                messages.append({"role": "system", "content": self.system_prompt})
+            if self.conversation_template:
+                # reference base.py:37: few-shot / instruction turns prefixed
+                # before each prompt
+                messages.extend(dict(m) for m in self.conversation_template)
+            if self.system_prompt and not any(
+                m.get("role") == "system" for m in messages
+            ):
+                messages.insert(0, {"role": "system", "content": self.system_prompt})
             messages.append({"role": "user", "content": prompt})
             return self.tokenizer.apply_chat_template(
                 messages, tokenize=False, add_generation_prompt=True
             )
+        if self.conversation_template:
+            prefix = "\n".join(
+                f"{m.get('role', 'user').capitalize()}: {m.get('content', '')}"
+                for m in self.conversation_template
+            )
+            return f"{prefix}\nUser: {prompt}"
         return prompt
 
     def score(self, sequences: torch.Tensor) -> np.ndarray:
