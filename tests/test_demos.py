@@ -11,6 +11,7 @@ REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
 DEMOS = [
     ("demos/single_agent/demo_off_policy.py", ["--max-steps", "2000"]),
     ("demos/single_agent/demo_on_policy.py", ["--max-steps", "4000"]),
+    ("demos/single_agent/demo_curriculum.py", ["--max-steps", "1500"]),
     ("demos/multi_agent/demo_multi_agent.py", ["--max-steps", "2000"]),
     ("demos/bandits/demo_bandit.py", ["--max-steps", "400"]),
     ("demos/llm/demo_llm_finetuning.py", ["--iterations", "2"]),
