@@ -27,9 +27,19 @@ class OptimizerWrapper:
         network_names: Optional[List[str]] = None,
         lr_name: str = "lr",
         multiagent: bool = False,
-        **optimizer_kwargs: Any,
+        lr_critic: Optional[float] = None,
+        is_llm_optimizer: bool = False,
+        optimizer_kwargs: Optional[Dict[str, Any]] = None,
+        **extra_kwargs: Any,
     ):
         self.optimizer_cls = optimizer_cls or torch.optim.Adam
+        # reference optimizer_wrapper.py surface: optimizer_kwargs may come
+        # as a dict argument (Accelerate-era calling convention) or as
+        # plain **kwargs; lr_critic tags dual-lr wrappers, is_llm_optimizer
+        # marks adapter-only parameter collection (both informational here)
+        optimizer_kwargs = {**(optimizer_kwargs or {}), **extra_kwargs}
+        self.lr_critic = lr_critic
+        self.is_llm_optimizer = bool(is_llm_optimizer)
         self.lr = lr
         self.network_names = network_names or []
         self.lr_name = lr_name
