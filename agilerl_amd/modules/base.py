@@ -95,13 +95,19 @@ class ModuleMeta(type):
                     try:
                         bound = __sig.bind(self, *args, **kwargs)
                         bound.apply_defaults()
+                        var_kw = {
+                            p.name for p in __sig.parameters.values()
+                            if p.kind is inspect.Parameter.VAR_KEYWORD
+                        }
                         captured = {
                             k: v
                             for k, v in list(bound.arguments.items())[1:]
-                            if k not in ("args", "kwargs")
+                            if k != "args" and k not in var_kw
                         }
-                        if "kwargs" in bound.arguments:
-                            captured.update(bound.arguments["kwargs"])
+                        # flatten the var-keyword dict (whatever it's named)
+                        # so clones re-pass its contents as real kwargs
+                        for k in var_kw:
+                            captured.update(bound.arguments.get(k, {}))
                         object.__setattr__(self, "_init_args", captured)
                     except TypeError:
                         object.__setattr__(self, "_init_args", {})

@@ -29,6 +29,7 @@ class DeterministicActor(EvolvableNetwork):
         head_config: Optional[Dict[str, Any]] = None,
         latent_dim: int = 64,
         device: str = "cpu",
+        **net_kwargs,
     ):
         self.action_space = action_space
         head_config = dict(head_config or {"hidden_size": [64]})
@@ -47,6 +48,7 @@ class DeterministicActor(EvolvableNetwork):
             head_config=head_config,
             latent_dim=latent_dim,
             device=device,
+            **net_kwargs,
         )
         if isinstance(action_space, Box):
             low = torch.as_tensor(action_space.low, dtype=torch.float32)
@@ -83,9 +85,14 @@ class StochasticActor(EvolvableNetwork):
         head_config: Optional[Dict[str, Any]] = None,
         latent_dim: int = 64,
         log_std_init: float = 0.0,
+        action_std_init: Optional[float] = None,
         squash_output: bool = False,
         device: str = "cpu",
+        **net_kwargs,
     ):
+        if action_std_init is not None:
+            # reference actors.py: initial LOG std despite the name
+            log_std_init = float(action_std_init)
         self.action_space = action_space
         dist = ActionDistribution(action_space, log_std_init=log_std_init, squash=squash_output)
         super().__init__(
@@ -95,6 +102,7 @@ class StochasticActor(EvolvableNetwork):
             head_config=head_config,
             latent_dim=latent_dim,
             device=device,
+            **net_kwargs,
         )
         self.dist_layer = dist.to(device)
 

@@ -169,16 +169,49 @@ class EvolvableNetwork(EvolvableModule):
         encoder_config: Optional[Dict[str, Any]] = None,
         head_config: Optional[Dict[str, Any]] = None,
         latent_dim: int = 64,
+        min_latent_dim: Optional[int] = None,
+        max_latent_dim: Optional[int] = None,
+        recurrent: bool = False,
+        simba: bool = False,
+        encoder_cls=None,
+        encoder=None,
+        encoder_name: Optional[str] = None,
         device: str = "cpu",
+        name: Optional[str] = None,
+        random_seed: Optional[int] = None,
     ):
-        super().__init__(device)
+        super().__init__(device, name=name, random_seed=random_seed)
         self.observation_space = observation_space
         self.num_outputs = int(num_outputs)
         self.latent_dim = int(latent_dim)
-        self.encoder_config = dict(encoder_config) if encoder_config else None
+        # reference networks/base.py surface: per-instance latent mutation
+        # bounds; simba/recurrent pick the encoder family when no explicit
+        # arch is configured; encoder/encoder_cls inject a prebuilt or
+        # custom-class encoder (encoder objects enable explicit sharing)
+        self.MIN_LATENT = int(min_latent_dim) if min_latent_dim is not None else type(self).MIN_LATENT
+        self.MAX_LATENT = int(max_latent_dim) if max_latent_dim is not None else type(self).MAX_LATENT
+        cfg = dict(encoder_config) if encoder_config else None
+        if cfg is None or "arch" not in cfg:
+            if simba:
+                cfg = dict(cfg or {})
+                cfg.setdefault("arch", "simba")
+            elif recurrent:
+                cfg = dict(cfg or {})
+                cfg.setdefault("arch", "lstm")
+        self.encoder_config = cfg
         self.head_config = dict(head_config) if head_config else None
+        self.encoder_name = encoder_name or "encoder"
 
-        self.encoder = build_encoder(observation_space, self.latent_dim, self.encoder_config, device)
+        if encoder is not None:
+            self.encoder = encoder.to(device) if hasattr(encoder, "to") else encoder
+        elif encoder_cls is not None:
+            self.encoder = encoder_cls(
+                observation_space, self.latent_dim, **(self.encoder_config or {})
+            ).to(device)
+        else:
+            self.encoder = build_encoder(
+                observation_space, self.latent_dim, self.encoder_config, device
+            )
         self.head_net = self._build_head()
 
     # ------------------------------------------------------------------

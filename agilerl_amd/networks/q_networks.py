@@ -31,6 +31,7 @@ class QNetwork(EvolvableNetwork):
         head_config: Optional[Dict[str, Any]] = None,
         latent_dim: int = 64,
         device: str = "cpu",
+        **net_kwargs,
     ):
         if not hasattr(action_space, "n"):
             raise TypeError(
@@ -45,6 +46,7 @@ class QNetwork(EvolvableNetwork):
             head_config=head_config,
             latent_dim=latent_dim,
             device=device,
+            **net_kwargs,
         )
 
 
@@ -68,7 +70,9 @@ class RainbowQNetwork(EvolvableNetwork):
         v_min: float = -10.0,
         v_max: float = 10.0,
         noise_std: float = 0.5,
+        support=None,
         device: str = "cpu",
+        **net_kwargs,
     ):
         self.action_space = action_space
         self.num_atoms = int(num_atoms)
@@ -86,9 +90,13 @@ class RainbowQNetwork(EvolvableNetwork):
             head_config=head_config,
             latent_dim=latent_dim,
             device=device,
+            **net_kwargs,
         )
         self.register_buffer(
-            "support", torch.linspace(self.v_min, self.v_max, self.num_atoms, device=device)
+            "support",
+            torch.as_tensor(support, dtype=torch.float32, device=device)
+            if support is not None
+            else torch.linspace(self.v_min, self.v_max, self.num_atoms, device=device)
         )
 
     def _build_head(self):
@@ -148,6 +156,7 @@ class ContinuousQNetwork(EvolvableNetwork):
         latent_dim: int = 64,
         action_dim: Optional[int] = None,
         device: str = "cpu",
+        **net_kwargs,
     ):
         self.action_space = action_space
         self.action_dim = int(action_dim) if action_dim is not None else flatdim(action_space)
@@ -158,6 +167,7 @@ class ContinuousQNetwork(EvolvableNetwork):
             head_config=head_config,
             latent_dim=latent_dim,
             device=device,
+            **net_kwargs,
         )
 
     def _build_head(self):

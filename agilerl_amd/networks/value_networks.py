@@ -21,6 +21,7 @@ class ValueNetwork(EvolvableNetwork):
         head_config: Optional[Dict[str, Any]] = None,
         latent_dim: int = 64,
         device: str = "cpu",
+        **net_kwargs,
     ):
         super().__init__(
             observation_space,
@@ -29,4 +30,5 @@ class ValueNetwork(EvolvableNetwork):
             head_config=head_config,
             latent_dim=latent_dim,
             device=device,
+            **net_kwargs,
         )

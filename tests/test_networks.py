@@ -111,3 +111,47 @@ def test_dict_obs_network():
     q = QNetwork(space, ACT)
     obs = {"vec": torch.randn(5, 4), "vec2": torch.randn(5, 6)}
     assert q(obs).shape == (5, 4)
+
+
+class TestReferenceNetworkSurface:
+    """Reference networks/base.py kwargs: latent bounds, encoder family
+    selectors, prebuilt-encoder injection, Rainbow explicit support."""
+
+    def test_simba_recurrent_selectors_and_bounds(self):
+        import torch
+
+        from agilerl_amd.networks import QNetwork
+        from agilerl_amd.networks.value_networks import ValueNetwork
+        from agilerl_amd.spaces import Box, Discrete
+
+        q = QNetwork(Box(-1, 1, (4,)), Discrete(2), simba=True,
+                     min_latent_dim=16, max_latent_dim=32, random_seed=3)
+        assert type(q.encoder).__name__ == "EvolvableSimBa"
+        assert q.MIN_LATENT == 16 and q.MAX_LATENT == 32
+        for _ in range(8):
+            q.apply_mutation("add_latent_node", numb_new_nodes=16)
+        assert q.latent_dim <= 32  # bounded by max_latent_dim
+        v = ValueNetwork(Box(-1, 1, (4,)), recurrent=True)
+        assert type(v.encoder).__name__ == "EvolvableLSTM"
+        c = q.clone()
+        x = torch.randn(2, 4)
+        assert torch.allclose(c(x), q(x))
+
+    def test_encoder_injection_shares_object(self):
+        from agilerl_amd.networks.actors import DeterministicActor
+        from agilerl_amd.networks.value_networks import ValueNetwork
+        from agilerl_amd.spaces import Box
+
+        actor = DeterministicActor(Box(-1, 1, (4,)), Box(-1, 1, (2,)))
+        value = ValueNetwork(Box(-1, 1, (4,)), encoder=actor.encoder)
+        assert value.encoder is actor.encoder
+
+    def test_rainbow_explicit_support(self):
+        import torch
+
+        from agilerl_amd.networks import RainbowQNetwork
+        from agilerl_amd.spaces import Box, Discrete
+
+        sup = torch.linspace(-5, 5, 51)
+        r = RainbowQNetwork(Box(-1, 1, (4,)), Discrete(2), support=sup)
+        assert torch.allclose(r.support, sup)
