@@ -59,14 +59,51 @@ def make_vect_envs(
     env_id: Optional[Union[str, Callable]] = None,
     num_envs: int = 1,
     seed: Optional[int] = None,
+    make_env: Optional[Callable] = None,
+    should_async_vector: bool = False,
+    extra_wrappers=None,
+    env_name: Optional[str] = None,
     **env_kwargs,
 ) -> VecEnv:
-    """Create a natively-batched vectorized env."""
+    """Create a vectorized env.
+
+    Default path: the first-party *natively batched* registry envs (one
+    tensor op steps all N copies — no subprocess fan-out needed).
+    Reference-compat paths (utils.py:222): ``make_env`` vectorizes a
+    user per-env factory via subprocess workers when
+    ``should_async_vector`` (shared-memory AsyncVectorEnv) or a simple
+    serial loop otherwise; ``extra_wrappers`` wrap each instance.
+    """
+    if env_name is not None and env_id is None:
+        env_id = env_name  # reference spelling
+    if make_env is not None:
+        factory = make_env
+        if extra_wrappers:
+            inner = factory
+
+            def factory():
+                e = inner()
+                for wrapper_cls in extra_wrappers:
+                    e = wrapper_cls(e)
+                return e
+
+        from ..vector.async_vec_env import AsyncVectorEnv
+
+        if should_async_vector:
+            return AsyncVectorEnv([factory for _ in range(num_envs)])
+        from ..vector.sync_vec_env import SyncVectorEnv
+
+        return SyncVectorEnv([factory for _ in range(num_envs)])
     if callable(env_id):
-        return env_id(num_envs=num_envs, seed=seed, **env_kwargs)
-    if env_id not in ENV_REGISTRY:
-        raise KeyError(
-            f"Unknown env id '{env_id}'. Registered: {sorted(ENV_REGISTRY)}. "
-            "Use register_env() for custom environments."
-        )
-    return ENV_REGISTRY[env_id](num_envs=num_envs, seed=seed, **env_kwargs)
+        env = env_id(num_envs=num_envs, seed=seed, **env_kwargs)
+    else:
+        if env_id not in ENV_REGISTRY:
+            raise KeyError(
+                f"Unknown env id '{env_id}'. Registered: {sorted(ENV_REGISTRY)}. "
+                "Use register_env() for custom environments."
+            )
+        env = ENV_REGISTRY[env_id](num_envs=num_envs, seed=seed, **env_kwargs)
+    if extra_wrappers:
+        for wrapper_cls in extra_wrappers:
+            env = wrapper_cls(env)
+    return env

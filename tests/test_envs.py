@@ -193,3 +193,61 @@ class TestLunarLanderContinuous:
         env.state[:, 5] = 0.0
         env.step(np.tile([0.0, 0.3], (2, 1)))  # below the 0.5 threshold
         assert np.allclose(env.state[:, 5], 0.0, atol=1e-9)
+
+
+class TestMakeVectEnvsReferencePaths:
+    """Reference utils.py:222 make_vect_envs surface: per-env factories
+    (sync/async vectorization) and extra_wrappers."""
+
+    class TinyEnv:
+        class _Obs:
+            shape = (2,)
+
+        class _Act:
+            n = 2
+            shape = ()
+
+            def sample(self):
+                return 0
+
+        observation_space = _Obs()
+        action_space = _Act()
+
+        def __init__(self):
+            self.t = 0
+
+        def reset(self, seed=None):
+            self.t = 0
+            return np.zeros(2, np.float32), {}
+
+        def step(self, a):
+            self.t += 1
+            done = self.t >= 5
+            return np.full(2, self.t, np.float32), 1.0, done, False, {}
+
+    def test_sync_factory_autoreset(self):
+        from agilerl_amd.envs.registry import make_vect_envs
+
+        env = make_vect_envs(make_env=self.TinyEnv, num_envs=3,
+                             should_async_vector=False)
+        obs, _ = env.reset(seed=0)
+        assert obs.shape == (3, 2)
+        for _ in range(5):
+            obs, r, te, tr, info = env.step(np.zeros(3, np.int64))
+        assert te.all()
+        assert (info["final_observation"][:, 0] == 5).all()
+        assert (obs[:, 0] == 0).all()  # autoreset happened
+
+    def test_extra_wrappers_on_batched_env(self):
+        from agilerl_amd.envs.registry import make_vect_envs
+        from agilerl_amd.wrappers.learning import Skill
+
+        class Doubler(Skill):
+            def skill_reward(self, obs, reward, terminated, truncated, info):
+                return reward * 2, terminated, truncated
+
+        env = make_vect_envs("CartPole-v1", num_envs=2, seed=0,
+                             extra_wrappers=[Doubler])
+        env.reset()
+        _, reward, _, _, _ = env.step(np.zeros(2, dtype=np.int64))
+        assert (reward == 2.0).all()
