@@ -315,3 +315,7 @@ def check_llm_policy_with_probe_env(agent, env, iterations: int = 15):
     last = _np.mean(rewards_hist[-5:])
     assert last > first, f"reward did not improve: {rewards_hist}"
     return rewards_hist
+
+
+# reference utils/probe_envs.py spelling
+check_policy_on_policy_with_probe_env = check_on_policy_with_probe_env

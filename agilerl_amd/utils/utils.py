@@ -28,6 +28,8 @@ __all__ = [
     "make_skill_vect_envs",
     "calculate_vectorized_scores",
     "print_hyperparams",
+    "init_wandb",
+    "init_loggers",
 ]
 
 
@@ -208,3 +210,27 @@ def print_hyperparams(pop) -> None:
               f"Last mutation: {getattr(agent, 'mut', 'None')}")
         for k in sorted(hps):
             print(f"  {k}: {hps[k]}")
+
+
+def init_wandb(project: str = "agilerl-amd", config=None, **kwargs):
+    """Start a wandb run if wandb is importable (reference utils.py:1428).
+    Returns the run or None (offline images have no wandb)."""
+    try:
+        import wandb
+    except ImportError:
+        import warnings
+
+        warnings.warn("wandb is not installed; init_wandb is a no-op")
+        return None
+    return wandb.init(project=project, config=config, **kwargs)
+
+
+def init_loggers(stdout: bool = True, csv_path=None, tensorboard_dir=None,
+                 wandb_project=None, prometheus_port=None):
+    """Build the logger sinks (reference utils.py:1499 init_loggers)."""
+    from ..logger import make_loggers
+
+    return make_loggers(stdout=stdout, csv_path=csv_path,
+                        tensorboard_dir=tensorboard_dir,
+                        wandb_project=wandb_project,
+                        prometheus_port=prometheus_port)
