@@ -17,7 +17,7 @@ import torch
 
 from ..components.rollout_buffer import RolloutBuffer
 
-__all__ = ["collect_rollouts", "collect_rollouts_device", "collect_rollouts_recurrent"]
+__all__ = ["collect_rollouts", "collect_rollouts_device", "collect_rollouts_recurrent", "collect_rollouts_llm"]
 
 
 def collect_rollouts(
@@ -155,3 +155,29 @@ def collect_rollouts_recurrent(
     )
     stats = {"mean_episode_return": float(np.mean(ep_returns))} if ep_returns else {}
     return obs, done, hidden, stats
+
+
+def collect_rollouts_llm(agent, env, n_batches: int = 1):
+    """One or more LLM prompt-batch rollouts (reference
+    rollouts/on_policy.py:265 collect_rollouts_llm): reset -> generate ->
+    score -> experiences dict ready for ``agent.learn``.
+
+    Returns (experiences, mean_reward) for the LAST batch when
+    ``n_batches == 1`` (the common case), else a list of experience
+    dicts.  Sampling logprobs/turn metadata captured by the paged engine
+    or multiturn envs ride inside the experiences.
+    """
+    from ..llm_envs.base import make_grpo_experiences
+
+    pad_id = getattr(agent.model.config, "pad_token_id", None) or 0
+    out = []
+    for _ in range(max(int(n_batches), 1)):
+        prompts = env.reset()
+        sequences = agent.get_action(prompts, training=True)
+        rewards = env.score(sequences)
+        experiences = make_grpo_experiences(env, sequences, rewards, pad_token_id=pad_id)
+        sampling = getattr(agent, "last_sampling_logps", None)
+        if sampling is not None and "sampling_logps" not in experiences:
+            experiences["sampling_logps"] = sampling
+        out.append((experiences, float(np.mean(rewards))))
+    return out[0] if len(out) == 1 else out

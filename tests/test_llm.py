@@ -1359,3 +1359,27 @@ class TestPpoLlmTurnGae(TestGrpoParityDeepening):
         assert a.importance_sampling_level == "turn"
         assert a.lr == 1e-5
         assert a.value_optimizer.param_groups[0]["lr"] == 1e-3
+
+
+class TestCollectRolloutsLlm(TestGrpoParityDeepening):
+    def test_single_batch(self):
+        from agilerl_amd.llm_envs import TokenReasoningGym
+        from agilerl_amd.rollouts.on_policy import collect_rollouts_llm
+
+        agent = self._tiny_agent(max_completion_tokens=8)
+        env = TokenReasoningGym(vocab_size=64, prompt_len=8,
+                                data_batch_size=2, group_size=2)
+        exp, reward = collect_rollouts_llm(agent, env)
+        assert {"ids", "action_mask", "rewards"} <= set(exp)
+        stats = agent.learn(exp)
+        assert np.isfinite(stats["loss"]) and np.isfinite(reward)
+
+    def test_multi_batch(self):
+        from agilerl_amd.llm_envs import TokenReasoningGym
+        from agilerl_amd.rollouts.on_policy import collect_rollouts_llm
+
+        agent = self._tiny_agent(max_completion_tokens=8)
+        env = TokenReasoningGym(vocab_size=64, prompt_len=8,
+                                data_batch_size=2, group_size=2)
+        batches = collect_rollouts_llm(agent, env, n_batches=2)
+        assert len(batches) == 2
