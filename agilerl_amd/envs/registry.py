@@ -14,6 +14,7 @@ from .cartpole import CartPoleVecEnv
 from .lunar_lander import LunarLanderContinuousVecEnv, LunarLanderVecEnv
 from .classic_control import AcrobotVecEnv, MountainCarContinuousVecEnv, MountainCarVecEnv
 from .pendulum import PendulumVecEnv
+from .toy_text import BlackjackVecEnv
 from .visual import BreakoutLiteVecEnv, CatchPongVecEnv
 from .probe import (
     ConstantRewardEnv,
@@ -41,6 +42,11 @@ ENV_REGISTRY: Dict[str, Callable[..., VecEnv]] = {
     "CatchPong-v0": CatchPongVecEnv,
     "BreakoutLite-v0": BreakoutLiteVecEnv,
     "PongLike-v0": CatchPongVecEnv,
+    "Blackjack-v1": BlackjackVecEnv,
+    # ALE ROMs are unavailable offline; the first-party BreakoutLite env
+    # stands in with the same visual-obs pipeline shape
+    "ALE/Breakout-v5": BreakoutLiteVecEnv,
+    "ALE/Pong-v5": CatchPongVecEnv,
     "probe/ConstantReward": ConstantRewardEnv,
     "probe/ObsDependentReward": ObsDependentRewardEnv,
     "probe/DiscountedReward": DiscountedRewardEnv,

@@ -114,18 +114,28 @@ def preprocess_observation(obs, space: Space, device) -> Any:
     uint8 /255, dict/tuple leaf conversion).  Shared by EvolvableNetwork and
     CustomNetworkAdapter."""
     if isinstance(space, (DictSpace, TupleSpace)):
-        dev = device if isinstance(device, torch.device) else torch.device(device)
-
-        def _leaf(x):
-            if not isinstance(x, torch.Tensor):
-                x = torch.as_tensor(np.asarray(x))
-            x = x.to(dev)
-            return x.float() if not x.is_floating_point() else x
-
-        if isinstance(obs, dict):
-            return {k: _leaf(v) for k, v in obs.items()}
-        if isinstance(obs, (tuple, list)):
-            return type(obs)(_leaf(v) for v in obs)
+        # recurse per sub-space so Discrete members one-hot to the width
+        # their sub-encoder was built for
+        if isinstance(space, DictSpace):
+            subspaces = space.spaces
+            if isinstance(obs, dict):
+                return {
+                    k: preprocess_observation(v, subspaces[k], device)
+                    if k in subspaces else v
+                    for k, v in obs.items()
+                }
+        else:
+            subs = list(space.spaces)
+            if isinstance(obs, dict):  # index-keyed storage of tuple obs
+                return {
+                    k: preprocess_observation(v, subs[int(k)], device)
+                    for k, v in obs.items()
+                }
+            if isinstance(obs, (tuple, list)):
+                return tuple(
+                    preprocess_observation(v, sp, device)
+                    for v, sp in zip(obs, subs)
+                )
         return obs
     if not isinstance(obs, torch.Tensor):
         obs = torch.as_tensor(np.asarray(obs))

@@ -34,6 +34,8 @@ PZ_REGISTRY = {
     "simple_speaker_listener": "agilerl_amd.envs.mpe.SpeakerListenerVecEnv",
     "simple_spread_v3": "agilerl_amd.envs.mpe.SimpleSpreadVecEnv",
     "simple_spread": "agilerl_amd.envs.mpe.SimpleSpreadVecEnv",
+    "cooperative_pong_v6": "agilerl_amd.envs.ma_pong.CooperativePongVecEnv",
+    "cooperative_pong": "agilerl_amd.envs.ma_pong.CooperativePongVecEnv",
 }
 
 

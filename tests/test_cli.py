@@ -185,6 +185,10 @@ class TestReferenceConfigEndToEnd:
         "dqn/dqn.yaml", "ppo/ppo.yaml", "dqn/dqn_rainbow.yaml",
         "td3.yaml", "ddpg/ddpg.yaml", "multi_agent/maddpg.yaml",
         "cqn.yaml", "dqn/dqn_mfpbt.yaml", "ppo/ppo_recurrent.yaml",
+        "multi_agent/matd3.yaml", "multi_agent/ippo.yaml",
+        "multi_agent/maddpg_mfpbt.yaml", "multi_agent/ippo_pong.yaml",
+        "ppo/ppo_mfpbt.yaml", "cqn_mfpbt.yaml",
+        "multi_input.yaml", "ppo/ppo_image.yaml",
     ])
     def test_reference_yaml_trains(self, rel):
         import yaml as _yaml
