@@ -126,15 +126,18 @@ def run_llm_workload(trainer, workload: str):
         )
         loop = finetune_llm_sft
     elif workload == "llm_preference":
-        from ...llm_envs.preference import SyntheticPreferenceGym
+        if getattr(spec, "dataset", None):
+            env = _dataset_preference_gym(spec, pop[0], env_kwargs)
+        else:
+            from ...llm_envs.preference import SyntheticPreferenceGym
 
-        env = SyntheticPreferenceGym(
-            vocab_size=vocab_size,
-            prompt_len=spec.max_prompt_tokens,
-            completion_len=spec.max_completion_tokens,
-            data_batch_size=spec.data_batch_size,
-            **env_kwargs,
-        )
+            env = SyntheticPreferenceGym(
+                vocab_size=vocab_size,
+                prompt_len=spec.max_prompt_tokens,
+                completion_len=spec.max_completion_tokens,
+                data_batch_size=spec.data_batch_size,
+                **env_kwargs,
+            )
         loop = finetune_llm_preference
     else:
         raise NotImplementedError(workload)
@@ -153,6 +156,62 @@ def run_llm_workload(trainer, workload: str):
         loggers=trainer.loggers,
         max_wall_seconds=t.max_wall_seconds,
     )
+
+
+def _dataset_preference_gym(spec, agent, env_kwargs):
+    """Build a PreferenceGym from a manifest `dataset:` (saved-to-disk)
+    with (prompt, chosen, rejected) columns — DPO-dataset aliases
+    question/instruction -> prompt accepted."""
+    import os
+
+    from ...llm_envs.preference import PreferenceGym
+
+    path = spec.dataset
+    if not os.path.exists(path):
+        import warnings
+
+        warnings.warn(
+            f"preference dataset '{path}' not found locally (no hub access); "
+            "falling back to the synthetic preference gym",
+            RuntimeWarning,
+        )
+        from ...llm_envs.preference import SyntheticPreferenceGym
+
+        vocab = agent.model.config.vocab_size
+        return SyntheticPreferenceGym(
+            vocab_size=vocab, prompt_len=spec.max_prompt_tokens,
+            completion_len=spec.max_completion_tokens,
+            data_batch_size=spec.data_batch_size,
+        )
+    import datasets as hf_datasets
+
+    ds = hf_datasets.load_from_disk(path)
+    if hasattr(ds, "keys") and "train" in ds:
+        ds = ds["train"]
+    cols = {c.lower(): c for c in ds.column_names}
+    prompt_col = next((cols[c] for c in ("prompt", "question", "instruction")
+                       if c in cols), None)
+    if prompt_col is None or "chosen" not in cols or "rejected" not in cols:
+        raise KeyError(
+            f"preference dataset needs prompt/chosen/rejected columns, got {ds.column_names}"
+        )
+    triples = list(zip(ds[prompt_col], ds[cols["chosen"]], ds[cols["rejected"]]))
+    tok_path = (env_kwargs or {}).pop("tokenizer_path", None)
+    if tok_path:
+        from transformers import AutoTokenizer
+
+        tokenizer = AutoTokenizer.from_pretrained(tok_path)
+    else:
+        tokenizer = getattr(agent, "tokenizer", None)
+    if tokenizer is None:
+        raise ValueError(
+            "a dataset-backed preference env needs a tokenizer: pass "
+            "`tokenizer_path` in environment.env_kwargs or give the "
+            "algorithm a tokenizer"
+        )
+    return PreferenceGym(triples, tokenizer,
+                         data_batch_size=spec.data_batch_size,
+                         max_tokens=spec.max_prompt_tokens + spec.max_completion_tokens)
 
 
 def _dataset_reasoning_gym(spec, agent, reward_fn, env_kwargs):

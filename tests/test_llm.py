@@ -1077,10 +1077,14 @@ class TestPpoLlmSamplingIs:
 
 
 class TestReferenceLlmManifestFields:
-    def test_reference_grpo_yaml_constructs_with_local_overrides(self, tmp_path):
-        """The actual reference grpo.yaml (vLLM-era fields, columns mapping,
-        reward_file_path, train_test_split) runs here with only a local
-        model + dataset substituted for the hub entries."""
+    @pytest.mark.parametrize("cfg", [
+        "grpo.yaml", "gspo.yaml", "cispo.yaml", "ppo_llm.yaml",
+        "reinforce_llm.yaml",
+    ])
+    def test_reference_llm_yaml_constructs_with_local_overrides(self, tmp_path, cfg):
+        """The actual reference LLM-finetuning YAMLs (vLLM-era fields,
+        columns mapping, reward_file_path, train_test_split) run here with
+        only a local model + dataset substituted for the hub entries."""
         import os
         import warnings
 
@@ -1090,7 +1094,7 @@ class TestReferenceLlmManifestFields:
         from agilerl_amd.models.manifest import TrainingManifest
         from agilerl_amd.training.trainer import LocalTrainer
 
-        ref = "/root/reference/configs/training/llm_finetuning/grpo.yaml"
+        ref = f"/root/reference/configs/training/llm_finetuning/{cfg}"
         if not os.path.exists(ref):
             pytest.skip("reference configs absent")
         doc = _yaml.safe_load(open(ref))
@@ -1138,8 +1142,11 @@ class TestReferenceLlmManifestFields:
             trainer = LocalTrainer(manifest, device="cpu")
             results = trainer.train()
         assert results is not None
-        dropped = [str(x.message) for x in w if "reference-only" in str(x.message)]
-        assert dropped and "use_vllm" in dropped[0]
+        if cfg == "grpo.yaml":
+            dropped = [str(x.message) for x in w
+                       if "reference-only" in str(x.message)
+                       or "paged-KV" in str(x.message)]
+            assert dropped
 
 
 class TestReferenceMultiturnManifest:
@@ -1383,3 +1390,58 @@ class TestCollectRolloutsLlm(TestGrpoParityDeepening):
                                 data_batch_size=2, group_size=2)
         batches = collect_rollouts_llm(agent, env, n_batches=2)
         assert len(batches) == 2
+
+
+class TestReferenceDpoManifest:
+    def test_reference_dpo_yaml_with_local_preference_dataset(self, tmp_path):
+        """The reference dpo.yaml trains against a local saved-to-disk
+        preference dataset (prompt/chosen/rejected columns)."""
+        import os
+
+        import yaml as _yaml
+        from datasets import Dataset
+
+        from agilerl_amd.models.manifest import TrainingManifest
+        from agilerl_amd.training.trainer import LocalTrainer
+
+        ref = "/root/reference/configs/training/llm_finetuning/dpo.yaml"
+        if not os.path.exists(ref):
+            pytest.skip("reference configs absent")
+        doc = _yaml.safe_load(open(ref))
+
+        texts = [f"question {i}" for i in range(6)]
+        ds = Dataset.from_dict({
+            "prompt": texts,
+            "chosen": [f"good answer {i}" for i in range(6)],
+            "rejected": [f"bad {i}" for i in range(6)],
+        })
+        ds_dir = tmp_path / "prefds"
+        ds.save_to_disk(str(ds_dir))
+        tok = _word_tokenizer(texts + [f"good answer {i}" for i in range(6)]
+                              + [f"bad {i}" for i in range(6)])
+        tok_dir = tmp_path / "tok"
+        tok.save_pretrained(str(tok_dir))
+
+        doc["environment"]["dataset"] = str(ds_dir)
+        doc["environment"].setdefault("env_kwargs", {})["tokenizer_path"] = str(tok_dir)
+        doc["environment"]["data_batch_size"] = 2
+        doc["algorithm"]["model_config"] = {
+            "model_type": "llama", "vocab_size": tok.vocab_size + 8,
+            "hidden_size": 32, "intermediate_size": 64,
+            "num_hidden_layers": 1, "num_attention_heads": 2,
+            "num_key_value_heads": 1, "max_position_embeddings": 128,
+            "pad_token_id": tok.pad_token_id,
+        }
+        doc["algorithm"]["lora_config"] = {"r": 2}
+        doc["algorithm"]["dtype"] = "float32"
+        doc["algorithm"]["micro_batch_size"] = 2
+        doc["training"].update({"max_steps": 2, "pop_size": 2, "evo_steps": 1})
+
+        algo = doc["algorithm"]
+        hp = {k: v for k, v in algo.items() if k != "name"}
+        doc["algorithm"] = {"name": algo["name"], "hyperparameters": hp}
+
+        manifest = TrainingManifest.model_validate(doc)
+        trainer = LocalTrainer(manifest, device="cpu")
+        results = trainer.train()
+        assert results is not None
