@@ -115,15 +115,18 @@ def run_llm_workload(trainer, workload: str):
             )
         loop = finetune_llm_reasoning
     elif workload == "llm_sft":
-        from ...llm_envs.sft import SyntheticSFTGym
+        if getattr(spec, "dataset", None):
+            env = _dataset_sft_gym(spec, pop[0], env_kwargs)
+        else:
+            from ...llm_envs.sft import SyntheticSFTGym
 
-        env = SyntheticSFTGym(
-            vocab_size=vocab_size,
-            prompt_len=spec.max_prompt_tokens,
-            completion_len=spec.max_completion_tokens,
-            data_batch_size=spec.data_batch_size,
-            **env_kwargs,
-        )
+            env = SyntheticSFTGym(
+                vocab_size=vocab_size,
+                prompt_len=spec.max_prompt_tokens,
+                completion_len=spec.max_completion_tokens,
+                data_batch_size=spec.data_batch_size,
+                **env_kwargs,
+            )
         loop = finetune_llm_sft
     elif workload == "llm_preference":
         if getattr(spec, "dataset", None):
@@ -156,6 +159,64 @@ def run_llm_workload(trainer, workload: str):
         loggers=trainer.loggers,
         max_wall_seconds=t.max_wall_seconds,
     )
+
+
+def _dataset_sft_gym(spec, agent, env_kwargs):
+    """SFTGym from a manifest `dataset:` (saved-to-disk): prompt column +
+    `response_column` (reference sft.yaml spelling) or completion/answer."""
+    import os
+
+    if not os.path.exists(spec.dataset):
+        import warnings
+
+        warnings.warn(
+            f"SFT dataset '{spec.dataset}' not found locally (no hub "
+            "access); falling back to the synthetic SFT gym",
+            RuntimeWarning,
+        )
+        from ...llm_envs.sft import SyntheticSFTGym
+
+        return SyntheticSFTGym(
+            vocab_size=agent.model.config.vocab_size,
+            prompt_len=spec.max_prompt_tokens,
+            completion_len=spec.max_completion_tokens,
+            data_batch_size=spec.data_batch_size,
+        )
+    import datasets as hf_datasets
+
+    from ...llm_envs.sft import SFTGym
+
+    ds = hf_datasets.load_from_disk(spec.dataset)
+    if hasattr(ds, "keys") and "train" in ds:
+        ds = ds["train"]
+    cols = {c.lower(): c for c in ds.column_names}
+    extra = getattr(spec, "model_extra", None) or {}
+    response_col = (env_kwargs or {}).pop("response_column", None) or         extra.get("response_column")
+    prompt_col = next((cols[c] for c in ("prompt", "question", "instruction")
+                       if c in cols), None)
+    if response_col is None:
+        response_col = next((cols[c] for c in ("completion", "response",
+                                               "answer", "chosen") if c in cols), None)
+    if prompt_col is None or response_col is None:
+        raise KeyError(
+            f"SFT dataset needs prompt + response columns, got {ds.column_names}"
+        )
+    pairs = list(zip(ds[prompt_col], ds[response_col]))
+    tok_path = (env_kwargs or {}).pop("tokenizer_path", None)
+    if tok_path:
+        from transformers import AutoTokenizer
+
+        tokenizer = AutoTokenizer.from_pretrained(tok_path)
+    else:
+        tokenizer = getattr(agent, "tokenizer", None)
+    if tokenizer is None:
+        raise ValueError(
+            "a dataset-backed SFT env needs a tokenizer: pass "
+            "`tokenizer_path` in environment.env_kwargs or give the "
+            "algorithm a tokenizer"
+        )
+    return SFTGym(pairs, tokenizer, data_batch_size=spec.data_batch_size,
+                  max_tokens=spec.max_prompt_tokens + spec.max_completion_tokens)
 
 
 def _dataset_preference_gym(spec, agent, env_kwargs):

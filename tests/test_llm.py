@@ -1445,3 +1445,58 @@ class TestReferenceDpoManifest:
         trainer = LocalTrainer(manifest, device="cpu")
         results = trainer.train()
         assert results is not None
+
+
+class TestReferenceSftManifest:
+    def test_reference_sft_yaml_with_local_dataset(self, tmp_path):
+        """The reference sft.yaml (response_column spelling) trains against
+        a local saved-to-disk dataset."""
+        import os
+
+        import yaml as _yaml
+        from datasets import Dataset
+
+        from agilerl_amd.models.manifest import TrainingManifest
+        from agilerl_amd.training.trainer import LocalTrainer
+
+        ref = "/root/reference/configs/training/llm_finetuning/../sft.yaml"
+        ref = os.path.normpath(ref)
+        if not os.path.exists(ref):
+            pytest.skip("reference configs absent")
+        doc = _yaml.safe_load(open(ref))
+
+        texts = [f"prompt {i}" for i in range(6)]
+        ds = Dataset.from_dict({
+            "prompt": texts,
+            "chosen": [f"target {i}" for i in range(6)],
+        })
+        ds_dir = tmp_path / "sftds"
+        ds.save_to_disk(str(ds_dir))
+        tok = _word_tokenizer(texts + [f"target {i}" for i in range(6)])
+        tok_dir = tmp_path / "tok"
+        tok.save_pretrained(str(tok_dir))
+
+        doc["environment"]["dataset"] = str(ds_dir)
+        doc["environment"].setdefault("env_kwargs", {})["tokenizer_path"] = str(tok_dir)
+        doc["environment"]["data_batch_size"] = 2
+        doc["algorithm"]["model_config"] = {
+            "model_type": "llama", "vocab_size": tok.vocab_size + 8,
+            "hidden_size": 32, "intermediate_size": 64,
+            "num_hidden_layers": 1, "num_attention_heads": 2,
+            "num_key_value_heads": 1, "max_position_embeddings": 128,
+            "pad_token_id": tok.pad_token_id,
+        }
+        doc["algorithm"]["lora_config"] = {"r": 2}
+        doc["algorithm"]["dtype"] = "float32"
+        doc["algorithm"]["micro_batch_size"] = 2
+        doc["algorithm"]["gradient_checkpointing"] = False
+        doc["training"].update({"max_steps": 2, "pop_size": 2, "evo_steps": 1})
+
+        algo = doc["algorithm"]
+        hp = {k: v for k, v in algo.items() if k != "name"}
+        doc["algorithm"] = {"name": algo["name"], "hyperparameters": hp}
+
+        manifest = TrainingManifest.model_validate(doc)
+        trainer = LocalTrainer(manifest, device="cpu")
+        results = trainer.train()
+        assert results is not None
