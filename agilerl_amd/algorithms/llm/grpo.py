@@ -202,10 +202,22 @@ class GRPO(LLMAlgorithm):
         if generation not in ("hf", "paged"):
             raise ValueError(f"generation must be 'hf' or 'paged', got {generation!r}")
         self.generation = generation
+        if isinstance(clip_coef, (list, tuple)):
+            # reference CISPO configs give RATIO bounds `clip_coef: [lo, hi]`
+            # (cispo_quant_bench.yaml); our epsilons are hi-1 / 1-lo
+            lo, hi = (float(x) for x in clip_coef)
+            clip_coef = hi - 1.0
+            if clip_coef_lower is None:
+                clip_coef_lower = 1.0 - lo
         self.clip_coef = float(clip_coef)
         self.clip_coef_lower = float(clip_coef_lower) if clip_coef_lower is not None else None
         self.beta = float(beta)  # k3 KL coefficient
         self.scale_rewards = bool(scale_rewards)
+        if loss_norm == "micro_batch":
+            # reference models/algorithms/grpo.py:45 spelling: each
+            # micro-batch normalized by its own action-token count — our
+            # "token" mode
+            loss_norm = "token"
         if loss_norm not in ("token", "sequence", "accumulation_window"):
             raise ValueError(f"unknown loss_norm {loss_norm!r}")
         self.loss_norm = loss_norm

@@ -20,11 +20,14 @@ __all__ = ["ReinforceLLM"]
 class ReinforceLLM(GRPO):
     CISPO = True  # fused kernel: loss = -w * adv * logp with w = clamp(ratio)
 
-    def __init__(self, *args, **kwargs):
+    def __init__(self, *args, gamma: float = 1.0, **kwargs):
         kwargs.setdefault("update_epochs", 1)
         kwargs.setdefault("clip_coef", 1e6)  # effectively unclipped weight ~ 1
         super().__init__(*args, **kwargs)
         self.algo = "ReinforceLLM"
+        # reference reinforce_llm.py gamma: discounts turn rewards into the
+        # return when the env emits per-turn rewards (1.0 = plain sum)
+        self.gamma = float(gamma)
 
     def _calculate_advantages(self, rewards: torch.Tensor) -> torch.Tensor:
         G = self.group_size

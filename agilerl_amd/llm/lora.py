@@ -184,10 +184,34 @@ def apply_lora(
 ) -> nn.Module:
     """Wrap every target nn.Linear in the model with LoraLinear (in place)."""
     targets = set(config.target_modules)
-    for parent_name, parent in list(model.named_modules()):
-        for child_name, child in list(parent.named_children()):
-            if isinstance(child, nn.Linear) and child_name in targets:
-                setattr(parent, child_name, LoraLinear(child, config, adapters))
+
+    def _wrap(target_names) -> int:
+        wrapped = 0
+        for parent_name, parent in list(model.named_modules()):
+            for child_name, child in list(parent.named_children()):
+                if isinstance(child, nn.Linear) and child_name in target_names:
+                    setattr(parent, child_name, LoraLinear(child, config, adapters))
+                    wrapped += 1
+        return wrapped
+
+    if _wrap(targets) == 0:
+        # reference configs for wrapped/multimodal models spell targets as
+        # `q_proj.linear`; retry with the leading segment against plain
+        # decoder layouts before giving up
+        stripped = {t.split(".", 1)[0] for t in targets}
+        if stripped != targets and _wrap(stripped) > 0:
+            import warnings
+
+            warnings.warn(
+                f"LoRA target_modules {sorted(targets)} matched nothing; "
+                f"matched the base names {sorted(stripped)} instead",
+                RuntimeWarning,
+            )
+        else:
+            raise ValueError(
+                f"LoRA target_modules {sorted(targets)} matched no nn.Linear "
+                "in the model — adapters would be empty"
+            )
     mark_only_adapter_trainable(model)
     return model
 

@@ -61,7 +61,19 @@ def run_llm_workload(trainer, workload: str):
     # to the local constructor surface with an explicit warning so
     # reference YAMLs construct instead of TypeError-ing
     batch_override = hps.pop("batch_size", None)
-    accepted = set(inspect.signature(algo_cls.__init__).parameters)
+    sig = inspect.signature(algo_cls.__init__)
+    has_var_kw = any(p.kind is inspect.Parameter.VAR_KEYWORD
+                     for p in sig.parameters.values())
+    if has_var_kw:
+        # subclasses like PPOLLM/ReinforceLLM forward **kwargs to GRPO;
+        # filter against the base surface instead of dropping everything
+        from ...algorithms.llm.grpo import GRPO
+
+        accepted = set(inspect.signature(GRPO.__init__).parameters)
+        accepted |= set(sig.parameters)
+        accepted |= {"vf_coef", "value_head_lr", "gamma", "gae_lambda"}
+    else:
+        accepted = set(sig.parameters)
     dropped = sorted(k for k in hps if k not in accepted)
     for k in dropped:
         hps.pop(k)
@@ -366,6 +378,20 @@ def _multiturn_env(spec, agent):
         mod, fn = factory_path.rsplit(".", 1)
         factory = getattr(importlib.import_module(mod), fn)
     elif "GuessTheNumber" in str(name):
+        factory = lambda: TokenGuessEnv(vocab_size=vocab)  # noqa: E731
+    elif str(name).startswith("game:"):
+        # reference GEM game ids (Sudoku, Wordle, ...) aren't shipped
+        # offline; stand in with the first-party guess game so the config
+        # trains, and say so loudly
+        import warnings
+
+        warnings.warn(
+            f"multiturn game {name!r} is not available offline; using the "
+            "first-party TokenGuessEnv stand-in (same multi-turn "
+            "token/reward contract). Set environment.env_kwargs."
+            "env_factory for a custom game.",
+            RuntimeWarning,
+        )
         factory = lambda: TokenGuessEnv(vocab_size=vocab)  # noqa: E731
     else:
         raise ValueError(

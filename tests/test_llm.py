@@ -1079,7 +1079,7 @@ class TestPpoLlmSamplingIs:
 class TestReferenceLlmManifestFields:
     @pytest.mark.parametrize("cfg", [
         "grpo.yaml", "gspo.yaml", "cispo.yaml", "ppo_llm.yaml",
-        "reinforce_llm.yaml",
+        "reinforce_llm.yaml", "ppo_llm_quant_bench.yaml",
     ])
     def test_reference_llm_yaml_constructs_with_local_overrides(self, tmp_path, cfg):
         """The actual reference LLM-finetuning YAMLs (vLLM-era fields,
@@ -1150,18 +1150,24 @@ class TestReferenceLlmManifestFields:
 
 
 class TestReferenceMultiturnManifest:
-    def test_reference_grpo_multiturn_yaml_runs(self):
-        """The actual reference grpo_multiturn.yaml trains here: env_type
-        routing to the multi-turn loop, GuessTheNumber env mapping, and the
-        NETWORK-section lora_config (lora_r spelling) — only the hub model
-        is swapped for a local random-init config."""
+    @pytest.mark.parametrize("cfg", [
+        "grpo_multiturn.yaml", "cispo_quant_bench.yaml",
+        "reinforce_quant_bench.yaml", "cispo_gemma4_group5.yaml",
+        "cispo_quant_bench_qwen.yaml",
+    ])
+    def test_reference_multiturn_yaml_runs(self, cfg):
+        """The reference multiturn YAMLs train here: env_type routing to the
+        multi-turn loop, game-env mapping (GuessTheNumber; unavailable GEM
+        games warn and use the first-party stand-in), NETWORK-section
+        lora_config, quantization fields warn-and-ignore — only the hub
+        model is swapped for a local random-init config."""
         import os
         import yaml as _yaml
 
         from agilerl_amd.models.manifest import TrainingManifest
         from agilerl_amd.training.trainer import LocalTrainer
 
-        ref = "/root/reference/configs/training/llm_finetuning/grpo_multiturn.yaml"
+        ref = f"/root/reference/configs/training/llm_finetuning/{cfg}"
         if not os.path.exists(ref):
             pytest.skip("reference configs absent")
         doc = _yaml.safe_load(open(ref))
@@ -1180,15 +1186,12 @@ class TestReferenceMultiturnManifest:
         doc["algorithm"] = {"name": algo["name"], "hyperparameters": hp}
         doc["environment"]["data_batch_size"] = 2
         doc["environment"]["group_size"] = 2
+        doc["environment"]["max_turns"] = 2  # budget knob (quant configs use 50)
         doc["training"].update({"max_steps": 2, "pop_size": 2, "evo_steps": 1})
         manifest = TrainingManifest.model_validate(doc)
         trainer = LocalTrainer(manifest, device="cpu")
         results = trainer.train()
         assert results is not None
-        # network-section lora config (lora_r: 16) reached the agents
-        agent = trainer.__dict__.get("_population", None)
-        # indirect check: the run completed through the multiturn loop with
-        # the reference's mutation/selection sections applied
 
 
 class TestGrpoReferenceSurface(TestGrpoParityDeepening):
