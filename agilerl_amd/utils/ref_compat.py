@@ -172,15 +172,26 @@ def is_reference_layout(ckpt: Dict[str, Any]) -> bool:
 def _filter_init_dict(cls, init: Dict[str, Any]) -> Tuple[Dict[str, Any], List[str]]:
     import inspect
 
-    params = inspect.signature(cls.__init__).parameters
-    accepts_kwargs = any(
-        p.kind == inspect.Parameter.VAR_KEYWORD for p in params.values()
-    )
-    if accepts_kwargs:
-        return dict(init), []
+    # union of NAMED parameters across the MRO: classes taking **kwargs
+    # forward them upward, so the accepted surface is every named param of
+    # every ancestor __init__ — unknown reference fields still drop instead
+    # of reaching a strict base constructor
+    named = set()
+    for klass in cls.__mro__:
+        init_fn = klass.__dict__.get("__init__")
+        if init_fn is None:
+            continue
+        try:
+            for p in inspect.signature(init_fn).parameters.values():
+                if p.kind in (inspect.Parameter.POSITIONAL_OR_KEYWORD,
+                              inspect.Parameter.KEYWORD_ONLY):
+                    named.add(p.name)
+        except (TypeError, ValueError):
+            continue
+    named.discard("self")
     kept, dropped = {}, []
     for k, v in init.items():
-        if k in params:
+        if k in named:
             kept[k] = _normalize_value(v)
         else:
             dropped.append(k)
